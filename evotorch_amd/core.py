@@ -1,0 +1,1163 @@
+"""Core runtime: Problem, SolutionBatch, Solution, ProblemBoundEvaluator.
+
+MI355X-native re-design of the reference's `core.py`
+(/root/reference/src/evotorch/core.py:365-5256). The major architectural
+departure: the reference parallelizes evaluation with Ray actors and an
+object store (core.py:115-348, 1977-2131); here the parallel substrate is
+SPMD — one process per GPU joined by RCCL over xGMI through
+`evotorch_amd.parallel.Comm`. A Problem may be attached to a Comm, after
+which `evaluate()` shards the population rows across ranks and all-gathers
+the fitnesses (P1 in SURVEY.md §2.8), and `sample_and_compute_gradients()`
+computes rank-local ES gradients merged by a single all-reduce (P2).
+
+Storage layout matches the reference's contract (verified by the aliasing
+tests): a SolutionBatch owns a contiguous 2-D `values` tensor of shape
+(popsize, solution_length) and a 2-D `evals` tensor of shape
+(popsize, num_objectives + eval_data_length) where NaN marks "not yet
+evaluated"; basic slicing returns shared-memory views.
+"""
+
+import logging
+import math
+from typing import Any, Callable, Iterable, List, Optional, Union
+
+import numpy as np
+import torch
+
+from .utils import (
+    Device,
+    DType,
+    Hook,
+    ObjectArray,
+    ReadOnlyTensor,
+    Serializable,
+    TensorMakerMixin,
+    as_read_only_tensor,
+    clone as _clone,
+    deep_clone,
+    is_dtype_bool,
+    is_dtype_object,
+    is_dtype_real,
+    is_sequence,
+    to_torch_dtype,
+)
+from .utils.misc import ensure_tensor_length_and_dtype, split_workload
+
+_logger = logging.getLogger(__name__)
+
+__all__ = ["Problem", "SolutionBatch", "SolutionBatchPieces", "Solution", "ProblemBoundEvaluator"]
+
+ObjectiveSense = Union[str, Iterable[str]]
+BoundsPair = Any
+
+
+def _normalize_sense(objective_sense: ObjectiveSense) -> List[str]:
+    if isinstance(objective_sense, str):
+        senses = [objective_sense]
+    else:
+        senses = list(objective_sense)
+    for s in senses:
+        if s not in ("min", "max"):
+            raise ValueError(f"Objective sense must be 'min' or 'max', got {s!r}")
+    return senses
+
+
+# ============================================================================
+# Pareto machinery (K7 in SURVEY.md §2.9)
+# ============================================================================
+
+
+def _domination_matrix(utils: torch.Tensor) -> torch.Tensor:
+    """Boolean (N, N) matrix: entry [i, j] is True iff solution i dominates
+    solution j. `utils` is (N, M) with *higher is better* for every column
+    (senses already folded in)."""
+    a = utils.unsqueeze(1)  # (N, 1, M)
+    b = utils.unsqueeze(0)  # (1, N, M)
+    ge = (a >= b).all(dim=-1)
+    gt = (a > b).any(dim=-1)
+    return ge & gt
+
+
+def _compute_pareto_ranks(utils: torch.Tensor, crowdsort: bool = True):
+    """Non-dominated ranking by iterative front peeling over domination
+    counts — formulated as masked reductions so the only host sync is the
+    loop-termination check (SURVEY.md §7 hard-parts note on K7).
+
+    Returns (ranks, crowd) where ranks[i] is the index of i's pareto front
+    (0 = best front) and crowd[i] is the crowding distance (or None)."""
+    n = utils.shape[0]
+    dom = _domination_matrix(utils)
+    dom_count = dom.sum(dim=0).to(torch.int64)  # how many dominate me
+    ranks = torch.full((n,), -1, dtype=torch.int64, device=utils.device)
+    assigned = torch.zeros(n, dtype=torch.bool, device=utils.device)
+    front_index = 0
+    domf = dom.to(torch.int64)
+    while not bool(assigned.all()):
+        current = (dom_count == 0) & (~assigned)
+        if not bool(current.any()):
+            # numerical corner: break ties by assigning the rest to one front
+            ranks[~assigned] = front_index
+            break
+        ranks[current] = front_index
+        assigned |= current
+        # remove the front's domination contributions
+        dom_count = dom_count - domf[current].sum(dim=0)
+        dom_count[assigned] = -1
+        front_index += 1
+    crowd = _crowding_distances(utils, ranks) if crowdsort else None
+    return ranks, crowd
+
+
+def _crowding_distances(utils: torch.Tensor, ranks: torch.Tensor) -> torch.Tensor:
+    """NSGA-II crowding distance, computed per front (reference
+    core.py:3432). Boundary solutions get +inf."""
+    n, m = utils.shape
+    crowd = torch.zeros(n, dtype=utils.dtype, device=utils.device)
+    for front in torch.unique(ranks):
+        idx = torch.nonzero(ranks == front, as_tuple=True)[0]
+        if len(idx) <= 2:
+            crowd[idx] = float("inf")
+            continue
+        sub = utils[idx]
+        for j in range(m):
+            order = sub[:, j].argsort()
+            sorted_vals = sub[order, j]
+            span = sorted_vals[-1] - sorted_vals[0]
+            if float(span) == 0.0:
+                continue
+            contrib = torch.zeros(len(idx), dtype=utils.dtype, device=utils.device)
+            contrib[order[0]] = float("inf")
+            contrib[order[-1]] = float("inf")
+            contrib[order[1:-1]] = (sorted_vals[2:] - sorted_vals[:-2]) / span
+            crowd[idx] = crowd[idx] + contrib
+    return crowd
+
+
+# ============================================================================
+# SolutionBatch
+# ============================================================================
+
+
+class SolutionBatch(Serializable):
+    """A population: a 2-D values tensor plus a 2-D evals tensor.
+
+    Mirrors the reference's container semantics
+    (/root/reference/src/evotorch/core.py:3590-4601): NaN rows of `evals`
+    mean "unevaluated"; slicing with a contiguous slice shares memory with
+    the parent batch; `split`/`concat`/`take`/`take_best` reshape
+    populations for GA-style algorithms.
+    """
+
+    def __init__(
+        self,
+        problem: Optional["Problem"] = None,
+        popsize: Optional[int] = None,
+        *,
+        device: Optional[Device] = None,
+        slice_of: Optional[tuple] = None,
+        like: Optional["SolutionBatch"] = None,
+        merging_of: Optional[Iterable["SolutionBatch"]] = None,
+        empty: bool = False,
+    ):
+        self._num_objs: int
+        self._eval_data_length: int
+        self._senses: List[str]
+
+        if slice_of is not None:
+            source, sl = slice_of
+            if isinstance(sl, slice):
+                start, stop, step = sl.indices(len(source))
+                if step != 1:
+                    raise ValueError("SolutionBatch slices must be contiguous (step 1); use take() for fancy indexing")
+            else:
+                start, stop = sl
+            self._values = source._values[start:stop]
+            self._evals = source._evals[start:stop]
+            self._num_objs = source._num_objs
+            self._eval_data_length = source._eval_data_length
+            self._senses = source._senses
+            return
+
+        if merging_of is not None:
+            batches = list(merging_of)
+            if len(batches) == 0:
+                raise ValueError("Cannot merge zero batches")
+            first = batches[0]
+            self._num_objs = first._num_objs
+            self._eval_data_length = first._eval_data_length
+            self._senses = first._senses
+            if isinstance(first._values, ObjectArray):
+                total = sum(len(b) for b in batches)
+                self._values = ObjectArray(total)
+                i = 0
+                for b in batches:
+                    for j in range(len(b)):
+                        self._values[i] = b._values[j]
+                        i += 1
+            else:
+                self._values = torch.cat([b._values for b in batches], dim=0)
+            self._evals = torch.cat([b._evals for b in batches], dim=0)
+            return
+
+        if like is not None:
+            n = popsize if popsize is not None else len(like)
+            self._num_objs = like._num_objs
+            self._eval_data_length = like._eval_data_length
+            self._senses = like._senses
+            if isinstance(like._values, ObjectArray):
+                self._values = ObjectArray(n)
+            else:
+                self._values = torch.empty((n, like._values.shape[1]), dtype=like._values.dtype, device=device if device is not None else like._values.device)
+            self._evals = torch.full(
+                (n, self._num_objs + self._eval_data_length), float("nan"), dtype=like._evals.dtype, device=device if device is not None else like._evals.device
+            )
+            return
+
+        if problem is None:
+            raise ValueError("SolutionBatch requires one of: problem, slice_of, like, merging_of")
+        if popsize is None:
+            popsize = 1
+        popsize = int(popsize)
+        self._num_objs = len(problem.senses)
+        self._eval_data_length = problem.eval_data_length
+        self._senses = list(problem.senses)
+        dev = device if device is not None else problem.device
+        if problem.dtype_is_object:
+            self._values = ObjectArray(popsize)
+        else:
+            self._values = torch.empty((popsize, problem.solution_length), dtype=problem.dtype, device=dev)
+        self._evals = torch.full((popsize, self._num_objs + self._eval_data_length), float("nan"), dtype=problem.eval_dtype, device=dev)
+        if not empty:
+            problem._fill(self._values)
+
+    # -- basic properties ----------------------------------------------------
+
+    def __len__(self) -> int:
+        return len(self._values) if isinstance(self._values, ObjectArray) else self._values.shape[0]
+
+    @property
+    def values_shape(self) -> tuple:
+        return tuple(self._values.shape)
+
+    @property
+    def eval_shape(self) -> tuple:
+        return tuple(self._evals.shape)
+
+    @property
+    def solution_length(self) -> Optional[int]:
+        if isinstance(self._values, ObjectArray):
+            return None
+        return self._values.shape[1]
+
+    @property
+    def objective_sense(self) -> Union[str, List[str]]:
+        return self._senses[0] if len(self._senses) == 1 else list(self._senses)
+
+    @property
+    def senses(self) -> List[str]:
+        return list(self._senses)
+
+    @property
+    def num_objectives(self) -> int:
+        return self._num_objs
+
+    @property
+    def eval_data_length(self) -> int:
+        return self._eval_data_length
+
+    @property
+    def device(self):
+        return self._evals.device
+
+    @property
+    def dtype(self):
+        return object if isinstance(self._values, ObjectArray) else self._values.dtype
+
+    @property
+    def eval_dtype(self):
+        return self._evals.dtype
+
+    @property
+    def is_multi_objective(self) -> bool:
+        return self._num_objs > 1
+
+    # -- value/eval access ---------------------------------------------------
+
+    def access_values(self, *, keep_evals: bool = False):
+        """Mutable access to the values tensor; evals are reset to NaN
+        unless `keep_evals=True` (reference core.py:3733)."""
+        if not keep_evals:
+            self.forget_evals()
+        return self._values
+
+    def access_evals(self, obj_index: Optional[int] = None) -> torch.Tensor:
+        if obj_index is None:
+            return self._evals
+        return self._evals[:, self._normalize_obj_index(obj_index)]
+
+    @property
+    def values(self):
+        if isinstance(self._values, ObjectArray):
+            return self._values.get_read_only_view()
+        return as_read_only_tensor(self._values)
+
+    @property
+    def evals(self) -> ReadOnlyTensor:
+        return as_read_only_tensor(self._evals)
+
+    @property
+    def unsafe_values(self):
+        """The raw (mutable) values tensor, WITHOUT forgetting the evals."""
+        return self._values
+
+    @property
+    def unsafe_evals(self) -> torch.Tensor:
+        return self._evals
+
+    def forget_evals(self, *, solutions: Optional[Union[slice, torch.Tensor]] = None):
+        if solutions is None:
+            self._evals.fill_(float("nan"))
+        else:
+            self._evals[solutions] = float("nan")
+
+    def set_values(self, values, *, solutions: Optional[Union[slice, torch.Tensor]] = None):
+        if solutions is None:
+            solutions = slice(None)
+        if isinstance(self._values, ObjectArray):
+            self._values[solutions] = values
+        else:
+            self._values[solutions] = torch.as_tensor(values, dtype=self._values.dtype, device=self._values.device)
+        self.forget_evals(solutions=solutions)
+
+    def set_evals(self, evals: torch.Tensor, eval_data: Optional[torch.Tensor] = None, *, solutions: Optional[Union[slice, torch.Tensor]] = None):
+        if solutions is None:
+            solutions = slice(None)
+        evals = torch.as_tensor(evals, dtype=self._evals.dtype, device=self._evals.device)
+        if evals.ndim == 1:
+            if self._num_objs != 1:
+                evals = evals.reshape(-1, self._num_objs)
+            else:
+                evals = evals.reshape(-1, 1)
+        if evals.shape[1] == self._num_objs + self._eval_data_length and eval_data is None:
+            self._evals[solutions] = evals
+            return
+        self._evals[solutions, : self._num_objs] = evals
+        if eval_data is not None:
+            eval_data = torch.as_tensor(eval_data, dtype=self._evals.dtype, device=self._evals.device)
+            self._evals[solutions, self._num_objs :] = eval_data
+
+    @property
+    def evals_are_ready(self) -> bool:
+        return not bool(torch.isnan(self._evals[:, : self._num_objs]).any())
+
+    # -- indexing ------------------------------------------------------------
+
+    def __getitem__(self, i):
+        if isinstance(i, slice):
+            return SolutionBatch(slice_of=(self, i))
+        if isinstance(i, (torch.Tensor, np.ndarray, list)):
+            return self.take(i)
+        return Solution(parent=self, index=int(i))
+
+    def __iter__(self):
+        for i in range(len(self)):
+            yield self[i]
+
+    def take(self, indices) -> "SolutionBatch":
+        """New batch whose rows are copies of the given rows."""
+        if isinstance(indices, torch.Tensor):
+            idx = indices.to(dtype=torch.int64)
+        else:
+            idx = torch.as_tensor(np.asarray(indices), dtype=torch.int64)
+        result = SolutionBatch(like=self, popsize=len(idx))
+        if isinstance(self._values, ObjectArray):
+            for out_i, src_i in enumerate(idx.tolist()):
+                result._values[out_i] = self._values[src_i]
+        else:
+            idx_dev = idx.to(self._values.device)
+            result._values.copy_(self._values[idx_dev])
+        result._evals.copy_(self._evals[idx.to(self._evals.device)])
+        return result
+
+    def take_best(self, n: Optional[int] = None, *, obj_index: Optional[int] = None) -> Union["SolutionBatch", "Solution"]:
+        """Best n solutions as a new batch (pareto-based for multi-objective
+        when obj_index is None); with n omitted, the single best Solution."""
+        if n is None:
+            indices = self.argsort(obj_index=obj_index)
+            return self[int(indices[0])]
+        indices = self.argsort(obj_index=obj_index)[:n]
+        return self.take(indices)
+
+    def split(self, num_pieces: Optional[int] = None, *, max_size: Optional[int] = None) -> "SolutionBatchPieces":
+        return SolutionBatchPieces(self, num_pieces=num_pieces, max_size=max_size)
+
+    @staticmethod
+    def cat(batches: Iterable["SolutionBatch"]) -> "SolutionBatch":
+        return SolutionBatch(merging_of=batches)
+
+    def concat(self, other: Union["SolutionBatch", Iterable["SolutionBatch"]]) -> "SolutionBatch":
+        others = [other] if isinstance(other, SolutionBatch) else list(other)
+        return SolutionBatch(merging_of=[self] + others)
+
+    def to(self, device: Device) -> "SolutionBatch":
+        device = torch.device(device)
+        if (not isinstance(self._values, ObjectArray)) and self._values.device == device and self._evals.device == device:
+            return self
+        result = SolutionBatch(like=self, popsize=len(self), device=device)
+        if isinstance(self._values, ObjectArray):
+            result._values = self._values.clone()
+        else:
+            result._values.copy_(self._values.to(device))
+        result._evals.copy_(self._evals.to(device))
+        return result
+
+    # -- sorting & utilities ---------------------------------------------------
+
+    def _normalize_obj_index(self, obj_index: Optional[int]) -> int:
+        if obj_index is None:
+            if self._num_objs != 1:
+                raise ValueError("obj_index must be given for a multi-objective batch")
+            return 0
+        obj_index = int(obj_index)
+        if obj_index < 0:
+            obj_index += self._num_objs
+        if not (0 <= obj_index < self._num_objs):
+            raise IndexError(f"Invalid obj_index {obj_index}")
+        return obj_index
+
+    def _utils_for_sorting(self) -> torch.Tensor:
+        """Evals folded so higher is better for every objective."""
+        utils = self._evals[:, : self._num_objs].clone()
+        for j, sense in enumerate(self._senses):
+            if sense == "min":
+                utils[:, j] = -utils[:, j]
+        return utils
+
+    def argsort(self, obj_index: Optional[int] = None) -> torch.Tensor:
+        """Indices from best to worst. For a multi-objective batch without
+        obj_index, sorts by (pareto rank, -crowding distance)."""
+        if self._num_objs > 1 and obj_index is None:
+            ranks, crowd = self.compute_pareto_ranks(crowdsort=True)
+            # lexicographic (pareto rank asc, crowding distance desc) via
+            # integer keys: position-in-crowd-order breaks ties within a front
+            n = ranks.shape[0]
+            crowd_order = torch.nan_to_num(crowd.to(torch.float64), posinf=1e300).argsort(descending=True)
+            crowd_pos = torch.empty_like(crowd_order)
+            crowd_pos.scatter_(0, crowd_order, torch.arange(n, device=ranks.device))
+            key = ranks * (n + 1) + crowd_pos
+            return key.argsort()
+        j = self._normalize_obj_index(obj_index)
+        evals = self._evals[:, j]
+        descending = self._senses[j] == "max"
+        return evals.argsort(descending=descending)
+
+    def argbest(self, obj_index: Optional[int] = None) -> torch.Tensor:
+        j = self._normalize_obj_index(obj_index)
+        evals = self._evals[:, j]
+        return evals.argmax() if self._senses[j] == "max" else evals.argmin()
+
+    def argworst(self, obj_index: Optional[int] = None) -> torch.Tensor:
+        j = self._normalize_obj_index(obj_index)
+        evals = self._evals[:, j]
+        return evals.argmin() if self._senses[j] == "max" else evals.argmax()
+
+    def compute_pareto_ranks(self, crowdsort: bool = True):
+        utils = self._utils_for_sorting()
+        return _compute_pareto_ranks(utils, crowdsort=crowdsort)
+
+    def arg_pareto_sort(self, crowdsort: bool = True):
+        """List of fronts (each a tensor of indices), best front first
+        (reference core.py:3554)."""
+        ranks, crowd = self.compute_pareto_ranks(crowdsort=crowdsort)
+        fronts = []
+        for front in torch.unique(ranks, sorted=True):
+            idx = torch.nonzero(ranks == front, as_tuple=True)[0]
+            if crowdsort and crowd is not None and len(idx) > 1:
+                idx = idx[crowd[idx].argsort(descending=True)]
+            fronts.append(idx)
+        return fronts
+
+    def utility(self, obj_index: Optional[int] = None, *, ranking_method: Optional[str] = None) -> torch.Tensor:
+        """Fitness-shaped utilities for one objective (reference
+        core.py:4207); higher utility = better regardless of sense."""
+        from .utils import ranking
+
+        j = self._normalize_obj_index(obj_index)
+        return ranking.rank(self._evals[:, j], ranking_method or "raw", higher_is_better=(self._senses[j] == "max"))
+
+    def utils(self, *, ranking_method: Optional[str] = None) -> torch.Tensor:
+        """Utilities for all objectives, shape (popsize, num_objectives)."""
+        cols = [self.utility(j, ranking_method=ranking_method) for j in range(self._num_objs)]
+        return torch.stack(cols, dim=-1)
+
+    # -- cloning / serialization ----------------------------------------------
+
+    def _get_cloned_state(self, *, memo: dict) -> dict:
+        return {
+            "_values": self._values.clone() if not isinstance(self._values, ObjectArray) else self._values.clone(),
+            "_evals": self._evals.clone(),
+            "_num_objs": self._num_objs,
+            "_eval_data_length": self._eval_data_length,
+            "_senses": list(self._senses),
+        }
+
+    def __repr__(self) -> str:
+        return f"<SolutionBatch popsize={len(self)} length={self.solution_length} device={self.device}>"
+
+
+class SolutionBatchPieces:
+    """Lazy view of a batch split into contiguous sub-batches (reference
+    core.py:4603-4729)."""
+
+    def __init__(self, batch: SolutionBatch, *, num_pieces: Optional[int] = None, max_size: Optional[int] = None):
+        self._batch = batch
+        n = len(batch)
+        if (num_pieces is None) == (max_size is None):
+            raise ValueError("Provide exactly one of num_pieces, max_size")
+        if max_size is not None:
+            num_pieces = max(1, math.ceil(n / int(max_size)))
+        sizes = split_workload(n, int(num_pieces))
+        self._ranges = []
+        start = 0
+        for s in sizes:
+            self._ranges.append((start, start + s))
+            start += s
+
+    def __len__(self) -> int:
+        return len(self._ranges)
+
+    def __getitem__(self, i: int) -> SolutionBatch:
+        start, stop = self._ranges[i]
+        return SolutionBatch(slice_of=(self._batch, (start, stop)))
+
+    def indices_of(self, i: int) -> tuple:
+        return self._ranges[i]
+
+    def __iter__(self):
+        for i in range(len(self)):
+            yield self[i]
+
+
+# ============================================================================
+# Solution
+# ============================================================================
+
+
+class Solution(Serializable):
+    """A single row view into a parent SolutionBatch (reference
+    core.py:4742-5107)."""
+
+    def __init__(self, parent: SolutionBatch, index: int):
+        if index < 0:
+            index += len(parent)
+        if not (0 <= index < len(parent)):
+            raise IndexError(f"Solution index {index} out of range")
+        self._batch = parent
+        self._index = index
+
+    @property
+    def values(self):
+        v = self._batch._values
+        if isinstance(v, ObjectArray):
+            return v[self._index]
+        return as_read_only_tensor(v[self._index])
+
+    @property
+    def evals(self) -> ReadOnlyTensor:
+        return as_read_only_tensor(self._batch._evals[self._index])
+
+    @property
+    def evaluation(self) -> torch.Tensor:
+        return as_read_only_tensor(self._batch._evals[self._index, 0])
+
+    def access_values(self, *, keep_evals: bool = False):
+        if not keep_evals:
+            self._batch.forget_evals(solutions=self._index)
+        v = self._batch._values
+        if isinstance(v, ObjectArray):
+            return v[self._index]
+        return v[self._index]
+
+    def set_values(self, values):
+        self._batch.set_values(values, solutions=self._index)
+
+    def set_evals(self, evals, eval_data=None):
+        evals = torch.as_tensor(evals, dtype=self._batch._evals.dtype, device=self._batch._evals.device).reshape(-1)
+        no = self._batch._num_objs
+        if len(evals) == no + self._batch._eval_data_length:
+            self._batch._evals[self._index] = evals
+            return
+        self._batch._evals[self._index, :no] = evals
+        if eval_data is not None:
+            self._batch._evals[self._index, no:] = torch.as_tensor(eval_data, dtype=self._batch._evals.dtype, device=self._batch._evals.device)
+
+    def set_evaluation(self, evaluation: float, eval_data=None):
+        self._batch._evals[self._index, 0] = float(evaluation)
+        if eval_data is not None:
+            self._batch._evals[self._index, self._batch._num_objs :] = torch.as_tensor(eval_data, dtype=self._batch._evals.dtype, device=self._batch._evals.device)
+
+    @property
+    def is_evaluated(self) -> bool:
+        return not bool(torch.isnan(self._batch._evals[self._index, : self._batch._num_objs]).any())
+
+    @property
+    def objective_sense(self):
+        return self._batch.objective_sense
+
+    @property
+    def senses(self) -> List[str]:
+        return self._batch.senses
+
+    @property
+    def dtype(self):
+        return self._batch.dtype
+
+    @property
+    def device(self):
+        return self._batch.device
+
+    def to_batch(self) -> SolutionBatch:
+        """1-row shared-memory view batch."""
+        return SolutionBatch(slice_of=(self._batch, (self._index, self._index + 1)))
+
+    def clone(self, *, memo: Optional[dict] = None) -> "Solution":
+        batch = self.to_batch()
+        new_batch = SolutionBatch(like=batch, popsize=1)
+        if isinstance(batch._values, ObjectArray):
+            new_batch._values[0] = batch._values[0]
+        else:
+            new_batch._values.copy_(batch._values)
+        new_batch._evals.copy_(batch._evals)
+        return Solution(parent=new_batch, index=0)
+
+    def _get_cloned_state(self, *, memo: dict) -> dict:
+        c = self.clone()
+        return {"_batch": c._batch, "_index": 0}
+
+    def __len__(self) -> int:
+        v = self._batch._values
+        return 0 if isinstance(v, ObjectArray) else v.shape[1]
+
+    def __getitem__(self, i):
+        return self.values[i]
+
+    def __repr__(self) -> str:
+        ev = "evaluated" if self.is_evaluated else "not evaluated"
+        return f"<Solution index={self._index} ({ev})>"
+
+
+# ============================================================================
+# Problem
+# ============================================================================
+
+
+class Problem(TensorMakerMixin, Serializable):
+    """The central object: objective sense(s), solution geometry, dtypes,
+    device, RNG, bounds, evaluation dispatch, and (through an attached
+    `evotorch_amd.parallel.Comm`) population-parallel evaluation and
+    distributed ES gradients.
+
+    Reference parity: /root/reference/src/evotorch/core.py:365-3415. The Ray
+    actor-pool constructor knobs (`num_actors`, `num_gpus_per_actor`, ...)
+    are intentionally absent; SPMD rank topology comes from the launcher
+    (torchrun) and `evotorch_amd.parallel`.
+    """
+
+    def __init__(
+        self,
+        objective_sense: ObjectiveSense,
+        objective_func: Optional[Callable] = None,
+        *,
+        initial_bounds: Optional[BoundsPair] = None,
+        bounds: Optional[BoundsPair] = None,
+        solution_length: Optional[int] = None,
+        dtype: Optional[DType] = None,
+        eval_dtype: Optional[DType] = None,
+        device: Optional[Device] = None,
+        eval_data_length: int = 0,
+        seed: Optional[int] = None,
+        store_solution_stats: Optional[bool] = None,
+        vectorized: Optional[bool] = None,
+    ):
+        self._senses = _normalize_sense(objective_sense)
+        self._objective_func = objective_func
+        self._vectorized = bool(getattr(objective_func, "__evotorch_vectorized__", False)) if vectorized is None else bool(vectorized)
+        fn_device = getattr(objective_func, "__evotorch_device__", None)
+        self._fitness_device = torch.device(fn_device) if fn_device is not None else None
+
+        self._dtype = to_torch_dtype(dtype) if (dtype is not None and not is_dtype_object(dtype)) else (object if dtype is not None else torch.float32)
+        if eval_dtype is not None:
+            self._eval_dtype = to_torch_dtype(eval_dtype)
+        else:
+            self._eval_dtype = self._dtype if (self._dtype != object and to_torch_dtype(self._dtype).is_floating_point) else torch.float32
+        self._device = torch.device(device) if device is not None else torch.device("cpu")
+        if self.dtype_is_object and self._device.type != "cpu":
+            raise ValueError("object-dtype problems must live on cpu")
+
+        self._solution_length = None if solution_length is None else int(solution_length)
+        if self._solution_length is None and not self.dtype_is_object:
+            if initial_bounds is None and bounds is None:
+                raise ValueError("Provide solution_length (and bounds or initial_bounds) for numeric problems")
+        self._eval_data_length = int(eval_data_length)
+
+        # Bounds
+        self._initial_lower_bounds = self._initial_upper_bounds = None
+        self._lower_bounds = self._upper_bounds = None
+        if bounds is not None:
+            self._lower_bounds, self._upper_bounds = self._normalize_bounds(bounds)
+        if initial_bounds is not None:
+            self._initial_lower_bounds, self._initial_upper_bounds = self._normalize_bounds(initial_bounds)
+        elif bounds is not None:
+            self._initial_lower_bounds, self._initial_upper_bounds = self._lower_bounds, self._upper_bounds
+
+        # RNG
+        self._seed = seed
+        self._generator: Optional[torch.Generator] = None
+        if seed is not None:
+            self._generator = torch.Generator(device=self._device)
+            self._generator.manual_seed(int(seed))
+
+        # Solution stat tracking
+        if store_solution_stats is None:
+            store_solution_stats = self._device.type == "cpu"
+        self._store_solution_stats = bool(store_solution_stats)
+        self._best: Optional[List[Optional[Solution]]] = None
+        self._worst: Optional[List[Optional[Solution]]] = None
+        self._best_evals: Optional[torch.Tensor] = None
+        self._worst_evals: Optional[torch.Tensor] = None
+
+        # Hooks (reference core.py:2175-2238)
+        self._before_eval_hook = Hook()
+        self._after_eval_hook = Hook()
+        self._before_grad_hook = Hook()
+        self._after_grad_hook = Hook()
+
+        # Counters
+        self._after_eval_status: dict = {}
+        self._after_grad_status: dict = {}
+
+        # SPMD comm (attached lazily; replaces the reference's actor pool)
+        self._comm = None
+
+    # -- configuration properties ---------------------------------------------
+
+    def _normalize_bounds(self, bounds: BoundsPair):
+        if self.dtype_is_object:
+            raise ValueError("Bounds are not supported for object-dtype problems")
+        try:
+            lb, ub = bounds
+        except Exception:
+            raise ValueError(f"Bounds must be a pair (lb, ub); got {bounds!r}") from None
+        length = self._solution_length
+        lb = ensure_tensor_length_and_dtype(lb, length, self._dtype, about="lower bound", device=self._device)
+        ub = ensure_tensor_length_and_dtype(ub, length, self._dtype, about="upper bound", device=self._device)
+        return lb, ub
+
+    @property
+    def senses(self) -> List[str]:
+        return list(self._senses)
+
+    @property
+    def objective_sense(self) -> Union[str, List[str]]:
+        return self._senses[0] if len(self._senses) == 1 else list(self._senses)
+
+    @property
+    def is_multi_objective(self) -> bool:
+        return len(self._senses) > 1
+
+    @property
+    def solution_length(self) -> Optional[int]:
+        return self._solution_length
+
+    @property
+    def eval_data_length(self) -> int:
+        return self._eval_data_length
+
+    @property
+    def dtype(self):
+        return self._dtype
+
+    @property
+    def dtype_is_object(self) -> bool:
+        return self._dtype == object
+
+    @property
+    def eval_dtype(self) -> torch.dtype:
+        return self._eval_dtype
+
+    @property
+    def device(self) -> torch.device:
+        return self._device
+
+    @property
+    def aux_device(self) -> torch.device:
+        """The device for heavy fitness computation: the first visible
+        accelerator if there is one, else the problem's own device
+        (reference core.py:1656-1692)."""
+        if torch.cuda.is_available():
+            return torch.device("cuda", torch.cuda.current_device())
+        return self._device
+
+    @property
+    def generator(self) -> Optional[torch.Generator]:
+        return self._generator
+
+    def manual_seed(self, seed: Optional[int] = None):
+        if seed is None:
+            self._generator = None
+        else:
+            self._generator = torch.Generator(device=self._device)
+            self._generator.manual_seed(int(seed))
+        return self
+
+    @property
+    def initial_lower_bounds(self):
+        return self._initial_lower_bounds
+
+    @property
+    def initial_upper_bounds(self):
+        return self._initial_upper_bounds
+
+    @property
+    def lower_bounds(self):
+        return self._lower_bounds
+
+    @property
+    def upper_bounds(self):
+        return self._upper_bounds
+
+    @property
+    def before_eval_hook(self) -> Hook:
+        return self._before_eval_hook
+
+    @property
+    def after_eval_hook(self) -> Hook:
+        return self._after_eval_hook
+
+    @property
+    def before_grad_hook(self) -> Hook:
+        return self._before_grad_hook
+
+    @property
+    def after_grad_hook(self) -> Hook:
+        return self._after_grad_hook
+
+    @property
+    def status(self) -> dict:
+        return {**self._after_eval_status, **self._after_grad_status}
+
+    # -- SPMD comm -------------------------------------------------------------
+
+    def use_comm(self, comm) -> "Problem":
+        """Attach an `evotorch_amd.parallel.Comm`; subsequent `evaluate`
+        calls shard the population rows across its ranks and all-gather the
+        evals, and `sample_and_compute_gradients` runs in SPMD mode."""
+        self._comm = comm
+        return self
+
+    @property
+    def comm(self):
+        return self._comm
+
+    @property
+    def is_main(self) -> bool:
+        return self._comm is None or self._comm.rank == 0
+
+    # -- validation helpers ------------------------------------------------------
+
+    def ensure_numeric(self):
+        if self.dtype_is_object:
+            raise ValueError("This operation requires a numeric (non-object) dtype problem")
+
+    def ensure_single_objective(self):
+        if self.is_multi_objective:
+            raise ValueError("This operation requires a single-objective problem")
+
+    def ensure_unbounded(self):
+        if self._lower_bounds is not None or self._upper_bounds is not None:
+            raise ValueError("This operation requires an unbounded problem")
+
+    # -- population generation ---------------------------------------------------
+
+    def generate_values(self, num_solutions: int):
+        """Fresh decision-variable rows, shape (num_solutions, L). Uses
+        `_fill` (uniform within initial bounds by default; subclasses
+        override)."""
+        if self.dtype_is_object:
+            result = ObjectArray(num_solutions)
+        else:
+            result = torch.empty((int(num_solutions), self._solution_length), dtype=self._dtype, device=self._device)
+        self._fill(result)
+        return result
+
+    def _fill(self, values):
+        """Fill a pre-allocated values container with fresh solutions
+        (reference core.py:1874). Default: uniform in initial bounds."""
+        if self.dtype_is_object:
+            raise NotImplementedError("object-dtype problems must override _fill or generate_values")
+        if self._initial_lower_bounds is None:
+            raise RuntimeError("Problem has no initial_bounds; cannot generate random solutions. Override _fill().")
+        from .utils.misc import make_uniform
+
+        make_uniform(lb=self._initial_lower_bounds, ub=self._initial_upper_bounds, generator=self._generator, out=values)
+
+    def generate_batch(
+        self,
+        popsize: Optional[int] = None,
+        *,
+        empty: bool = False,
+        device: Optional[Device] = None,
+    ) -> SolutionBatch:
+        return SolutionBatch(self, popsize, device=device, empty=empty)
+
+    def populate(self, popsize: int) -> SolutionBatch:
+        return self.generate_batch(popsize)
+
+    # -- evaluation ---------------------------------------------------------------
+
+    def _evaluate_batch(self, batch: SolutionBatch):
+        """Evaluate every solution of `batch`, writing into its evals.
+        Default: vectorized objective_func if declared, else per-solution
+        `_evaluate` (reference core.py:2602-2621)."""
+        if self._vectorized and self._objective_func is not None:
+            self._evaluate_vectorized(batch, self._objective_func)
+            return
+        if self._objective_func is not None:
+            for sln in batch:
+                result = self._objective_func(sln.values)
+                self._write_solution_result(sln, result)
+            return
+        for sln in batch:
+            self._evaluate(sln)
+
+    def _evaluate_vectorized(self, batch: SolutionBatch, fn: Callable):
+        values = batch.access_values(keep_evals=True)
+        target_device = self._fitness_device
+        moved = values.to(target_device) if (target_device is not None and not isinstance(values, ObjectArray)) else values
+        result = fn(moved)
+        if isinstance(result, tuple):
+            evals, eval_data = result
+            batch.set_evals(torch.as_tensor(evals).to(batch.device), torch.as_tensor(eval_data).to(batch.device))
+        else:
+            batch.set_evals(torch.as_tensor(result).to(batch.device))
+
+    def _write_solution_result(self, sln: Solution, result):
+        if isinstance(result, tuple):
+            evals, eval_data = result
+            sln.set_evals(torch.as_tensor(evals, dtype=self._eval_dtype).reshape(-1), torch.as_tensor(eval_data, dtype=self._eval_dtype).reshape(-1))
+        elif isinstance(result, torch.Tensor) and result.ndim >= 1 and result.numel() == len(self._senses):
+            sln.set_evals(result.reshape(-1))
+        elif result is None:
+            if not sln.is_evaluated:
+                raise RuntimeError("Fitness function returned None but did not fill the solution's evals")
+        else:
+            sln.set_evaluation(float(result))
+
+    def _evaluate(self, solution: Solution):
+        """Per-solution evaluation; override in subclasses when no
+        objective_func is given (reference core.py:2613)."""
+        raise NotImplementedError("Either provide objective_func or override _evaluate/_evaluate_batch")
+
+    def evaluate(self, batch: Union[SolutionBatch, Solution]):
+        """Evaluate a batch. With an attached Comm and world_size > 1, each
+        rank evaluates a contiguous shard of the rows and the evals are
+        joined with one all_gather over xGMI (P1 in SURVEY.md §2.8)."""
+        if isinstance(batch, Solution):
+            batch = batch.to_batch()
+        self._before_eval_hook(batch)
+        comm = self._comm
+        if comm is not None and comm.world_size > 1 and len(batch) >= comm.world_size:
+            self._evaluate_sharded(batch, comm)
+        else:
+            self._evaluate_batch(batch)
+        if self._store_solution_stats:
+            self._update_solution_stats(batch)
+        self._after_eval_status = self._after_eval_hook.accumulate_dict(batch)
+
+    def _evaluate_sharded(self, batch: SolutionBatch, comm):
+        pieces = batch.split(comm.world_size)
+        my_piece = pieces[comm.rank]
+        self._evaluate_batch(my_piece)
+        comm.all_gather_rows(batch.unsafe_evals, [pieces.indices_of(i) for i in range(len(pieces))])
+
+    def _update_solution_stats(self, batch: SolutionBatch):
+        nobj = len(self._senses)
+        if self._best is None:
+            self._best = [None] * nobj
+            self._worst = [None] * nobj
+        for j, sense in enumerate(self._senses):
+            evals = batch._evals[:, j]
+            if bool(torch.isnan(evals).all()):
+                continue
+            bi = int(batch.argbest(j))
+            wi = int(batch.argworst(j))
+            cand_best = batch[bi]
+            cand_worst = batch[wi]
+            def better(a: float, b: float) -> bool:
+                return a > b if sense == "max" else a < b
+            if self._best[j] is None or better(float(cand_best.evals[j]), float(self._best[j].evals[j])):
+                self._best[j] = cand_best.clone()
+            if self._worst[j] is None or better(float(self._worst[j].evals[j]), float(cand_worst.evals[j])):
+                self._worst[j] = cand_worst.clone()
+
+    @property
+    def stores_solution_stats(self) -> bool:
+        return self._store_solution_stats
+
+    @property
+    def best(self):
+        if self._best is None:
+            return None
+        return self._best[0] if len(self._senses) == 1 else list(self._best)
+
+    @property
+    def worst(self):
+        if self._worst is None:
+            return None
+        return self._worst[0] if len(self._senses) == 1 else list(self._worst)
+
+    # -- distribution-gradient machinery (P2) --------------------------------------
+
+    def sample_and_compute_gradients(
+        self,
+        distribution,
+        popsize: int,
+        *,
+        obj_index: Optional[int] = None,
+        ranking_method: Optional[str] = None,
+        num_interactions: Optional[int] = None,
+        popsize_max: Optional[int] = None,
+        ensure_even_popsize: bool = False,
+    ) -> dict:
+        """Sample a population from `distribution`, evaluate it, and return
+        ``{"gradients": ..., "num_solutions": ..., "mean_eval": ...}``.
+
+        SPMD mode (Comm attached, world > 1): each rank samples and
+        evaluates ``popsize / world`` solutions using its own slice of the
+        counter-based RNG stream; the utilities are ranked *globally* via an
+        all-gather of the (tiny) fitness vector, after which each rank
+        computes a partial gradient that is merged by a single all-reduce —
+        the RCCL collapse of the reference's actor round-trip
+        (core.py:2762-3074 → SURVEY.md §3.3).
+        """
+        obj_index = 0 if obj_index is None else int(obj_index)
+        comm = self._comm
+        self._before_grad_hook()
+        if comm is not None and comm.world_size > 1:
+            result = self._sample_and_compute_gradients_sharded(distribution, int(popsize), obj_index, ranking_method, comm, num_interactions, popsize_max, ensure_even_popsize)
+        else:
+            result = self._sample_and_compute_gradients(distribution, int(popsize), obj_index, ranking_method, num_interactions, popsize_max, ensure_even_popsize)
+        self._after_grad_status = self._after_grad_hook.accumulate_dict(result)
+        return result
+
+    def _sample_popsize(self, distribution, popsize: int, num_interactions, popsize_max, ensure_even: bool):
+        if ensure_even and (popsize % 2 != 0):
+            popsize += 1
+        return popsize
+
+    def _sample_and_compute_gradients(
+        self, distribution, popsize: int, obj_index: int, ranking_method, num_interactions=None, popsize_max=None, ensure_even_popsize: bool = False
+    ) -> dict:
+        popsize = self._sample_popsize(distribution, popsize, num_interactions, popsize_max, ensure_even_popsize)
+        batch = self.generate_batch(popsize, empty=True)
+        distribution.sample(out=batch.access_values(), generator=self._generator)
+        self.evaluate(batch)
+        fitnesses = batch._evals[:, obj_index]
+        sense = self._senses[obj_index]
+        grads = distribution.compute_gradients(batch._values, fitnesses, objective_sense=sense, ranking_method=ranking_method)
+        return {
+            "gradients": grads,
+            "num_solutions": popsize,
+            "mean_eval": float(torch.nanmean(fitnesses)),
+        }
+
+    def _sample_and_compute_gradients_sharded(
+        self, distribution, popsize: int, obj_index: int, ranking_method, comm, num_interactions=None, popsize_max=None, ensure_even_popsize: bool = False
+    ) -> dict:
+        world = comm.world_size
+        local_popsize = popsize // world
+        if ensure_even_popsize and local_popsize % 2 != 0:
+            local_popsize += 1
+        batch = self.generate_batch(local_popsize, empty=True)
+        distribution.sample(out=batch.access_values(), generator=self._generator)
+        self._before_eval_hook(batch)
+        self._evaluate_batch(batch)
+        if self._store_solution_stats:
+            self._update_solution_stats(batch)
+        self._after_eval_status = self._after_eval_hook.accumulate_dict(batch)
+        local_fit = batch._evals[:, obj_index]
+        # Global ranking: gather all fitnesses (N floats — latency-bound, tiny)
+        all_fit = comm.all_gather_vector(local_fit)
+        sense = self._senses[obj_index]
+        from .utils import ranking
+
+        all_utils = ranking.rank(all_fit, ranking_method or "raw", higher_is_better=(sense == "max"))
+        my_utils = all_utils[comm.rank * local_popsize : (comm.rank + 1) * local_popsize].to(dtype=local_fit.dtype)
+        grads = distribution._compute_gradients(batch._values, my_utils, ranking_used=(ranking_method or "raw"))
+        # Partial gradients sum over local samples; merge & normalize by the
+        # global popsize with one fused all-reduce.
+        total = world * local_popsize
+        for k in grads:
+            grads[k] = grads[k] * (local_popsize / total)
+        comm.all_reduce_container(grads)
+        return {
+            "gradients": grads,
+            "num_solutions": total,
+            "mean_eval": float(torch.nanmean(all_fit)),
+        }
+
+    # -- functional adapter ---------------------------------------------------------
+
+    def make_callable_evaluator(self, *, obj_index: Optional[int] = None) -> "ProblemBoundEvaluator":
+        return ProblemBoundEvaluator(self, obj_index=obj_index)
+
+    # -- cloning / pickling -----------------------------------------------------------
+
+    def _get_cloned_state(self, *, memo: dict) -> dict:
+        state = {}
+        for k, v in self.__dict__.items():
+            if k in ("_generator", "_comm"):
+                state[k] = None
+            else:
+                state[k] = deep_clone(v, otherwise_deepcopy=True, memo=memo)
+        return state
+
+    def __setstate__(self, state: dict):
+        super().__setstate__(state)
+        if self._seed is not None and self._generator is None:
+            self._generator = torch.Generator(device=self._device)
+            self._generator.manual_seed(int(self._seed))
+
+    def __repr__(self) -> str:
+        return f"<{type(self).__name__} senses={self._senses} length={self._solution_length} dtype={self._dtype} device={self._device}>"
+
+
+class ProblemBoundEvaluator:
+    """Adapts a Problem into a pure callable ``f(values_2d) -> evals`` for
+    the functional API (reference core.py:5109-5256). Supports an extra
+    leading batch dimension (the batched-search axis)."""
+
+    def __init__(self, problem: Problem, *, obj_index: Optional[int] = None):
+        self._problem = problem
+        self._obj_index = 0 if obj_index is None else problem._senses.index(problem._senses[obj_index]) if isinstance(obj_index, int) else 0
+        if obj_index is not None:
+            self._obj_index = int(obj_index)
+
+    @property
+    def problem(self) -> Problem:
+        return self._problem
+
+    def __call__(self, values: torch.Tensor) -> torch.Tensor:
+        problem = self._problem
+        if values.ndim == 2:
+            batch = problem.generate_batch(values.shape[0], empty=True)
+            batch.access_values().copy_(values.to(batch.device, dtype=batch.dtype))
+            problem.evaluate(batch)
+            return batch._evals[:, self._obj_index].to(values.device)
+        if values.ndim > 2:
+            lead = values.shape[:-2]
+            flat = values.reshape(-1, values.shape[-1])
+            # Evaluate each (batched search) population in one flat batch
+            out = self(flat)
+            return out.reshape(lead + (values.shape[-2],))
+        raise ValueError(f"Expected values of ndim >= 2, got shape {tuple(values.shape)}")

@@ -1,0 +1,27 @@
+"""Hot-op dispatch: hand-written CDNA4 HIP kernels with eager CPU references.
+
+See `evotorch_amd/ops/dispatch.py` for the routing contract and
+`evotorch_amd/ops/hip/` for the gfx950 kernel sources.
+"""
+
+from .dispatch import (
+    clipup_step_,
+    es_gradients,
+    fused_adam_step_,
+    hip_available,
+    hip_required,
+    load_hip,
+    sample_gaussian,
+    snes_gradients,
+)
+
+__all__ = [
+    "clipup_step_",
+    "es_gradients",
+    "fused_adam_step_",
+    "hip_available",
+    "hip_required",
+    "load_hip",
+    "sample_gaussian",
+    "snes_gradients",
+]

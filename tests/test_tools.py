@@ -1,0 +1,188 @@
+"""L1 utils tests (mirrors reference tests/test_tools_misc.py,
+test_ranking.py, test_hook.py, test_read_only_tensor.py,
+test_objectarray.py, test_cloning.py, test_constraint_penalization.py)."""
+
+import numpy as np
+import pytest
+import torch
+
+from evotorch_amd.utils import (
+    Hook,
+    ObjectArray,
+    ReadOnlyTensor,
+    as_immutable,
+    as_read_only_tensor,
+    clip_tensor,
+    clone,
+    deep_clone,
+    is_dtype_object,
+    make_gaussian,
+    make_tensor,
+    make_uniform,
+    modify_tensor,
+    rank,
+    to_torch_dtype,
+)
+from evotorch_amd.utils.constraints import log_barrier, penalty, violation
+from evotorch_amd.utils.misc import split_workload, stdev_from_radius
+
+
+def test_to_torch_dtype():
+    assert to_torch_dtype("float32") is torch.float32
+    assert to_torch_dtype(np.float64) is torch.float64
+    assert to_torch_dtype(float) is torch.float32
+    assert to_torch_dtype(torch.bfloat16) is torch.bfloat16
+    assert is_dtype_object(object)
+    assert is_dtype_object("object")
+    assert not is_dtype_object("float32")
+
+
+def test_make_uniform_bounds():
+    t = make_uniform(1000, lb=-2.0, ub=3.0, dtype="float32")
+    assert t.shape == (1000,)
+    assert float(t.min()) >= -2.0
+    assert float(t.max()) <= 3.0
+
+
+def test_make_gaussian_symmetric():
+    g = torch.Generator().manual_seed(7)
+    t = make_gaussian(10, 5, center=torch.ones(5), stdev=2.0, symmetric=True, generator=g)
+    # second half mirrors: row i and row i+5 average to the center
+    mid = (t[:5] + t[5:]) / 2.0
+    assert torch.allclose(mid, torch.ones(5, dtype=t.dtype).expand(5, 5), atol=1e-6)
+
+
+def test_rank_centered():
+    f = torch.tensor([3.0, 1.0, 2.0])
+    u = rank(f, "centered", higher_is_better=True)
+    assert torch.allclose(u, torch.tensor([0.5, -0.5, 0.0]))
+    u2 = rank(f, "centered", higher_is_better=False)
+    assert torch.allclose(u2, torch.tensor([-0.5, 0.5, 0.0]))
+
+
+def test_rank_nes_sums_to_zero():
+    f = torch.randn(101)
+    u = rank(f, "nes", higher_is_better=True)
+    assert abs(float(u.sum())) < 1e-5
+    # best solution gets the highest utility
+    assert int(u.argmax()) == int(f.argmax())
+
+
+def test_rank_batched():
+    f = torch.randn(4, 50)
+    u = rank(f, "centered", higher_is_better=True)
+    assert u.shape == (4, 50)
+    for i in range(4):
+        expected = rank(f[i], "centered", higher_is_better=True)
+        assert torch.allclose(u[i], expected)
+
+
+def test_modify_tensor_max_change():
+    original = torch.tensor([1.0, -2.0, 4.0])
+    target = torch.tensor([10.0, -10.0, 4.1])
+    out = modify_tensor(original, target, max_change=0.5)
+    assert torch.allclose(out, torch.tensor([1.5, -3.0, 4.1]))
+
+
+def test_clip_tensor():
+    x = torch.tensor([-5.0, 0.0, 5.0])
+    assert torch.allclose(clip_tensor(x, lb=-1.0, ub=1.0), torch.tensor([-1.0, 0.0, 1.0]))
+
+
+def test_split_workload():
+    assert split_workload(10, 3) == [4, 3, 3]
+    assert sum(split_workload(1000, 7)) == 1000
+
+
+def test_stdev_from_radius():
+    assert stdev_from_radius(4.0, 16) == pytest.approx(1.0)
+
+
+def test_hook_accumulates_dicts():
+    h = Hook()
+    h.append(lambda: {"a": 1})
+    h.append(lambda: {"b": 2})
+    assert h() == {"a": 1, "b": 2}
+    assert h.accumulate_dict() == {"a": 1, "b": 2}
+
+
+def test_hook_args():
+    seen = []
+    h = Hook([lambda x: seen.append(x)])
+    h(42)
+    assert seen == [42]
+
+
+def test_read_only_tensor_blocks_mutation():
+    t = as_read_only_tensor(torch.zeros(5))
+    assert isinstance(t, ReadOnlyTensor)
+    with pytest.raises(Exception):
+        t[0] = 1.0
+    with pytest.raises(Exception):
+        t.fill_(1.0)
+    c = t.clone()
+    c[0] = 1.0  # clone is mutable
+    assert float(c[0]) == 1.0
+    # views stay read-only
+    v = t[1:3]
+    with pytest.raises(Exception):
+        v[0] = 1.0
+    # arithmetic produces plain tensors
+    r = t + 1.0
+    r[0] = 5.0
+
+
+def test_object_array_immutability_and_views():
+    arr = ObjectArray(4)
+    src = [1, 2, 3]
+    arr[0] = src
+    src.append(4)  # must not affect stored clone
+    assert list(arr[0]) == [1, 2, 3]
+    view = arr[1:3]
+    view[0] = "hello"
+    assert arr[1] == "hello"
+    ro = arr.get_read_only_view()
+    with pytest.raises(RuntimeError):
+        ro[0] = 5
+
+
+def test_as_immutable():
+    d = as_immutable({"a": [1, 2], "b": torch.ones(3)})
+    with pytest.raises(Exception):
+        d["b"][0] = 5
+    assert list(d["a"]) == [1, 2]
+
+
+def test_deep_clone_tensors_and_cycles():
+    x = {"t": torch.ones(3)}
+    x["self"] = x
+    c = deep_clone(x)
+    assert c["self"] is c
+    c["t"][0] = 9.0
+    assert float(x["t"][0]) == 1.0
+    t = torch.arange(3.0)
+    assert torch.equal(clone(t), t)
+
+
+def test_make_tensor_object_dtype():
+    arr = make_tensor([1, "two", [3]], dtype=object)
+    assert isinstance(arr, ObjectArray)
+    assert arr[1] == "two"
+
+
+def test_violation():
+    assert float(violation(3.0, "<=", 2.0)) == pytest.approx(1.0)
+    assert float(violation(1.0, "<=", 2.0)) == 0.0
+    assert float(violation(1.0, ">=", 2.0)) == pytest.approx(1.0)
+    assert float(violation(1.0, "==", 3.0)) == pytest.approx(2.0)
+    v = violation(torch.tensor([1.0, 3.0]), "<=", torch.tensor([2.0, 2.0]))
+    assert torch.allclose(v, torch.tensor([0.0, 1.0]))
+
+
+def test_penalty_and_log_barrier():
+    p = penalty(3.0, "<=", 2.0, penalty_sign="-", linear=2.0)
+    assert float(p) == pytest.approx(-2.0)
+    lb = log_barrier(1.0, "<=", 2.0, penalty_sign="-")
+    assert float(lb) == pytest.approx(0.0)  # log(1) = 0
+    lb2 = log_barrier(2.5, "<=", 2.0, penalty_sign="-", inf=100.0)
+    assert float(lb2) == pytest.approx(-100.0)
