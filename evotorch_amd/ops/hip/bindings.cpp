@@ -1,0 +1,26 @@
+// Python bindings for the gfx950 kernel extension `evotorch_amd._C`.
+#include <torch/extension.h>
+
+namespace ea {
+void sample_gaussian(torch::Tensor out, torch::Tensor mu, torch::Tensor sigma, bool symmetric, int64_t seed);
+std::vector<torch::Tensor> es_gradients(torch::Tensor samples, torch::Tensor mu, torch::Tensor sigma,
+                                        torch::Tensor weights, bool symmetric);
+std::vector<torch::Tensor> snes_gradients(torch::Tensor samples, torch::Tensor mu, torch::Tensor sigma,
+                                          torch::Tensor weights);
+void clipup_step(torch::Tensor velocity, torch::Tensor grad, double step_size, double max_speed, double momentum);
+void adam_step(torch::Tensor step_out, torch::Tensor grad, torch::Tensor m, torch::Tensor v, int64_t step_count,
+               double stepsize, double beta1, double beta2, double epsilon);
+torch::Tensor rollout_linear(torch::Tensor params, torch::Tensor env_blob, torch::Tensor obs_stats_out,
+                             int64_t obs_dim, int64_t act_dim, int64_t rank, int64_t steps, double alive_bonus,
+                             double act_cost, int64_t init_seed, int64_t member_offset);
+}  // namespace ea
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    m.doc() = "evotorch_amd gfx950 (MI355X / CDNA4) HIP kernels";
+    m.def("sample_gaussian", &ea::sample_gaussian, "K1: philox Gaussian population sampling (plain/antithetic)");
+    m.def("es_gradients", &ea::es_gradients, "K3: fused (mu, sigma) ES gradient reduction");
+    m.def("snes_gradients", &ea::snes_gradients, "K3: SNES raw-noise gradient reduction");
+    m.def("clipup_step", &ea::clipup_step, "K4: fused ClipUp velocity update (no host sync)");
+    m.def("adam_step", &ea::adam_step, "K4: fused Adam ascent step");
+    m.def("rollout_linear", &ea::rollout_linear, "K10+K11: fused linear-policy episode rollout (synthetic env)");
+}

@@ -1,0 +1,160 @@
+"""SPMD comm-layer tests over gloo with world_size 2 — the in-process
+seam replacing the reference's `ray local_mode` trick
+(tests/conftest.py:28-40 of the reference)."""
+
+import os
+
+import pytest
+import torch
+import torch.multiprocessing as mp
+
+PORT = 29611
+
+
+def _worker(rank: int, world: int, fn_name: str, port: int, q):
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    torch.manual_seed(100 + rank)
+    import evotorch_amd.parallel.comm as comm_mod
+
+    comm_mod._global_comm = None
+    comm = comm_mod.Comm(backend="gloo", device=torch.device("cpu"))
+    try:
+        result = globals()[fn_name](comm, rank, world)
+        q.put((rank, "ok", result))
+    except Exception as e:  # pragma: no cover
+        import traceback
+
+        q.put((rank, "error", traceback.format_exc()))
+    finally:
+        torch.distributed.destroy_process_group()
+
+
+_port_counter = [PORT]
+
+
+def _run_world(fn_name: str, world: int = 2):
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    _port_counter[0] += 1
+    port = _port_counter[0]
+    procs = [ctx.Process(target=_worker, args=(r, world, fn_name, port, q)) for r in range(world)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(world):
+        rank, status, payload = q.get(timeout=180)
+        assert status == "ok", f"rank {rank} failed:\n{payload}"
+        results[rank] = payload
+    for p in procs:
+        p.join(timeout=60)
+    return results
+
+
+# -- worker bodies -----------------------------------------------------------
+
+
+def _body_collectives(comm, rank, world):
+    t = torch.full((4,), float(rank + 1))
+    comm.all_reduce_(t)
+    assert torch.allclose(t, torch.full((4,), 3.0))
+
+    v = torch.arange(3, dtype=torch.float32) + rank * 10
+    gathered = comm.all_gather_vector(v)
+    assert gathered.shape == (6,)
+    assert torch.allclose(gathered[:3], torch.arange(3, dtype=torch.float32))
+    assert torch.allclose(gathered[3:], torch.arange(3, dtype=torch.float32) + 10)
+
+    grads = {"mu": torch.full((5,), float(rank)), "sigma": torch.full((2,), 1.0)}
+    comm.all_reduce_container(grads)
+    assert torch.allclose(grads["mu"], torch.full((5,), 1.0))
+    assert torch.allclose(grads["sigma"], torch.full((2,), 2.0))
+
+    full = torch.zeros(5, 2)
+    ranges = [(0, 3), (3, 5)]
+    a, b = ranges[rank]
+    full[a:b] = rank + 1.0
+    comm.all_gather_rows(full, ranges)
+    assert torch.allclose(full[:3], torch.full((3, 2), 1.0))
+    assert torch.allclose(full[3:], torch.full((2, 2), 2.0))
+    return True
+
+
+def _body_sharded_evaluate(comm, rank, world):
+    from evotorch_amd import Problem
+    from evotorch_amd.decorators import vectorized
+
+    @vectorized
+    def sphere(x):
+        return (x**2).sum(-1)
+
+    prob = Problem("min", sphere, solution_length=6, initial_bounds=(-1, 1), seed=5)
+    prob.use_comm(comm)
+    batch = prob.generate_batch(7)  # same values on both ranks (same seed)
+    prob.evaluate(batch)
+    evals = batch.unsafe_evals[:, 0]
+    expected = (batch.unsafe_values**2).sum(-1)
+    assert torch.allclose(evals, expected, atol=1e-5)
+    return evals.tolist()
+
+
+def _body_distributed_pgpe(comm, rank, world):
+    from evotorch_amd import Problem
+    from evotorch_amd.algorithms import PGPE
+    from evotorch_amd.decorators import vectorized
+
+    @vectorized
+    def sphere(x):
+        return (x**2).sum(-1)
+
+    prob = Problem("min", sphere, solution_length=8, initial_bounds=(-1, 1), seed=50 + rank)
+    prob.use_comm(comm)
+    searcher = PGPE(prob, popsize=40, center_learning_rate=0.2, stdev_learning_rate=0.1,
+                    stdev_init=1.0, distributed=True, center_init=torch.ones(8))
+    for _ in range(10):
+        searcher.step()
+    center = searcher.status["center"]
+    # all ranks hold identical distributions after all-reduced updates
+    center_t = torch.Tensor.as_subclass(center, torch.Tensor).clone()
+    ref = center_t.clone()
+    comm.broadcast_(ref, src=0)
+    assert torch.allclose(center_t, ref, atol=1e-6), "ranks diverged"
+    assert searcher.status["mean_eval"] < 8.0  # descended from ~8 (center ones)
+    return center_t.tolist()
+
+
+def _body_obs_norm_allreduce(comm, rank, world):
+    from evotorch_amd.neuroevolution import SyntheticRolloutProblem
+
+    prob = SyntheticRolloutProblem(seed=60 + rank, episode_length=5)
+    prob.use_comm(comm)
+    result = prob.sample_and_compute_gradients(_make_dist(prob), 8, ranking_method="centered")
+    assert result["num_solutions"] == 8
+    # both ranks merged both shards' stats: 8 members * 5 steps = 40 obs
+    assert prob.obs_norm.count == 40.0
+    return prob.obs_norm.count
+
+
+def _make_dist(prob):
+    from evotorch_amd.distributions import SymmetricSeparableGaussian
+
+    L = prob.solution_length
+    return SymmetricSeparableGaussian({"mu": torch.zeros(L), "sigma": torch.ones(L) * 0.1})
+
+
+# -- tests -------------------------------------------------------------------
+
+
+@pytest.mark.parametrize(
+    "body",
+    ["_body_collectives", "_body_sharded_evaluate", "_body_distributed_pgpe", "_body_obs_norm_allreduce"],
+)
+def test_world2(body):
+    results = _run_world(body, world=2)
+    assert len(results) == 2
+    if body == "_body_sharded_evaluate":
+        assert results[0] == results[1]  # both ranks hold the full eval vector
+    if body == "_body_distributed_pgpe":
+        assert results[0] == results[1]
