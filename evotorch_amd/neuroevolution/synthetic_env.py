@@ -67,10 +67,19 @@ class SyntheticEnvSpec:
         return self.act_dim * self.obs_dim + self.act_dim
 
     def env_blob(self, mean: torch.Tensor, std: torch.Tensor, device=None) -> torch.Tensor:
-        """Pack (V, U_T, D2_T, c, wr, mean, std) contiguously as fp32."""
-        device = device or self.device
-        parts = [self.V, self.U_T, self.D2_T, self.c, self.wr, mean.reshape(-1), std.reshape(-1)]
-        return torch.cat([p.reshape(-1).to(torch.float32) for p in parts]).to(device).contiguous()
+        """Pack (V, U_T, D2_T, c, wr, mean, std) contiguously as fp32. The
+        static matrices are cached per device; only mean/std change per
+        generation."""
+        device = torch.device(device or self.device)
+        cache = getattr(self, "_matrix_cache", None)
+        if cache is None or cache[0] != device:
+            mats = torch.cat(
+                [p.reshape(-1).to(torch.float32) for p in (self.V, self.U_T, self.D2_T, self.c, self.wr)]
+            ).to(device)
+            cache = (device, mats)
+            self._matrix_cache = cache
+        tail = torch.cat([mean.reshape(-1).to(torch.float32), std.reshape(-1).to(torch.float32)]).to(device)
+        return torch.cat([cache[1], tail]).contiguous()
 
     def initial_obs(self, n_members: int, member_offset: int, init_seed: int, device=None) -> torch.Tensor:
         """0.1 * N(0,1), matching the kernel's philox stream exactly."""
