@@ -116,27 +116,30 @@ def rollout_eager(
         x = x.to(device, torch.float32)
         return x.to(torch.bfloat16).to(torch.float32) if bf16_operands else x
 
+    # bf16 OPERANDS / fp32 accumulation, exactly as the kernel stores them
     W = q(params[:, : A * O].reshape(N, A, O))
     b = params[:, A * O :].to(device, dt)
     V = q(spec.V)
     U_T = q(spec.U_T)
     D2_T = q(spec.D2_T)
-    c = q(spec.c)
-    wr = q(spec.wr)
-    mean_q = q(mean)
-    std_q = q(std)
+    c = spec.c.to(device, dt)
+    wr = spec.wr.to(device, dt)
+    mean_f = mean.to(device, dt)
+    inv_std = 1.0 / std.to(device, dt)
 
-    obs = spec.initial_obs(N, member_offset, init_seed, device=device).to(dt)
+    obs = q(spec.initial_obs(N, member_offset, init_seed, device=device))
     fitness = torch.zeros(N, dtype=dt, device=device)
     obs_sum = torch.zeros(O, dtype=dt, device=device)
     obs_sumsq = torch.zeros(O, dtype=dt, device=device)
     for _ in range(steps):
-        obs_n = (obs - mean_q) / std_q
+        obs_n = q((obs - mean_f) * inv_std)
         act = torch.clamp(torch.einsum("nao,no->na", W, obs_n) + b, -1.0, 1.0)
-        h = obs @ V.T  # (N, R)
-        o_new = torch.tanh(h @ U_T + act @ D2_T + c)
-        fitness = fitness + o_new @ wr + spec.alive_bonus - spec.act_cost * (act**2).sum(-1) / A
+        act_cost_term = spec.act_cost * (act**2).sum(-1) / A
+        act_b = q(act)
+        h = q(obs @ V.T)  # (N, R)
+        o_new = torch.tanh(h @ U_T + act_b @ D2_T + c)
+        fitness = fitness + o_new @ wr + spec.alive_bonus - act_cost_term
         obs_sum += o_new.sum(0)
         obs_sumsq += (o_new**2).sum(0)
-        obs = o_new
+        obs = q(o_new)
     return fitness, (float(N * steps), obs_sum, obs_sumsq)

@@ -2,27 +2,32 @@
 // policies through the synthetic vectorized environment, one kernel launch
 // per generation (SURVEY.md §3.4 — the VecGymNE hot loop, collapsed).
 //
-// MI355X design: one workgroup per population member. The member's policy
-// weights AND the (shared) environment matrices are staged once into LDS as
-// bf16 and stay resident for the entire T-step episode — the inner loop
-// touches no global memory at all, so the rollout is pure VALU compute
-// instead of the reference's per-step kernel round-trips through HBM
-// (policy W re-read every step). Observation-normalization statistics
-// (sum, sumsq) accumulate per-thread and are merged with one atomic pass
-// at the end (K11; they become a single RCCL all-reduce across ranks —
-// SURVEY.md §2.8 P5).
+// MI355X design (v2): one workgroup per population member. The member's
+// policy weights AND the (shared) environment matrices are staged once
+// into LDS and stay resident for the entire T-step episode — the inner
+// loop touches no global memory at all. All matrix operands are bf16 and
+// every inner product runs on `v_dot2_f32_bf16` (2 bf16 MACs/instruction,
+// fp32 accumulate, no per-element converts). LDS layouts:
+//   * wave-reduced dots (policy W, dynamics factor V): row-major [out][O],
+//     lanes read consecutive bf16x2 along O — conflict-free.
+//   * per-thread dots (U, D2): per-OUTPUT rows padded to stride 18
+//     (36 B ⇒ bank index advances by 9 per lane, gcd(9,32)=1 ⇒ ≤2-way,
+//     i.e. free on CDNA4 — see cdna_hip_programming.md §6 G4).
+// Observation-normalization statistics (sum, sumsq) accumulate per-thread
+// in registers and are merged with one atomic pass at the end (K11; they
+// become a single RCCL all-reduce across ranks — SURVEY.md §2.8 P5).
 //
 // Environment spec (must match the eager reference in
-// evotorch_amd/neuroevolution/synthetic_env.py):
-//   state  o ∈ R^O, action a = clip(W·obs_norm(o) + b, -1, 1) ∈ R^A
-//   h  = V·o                       (V: R×O, low-rank dynamics factor)
-//   o' = tanh(Uᵀ·h + D2ᵀ·a + c)    (U stored transposed [R][O], D2 [A][O])
-//   r  = wr·o' + alive_bonus − act_cost·‖a‖²/A
-//   fitness = Σ_t r_t over T steps; o_0 = 0.1·philox_normal(member)
-// All matrices are bf16 in LDS; accumulation is fp32.
+// evotorch_amd/neuroevolution/synthetic_env.py::rollout_eager, which
+// quantizes the same operands to bf16):
+//   obs_n = bf16((obs − mean) · inv_std)
+//   a     = clip(W·obs_n + b, −1, 1)        (W bf16, accum fp32)
+//   h     = bf16(V·obs)
+//   o'    = tanh(Σ_i U_T[i]·h[i] + Σ_m D2_T[m]·a[m] + c)
+//   r     = wr·o' + alive_bonus − act_cost·‖a‖²/A
+//   o₀    = 0.1·philox_normal(member);  fitness = Σ_t r_t
 
 #include <hip/hip_runtime.h>
-#include <hip/hip_bf16.h>
 #include <torch/extension.h>
 #include <ATen/cuda/CUDAContext.h>
 
@@ -33,12 +38,14 @@ namespace ea {
 
 #define CHECK_GPU(x) TORCH_CHECK(x.is_cuda(), #x " must be a ROCm tensor")
 
-__device__ __forceinline__ float bf2f(__hip_bfloat16 v) { return __bfloat162float(v); }
-__device__ __forceinline__ __hip_bfloat16 f2bf(float v) { return __float2bfloat16(v); }
+typedef __attribute__((ext_vector_type(2))) __bf16 bf16x2;
+
+__device__ __forceinline__ __bf16 f2b(float v) { return (__bf16)v; }
+__device__ __forceinline__ float b2f(__bf16 v) { return (float)v; }
 
 struct RolloutArgs {
     const float* params;      // [n_members][A*O + A]  (W row-major, then b)
-    const float* env_blob;    // packed bf16/fp32 env data, see offsets below
+    const float* env_blob;    // packed fp32 env data (see env_blob())
     float* fitness_out;       // [n_members]
     float* obs_stats_out;     // [2][O]  (sum, sumsq) — atomically accumulated
     int n_members;
@@ -48,53 +55,63 @@ struct RolloutArgs {
     unsigned long long init_seed;
 };
 
-// env_blob layout (all fp32, converted to bf16 while staging into LDS):
-//   V     [rank][O]
-//   U_T   [rank][O]   (U transposed: U_T[i][j] = U[j][i])
-//   D2_T  [act][O]
-//   c     [O]
-//   wr    [O]
-//   mean  [O]
-//   std   [O]         (already max(std, eps))
+// env_blob layout (fp32): V [R][O] · U_T [R][O] · D2_T [A][O] · c [O] ·
+// wr [O] · mean [O] · std [O]
 
 __global__ __launch_bounds__(256, 2) void rollout_linear_kernel(RolloutArgs args) {
     const int O = args.obs_dim, A = args.act_dim, R = args.rank;
     const int tid = threadIdx.x;
     const int member = blockIdx.x;
     if (member >= args.n_members) return;
+    const int A_PAD = (A + 2) & ~1;  // D2 per-output row stride (18 for A=17)
+    const int R_PAD = (R + 2) & ~1;  // U per-output row stride
 
     extern __shared__ unsigned char lds_raw[];
-    __hip_bfloat16* W_l = reinterpret_cast<__hip_bfloat16*>(lds_raw);            // [A][O]
-    __hip_bfloat16* V_l = W_l + A * O;                                           // [R][O]
-    __hip_bfloat16* UT_l = V_l + R * O;                                          // [R][O]
-    __hip_bfloat16* D2T_l = UT_l + R * O;                                        // [A][O]
-    __hip_bfloat16* c_l = D2T_l + A * O;                                         // [O]
-    __hip_bfloat16* wr_l = c_l + O;                                              // [O]
-    __hip_bfloat16* mean_l = wr_l + O;                                           // [O]
-    __hip_bfloat16* std_l = mean_l + O;                                          // [O]
-    float* b_l = reinterpret_cast<float*>(std_l + O);                            // [A]
-    float* obs = b_l + A;                                                        // [O]
-    float* obs_n = obs + O;                                                      // [O]
-    float* h_l = obs_n + O;                                                      // [R]
-    float* act_l = h_l + R;                                                      // [A]
-    float* scratch = act_l + A;                                                  // [8]
+    __bf16* W_l = reinterpret_cast<__bf16*>(lds_raw);   // [A][O] row-major
+    __bf16* V_l = W_l + A * O;                          // [R][O] row-major
+    __bf16* U_row = V_l + R * O;                        // [O][R_PAD]
+    __bf16* D2_row = U_row + O * R_PAD;                 // [O][A_PAD]
+    __bf16* obs_b = D2_row + O * A_PAD;                 // [O]    bf16 state
+    __bf16* obsn_b = obs_b + O;                         // [O]    normalized
+    __bf16* h_b = obsn_b + O;                           // [R_PAD]
+    __bf16* act_b = h_b + R_PAD;                        // [A_PAD]
+    float* c_l = reinterpret_cast<float*>(act_b + A_PAD);  // [O]
+    float* wr_l = c_l + O;                              // [O]
+    float* mean_l = wr_l + O;                           // [O]
+    float* istd_l = mean_l + O;                         // [O]
+    float* b_l = istd_l + O;                            // [A]
+    float* scratch = b_l + A;                           // [8]
 
-    // ---- stage: policy params (per member) + env matrices (shared) ----
+    // ---- stage ----
     const float* my_params = args.params + (long)member * (A * O + A);
-    for (int i = tid; i < A * O; i += blockDim.x) W_l[i] = f2bf(my_params[i]);
+    for (int i = tid; i < A * O; i += blockDim.x) W_l[i] = f2b(my_params[i]);
     for (int i = tid; i < A; i += blockDim.x) b_l[i] = my_params[A * O + i];
     {
         const float* e = args.env_blob;
         const int RO = R * O, AO = A * O;
-        for (int i = tid; i < RO; i += blockDim.x) V_l[i] = f2bf(e[i]);
-        for (int i = tid; i < RO; i += blockDim.x) UT_l[i] = f2bf(e[RO + i]);
-        for (int i = tid; i < AO; i += blockDim.x) D2T_l[i] = f2bf(e[2 * RO + i]);
+        for (int i = tid; i < RO; i += blockDim.x) V_l[i] = f2b(e[i]);
+        // U arrives as U_T [R][O]; store per-output rows U_row[j][i] = U_T[i][j]
+        for (int i = tid; i < RO; i += blockDim.x) {
+            const int r = i / O, j = i % O;
+            U_row[j * R_PAD + r] = f2b(e[RO + i]);
+        }
+        for (int i = tid; i < AO; i += blockDim.x) {
+            const int m = i / O, j = i % O;
+            D2_row[j * A_PAD + m] = f2b(e[2 * RO + i]);
+        }
         const float* tail = e + 2 * RO + AO;
-        for (int i = tid; i < O; i += blockDim.x) {
-            c_l[i] = f2bf(tail[i]);
-            wr_l[i] = f2bf(tail[O + i]);
-            mean_l[i] = f2bf(tail[2 * O + i]);
-            std_l[i] = f2bf(tail[3 * O + i]);
+        for (int j = tid; j < O; j += blockDim.x) {
+            c_l[j] = tail[j];
+            wr_l[j] = tail[O + j];
+            mean_l[j] = tail[2 * O + j];
+            istd_l[j] = 1.0f / tail[3 * O + j];
+        }
+        // zero the pads once so dot2 over padded rows is exact
+        for (int i = tid; i < R_PAD; i += blockDim.x) h_b[i] = f2b(0.0f);
+        for (int i = tid; i < A_PAD; i += blockDim.x) act_b[i] = f2b(0.0f);
+        for (int j = tid; j < O; j += blockDim.x) {
+            for (int r = R; r < R_PAD; ++r) U_row[j * R_PAD + r] = f2b(0.0f);
+            for (int m = A; m < A_PAD; ++m) D2_row[j * A_PAD + m] = f2b(0.0f);
         }
     }
     // initial observation: 0.1 * N(0,1), deterministic per global member id
@@ -105,7 +122,7 @@ __global__ __launch_bounds__(256, 2) void rollout_linear_kernel(RolloutArgs args
 #pragma unroll
         for (int u = 0; u < 4; ++u) {
             const int j = j4 * 4 + u;
-            if (j < O) obs[j] = 0.1f * z[u];
+            if (j < O) obs_b[j] = f2b(0.1f * z[u]);
         }
     }
     __syncthreads();
@@ -113,57 +130,131 @@ __global__ __launch_bounds__(256, 2) void rollout_linear_kernel(RolloutArgs args
     const int lane = tid & (kWaveSize - 1);
     const int wave = tid / kWaveSize;
     const int nwaves = blockDim.x / kWaveSize;
+    const int n_pairs = O / 2;          // O must be even
+    const int n_outputs = A + R;        // group-reduced dots per step
 
-    float fit_part = 0.0f;       // per-thread fitness partial (dims owned)
-    float actsq_part = 0.0f;     // per-thread Σ_t a²   (acts owned by wave 0 lanes)
-    float stat_sum[8], stat_sumsq[8];
-    const int dims_per_thread = (O + blockDim.x - 1) / blockDim.x;
+    // phase-1 layout: 8-lane groups, one output per group per round — the
+    // wave computes 8 dots concurrently with only a 3-level shuffle
+    // reduction (a 64-lane tree would serialize 6 dependent shuffles per
+    // output; measured 10:1 SQ_WAIT:SQ_BUSY). With 4 waves × 8 groups,
+    // all 33 outputs finish in ceil(33/32) = 2 rounds.
+    constexpr int kGroup = 8;
+    constexpr int kRounds = 2;  // supports up to 64 outputs
+    const int group = lane / kGroup;
+    const int glane = lane % kGroup;
+    const int groups_per_block = (int)(blockDim.x / kGroup);  // threads/8
+
+    const __bf16* my_row[kRounds];
+    bool my_is_act[kRounds];
+    bool my_valid[kRounds];
+    int my_out[kRounds];
 #pragma unroll
-    for (int u = 0; u < 8; ++u) { stat_sum[u] = 0.0f; stat_sumsq[u] = 0.0f; }
+    for (int r = 0; r < kRounds; ++r) {
+        const int out = (wave * kGroup + group) + r * groups_per_block;
+        my_valid[r] = out < n_outputs;
+        my_out[r] = my_valid[r] ? out : 0;
+        my_is_act[r] = my_out[r] < A;
+        my_row[r] = my_is_act[r] ? (W_l + my_out[r] * O) : (V_l + (my_out[r] - A) * O);
+    }
+
+    float fit_part = 0.0f;
+    float actsq_part = 0.0f;
+    float stat_sum[2] = {0.0f, 0.0f}, stat_sumsq[2] = {0.0f, 0.0f};
+
+    // initial normalization (later steps fuse it into phase 2's epilogue)
+    for (int j = tid; j < O; j += blockDim.x) {
+        obsn_b[j] = f2b((b2f(obs_b[j]) - mean_l[j]) * istd_l[j]);
+    }
+    __syncthreads();
 
     for (int t = 0; t < args.steps; ++t) {
-        // phase 1: obs normalization (owned dims)
-        for (int j = tid; j < O; j += blockDim.x) {
-            obs_n[j] = (obs[j] - bf2f(mean_l[j])) / bf2f(std_l[j]);
+        // phase 1: all of this wave's dots concurrently (policy rows read
+        // obsn, dynamics-factor rows read obs; both vectors loaded once
+        // per pair-column and selected per output)
+        float acc[kRounds];
+#pragma unroll
+        for (int r = 0; r < kRounds; ++r) acc[r] = 0.0f;
+#pragma unroll
+        for (int r = 0; r < kRounds; ++r) {
+            if (!my_valid[r]) continue;
+            const __bf16* row = my_row[r];
+            const __bf16* vec = my_is_act[r] ? obsn_b : obs_b;
+            float a0 = 0.0f, a1 = 0.0f;  // two chains: halve the serial depth
+            const int full_iters = n_pairs / kGroup;   // guard-free iterations
+            const int tail = n_pairs % kGroup;
+            for (int i = 0; i + 1 < full_iters; i += 2) {
+                const int p0 = glane + i * kGroup;
+                const int p1 = glane + (i + 1) * kGroup;
+                a0 = __builtin_amdgcn_fdot2_f32_bf16(
+                    *reinterpret_cast<const bf16x2*>(row + 2 * p0),
+                    *reinterpret_cast<const bf16x2*>(vec + 2 * p0), a0, false);
+                a1 = __builtin_amdgcn_fdot2_f32_bf16(
+                    *reinterpret_cast<const bf16x2*>(row + 2 * p1),
+                    *reinterpret_cast<const bf16x2*>(vec + 2 * p1), a1, false);
+            }
+            if (full_iters & 1) {
+                const int p0 = glane + (full_iters - 1) * kGroup;
+                a0 = __builtin_amdgcn_fdot2_f32_bf16(
+                    *reinterpret_cast<const bf16x2*>(row + 2 * p0),
+                    *reinterpret_cast<const bf16x2*>(vec + 2 * p0), a0, false);
+            }
+            if (tail && glane < tail) {
+                const int p0 = glane + full_iters * kGroup;
+                a1 = __builtin_amdgcn_fdot2_f32_bf16(
+                    *reinterpret_cast<const bf16x2*>(row + 2 * p0),
+                    *reinterpret_cast<const bf16x2*>(vec + 2 * p0), a1, false);
+            }
+            acc[r] = a0 + a1;
         }
-        __syncthreads();
-
-        // phase 2: policy  a[j] = clip(W[j]·obs_n + b[j])  and  h[i] = V[i]·obs
-        for (int j = wave; j < A; j += nwaves) {
-            float acc = 0.0f;
-            for (int k = lane; k < O; k += kWaveSize) acc = fmaf(bf2f(W_l[j * O + k]), obs_n[k], acc);
-            acc = wave_reduce_sum(acc);
-            if (lane == 0) {
-                float a = acc + b_l[j];
-                a = fminf(fmaxf(a, -1.0f), 1.0f);
-                act_l[j] = a;
+        // 3-level in-group reductions (both rounds' shuffles overlap)
+#pragma unroll
+        for (int offset = kGroup / 2; offset > 0; offset >>= 1) {
+#pragma unroll
+            for (int r = 0; r < kRounds; ++r) {
+                acc[r] += __shfl_down(acc[r], offset, kWaveSize);
             }
         }
-        for (int i = wave; i < R; i += nwaves) {
-            float acc = 0.0f;
-            for (int k = lane; k < O; k += kWaveSize) acc = fmaf(bf2f(V_l[i * O + k]), obs[k], acc);
-            acc = wave_reduce_sum(acc);
-            if (lane == 0) h_l[i] = acc;
+        if (glane == 0) {
+#pragma unroll
+            for (int r = 0; r < kRounds; ++r) {
+                if (!my_valid[r]) continue;
+                if (my_is_act[r]) {
+                    float a = fminf(fmaxf(acc[r] + b_l[my_out[r]], -1.0f), 1.0f);
+                    act_b[my_out[r]] = f2b(a);
+                    actsq_part = fmaf(a, a, actsq_part);
+                } else {
+                    h_b[my_out[r] - A] = f2b(acc[r]);
+                }
+            }
         }
         __syncthreads();
 
-        // action cost accumulation (wave 0, lanes over A)
-        if (wave == 0 && lane < A) {
-            const float a = act_l[lane];
-            actsq_part = fmaf(a, a, actsq_part);
-        }
-
-        // phase 3: dynamics  o'[j] = tanh(c[j] + Σ_i U_T[i][j] h[i] + Σ_m D2_T[m][j] a[m])
+        // phase 2: per-thread dynamics rows (padded-stride LDS reads) with
+        // the NEXT step's normalization fused into the epilogue
         for (int j = tid; j < O; j += blockDim.x) {
-            float v = bf2f(c_l[j]);
-            for (int i = 0; i < R; ++i) v = fmaf(bf2f(UT_l[i * O + j]), h_l[i], v);
-            for (int m = 0; m < A; ++m) v = fmaf(bf2f(D2T_l[m * O + j]), act_l[m], v);
-            const float o_new = tanhf(v);
-            fit_part = fmaf(bf2f(wr_l[j]), o_new, fit_part);
-            const int slot = (j - tid) / (int)blockDim.x;
+            float vacc = c_l[j];
+            const __bf16* urow = U_row + j * R_PAD;
+            const __bf16* drow = D2_row + j * A_PAD;
+#pragma unroll 4
+            for (int p = 0; p < R_PAD / 2; ++p) {
+                vacc = __builtin_amdgcn_fdot2_f32_bf16(
+                    *reinterpret_cast<const bf16x2*>(urow + 2 * p),
+                    *reinterpret_cast<const bf16x2*>(h_b + 2 * p), vacc, false);
+            }
+#pragma unroll 3
+            for (int p = 0; p < A_PAD / 2; ++p) {
+                vacc = __builtin_amdgcn_fdot2_f32_bf16(
+                    *reinterpret_cast<const bf16x2*>(drow + 2 * p),
+                    *reinterpret_cast<const bf16x2*>(act_b + 2 * p), vacc, false);
+            }
+            const float o_new = tanhf(vacc);
+            fit_part = fmaf(wr_l[j], o_new, fit_part);
+            const int slot = j >= (int)blockDim.x;
             stat_sum[slot] += o_new;
             stat_sumsq[slot] = fmaf(o_new, o_new, stat_sumsq[slot]);
-            obs[j] = o_new;  // in-place: raw obs of step t is dead after phase 2
+            const __bf16 ob = f2b(o_new);
+            obs_b[j] = ob;
+            obsn_b[j] = f2b((b2f(ob) - mean_l[j]) * istd_l[j]);
         }
         __syncthreads();
     }
@@ -177,7 +268,7 @@ __global__ __launch_bounds__(256, 2) void rollout_linear_kernel(RolloutArgs args
             total + args.alive_bonus * args.steps - args.act_cost * act_total / (float)A;
     }
     for (int j = tid; j < O; j += blockDim.x) {
-        const int slot = (j - tid) / (int)blockDim.x;
+        const int slot = j >= (int)blockDim.x;
         atomicAdd(&args.obs_stats_out[j], stat_sum[slot]);
         atomicAdd(&args.obs_stats_out[O + j], stat_sumsq[slot]);
     }
@@ -192,6 +283,8 @@ torch::Tensor rollout_linear(torch::Tensor params, torch::Tensor env_blob, torch
     const int n = (int)params.size(0);
     const int O = (int)obs_dim, A = (int)act_dim, R = (int)rank;
     TORCH_CHECK(params.size(1) == (int64_t)A * O + A, "param length mismatch");
+    TORCH_CHECK(O % 2 == 0, "obs_dim must be even (bf16x2 packing)");
+    TORCH_CHECK(O <= 512, "obs_dim too large for the 2-slot stat accumulators");
     auto fitness = torch::empty({n}, params.options());
 
     RolloutArgs args;
@@ -207,12 +300,13 @@ torch::Tensor rollout_linear(torch::Tensor params, torch::Tensor env_blob, torch
     args.act_cost = (float)act_cost;
     args.init_seed = (unsigned long long)init_seed;
 
-    const size_t bf_elems = (size_t)(A * O) + 2 * (size_t)(R * O) + (size_t)(A * O) + 4 * (size_t)O;
-    const size_t f32_elems = (size_t)A + 2 * (size_t)O + (size_t)R + (size_t)A + 8;
+    const int A_PAD = (A + 2) & ~1, R_PAD = (R + 2) & ~1;
+    const size_t bf_elems = (size_t)A * O + (size_t)R * O + (size_t)O * R_PAD + (size_t)O * A_PAD +
+                            2 * (size_t)O + R_PAD + A_PAD;
+    const size_t f32_elems = 4 * (size_t)O + (size_t)A + 8;
     const size_t lds_bytes = bf_elems * 2 + f32_elems * 4 + 64;
     TORCH_CHECK(lds_bytes <= 64 * 1024, "rollout LDS footprint too large: ", lds_bytes,
                 " bytes (reduce rank / dims)");
-    TORCH_CHECK((O + 255) / 256 <= 8, "obs_dim too large for stat slots");
     auto stream = at::cuda::getCurrentCUDAStream();
     hipLaunchKernelGGL(rollout_linear_kernel, dim3(n), dim3(256), lds_bytes, stream, args);
     return fitness;
