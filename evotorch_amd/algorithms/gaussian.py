@@ -259,7 +259,8 @@ class GaussianSearchAlgorithm(SearchAlgorithm, SinglePopulationAlgorithmMixin):
 
     def _get_mean_eval(self) -> Optional[float]:
         if self._population is None:
-            return self._mean_eval
+            me = self._mean_eval
+            return float(me) if isinstance(me, torch.Tensor) else me
         return float(torch.mean(torch.Tensor.as_subclass(self._population.evals, torch.Tensor)[:, self._obj_index]))
 
 

@@ -1068,7 +1068,7 @@ class Problem(TensorMakerMixin, Serializable):
         return {
             "gradients": grads,
             "num_solutions": popsize,
-            "mean_eval": float(torch.nanmean(fitnesses)),
+            "mean_eval": torch.nanmean(fitnesses),  # 0-dim tensor: no host sync
         }
 
     def _sample_and_compute_gradients_sharded(
@@ -1103,7 +1103,7 @@ class Problem(TensorMakerMixin, Serializable):
         return {
             "gradients": grads,
             "num_solutions": total,
-            "mean_eval": float(torch.nanmean(all_fit)),
+            "mean_eval": torch.nanmean(all_fit),  # 0-dim tensor: no host sync
         }
 
     # -- functional adapter ---------------------------------------------------------

@@ -27,6 +27,7 @@ class RunningNorm:
         self._count = torch.zeros((), dtype=torch.float64, device=self._device)
         self._sum = torch.zeros(self._shape, dtype=dtype, device=self._device)
         self._sum_sq = torch.zeros(self._shape, dtype=dtype, device=self._device)
+        self._has_data = False
 
     @property
     def shape(self):
@@ -43,6 +44,12 @@ class RunningNorm:
     @property
     def count(self) -> float:
         return float(self._count)
+
+    @property
+    def has_data(self) -> bool:
+        """Host-side flag: True once any observation was accumulated (no
+        device sync, unlike `count`)."""
+        return self._has_data
 
     @property
     def sum(self) -> torch.Tensor:
@@ -71,12 +78,17 @@ class RunningNorm:
             self._count += x._count.to(self._count.device)
             self._sum += x._sum.to(self._sum.device, self._dtype)
             self._sum_sq += x._sum_sq.to(self._sum_sq.device, self._dtype)
+            self._has_data = self._has_data or x._has_data
             return
         if isinstance(x, tuple) and len(x) == 3:
             count, s, ss = x
-            self._count += float(count)
+            if isinstance(count, torch.Tensor):
+                self._count += count.to(self._count.device, torch.float64)
+            else:
+                self._count += float(count)
             self._sum += torch.as_tensor(s, dtype=self._dtype, device=self._device)
             self._sum_sq += torch.as_tensor(ss, dtype=self._dtype, device=self._device)
+            self._has_data = True
             return
         x = torch.as_tensor(x, dtype=self._dtype, device=self._device)
         if x.ndim == len(self._shape):
@@ -86,10 +98,11 @@ class RunningNorm:
         self._count += x.shape[0]
         self._sum += x.sum(dim=0)
         self._sum_sq += (x**2).sum(dim=0)
+        self._has_data = self._has_data or x.shape[0] > 0
 
     def normalize(self, x: torch.Tensor) -> torch.Tensor:
         x = torch.as_tensor(x, dtype=self._dtype, device=self._device)
-        if self.count == 0:
+        if not self._has_data:
             result = x
         else:
             result = (x - self.mean) / self.stdev
@@ -110,6 +123,7 @@ class RunningNorm:
         self._count.zero_()
         self._sum.zero_()
         self._sum_sq.zero_()
+        self._has_data = False
 
     def to(self, device) -> "RunningNorm":
         device = torch.device(device)
@@ -119,6 +133,7 @@ class RunningNorm:
         out._count = self._count.to(device)
         out._sum = self._sum.to(device)
         out._sum_sq = self._sum_sq.to(device)
+        out._has_data = self._has_data
         return out
 
     def to_layer(self) -> "ObsNormLayer":

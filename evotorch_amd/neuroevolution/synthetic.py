@@ -63,7 +63,7 @@ class SyntheticRolloutProblem(Problem):
         return self._obs_norm
 
     def _norm_mean_std(self):
-        if self._obs_norm_enabled and self._obs_norm.count > 0:
+        if self._obs_norm_enabled and self._obs_norm.has_data:
             return self._obs_norm.mean, self._obs_norm.stdev
         O = self._spec.obs_dim
         return (
@@ -81,7 +81,9 @@ class SyntheticRolloutProblem(Problem):
         values = batch.access_values(keep_evals=True)
         n = len(batch)
         spec = self._spec
-        init_seed = int(torch.randint(0, 2**31, (1,), generator=self._generator, device=self._generator.device if self._generator is not None else "cpu").item())
+        from ..ops.dispatch import _seed_from_generator
+
+        init_seed = _seed_from_generator(self._generator, self._device) & 0x7FFFFFFF
         member_offset = 0
         comm = self._comm
         if comm is not None and comm.world_size > 1:
@@ -108,7 +110,7 @@ class SyntheticRolloutProblem(Problem):
                 init_seed,
                 member_offset,
             )
-            triple = (float(n * spec.episode_length), obs_stats[: spec.obs_dim], obs_stats[spec.obs_dim :])
+            triple = (float(n) * spec.episode_length, obs_stats[: spec.obs_dim], obs_stats[spec.obs_dim :])
         else:
             mean, std = self._norm_mean_std()
             fitness, triple = rollout_eager(
@@ -141,7 +143,7 @@ class SyntheticRolloutProblem(Problem):
         if comm is not None and comm.world_size > 1:
             packed = torch.cat([torch.tensor([count], dtype=torch.float32, device=s.device), s.reshape(-1), ss.reshape(-1)])
             comm.all_reduce_(packed)
-            count = float(packed[0])
+            count = packed[0]  # stays a device tensor: no host sync
             s = packed[1 : 1 + self._spec.obs_dim]
             ss = packed[1 + self._spec.obs_dim :]
         self._obs_norm.update((count, s, ss))

@@ -73,13 +73,32 @@ def _allow_eager_on_gpu() -> bool:
     return os.environ.get("EVOTORCH_AMD_ALLOW_EAGER_GPU", "0") == "1"
 
 
+_SPLITMIX_INC = 0x9E3779B97F4A7C15
+_seed_fallback_counter = [0x1234ABCD]
+_gen_counters: dict = {}
+
+
+def _splitmix64(x: int) -> int:
+    x = (x + _SPLITMIX_INC) & 0xFFFFFFFFFFFFFFFF
+    z = x
+    z = ((z ^ (z >> 30)) * 0xBF58476D1CE4E5B9) & 0xFFFFFFFFFFFFFFFF
+    z = ((z ^ (z >> 27)) * 0x94D049BB133111EB) & 0xFFFFFFFFFFFFFFFF
+    return (z ^ (z >> 31)) & 0x7FFFFFFFFFFFFFFF
+
+
 def _seed_from_generator(generator: Optional[torch.Generator], device: torch.device) -> int:
-    """Derive a fresh 63-bit seed from (and advancing) the given generator,
-    so kernel-sampled streams stay reproducible under the same generator
-    discipline as the eager path."""
+    """Derive a fresh 63-bit seed deterministically from the generator's
+    initial seed plus a host-side call counter (splitmix64 mix) — NO device
+    RNG draw, so kernel launches never force a host-device sync."""
     if generator is not None:
-        return int(torch.randint(0, 2**62, (1,), generator=generator, device=generator.device).item())
-    return int(torch.randint(0, 2**62, (1,), device="cpu").item())
+        key = id(generator)
+        counter = _gen_counters.get(key)
+        if counter is None:
+            counter = int(generator.initial_seed()) & 0xFFFFFFFFFFFFFFFF
+        _gen_counters[key] = counter + 1
+        return _splitmix64(counter)
+    _seed_fallback_counter[0] += 1
+    return _splitmix64(_seed_fallback_counter[0])
 
 
 # ============================================================================

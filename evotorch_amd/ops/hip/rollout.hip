@@ -232,22 +232,24 @@ __global__ __launch_bounds__(256, 2) void rollout_linear_kernel(RolloutArgs args
         // phase 2: per-thread dynamics rows (padded-stride LDS reads) with
         // the NEXT step's normalization fused into the epilogue
         for (int j = tid; j < O; j += blockDim.x) {
-            float vacc = c_l[j];
+            // two independent dot chains (U·h and D2·a) halve the serial
+            // dot2 dependency depth; joined at the end
+            float uacc = c_l[j], dacc = 0.0f;
             const __bf16* urow = U_row + j * R_PAD;
             const __bf16* drow = D2_row + j * A_PAD;
 #pragma unroll 4
             for (int p = 0; p < R_PAD / 2; ++p) {
-                vacc = __builtin_amdgcn_fdot2_f32_bf16(
+                uacc = __builtin_amdgcn_fdot2_f32_bf16(
                     *reinterpret_cast<const bf16x2*>(urow + 2 * p),
-                    *reinterpret_cast<const bf16x2*>(h_b + 2 * p), vacc, false);
+                    *reinterpret_cast<const bf16x2*>(h_b + 2 * p), uacc, false);
             }
 #pragma unroll 3
             for (int p = 0; p < A_PAD / 2; ++p) {
-                vacc = __builtin_amdgcn_fdot2_f32_bf16(
+                dacc = __builtin_amdgcn_fdot2_f32_bf16(
                     *reinterpret_cast<const bf16x2*>(drow + 2 * p),
-                    *reinterpret_cast<const bf16x2*>(act_b + 2 * p), vacc, false);
+                    *reinterpret_cast<const bf16x2*>(act_b + 2 * p), dacc, false);
             }
-            const float o_new = tanhf(vacc);
+            const float o_new = tanhf(uacc + dacc);
             fit_part = fmaf(wr_l[j], o_new, fit_part);
             const int slot = j >= (int)blockDim.x;
             stat_sum[slot] += o_new;
