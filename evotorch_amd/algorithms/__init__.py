@@ -1,17 +1,35 @@
 """Search algorithms (L5). Reference parity:
 /root/reference/src/evotorch/algorithms/__init__.py."""
 
+from .cmaes import CMAES
+from .ga import Cosyne, GeneticAlgorithm, SteadyStateGA
 from .gaussian import CEM, PGPE, SNES, XNES, GaussianSearchAlgorithm
+from .mapelites import MAPElites
+from .restarter import IPOP, ModifyingRestart, Restart
 from .searchalgorithm import LazyReporter, LazyStatusDict, SearchAlgorithm, SinglePopulationAlgorithmMixin
+
+try:
+    from .pycmaes import PyCMAES  # requires the optional `cma` package at use time
+except ImportError:  # pragma: no cover
+    PyCMAES = None
 
 __all__ = [
     "CEM",
-    "PGPE",
-    "SNES",
-    "XNES",
+    "CMAES",
+    "Cosyne",
     "GaussianSearchAlgorithm",
+    "GeneticAlgorithm",
+    "IPOP",
     "LazyReporter",
     "LazyStatusDict",
+    "MAPElites",
+    "ModifyingRestart",
+    "PGPE",
+    "PyCMAES",
+    "Restart",
+    "SNES",
     "SearchAlgorithm",
     "SinglePopulationAlgorithmMixin",
+    "SteadyStateGA",
+    "XNES",
 ]

@@ -1,0 +1,253 @@
+"""CMA-ES: native torch implementation (Hansen 2016 tutorial formulas).
+
+Reference parity: /root/reference/src/evotorch/algorithms/cmaes.py:90-567 —
+rank-μ + rank-1 covariance update, active CMA, separable mode, CSA
+step-size control, and Cholesky (not eigen) decomposition for sampling.
+
+MI355X notes: sampling is X = m + σ·(Z Aᵀ) and the rank-μ update is
+(√w ⊙ Y)ᵀ(√w ⊙ Y) — both plain dense GEMMs served by rocBLAS (K5 in
+SURVEY.md §2.9; library GEMMs are the right tool for unfused dense
+products). The Cholesky factorization (d ≤ ~8k) runs on rocSOLVER through
+`torch.linalg.cholesky`. The whitened evolution path uses z_w = Σ w_i z_i
+directly (A⁻¹y_w = z_w), avoiding any triangular solve.
+"""
+
+import math
+from typing import Optional
+
+import torch
+
+from ..core import Problem, SolutionBatch
+from ..utils import RealOrVector, to_stdev_init
+from ..utils.misc import ensure_tensor_length_and_dtype
+from .searchalgorithm import SearchAlgorithm, SinglePopulationAlgorithmMixin
+
+__all__ = ["CMAES"]
+
+
+class CMAES(SearchAlgorithm, SinglePopulationAlgorithmMixin):
+    def __init__(
+        self,
+        problem: Problem,
+        *,
+        stdev_init: Optional[RealOrVector] = None,
+        radius_init: Optional[RealOrVector] = None,
+        popsize: Optional[int] = None,
+        center_init: Optional[RealOrVector] = None,
+        c_m: float = 1.0,
+        c_sigma: Optional[float] = None,
+        c_sigma_ratio: float = 1.0,
+        damp_sigma: Optional[float] = None,
+        damp_sigma_ratio: float = 1.0,
+        c_c: Optional[float] = None,
+        c_c_ratio: float = 1.0,
+        c_1: Optional[float] = None,
+        c_1_ratio: float = 1.0,
+        c_mu: Optional[float] = None,
+        c_mu_ratio: float = 1.0,
+        active: bool = True,
+        csa_squared: bool = False,
+        stdev_min: Optional[float] = None,
+        stdev_max: Optional[float] = None,
+        separable: bool = False,
+        limit_C_decomposition: bool = True,
+        obj_index: Optional[int] = None,
+    ):
+        problem.ensure_numeric()
+        problem.ensure_unbounded()
+        SearchAlgorithm.__init__(
+            self,
+            problem,
+            center=lambda: self._m,
+            stdev=lambda: self._sigma * (self._sigma_diag() if self._separable else torch.sqrt(torch.diagonal(self._C))),
+            sigma=lambda: float(self._sigma),
+            mean_eval=self._get_mean_eval,
+        )
+        n = problem.solution_length
+        self._n = n
+        self._obj_index = 0 if obj_index is None else int(obj_index)
+        device = problem.device
+        dtype = problem.dtype
+
+        stdev_spec = to_stdev_init(solution_length=n, stdev_init=stdev_init, radius_init=radius_init)
+        sigma_vec = ensure_tensor_length_and_dtype(stdev_spec, n, dtype, about="stdev_init", device=device)
+        self._sigma = torch.as_tensor(float(sigma_vec.mean()), dtype=dtype, device=device)
+        init_scale = (sigma_vec / self._sigma) ** 2  # fold anisotropy into C
+
+        if center_init is None:
+            self._m = problem.generate_values(1).reshape(-1)
+        else:
+            self._m = ensure_tensor_length_and_dtype(center_init, n, dtype, about="center_init", device=device)
+
+        if popsize is None:
+            popsize = int(4 + math.floor(3 * math.log(n)))
+        self._popsize = int(popsize)
+        mu = self._popsize // 2
+
+        # recombination weights (Hansen 2016 eq. 49-53; active CMA keeps the
+        # negative tail)
+        raw = torch.tensor([math.log((self._popsize + 1) / 2.0) - math.log(i + 1) for i in range(self._popsize)], dtype=dtype, device=device)
+        pos = raw[:mu]
+        self._mu_eff = float(pos.sum() ** 2 / (pos**2).sum())
+        w_pos = pos / pos.sum()
+
+        self._c_sigma = c_sigma if c_sigma is not None else c_sigma_ratio * (self._mu_eff + 2.0) / (n + self._mu_eff + 5.0)
+        self._damp_sigma = (
+            damp_sigma
+            if damp_sigma is not None
+            else damp_sigma_ratio * (1.0 + 2.0 * max(0.0, math.sqrt((self._mu_eff - 1.0) / (n + 1.0)) - 1.0) + self._c_sigma)
+        )
+        self._c_c = c_c if c_c is not None else c_c_ratio * (4.0 + self._mu_eff / n) / (n + 4.0 + 2.0 * self._mu_eff / n)
+        alpha_cov = 2.0
+        self._c_1 = c_1 if c_1 is not None else c_1_ratio * alpha_cov / ((n + 1.3) ** 2 + self._mu_eff)
+        self._c_mu = (
+            c_mu
+            if c_mu is not None
+            else c_mu_ratio
+            * min(
+                1.0 - self._c_1,
+                alpha_cov * (0.25 + self._mu_eff + 1.0 / self._mu_eff - 2.0) / ((n + 2.0) ** 2 + alpha_cov * self._mu_eff / 2.0),
+            )
+        )
+        self._active = bool(active)
+        if self._active:
+            neg = raw[mu:]
+            mu_eff_neg = float(neg.sum() ** 2 / (neg**2).sum())
+            alpha_mu_neg = 1.0 + self._c_1 / self._c_mu
+            alpha_mueff_neg = 1.0 + 2.0 * mu_eff_neg / (self._mu_eff + 2.0)
+            alpha_posdef_neg = (1.0 - self._c_1 - self._c_mu) / (n * self._c_mu)
+            w_neg = neg / neg.abs().sum() * min(alpha_mu_neg, alpha_mueff_neg, alpha_posdef_neg)
+            self._weights = torch.cat([w_pos, w_neg])
+        else:
+            self._weights = torch.cat([w_pos, torch.zeros(self._popsize - mu, dtype=dtype, device=device)])
+        self._mu = mu
+        self._c_m = float(c_m)
+        self._csa_squared = bool(csa_squared)
+        self._stdev_min = stdev_min
+        self._stdev_max = stdev_max
+        self._separable = bool(separable)
+
+        self._p_sigma = torch.zeros(n, dtype=dtype, device=device)
+        self._p_c = torch.zeros(n, dtype=dtype, device=device)
+        if self._separable:
+            self._C = init_scale.clone()  # diagonal
+            self._A = torch.sqrt(self._C)
+        else:
+            self._C = torch.diag(init_scale.clone())
+            self._A = torch.diag(torch.sqrt(init_scale.clone()))
+        self._chi_n = math.sqrt(n) * (1.0 - 1.0 / (4.0 * n) + 1.0 / (21.0 * n * n))
+        # amortize the O(n^3) Cholesky (reference limit_C_decomposition)
+        self._decompose_interval = max(1, int(1.0 / ((self._c_1 + self._c_mu) * n * 10.0))) if limit_C_decomposition else 1
+        self._steps_since_decompose = 0
+        self._population: Optional[SolutionBatch] = None
+        SinglePopulationAlgorithmMixin.__init__(self, exclude={"mean_eval"})
+
+    # -- properties ---------------------------------------------------------
+
+    @property
+    def population(self) -> Optional[SolutionBatch]:
+        return self._population
+
+    @property
+    def obj_index(self) -> int:
+        return self._obj_index
+
+    def _sigma_diag(self):
+        return torch.sqrt(self._C) if self._separable else torch.sqrt(torch.diagonal(self._C))
+
+    def _get_mean_eval(self):
+        if self._population is None:
+            return None
+        return float(torch.mean(torch.Tensor.as_subclass(self._population.evals, torch.Tensor)[:, self._obj_index]))
+
+    # -- stepping -----------------------------------------------------------
+
+    def _sample(self):
+        n, lam = self._n, self._popsize
+        problem = self.problem
+        g = problem.generator
+        z = torch.empty((lam, n), dtype=self._m.dtype, device=self._m.device)
+        z.normal_(generator=g if (g is not None and g.device == z.device) else None)
+        if self._separable:
+            y = z * self._A  # A is the diagonal sqrt
+        else:
+            y = z @ self._A.T  # y ~ N(0, C) with C = A Aᵀ
+        return z, y, self._m + self._sigma * y
+
+    def _step(self):
+        problem = self.problem
+        n, lam = self._n, self._popsize
+        z, y, x = self._sample()
+        batch = SolutionBatch(problem, popsize=lam, device=self._m.device, empty=True)
+        batch.access_values().copy_(x)
+        problem.evaluate(batch)
+        self._population = batch
+        order = batch.argsort(obj_index=self._obj_index)
+        z = z[order]
+        y = y[order]
+
+        w = self._weights
+        w_pos = w[: self._mu]
+        y_w = w_pos @ y[: self._mu]
+        z_w = w_pos @ z[: self._mu]
+
+        # mean update
+        self._m = self._m + self._c_m * self._sigma * y_w
+
+        # step-size control (CSA)
+        cs = self._c_sigma
+        self._p_sigma = (1.0 - cs) * self._p_sigma + math.sqrt(cs * (2.0 - cs) * self._mu_eff) * z_w
+        ps_norm = torch.linalg.vector_norm(self._p_sigma)
+        if self._csa_squared:
+            exponent = (cs / self._damp_sigma) * ((ps_norm**2 / n) - 1.0) / 2.0
+        else:
+            exponent = (cs / self._damp_sigma) * (ps_norm / self._chi_n - 1.0)
+        self._sigma = self._sigma * torch.exp(torch.clamp(exponent, max=1.0))
+        if self._stdev_min is not None:
+            self._sigma = torch.clamp(self._sigma, min=self._stdev_min)
+        if self._stdev_max is not None:
+            self._sigma = torch.clamp(self._sigma, max=self._stdev_max)
+
+        # covariance update
+        cc = self._c_c
+        hs = float(ps_norm / math.sqrt(1.0 - (1.0 - cs) ** (2 * (self._steps_count + 1)))) < (1.4 + 2.0 / (n + 1.0)) * self._chi_n
+        hs_f = 1.0 if hs else 0.0
+        self._p_c = (1.0 - cc) * self._p_c + hs_f * math.sqrt(cc * (2.0 - cc) * self._mu_eff) * y_w
+        c1, cmu = self._c_1, self._c_mu
+        delta_hs = (1.0 - hs_f) * cc * (2.0 - cc)
+        # active CMA: negative weights are scaled by n/||z_i||^2
+        w_adj = w.clone()
+        if self._active:
+            neg_mask = w < 0
+            z_norm_sq = (z**2).sum(dim=-1)
+            w_adj = torch.where(neg_mask, w * n / torch.clamp(z_norm_sq, min=1e-12), w)
+        if self._separable:
+            rank_mu = (w_adj.unsqueeze(-1) * y**2).sum(dim=0)
+            self._C = (1.0 + c1 * delta_hs - c1 - cmu * w_adj.sum()) * self._C + c1 * self._p_c**2 + cmu * rank_mu
+            self._C = torch.clamp(self._C, min=1e-20)
+            self._A = torch.sqrt(self._C)
+        else:
+            # rank-μ as a GEMM: Yᵀ diag(w) Y  (rocBLAS; MFMA-shaped)
+            rank_mu = (y * w_adj.unsqueeze(-1)).T @ y
+            rank_one = torch.outer(self._p_c, self._p_c)
+            self._C = (1.0 + c1 * delta_hs - c1 - cmu * w_adj.sum()) * self._C + c1 * rank_one + cmu * rank_mu
+            self._C = 0.5 * (self._C + self._C.T)
+            self._steps_since_decompose += 1
+            if self._steps_since_decompose >= self._decompose_interval:
+                self._A = self._cholesky(self._C)
+                self._steps_since_decompose = 0
+
+    def _cholesky(self, C: torch.Tensor) -> torch.Tensor:
+        try:
+            return torch.linalg.cholesky(C)
+        except Exception:
+            # regularize on failure
+            n = C.shape[0]
+            jitter = 1e-12
+            eye = torch.eye(n, dtype=C.dtype, device=C.device)
+            for _ in range(12):
+                try:
+                    return torch.linalg.cholesky(C + jitter * eye)
+                except Exception:
+                    jitter *= 10.0
+            raise

@@ -1,0 +1,121 @@
+"""Restart meta-algorithms: Restart, ModifyingRestart, IPOP.
+
+Reference parity: /root/reference/src/evotorch/algorithms/restarter/
+(restart.py:21, modify_restart.py:23,34). The inner searcher is
+re-instantiated whenever it reports termination; IPOP doubles the
+population size on each restart (Auger & Hansen 2005).
+"""
+
+from typing import Callable, Optional
+
+import torch
+
+from ..core import Problem
+from .searchalgorithm import SearchAlgorithm
+
+__all__ = ["Restart", "ModifyingRestart", "IPOP"]
+
+
+class Restart(SearchAlgorithm):
+    """Repeatedly runs a searcher built by `algorithm_factory`; when
+    `termination_criterion(searcher) -> bool` fires (default: the inner
+    searcher exposes a truthy status item "terminated", or its stdev
+    collapsed below `min_stdev`), a fresh searcher is created."""
+
+    def __init__(
+        self,
+        problem: Problem,
+        algorithm_factory: Callable[..., SearchAlgorithm],
+        *,
+        algorithm_args: Optional[dict] = None,
+        termination_criterion: Optional[Callable[[SearchAlgorithm], bool]] = None,
+        min_stdev: Optional[float] = None,
+        max_inner_steps: Optional[int] = None,
+    ):
+        super().__init__(
+            problem,
+            num_restarts=lambda: self._num_restarts,
+            inner_iter=lambda: self._inner_steps,
+        )
+        self._factory = algorithm_factory
+        self._algorithm_args = dict(algorithm_args or {})
+        self._criterion = termination_criterion
+        self._min_stdev = min_stdev
+        self._max_inner_steps = max_inner_steps
+        self._num_restarts = 0
+        self._inner_steps = 0
+        self._searcher: Optional[SearchAlgorithm] = None
+
+    @property
+    def search_algorithm(self) -> Optional[SearchAlgorithm]:
+        return self._searcher
+
+    @property
+    def num_restarts(self) -> int:
+        return self._num_restarts
+
+    def _current_args(self) -> dict:
+        return dict(self._algorithm_args)
+
+    def _make_searcher(self) -> SearchAlgorithm:
+        return self._factory(self.problem, **self._current_args())
+
+    def _terminated(self) -> bool:
+        s = self._searcher
+        if self._criterion is not None and self._criterion(s):
+            return True
+        if "terminated" in s.status and bool(s.status["terminated"]):
+            return True
+        if self._min_stdev is not None and "stdev" in s.status:
+            stdev = s.status["stdev"]
+            if isinstance(stdev, torch.Tensor) and float(torch.Tensor.as_subclass(stdev, torch.Tensor).max()) < self._min_stdev:
+                return True
+        if self._max_inner_steps is not None and self._inner_steps >= self._max_inner_steps:
+            return True
+        return False
+
+    def _on_restart(self):
+        """Hook for subclasses: adjust `self._algorithm_args` before the
+        next inner searcher is created."""
+
+    def _step(self):
+        if self._searcher is None:
+            self._searcher = self._make_searcher()
+            self._inner_steps = 0
+        self._searcher.step()
+        self._inner_steps += 1
+        for k, v in self._searcher.status.items():
+            if k != "iter":
+                self.update_status({k: v})
+        if self._terminated():
+            self._num_restarts += 1
+            self._on_restart()
+            self._searcher = None
+
+
+class ModifyingRestart(Restart):
+    """Restart variant that calls `modify(self)` before each restart to
+    adjust the inner algorithm's arguments (reference
+    modify_restart.py:23)."""
+
+    def __init__(self, problem, algorithm_factory, *, modify: Optional[Callable[["ModifyingRestart"], None]] = None, **kwargs):
+        super().__init__(problem, algorithm_factory, **kwargs)
+        self._modify = modify
+
+    def _on_restart(self):
+        if self._modify is not None:
+            self._modify(self)
+
+
+class IPOP(ModifyingRestart):
+    """IPOP restart strategy: double `popsize` on every restart
+    (reference modify_restart.py:34)."""
+
+    def __init__(self, problem, algorithm_factory, *, popsize_multiplier: float = 2.0, **kwargs):
+        super().__init__(problem, algorithm_factory, **kwargs)
+        self._popsize_multiplier = float(popsize_multiplier)
+
+    def _on_restart(self):
+        if "popsize" in self._algorithm_args and self._algorithm_args["popsize"]:
+            self._algorithm_args["popsize"] = int(self._algorithm_args["popsize"] * self._popsize_multiplier)
+        super()._on_restart()
