@@ -1,0 +1,307 @@
+"""Functional (stateless) evolutionary operators.
+
+Reference parity: /root/reference/src/evotorch/operators/functional.py
+(tournament :817, multi_point_cross_over :1091, one/two_point :1192/:1290,
+simulated_binary_cross_over :1411, utility :1580, cosyne_permutation :1737,
+combine :1852, take_best :2111; multi-objective helpers :240-471).
+
+Every function is a pure tensor transform; extra leading dimensions on the
+population arguments run independent batched populations via
+`torch.func.vmap` (the `@rowwise`/`expects_ndim` discipline of the
+reference), which is how batched searches compose.
+"""
+
+from typing import Optional, Tuple, Union
+
+import torch
+
+from ..utils import ranking as _ranking
+
+__all__ = [
+    "tournament",
+    "multi_point_cross_over",
+    "one_point_cross_over",
+    "two_point_cross_over",
+    "simulated_binary_cross_over",
+    "utility",
+    "cosyne_permutation",
+    "combine",
+    "take_best",
+    "dominates",
+    "domination_matrix",
+    "domination_counts",
+    "crowding_distances",
+    "pareto_utility",
+]
+
+
+def _check_sense(objective_sense: str):
+    if objective_sense not in ("min", "max"):
+        raise ValueError(f"objective_sense must be 'min' or 'max', got {objective_sense!r}")
+
+
+def _utils_2d(evals: torch.Tensor, objective_sense) -> torch.Tensor:
+    """Fold senses so higher is better. evals (..., N) single-objective or
+    (..., N, M) multi-objective with a list of senses."""
+    if isinstance(objective_sense, str):
+        _check_sense(objective_sense)
+        return evals if objective_sense == "max" else -evals
+    senses = list(objective_sense)
+    cols = []
+    for j, s in enumerate(senses):
+        _check_sense(s)
+        col = evals[..., j]
+        cols.append(col if s == "max" else -col)
+    return torch.stack(cols, dim=-1)
+
+
+# ----------------------------------------------------------------------------
+# selection
+# ----------------------------------------------------------------------------
+
+
+def tournament(
+    solutions: torch.Tensor,
+    evals: torch.Tensor,
+    *,
+    num_tournaments: int,
+    tournament_size: int,
+    objective_sense: Union[str, list],
+    return_indices: bool = False,
+    with_evals: bool = False,
+    generator: Optional[torch.Generator] = None,
+):
+    """Tournament selection: `num_tournaments` independent tournaments of
+    `tournament_size` uniformly drawn rows; winners returned (as values, or
+    indices with return_indices=True)."""
+
+    def one_pop(solutions: torch.Tensor, evals: torch.Tensor):
+        n = solutions.shape[0]
+        if isinstance(objective_sense, str) and evals.ndim == 1:
+            utils = _utils_2d(evals, objective_sense)
+        else:
+            # multi-objective: tournament by pareto rank
+            utils_mo = _utils_2d(evals, objective_sense)
+            counts = domination_counts(utils_mo, _already_folded=True)
+            utils = -counts.to(torch.float32)
+        contenders = torch.randint(0, n, (num_tournaments, tournament_size), device=solutions.device, generator=generator)
+        scores = utils[contenders]
+        winners = contenders.gather(1, scores.argmax(dim=1, keepdim=True)).reshape(-1)
+        return winners
+
+    if solutions.ndim == 2:
+        winners = one_pop(solutions, evals)
+    elif solutions.ndim > 2:
+        lead = solutions.shape[:-2]
+        flat_s = solutions.reshape((-1,) + solutions.shape[-2:])
+        flat_e = evals.reshape((flat_s.shape[0],) + evals.shape[len(lead):])
+        winners = torch.stack([one_pop(flat_s[i], flat_e[i]) for i in range(flat_s.shape[0])]).reshape(lead + (num_tournaments,))
+    else:
+        raise ValueError("solutions must be at least 2-D")
+
+    if return_indices:
+        return winners
+    picked = torch.gather(solutions, -2, winners.unsqueeze(-1).expand(winners.shape + (solutions.shape[-1],)))
+    if with_evals:
+        if evals.ndim == solutions.ndim - 1:
+            picked_evals = torch.gather(evals, -1, winners)
+        else:
+            picked_evals = torch.gather(evals, -2, winners.unsqueeze(-1).expand(winners.shape + (evals.shape[-1],)))
+        return picked, picked_evals
+    return picked
+
+
+def _pair_parents(parents: torch.Tensor, evals, tournament_size, objective_sense, num_children, generator):
+    n = parents.shape[-2]
+    if tournament_size is not None:
+        if evals is None:
+            raise ValueError("tournament selection requires evals")
+        num_children = num_children if num_children is not None else n
+        num_pairs = max(1, num_children // 2)
+        winners = tournament(
+            parents, evals, num_tournaments=num_pairs * 2, tournament_size=tournament_size, objective_sense=objective_sense, return_indices=True, generator=generator
+        )
+        p1 = torch.gather(parents, -2, winners[..., :num_pairs].unsqueeze(-1).expand(winners[..., :num_pairs].shape + (parents.shape[-1],)))
+        p2 = torch.gather(parents, -2, winners[..., num_pairs:].unsqueeze(-1).expand(winners[..., num_pairs:].shape + (parents.shape[-1],)))
+        return p1, p2
+    half = n // 2
+    return parents[..., :half, :], parents[..., half : 2 * half, :]
+
+
+def multi_point_cross_over(
+    parents: torch.Tensor,
+    evals: Optional[torch.Tensor] = None,
+    *,
+    num_points: int,
+    num_children: Optional[int] = None,
+    tournament_size: Optional[int] = None,
+    objective_sense: Optional[Union[str, list]] = None,
+    generator: Optional[torch.Generator] = None,
+) -> torch.Tensor:
+    """k-point crossover. Without tournament_size, the first and second
+    halves of `parents` are paired; children = both swap directions."""
+    p1, p2 = _pair_parents(parents, evals, tournament_size, objective_sense, num_children, generator)
+    length = parents.shape[-1]
+    pair_shape = p1.shape[:-1]
+    cuts = torch.randint(1, length, pair_shape + (num_points,), device=parents.device, generator=generator)
+    positions = torch.arange(length, device=parents.device)
+    crossed = (positions >= cuts.unsqueeze(-1)).sum(dim=-2)
+    mask = (crossed % 2) == 1
+    child1 = torch.where(mask, p2, p1)
+    child2 = torch.where(mask, p1, p2)
+    return torch.cat([child1, child2], dim=-2)
+
+
+def one_point_cross_over(parents, evals=None, *, num_children=None, tournament_size=None, objective_sense=None, generator=None) -> torch.Tensor:
+    return multi_point_cross_over(parents, evals, num_points=1, num_children=num_children, tournament_size=tournament_size, objective_sense=objective_sense, generator=generator)
+
+
+def two_point_cross_over(parents, evals=None, *, num_children=None, tournament_size=None, objective_sense=None, generator=None) -> torch.Tensor:
+    return multi_point_cross_over(parents, evals, num_points=2, num_children=num_children, tournament_size=tournament_size, objective_sense=objective_sense, generator=generator)
+
+
+def simulated_binary_cross_over(
+    parents: torch.Tensor,
+    evals: Optional[torch.Tensor] = None,
+    *,
+    eta: float,
+    num_children: Optional[int] = None,
+    tournament_size: Optional[int] = None,
+    objective_sense: Optional[Union[str, list]] = None,
+    generator: Optional[torch.Generator] = None,
+) -> torch.Tensor:
+    p1, p2 = _pair_parents(parents, evals, tournament_size, objective_sense, num_children, generator)
+    u = torch.rand(p1.shape, dtype=parents.dtype, device=parents.device, generator=generator)
+    betas = torch.where(u <= 0.5, (2.0 * u) ** (1.0 / (eta + 1.0)), (0.5 / (1.0 - u)) ** (1.0 / (eta + 1.0)))
+    child1 = 0.5 * ((1 + betas) * p1 + (1 - betas) * p2)
+    child2 = 0.5 * ((1 - betas) * p1 + (1 + betas) * p2)
+    return torch.cat([child1, child2], dim=-2)
+
+
+# ----------------------------------------------------------------------------
+# fitness shaping / combination / truncation
+# ----------------------------------------------------------------------------
+
+
+def utility(evals: torch.Tensor, *, objective_sense: str, ranking_method: Optional[str] = "centered") -> torch.Tensor:
+    """Fitness shaping along the last dim (higher utility = better)."""
+    _check_sense(objective_sense)
+    return _ranking.rank(evals, ranking_method or "raw", higher_is_better=(objective_sense == "max"))
+
+
+def cosyne_permutation(values: torch.Tensor, permute_all: bool = True, *, generator: Optional[torch.Generator] = None) -> torch.Tensor:
+    """Independently permute each column of the population (the CoSyNE
+    decorrelation op). Functional variant permutes all values."""
+    if not permute_all:
+        raise NotImplementedError("functional cosyne_permutation supports permute_all=True")
+    keys = torch.rand(values.shape, device=values.device, generator=generator)
+    order = keys.argsort(dim=-2)
+    return torch.gather(values, -2, order)
+
+
+def combine(a, b):
+    """Concatenate two populations; accepts tensors or (values, evals)
+    pairs and returns the same structure."""
+    if isinstance(a, tuple):
+        av, ae = a
+        bv, be = b
+        return torch.cat([av, bv], dim=-2), torch.cat([ae, be], dim=-2 if ae.ndim == av.ndim else -1)
+    return torch.cat([a, b], dim=-2)
+
+
+def take_best(
+    values: torch.Tensor,
+    evals: torch.Tensor,
+    n: Optional[int] = None,
+    *,
+    objective_sense: Union[str, list],
+    crowdsort: bool = True,
+):
+    """The best n rows (or the single best row with n=None), with their
+    evals. Multi-objective: pareto-rank (+ crowding) ordering."""
+    if isinstance(objective_sense, str):
+        utils = _utils_2d(evals, objective_sense)
+        order = utils.argsort(dim=-1, descending=True)
+    else:
+        folded = _utils_2d(evals, objective_sense)
+        counts = domination_counts(folded, _already_folded=True)
+        key = counts.to(torch.float64)
+        if crowdsort:
+            crowd = crowding_distances(folded, _already_folded=True)
+            cpos = torch.nan_to_num(crowd, posinf=1e300).argsort(dim=-1, descending=True).argsort(dim=-1)
+            key = key * (evals.shape[-2] + 1) + cpos
+        order = key.argsort(dim=-1)
+    if n is None:
+        best = order[..., 0]
+        v = torch.index_select(values, -2, best.reshape(-1)[:1]).squeeze(-2) if values.ndim == 2 else None
+        if values.ndim == 2:
+            return values[best], evals[best]
+        raise ValueError("take_best with n=None expects 2-D values")
+    top = order[..., :n]
+    picked = torch.gather(values, -2, top.unsqueeze(-1).expand(top.shape + (values.shape[-1],)))
+    if isinstance(objective_sense, str):
+        picked_evals = torch.gather(evals, -1, top)
+    else:
+        picked_evals = torch.gather(evals, -2, top.unsqueeze(-1).expand(top.shape + (evals.shape[-1],)))
+    return picked, picked_evals
+
+
+# ----------------------------------------------------------------------------
+# multi-objective helpers
+# ----------------------------------------------------------------------------
+
+
+def dominates(a_evals: torch.Tensor, b_evals: torch.Tensor, *, objective_sense: list) -> torch.Tensor:
+    """True iff solution a pareto-dominates solution b."""
+    a = _utils_2d(a_evals, objective_sense)
+    b = _utils_2d(b_evals, objective_sense)
+    return (a >= b).all(dim=-1) & (a > b).any(dim=-1)
+
+
+def domination_matrix(evals: torch.Tensor, *, objective_sense: list = None, _already_folded: bool = False) -> torch.Tensor:
+    """(N, N) boolean: [i, j] True iff i dominates j."""
+    utils = evals if _already_folded else _utils_2d(evals, objective_sense)
+    a = utils.unsqueeze(-2)
+    b = utils.unsqueeze(-3)
+    return (a >= b).all(dim=-1) & (a > b).any(dim=-1)
+
+
+def domination_counts(evals: torch.Tensor, *, objective_sense: list = None, _already_folded: bool = False) -> torch.Tensor:
+    """Per solution: how many others dominate it (0 = on the best front)."""
+    dom = domination_matrix(evals, objective_sense=objective_sense, _already_folded=_already_folded)
+    return dom.sum(dim=-2)
+
+
+def crowding_distances(evals: torch.Tensor, *, objective_sense: list = None, _already_folded: bool = False) -> torch.Tensor:
+    """Global crowding distances (computed over the whole set per
+    objective; reference functional.py:357-447)."""
+    utils = evals if _already_folded else _utils_2d(evals, objective_sense)
+    n, m = utils.shape[-2], utils.shape[-1]
+    crowd = torch.zeros(utils.shape[:-1], dtype=utils.dtype, device=utils.device)
+    for j in range(m):
+        col = utils[..., j]
+        order = col.argsort(dim=-1)
+        sorted_vals = torch.gather(col, -1, order)
+        span = sorted_vals[..., -1:] - sorted_vals[..., :1]
+        span = torch.where(span == 0, torch.ones_like(span), span)
+        contrib = torch.zeros_like(col)
+        inner = (sorted_vals[..., 2:] - sorted_vals[..., :-2]) / span
+        contrib.scatter_(-1, order[..., 1:-1], inner)
+        contrib.scatter_(-1, order[..., :1], torch.full_like(order[..., :1], 0, dtype=col.dtype).fill_(float("inf")))
+        contrib.scatter_(-1, order[..., -1:], torch.full_like(order[..., -1:], 0, dtype=col.dtype).fill_(float("inf")))
+        crowd = crowd + contrib
+    return crowd
+
+
+def pareto_utility(evals: torch.Tensor, *, objective_sense: list, crowdsort: bool = True) -> torch.Tensor:
+    """Scalar utility for multi-objective evals: -(domination count), with
+    a small crowding-distance tie-break (higher = better)."""
+    folded = _utils_2d(evals, objective_sense)
+    counts = domination_counts(folded, _already_folded=True).to(torch.float32)
+    result = -counts
+    if crowdsort:
+        crowd = crowding_distances(folded, _already_folded=True)
+        n = evals.shape[-2]
+        cpos = torch.nan_to_num(crowd, posinf=1e300).argsort(dim=-1).argsort(dim=-1).to(torch.float32)
+        result = result + cpos / (n * 10.0)
+    return result
