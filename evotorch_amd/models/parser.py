@@ -1,0 +1,103 @@
+"""`str_to_net`: the tiny network-description DSL.
+
+Reference parity: /root/reference/src/evotorch/neuroevolution/net/
+parser.py:218 — Python-AST-parsed expressions like
+
+    "Linear(obs_length, 64) >> Tanh() >> Linear(64, act_length)"
+
+with named constants substituted, `>>` composing modules sequentially.
+Resolvable names: torch.nn modules (Linear, Tanh, ReLU, ...), the custom
+layers of evotorch_amd.models.layers, and any keyword constants passed in.
+"""
+
+import ast
+from typing import Any
+
+import torch
+from torch import nn
+
+from . import layers as _layers
+from .multilayered import MultiLayered
+
+__all__ = ["str_to_net", "NetParsingError"]
+
+
+class NetParsingError(Exception):
+    def __init__(self, message: str, node: ast.AST = None):
+        if node is not None:
+            message = f"{message} (at line {getattr(node, 'lineno', '?')}, col {getattr(node, 'col_offset', '?')})"
+        super().__init__(message)
+
+
+def _resolve_name(name: str, constants: dict, node) -> Any:
+    if name in constants:
+        return constants[name]
+    if hasattr(_layers, name):
+        return getattr(_layers, name)
+    if hasattr(nn, name):
+        return getattr(nn, name)
+    if hasattr(torch, name):
+        return getattr(torch, name)
+    raise NetParsingError(f"Cannot resolve name {name!r}", node)
+
+
+def _eval_node(node: ast.AST, constants: dict) -> Any:
+    if isinstance(node, ast.Expression):
+        return _eval_node(node.body, constants)
+    if isinstance(node, ast.BinOp):
+        if isinstance(node.op, ast.RShift):
+            left = _eval_node(node.left, constants)
+            right = _eval_node(node.right, constants)
+            return _compose(left, right)
+        left = _eval_node(node.left, constants)
+        right = _eval_node(node.right, constants)
+        ops = {ast.Add: lambda a, b: a + b, ast.Sub: lambda a, b: a - b, ast.Mult: lambda a, b: a * b, ast.Div: lambda a, b: a / b, ast.FloorDiv: lambda a, b: a // b, ast.Pow: lambda a, b: a**b, ast.Mod: lambda a, b: a % b}
+        for op_type, fn in ops.items():
+            if isinstance(node.op, op_type):
+                return fn(left, right)
+        raise NetParsingError(f"Unsupported operator {type(node.op).__name__}", node)
+    if isinstance(node, ast.UnaryOp):
+        operand = _eval_node(node.operand, constants)
+        if isinstance(node.op, ast.USub):
+            return -operand
+        if isinstance(node.op, ast.UAdd):
+            return +operand
+        raise NetParsingError(f"Unsupported unary operator {type(node.op).__name__}", node)
+    if isinstance(node, ast.Call):
+        if not isinstance(node.func, ast.Name):
+            raise NetParsingError("Only simple names can be called", node)
+        fn = _resolve_name(node.func.id, constants, node)
+        args = [_eval_node(a, constants) for a in node.args]
+        kwargs = {kw.arg: _eval_node(kw.value, constants) for kw in node.keywords}
+        return fn(*args, **kwargs)
+    if isinstance(node, ast.Name):
+        return _resolve_name(node.id, constants, node)
+    if isinstance(node, ast.Constant):
+        return node.value
+    if isinstance(node, (ast.List, ast.Tuple)):
+        items = [_eval_node(e, constants) for e in node.elts]
+        return items if isinstance(node, ast.List) else tuple(items)
+    if isinstance(node, ast.Attribute):
+        value = _eval_node(node.value, constants)
+        return getattr(value, node.attr)
+    raise NetParsingError(f"Unsupported syntax: {type(node).__name__}", node)
+
+
+def _compose(left, right) -> nn.Module:
+    if not isinstance(left, nn.Module) or not isinstance(right, nn.Module):
+        raise NetParsingError(">> requires nn.Module operands")
+    left_mods = list(left) if isinstance(left, MultiLayered) else [left]
+    right_mods = list(right) if isinstance(right, MultiLayered) else [right]
+    return MultiLayered(*(left_mods + right_mods))
+
+
+def str_to_net(s: str, **constants) -> nn.Module:
+    """Parse a network DSL string into an nn.Module."""
+    try:
+        tree = ast.parse(s.strip(), mode="eval")
+    except SyntaxError as e:
+        raise NetParsingError(f"Could not parse network string: {e}") from e
+    result = _eval_node(tree, dict(constants))
+    if not isinstance(result, nn.Module):
+        raise NetParsingError(f"Expression did not produce an nn.Module but {type(result)}")
+    return result

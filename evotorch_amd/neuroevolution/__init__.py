@@ -1,22 +1,32 @@
 """Neuroevolution problem domains (L3). Reference parity:
 /root/reference/src/evotorch/neuroevolution/__init__.py."""
 
+from .neproblem import NEProblem
 from .runningnorm import ObsNormLayer, RunningNorm
+from .supervisedne import SupervisedNE
 from .synthetic import SyntheticRolloutProblem
 from .synthetic_env import SyntheticEnvSpec, rollout_eager
+from .vecenv import GymVectorEnvAdapter, SyntheticTorchEnv, VecEnvNE, VecGymNE
 
 __all__ = [
+    "GymVectorEnvAdapter",
+    "NEProblem",
     "ObsNormLayer",
     "RunningNorm",
+    "SupervisedNE",
     "SyntheticEnvSpec",
     "SyntheticRolloutProblem",
+    "SyntheticTorchEnv",
+    "VecEnvNE",
+    "VecGymNE",
     "rollout_eager",
 ]
 
-try:  # full NEProblem family lands with the net/ infrastructure
-    from .neproblem import NEProblem  # noqa: F401
-    from .supervisedne import SupervisedNE  # noqa: F401
 
-    __all__ += ["NEProblem", "SupervisedNE"]
-except ImportError:
-    pass
+def __getattr__(name):
+    # GymNE needs gymnasium; import lazily so the package works offline
+    if name == "GymNE":
+        from .gymne import GymNE
+
+        return GymNE
+    raise AttributeError(f"module {__name__!r} has no attribute {name!r}")

@@ -1,0 +1,174 @@
+"""GymNE: classic per-solution episode rollouts in (gymnasium) envs.
+
+Reference parity: /root/reference/src/evotorch/neuroevolution/
+gymne.py:64-730 — one env per worker, per-solution rollout loop, CPU-side
+observation normalization (RunningNorm here plays the RunningStat role),
+alive-bonus removal via `decrease_rewards_by`, episode/interaction
+counters, and `to_policy` exporting ObsNorm + ActClip wrapped modules.
+
+gymnasium is an optional dependency (absent in the offline image); the
+import is deferred to construction time.
+"""
+
+from typing import Callable, Optional, Union
+
+import torch
+from torch import nn
+
+from ..core import Solution, SolutionBatch
+from ..models import ensure_stateful
+from .neproblem import NEProblem
+from .runningnorm import ObsNormLayer, RunningNorm
+
+__all__ = ["GymNE", "ActClipLayer"]
+
+
+class ActClipLayer(nn.Module):
+    """Clips actions into the env's action-space box (reference
+    net/rl.py:130)."""
+
+    def __init__(self, lb, ub):
+        super().__init__()
+        self.register_buffer("lb", torch.as_tensor(lb, dtype=torch.float32))
+        self.register_buffer("ub", torch.as_tensor(ub, dtype=torch.float32))
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return torch.clamp(x, self.lb, self.ub)
+
+
+class GymNE(NEProblem):
+    def __init__(
+        self,
+        env: Optional[Union[str, Callable]] = None,
+        network: Optional[Union[str, nn.Module, Callable]] = None,
+        *,
+        env_name: Optional[str] = None,
+        network_args: Optional[dict] = None,
+        env_config: Optional[dict] = None,
+        observation_normalization: bool = False,
+        num_episodes: int = 1,
+        episode_length: Optional[int] = None,
+        decrease_rewards_by: Optional[float] = None,
+        alive_bonus_schedule: Optional[tuple] = None,
+        action_noise_stdev: Optional[float] = None,
+        initial_bounds=(-0.00001, 0.00001),
+        seed: Optional[int] = None,
+    ):
+        try:
+            import gymnasium  # noqa: F401
+        except ImportError as e:
+            raise ImportError("GymNE requires gymnasium, which is not installed in this image; use VecEnvNE with a torch-native env instead") from e
+        if env is None:
+            env = env_name
+        if env is None:
+            raise ValueError("Provide env (or env_name)")
+        self._env_def = env
+        self._env_config = dict(env_config or {})
+        self._env = None
+        probe = self._make_env()
+        self._obs_dim = int(probe.observation_space.shape[0])
+        self._act_dim = int(probe.action_space.shape[0]) if hasattr(probe.action_space, "shape") and probe.action_space.shape else int(probe.action_space.n)
+        self._act_space = probe.action_space
+        self._env = probe
+        self._num_episodes = int(num_episodes)
+        self._episode_length = episode_length
+        self._decrease_rewards_by = 0.0 if decrease_rewards_by is None else float(decrease_rewards_by)
+        self._alive_bonus_schedule = alive_bonus_schedule
+        self._action_noise_stdev = action_noise_stdev
+        self._obs_norm_enabled = bool(observation_normalization)
+        super().__init__("max", network, network_args=network_args, initial_bounds=initial_bounds, seed=seed, store_solution_stats=False)
+        self._obs_norm = RunningNorm(shape=self._obs_dim, device="cpu")
+        self.last_eval_interaction_count = 0
+        self._total_interactions = 0
+        self._episode_count = 0
+        self.after_eval_hook.append(
+            lambda b: {"total_interaction_count": self._total_interactions, "total_episode_count": self._episode_count}
+        )
+
+    def _network_constants(self) -> dict:
+        return {"obs_length": self._obs_dim, "act_length": self._act_dim, "obs_space": getattr(self._env, "observation_space", None), "act_space": self._act_space}
+
+    def _make_env(self):
+        import gymnasium as gym
+
+        if isinstance(self._env_def, str):
+            return gym.make(self._env_def, **self._env_config)
+        return self._env_def(**self._env_config)
+
+    @property
+    def obs_norm(self) -> RunningNorm:
+        return self._obs_norm
+
+    def observation_normalization_data(self):
+        return {"mean": self._obs_norm.mean, "stdev": self._obs_norm.stdev, "count": self._obs_norm.count}
+
+    def _alive_bonus(self, t: int) -> float:
+        if self._alive_bonus_schedule is None:
+            return 0.0
+        t0, t1, bonus = self._alive_bonus_schedule
+        if t < t0:
+            return 0.0
+        if t >= t1:
+            return float(bonus)
+        return float(bonus) * (t - t0) / max(1, (t1 - t0))
+
+    def _rollout(self, policy: nn.Module) -> float:
+        import numpy as np
+
+        env = self._env
+        total = 0.0
+        obs, _ = env.reset(seed=None)
+        t = 0
+        policy = ensure_stateful(policy)
+        policy.reset()
+        while True:
+            obs_t = torch.as_tensor(np.asarray(obs), dtype=torch.float32)
+            if self._obs_norm_enabled:
+                obs_in = self._obs_norm.update_and_normalize(obs_t)
+            else:
+                obs_in = obs_t
+            with torch.no_grad():
+                act = policy(obs_in)
+            if self._action_noise_stdev is not None:
+                act = act + torch.randn_like(act) * self._action_noise_stdev
+            if hasattr(self._act_space, "low"):
+                act = torch.clamp(act, torch.as_tensor(self._act_space.low), torch.as_tensor(self._act_space.high))
+                action = np.asarray(act)
+            else:
+                action = int(act.argmax())
+            obs, reward, terminated, truncated, _ = env.step(action)
+            total += float(reward) - self._decrease_rewards_by + self._alive_bonus(t)
+            t += 1
+            self._total_interactions += 1
+            self.last_eval_interaction_count += 1
+            if terminated or truncated:
+                break
+            if self._episode_length is not None and t >= self._episode_length:
+                break
+        self._episode_count += 1
+        return total
+
+    def _evaluate_network(self, network: nn.Module) -> float:
+        total = 0.0
+        for _ in range(self._num_episodes):
+            total += self._rollout(network)
+        return total / self._num_episodes
+
+    def _evaluate_batch(self, batch: SolutionBatch):
+        self.last_eval_interaction_count = 0
+        super()._evaluate_batch(batch)
+
+    def run(self, solution, *, render: bool = False) -> float:
+        """Roll out one solution (optionally rendering)."""
+        net = self.parameterize_net(torch.as_tensor(solution if not isinstance(solution, Solution) else torch.Tensor.as_subclass(solution.values, torch.Tensor)))
+        return self._rollout(net)
+
+    def to_policy(self, x, *, clip_actions: bool = True) -> nn.Module:
+        module = self.make_net(x)
+        layers = []
+        if self._obs_norm_enabled and self._obs_norm.has_data:
+            layers.append(self._obs_norm.to_layer())
+        layers.append(module)
+        if clip_actions and hasattr(self._act_space, "low"):
+            layers.append(ActClipLayer(self._act_space.low, self._act_space.high))
+        return nn.Sequential(*layers)

@@ -1,0 +1,267 @@
+"""Vectorized torch-native environments + the VecEnvNE problem (the
+VecGymNE-equivalent).
+
+Reference parity: /root/reference/src/evotorch/neuroevolution/
+vecgymne.py:95-1073 and net/vecrl.py (TorchWrapper :362, SyncVectorEnv
+:1541). Since no simulator ships offline, the environment contract here is
+torch-native from the start:
+
+    env.reset(seed) -> obs (num_envs, obs_dim)
+    env.step(actions) -> (obs, reward, done)          all torch tensors
+
+`SyntheticTorchEnv` implements the contract over `SyntheticEnvSpec`;
+`GymVectorEnvAdapter` (import-guarded) wraps a gymnasium vector env. The
+VecEnvNE hot loop is SURVEY.md §3.4: one vmapped population forward per
+simulator step, masked reward accumulation, masked recurrent-state resets,
+on-device RunningNorm.
+"""
+
+from typing import Callable, Optional, Union
+
+import torch
+from torch import nn
+
+from ..core import SolutionBatch
+from ..models import Policy
+from .neproblem import NEProblem
+from .runningnorm import ObsNormLayer, RunningNorm
+from .synthetic_env import SyntheticEnvSpec
+
+__all__ = ["SyntheticTorchEnv", "GymVectorEnvAdapter", "VecEnvNE", "VecGymNE"]
+
+
+class SyntheticTorchEnv:
+    """Batched torch env over the synthetic low-rank dynamics."""
+
+    def __init__(self, spec: Optional[SyntheticEnvSpec] = None, *, num_envs: int, device="cpu", **spec_kwargs):
+        self.spec = spec if spec is not None else SyntheticEnvSpec(**spec_kwargs)
+        self.num_envs = int(num_envs)
+        self.device = torch.device(device)
+        self.obs_dim = self.spec.obs_dim
+        self.act_dim = self.spec.act_dim
+        s = self.spec
+        self._V = s.V.to(self.device)
+        self._U_T = s.U_T.to(self.device)
+        self._D2_T = s.D2_T.to(self.device)
+        self._c = s.c.to(self.device)
+        self._wr = s.wr.to(self.device)
+        self._obs: Optional[torch.Tensor] = None
+        self._t = 0
+
+    def reset(self, seed: int = 0) -> torch.Tensor:
+        self._obs = self.spec.initial_obs(self.num_envs, 0, int(seed), device=self.device)
+        self._t = 0
+        return self._obs
+
+    def step(self, actions: torch.Tensor):
+        s = self.spec
+        act = torch.clamp(actions.to(self.device), -1.0, 1.0)
+        h = self._obs @ self._V.T
+        o_new = torch.tanh(h @ self._U_T + act @ self._D2_T + self._c)
+        reward = o_new @ self._wr + s.alive_bonus - s.act_cost * (act**2).sum(-1) / s.act_dim
+        self._obs = o_new
+        self._t += 1
+        done = torch.full((self.num_envs,), self._t >= s.episode_length, dtype=torch.bool, device=self.device)
+        return o_new, reward, done
+
+
+class GymVectorEnvAdapter:
+    """Wraps a gymnasium (Sync/Async)VectorEnv into the torch contract
+    (reference net/vecrl.py:362 TorchWrapper). Import-guarded: gymnasium
+    is optional and absent in the offline image."""
+
+    def __init__(self, env, *, device="cpu"):
+        import numpy as np  # noqa: F401
+
+        self._env = env
+        self.device = torch.device(device)
+        self.num_envs = env.num_envs
+        self.obs_dim = int(env.single_observation_space.shape[0])
+        self.act_dim = int(env.single_action_space.shape[0])
+
+    def reset(self, seed: int = 0) -> torch.Tensor:
+        obs, _ = self._env.reset(seed=int(seed))
+        return torch.as_tensor(obs, dtype=torch.float32, device=self.device)
+
+    def step(self, actions: torch.Tensor):
+        import numpy as np
+
+        obs, reward, terminated, truncated, _ = self._env.step(np.asarray(actions.detach().cpu()))
+        return (
+            torch.as_tensor(obs, dtype=torch.float32, device=self.device),
+            torch.as_tensor(reward, dtype=torch.float32, device=self.device),
+            torch.as_tensor(np.logical_or(terminated, truncated), dtype=torch.bool, device=self.device),
+        )
+
+
+class VecEnvNE(NEProblem):
+    """Whole-population vectorized rollouts: one env row per solution, one
+    vmapped policy forward per step."""
+
+    def __init__(
+        self,
+        env: Union[Callable[[int], object], object, str],
+        network: Union[str, nn.Module, Callable[[], nn.Module]],
+        *,
+        network_args: Optional[dict] = None,
+        max_num_steps: Optional[int] = None,
+        observation_normalization: bool = True,
+        decrease_rewards_by: float = 0.0,
+        alive_bonus_schedule: Optional[tuple] = None,
+        device=None,
+        seed: Optional[int] = None,
+        initial_bounds=(-0.00001, 0.00001),
+    ):
+        if isinstance(env, str):
+            env_name = env
+            def factory(num_envs: int):
+                import gymnasium as gym
+
+                return GymVectorEnvAdapter(gym.make_vec(env_name, num_envs=num_envs), device=device or "cpu")
+
+            self._env_factory = factory
+            self._env = None
+            probe = factory(1)
+            self._obs_dim, self._act_dim = probe.obs_dim, probe.act_dim
+        elif callable(env) and not hasattr(env, "step"):
+            self._env_factory = env
+            self._env = None
+            probe = env(1)
+            self._obs_dim, self._act_dim = probe.obs_dim, probe.act_dim
+        else:
+            self._env_factory = None
+            self._env = env
+            self._obs_dim, self._act_dim = env.obs_dim, env.act_dim
+
+        self._max_num_steps = max_num_steps
+        self._obs_norm_enabled = bool(observation_normalization)
+        self._decrease_rewards_by = float(decrease_rewards_by)
+        self._alive_bonus_schedule = alive_bonus_schedule
+        super().__init__(
+            "max",
+            network,
+            network_args=network_args,
+            initial_bounds=initial_bounds,
+            device=device,
+            seed=seed,
+            store_solution_stats=False,
+        )
+        self._obs_norm = RunningNorm(shape=self._obs_dim, device=self.network_device)
+        self._policy: Optional[Policy] = None
+        self.last_eval_interaction_count = 0
+        self._total_interactions = 0
+        self._episode_count = 0
+        self._pending_stats: Optional[RunningNorm] = None
+        self.after_eval_hook.append(self._counters_status)
+
+    # -- plumbing -------------------------------------------------------------
+
+    def _network_constants(self) -> dict:
+        return {"obs_length": self._obs_dim, "act_length": self._act_dim, "obs_space": None, "act_space": None}
+
+    @property
+    def obs_norm(self) -> RunningNorm:
+        return self._obs_norm
+
+    def observation_normalization_data(self):
+        return {"mean": self._obs_norm.mean.cpu(), "stdev": self._obs_norm.stdev.cpu(), "count": self._obs_norm.count}
+
+    def _get_env(self, num_envs: int):
+        if self._env is not None and self._env.num_envs == num_envs:
+            return self._env
+        if self._env_factory is None:
+            raise ValueError(f"The fixed env has {self._env.num_envs} rows but the batch needs {num_envs}; provide an env factory instead")
+        self._env = self._env_factory(num_envs)
+        return self._env
+
+    def _alive_bonus(self, t: int) -> float:
+        if self._alive_bonus_schedule is None:
+            return 0.0
+        t0, t1, bonus = self._alive_bonus_schedule
+        if t < t0:
+            return 0.0
+        if t >= t1:
+            return float(bonus)
+        return float(bonus) * (t - t0) / max(1, (t1 - t0))
+
+    def _counters_status(self, batch) -> dict:
+        self._merge_pending_stats()
+        return {
+            "total_interaction_count": self._total_interactions,
+            "total_episode_count": self._episode_count,
+        }
+
+    def _merge_pending_stats(self):
+        if self._pending_stats is None or not self._obs_norm_enabled:
+            self._pending_stats = None
+            return
+        pending = self._pending_stats
+        self._pending_stats = None
+        comm = self._comm
+        if comm is not None and comm.world_size > 1:
+            c, s, ss = pending.stats_triple()
+            packed = torch.cat([c.to(s.device, s.dtype).reshape(1), s.reshape(-1), ss.reshape(-1)])
+            comm.all_reduce_(packed)
+            self._obs_norm.update((packed[0], packed[1 : 1 + self._obs_dim], packed[1 + self._obs_dim :]))
+        else:
+            self._obs_norm.update(pending)
+
+    # -- the hot loop ---------------------------------------
+
+    def _evaluate_batch(self, batch: SolutionBatch):
+        n = len(batch)
+        env = self._get_env(n)
+        device = self.network_device
+        if self._policy is None:
+            self._policy = self.make_functional_policy()
+        policy = self._policy
+        params = batch.access_values(keep_evals=True).to(device, torch.float32)
+        policy.set_parameters(params)
+
+        from ..ops.dispatch import _seed_from_generator
+
+        episode_seed = _seed_from_generator(self._generator, self._device) & 0x7FFFFFFF
+        obs = env.reset(seed=episode_seed).to(device)
+        fitness = torch.zeros(n, dtype=torch.float32, device=device)
+        active = torch.ones(n, dtype=torch.bool, device=device)
+        pending = RunningNorm(shape=self._obs_dim, device=device)
+        max_steps = self._max_num_steps or getattr(env.spec, "episode_length", None) or 1000
+        steps_done = 0
+        for t in range(max_steps):
+            if self._obs_norm_enabled:
+                pending.update(obs, mask=active)
+                obs_in = self._obs_norm.normalize(obs)
+            else:
+                obs_in = obs
+            actions = policy(obs_in)
+            obs, reward, done = env.step(actions)
+            obs = obs.to(device)
+            bonus = self._alive_bonus(t)
+            step_reward = reward.to(device) - self._decrease_rewards_by + bonus
+            fitness = fitness + step_reward * active
+            steps_done += int(active.sum())
+            newly_done = done.to(device) & active
+            if bool(newly_done.any()):
+                policy.reset(newly_done)
+            active = active & ~done.to(device)
+            if not bool(active.any()):
+                break
+        batch.set_evals(fitness.to(self._eval_dtype).to(batch.device))
+        self._pending_stats = pending
+        self.last_eval_interaction_count = steps_done
+        self._total_interactions += steps_done
+        self._episode_count += n
+
+    # -- policy export --------------------------------------------------------
+
+    def to_policy(self, x: torch.Tensor) -> nn.Module:
+        if self._policy is None:
+            self._policy = self.make_functional_policy()
+        module = self._policy.to_torch_module(torch.as_tensor(x, dtype=torch.float32).cpu())
+        if self._obs_norm_enabled and self._obs_norm.has_data:
+            norm_cpu = self._obs_norm.to("cpu")
+            return nn.Sequential(norm_cpu.to_layer(), module)
+        return module
+
+
+VecGymNE = VecEnvNE  # reference-compatible alias (vecgymne.py:95)

