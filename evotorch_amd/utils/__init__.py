@@ -60,6 +60,8 @@ from .misc import (
     to_torch_dtype,
 )
 from .objectarray import ObjectArray, as_object_array
+from .structures import CBag, CDict, CList, CMemory, Structure
+from .tensorframe import TensorFrame
 from .ranking import rank, ranking_method_exists
 from .readonlytensor import ReadOnlyTensor, as_read_only_tensor, read_only_tensor
 from .recursiveprintable import RecursivePrintable

@@ -1,0 +1,192 @@
+"""TensorFrame: a pandas-like columnar container of torch tensors.
+
+Reference parity: /root/reference/src/evotorch/tools/tensorframe.py:53-1338
+(sort/argsort :807/:868, hstack/vstack :881/:922, `each` vmapped row-fn
+:953, `pick` indexer :1270). All columns live on one device; row-wise
+operations are batched tensor ops, never python loops.
+"""
+
+from collections.abc import Mapping
+from typing import Callable, Iterable, Optional, Union
+
+import torch
+
+from .recursiveprintable import RecursivePrintable
+
+__all__ = ["TensorFrame"]
+
+
+class _PickIndexer:
+    def __init__(self, frame: "TensorFrame"):
+        self._frame = frame
+
+    def __getitem__(self, index) -> "TensorFrame":
+        frame = self._frame
+        if isinstance(index, tuple) and len(index) == 2:
+            rows, cols = index
+            if isinstance(cols, str):
+                cols = [cols]
+            sub = TensorFrame({c: frame._columns[c] for c in cols})
+            return sub.pick[rows]
+        if isinstance(index, (int,)):
+            index = slice(index, index + 1)
+        new_cols = {}
+        for name, col in frame._columns.items():
+            if isinstance(index, torch.Tensor) and index.dtype == torch.bool:
+                new_cols[name] = col[index]
+            else:
+                new_cols[name] = col[index]
+        return TensorFrame(new_cols)
+
+
+class TensorFrame(RecursivePrintable):
+    def __init__(self, data: Optional[Union[Mapping, "TensorFrame"]] = None, *, read_only: bool = False, device=None, **kwargs):
+        if isinstance(data, TensorFrame):
+            columns = dict(data._columns)
+        elif data is not None:
+            columns = dict(data)
+        else:
+            columns = {}
+        columns.update(kwargs)
+        self._columns = {}
+        n = None
+        for name, col in columns.items():
+            t = torch.as_tensor(col) if not isinstance(col, torch.Tensor) else col
+            if device is not None:
+                t = t.to(device)
+            if t.ndim == 0:
+                t = t.reshape(1)
+            if n is None:
+                n = t.shape[0]
+            elif t.shape[0] != n:
+                if t.shape[0] == 1:
+                    t = t.expand((n,) + t.shape[1:]).clone()
+                else:
+                    raise ValueError(f"Column {name!r} has {t.shape[0]} rows, expected {n}")
+            self._columns[str(name)] = t
+        self._read_only = bool(read_only)
+
+    # -- basics --------------------------------------------------------------
+
+    @property
+    def columns(self) -> list:
+        return list(self._columns.keys())
+
+    @property
+    def is_read_only(self) -> bool:
+        return self._read_only
+
+    def __len__(self) -> int:
+        for col in self._columns.values():
+            return col.shape[0]
+        return 0
+
+    @property
+    def device(self):
+        for col in self._columns.values():
+            return col.device
+        return torch.device("cpu")
+
+    def __getitem__(self, name: str) -> torch.Tensor:
+        if isinstance(name, str):
+            return self._columns[name]
+        return self.pick[name]
+
+    def __setitem__(self, name: str, value):
+        if self._read_only:
+            raise RuntimeError("This TensorFrame is read-only")
+        value = torch.as_tensor(value)
+        if value.ndim == 0 and len(self) > 0:
+            value = value.expand(len(self)).clone()
+        if len(self._columns) > 0 and value.shape[0] != len(self):
+            raise ValueError(f"Column of {value.shape[0]} rows cannot join a frame of {len(self)} rows")
+        self._columns[str(name)] = value
+
+    def __contains__(self, name: str) -> bool:
+        return name in self._columns
+
+    def __getattr__(self, name: str):
+        cols = object.__getattribute__(self, "_columns") if "_columns" in self.__dict__ else {}
+        if name in cols:
+            return cols[name]
+        raise AttributeError(name)
+
+    @property
+    def pick(self) -> _PickIndexer:
+        """Row indexer: frame.pick[rows] / frame.pick[rows, cols]."""
+        return _PickIndexer(self)
+
+    # -- transforms -----------------------------------------------------------
+
+    def with_columns(self, **kwargs) -> "TensorFrame":
+        new = dict(self._columns)
+        for k, v in kwargs.items():
+            new[k] = torch.as_tensor(v)
+        return TensorFrame(new)
+
+    def without_columns(self, *names: str) -> "TensorFrame":
+        return TensorFrame({k: v for k, v in self._columns.items() if k not in names})
+
+    def to(self, device) -> "TensorFrame":
+        return TensorFrame({k: v.to(device) for k, v in self._columns.items()})
+
+    def argsort(self, by: str, *, descending: bool = False) -> torch.Tensor:
+        return self._columns[by].argsort(descending=descending)
+
+    def sort(self, by: str, *, descending: bool = False) -> "TensorFrame":
+        order = self.argsort(by, descending=descending)
+        return self.pick[order]
+
+    def sort_values(self, by: str, *, ascending: bool = True) -> "TensorFrame":
+        return self.sort(by, descending=not ascending)
+
+    def hstack(self, other: "TensorFrame", *, override: bool = False) -> "TensorFrame":
+        new = dict(self._columns)
+        for k, v in other._columns.items():
+            if k in new and not override:
+                raise ValueError(f"Column {k!r} exists in both frames (use override=True)")
+            new[k] = v
+        return TensorFrame(new)
+
+    def vstack(self, other: "TensorFrame") -> "TensorFrame":
+        if set(self.columns) != set(other.columns):
+            raise ValueError("vstack requires identical column sets")
+        return TensorFrame({k: torch.cat([self._columns[k], other._columns[k]], dim=0) for k in self._columns})
+
+    def each(self, fn: Callable, *, join: bool = False, override: bool = False) -> "TensorFrame":
+        """Apply a row-wise function (receiving a dict of row tensors) via
+        vmap; returns a frame of the (dict) outputs. With join=True the
+        outputs are hstacked onto self."""
+
+        keys = list(self._columns.keys())
+
+        def row_fn(*tensors):
+            row = dict(zip(keys, tensors))
+            out = fn(row)
+            if not isinstance(out, Mapping):
+                raise TypeError("each(fn): fn must return a dict")
+            return tuple(out[k] for k in sorted(out.keys()))
+
+        # probe output keys with row 0
+        probe = fn({k: v[0] for k, v in self._columns.items()})
+        out_keys = sorted(probe.keys())
+        outputs = torch.func.vmap(row_fn)(*[self._columns[k] for k in keys])
+        result = TensorFrame({k: t for k, t in zip(out_keys, outputs)})
+        if join:
+            return self.hstack(result, override=override)
+        return result
+
+    def get_read_only_view(self) -> "TensorFrame":
+        return TensorFrame(self._columns, read_only=True)
+
+    def clone(self, *, memo: Optional[dict] = None) -> "TensorFrame":
+        return TensorFrame({k: v.clone() for k, v in self._columns.items()})
+
+    def to_pandas(self):
+        import pandas as pd
+
+        return pd.DataFrame({k: v.detach().cpu().numpy() if v.ndim == 1 else list(v.detach().cpu().numpy()) for k, v in self._columns.items()})
+
+    def to_string(self, *, max_depth: int = 10) -> str:
+        cols = ", ".join(f"{k}: {tuple(v.shape)}" for k, v in self._columns.items())
+        return f"TensorFrame({len(self)} rows; {cols})"

@@ -1,0 +1,154 @@
+"""Batched structures + TensorFrame tests (mirrors reference
+tests/test_structures.py and test_tensorframe.py)."""
+
+import pytest
+import torch
+
+from evotorch_amd.utils import CBag, CDict, CList, CMemory, TensorFrame
+
+
+def test_cmemory_basic():
+    m = CMemory(3, num_keys=5, fill_with=0.0)
+    m.set_(2, torch.tensor([1.0, 2.0, 3.0]))
+    assert torch.allclose(m[2], torch.tensor([1.0, 2.0, 3.0]))
+    m.add_(2, 1.0)
+    assert torch.allclose(m[2], torch.tensor([2.0, 3.0, 4.0]))
+    assert torch.allclose(m[0], torch.zeros(3))
+
+
+def test_cmemory_batched_masked():
+    m = CMemory(num_keys=4, batch_size=3, fill_with=0.0)
+    keys = torch.tensor([0, 1, 2])
+    m.set_(keys, torch.tensor([10.0, 20.0, 30.0]))
+    got = m.get(keys)
+    assert torch.allclose(got, torch.tensor([10.0, 20.0, 30.0]))
+    # masked write: only batch elements 0 and 2
+    m.set_(keys, torch.tensor([-1.0, -2.0, -3.0]), where=torch.tensor([True, False, True]))
+    assert torch.allclose(m.get(keys), torch.tensor([-1.0, 20.0, -3.0]))
+
+
+def test_cmemory_tuple_keys():
+    m = CMemory(num_keys=(3, 5), fill_with=0.0)
+    m.set_(torch.tensor([2, 4]), 7.0)
+    assert float(m.get(torch.tensor([2, 4]))) == 7.0
+    with pytest.raises(KeyError):
+        m.get(torch.tensor([3, 0]))
+
+
+def test_cmemory_key_offset():
+    m = CMemory(num_keys=10, key_offset=1)
+    m.set_(1, 5.0)
+    assert float(m[1]) == 5.0
+    with pytest.raises(KeyError):
+        m.get(0)
+
+
+def test_cdict_presence_and_default():
+    d = CDict(num_keys=6, batch_size=2, fill_with=0.0)
+    keys = torch.tensor([1, 3])
+    assert not bool(d.contains(keys).any())
+    d.set_(keys, torch.tensor([4.0, 5.0]))
+    assert bool(d.contains(keys).all())
+    other = torch.tensor([2, 2])
+    got = d.get(other, default=-1.0)
+    assert torch.allclose(got, torch.tensor([-1.0, -1.0]))
+    d.clear()
+    assert not bool(d.contains(keys).any())
+
+
+def test_clist_push_pop():
+    lst = CList(max_length=4, batch_size=2)
+    lst.append_(torch.tensor([1.0, 10.0]))
+    lst.append_(torch.tensor([2.0, 20.0]))
+    assert lst.length.tolist() == [2, 2]
+    assert torch.allclose(lst[0], torch.tensor([1.0, 10.0]))
+    assert torch.allclose(lst[-1], torch.tensor([2.0, 20.0]))
+    popped = lst.pop_()
+    assert torch.allclose(popped, torch.tensor([2.0, 20.0]))
+    assert lst.length.tolist() == [1, 1]
+    lst.appendleft_(torch.tensor([0.0, 0.5]))
+    assert torch.allclose(lst[0], torch.tensor([0.0, 0.5]))
+    left = lst.popleft_()
+    assert torch.allclose(left, torch.tensor([0.0, 0.5]))
+
+
+def test_clist_masked_append():
+    lst = CList(max_length=3, batch_size=2)
+    lst.append_(torch.tensor([1.0, 1.0]))
+    lst.append_(torch.tensor([2.0, 2.0]), where=torch.tensor([True, False]))
+    assert lst.length.tolist() == [2, 1]
+
+
+def test_clist_capacity():
+    lst = CList(max_length=2, batch_size=1)
+    for v in [1.0, 2.0, 3.0]:
+        lst.append_(torch.tensor([v]))
+    assert lst.length.tolist() == [2]  # third append ignored (full)
+
+
+def test_cbag_pop_returns_pushed_values():
+    g = torch.Generator().manual_seed(0)
+    bag = CBag(max_length=5, batch_size=1, generator=g)
+    for v in [1.0, 2.0, 3.0]:
+        bag.push_(torch.tensor([v]))
+    seen = set()
+    for _ in range(3):
+        seen.add(float(bag.pop_()[0]))
+    assert seen == {1.0, 2.0, 3.0}
+    assert bag.length.tolist() == [0]
+
+
+# -- TensorFrame -------------------------------------------------------------
+
+
+def test_tensorframe_basics():
+    tf = TensorFrame({"a": torch.arange(5.0), "b": torch.arange(5.0) * 2})
+    assert len(tf) == 5
+    assert tf.columns == ["a", "b"]
+    assert torch.allclose(tf["b"], torch.arange(5.0) * 2)
+    tf["c"] = torch.ones(5)
+    assert "c" in tf
+
+
+def test_tensorframe_sort_and_pick():
+    tf = TensorFrame({"x": torch.tensor([3.0, 1.0, 2.0]), "y": torch.tensor([30.0, 10.0, 20.0])})
+    s = tf.sort("x")
+    assert s["y"].tolist() == [10.0, 20.0, 30.0]
+    sub = tf.pick[torch.tensor([0, 2])]
+    assert sub["x"].tolist() == [3.0, 2.0]
+    sub2 = tf.pick[torch.tensor([True, False, True])]
+    assert sub2["y"].tolist() == [30.0, 20.0]
+    one_col = tf.pick[torch.tensor([0, 1]), "y"]
+    assert one_col.columns == ["y"]
+
+
+def test_tensorframe_stack():
+    a = TensorFrame({"x": torch.zeros(2)})
+    b = TensorFrame({"y": torch.ones(2)})
+    h = a.hstack(b)
+    assert set(h.columns) == {"x", "y"}
+    c = TensorFrame({"x": torch.ones(3)})
+    v = a.vstack(c)
+    assert len(v) == 5
+    with pytest.raises(ValueError):
+        a.vstack(b)
+
+
+def test_tensorframe_each():
+    tf = TensorFrame({"x": torch.arange(4.0), "y": torch.ones(4)})
+    out = tf.each(lambda row: {"z": row["x"] + row["y"]})
+    assert out["z"].tolist() == [1.0, 2.0, 3.0, 4.0]
+    joined = tf.each(lambda row: {"z": row["x"] * 2}, join=True)
+    assert set(joined.columns) == {"x", "y", "z"}
+
+
+def test_tensorframe_read_only():
+    tf = TensorFrame({"a": torch.zeros(3)}).get_read_only_view()
+    with pytest.raises(RuntimeError):
+        tf["b"] = torch.ones(3)
+
+
+def test_tensorframe_to_pandas():
+    tf = TensorFrame({"a": torch.arange(3.0)})
+    df = tf.to_pandas()
+    assert list(df["a"]) == [0.0, 1.0, 2.0]
