@@ -79,6 +79,18 @@ def _domination_matrix(utils: torch.Tensor) -> torch.Tensor:
 
 
 def _compute_pareto_ranks(utils: torch.Tensor, crowdsort: bool = True):
+    """Device-dispatching wrapper: HIP front-peel kernels on ROCm (K7),
+    eager masked reductions on CPU."""
+    if utils.device.type == "cuda":
+        from . import ops
+
+        ranks = ops.pareto_ranks(utils)
+        crowd = _crowding_distances(utils, ranks) if crowdsort else None
+        return ranks, crowd
+    return _compute_pareto_ranks_eager(utils, crowdsort)
+
+
+def _compute_pareto_ranks_eager(utils: torch.Tensor, crowdsort: bool = True):
     """Non-dominated ranking by iterative front peeling over domination
     counts — formulated as masked reductions so the only host sync is the
     loop-termination check (SURVEY.md §7 hard-parts note on K7).

@@ -6,11 +6,13 @@ See `evotorch_amd/ops/dispatch.py` for the routing contract and
 
 from .dispatch import (
     clipup_step_,
+    domination_counts,
     es_gradients,
     fused_adam_step_,
     hip_available,
     hip_required,
     load_hip,
+    pareto_ranks,
     sample_gaussian,
     snes_gradients,
 )
@@ -24,4 +26,6 @@ __all__ = [
     "load_hip",
     "sample_gaussian",
     "snes_gradients",
+    "pareto_ranks",
+    "domination_counts",
 ]

@@ -29,6 +29,8 @@ __all__ = [
     "snes_gradients",
     "clipup_step_",
     "fused_adam_step_",
+    "pareto_ranks",
+    "domination_counts",
 ]
 
 _hip_module = None
@@ -265,3 +267,30 @@ def fused_adam_step_(
     vhat = v32 / (1.0 - beta2**step_count)
     param_step_out.copy_((stepsize * mhat / (vhat.sqrt() + epsilon)).to(param_step_out.dtype))
     return param_step_out
+
+
+# ============================================================================
+# K7 — NSGA-II non-dominated sorting
+# ============================================================================
+
+
+def domination_counts(utils: torch.Tensor) -> torch.Tensor:
+    """Per solution: how many others dominate it. `utils` is (N, M) with
+    higher-is-better columns."""
+    if utils.device.type == "cuda" and not _allow_eager_on_gpu():
+        return hip_required().domination_counts(utils).to(torch.int64)
+    a = utils.unsqueeze(1)
+    b = utils.unsqueeze(0)
+    dom = (a >= b).all(dim=-1) & (a > b).any(dim=-1)
+    return dom.sum(dim=0).to(torch.int64)
+
+
+def pareto_ranks(utils: torch.Tensor) -> torch.Tensor:
+    """Front index per solution (0 = non-dominated front), computed on GPU
+    by count + front peeling without the N x N matrix."""
+    if utils.device.type == "cuda" and not _allow_eager_on_gpu():
+        return hip_required().pareto_ranks(utils)
+    from ..core import _compute_pareto_ranks_eager
+
+    ranks, _ = _compute_pareto_ranks_eager(utils, crowdsort=False)
+    return ranks

@@ -13,6 +13,8 @@ void adam_step(torch::Tensor step_out, torch::Tensor grad, torch::Tensor m, torc
 torch::Tensor rollout_linear(torch::Tensor params, torch::Tensor env_blob, torch::Tensor obs_stats_out,
                              int64_t obs_dim, int64_t act_dim, int64_t rank, int64_t steps, double alive_bonus,
                              double act_cost, int64_t init_seed, int64_t member_offset);
+torch::Tensor domination_counts(torch::Tensor utils);
+torch::Tensor pareto_ranks(torch::Tensor utils);
 }  // namespace ea
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -23,4 +25,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("clipup_step", &ea::clipup_step, "K4: fused ClipUp velocity update (no host sync)");
     m.def("adam_step", &ea::adam_step, "K4: fused Adam ascent step");
     m.def("rollout_linear", &ea::rollout_linear, "K10+K11: fused linear-policy episode rollout (synthetic env)");
+    m.def("domination_counts", &ea::domination_counts, "K7: NSGA-II domination counts (no N x N matrix)");
+    m.def("pareto_ranks", &ea::pareto_ranks, "K7: NSGA-II non-dominated sorting by front peeling");
 }

@@ -1,0 +1,145 @@
+// K7: NSGA-II non-dominated sorting — domination counts
+// and iterative front peeling WITHOUT materializing the N×N domination
+// matrix (the eager path's (N, N, M) broadcast explodes past ~10k
+// solutions; these kernels are O(N²·M) compute but O(N) memory).
+//
+// utils is (N, M) fp32 with HIGHER IS BETTER for every column (senses
+// already folded by the caller). Each thread owns one solution j, caches
+// its own objective row in registers, and streams the other rows — row i
+// is read by all threads of a wave simultaneously (broadcast, L2-cached).
+//
+// Peeling: counts[j] = #dominators. Front k = {count == 0}. The peel
+// kernel subtracts each front member's contributions; every solution is
+// subtracted exactly once over the whole sort, so total peel work is one
+// more O(N²·M) pass — host sync only once per front.
+
+#include <hip/hip_runtime.h>
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+
+namespace ea {
+
+constexpr int kMaxObjCache = 16;
+
+__device__ __forceinline__ bool dominates_rows(const float* __restrict__ a, const float* __restrict__ b, int m) {
+    bool ge_all = true, gt_any = false;
+    for (int k = 0; k < m; ++k) {
+        ge_all &= (a[k] >= b[k]);
+        gt_any |= (a[k] > b[k]);
+    }
+    return ge_all && gt_any;
+}
+
+__global__ void domination_counts_kernel(const float* __restrict__ utils, int* __restrict__ counts, int64_t n, int m) {
+    const int64_t j = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+    if (j >= n) return;
+    float mine[kMaxObjCache];
+    const bool cached = m <= kMaxObjCache;
+    if (cached) {
+        for (int k = 0; k < m; ++k) mine[k] = utils[j * m + k];
+    }
+    int count = 0;
+    for (int64_t i = 0; i < n; ++i) {
+        const float* other = utils + i * m;
+        bool ge_all = true, gt_any = false;
+        if (cached) {
+            for (int k = 0; k < m; ++k) {
+                const float o = other[k];
+                ge_all &= (o >= mine[k]);
+                gt_any |= (o > mine[k]);
+            }
+        } else {
+            const float* me = utils + j * m;
+            for (int k = 0; k < m; ++k) {
+                const float o = other[k];
+                ge_all &= (o >= me[k]);
+                gt_any |= (o > me[k]);
+            }
+        }
+        count += (ge_all && gt_any) ? 1 : 0;
+    }
+    counts[j] = count;
+}
+
+// Assign front `front_index` to zero-count unassigned solutions, then
+// subtract their domination contributions from everyone else. Returns (in
+// num_assigned[0]) how many were assigned this round.
+__global__ void peel_front_kernel(const float* __restrict__ utils, int* __restrict__ counts,
+                                  int64_t* __restrict__ ranks, const uint8_t* __restrict__ front_mask,
+                                  int64_t n, int m, int64_t front_index) {
+    const int64_t j = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+    if (j >= n) return;
+    if (front_mask[j]) {
+        ranks[j] = front_index;
+        counts[j] = -1;  // assigned marker
+        return;
+    }
+    if (counts[j] < 0) return;  // already assigned earlier
+    float mine[kMaxObjCache];
+    const bool cached = m <= kMaxObjCache;
+    if (cached) {
+        for (int k = 0; k < m; ++k) mine[k] = utils[j * m + k];
+    }
+    int removed = 0;
+    for (int64_t i = 0; i < n; ++i) {
+        if (!front_mask[i]) continue;
+        const float* other = utils + i * m;
+        const float* me = cached ? mine : (utils + j * m);
+        bool ge_all = true, gt_any = false;
+        for (int k = 0; k < m; ++k) {
+            const float o = other[k];
+            ge_all &= (o >= me[k]);
+            gt_any |= (o > me[k]);
+        }
+        removed += (ge_all && gt_any) ? 1 : 0;
+    }
+    counts[j] -= removed;
+}
+
+torch::Tensor domination_counts(torch::Tensor utils) {
+    TORCH_CHECK(utils.is_cuda() && utils.dim() == 2, "utils must be a 2-D ROCm tensor");
+    auto utils_f = utils.to(torch::kFloat32).contiguous();
+    const int64_t n = utils_f.size(0);
+    const int m = (int)utils_f.size(1);
+    auto counts = torch::empty({n}, utils_f.options().dtype(torch::kInt32));
+    const int threads = 256;
+    const int blocks = (int)((n + threads - 1) / threads);
+    auto stream = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(domination_counts_kernel, dim3(blocks), dim3(threads), 0, stream, utils_f.data_ptr<float>(),
+                       counts.data_ptr<int>(), n, m);
+    return counts;
+}
+
+torch::Tensor pareto_ranks(torch::Tensor utils) {
+    TORCH_CHECK(utils.is_cuda() && utils.dim() == 2, "utils must be a 2-D ROCm tensor");
+    auto utils_f = utils.to(torch::kFloat32).contiguous();
+    const int64_t n = utils_f.size(0);
+    const int m = (int)utils_f.size(1);
+    auto counts = domination_counts(utils_f);
+    auto ranks = torch::full({n}, -1, utils_f.options().dtype(torch::kInt64));
+    const int threads = 256;
+    const int blocks = (int)((n + threads - 1) / threads);
+    auto stream = at::cuda::getCurrentCUDAStream();
+    int64_t front_index = 0;
+    while (true) {
+        auto front_mask = (counts == 0);  // unassigned & undominated
+        // ONE host sync per front: are we done / does the front exist?
+        const int64_t front_size = front_mask.sum().item<int64_t>();
+        if (front_size == 0) {
+            // numerical corner: assign any stragglers to the current front
+            auto remaining = counts >= 0;
+            if (remaining.sum().item<int64_t>() == 0) break;
+            ranks.masked_fill_(remaining, front_index);
+            break;
+        }
+        auto mask_u8 = front_mask.to(torch::kUInt8).contiguous();
+        hipLaunchKernelGGL(peel_front_kernel, dim3(blocks), dim3(threads), 0, stream, utils_f.data_ptr<float>(),
+                           counts.data_ptr<int>(), ranks.data_ptr<int64_t>(), mask_u8.data_ptr<uint8_t>(), n, m,
+                           front_index);
+        ++front_index;
+        if (front_index > n) break;  // safety
+    }
+    return ranks;
+}
+
+}  // namespace ea

@@ -245,3 +245,50 @@ class TestEndToEndGPU:
         first = searcher.status["mean_eval"]
         searcher.run(50)
         assert searcher.status["mean_eval"] < first * 0.5
+
+
+@requires_gpu
+class TestParetoKernels:
+    def test_matches_eager(self, C):
+        import os
+
+        torch.manual_seed(7)
+        n, m = 2000, 3
+        utils = torch.randn(n, m, device="cuda")
+        counts = C.domination_counts(utils)
+        ranks = C.pareto_ranks(utils)
+        os.environ["EVOTORCH_AMD_ALLOW_EAGER_GPU"] = "1"
+        try:
+            from evotorch_amd.core import _compute_pareto_ranks_eager
+
+            eranks, _ = _compute_pareto_ranks_eager(utils, crowdsort=False)
+            a = utils.unsqueeze(1)
+            b = utils.unsqueeze(0)
+            dom = (a >= b).all(dim=-1) & (a > b).any(dim=-1)
+            ecounts = dom.sum(dim=0)
+        finally:
+            os.environ["EVOTORCH_AMD_ALLOW_EAGER_GPU"] = "0"
+        assert torch.equal(counts.to(torch.int64), ecounts.to(torch.int64))
+        assert torch.equal(ranks, eranks)
+
+    def test_nsga2_ga_on_gpu(self):
+        from evotorch_amd import Problem
+        from evotorch_amd.algorithms import GeneticAlgorithm
+        from evotorch_amd.decorators import vectorized
+        from evotorch_amd.operators import GaussianMutation, SimulatedBinaryCrossOver
+
+        @vectorized
+        def two_obj(x):
+            f1 = (x**2).sum(-1)
+            f2 = ((x - 2.0) ** 2).sum(-1)
+            return torch.stack([f1, f2], dim=-1)
+
+        prob = Problem(["min", "min"], two_obj, solution_length=8, initial_bounds=(-4, 4), seed=3, device="cuda:0")
+        ga = GeneticAlgorithm(
+            prob,
+            popsize=512,
+            operators=[SimulatedBinaryCrossOver(prob, tournament_size=3, eta=8.0), GaussianMutation(prob, stdev=0.2)],
+        )
+        ga.run(10)
+        ranks, _ = ga.population.compute_pareto_ranks()
+        assert float((ranks == 0).float().mean()) > 0.3
