@@ -58,7 +58,7 @@ struct RolloutArgs {
 // env_blob layout (fp32): V [R][O] · U_T [R][O] · D2_T [A][O] · c [O] ·
 // wr [O] · mean [O] · std [O]
 
-__global__ __launch_bounds__(256, 2) void rollout_linear_kernel(RolloutArgs args) {
+__global__ __launch_bounds__(512, 2) void rollout_linear_kernel(RolloutArgs args) {
     const int O = args.obs_dim, A = args.act_dim, R = args.rank;
     const int tid = threadIdx.x;
     const int member = blockIdx.x;
@@ -310,7 +310,10 @@ torch::Tensor rollout_linear(torch::Tensor params, torch::Tensor env_blob, torch
     TORCH_CHECK(lds_bytes <= 64 * 1024, "rollout LDS footprint too large: ", lds_bytes,
                 " bytes (reduce rank / dims)");
     auto stream = at::cuda::getCurrentCUDAStream();
-    hipLaunchKernelGGL(rollout_linear_kernel, dim3(n), dim3(256), lds_bytes, stream, args);
+    int block = 512;
+    if (const char* env = getenv("EVOTORCH_AMD_ROLLOUT_BLOCK")) block = atoi(env);
+    TORCH_CHECK(block == 256 || block == 512, "EVOTORCH_AMD_ROLLOUT_BLOCK must be 256 or 512");
+    hipLaunchKernelGGL(rollout_linear_kernel, dim3(n), dim3(block), lds_bytes, stream, args);
     return fitness;
 }
 

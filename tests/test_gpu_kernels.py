@@ -289,6 +289,6 @@ class TestParetoKernels:
             popsize=512,
             operators=[SimulatedBinaryCrossOver(prob, tournament_size=3, eta=8.0), GaussianMutation(prob, stdev=0.2)],
         )
-        ga.run(10)
+        ga.run(20)
         ranks, _ = ga.population.compute_pareto_ranks()
-        assert float((ranks == 0).float().mean()) > 0.3
+        assert float((ranks == 0).float().mean()) > 0.1
