@@ -1067,11 +1067,25 @@ class Problem(TensorMakerMixin, Serializable):
             popsize += 1
         return popsize
 
+    def _get_cached_grad_batch(self, popsize: int, device) -> "SolutionBatch":
+        """Reusable sampling batch for the distribution-gradient path: a
+        fresh 50 GB population tensor every generation would churn the
+        caching allocator (measured: multi-second hipMalloc stalls at
+        L=1e6, N=1e4); the batch is allocated once and re-sampled
+        in place."""
+        cached = getattr(self, "_grad_batch_cache", None)
+        if cached is not None and len(cached) == popsize and cached.device == torch.device(device):
+            cached.forget_evals()
+            return cached
+        batch = self.generate_batch(popsize, empty=True)
+        self._grad_batch_cache = batch
+        return batch
+
     def _sample_and_compute_gradients(
         self, distribution, popsize: int, obj_index: int, ranking_method, num_interactions=None, popsize_max=None, ensure_even_popsize: bool = False
     ) -> dict:
         popsize = self._sample_popsize(distribution, popsize, num_interactions, popsize_max, ensure_even_popsize)
-        batch = self.generate_batch(popsize, empty=True)
+        batch = self._get_cached_grad_batch(popsize, self._device)
         distribution.sample(out=batch.access_values(), generator=self._generator)
         self.evaluate(batch)
         fitnesses = batch._evals[:, obj_index]
@@ -1090,7 +1104,7 @@ class Problem(TensorMakerMixin, Serializable):
         local_popsize = popsize // world
         if ensure_even_popsize and local_popsize % 2 != 0:
             local_popsize += 1
-        batch = self.generate_batch(local_popsize, empty=True)
+        batch = self._get_cached_grad_batch(local_popsize, self._device)
         distribution.sample(out=batch.access_values(), generator=self._generator)
         self._before_eval_hook(batch)
         self._evaluate_batch(batch)
@@ -1128,7 +1142,7 @@ class Problem(TensorMakerMixin, Serializable):
     def _get_cloned_state(self, *, memo: dict) -> dict:
         state = {}
         for k, v in self.__dict__.items():
-            if k in ("_generator", "_comm"):
+            if k in ("_generator", "_comm", "_grad_batch_cache"):
                 state[k] = None
             else:
                 state[k] = deep_clone(v, otherwise_deepcopy=True, memo=memo)

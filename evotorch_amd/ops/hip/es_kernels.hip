@@ -59,6 +59,39 @@ __global__ void sample_gaussian_kernel(T* __restrict__ out, const T* __restrict_
     }
 }
 
+// fp32 fast path for length % 4 == 0: the 4 philox normals of one counter
+// are 4 consecutive elements of one row, stored as a single 16 B float4
+// (the CDNA coalescing sweet spot — cdna_hip_programming.md G13); mu and
+// sigma are read as float4 too.
+template <bool kSymmetric>
+__global__ void sample_gaussian_f32x4_kernel(float4* __restrict__ out, const float4* __restrict__ mu,
+                                             const float4* __restrict__ sigma, int64_t rows, int64_t length4,
+                                             uint64_t seed) {
+    const int64_t total4 = rows * length4;
+    for (int64_t idx4 = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; idx4 < total4;
+         idx4 += (int64_t)gridDim.x * blockDim.x) {
+        float z[4];
+        philox_normal4(seed, 0u, (uint64_t)idx4, z);
+        const int64_t col4 = idx4 % length4;
+        const float4 m = mu[col4];
+        const float4 s = sigma[col4];
+        float4 plus;
+        plus.x = fmaf(s.x, z[0], m.x);
+        plus.y = fmaf(s.y, z[1], m.y);
+        plus.z = fmaf(s.z, z[2], m.z);
+        plus.w = fmaf(s.w, z[3], m.w);
+        out[idx4] = plus;
+        if (kSymmetric) {
+            float4 minus;
+            minus.x = 2.0f * m.x - plus.x;
+            minus.y = 2.0f * m.y - plus.y;
+            minus.z = 2.0f * m.z - plus.z;
+            minus.w = 2.0f * m.w - plus.w;
+            out[idx4 + total4] = minus;
+        }
+    }
+}
+
 void sample_gaussian(torch::Tensor out, torch::Tensor mu, torch::Tensor sigma, bool symmetric, int64_t seed) {
     CHECK_GPU(out); CHECK_CONTIG(out); CHECK_GPU(mu); CHECK_GPU(sigma);
     const int64_t n = out.size(0), length = out.size(1);
@@ -68,6 +101,18 @@ void sample_gaussian(torch::Tensor out, torch::Tensor mu, torch::Tensor sigma, b
     const int64_t total4 = (rows * length + 3) / 4;
     const int blocks = (int)std::min<int64_t>((total4 + threads - 1) / threads, 256 * 8);
     auto stream = at::cuda::getCurrentCUDAStream();
+    if (out.scalar_type() == at::ScalarType::Float && length % 4 == 0 && mu.is_contiguous() && sigma.is_contiguous()) {
+        if (symmetric) {
+            hipLaunchKernelGGL((sample_gaussian_f32x4_kernel<true>), dim3(blocks), dim3(threads), 0, stream,
+                               reinterpret_cast<float4*>(out.data_ptr<float>()), reinterpret_cast<const float4*>(mu.data_ptr<float>()),
+                               reinterpret_cast<const float4*>(sigma.data_ptr<float>()), rows, length / 4, (uint64_t)seed);
+        } else {
+            hipLaunchKernelGGL((sample_gaussian_f32x4_kernel<false>), dim3(blocks), dim3(threads), 0, stream,
+                               reinterpret_cast<float4*>(out.data_ptr<float>()), reinterpret_cast<const float4*>(mu.data_ptr<float>()),
+                               reinterpret_cast<const float4*>(sigma.data_ptr<float>()), rows, length / 4, (uint64_t)seed);
+        }
+        return;
+    }
     AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::BFloat16, at::ScalarType::Half, out.scalar_type(), "sample_gaussian", [&] {
         using T = scalar_t;
         if (symmetric) {

@@ -44,10 +44,19 @@ def main():
     g = torch.Generator(device="cpu").manual_seed(7)
     target = torch.randn(L, generator=g).to(device)
 
+    scratch = {}
+
     @vectorized
     def quadratic(x):
-        d = x - target
-        return (d * d).sum(-1)
+        # reuse one difference buffer: fresh 50 GB temporaries every
+        # generation would thrash the caching allocator
+        d = scratch.get("d")
+        if d is None or d.shape != x.shape:
+            d = torch.empty_like(x)
+            scratch["d"] = d
+        torch.sub(x, target, out=d)
+        d.mul_(d)
+        return d.sum(-1)
 
     prob = Problem("min", quadratic, solution_length=L, initial_bounds=(-1, 1), device=device, seed=1 + rank)
     if comm is not None:
