@@ -208,10 +208,10 @@ class CMAES(SearchAlgorithm, SinglePopulationAlgorithmMixin):
         if self._stdev_max is not None:
             self._sigma = torch.clamp(self._sigma, max=self._stdev_max)
 
-        # covariance update
+        # covariance update (hs stays a device tensor: no per-gen host sync)
         cc = self._c_c
-        hs = float(ps_norm / math.sqrt(1.0 - (1.0 - cs) ** (2 * (self._steps_count + 1)))) < (1.4 + 2.0 / (n + 1.0)) * self._chi_n
-        hs_f = 1.0 if hs else 0.0
+        hs_threshold = (1.4 + 2.0 / (n + 1.0)) * self._chi_n * math.sqrt(1.0 - (1.0 - cs) ** (2 * (self._steps_count + 1)))
+        hs_f = (ps_norm < hs_threshold).to(self._p_c.dtype)
         self._p_c = (1.0 - cc) * self._p_c + hs_f * math.sqrt(cc * (2.0 - cc) * self._mu_eff) * y_w
         c1, cmu = self._c_1, self._c_mu
         delta_hs = (1.0 - hs_f) * cc * (2.0 - cc)
