@@ -1,0 +1,31 @@
+"""Run each example end-to-end with tiny budgets (mirrors the reference's
+tests/test_examples.py integration strategy)."""
+
+import os
+import subprocess
+import sys
+
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+CASES = [
+    ("examples/rastrigin_snes.py", ["--generations", "3"]),
+    ("examples/multiobjective_nsga2.py", []),
+    ("examples/mapelites_illumination.py", []),
+    ("examples/functional_api_batched.py", []),
+    ("examples/synthetic_humanoid_pgpe.py", ["--generations", "2", "--popsize", "16", "--device", "cpu"]),
+    ("examples/mnist30k_distributed.py", ["--generations", "1", "--popsize", "8"]),
+]
+
+
+@pytest.mark.parametrize("script,args", CASES, ids=[c[0] for c in CASES])
+def test_example_runs(script, args, tmp_path):
+    result = subprocess.run(
+        [sys.executable, os.path.join(REPO, script), *args],
+        cwd=tmp_path,  # checkpoints etc. land in a temp dir
+        capture_output=True,
+        text=True,
+        timeout=420,
+    )
+    assert result.returncode == 0, f"{script} failed:\n{result.stderr[-2000:]}"
