@@ -25,7 +25,7 @@ __version__ = "0.1.0"
 from . import utils
 from .core import Problem, ProblemBoundEvaluator, Solution, SolutionBatch
 
-from . import algorithms, decorators, distributions, logging, operators, optimizers, testing  # noqa: E402
+from . import algorithms, decorators, distributions, logging, models, neuroevolution, operators, ops, optimizers, parallel, testing, tools  # noqa: E402
 
 __all__ = [
     "__version__",
@@ -37,8 +37,13 @@ __all__ = [
     "decorators",
     "distributions",
     "logging",
+    "models",
+    "neuroevolution",
     "operators",
+    "ops",
     "optimizers",
+    "parallel",
     "testing",
+    "tools",
     "utils",
 ]
