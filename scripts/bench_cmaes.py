@@ -20,6 +20,7 @@ def main():
     p.add_argument("--popsize", type=int, default=64)
     p.add_argument("--steps", type=int, default=20)
     p.add_argument("--warmup", type=int, default=3)
+    p.add_argument("--every-gen-decomposition", action="store_true", help="Cholesky every generation (default: reference-style amortized interval)")
     args = p.parse_args()
 
     from evotorch_amd import Problem
@@ -33,7 +34,7 @@ def main():
         return (x**2).sum(-1)
 
     prob = Problem("min", sphere, solution_length=args.dim, initial_bounds=(-3, 3), device=device, seed=1)
-    searcher = CMAES(prob, stdev_init=1.0, popsize=args.popsize, limit_C_decomposition=False)
+    searcher = CMAES(prob, stdev_init=1.0, popsize=args.popsize, limit_C_decomposition=not args.every_gen_decomposition)
     for _ in range(args.warmup):
         searcher.step()
     if device != "cpu":

@@ -1,0 +1,105 @@
+"""Decorator semantics (mirrors reference tests/test_decorators.py and
+test_expects_ndim.py)."""
+
+import pytest
+import torch
+
+from evotorch_amd.decorators import expects_ndim, on_aux_device, on_cuda, on_device, pass_info, rowwise, vectorized
+
+
+def test_vectorized_flag():
+    @vectorized
+    def f(x):
+        return x.sum(-1)
+
+    assert f.__evotorch_vectorized__ is True
+
+
+def test_on_device_flags():
+    @on_device("cuda:1")
+    def f(x):
+        return x
+
+    assert f.__evotorch_device__ == "cuda:1"
+
+    @on_cuda
+    def g(x):
+        return x
+
+    assert g.__evotorch_device__ == "cuda"
+
+    @on_cuda(2)
+    def h(x):
+        return x
+
+    assert h.__evotorch_device__ == "cuda:2"
+
+    @on_aux_device
+    def k(x):
+        return x
+
+    assert k.__evotorch_on_aux_device__ is True
+
+
+def test_pass_info():
+    @pass_info
+    def net_factory(**kwargs):
+        return kwargs
+
+    assert net_factory.__evotorch_pass_info__ is True
+
+
+def test_expects_ndim_batches_extra_dims():
+    def dot(a, b):
+        return (a * b).sum()
+
+    batched = expects_ndim(dot, (1, 1))
+    a = torch.randn(5, 3)
+    b = torch.randn(3)
+    out = batched(a, b)
+    assert out.shape == (5,)
+    expected = a @ b
+    assert torch.allclose(out, expected, atol=1e-5)
+    # two batch levels
+    a2 = torch.randn(4, 5, 3)
+    out2 = batched(a2, b)
+    assert out2.shape == (4, 5)
+
+
+def test_expects_ndim_scalar_args():
+    def scale(k, v):
+        return k * v
+
+    batched = expects_ndim(scale, (None, 1))
+    v = torch.randn(6, 4)
+    out = batched(2.0, v)
+    assert torch.allclose(out, v * 2)
+
+
+def test_expects_ndim_validates():
+    def f(v):
+        return v.sum()
+
+    wrapped = expects_ndim(f, (2,))
+    with pytest.raises(ValueError):
+        wrapped(torch.randn(5))  # ndim too small
+
+
+def test_rowwise():
+    @rowwise
+    def normalize(x):
+        return x / x.sum()
+
+    row = torch.tensor([1.0, 3.0])
+    assert torch.allclose(normalize(row), torch.tensor([0.25, 0.75]))
+    mat = torch.tensor([[1.0, 1.0], [1.0, 3.0]])
+    out = normalize(mat)
+    assert torch.allclose(out, torch.tensor([[0.5, 0.5], [0.25, 0.75]]))
+
+
+def test_tools_alias():
+    import evotorch_amd.tools as tools
+    import evotorch_amd.utils as utils
+
+    assert tools.rank is utils.rank
+    assert tools.ObjectArray is utils.ObjectArray
