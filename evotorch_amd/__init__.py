@@ -22,6 +22,20 @@ Layer map:
 
 __version__ = "0.1.0"
 
+import logging as _py_logging
+import os as _os
+
+_verbose = {"-1": -1, "0": _py_logging.WARNING, "1": _py_logging.INFO, "2": _py_logging.DEBUG}.get(
+    str(_os.environ.get("EVOTORCH_AMD_VERBOSE_LEVEL", "1"))
+)
+if _verbose is not None and _verbose >= 0:
+    _logger = _py_logging.getLogger("evotorch_amd")
+    if not _logger.handlers:
+        _handler = _py_logging.StreamHandler()
+        _handler.setFormatter(_py_logging.Formatter("[%(asctime)s] %(name)s %(levelname)s: %(message)s"))
+        _logger.addHandler(_handler)
+    _logger.setLevel(_verbose)
+
 from . import utils
 from .core import Problem, ProblemBoundEvaluator, Solution, SolutionBatch
 

@@ -70,9 +70,16 @@ __global__ __launch_bounds__(512, 2) void rollout_linear_kernel(RolloutArgs args
     extern __shared__ unsigned char lds_raw[];
     __bf16* W_l = reinterpret_cast<__bf16*>(lds_raw);   // [A][O] row-major
     __bf16* V_l = W_l + A * O;                          // [R][O] row-major
-    __bf16* U_row = V_l + R * O;                        // [O][R_PAD]
-    __bf16* D2_row = U_row + O * R_PAD;                 // [O][A_PAD]
-    __bf16* obs_b = D2_row + O * A_PAD;                 // [O]    bf16 state
+    // U and D2 are stored PAIR-INTERLEAVED COLUMN-MAJOR: element [i2][j]
+    // is the bf16x2 (U_T[2*i2][j], U_T[2*i2+1][j]). Phase 2's thread j then
+    // reads one 4 B bf16x2 per dot2 at consecutive-lane-consecutive-word
+    // addresses — conflict-free b32 LDS reads. (The previous padded
+    // per-output-row layout let the compiler merge the 9 contiguous pair
+    // reads into strided ds_read_b128s: 1.16e9 bank-conflict cycles per
+    // bench run, 58% of all LDS instructions.)
+    bf16x2* U_pair = reinterpret_cast<bf16x2*>(V_l + R * O);  // [R_PAD/2][O]
+    bf16x2* D2_pair = U_pair + (R_PAD / 2) * O;               // [A_PAD/2][O]
+    __bf16* obs_b = reinterpret_cast<__bf16*>(D2_pair + (A_PAD / 2) * O);  // [O]
     __bf16* obsn_b = obs_b + O;                         // [O]    normalized
     __bf16* h_b = obsn_b + O;                           // [R_PAD]
     __bf16* act_b = h_b + R_PAD;                        // [A_PAD]
@@ -91,14 +98,20 @@ __global__ __launch_bounds__(512, 2) void rollout_linear_kernel(RolloutArgs args
         const float* e = args.env_blob;
         const int RO = R * O, AO = A * O;
         for (int i = tid; i < RO; i += blockDim.x) V_l[i] = f2b(e[i]);
-        // U arrives as U_T [R][O]; store per-output rows U_row[j][i] = U_T[i][j]
-        for (int i = tid; i < RO; i += blockDim.x) {
-            const int r = i / O, j = i % O;
-            U_row[j * R_PAD + r] = f2b(e[RO + i]);
+        // U arrives as U_T [R][O]; store bf16x2 pairs along R, column-major
+        for (int i = tid; i < (R_PAD / 2) * O; i += blockDim.x) {
+            const int i2 = i / O, j = i % O;
+            bf16x2 v2;
+            v2.x = (2 * i2 < R) ? f2b(e[RO + (2 * i2) * O + j]) : f2b(0.0f);
+            v2.y = (2 * i2 + 1 < R) ? f2b(e[RO + (2 * i2 + 1) * O + j]) : f2b(0.0f);
+            U_pair[i2 * O + j] = v2;
         }
-        for (int i = tid; i < AO; i += blockDim.x) {
-            const int m = i / O, j = i % O;
-            D2_row[j * A_PAD + m] = f2b(e[2 * RO + i]);
+        for (int i = tid; i < (A_PAD / 2) * O; i += blockDim.x) {
+            const int m2 = i / O, j = i % O;
+            bf16x2 v2;
+            v2.x = (2 * m2 < A) ? f2b(e[2 * RO + (2 * m2) * O + j]) : f2b(0.0f);
+            v2.y = (2 * m2 + 1 < A) ? f2b(e[2 * RO + (2 * m2 + 1) * O + j]) : f2b(0.0f);
+            D2_pair[m2 * O + j] = v2;
         }
         const float* tail = e + 2 * RO + AO;
         for (int j = tid; j < O; j += blockDim.x) {
@@ -107,13 +120,9 @@ __global__ __launch_bounds__(512, 2) void rollout_linear_kernel(RolloutArgs args
             mean_l[j] = tail[2 * O + j];
             istd_l[j] = 1.0f / tail[3 * O + j];
         }
-        // zero the pads once so dot2 over padded rows is exact
+        // zero the h/act pads once so dot2 over padded vectors is exact
         for (int i = tid; i < R_PAD; i += blockDim.x) h_b[i] = f2b(0.0f);
         for (int i = tid; i < A_PAD; i += blockDim.x) act_b[i] = f2b(0.0f);
-        for (int j = tid; j < O; j += blockDim.x) {
-            for (int r = R; r < R_PAD; ++r) U_row[j * R_PAD + r] = f2b(0.0f);
-            for (int m = A; m < A_PAD; ++m) D2_row[j * A_PAD + m] = f2b(0.0f);
-        }
     }
     // initial observation: 0.1 * N(0,1), deterministic per global member id
     const unsigned long long gmember = (unsigned long long)(args.member_offset + member);
@@ -234,20 +243,19 @@ __global__ __launch_bounds__(512, 2) void rollout_linear_kernel(RolloutArgs args
         // the NEXT step's normalization fused into the epilogue
         for (int j = tid; j < O; j += blockDim.x) {
             // two independent dot chains (U·h and D2·a) halve the serial
-            // dot2 dependency depth; joined at the end
+            // dot2 dependency depth; joined at the end. Each operand read
+            // is one bf16x2 at lane-consecutive addresses (conflict-free).
             float uacc = c_l[j], dacc = 0.0f;
-            const __bf16* urow = U_row + j * R_PAD;
-            const __bf16* drow = D2_row + j * A_PAD;
-#pragma unroll 4
+#pragma unroll
             for (int p = 0; p < R_PAD / 2; ++p) {
                 uacc = __builtin_amdgcn_fdot2_f32_bf16(
-                    *reinterpret_cast<const bf16x2*>(urow + 2 * p),
+                    U_pair[p * O + j],
                     *reinterpret_cast<const bf16x2*>(h_b + 2 * p), uacc, false);
             }
-#pragma unroll 3
+#pragma unroll
             for (int p = 0; p < A_PAD / 2; ++p) {
                 dacc = __builtin_amdgcn_fdot2_f32_bf16(
-                    *reinterpret_cast<const bf16x2*>(drow + 2 * p),
+                    D2_pair[p * O + j],
                     *reinterpret_cast<const bf16x2*>(act_b + 2 * p), dacc, false);
             }
             const float o_new = tanhf(uacc + dacc);

@@ -1,0 +1,85 @@
+"""End-to-end GPU runs of the wider algorithm portfolio (beyond the
+kernel-numerics suite)."""
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+requires_gpu = pytest.mark.skipif(not torch.cuda.is_available(), reason="needs ROCm GPU")
+
+
+@requires_gpu
+def test_cmaes_gpu_converges():
+    from evotorch_amd import Problem
+    from evotorch_amd.algorithms import CMAES
+    from evotorch_amd.decorators import vectorized
+
+    @vectorized
+    def sphere(x):
+        return (x**2).sum(-1)
+
+    prob = Problem("min", sphere, solution_length=256, initial_bounds=(-3, 3), device="cuda:0", seed=2)
+    searcher = CMAES(prob, stdev_init=2.0, popsize=64)
+    searcher.step()
+    first = searcher.status["pop_best_eval"]
+    searcher.run(60)
+    assert searcher.status["pop_best_eval"] < first * 0.2
+
+
+@requires_gpu
+def test_vecenvne_gpu_rollout():
+    from evotorch_amd.algorithms import PGPE
+    from evotorch_amd.neuroevolution import SyntheticTorchEnv, VecEnvNE
+
+    def env_factory(num_envs):
+        return SyntheticTorchEnv(num_envs=num_envs, episode_length=20, obs_dim=32, act_dim=8, rank=8, device="cuda:0")
+
+    prob = VecEnvNE(env_factory, "Linear(obs_length, 16) >> Tanh() >> Linear(16, act_length)",
+                    device="cuda:0", seed=3)
+    searcher = PGPE(prob, popsize=64, center_learning_rate=0.05, stdev_learning_rate=0.1,
+                    stdev_init=0.1, distributed=True)
+    searcher.run(3)
+    assert searcher.status["iter"] == 3
+    assert prob.obs_norm.has_data
+
+
+@requires_gpu
+def test_supervisedne_gpu():
+    import torch.nn as nn
+
+    from evotorch_amd.algorithms import CEM
+    from evotorch_amd.neuroevolution import SupervisedNE
+
+    torch.manual_seed(0)
+    x = torch.randn(512, 8)
+    y = x @ torch.randn(8, 1)
+    ds = torch.utils.data.TensorDataset(x, y)
+    prob = SupervisedNE(ds, nn.Linear(8, 1), nn.MSELoss(), minibatch_size=128,
+                        common_minibatch=True, device="cuda:0", seed=4)
+    searcher = CEM(prob, popsize=64, parenthood_ratio=0.25, stdev_init=0.5)
+    searcher.step()
+    first = searcher.status["mean_eval"]
+    searcher.run(20)
+    assert searcher.status["mean_eval"] < first
+
+
+@requires_gpu
+def test_ga_singleobj_gpu():
+    from evotorch_amd import Problem
+    from evotorch_amd.algorithms import GeneticAlgorithm
+    from evotorch_amd.decorators import vectorized
+    from evotorch_amd.operators import GaussianMutation, OnePointCrossOver
+
+    @vectorized
+    def sphere(x):
+        return (x**2).sum(-1)
+
+    prob = Problem("min", sphere, solution_length=64, initial_bounds=(-3, 3), device="cuda:0", seed=5)
+    ga = GeneticAlgorithm(prob, popsize=256, operators=[
+        OnePointCrossOver(prob, tournament_size=4),
+        GaussianMutation(prob, stdev=0.2),
+    ])
+    ga.step()
+    first = ga.status["pop_best_eval"]
+    ga.run(20)
+    assert ga.status["pop_best_eval"] < first
