@@ -158,3 +158,33 @@ def test_world2(body):
         assert results[0] == results[1]  # both ranks hold the full eval vector
     if body == "_body_distributed_pgpe":
         assert results[0] == results[1]
+
+
+def test_bench_entry_torchrun_world2(tmp_path):
+    """The driver launches bench.py via torch.distributed.run; validate that
+    exact entry path (world 2, gloo on CPU) end to end."""
+    import json
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    result = subprocess.run(
+        [
+            sys.executable, "-m", "torch.distributed.run",
+            "--nnodes=1", "--nproc-per-node", "2",
+            "--master-addr", "127.0.0.1", "--master-port", "29877",
+            os.path.join(repo, "bench.py"),
+            "--gpus", "2", "--steps", "2", "--warmup", "1",
+            "--popsize-per-gpu", "16", "--episode-length", "5",
+        ],
+        cwd=repo,
+        capture_output=True,
+        text=True,
+        timeout=300,
+    )
+    assert result.returncode == 0, result.stderr[-3000:]
+    line = [l for l in result.stdout.splitlines() if l.startswith("{")][-1]
+    payload = json.loads(line)
+    assert payload["config"]["global_batch"] == 32
+    assert payload["config"]["parallelism"] == "dp2"
+    assert payload["value"] > 0
