@@ -4,6 +4,7 @@
 from .cmaes import CMAES
 from .ga import Cosyne, GeneticAlgorithm, SteadyStateGA
 from .gaussian import CEM, PGPE, SNES, XNES, GaussianSearchAlgorithm
+from .graphed import GraphedSearch
 from .mapelites import MAPElites
 from .restarter import IPOP, ModifyingRestart, Restart
 from .searchalgorithm import LazyReporter, LazyStatusDict, SearchAlgorithm, SinglePopulationAlgorithmMixin
@@ -18,6 +19,7 @@ __all__ = [
     "CMAES",
     "Cosyne",
     "GaussianSearchAlgorithm",
+    "GraphedSearch",
     "GeneticAlgorithm",
     "IPOP",
     "LazyReporter",

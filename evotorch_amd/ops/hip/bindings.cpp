@@ -3,6 +3,8 @@
 
 namespace ea {
 void sample_gaussian(torch::Tensor out, torch::Tensor mu, torch::Tensor sigma, bool symmetric, int64_t seed);
+void sample_gaussian_graphsafe(torch::Tensor out, torch::Tensor mu, torch::Tensor sigma, bool symmetric,
+                               torch::Tensor seed_buf);
 std::vector<torch::Tensor> es_gradients(torch::Tensor samples, torch::Tensor mu, torch::Tensor sigma,
                                         torch::Tensor weights, bool symmetric);
 std::vector<torch::Tensor> snes_gradients(torch::Tensor samples, torch::Tensor mu, torch::Tensor sigma,
@@ -20,6 +22,8 @@ torch::Tensor pareto_ranks(torch::Tensor utils);
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.doc() = "evotorch_amd gfx950 (MI355X / CDNA4) HIP kernels";
     m.def("sample_gaussian", &ea::sample_gaussian, "K1: philox Gaussian population sampling (plain/antithetic)");
+    m.def("sample_gaussian_graphsafe", &ea::sample_gaussian_graphsafe,
+          "K1 (hipGraph-safe): seed read from and advanced in device memory");
     m.def("es_gradients", &ea::es_gradients, "K3: fused (mu, sigma) ES gradient reduction");
     m.def("snes_gradients", &ea::snes_gradients, "K3: SNES raw-noise gradient reduction");
     m.def("clipup_step", &ea::clipup_step, "K4: fused ClipUp velocity update (no host sync)");

@@ -33,10 +33,29 @@ namespace ea {
 // K1: sampling
 // ---------------------------------------------------------------------------
 
+// Graph-safe seeding: when `seed_ptr` is non-null the philox seed is read
+// from device memory (and a separate bump kernel advances it), so a
+// hipGraph replay of the sampling kernel produces a fresh population each
+// replay — a by-value seed would be frozen into the captured graph.
+__device__ __forceinline__ uint64_t resolve_seed(uint64_t seed, const unsigned long long* seed_ptr) {
+    return seed_ptr ? (uint64_t)*seed_ptr : seed;
+}
+
+__global__ void bump_seed_kernel(unsigned long long* seed_ptr) {
+    // splitmix64 advance (single thread)
+    unsigned long long x = *seed_ptr + 0x9E3779B97F4A7C15ull;
+    unsigned long long z = x;
+    z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ull;
+    z = (z ^ (z >> 27)) * 0x94D049BB133111EBull;
+    *seed_ptr = (z ^ (z >> 31)) & 0x7FFFFFFFFFFFFFFFull;
+}
+
 template <typename T, bool kSymmetric>
 __global__ void sample_gaussian_kernel(T* __restrict__ out, const T* __restrict__ mu, const T* __restrict__ sigma,
                                        int64_t rows,  // = N (plain) or N/2 (symmetric)
-                                       int64_t length, uint64_t seed) {
+                                       int64_t length, uint64_t seed_in,
+                                       const unsigned long long* __restrict__ seed_ptr) {
+    const uint64_t seed = resolve_seed(seed_in, seed_ptr);
     const int64_t total4 = (rows * length + 3) / 4;
     for (int64_t idx4 = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; idx4 < total4;
          idx4 += (int64_t)gridDim.x * blockDim.x) {
@@ -66,7 +85,8 @@ __global__ void sample_gaussian_kernel(T* __restrict__ out, const T* __restrict_
 template <bool kSymmetric>
 __global__ void sample_gaussian_f32x4_kernel(float4* __restrict__ out, const float4* __restrict__ mu,
                                              const float4* __restrict__ sigma, int64_t rows, int64_t length4,
-                                             uint64_t seed) {
+                                             uint64_t seed_in, const unsigned long long* __restrict__ seed_ptr) {
+    const uint64_t seed = resolve_seed(seed_in, seed_ptr);
     const int64_t total4 = rows * length4;
     for (int64_t idx4 = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; idx4 < total4;
          idx4 += (int64_t)gridDim.x * blockDim.x) {
@@ -92,7 +112,8 @@ __global__ void sample_gaussian_f32x4_kernel(float4* __restrict__ out, const flo
     }
 }
 
-void sample_gaussian(torch::Tensor out, torch::Tensor mu, torch::Tensor sigma, bool symmetric, int64_t seed) {
+void sample_gaussian_impl(torch::Tensor out, torch::Tensor mu, torch::Tensor sigma, bool symmetric, int64_t seed,
+                          const unsigned long long* seed_ptr) {
     CHECK_GPU(out); CHECK_CONTIG(out); CHECK_GPU(mu); CHECK_GPU(sigma);
     const int64_t n = out.size(0), length = out.size(1);
     TORCH_CHECK(!symmetric || n % 2 == 0, "symmetric sampling needs even popsize");
@@ -105,11 +126,11 @@ void sample_gaussian(torch::Tensor out, torch::Tensor mu, torch::Tensor sigma, b
         if (symmetric) {
             hipLaunchKernelGGL((sample_gaussian_f32x4_kernel<true>), dim3(blocks), dim3(threads), 0, stream,
                                reinterpret_cast<float4*>(out.data_ptr<float>()), reinterpret_cast<const float4*>(mu.data_ptr<float>()),
-                               reinterpret_cast<const float4*>(sigma.data_ptr<float>()), rows, length / 4, (uint64_t)seed);
+                               reinterpret_cast<const float4*>(sigma.data_ptr<float>()), rows, length / 4, (uint64_t)seed, seed_ptr);
         } else {
             hipLaunchKernelGGL((sample_gaussian_f32x4_kernel<false>), dim3(blocks), dim3(threads), 0, stream,
                                reinterpret_cast<float4*>(out.data_ptr<float>()), reinterpret_cast<const float4*>(mu.data_ptr<float>()),
-                               reinterpret_cast<const float4*>(sigma.data_ptr<float>()), rows, length / 4, (uint64_t)seed);
+                               reinterpret_cast<const float4*>(sigma.data_ptr<float>()), rows, length / 4, (uint64_t)seed, seed_ptr);
         }
         return;
     }
@@ -117,12 +138,29 @@ void sample_gaussian(torch::Tensor out, torch::Tensor mu, torch::Tensor sigma, b
         using T = scalar_t;
         if (symmetric) {
             hipLaunchKernelGGL((sample_gaussian_kernel<T, true>), dim3(blocks), dim3(threads), 0, stream,
-                               out.data_ptr<T>(), mu.data_ptr<T>(), sigma.data_ptr<T>(), rows, length, (uint64_t)seed);
+                               out.data_ptr<T>(), mu.data_ptr<T>(), sigma.data_ptr<T>(), rows, length, (uint64_t)seed, seed_ptr);
         } else {
             hipLaunchKernelGGL((sample_gaussian_kernel<T, false>), dim3(blocks), dim3(threads), 0, stream,
-                               out.data_ptr<T>(), mu.data_ptr<T>(), sigma.data_ptr<T>(), rows, length, (uint64_t)seed);
+                               out.data_ptr<T>(), mu.data_ptr<T>(), sigma.data_ptr<T>(), rows, length, (uint64_t)seed, seed_ptr);
         }
     });
+}
+
+void sample_gaussian(torch::Tensor out, torch::Tensor mu, torch::Tensor sigma, bool symmetric, int64_t seed) {
+    sample_gaussian_impl(out, mu, sigma, symmetric, seed, nullptr);
+}
+
+// Graph-safe variant: the seed lives in `seed_buf` (int64 tensor of 1
+// element) and is advanced ON DEVICE after sampling, so hipGraph replays
+// draw fresh populations.
+void sample_gaussian_graphsafe(torch::Tensor out, torch::Tensor mu, torch::Tensor sigma, bool symmetric,
+                               torch::Tensor seed_buf) {
+    TORCH_CHECK(seed_buf.is_cuda() && seed_buf.scalar_type() == at::ScalarType::Long && seed_buf.numel() >= 1,
+                "seed_buf must be a cuda int64 tensor");
+    auto* ptr = reinterpret_cast<unsigned long long*>(seed_buf.data_ptr<int64_t>());
+    sample_gaussian_impl(out, mu, sigma, symmetric, 0, ptr);
+    auto stream = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(bump_seed_kernel, dim3(1), dim3(1), 0, stream, ptr);
 }
 
 // ---------------------------------------------------------------------------

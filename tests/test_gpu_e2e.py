@@ -83,3 +83,45 @@ def test_ga_singleobj_gpu():
     first = ga.status["pop_best_eval"]
     ga.run(20)
     assert ga.status["pop_best_eval"] < first
+
+
+@requires_gpu
+def test_graphed_snes_converges_and_resamples():
+    from evotorch_amd import Problem
+    from evotorch_amd.algorithms import SNES, GraphedSearch
+    from evotorch_amd.decorators import vectorized
+
+    @vectorized
+    def sphere(x):
+        return (x**2).sum(-1)
+
+    prob = Problem("min", sphere, solution_length=100, initial_bounds=(-5, 5), device="cuda:0", seed=1)
+    searcher = SNES(prob, stdev_init=5.0, popsize=500)
+    graphed = GraphedSearch(searcher)
+    graphed.capture()
+    first = graphed.mean_eval
+    pop_before = searcher.population.unsafe_values.clone()
+    graphed.run(100)
+    pop_after = searcher.population.unsafe_values
+    # replays resample (graph-safe seed bump) and the search descends
+    assert not torch.allclose(pop_before, pop_after)
+    assert graphed.mean_eval < first * 0.5, (first, graphed.mean_eval)
+    assert searcher.status["iter"] >= 100
+
+
+@requires_gpu
+def test_graphed_pgpe_clipup():
+    from evotorch_amd import Problem
+    from evotorch_amd.algorithms import PGPE, GraphedSearch
+    from evotorch_amd.decorators import vectorized
+
+    @vectorized
+    def sphere(x):
+        return (x**2).sum(-1)
+
+    prob = Problem("min", sphere, solution_length=64, initial_bounds=(-3, 3), device="cuda:0", seed=2)
+    searcher = PGPE(prob, popsize=100, center_learning_rate=0.2, stdev_learning_rate=0.1,
+                    stdev_init=2.0, optimizer="clipup")
+    graphed = GraphedSearch(searcher)
+    graphed.run(150)
+    assert graphed.mean_eval < 20.0
