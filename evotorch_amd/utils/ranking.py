@@ -50,11 +50,15 @@ def linear(fitnesses: torch.Tensor, *, higher_is_better: bool = True) -> torch.T
 def nes(fitnesses: torch.Tensor, *, higher_is_better: bool = True) -> torch.Tensor:
     """NES log-utilities (Wierstra et al. 2014), shifted to sum to ~0:
     ``u_i = max(0, log(n/2+1) - log(rank_from_best_i)) / Z - 1/n``."""
+    import math
+
     n = fitnesses.shape[-1]
     ranks = _ranks_ascending(fitnesses, higher_is_better)
-    # rank-from-best: best solution gets 1
+    # rank-from-best: best solution gets 1. The log(n/2+1) constant stays a
+    # python float (a device-tensor construction here would be an H2D copy,
+    # which hipGraph capture forbids).
     rank_from_best = (n - ranks).to(torch.float32)
-    util = torch.clamp(torch.log(torch.tensor(n / 2.0 + 1.0, device=fitnesses.device)) - torch.log(rank_from_best), min=0.0)
+    util = torch.clamp(math.log(n / 2.0 + 1.0) - torch.log(rank_from_best), min=0.0)
     denom = util.sum(dim=-1, keepdim=True)
     return util / denom - 1.0 / n
 
