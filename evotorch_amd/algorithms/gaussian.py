@@ -28,6 +28,7 @@ from ..distributions import (
 from ..optimizers import get_optimizer_class
 from ..utils import RealOrVector, modify_tensor, to_stdev_init
 from ..utils.misc import ensure_tensor_length_and_dtype
+from ..utils.profiling import record_range
 from .searchalgorithm import SearchAlgorithm, SinglePopulationAlgorithmMixin
 
 __all__ = ["GaussianSearchAlgorithm", "PGPE", "SNES", "CEM", "XNES"]
@@ -202,10 +203,12 @@ class GaussianSearchAlgorithm(SearchAlgorithm, SinglePopulationAlgorithmMixin):
             self._fill_and_eval_pop()
         fitnesses = self._population.access_evals()[:, self._obj_index]
         obj_sense = self.problem.senses[self._obj_index]
-        gradients = self._distribution.compute_gradients(
-            self._population.unsafe_values, fitnesses, objective_sense=obj_sense, ranking_method=self._ranking_method
-        )
-        self._update_distribution(gradients)
+        with record_range("rank+grad"):
+            gradients = self._distribution.compute_gradients(
+                self._population.unsafe_values, fitnesses, objective_sense=obj_sense, ranking_method=self._ranking_method
+            )
+        with record_range("update"):
+            self._update_distribution(gradients)
         self._fill_and_eval_pop()
 
     def _update_distribution(self, gradients: dict):

@@ -42,6 +42,7 @@ from .utils import (
     to_torch_dtype,
 )
 from .utils.misc import ensure_tensor_length_and_dtype, split_workload
+from .utils.profiling import record_range
 
 _logger = logging.getLogger(__name__)
 
@@ -1086,8 +1087,10 @@ class Problem(TensorMakerMixin, Serializable):
     ) -> dict:
         popsize = self._sample_popsize(distribution, popsize, num_interactions, popsize_max, ensure_even_popsize)
         batch = self._get_cached_grad_batch(popsize, self._device)
-        distribution.sample(out=batch.access_values(), generator=self._generator)
-        self.evaluate(batch)
+        with record_range("sample"):
+            distribution.sample(out=batch.access_values(), generator=self._generator)
+        with record_range("evaluate"):
+            self.evaluate(batch)
         fitnesses = batch._evals[:, obj_index]
         sense = self._senses[obj_index]
         grads = distribution.compute_gradients(batch._values, fitnesses, objective_sense=sense, ranking_method=ranking_method)
