@@ -2,19 +2,26 @@
 // policies through the synthetic vectorized environment, one kernel launch
 // per generation (SURVEY.md §3.4 — the VecGymNE hot loop, collapsed).
 //
-// MI355X design (v2): one workgroup per population member. The member's
-// policy weights AND the (shared) environment matrices are staged once
-// into LDS and stay resident for the entire T-step episode — the inner
-// loop touches no global memory at all. All matrix operands are bf16 and
-// every inner product runs on `v_dot2_f32_bf16` (2 bf16 MACs/instruction,
-// fp32 accumulate, no per-element converts). LDS layouts:
-//   * wave-reduced dots (policy W, dynamics factor V): row-major [out][O],
-//     lanes read consecutive bf16x2 along O — conflict-free.
-//   * per-thread dots (U, D2): per-OUTPUT rows padded to stride 18
-//     (36 B ⇒ bank index advances by 9 per lane, gcd(9,32)=1 ⇒ ≤2-way,
-//     i.e. free on CDNA4 — see cdna_hip_programming.md §6 G4).
+// MI355X design (v5): one workgroup rolls out kMembers (default 2)
+// population members for the entire T-step episode with policy weights
+// AND the shared environment matrices resident in LDS — the inner loop
+// never touches global memory. Multi-member blocks exist because one
+// member's per-step work underfills a 512-thread block (phase 1 has only
+// 33 reduction outputs for 64 lane-groups; phase 2 has 376 dims for 512
+// lanes): with 2 members both phases nearly double their parallel work
+// per barrier, and gfx950 allows the required ~75 KB of dynamic LDS per
+// workgroup (sharedMemPerBlock = 160 KB, probed).
+//
+// All inner products run on `v_dot2_f32_bf16` (2 bf16 MACs/instruction,
+// fp32 accumulate). LDS access patterns:
+//   * group-reduced dots (policy W, dynamics factor V): row-major
+//     [out][O]; 8-lane groups read consecutive bf16x2 — conflict-free,
+//     with a 3-level shuffle reduction (64-lane trees serialized 6
+//     dependent shuffles; measured 10:1 SQ_WAIT:SQ_BUSY).
+//   * per-thread dots (U, D2): PAIR-INTERLEAVED COLUMN-MAJOR [K/2][O]
+//     bf16x2 — one 4 B read per dot2 at lane-consecutive addresses.
 // Observation-normalization statistics (sum, sumsq) accumulate per-thread
-// in registers and are merged with one atomic pass at the end (K11; they
+// in registers and merge with one atomic pass at the end (K11; they
 // become a single RCCL all-reduce across ranks — SURVEY.md §2.8 P5).
 //
 // Environment spec (must match the eager reference in
@@ -31,6 +38,8 @@
 #include <torch/extension.h>
 #include <ATen/cuda/CUDAContext.h>
 
+#include <cstdlib>
+
 #include "philox.h"
 #include "reduce.h"
 
@@ -42,6 +51,14 @@ typedef __attribute__((ext_vector_type(2))) __bf16 bf16x2;
 
 __device__ __forceinline__ __bf16 f2b(float v) { return (__bf16)v; }
 __device__ __forceinline__ float b2f(__bf16 v) { return (float)v; }
+
+// runtime member-index -> pointer select WITHOUT a runtime-indexed array
+// (which would demote to scratch memory — common-mistake #20); for
+// kMembers<=2 this folds to a single v_cndmask per operand.
+template <typename T>
+__device__ __forceinline__ T* sel2(int m, T* p0, T* p1) {
+    return m == 0 ? p0 : p1;
+}
 
 struct RolloutArgs {
     const float* params;      // [n_members][A*O + A]  (W row-major, then b)
@@ -58,47 +75,57 @@ struct RolloutArgs {
 // env_blob layout (fp32): V [R][O] · U_T [R][O] · D2_T [A][O] · c [O] ·
 // wr [O] · mean [O] · std [O]
 
-template <int kGroup>
+template <int kGroup, int kMembers>
 __global__ __launch_bounds__(512, 2) void rollout_linear_kernel(RolloutArgs args) {
     const int O = args.obs_dim, A = args.act_dim, R = args.rank;
     const int tid = threadIdx.x;
-    const int member = blockIdx.x;
-    if (member >= args.n_members) return;
-    const int A_PAD = (A + 2) & ~1;  // D2 per-output row stride (18 for A=17)
-    const int R_PAD = (R + 2) & ~1;  // U per-output row stride
+    const int base_member = blockIdx.x * kMembers;
+    if (base_member >= args.n_members) return;
+    const int live_members = min(kMembers, args.n_members - base_member);
+    const int A_PAD = (A + 2) & ~1;
+    const int R_PAD = (R + 2) & ~1;
 
     extern __shared__ unsigned char lds_raw[];
-    __bf16* W_l = reinterpret_cast<__bf16*>(lds_raw);   // [A][O] row-major
-    __bf16* V_l = W_l + A * O;                          // [R][O] row-major
-    // U and D2 are stored PAIR-INTERLEAVED COLUMN-MAJOR: element [i2][j]
-    // is the bf16x2 (U_T[2*i2][j], U_T[2*i2+1][j]). Phase 2's thread j then
-    // reads one 4 B bf16x2 per dot2 at consecutive-lane-consecutive-word
-    // addresses — conflict-free b32 LDS reads. (The previous padded
-    // per-output-row layout let the compiler merge the 9 contiguous pair
-    // reads into strided ds_read_b128s: 1.16e9 bank-conflict cycles per
-    // bench run, 58% of all LDS instructions.)
-    bf16x2* U_pair = reinterpret_cast<bf16x2*>(V_l + R * O);  // [R_PAD/2][O]
-    bf16x2* D2_pair = U_pair + (R_PAD / 2) * O;               // [A_PAD/2][O]
-    __bf16* obs_b = reinterpret_cast<__bf16*>(D2_pair + (A_PAD / 2) * O);  // [O]
-    __bf16* obsn_b = obs_b + O;                         // [O]    normalized
-    __bf16* h_b = obsn_b + O;                           // [R_PAD]
-    __bf16* act_b = h_b + R_PAD;                        // [A_PAD]
-    float* c_l = reinterpret_cast<float*>(act_b + A_PAD);  // [O]
-    float* wr_l = c_l + O;                              // [O]
-    float* mean_l = wr_l + O;                           // [O]
-    float* istd_l = mean_l + O;                         // [O]
-    float* b_l = istd_l + O;                            // [A]
-    float* scratch = b_l + A;                           // [8]
+    // ---- shared environment ----
+    __bf16* V_l = reinterpret_cast<__bf16*>(lds_raw);          // [R][O]
+    bf16x2* U_pair = reinterpret_cast<bf16x2*>(V_l + R * O);   // [R_PAD/2][O]
+    bf16x2* D2_pair = U_pair + (R_PAD / 2) * O;                // [A_PAD/2][O]
+    float* c_l = reinterpret_cast<float*>(D2_pair + (A_PAD / 2) * O);  // [O]
+    float* wr_l = c_l + O;                                     // [O]
+    float* mean_l = wr_l + O;                                  // [O]
+    float* istd_l = mean_l + O;                                // [O]
+    float* scratch = istd_l + O;                               // [16]
+    // ---- per-member state ----
+    __bf16* W_l[kMembers];
+    float* b_l[kMembers];
+    __bf16* obs_b[kMembers];
+    __bf16* obsn_b[kMembers];
+    __bf16* h_b[kMembers];
+    __bf16* act_b[kMembers];
+    {
+        unsigned char* cursor = reinterpret_cast<unsigned char*>(scratch + 16);
+#pragma unroll
+        for (int m = 0; m < kMembers; ++m) {
+            W_l[m] = reinterpret_cast<__bf16*>(cursor);
+            cursor += (size_t)A * O * 2;
+            b_l[m] = reinterpret_cast<float*>(cursor);
+            cursor += (size_t)A * 4;
+            obs_b[m] = reinterpret_cast<__bf16*>(cursor);
+            cursor += (size_t)O * 2;
+            obsn_b[m] = reinterpret_cast<__bf16*>(cursor);
+            cursor += (size_t)O * 2;
+            h_b[m] = reinterpret_cast<__bf16*>(cursor);
+            cursor += (size_t)R_PAD * 2;
+            act_b[m] = reinterpret_cast<__bf16*>(cursor);
+            cursor += (size_t)A_PAD * 2;
+        }
+    }
 
-    // ---- stage ----
-    const float* my_params = args.params + (long)member * (A * O + A);
-    for (int i = tid; i < A * O; i += blockDim.x) W_l[i] = f2b(my_params[i]);
-    for (int i = tid; i < A; i += blockDim.x) b_l[i] = my_params[A * O + i];
+    // ---- stage shared env ----
     {
         const float* e = args.env_blob;
         const int RO = R * O, AO = A * O;
         for (int i = tid; i < RO; i += blockDim.x) V_l[i] = f2b(e[i]);
-        // U arrives as U_T [R][O]; store bf16x2 pairs along R, column-major
         for (int i = tid; i < (R_PAD / 2) * O; i += blockDim.x) {
             const int i2 = i / O, j = i % O;
             bf16x2 v2;
@@ -120,67 +147,91 @@ __global__ __launch_bounds__(512, 2) void rollout_linear_kernel(RolloutArgs args
             mean_l[j] = tail[2 * O + j];
             istd_l[j] = 1.0f / tail[3 * O + j];
         }
-        // zero the h/act pads once so dot2 over padded vectors is exact
-        for (int i = tid; i < R_PAD; i += blockDim.x) h_b[i] = f2b(0.0f);
-        for (int i = tid; i < A_PAD; i += blockDim.x) act_b[i] = f2b(0.0f);
     }
-    // initial observation: 0.1 * N(0,1), deterministic per global member id
-    const unsigned long long gmember = (unsigned long long)(args.member_offset + member);
-    for (int j4 = tid; j4 * 4 < O; j4 += blockDim.x) {
-        float z[4];
-        philox_normal4(args.init_seed, (uint32_t)gmember, (uint64_t)j4, z);
+    // ---- stage members: weights, pads, initial observations ----
 #pragma unroll
-        for (int u = 0; u < 4; ++u) {
-            const int j = j4 * 4 + u;
-            if (j < O) obs_b[j] = f2b(0.1f * z[u]);
+    for (int m = 0; m < kMembers; ++m) {
+        if (m >= live_members) break;
+        const float* my_params = args.params + (long)(base_member + m) * (A * O + A);
+        for (int i = tid; i < A * O; i += blockDim.x) W_l[m][i] = f2b(my_params[i]);
+        for (int i = tid; i < A; i += blockDim.x) b_l[m][i] = my_params[A * O + i];
+        for (int i = tid; i < R_PAD; i += blockDim.x) h_b[m][i] = f2b(0.0f);
+        for (int i = tid; i < A_PAD; i += blockDim.x) act_b[m][i] = f2b(0.0f);
+        const unsigned long long gmember = (unsigned long long)(args.member_offset + base_member + m);
+        for (int j4 = tid; j4 * 4 < O; j4 += blockDim.x) {
+            float z[4];
+            philox_normal4(args.init_seed, (uint32_t)gmember, (uint64_t)j4, z);
+#pragma unroll
+            for (int u = 0; u < 4; ++u) {
+                const int j = j4 * 4 + u;
+                if (j < O) obs_b[m][j] = f2b(0.1f * z[u]);
+            }
         }
     }
     __syncthreads();
 
     const int lane = tid & (kWaveSize - 1);
     const int wave = tid / kWaveSize;
-    const int nwaves = blockDim.x / kWaveSize;
-    const int n_pairs = O / 2;          // O must be even
-    const int n_outputs = A + R;        // group-reduced dots per step
-
-    // phase-1 layout: 8-lane groups, one output per group per round — the
-    // wave computes 8 dots concurrently with only a 3-level shuffle
-    // reduction (a 64-lane tree would serialize 6 dependent shuffles per
-    // output; measured 10:1 SQ_WAIT:SQ_BUSY). With 4 waves × 8 groups,
-    // all 33 outputs finish in ceil(33/32) = 2 rounds.
-    constexpr int kRounds = 2;  // supports up to 2 * threads/kGroup outputs
+    const int n_pairs = O / 2;              // O must be even
+    const int outs_per_member = A + R;
+    const int n_outputs = live_members * outs_per_member;
+    constexpr int kRounds = 2;              // up to 2 * threads/kGroup outputs
     const int group = lane / kGroup;
     const int glane = lane % kGroup;
-    const int groups_per_block = (int)(blockDim.x / kGroup);  // threads/8
+    const int groups_per_block = (int)(blockDim.x / kGroup);
+    const int groups_per_wave = kWaveSize / kGroup;
 
+    // per-round output assignment, hoisted: everything below is FULLY
+    // UNROLLED over kRounds (runtime-indexed arrays would demote to
+    // scratch — cdna_hip_programming.md common-mistake #20)
     const __bf16* my_row[kRounds];
+    const __bf16* my_vec[kRounds];
+    float* my_bias[kRounds];
+    __bf16* my_dst[kRounds];
     bool my_is_act[kRounds];
     bool my_valid[kRounds];
-    int my_out[kRounds];
-    const int groups_per_wave = kWaveSize / kGroup;
+    int my_member[kRounds];
 #pragma unroll
     for (int r = 0; r < kRounds; ++r) {
         const int out = (wave * groups_per_wave + group) + r * groups_per_block;
         my_valid[r] = out < n_outputs;
-        my_out[r] = my_valid[r] ? out : 0;
-        my_is_act[r] = my_out[r] < A;
-        my_row[r] = my_is_act[r] ? (W_l + my_out[r] * O) : (V_l + (my_out[r] - A) * O);
+        const int safe_out = my_valid[r] ? out : 0;
+        const int m = safe_out / outs_per_member;
+        const int lo = safe_out % outs_per_member;
+        my_member[r] = m;
+        my_is_act[r] = lo < A;
+        __bf16* w_m = sel2(m, W_l[0], W_l[kMembers - 1]);
+        float* b_m = sel2(m, b_l[0], b_l[kMembers - 1]);
+        __bf16* obs_m = sel2(m, obs_b[0], obs_b[kMembers - 1]);
+        __bf16* obsn_m = sel2(m, obsn_b[0], obsn_b[kMembers - 1]);
+        __bf16* h_m = sel2(m, h_b[0], h_b[kMembers - 1]);
+        __bf16* act_m = sel2(m, act_b[0], act_b[kMembers - 1]);
+        my_row[r] = my_is_act[r] ? (w_m + lo * O) : (V_l + (lo - A) * O);
+        my_vec[r] = my_is_act[r] ? obsn_m : obs_m;
+        my_bias[r] = my_is_act[r] ? (b_m + lo) : nullptr;
+        my_dst[r] = my_is_act[r] ? (act_m + lo) : (h_m + (lo - A));
     }
 
-    float fit_part = 0.0f;
-    float actsq_part = 0.0f;
+    float fit_part[kMembers];
+    float actsq_part[kMembers];
+#pragma unroll
+    for (int m = 0; m < kMembers; ++m) {
+        fit_part[m] = 0.0f;
+        actsq_part[m] = 0.0f;
+    }
     float stat_sum[2] = {0.0f, 0.0f}, stat_sumsq[2] = {0.0f, 0.0f};
 
     // initial normalization (later steps fuse it into phase 2's epilogue)
-    for (int j = tid; j < O; j += blockDim.x) {
-        obsn_b[j] = f2b((b2f(obs_b[j]) - mean_l[j]) * istd_l[j]);
+    for (int j = tid; j < live_members * O; j += blockDim.x) {
+        const int m = j / O, jo = j % O;
+        __bf16* obs_m = sel2(m, obs_b[0], obs_b[kMembers - 1]);
+        __bf16* obsn_m = sel2(m, obsn_b[0], obsn_b[kMembers - 1]);
+        obsn_m[jo] = f2b((b2f(obs_m[jo]) - mean_l[jo]) * istd_l[jo]);
     }
     __syncthreads();
 
     for (int t = 0; t < args.steps; ++t) {
-        // phase 1: all of this wave's dots concurrently (policy rows read
-        // obsn, dynamics-factor rows read obs; both vectors loaded once
-        // per pair-column and selected per output)
+        // ---- phase 1: group-reduced dots (policy actions + dynamics h) ----
         float acc[kRounds];
 #pragma unroll
         for (int r = 0; r < kRounds; ++r) acc[r] = 0.0f;
@@ -188,9 +239,9 @@ __global__ __launch_bounds__(512, 2) void rollout_linear_kernel(RolloutArgs args
         for (int r = 0; r < kRounds; ++r) {
             if (!my_valid[r]) continue;
             const __bf16* row = my_row[r];
-            const __bf16* vec = my_is_act[r] ? obsn_b : obs_b;
+            const __bf16* vec = my_vec[r];
             float a0 = 0.0f, a1 = 0.0f;  // two chains: halve the serial depth
-            const int full_iters = n_pairs / kGroup;   // guard-free iterations
+            const int full_iters = n_pairs / kGroup;
             const int tail = n_pairs % kGroup;
             for (int i = 0; i + 1 < full_iters; i += 2) {
                 const int p0 = glane + i * kGroup;
@@ -216,7 +267,7 @@ __global__ __launch_bounds__(512, 2) void rollout_linear_kernel(RolloutArgs args
             }
             acc[r] = a0 + a1;
         }
-        // 3-level in-group reductions (both rounds' shuffles overlap)
+        // in-group reductions (rounds' shuffles overlap)
 #pragma unroll
         for (int offset = kGroup / 2; offset > 0; offset >>= 1) {
 #pragma unroll
@@ -229,60 +280,82 @@ __global__ __launch_bounds__(512, 2) void rollout_linear_kernel(RolloutArgs args
             for (int r = 0; r < kRounds; ++r) {
                 if (!my_valid[r]) continue;
                 if (my_is_act[r]) {
-                    float a = fminf(fmaxf(acc[r] + b_l[my_out[r]], -1.0f), 1.0f);
-                    act_b[my_out[r]] = f2b(a);
-                    actsq_part = fmaf(a, a, actsq_part);
+                    const float a = fminf(fmaxf(acc[r] + *my_bias[r], -1.0f), 1.0f);
+                    *my_dst[r] = f2b(a);
+#pragma unroll
+                    for (int m = 0; m < kMembers; ++m) {
+                        if (my_member[r] == m) actsq_part[m] = fmaf(a, a, actsq_part[m]);
+                    }
                 } else {
-                    h_b[my_out[r] - A] = f2b(acc[r]);
+                    *my_dst[r] = f2b(acc[r]);
                 }
             }
         }
         __syncthreads();
 
-        // phase 2: per-thread dynamics rows (padded-stride LDS reads) with
-        // the NEXT step's normalization fused into the epilogue
-        for (int j = tid; j < O; j += blockDim.x) {
-            // two independent dot chains (U·h and D2·a) halve the serial
-            // dot2 dependency depth; joined at the end. Each operand read
-            // is one bf16x2 at lane-consecutive addresses (conflict-free).
-            float uacc = c_l[j], dacc = 0.0f;
+        // ---- phase 2: per-thread dynamics dims, norm fused in epilogue ----
+        for (int j = tid; j < live_members * O; j += blockDim.x) {
+            const int m = j / O, jo = j % O;
+            float uacc = c_l[jo], dacc = 0.0f;
+            const __bf16* hv = sel2(m, h_b[0], h_b[kMembers - 1]);
+            const __bf16* av = sel2(m, act_b[0], act_b[kMembers - 1]);
 #pragma unroll
             for (int p = 0; p < R_PAD / 2; ++p) {
                 uacc = __builtin_amdgcn_fdot2_f32_bf16(
-                    U_pair[p * O + j],
-                    *reinterpret_cast<const bf16x2*>(h_b + 2 * p), uacc, false);
+                    U_pair[p * O + jo], *reinterpret_cast<const bf16x2*>(hv + 2 * p), uacc, false);
             }
 #pragma unroll
             for (int p = 0; p < A_PAD / 2; ++p) {
                 dacc = __builtin_amdgcn_fdot2_f32_bf16(
-                    D2_pair[p * O + j],
-                    *reinterpret_cast<const bf16x2*>(act_b + 2 * p), dacc, false);
+                    D2_pair[p * O + jo], *reinterpret_cast<const bf16x2*>(av + 2 * p), dacc, false);
             }
             const float o_new = tanhf(uacc + dacc);
-            fit_part = fmaf(wr_l[j], o_new, fit_part);
+#pragma unroll
+            for (int mm = 0; mm < kMembers; ++mm) {
+                if (m == mm) fit_part[mm] = fmaf(wr_l[jo], o_new, fit_part[mm]);
+            }
             const int slot = j >= (int)blockDim.x;
             stat_sum[slot] += o_new;
             stat_sumsq[slot] = fmaf(o_new, o_new, stat_sumsq[slot]);
             const __bf16 ob = f2b(o_new);
-            obs_b[j] = ob;
-            obsn_b[j] = f2b((b2f(ob) - mean_l[j]) * istd_l[j]);
+            __bf16* obs_m = sel2(m, obs_b[0], obs_b[kMembers - 1]);
+            __bf16* obsn_m = sel2(m, obsn_b[0], obsn_b[kMembers - 1]);
+            obs_m[jo] = ob;
+            obsn_m[jo] = f2b((b2f(ob) - mean_l[jo]) * istd_l[jo]);
         }
         __syncthreads();
     }
 
-    // ---- wrap-up: fitness reduction + obs-stat atomics ----
-    float total = block_reduce_sum<false>(fit_part, scratch);
-    __syncthreads();
-    float act_total = block_reduce_sum<false>(actsq_part, scratch);
-    if (tid == 0) {
-        args.fitness_out[member] =
-            total + args.alive_bonus * args.steps - args.act_cost * act_total / (float)A;
+    // ---- wrap-up: per-member fitness reductions + obs-stat atomics ----
+#pragma unroll
+    for (int m = 0; m < kMembers; ++m) {
+        if (m >= live_members) break;
+        const float total = block_reduce_sum<false>(fit_part[m], scratch);
+        __syncthreads();
+        const float act_total = block_reduce_sum<false>(actsq_part[m], scratch);
+        __syncthreads();
+        if (tid == 0) {
+            args.fitness_out[base_member + m] =
+                total + args.alive_bonus * args.steps - args.act_cost * act_total / (float)A;
+        }
     }
-    for (int j = tid; j < O; j += blockDim.x) {
+    for (int j = tid; j < live_members * O; j += blockDim.x) {
         const int slot = j >= (int)blockDim.x;
-        atomicAdd(&args.obs_stats_out[j], stat_sum[slot]);
-        atomicAdd(&args.obs_stats_out[O + j], stat_sumsq[slot]);
+        const int jo = j % O;
+        atomicAdd(&args.obs_stats_out[jo], stat_sum[slot]);
+        atomicAdd(&args.obs_stats_out[O + jo], stat_sumsq[slot]);
     }
+}
+
+template <int kGroup, int kMembers>
+static void launch_rollout(int n_blocks, int block, size_t lds_bytes, hipStream_t stream, const RolloutArgs& args) {
+    static bool attr_set = false;
+    if (!attr_set && lds_bytes > 64 * 1024) {
+        (void)hipFuncSetAttribute(reinterpret_cast<const void*>(&rollout_linear_kernel<kGroup, kMembers>),
+                                  hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
+        attr_set = true;
+    }
+    hipLaunchKernelGGL((rollout_linear_kernel<kGroup, kMembers>), dim3(n_blocks), dim3(block), lds_bytes, stream, args);
 }
 
 torch::Tensor rollout_linear(torch::Tensor params, torch::Tensor env_blob, torch::Tensor obs_stats_out,
@@ -295,7 +368,6 @@ torch::Tensor rollout_linear(torch::Tensor params, torch::Tensor env_blob, torch
     const int O = (int)obs_dim, A = (int)act_dim, R = (int)rank;
     TORCH_CHECK(params.size(1) == (int64_t)A * O + A, "param length mismatch");
     TORCH_CHECK(O % 2 == 0, "obs_dim must be even (bf16x2 packing)");
-    TORCH_CHECK(O <= 512, "obs_dim too large for the 2-slot stat accumulators");
     auto fitness = torch::empty({n}, params.options());
 
     RolloutArgs args;
@@ -311,23 +383,23 @@ torch::Tensor rollout_linear(torch::Tensor params, torch::Tensor env_blob, torch
     args.act_cost = (float)act_cost;
     args.init_seed = (unsigned long long)init_seed;
 
+    int members = 2;
+    if (const char* env = getenv("EVOTORCH_AMD_ROLLOUT_MEMBERS")) members = atoi(env);
+    TORCH_CHECK(members == 1 || members == 2, "EVOTORCH_AMD_ROLLOUT_MEMBERS must be 1 or 2");
+    const int block = 512;
+    TORCH_CHECK(O * members <= 2 * block, "obs_dim too large for the 2-slot stat accumulators");
+
     const int A_PAD = (A + 2) & ~1, R_PAD = (R + 2) & ~1;
-    const size_t bf_elems = (size_t)A * O + (size_t)R * O + (size_t)O * R_PAD + (size_t)O * A_PAD +
-                            2 * (size_t)O + R_PAD + A_PAD;
-    const size_t f32_elems = 4 * (size_t)O + (size_t)A + 8;
-    const size_t lds_bytes = bf_elems * 2 + f32_elems * 4 + 64;
-    TORCH_CHECK(lds_bytes <= 64 * 1024, "rollout LDS footprint too large: ", lds_bytes,
-                " bytes (reduce rank / dims)");
+    const size_t shared_bf = (size_t)R * O + (size_t)R_PAD * O + (size_t)A_PAD * O;
+    const size_t member_bytes = ((size_t)A * O + 2 * (size_t)O + R_PAD + A_PAD) * 2 + (size_t)A * 4;
+    const size_t lds_bytes = shared_bf * 2 + 4 * (size_t)O * 4 + 16 * 4 + 64 + (size_t)members * member_bytes;
+    TORCH_CHECK(lds_bytes <= 160 * 1024, "rollout LDS footprint too large: ", lds_bytes, " bytes");
     auto stream = at::cuda::getCurrentCUDAStream();
-    int block = 512;
-    if (const char* env = getenv("EVOTORCH_AMD_ROLLOUT_BLOCK")) block = atoi(env);
-    TORCH_CHECK(block == 256 || block == 512, "EVOTORCH_AMD_ROLLOUT_BLOCK must be 256 or 512");
-    int group = 8;  // A/B-measured faster than 16 at O=376 (14.2 vs 15.5 us/step)
-    if (const char* env = getenv("EVOTORCH_AMD_ROLLOUT_GROUP")) group = atoi(env);
-    if (group == 8) {
-        hipLaunchKernelGGL((rollout_linear_kernel<8>), dim3(n), dim3(block), lds_bytes, stream, args);
+    const int n_blocks = (n + members - 1) / members;
+    if (members == 2) {
+        launch_rollout<8, 2>(n_blocks, block, lds_bytes, stream, args);
     } else {
-        hipLaunchKernelGGL((rollout_linear_kernel<16>), dim3(n), dim3(block), lds_bytes, stream, args);
+        launch_rollout<8, 1>(n_blocks, block, lds_bytes, stream, args);
     }
     return fitness;
 }
