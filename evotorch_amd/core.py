@@ -1168,9 +1168,9 @@ class ProblemBoundEvaluator:
 
     def __init__(self, problem: Problem, *, obj_index: Optional[int] = None):
         self._problem = problem
-        self._obj_index = 0 if obj_index is None else problem._senses.index(problem._senses[obj_index]) if isinstance(obj_index, int) else 0
-        if obj_index is not None:
-            self._obj_index = int(obj_index)
+        self._obj_index = 0 if obj_index is None else int(obj_index)
+        if not (0 <= self._obj_index < len(problem.senses)):
+            raise IndexError(f"obj_index {obj_index} out of range for {len(problem.senses)} objectives")
 
     @property
     def problem(self) -> Problem:

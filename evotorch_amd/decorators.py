@@ -84,65 +84,70 @@ def _call_with_ndims(fn: Callable, ndims: tuple, args: tuple, randomness: str):
     return torch.func.vmap(inner, in_dims=in_dims, randomness=randomness)(*prepared)
 
 
-def expects_ndim(
-    fn: Optional[Callable] = None,
-    expected_ndims: Optional[Iterable[Optional[int]]] = None,
-    *,
-    allow_smaller_ndim: bool = False,
-    randomness: str = "error",
-) -> Callable:
-    """Wrap `fn` so that each positional argument is validated against its
-    expected ndim and extra leftmost dimensions are vmapped over.
+def expects_ndim(*spec, allow_smaller_ndim: bool = False, randomness: str = "error") -> Callable:
+    """Wrap a function so that each positional argument is validated
+    against its expected ndim and extra leftmost dimensions are vmapped
+    over (reference decorators.py:613).
 
-    Usable both as `expects_ndim(fn, (None, 1, 1))` and as a decorator
-    `@expects_ndim(1, 1)`.
+    Usable in all of the reference's forms:
+      * `expects_ndim(fn, (None, 1, 1))`           — direct wrap
+      * `@expects_ndim(1, 1)` / `@expects_ndim(0, (1,), 2)` — decorator
+        with per-argument ndims as positional arguments (tuples are
+        flattened to their single element when length-1; `None` = leave
+        the argument untouched)
     """
-    # decorator-style: expects_ndim(1, 1, ...) or expects_ndim((1,1))
-    if fn is not None and not callable(fn):
-        if expected_ndims is None:
-            expected_ndims = (fn,)
+
+    def norm(nd):
+        if isinstance(nd, (tuple, list)):
+            if len(nd) != 1:
+                raise ValueError(f"Cannot interpret the ndim spec {nd!r}")
+            return int(nd[0])
+        return None if nd is None else int(nd)
+
+    if len(spec) >= 1 and callable(spec[0]):
+        fn = spec[0]
+        if len(spec) == 2 and isinstance(spec[1], (tuple, list)):
+            ndims = tuple(None if n is None else int(n) for n in spec[1])
         else:
-            expected_ndims = (fn, expected_ndims)
-        fn = None
-    if fn is None:
-        ndims_outer = expected_ndims
+            ndims = tuple(norm(n) for n in spec[1:])
 
-        def decorator(inner_fn: Callable, _nd=ndims_outer) -> Callable:
-            return expects_ndim(inner_fn, _nd, allow_smaller_ndim=allow_smaller_ndim, randomness=randomness)
+        @functools.wraps(fn)
+        def wrapped(*args):
+            if len(args) != len(ndims):
+                raise TypeError(f"{fn.__name__} expects {len(ndims)} positional arguments, got {len(args)}")
+            return _call_with_ndims(fn, ndims, args, randomness)
 
-        # Support @expects_ndim(1, None, 2) with multiple scalar args
-        def flexible_decorator(*args, **kwargs):
-            if len(args) == 1 and callable(args[0]) and not kwargs:
-                return decorator(args[0])
-            raise TypeError("expects_ndim decorator takes exactly the function")
+        wrapped.__expects_ndim__ = ndims
+        return wrapped
 
-        return flexible_decorator
+    # decorator form: the positional args are the ndim spec
+    ndim_spec = spec[0] if (len(spec) == 1 and isinstance(spec[0], (tuple, list)) and len(spec[0]) != 1) else spec
 
-    ndims = tuple(expected_ndims) if expected_ndims is not None else ()
+    def decorator(inner_fn: Callable) -> Callable:
+        return expects_ndim(inner_fn, tuple(ndim_spec), allow_smaller_ndim=allow_smaller_ndim, randomness=randomness)
 
-    @functools.wraps(fn)
-    def wrapped(*args):
-        if len(args) != len(ndims):
-            raise TypeError(f"{fn.__name__} expects {len(ndims)} positional arguments, got {len(args)}")
-        return _call_with_ndims(fn, ndims, args, randomness)
-
-    wrapped.__expects_ndim__ = ndims
-    return wrapped
+    return decorator
 
 
 def _expects_ndim_varargs(fn: Callable, per_arg_ndim: int, randomness: str) -> Callable:
     @functools.wraps(fn)
     def wrapped(*args):
-        ndims = tuple(per_arg_ndim if isinstance(a, (torch.Tensor,)) or _is_numeric(a) else None for a in args)
+        # tensors/arrays are treated as rows to map over; python scalars
+        # pass through untouched
+        ndims = tuple(per_arg_ndim if _is_rowlike(a) else None for a in args)
         return _call_with_ndims(fn, ndims, args, randomness)
 
     return wrapped
 
 
-def _is_numeric(a) -> bool:
+def _is_rowlike(a) -> bool:
     import numpy as np
 
-    return isinstance(a, (int, float, np.ndarray, list, tuple))
+    if isinstance(a, torch.Tensor):
+        return True
+    if isinstance(a, np.ndarray):
+        return a.ndim >= 1
+    return isinstance(a, (list, tuple))
 
 
 def rowwise(fn: Optional[Callable] = None, *, randomness: str = "error") -> Callable:

@@ -103,3 +103,32 @@ def test_tools_alias():
 
     assert tools.rank is utils.rank
     assert tools.ObjectArray is utils.ObjectArray
+
+
+def test_expects_ndim_decorator_variadic():
+    @expects_ndim(1, None, 1)
+    def axpy(x, alpha, y):
+        return alpha * x + y
+
+    x = torch.randn(4, 3)
+    y = torch.randn(3)
+    out = axpy(x, 2.0, y)
+    assert out.shape == (4, 3)
+    assert torch.allclose(out, 2.0 * x + y, atol=1e-6)
+
+
+def test_expects_ndim_direct_tuple():
+    def dot(a, b):
+        return (a * b).sum()
+
+    f = expects_ndim(dot, (1, 1))
+    assert float(f(torch.ones(3), torch.ones(3))) == 3.0
+
+
+def test_rowwise_with_scalar_arg():
+    @rowwise
+    def scale_row(x, k):
+        return x * k
+
+    out = scale_row(torch.ones(5, 2), 3.0)
+    assert torch.allclose(out, torch.full((5, 2), 3.0))
