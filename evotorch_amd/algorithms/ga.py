@@ -7,7 +7,7 @@ Cosyne :893). The NSGA-II selection path uses the batched domination-count
 front peeling of evotorch_amd.core (K7 in SURVEY.md §2.9).
 """
 
-from typing import Callable, Iterable, List, Optional, Union
+from typing import Iterable, Optional
 
 import torch
 

@@ -33,12 +33,8 @@ from .utils import (
     Serializable,
     TensorMakerMixin,
     as_read_only_tensor,
-    clone as _clone,
     deep_clone,
-    is_dtype_bool,
     is_dtype_object,
-    is_dtype_real,
-    is_sequence,
     to_torch_dtype,
 )
 from .utils.misc import ensure_tensor_length_and_dtype, split_workload

@@ -9,7 +9,7 @@ for one population runs as B independent batched searches.
 """
 
 import functools
-from typing import Callable, Iterable, Optional, Union
+from typing import Callable, Optional, Union
 
 import torch
 

@@ -13,8 +13,6 @@ lives in evotorch_amd/ops/hip/rollout.hip.
 """
 
 import copy
-from typing import Optional
-
 import torch
 from torch import nn
 

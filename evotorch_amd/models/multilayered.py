@@ -7,7 +7,7 @@ multilayered.py:21. A layer whose forward takes (x, h) and returns
 """
 
 import inspect
-from typing import Any, Optional, Tuple, Union
+from typing import Optional, Tuple, Union
 
 import torch
 from torch import nn

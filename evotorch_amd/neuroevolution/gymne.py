@@ -18,7 +18,7 @@ from torch import nn
 from ..core import Solution, SolutionBatch
 from ..models import ensure_stateful
 from .neproblem import NEProblem
-from .runningnorm import ObsNormLayer, RunningNorm
+from .runningnorm import RunningNorm
 
 __all__ = ["GymNE", "ActClipLayer"]
 

@@ -8,14 +8,14 @@ evaluation function; the network lives on the problem's aux device (the
 GPU) while the population may live elsewhere.
 """
 
-from typing import Callable, Iterable, Optional, Union
+from typing import Callable, Optional, Union
 
 import torch
 from torch import nn
 
-from ..core import Problem, Solution, SolutionBatch
-from ..models import Policy, count_parameters, fill_parameters, make_functional_module, parameter_vector, str_to_net
-from ..utils.misc import is_sequence, pass_info_if_needed
+from ..core import Problem, Solution
+from ..models import Policy, count_parameters, fill_parameters, str_to_net
+from ..utils.misc import pass_info_if_needed
 
 __all__ = ["NEProblem", "BaseNEProblem"]
 

@@ -24,7 +24,7 @@ from torch import nn
 from ..core import SolutionBatch
 from ..models import Policy
 from .neproblem import NEProblem
-from .runningnorm import ObsNormLayer, RunningNorm
+from .runningnorm import RunningNorm
 from .synthetic_env import SyntheticEnvSpec
 
 __all__ = ["SyntheticTorchEnv", "GymVectorEnvAdapter", "VecEnvNE", "VecGymNE"]

@@ -11,7 +11,7 @@ population arguments run independent batched populations via
 reference), which is how batched searches compose.
 """
 
-from typing import Optional, Tuple, Union
+from typing import Optional, Union
 
 import torch
 

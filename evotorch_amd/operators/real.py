@@ -12,7 +12,7 @@ import torch
 
 from ..core import Problem, SolutionBatch
 from ..utils import RealOrVector
-from ..utils.misc import ensure_tensor_length_and_dtype, make_gaussian, make_uniform
+from ..utils.misc import make_gaussian, make_uniform
 from .base import CopyingOperator, CrossOver
 
 __all__ = [

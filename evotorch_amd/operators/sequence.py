@@ -5,7 +5,7 @@ Reference parity: /root/reference/src/evotorch/operators/sequence.py:25.
 
 import torch
 
-from ..core import Problem, SolutionBatch
+from ..core import SolutionBatch
 from ..utils import ObjectArray
 from .base import CrossOver
 

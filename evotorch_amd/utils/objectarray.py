@@ -8,7 +8,7 @@ store-immutable-clones policy), and `__getitem__` with a slice returns a
 *view* sharing storage, mirroring SolutionBatch's shared-memory slicing.
 """
 
-from typing import Any, Iterable, Optional, Union
+from typing import Iterable, Optional, Union
 
 import numpy as np
 

@@ -6,7 +6,7 @@ and `SolutionBatch` accessors hand these out so that user code cannot
 silently corrupt population storage shared with HIP kernels.
 """
 
-from typing import Optional, Union
+from typing import Union
 
 import torch
 

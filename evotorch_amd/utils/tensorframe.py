@@ -7,7 +7,7 @@ operations are batched tensor ops, never python loops.
 """
 
 from collections.abc import Mapping
-from typing import Callable, Iterable, Optional, Union
+from typing import Callable, Optional, Union
 
 import torch
 
