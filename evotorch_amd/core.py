@@ -283,6 +283,10 @@ class SolutionBatch(Serializable):
         return object if isinstance(self._values, ObjectArray) else self._values.dtype
 
     @property
+    def values_dtype(self):
+        return self.dtype
+
+    @property
     def eval_dtype(self):
         return self._evals.dtype
 
@@ -775,6 +779,22 @@ class Problem(TensorMakerMixin, Serializable):
         return len(self._senses) > 1
 
     @property
+    def is_single_objective(self) -> bool:
+        return len(self._senses) == 1
+
+    def get_obj_order_descending(self) -> List[bool]:
+        """Per objective: True iff higher is better (reference
+        core.py:1763)."""
+        return [s == "max" for s in self._senses]
+
+    def ensure_tensor_length_and_dtype(self, t, *, about: Optional[str] = None, allow_scalar: bool = False):
+        """Coerce `t` into a solution-length vector of this problem's dtype
+        and device (reference exposes this as a Problem method)."""
+        return ensure_tensor_length_and_dtype(
+            t, self._solution_length, self._dtype, about=about, allow_scalar=allow_scalar, device=self._device
+        )
+
+    @property
     def solution_length(self) -> Optional[int]:
         return self._solution_length
 
@@ -810,6 +830,12 @@ class Problem(TensorMakerMixin, Serializable):
     @property
     def generator(self) -> Optional[torch.Generator]:
         return self._generator
+
+    @property
+    def has_own_generator(self) -> bool:
+        """True iff this Problem was given a seed (and so owns a private
+        RNG stream instead of using torch's global one)."""
+        return self._generator is not None
 
     def manual_seed(self, seed: Optional[int] = None):
         if seed is None:
