@@ -247,3 +247,40 @@ def test_restart_and_ipop():
     restarter.run(12)
     assert restarter.num_restarts >= 2
     assert restarter._algorithm_args["popsize"] >= 16
+
+
+def test_object_dtype_ga_with_cut_and_splice():
+    """Variable-length sequence evolution (object dtype) end to end —
+    mirrors the reference's CutAndSplice usage (operators/sequence.py:25)."""
+    import numpy as np
+
+    from evotorch_amd.operators import CutAndSplice
+
+    target = [1.0, 2.0, 3.0, 4.0, 5.0]
+
+    class SeqProblem(Problem):
+        def __init__(self):
+            super().__init__(objective_sense="min", dtype=object, eval_dtype=torch.float32, seed=3)
+            self._rng = np.random.default_rng(0)
+
+        def _fill(self, values):
+            for i in range(len(values)):
+                n = int(self._rng.integers(1, 8))
+                values[i] = [float(x) for x in self._rng.uniform(0, 6, n)]
+
+        def _evaluate(self, solution):
+            seq = list(solution.values)
+            # distance to target: elementwise + length penalty
+            cost = abs(len(seq) - len(target)) * 5.0
+            for a, b in zip(seq, target):
+                cost += abs(float(a) - b)
+            solution.set_evaluation(cost)
+
+    prob = SeqProblem()
+    ga = GeneticAlgorithm(prob, popsize=40, operators=[CutAndSplice(prob, tournament_size=4)])
+    ga.step()
+    first = ga.status["pop_best_eval"]
+    ga.run(25)
+    assert ga.status["pop_best_eval"] <= first
+    best = ga.population.take_best()
+    assert isinstance(list(best.values), list)
