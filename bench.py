@@ -38,6 +38,8 @@ def parse_args():
     p.add_argument("--popsize-per-gpu", type=int, default=4000)
     p.add_argument("--episode-length", type=int, default=200)
     p.add_argument("--seed", type=int, default=1)
+    p.add_argument("--policy", choices=["linear", "mlp64"], default="linear",
+                   help="linear = Humanoid-v4 linear policy (the headline metric); mlp64 = the paper's MLP-64-tanh brax config")
     return p.parse_args()
 
 
@@ -69,6 +71,7 @@ def main():
         seed=args.seed + rank,
         episode_length=args.episode_length,
         observation_normalization=True,
+        policy_hidden=64 if args.policy == "mlp64" else 0,
     )
     if comm is not None:
         problem.use_comm(comm)
@@ -125,7 +128,7 @@ def main():
             "dtype": "bf16",
             "data": "synthetic (offline low-rank neural dynamics, Humanoid-v4 obs/act geometry; no simulator available offline)",
             "config": {
-                "model": "linear policy obs376->act17 (6409 params), PGPE+ClipUp, obs-norm on",
+                "model": ("linear policy obs376->act17 (6409 params), PGPE+ClipUp, obs-norm on" if args.policy == "linear" else "MLP-64-tanh policy obs376->64->act17 (25233 params), PGPE+ClipUp, obs-norm on"),
                 "global_batch": total_popsize,
                 "seq_len": args.episode_length,
                 "parallelism": f"dp{world}",

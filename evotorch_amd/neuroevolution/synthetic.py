@@ -109,6 +109,7 @@ class SyntheticRolloutProblem(Problem):
                 spec.act_cost,
                 init_seed,
                 member_offset,
+                spec.policy_hidden,
             )
             triple = (float(n) * spec.episode_length, obs_stats[: spec.obs_dim], obs_stats[spec.obs_dim :])
         else:
@@ -152,15 +153,25 @@ class SyntheticRolloutProblem(Problem):
 
     def to_policy(self, x: torch.Tensor) -> torch.nn.Module:
         spec = self._spec
-        O, A = spec.obs_dim, spec.act_dim
+        O, A, H = spec.obs_dim, spec.act_dim, spec.policy_hidden
         x = torch.as_tensor(x, dtype=torch.float32).detach().cpu().reshape(-1)
-        linear = torch.nn.Linear(O, A)
-        with torch.no_grad():
-            linear.weight.copy_(x[: A * O].reshape(A, O))
-            linear.bias.copy_(x[A * O :])
         mean, std = self._norm_mean_std()
         layers = []
         if self._obs_norm_enabled:
             layers.append(ObsNormLayer(mean.cpu(), std.cpu()))
-        layers += [linear, torch.nn.Hardtanh()]
+        with torch.no_grad():
+            if H > 0:
+                l1 = torch.nn.Linear(O, H)
+                l2 = torch.nn.Linear(H, A)
+                off = 0
+                l1.weight.copy_(x[off : off + H * O].reshape(H, O)); off += H * O
+                l1.bias.copy_(x[off : off + H]); off += H
+                l2.weight.copy_(x[off : off + A * H].reshape(A, H)); off += A * H
+                l2.bias.copy_(x[off:])
+                layers += [l1, torch.nn.Tanh(), l2, torch.nn.Hardtanh()]
+            else:
+                linear = torch.nn.Linear(O, A)
+                linear.weight.copy_(x[: A * O].reshape(A, O))
+                linear.bias.copy_(x[A * O :])
+                layers += [linear, torch.nn.Hardtanh()]
         return torch.nn.Sequential(*layers)

@@ -43,11 +43,13 @@ class SyntheticEnvSpec:
         alive_bonus: float = 1.0,
         act_cost: float = 0.05,
         env_seed: int = 1234,
+        policy_hidden: int = 0,   # 0 = linear policy; H = one tanh hidden layer
         device="cpu",
     ):
         self.obs_dim = int(obs_dim)
         self.act_dim = int(act_dim)
         self.rank = int(rank)
+        self.policy_hidden = int(policy_hidden)
         self.episode_length = int(episode_length)
         self.alive_bonus = float(alive_bonus)
         self.act_cost = float(act_cost)
@@ -64,6 +66,9 @@ class SyntheticEnvSpec:
 
     @property
     def solution_length(self) -> int:
+        if self.policy_hidden > 0:
+            H = self.policy_hidden
+            return H * self.obs_dim + H + self.act_dim * H + self.act_dim
         return self.act_dim * self.obs_dim + self.act_dim
 
     def env_blob(self, mean: torch.Tensor, std: torch.Tensor, device=None) -> torch.Tensor:
@@ -117,8 +122,16 @@ def rollout_eager(
         return x.to(torch.bfloat16).to(torch.float32) if bf16_operands else x
 
     # bf16 OPERANDS / fp32 accumulation, exactly as the kernel stores them
-    W = q(params[:, : A * O].reshape(N, A, O))
-    b = params[:, A * O :].to(device, dt)
+    H = spec.policy_hidden
+    if H > 0:
+        off = 0
+        W1 = q(params[:, off : off + H * O].reshape(N, H, O)); off += H * O
+        b1 = params[:, off : off + H].to(device, dt); off += H
+        W2 = q(params[:, off : off + A * H].reshape(N, A, H)); off += A * H
+        b = params[:, off :].to(device, dt)
+    else:
+        W = q(params[:, : A * O].reshape(N, A, O))
+        b = params[:, A * O :].to(device, dt)
     V = q(spec.V)
     U_T = q(spec.U_T)
     D2_T = q(spec.D2_T)
@@ -133,7 +146,11 @@ def rollout_eager(
     obs_sumsq = torch.zeros(O, dtype=dt, device=device)
     for _ in range(steps):
         obs_n = q((obs - mean_f) * inv_std)
-        act = torch.clamp(torch.einsum("nao,no->na", W, obs_n) + b, -1.0, 1.0)
+        if H > 0:
+            hid = q(torch.tanh(torch.einsum("nho,no->nh", W1, obs_n) + b1))
+            act = torch.clamp(torch.einsum("nah,nh->na", W2, hid) + b, -1.0, 1.0)
+        else:
+            act = torch.clamp(torch.einsum("nao,no->na", W, obs_n) + b, -1.0, 1.0)
         act_cost_term = spec.act_cost * (act**2).sum(-1) / A
         act_b = q(act)
         h = q(obs @ V.T)  # (N, R)

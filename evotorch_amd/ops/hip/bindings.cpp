@@ -14,7 +14,7 @@ void adam_step(torch::Tensor step_out, torch::Tensor grad, torch::Tensor m, torc
                double stepsize, double beta1, double beta2, double epsilon);
 torch::Tensor rollout_linear(torch::Tensor params, torch::Tensor env_blob, torch::Tensor obs_stats_out,
                              int64_t obs_dim, int64_t act_dim, int64_t rank, int64_t steps, double alive_bonus,
-                             double act_cost, int64_t init_seed, int64_t member_offset);
+                             double act_cost, int64_t init_seed, int64_t member_offset, int64_t policy_hidden);
 torch::Tensor domination_counts(torch::Tensor utils);
 torch::Tensor pareto_ranks(torch::Tensor utils);
 }  // namespace ea
@@ -28,7 +28,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("snes_gradients", &ea::snes_gradients, "K3: SNES raw-noise gradient reduction");
     m.def("clipup_step", &ea::clipup_step, "K4: fused ClipUp velocity update (no host sync)");
     m.def("adam_step", &ea::adam_step, "K4: fused Adam ascent step");
-    m.def("rollout_linear", &ea::rollout_linear, "K10+K11: fused linear-policy episode rollout (synthetic env)");
+    m.def("rollout_linear", &ea::rollout_linear,
+          "K10+K11: fused policy episode rollout (linear or MLP-H, synthetic env)",
+          pybind11::arg("params"), pybind11::arg("env_blob"), pybind11::arg("obs_stats_out"),
+          pybind11::arg("obs_dim"), pybind11::arg("act_dim"), pybind11::arg("rank"), pybind11::arg("steps"),
+          pybind11::arg("alive_bonus"), pybind11::arg("act_cost"), pybind11::arg("init_seed"),
+          pybind11::arg("member_offset"), pybind11::arg("policy_hidden") = 0);
     m.def("domination_counts", &ea::domination_counts, "K7: NSGA-II domination counts (no N x N matrix)");
     m.def("pareto_ranks", &ea::pareto_ranks, "K7: NSGA-II non-dominated sorting by front peeling");
 }

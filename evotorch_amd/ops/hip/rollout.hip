@@ -1,21 +1,22 @@
-// K10+K11 fused: whole-generation rollout of a population of linear
-// policies through the synthetic vectorized environment, one kernel launch
-// per generation (SURVEY.md §3.4 — the VecGymNE hot loop, collapsed).
+// K10+K11 fused: whole-generation rollout of a population of policies
+// through the synthetic vectorized environment, one kernel launch per
+// generation (SURVEY.md §3.4 — the VecGymNE hot loop, collapsed).
 //
-// MI355X design (v5): one workgroup rolls out kMembers (default 2)
+// Policies: linear (obs -> act) or one-hidden-layer tanh MLP
+// (obs -> H -> act; H=64 is the reference paper's brax-humanoid config).
+//
+// MI355X design (v6): one workgroup rolls out kMembers (default 2)
 // population members for the entire T-step episode with policy weights
 // AND the shared environment matrices resident in LDS — the inner loop
 // never touches global memory. Multi-member blocks exist because one
-// member's per-step work underfills a 512-thread block (phase 1 has only
-// 33 reduction outputs for 64 lane-groups; phase 2 has 376 dims for 512
-// lanes): with 2 members both phases nearly double their parallel work
-// per barrier, and gfx950 allows the required ~75 KB of dynamic LDS per
-// workgroup (sharedMemPerBlock = 160 KB, probed).
+// member's per-step work underfills a 512-thread block; gfx950 allows the
+// required LDS (sharedMemPerBlock = 160 KB, probed; ~75 KB for 2 linear
+// members, ~146 KB for 2 MLP-64 members).
 //
 // All inner products run on `v_dot2_f32_bf16` (2 bf16 MACs/instruction,
 // fp32 accumulate). LDS access patterns:
-//   * group-reduced dots (policy W, dynamics factor V): row-major
-//     [out][O]; 8-lane groups read consecutive bf16x2 — conflict-free,
+//   * group-reduced dots (policy layers, dynamics factor V): row-major
+//     [out][len]; 8-lane groups read consecutive bf16x2 — conflict-free,
 //     with a 3-level shuffle reduction (64-lane trees serialized 6
 //     dependent shuffles; measured 10:1 SQ_WAIT:SQ_BUSY).
 //   * per-thread dots (U, D2): PAIR-INTERLEAVED COLUMN-MAJOR [K/2][O]
@@ -28,7 +29,8 @@
 // evotorch_amd/neuroevolution/synthetic_env.py::rollout_eager, which
 // quantizes the same operands to bf16):
 //   obs_n = bf16((obs − mean) · inv_std)
-//   a     = clip(W·obs_n + b, −1, 1)        (W bf16, accum fp32)
+//   hid   = bf16(tanh(W1·obs_n + b1))            (MLP mode only)
+//   a     = clip(W2·hid + b2, −1, 1)   |  clip(W·obs_n + b, −1, 1)
 //   h     = bf16(V·obs)
 //   o'    = tanh(Σ_i U_T[i]·h[i] + Σ_m D2_T[m]·a[m] + c)
 //   r     = wr·o' + alive_bonus − act_cost·‖a‖²/A
@@ -61,15 +63,27 @@ __device__ __forceinline__ T* sel2(int m, T* p0, T* p1) {
 }
 
 struct RolloutArgs {
-    const float* params;      // [n_members][A*O + A]  (W row-major, then b)
+    const float* params;      // [n_members][L]
     const float* env_blob;    // packed fp32 env data (see env_blob())
     float* fitness_out;       // [n_members]
     float* obs_stats_out;     // [2][O]  (sum, sumsq) — atomically accumulated
     int n_members;
     long member_offset;       // global id of member 0 (rank sharding)
-    int obs_dim, act_dim, rank, steps;
+    int obs_dim, act_dim, rank, steps, hidden;
     float alive_bonus, act_cost;
     unsigned long long init_seed;
+};
+
+// one group-reduced dot output: 8 lanes compute row·vec, lane 0 of the
+// group applies the epilogue selected by `kind`
+struct OutDesc {
+    const __bf16* row;
+    const __bf16* vec;
+    __bf16* dst;
+    const float* bias;   // null for kind 0
+    int kind;            // 0 = plain store, 1 = action (bias+clip), 2 = hidden (bias+tanh)
+    int member;
+    bool valid;
 };
 
 // env_blob layout (fp32): V [R][O] · U_T [R][O] · D2_T [A][O] · c [O] ·
@@ -77,13 +91,14 @@ struct RolloutArgs {
 
 template <int kGroup, int kMembers>
 __global__ __launch_bounds__(512, 2) void rollout_linear_kernel(RolloutArgs args) {
-    const int O = args.obs_dim, A = args.act_dim, R = args.rank;
+    const int O = args.obs_dim, A = args.act_dim, R = args.rank, H = args.hidden;
     const int tid = threadIdx.x;
     const int base_member = blockIdx.x * kMembers;
     if (base_member >= args.n_members) return;
     const int live_members = min(kMembers, args.n_members - base_member);
     const int A_PAD = (A + 2) & ~1;
     const int R_PAD = (R + 2) & ~1;
+    const long param_len = (H > 0) ? ((long)H * O + H + (long)A * H + A) : ((long)A * O + A);
 
     extern __shared__ unsigned char lds_raw[];
     // ---- shared environment ----
@@ -96,8 +111,11 @@ __global__ __launch_bounds__(512, 2) void rollout_linear_kernel(RolloutArgs args
     float* istd_l = mean_l + O;                                // [O]
     float* scratch = istd_l + O;                               // [16]
     // ---- per-member state ----
-    __bf16* W_l[kMembers];
-    float* b_l[kMembers];
+    __bf16* W1_l[kMembers];   // linear: [A][O]; MLP: [H][O]
+    float* b1_l[kMembers];    // linear: [A];    MLP: [H]
+    __bf16* W2_l[kMembers];   // MLP only: [A][H]
+    float* b2_l[kMembers];    // MLP only: [A]
+    __bf16* hid_b[kMembers];  // MLP only: [H]
     __bf16* obs_b[kMembers];
     __bf16* obsn_b[kMembers];
     __bf16* h_b[kMembers];
@@ -106,10 +124,23 @@ __global__ __launch_bounds__(512, 2) void rollout_linear_kernel(RolloutArgs args
         unsigned char* cursor = reinterpret_cast<unsigned char*>(scratch + 16);
 #pragma unroll
         for (int m = 0; m < kMembers; ++m) {
-            W_l[m] = reinterpret_cast<__bf16*>(cursor);
-            cursor += (size_t)A * O * 2;
-            b_l[m] = reinterpret_cast<float*>(cursor);
-            cursor += (size_t)A * 4;
+            const int w1_rows = (H > 0) ? H : A;
+            W1_l[m] = reinterpret_cast<__bf16*>(cursor);
+            cursor += (size_t)w1_rows * O * 2;
+            b1_l[m] = reinterpret_cast<float*>(cursor);
+            cursor += (size_t)w1_rows * 4;
+            if (H > 0) {
+                W2_l[m] = reinterpret_cast<__bf16*>(cursor);
+                cursor += (size_t)A * H * 2;
+                b2_l[m] = reinterpret_cast<float*>(cursor);
+                cursor += (size_t)A * 4;
+                hid_b[m] = reinterpret_cast<__bf16*>(cursor);
+                cursor += (size_t)H * 2;
+            } else {
+                W2_l[m] = nullptr;
+                b2_l[m] = nullptr;
+                hid_b[m] = nullptr;
+            }
             obs_b[m] = reinterpret_cast<__bf16*>(cursor);
             cursor += (size_t)O * 2;
             obsn_b[m] = reinterpret_cast<__bf16*>(cursor);
@@ -152,9 +183,16 @@ __global__ __launch_bounds__(512, 2) void rollout_linear_kernel(RolloutArgs args
 #pragma unroll
     for (int m = 0; m < kMembers; ++m) {
         if (m >= live_members) break;
-        const float* my_params = args.params + (long)(base_member + m) * (A * O + A);
-        for (int i = tid; i < A * O; i += blockDim.x) W_l[m][i] = f2b(my_params[i]);
-        for (int i = tid; i < A; i += blockDim.x) b_l[m][i] = my_params[A * O + i];
+        const float* my_params = args.params + (long)(base_member + m) * param_len;
+        if (H > 0) {
+            for (int i = tid; i < H * O; i += blockDim.x) W1_l[m][i] = f2b(my_params[i]);
+            for (int i = tid; i < H; i += blockDim.x) b1_l[m][i] = my_params[H * O + i];
+            for (int i = tid; i < A * H; i += blockDim.x) W2_l[m][i] = f2b(my_params[H * O + H + i]);
+            for (int i = tid; i < A; i += blockDim.x) b2_l[m][i] = my_params[H * O + H + A * H + i];
+        } else {
+            for (int i = tid; i < A * O; i += blockDim.x) W1_l[m][i] = f2b(my_params[i]);
+            for (int i = tid; i < A; i += blockDim.x) b1_l[m][i] = my_params[A * O + i];
+        }
         for (int i = tid; i < R_PAD; i += blockDim.x) h_b[m][i] = f2b(0.0f);
         for (int i = tid; i < A_PAD; i += blockDim.x) act_b[m][i] = f2b(0.0f);
         const unsigned long long gmember = (unsigned long long)(args.member_offset + base_member + m);
@@ -172,44 +210,75 @@ __global__ __launch_bounds__(512, 2) void rollout_linear_kernel(RolloutArgs args
 
     const int lane = tid & (kWaveSize - 1);
     const int wave = tid / kWaveSize;
-    const int n_pairs = O / 2;              // O must be even
-    const int outs_per_member = A + R;
-    const int n_outputs = live_members * outs_per_member;
-    constexpr int kRounds = 2;              // up to 2 * threads/kGroup outputs
     const int group = lane / kGroup;
     const int glane = lane % kGroup;
     const int groups_per_block = (int)(blockDim.x / kGroup);
     const int groups_per_wave = kWaveSize / kGroup;
+    const int my_group = wave * groups_per_wave + group;
 
-    // per-round output assignment, hoisted: everything below is FULLY
-    // UNROLLED over kRounds (runtime-indexed arrays would demote to
-    // scratch — cdna_hip_programming.md common-mistake #20)
-    const __bf16* my_row[kRounds];
-    const __bf16* my_vec[kRounds];
-    float* my_bias[kRounds];
-    __bf16* my_dst[kRounds];
-    bool my_is_act[kRounds];
-    bool my_valid[kRounds];
-    int my_member[kRounds];
+    // ---- phase-A table: dots over obs-side vectors -------------------------
+    // Per member: H hidden outputs (MLP) or A action outputs (linear),
+    // followed by R dynamics-factor outputs. Up to 3 rounds (e.g. 2 MLP-64
+    // members: 2*(64+16) = 160 outputs over 64 groups).
+    constexpr int kRoundsA = 3;
+    const int a_per_member = (H > 0 ? H : A) + R;
+    const int n_outs_a = live_members * a_per_member;
+    OutDesc descA[kRoundsA];
 #pragma unroll
-    for (int r = 0; r < kRounds; ++r) {
-        const int out = (wave * groups_per_wave + group) + r * groups_per_block;
-        my_valid[r] = out < n_outputs;
-        const int safe_out = my_valid[r] ? out : 0;
-        const int m = safe_out / outs_per_member;
-        const int lo = safe_out % outs_per_member;
-        my_member[r] = m;
-        my_is_act[r] = lo < A;
-        __bf16* w_m = sel2(m, W_l[0], W_l[kMembers - 1]);
-        float* b_m = sel2(m, b_l[0], b_l[kMembers - 1]);
+    for (int r = 0; r < kRoundsA; ++r) {
+        const int out = my_group + r * groups_per_block;
+        OutDesc d;
+        d.valid = out < n_outs_a;
+        const int safe_out = d.valid ? out : 0;
+        const int m = safe_out / a_per_member;
+        const int lo = safe_out % a_per_member;
+        d.member = m;
+        const int first_count = (H > 0 ? H : A);
+        __bf16* w1_m = sel2(m, W1_l[0], W1_l[kMembers - 1]);
+        float* b1_m = sel2(m, b1_l[0], b1_l[kMembers - 1]);
         __bf16* obs_m = sel2(m, obs_b[0], obs_b[kMembers - 1]);
         __bf16* obsn_m = sel2(m, obsn_b[0], obsn_b[kMembers - 1]);
-        __bf16* h_m = sel2(m, h_b[0], h_b[kMembers - 1]);
-        __bf16* act_m = sel2(m, act_b[0], act_b[kMembers - 1]);
-        my_row[r] = my_is_act[r] ? (w_m + lo * O) : (V_l + (lo - A) * O);
-        my_vec[r] = my_is_act[r] ? obsn_m : obs_m;
-        my_bias[r] = my_is_act[r] ? (b_m + lo) : nullptr;
-        my_dst[r] = my_is_act[r] ? (act_m + lo) : (h_m + (lo - A));
+        __bf16* henv_m = sel2(m, h_b[0], h_b[kMembers - 1]);
+        if (lo < first_count) {
+            d.row = w1_m + lo * O;
+            d.vec = obsn_m;
+            d.bias = b1_m + lo;
+            if (H > 0) {
+                __bf16* hid_m = sel2(m, hid_b[0], hid_b[kMembers - 1]);
+                d.dst = hid_m + lo;
+                d.kind = 2;
+            } else {
+                __bf16* act_m = sel2(m, act_b[0], act_b[kMembers - 1]);
+                d.dst = act_m + lo;
+                d.kind = 1;
+            }
+        } else {
+            d.row = V_l + (lo - first_count) * O;
+            d.vec = obs_m;
+            d.bias = nullptr;
+            d.dst = henv_m + (lo - first_count);
+            d.kind = 0;
+        }
+        descA[r] = d;
+    }
+    // ---- phase-B table (MLP only): action dots over the hidden vector ------
+    constexpr int kRoundsB = 1;
+    OutDesc descB[kRoundsB];
+#pragma unroll
+    for (int r = 0; r < kRoundsB; ++r) {
+        const int out = my_group + r * groups_per_block;
+        OutDesc d;
+        d.valid = (H > 0) && (out < live_members * A);
+        const int safe_out = d.valid ? out : 0;
+        const int m = safe_out / A;
+        const int lo = safe_out % A;
+        d.member = m;
+        d.row = (H > 0) ? sel2(m, W2_l[0], W2_l[kMembers - 1]) + lo * H : nullptr;
+        d.vec = (H > 0) ? sel2(m, hid_b[0], hid_b[kMembers - 1]) : nullptr;
+        d.bias = (H > 0) ? sel2(m, b2_l[0], b2_l[kMembers - 1]) + lo : nullptr;
+        d.dst = sel2(m, act_b[0], act_b[kMembers - 1]) + lo;
+        d.kind = 1;
+        descB[r] = d;
     }
 
     float fit_part[kMembers];
@@ -230,70 +299,84 @@ __global__ __launch_bounds__(512, 2) void rollout_linear_kernel(RolloutArgs args
     }
     __syncthreads();
 
+    // group-dot worker: acc[r] = descs[r].row · descs[r].vec  (length
+    // 2*n_pairs), reduced across the kGroup lanes of the group
+#define GROUP_DOTS(descs, NR, n_pairs_expr, acc)                                              \
+    {                                                                                         \
+        const int np = (n_pairs_expr);                                                        \
+        _Pragma("unroll") for (int r = 0; r < (NR); ++r) acc[r] = 0.0f;                       \
+        _Pragma("unroll") for (int r = 0; r < (NR); ++r) {                                    \
+            if (!descs[r].valid) continue;                                                    \
+            const __bf16* row = descs[r].row;                                                 \
+            const __bf16* vec = descs[r].vec;                                                 \
+            float a0 = 0.0f, a1 = 0.0f;                                                       \
+            const int full_iters = np / kGroup;                                               \
+            const int tail = np % kGroup;                                                     \
+            for (int i = 0; i + 1 < full_iters; i += 2) {                                     \
+                const int p0 = glane + i * kGroup;                                            \
+                const int p1 = glane + (i + 1) * kGroup;                                      \
+                a0 = __builtin_amdgcn_fdot2_f32_bf16(                                         \
+                    *reinterpret_cast<const bf16x2*>(row + 2 * p0),                           \
+                    *reinterpret_cast<const bf16x2*>(vec + 2 * p0), a0, false);               \
+                a1 = __builtin_amdgcn_fdot2_f32_bf16(                                         \
+                    *reinterpret_cast<const bf16x2*>(row + 2 * p1),                           \
+                    *reinterpret_cast<const bf16x2*>(vec + 2 * p1), a1, false);               \
+            }                                                                                 \
+            if (full_iters & 1) {                                                             \
+                const int p0 = glane + (full_iters - 1) * kGroup;                             \
+                a0 = __builtin_amdgcn_fdot2_f32_bf16(                                         \
+                    *reinterpret_cast<const bf16x2*>(row + 2 * p0),                           \
+                    *reinterpret_cast<const bf16x2*>(vec + 2 * p0), a0, false);               \
+            }                                                                                 \
+            if (tail && glane < tail) {                                                       \
+                const int p0 = glane + full_iters * kGroup;                                   \
+                a1 = __builtin_amdgcn_fdot2_f32_bf16(                                         \
+                    *reinterpret_cast<const bf16x2*>(row + 2 * p0),                           \
+                    *reinterpret_cast<const bf16x2*>(vec + 2 * p0), a1, false);               \
+            }                                                                                 \
+            acc[r] = a0 + a1;                                                                 \
+        }                                                                                     \
+        _Pragma("unroll") for (int offset = kGroup / 2; offset > 0; offset >>= 1) {           \
+            _Pragma("unroll") for (int r = 0; r < (NR); ++r) {                                \
+                acc[r] += __shfl_down(acc[r], offset, kWaveSize);                             \
+            }                                                                                 \
+        }                                                                                     \
+    }
+
+#define DOT_EPILOGUE(descs, NR, acc)                                                          \
+    if (glane == 0) {                                                                         \
+        _Pragma("unroll") for (int r = 0; r < (NR); ++r) {                                    \
+            if (!descs[r].valid) continue;                                                    \
+            if (descs[r].kind == 1) {                                                         \
+                const float a = fminf(fmaxf(acc[r] + *descs[r].bias, -1.0f), 1.0f);           \
+                *descs[r].dst = f2b(a);                                                       \
+                _Pragma("unroll") for (int m = 0; m < kMembers; ++m) {                        \
+                    if (descs[r].member == m) actsq_part[m] = fmaf(a, a, actsq_part[m]);      \
+                }                                                                             \
+            } else if (descs[r].kind == 2) {                                                  \
+                *descs[r].dst = f2b(tanhf(acc[r] + *descs[r].bias));                          \
+            } else {                                                                          \
+                *descs[r].dst = f2b(acc[r]);                                                  \
+            }                                                                                 \
+        }                                                                                     \
+    }
+
     for (int t = 0; t < args.steps; ++t) {
-        // ---- phase 1: group-reduced dots (policy actions + dynamics h) ----
-        float acc[kRounds];
-#pragma unroll
-        for (int r = 0; r < kRounds; ++r) acc[r] = 0.0f;
-#pragma unroll
-        for (int r = 0; r < kRounds; ++r) {
-            if (!my_valid[r]) continue;
-            const __bf16* row = my_row[r];
-            const __bf16* vec = my_vec[r];
-            float a0 = 0.0f, a1 = 0.0f;  // two chains: halve the serial depth
-            const int full_iters = n_pairs / kGroup;
-            const int tail = n_pairs % kGroup;
-            for (int i = 0; i + 1 < full_iters; i += 2) {
-                const int p0 = glane + i * kGroup;
-                const int p1 = glane + (i + 1) * kGroup;
-                a0 = __builtin_amdgcn_fdot2_f32_bf16(
-                    *reinterpret_cast<const bf16x2*>(row + 2 * p0),
-                    *reinterpret_cast<const bf16x2*>(vec + 2 * p0), a0, false);
-                a1 = __builtin_amdgcn_fdot2_f32_bf16(
-                    *reinterpret_cast<const bf16x2*>(row + 2 * p1),
-                    *reinterpret_cast<const bf16x2*>(vec + 2 * p1), a1, false);
-            }
-            if (full_iters & 1) {
-                const int p0 = glane + (full_iters - 1) * kGroup;
-                a0 = __builtin_amdgcn_fdot2_f32_bf16(
-                    *reinterpret_cast<const bf16x2*>(row + 2 * p0),
-                    *reinterpret_cast<const bf16x2*>(vec + 2 * p0), a0, false);
-            }
-            if (tail && glane < tail) {
-                const int p0 = glane + full_iters * kGroup;
-                a1 = __builtin_amdgcn_fdot2_f32_bf16(
-                    *reinterpret_cast<const bf16x2*>(row + 2 * p0),
-                    *reinterpret_cast<const bf16x2*>(vec + 2 * p0), a1, false);
-            }
-            acc[r] = a0 + a1;
-        }
-        // in-group reductions (rounds' shuffles overlap)
-#pragma unroll
-        for (int offset = kGroup / 2; offset > 0; offset >>= 1) {
-#pragma unroll
-            for (int r = 0; r < kRounds; ++r) {
-                acc[r] += __shfl_down(acc[r], offset, kWaveSize);
-            }
-        }
-        if (glane == 0) {
-#pragma unroll
-            for (int r = 0; r < kRounds; ++r) {
-                if (!my_valid[r]) continue;
-                if (my_is_act[r]) {
-                    const float a = fminf(fmaxf(acc[r] + *my_bias[r], -1.0f), 1.0f);
-                    *my_dst[r] = f2b(a);
-#pragma unroll
-                    for (int m = 0; m < kMembers; ++m) {
-                        if (my_member[r] == m) actsq_part[m] = fmaf(a, a, actsq_part[m]);
-                    }
-                } else {
-                    *my_dst[r] = f2b(acc[r]);
-                }
-            }
-        }
+        // ---- phase A: hidden (or action) + dynamics-factor dots ----
+        float accA[kRoundsA];
+        GROUP_DOTS(descA, kRoundsA, O / 2, accA);
+        DOT_EPILOGUE(descA, kRoundsA, accA);
         __syncthreads();
 
-        // ---- phase 2: per-thread dynamics dims, norm fused in epilogue ----
+        // ---- phase B (MLP only): action dots over the hidden vector ----
+        if (H > 0) {
+            float accB[kRoundsB];
+            GROUP_DOTS(descB, kRoundsB, H / 2, accB);
+            DOT_EPILOGUE(descB, kRoundsB, accB);
+            __syncthreads();
+        }
+
+        // ---- dynamics phase: per-thread dims, norm fused in epilogue ----
         for (int j = tid; j < live_members * O; j += blockDim.x) {
             const int m = j / O, jo = j % O;
             float uacc = c_l[jo], dacc = 0.0f;
@@ -325,6 +408,9 @@ __global__ __launch_bounds__(512, 2) void rollout_linear_kernel(RolloutArgs args
         }
         __syncthreads();
     }
+
+#undef GROUP_DOTS
+#undef DOT_EPILOGUE
 
     // ---- wrap-up: per-member fitness reductions + obs-stat atomics ----
 #pragma unroll
@@ -360,14 +446,16 @@ static void launch_rollout(int n_blocks, int block, size_t lds_bytes, hipStream_
 
 torch::Tensor rollout_linear(torch::Tensor params, torch::Tensor env_blob, torch::Tensor obs_stats_out,
                              int64_t obs_dim, int64_t act_dim, int64_t rank, int64_t steps, double alive_bonus,
-                             double act_cost, int64_t init_seed, int64_t member_offset) {
+                             double act_cost, int64_t init_seed, int64_t member_offset, int64_t policy_hidden) {
     CHECK_GPU(params);
     TORCH_CHECK(params.is_contiguous() && params.dim() == 2, "params must be contiguous [N][L]");
     TORCH_CHECK(params.scalar_type() == at::ScalarType::Float, "params must be fp32");
     const int n = (int)params.size(0);
-    const int O = (int)obs_dim, A = (int)act_dim, R = (int)rank;
-    TORCH_CHECK(params.size(1) == (int64_t)A * O + A, "param length mismatch");
+    const int O = (int)obs_dim, A = (int)act_dim, R = (int)rank, H = (int)policy_hidden;
+    const int64_t expected_len = (H > 0) ? ((int64_t)H * O + H + (int64_t)A * H + A) : ((int64_t)A * O + A);
+    TORCH_CHECK(params.size(1) == expected_len, "param length mismatch: got ", params.size(1), " expected ", expected_len);
     TORCH_CHECK(O % 2 == 0, "obs_dim must be even (bf16x2 packing)");
+    TORCH_CHECK(H % 2 == 0, "policy_hidden must be even (bf16x2 packing)");
     auto fitness = torch::empty({n}, params.options());
 
     RolloutArgs args;
@@ -377,23 +465,30 @@ torch::Tensor rollout_linear(torch::Tensor params, torch::Tensor env_blob, torch
     args.obs_stats_out = obs_stats_out.data_ptr<float>();
     args.n_members = n;
     args.member_offset = (long)member_offset;
-    args.obs_dim = O; args.act_dim = A; args.rank = R;
+    args.obs_dim = O; args.act_dim = A; args.rank = R; args.hidden = H;
     args.steps = (int)steps;
     args.alive_bonus = (float)alive_bonus;
     args.act_cost = (float)act_cost;
     args.init_seed = (unsigned long long)init_seed;
 
+    const int block = 512;
+    const int A_PAD = (A + 2) & ~1, R_PAD = (R + 2) & ~1;
+    const size_t shared_bytes = ((size_t)R * O + (size_t)R_PAD * O + (size_t)A_PAD * O) * 2 + 4 * (size_t)O * 4 + 16 * 4 + 64;
+    const int w1_rows = (H > 0) ? H : A;
+    size_t member_bytes = ((size_t)w1_rows * O + 2 * (size_t)O + R_PAD + A_PAD) * 2 + (size_t)w1_rows * 4;
+    if (H > 0) member_bytes += (size_t)A * H * 2 + (size_t)A * 4 + (size_t)H * 2;
+
     int members = 2;
     if (const char* env = getenv("EVOTORCH_AMD_ROLLOUT_MEMBERS")) members = atoi(env);
     TORCH_CHECK(members == 1 || members == 2, "EVOTORCH_AMD_ROLLOUT_MEMBERS must be 1 or 2");
-    const int block = 512;
+    if (shared_bytes + 2 * member_bytes > 160 * 1024) members = 1;
+    const size_t lds_bytes = shared_bytes + (size_t)members * member_bytes;
+    TORCH_CHECK(lds_bytes <= 160 * 1024, "rollout LDS footprint too large: ", lds_bytes, " bytes");
+    // phase-A round capacity: up to 3 rounds of (threads/8) outputs
+    TORCH_CHECK(members * ((H > 0 ? H : A) + R) <= 3 * block / 8, "policy too wide for the 3-round phase-A table");
+    TORCH_CHECK(members * A <= block / 8, "act_dim too large for the 1-round phase-B table");
     TORCH_CHECK(O * members <= 2 * block, "obs_dim too large for the 2-slot stat accumulators");
 
-    const int A_PAD = (A + 2) & ~1, R_PAD = (R + 2) & ~1;
-    const size_t shared_bf = (size_t)R * O + (size_t)R_PAD * O + (size_t)A_PAD * O;
-    const size_t member_bytes = ((size_t)A * O + 2 * (size_t)O + R_PAD + A_PAD) * 2 + (size_t)A * 4;
-    const size_t lds_bytes = shared_bf * 2 + 4 * (size_t)O * 4 + 16 * 4 + 64 + (size_t)members * member_bytes;
-    TORCH_CHECK(lds_bytes <= 160 * 1024, "rollout LDS footprint too large: ", lds_bytes, " bytes");
     auto stream = at::cuda::getCurrentCUDAStream();
     const int n_blocks = (n + members - 1) / members;
     if (members == 2) {

@@ -292,3 +292,36 @@ class TestParetoKernels:
         ga.run(20)
         ranks, _ = ga.population.compute_pareto_ranks()
         assert float((ranks == 0).float().mean()) > 0.1
+
+
+@requires_gpu
+class TestMlpRollout:
+    def test_mlp64_matches_eager(self, C):
+        from evotorch_amd.neuroevolution.synthetic_env import SyntheticEnvSpec, rollout_eager
+
+        spec = SyntheticEnvSpec(episode_length=10, policy_hidden=64, device="cuda")
+        torch.manual_seed(9)
+        n = 16
+        params = 0.1 * torch.randn(n, spec.solution_length, device="cuda")
+        mean = torch.zeros(spec.obs_dim, device="cuda")
+        std = torch.ones(spec.obs_dim, device="cuda")
+        blob = spec.env_blob(mean, std, device="cuda")
+        obs_stats = torch.zeros(2 * spec.obs_dim, device="cuda")
+        fit = C.rollout_linear(params, blob, obs_stats, spec.obs_dim, spec.act_dim, spec.rank,
+                               spec.episode_length, spec.alive_bonus, spec.act_cost, 55, 0,
+                               spec.policy_hidden)
+        efit, _ = rollout_eager(spec, params, mean, std, init_seed=55)
+        assert torch.allclose(fit, efit, rtol=2e-2, atol=2e-2), (fit[:4], efit[:4])
+
+    def test_mlp_problem_end_to_end(self):
+        from evotorch_amd.algorithms import PGPE
+        from evotorch_amd.neuroevolution import SyntheticRolloutProblem
+
+        problem = SyntheticRolloutProblem(device="cuda:0", seed=12, episode_length=50, policy_hidden=64)
+        searcher = PGPE(problem, popsize=128, radius_init=2.25, center_learning_rate=0.1,
+                        stdev_learning_rate=0.1, optimizer="clipup", distributed=True)
+        searcher.step()
+        first = searcher.status["mean_eval"]
+        for _ in range(20):
+            searcher.step()
+        assert searcher.status["mean_eval"] > first - 5.0

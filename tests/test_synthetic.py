@@ -107,3 +107,15 @@ def test_pgpe_on_synthetic_cpu_improves():
     for _ in range(15):
         searcher.step()
     assert searcher.status["mean_eval"] > first - 1.0  # no collapse; usually improves
+
+
+def test_mlp_policy_eager_and_to_policy():
+    prob = SyntheticRolloutProblem(seed=4, episode_length=5, policy_hidden=64)
+    assert prob.solution_length == 64 * 376 + 64 + 17 * 64 + 17
+    batch = prob.generate_batch(4)
+    prob.evaluate(batch)
+    assert batch.evals_are_ready
+    policy = prob.to_policy(torch.Tensor.as_subclass(batch[0].values, torch.Tensor))
+    act = policy(torch.randn(prob.spec.obs_dim))
+    assert act.shape == (prob.spec.act_dim,)
+    assert float(act.abs().max()) <= 1.0
