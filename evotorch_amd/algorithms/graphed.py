@@ -37,7 +37,7 @@ __all__ = ["GraphedSearch"]
 
 
 class GraphedSearch:
-    def __init__(self, searcher: GaussianSearchAlgorithm, *, warmup: int = 3):
+    def __init__(self, searcher: GaussianSearchAlgorithm, *, warmup: int = 3, generations_per_capture: int = 1):
         if not isinstance(searcher, GaussianSearchAlgorithm):
             raise TypeError("GraphedSearch supports GaussianSearchAlgorithm searchers")
         if searcher._distributed:
@@ -61,6 +61,7 @@ class GraphedSearch:
         self._opt = opt
         self._graph: Optional[torch.cuda.CUDAGraph] = None
         self._warmup = int(warmup)
+        self._gens_per_capture = max(1, int(generations_per_capture))
 
         from .. import ops
         from ..core import SolutionBatch
@@ -136,8 +137,9 @@ class GraphedSearch:
         torch.cuda.current_stream().wait_stream(side)
         self._graph = torch.cuda.CUDAGraph()
         with torch.cuda.graph(self._graph):
-            self._step_body()
-        self._searcher._steps_count += 1
+            for _ in range(self._gens_per_capture):
+                self._step_body()
+        self._searcher._steps_count += self._gens_per_capture
         return self
 
     def run(self, num_generations: int):
@@ -146,11 +148,12 @@ class GraphedSearch:
         if self._graph is None:
             self.capture()
         g = self._graph
-        for _ in range(int(num_generations)):
+        replays = (int(num_generations) + self._gens_per_capture - 1) // self._gens_per_capture
+        for _ in range(replays):
             g.replay()
         torch.cuda.synchronize()
         searcher = self._searcher
-        searcher._steps_count += int(num_generations)
+        searcher._steps_count += replays * self._gens_per_capture
         searcher.clear_status()
         searcher.update_status({"iter": searcher._steps_count, "mean_eval": float(self._mean_eval_buf)})
         return self

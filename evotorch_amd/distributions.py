@@ -201,6 +201,16 @@ class SeparableGaussian(Distribution):
 
         ops.sample_gaussian(out, self.mu, self.sigma, symmetric=self._symmetric, generator=generator)
 
+    @classmethod
+    def functional_sample(cls, num_solutions: int, parameters: dict) -> torch.Tensor:
+        """Pure out-of-place sampling (vmap-safe under
+        randomness='different'; reference distributions.py:431)."""
+        mu = parameters["mu"]
+        sigma = parameters["sigma"]
+        n = int(num_solutions)
+        z = torch.randn((n,) + mu.shape[-1:], dtype=mu.dtype, device=mu.device)
+        return mu.unsqueeze(-2) + sigma.unsqueeze(-2) * z
+
     def _centered_weights(self, weights: torch.Tensor, ranking_used: Optional[str]) -> torch.Tensor:
         if ranking_used not in ("centered", "normalized"):
             weights = weights - weights.mean()
@@ -275,6 +285,18 @@ class SymmetricSeparableGaussian(SeparableGaussian):
         if out is not None and out.shape[0] % 2 != 0:
             raise ValueError(f"Symmetric sampling needs an even population, got {out.shape[0]}")
         return super().sample(num_solutions, out=out, generator=generator)
+
+    @classmethod
+    def functional_sample(cls, num_solutions: int, parameters: dict) -> torch.Tensor:
+        n = int(num_solutions)
+        if n % 2 != 0:
+            raise ValueError(f"Symmetric sampling needs an even population, got {n}")
+        mu = parameters["mu"]
+        sigma = parameters["sigma"]
+        z = torch.randn((n // 2,) + mu.shape[-1:], dtype=mu.dtype, device=mu.device)
+        plus = mu.unsqueeze(-2) + sigma.unsqueeze(-2) * z
+        minus = 2.0 * mu.unsqueeze(-2) - plus
+        return torch.cat([plus, minus], dim=-2)
 
     def _compute_gradients(self, samples: torch.Tensor, weights: torch.Tensor, ranking_used: Optional[str]) -> dict:
         if "parenthood_ratio" in self._parameters:
@@ -410,6 +432,9 @@ def make_functional_sampler(distribution_class: Type[Distribution], *, required_
     def _sample(num_solutions: int, *args) -> torch.Tensor:
         params = dict(zip(required_parameters, args))
         params.update(fixed_parameters)
+        fs = getattr(distribution_class, "functional_sample", NotImplemented)
+        if fs is not NotImplemented and fs is not None:
+            return fs(num_solutions, params)
         dist = distribution_class(params)
         return dist.sample(int(num_solutions))
 

@@ -39,9 +39,9 @@ def main():
     torch.cuda.synchronize()
     eager_dt = time.perf_counter() - t0
     eager_mean = s.status["mean_eval"]
-    # graphed
+    # graphed (10 generations per captured graph)
     s2 = make(1)
-    graphed = GraphedSearch(s2)
+    graphed = GraphedSearch(s2, generations_per_capture=10)
     graphed.capture()
     graphed.run(10)
     t0 = time.perf_counter()
