@@ -188,9 +188,13 @@ def test_make_functional_sampler_and_grad_estimator():
     grad = make_functional_grad_estimator(
         SeparableGaussian, required_parameters=["mu", "sigma"], objective_sense="min", ranking_method="centered"
     )
-    fit = (pop**2).sum(-1)
-    mu_g, sigma_g = grad(pop, fit, mu, sigma)
+    center = torch.ones(6) * 2.0
+    pop2 = sample(200, center, sigma)
+    fit = (pop2**2).sum(-1)
+    mu_g, sigma_g = grad(pop2, fit, center, sigma)
     assert mu_g.shape == (6,)
     assert sigma_g.shape == (6,)
-    # descending against the gradient reduces the sphere value
-    assert float(((mu + 0.5 * mu_g) ** 2).sum()) <= float((mu**2).sum()) + 1.0
+    # following the (ascent) gradient of the min-sense utilities moves the
+    # center toward the sphere optimum
+    stepped = center + 0.05 * mu_g
+    assert float((stepped**2).sum()) < float((center**2).sum())
