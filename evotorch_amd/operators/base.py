@@ -105,9 +105,14 @@ class CrossOver(CopyingOperator):
         num_pairings = max(1, num_children // 2)
 
         if self._problem.is_multi_objective and self._obj_index is None:
-            ranks, _ = batch.compute_pareto_ranks(crowdsort=False)
-            # lower pareto rank is better: negate for "higher is better"
-            utils = -ranks.to(torch.float32)
+            # crowded-comparison operator (Deb 2002): primary key = pareto
+            # rank, tie-break = crowding distance within the front
+            ranks, crowd = batch.compute_pareto_ranks(crowdsort=True)
+            n = len(batch)
+            crowd_order = torch.nan_to_num(crowd.to(torch.float64), posinf=1e300).argsort(descending=True)
+            crowd_pos = torch.empty_like(crowd_order)
+            crowd_pos.scatter_(0, crowd_order, torch.arange(n, device=ranks.device))
+            utils = -(ranks.to(torch.float64) * (n + 1) + crowd_pos).to(torch.float32)
         else:
             utils = batch.utility(self._obj_index, ranking_method="centered")
 

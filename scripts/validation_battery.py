@@ -16,7 +16,7 @@ from evotorch_amd import Problem
 from evotorch_amd.algorithms import CMAES, PGPE, SNES, GeneticAlgorithm, GraphedSearch
 from evotorch_amd.decorators import vectorized
 from evotorch_amd.neuroevolution import SyntheticRolloutProblem
-from evotorch_amd.operators import GaussianMutation, SimulatedBinaryCrossOver
+from evotorch_amd.operators import GaussianMutation, PolynomialMutation, SimulatedBinaryCrossOver
 
 DEVICE = "cuda:0" if torch.cuda.is_available() else "cpu"
 
@@ -94,8 +94,8 @@ def main():
     prob = Problem(["min", "min"], zdt1, solution_length=12, initial_bounds=(0.0, 1.0), bounds=(0.0, 1.0),
                    device=DEVICE, seed=5)
     ga = GeneticAlgorithm(prob, popsize=512, operators=[
-        SimulatedBinaryCrossOver(prob, tournament_size=3, eta=10.0),
-        GaussianMutation(prob, stdev=0.05),
+        SimulatedBinaryCrossOver(prob, tournament_size=2, eta=15.0),
+        PolynomialMutation(prob, eta=20.0),
     ])
     t0 = time.perf_counter()
     ga.run(100)
