@@ -127,3 +127,40 @@ def test_graphed_pgpe_clipup():
     first = graphed.mean_eval
     graphed.run(300)
     assert graphed.mean_eval < first * 0.5, (first, graphed.mean_eval)
+
+
+@requires_gpu
+def test_structures_on_gpu():
+    from evotorch_amd.utils import CDict, CList, CMemory
+
+    m = CMemory(4, num_keys=8, batch_size=16, fill_with=0.0, device="cuda:0")
+    keys = torch.randint(0, 8, (16,), device="cuda:0")
+    m.set_(keys, torch.randn(16, 4, device="cuda:0"))
+    got = m.get(keys)
+    assert got.shape == (16, 4)
+
+    lst = CList(max_length=8, batch_size=32, device="cuda:0")
+    lst.append_(torch.arange(32.0, device="cuda:0"))
+    lst.append_(torch.arange(32.0, device="cuda:0") * 2, where=torch.arange(32, device="cuda:0") % 2 == 0)
+    assert lst.length.sum().item() == 32 + 16
+
+    d = CDict(num_keys=5, batch_size=8, fill_with=0.0, device="cuda:0")
+    k = torch.zeros(8, dtype=torch.int64, device="cuda:0")
+    d.set_(k, 3.0)
+    assert bool(d.contains(k).all())
+
+
+@requires_gpu
+def test_genetic_programming_example_gpu():
+    import os
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    result = subprocess.run(
+        [sys.executable, os.path.join(repo, "examples", "genetic_programming.py"),
+         "--generations", "10", "--popsize", "128", "--device", "cuda:0"],
+        capture_output=True, text=True, timeout=300,
+    )
+    assert result.returncode == 0, result.stderr[-2000:]
+    assert "best MSE" in result.stdout
