@@ -120,24 +120,30 @@ torch::Tensor pareto_ranks(torch::Tensor utils) {
     const int threads = 256;
     const int blocks = (int)((n + threads - 1) / threads);
     auto stream = at::cuda::getCurrentCUDAStream();
+    // Peel in blind batches of kPeelBatch fronts with ONE host sync per
+    // batch (a sync per front costs hundreds of round-trips at large N:
+    // a random 16k population can have hundreds of fronts). A peel with
+    // an empty front mask is a no-op, so over-issuing is safe.
+    constexpr int kPeelBatch = 8;
     int64_t front_index = 0;
-    while (true) {
-        auto front_mask = (counts == 0);  // unassigned & undominated
-        // ONE host sync per front: are we done / does the front exist?
-        const int64_t front_size = front_mask.sum().item<int64_t>();
-        if (front_size == 0) {
-            // numerical corner: assign any stragglers to the current front
-            auto remaining = counts >= 0;
-            if (remaining.sum().item<int64_t>() == 0) break;
-            ranks.masked_fill_(remaining, front_index);
+    while (front_index <= n) {
+        for (int k = 0; k < kPeelBatch; ++k) {
+            auto front_mask = (counts == 0);  // unassigned & undominated (device-side)
+            auto mask_u8 = front_mask.to(torch::kUInt8).contiguous();
+            hipLaunchKernelGGL(peel_front_kernel, dim3(blocks), dim3(threads), 0, stream, utils_f.data_ptr<float>(),
+                               counts.data_ptr<int>(), ranks.data_ptr<int64_t>(), mask_u8.data_ptr<uint8_t>(), n, m,
+                               front_index);
+            ++front_index;
+        }
+        const int64_t remaining = (counts >= 0).sum().item<int64_t>();  // one sync per batch
+        if (remaining == 0) break;
+        // numerical corner: no zero-count candidates among the remaining —
+        // assign the stragglers and stop
+        const int64_t candidates = (counts == 0).sum().item<int64_t>();
+        if (candidates == 0) {
+            ranks.masked_fill_(counts >= 0, front_index);
             break;
         }
-        auto mask_u8 = front_mask.to(torch::kUInt8).contiguous();
-        hipLaunchKernelGGL(peel_front_kernel, dim3(blocks), dim3(threads), 0, stream, utils_f.data_ptr<float>(),
-                           counts.data_ptr<int>(), ranks.data_ptr<int64_t>(), mask_u8.data_ptr<uint8_t>(), n, m,
-                           front_index);
-        ++front_index;
-        if (front_index > n) break;  // safety
     }
     return ranks;
 }

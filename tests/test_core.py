@@ -207,3 +207,39 @@ def test_problem_bound_evaluator():
     out3 = ev(x3)
     assert out3.shape == (2, 5)
     assert torch.allclose(out3, (x3**2).sum(-1), atol=1e-5)
+
+
+def test_crowding_distances_match_bruteforce():
+    from evotorch_amd.core import _crowding_distances
+
+    torch.manual_seed(0)
+    for trial in range(3):
+        n, m = 60, 2 + trial % 2
+        utils = torch.randn(n, m)
+        # synthetic front assignment with varied sizes
+        ranks = torch.randint(0, 7, (n,))
+
+        # brute force reference (the per-front formulation)
+        crowd_ref = torch.zeros(n)
+        for front in torch.unique(ranks):
+            idx = torch.nonzero(ranks == front, as_tuple=True)[0]
+            if len(idx) <= 2:
+                crowd_ref[idx] = float("inf")
+                continue
+            sub = utils[idx]
+            for j in range(m):
+                order = sub[:, j].argsort()
+                sv = sub[order, j]
+                span = sv[-1] - sv[0]
+                if float(span) == 0.0:
+                    continue
+                contrib = torch.zeros(len(idx))
+                contrib[order[0]] = float("inf")
+                contrib[order[-1]] = float("inf")
+                contrib[order[1:-1]] = (sv[2:] - sv[:-2]) / span
+                crowd_ref[idx] += contrib
+
+        crowd = _crowding_distances(utils, ranks)
+        finite = torch.isfinite(crowd_ref)
+        assert torch.equal(torch.isfinite(crowd), finite)
+        assert torch.allclose(crowd[finite], crowd_ref[finite], atol=1e-5)
