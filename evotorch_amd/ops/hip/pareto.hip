@@ -61,29 +61,48 @@ __global__ void domination_counts_kernel(const float* __restrict__ utils, int* _
     counts[j] = count;
 }
 
-// Assign front `front_index` to zero-count unassigned solutions, then
-// subtract their domination contributions from everyone else. Returns (in
-// num_assigned[0]) how many were assigned this round.
-__global__ void peel_front_kernel(const float* __restrict__ utils, int* __restrict__ counts,
-                                  int64_t* __restrict__ ranks, const uint8_t* __restrict__ front_mask,
-                                  int64_t n, int m, int64_t front_index) {
+// Front peeling without host syncs, in two device-side stages per front:
+//   compact: zero-count unassigned solutions claim a slot in front_list
+//            (atomic counter), get their rank assigned, and are marked.
+//   subtract: every still-unassigned solution subtracts the domination
+//            contributions of the compacted front members ONLY — total
+//            subtract work over the whole sort is one O(N²·M) pass
+//            (each solution appears in exactly one front), instead of the
+//            O(N²·fronts) full-mask scans of a mask-based peel.
+
+__global__ void compact_front_kernel(int* __restrict__ counts, int64_t* __restrict__ ranks,
+                                     int* __restrict__ front_list, int* __restrict__ front_count, int64_t n,
+                                     int64_t front_index) {
     const int64_t j = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+    if (j == 0 && front_index == 0) {
+        // nothing: front_count reset is handled by reset_count_kernel
+    }
     if (j >= n) return;
-    if (front_mask[j]) {
+    if (counts[j] == 0) {
+        const int pos = atomicAdd(front_count, 1);
+        front_list[pos] = (int)j;
         ranks[j] = front_index;
         counts[j] = -1;  // assigned marker
-        return;
     }
-    if (counts[j] < 0) return;  // already assigned earlier
+}
+
+__global__ void reset_count_kernel(int* __restrict__ front_count) { *front_count = 0; }
+
+__global__ void subtract_front_kernel(const float* __restrict__ utils, int* __restrict__ counts,
+                                      const int* __restrict__ front_list, const int* __restrict__ front_count,
+                                      int64_t n, int m) {
+    const int64_t j = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+    if (j >= n) return;
+    if (counts[j] < 0) return;  // already assigned
     float mine[kMaxObjCache];
     const bool cached = m <= kMaxObjCache;
     if (cached) {
         for (int k = 0; k < m; ++k) mine[k] = utils[j * m + k];
     }
+    const int fc = *front_count;
     int removed = 0;
-    for (int64_t i = 0; i < n; ++i) {
-        if (!front_mask[i]) continue;
-        const float* other = utils + i * m;
+    for (int t = 0; t < fc; ++t) {
+        const float* other = utils + (int64_t)front_list[t] * m;
         const float* me = cached ? mine : (utils + j * m);
         bool ge_all = true, gt_any = false;
         for (int k = 0; k < m; ++k) {
@@ -123,16 +142,19 @@ torch::Tensor pareto_ranks(torch::Tensor utils) {
     // Peel in blind batches of kPeelBatch fronts with ONE host sync per
     // batch (a sync per front costs hundreds of round-trips at large N:
     // a random 16k population can have hundreds of fronts). A peel with
-    // an empty front mask is a no-op, so over-issuing is safe.
-    constexpr int kPeelBatch = 8;
+    // an empty front is a no-op, so over-issuing is safe.
+    auto front_list = torch::empty({n}, utils_f.options().dtype(torch::kInt32));
+    auto front_count = torch::zeros({1}, utils_f.options().dtype(torch::kInt32));
+    constexpr int kPeelBatch = 16;
     int64_t front_index = 0;
     while (front_index <= n) {
         for (int k = 0; k < kPeelBatch; ++k) {
-            auto front_mask = (counts == 0);  // unassigned & undominated (device-side)
-            auto mask_u8 = front_mask.to(torch::kUInt8).contiguous();
-            hipLaunchKernelGGL(peel_front_kernel, dim3(blocks), dim3(threads), 0, stream, utils_f.data_ptr<float>(),
-                               counts.data_ptr<int>(), ranks.data_ptr<int64_t>(), mask_u8.data_ptr<uint8_t>(), n, m,
+            hipLaunchKernelGGL(reset_count_kernel, dim3(1), dim3(1), 0, stream, front_count.data_ptr<int>());
+            hipLaunchKernelGGL(compact_front_kernel, dim3(blocks), dim3(threads), 0, stream, counts.data_ptr<int>(),
+                               ranks.data_ptr<int64_t>(), front_list.data_ptr<int>(), front_count.data_ptr<int>(), n,
                                front_index);
+            hipLaunchKernelGGL(subtract_front_kernel, dim3(blocks), dim3(threads), 0, stream, utils_f.data_ptr<float>(),
+                               counts.data_ptr<int>(), front_list.data_ptr<int>(), front_count.data_ptr<int>(), n, m);
             ++front_index;
         }
         const int64_t remaining = (counts >= 0).sum().item<int64_t>();  // one sync per batch
