@@ -338,7 +338,9 @@ __global__ __launch_bounds__(512, 2) void rollout_linear_kernel(RolloutArgs args
         }                                                                                     \
         _Pragma("unroll") for (int offset = kGroup / 2; offset > 0; offset >>= 1) {           \
             _Pragma("unroll") for (int r = 0; r < (NR); ++r) {                                \
-                acc[r] += __shfl_down(acc[r], offset, kWaveSize);                             \
+                /* validity is GROUP-uniform, so this divergence masks      */                \
+                /* whole 8-lane groups: invalid rounds issue no bpermutes   */                \
+                if (descs[r].valid) acc[r] += __shfl_down(acc[r], offset, kWaveSize);         \
             }                                                                                 \
         }                                                                                     \
     }
