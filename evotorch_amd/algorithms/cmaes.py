@@ -244,6 +244,23 @@ class CMAES(SearchAlgorithm, SinglePopulationAlgorithmMixin):
                 self._A = self._cholesky(self._C)
                 self._steps_since_decompose = 0
 
+    def _state_items(self) -> dict:
+        return {
+            "m": self._m, "sigma": self._sigma, "C": self._C, "A": self._A,
+            "p_sigma": self._p_sigma, "p_c": self._p_c,
+            "steps_since_decompose": self._steps_since_decompose,
+        }
+
+    def _load_state_items(self, items: dict):
+        for name in ("m", "sigma", "C", "A", "p_sigma", "p_c"):
+            target = getattr(self, f"_{name}")
+            value = torch.as_tensor(items[name]).to(target.device, target.dtype)
+            if target.ndim == 0:
+                setattr(self, f"_{name}", value)
+            else:
+                target.copy_(value)
+        self._steps_since_decompose = int(items.get("steps_since_decompose", 0))
+
     def _device_is_gpu(self) -> bool:
         return self._m.device.type == "cuda"
 

@@ -102,6 +102,22 @@ class GeneticAlgorithm(SearchAlgorithm, SinglePopulationAlgorithmMixin, Extended
     def popsize(self) -> int:
         return self._popsize
 
+    def _state_items(self) -> dict:
+        if self._population is None:
+            return {}
+        return {"values": self._population.unsafe_values, "evals": self._population.unsafe_evals}
+
+    def _load_state_items(self, items: dict):
+        if "values" not in items:
+            self._population = None
+            return
+        problem = self.problem
+        values = torch.as_tensor(items["values"])
+        batch = SolutionBatch(problem, popsize=values.shape[0], empty=True)
+        batch.access_values().copy_(values.to(batch.device, batch.dtype))
+        batch.unsafe_evals.copy_(torch.as_tensor(items["evals"]).to(batch.device))
+        self._population = batch
+
     def _step(self):
         problem = self.problem
         if self._population is None:

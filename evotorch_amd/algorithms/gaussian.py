@@ -254,6 +254,33 @@ class GaussianSearchAlgorithm(SearchAlgorithm, SinglePopulationAlgorithmMixin):
     def obj_index(self) -> int:
         return self._obj_index
 
+    def _state_items(self) -> dict:
+        items = {"distribution": dict(self._distribution.parameters)}
+        opt = self._optimizer
+        if opt is not None:
+            opt_state = {}
+            for attr in ("_velocity", "_m", "_v"):
+                if getattr(opt, attr, None) is not None:
+                    opt_state[attr] = getattr(opt, attr)
+            if hasattr(opt, "_t"):
+                opt_state["_t"] = opt._t
+            items["optimizer"] = opt_state
+        return items
+
+    def _load_state_items(self, items: dict):
+        dist_params = items.get("distribution", {})
+        params = {
+            k: (torch.as_tensor(v).to(self._distribution.device, self._distribution.dtype) if isinstance(v, torch.Tensor) else v)
+            for k, v in dist_params.items()
+        }
+        self._distribution = type(self._distribution)(params, dtype=self._distribution.dtype, device=self._distribution.device)
+        opt = self._optimizer
+        for attr, v in items.get("optimizer", {}).items():
+            if attr == "_t":
+                opt._t = int(v)
+            elif getattr(opt, attr, None) is not None:
+                getattr(opt, attr).copy_(torch.as_tensor(v).to(getattr(opt, attr).device))
+
     def _get_mu(self) -> torch.Tensor:
         return self._distribution.parameters["mu"]
 

@@ -167,6 +167,51 @@ class SearchAlgorithm(LazyReporter):
     def reset_first_step_datetime(self):
         self._first_step_datetime = None
 
+    # -- checkpoint / resume -------------------------------------------------
+    # (a green-field addition relative to the reference, which only offers
+    # whole-object pickling and PicklingLogger snapshots — SURVEY.md §5.4)
+
+    def _state_items(self) -> dict:
+        """Subclass hook: the tensors/values that define search progress."""
+        return {}
+
+    def _load_state_items(self, state: dict):
+        raise NotImplementedError(f"{type(self).__name__} does not support load_state_dict")
+
+    def state_dict(self) -> dict:
+        """Resumable search state (cpu tensors): algorithm parameters plus
+        any problem-side state (observation normalization, counters)."""
+        import torch as _torch
+
+        def to_cpu(x):
+            if isinstance(x, _torch.Tensor):
+                return x.detach().cpu().clone()
+            if isinstance(x, dict):
+                return {k: to_cpu(v) for k, v in x.items()}
+            return x
+
+        state = {"steps_count": self._steps_count, "algorithm": type(self).__name__}
+        state["items"] = to_cpu(self._state_items())
+        problem = self._problem
+        if hasattr(problem, "obs_norm"):
+            rn = problem.obs_norm
+            c, s, ss = rn.stats_triple()
+            state["obs_norm"] = {"count": c.cpu(), "sum": s.cpu(), "sum_sq": ss.cpu()}
+        return state
+
+    def load_state_dict(self, state: dict):
+        if state.get("algorithm") not in (type(self).__name__, None):
+            raise ValueError(f"Checkpoint is for {state.get('algorithm')}, not {type(self).__name__}")
+        self._steps_count = int(state.get("steps_count", 0))
+        self._load_state_items(state.get("items", {}))
+        problem = self._problem
+        if "obs_norm" in state and hasattr(problem, "obs_norm"):
+            rn = problem.obs_norm
+            rn.reset()
+            o = state["obs_norm"]
+            rn.update((o["count"], o["sum"], o["sum_sq"]))
+        return self
+
 
 class SinglePopulationAlgorithmMixin:
     """Adds pop_best / mean_eval / median_eval / pop_best_eval status items

@@ -136,3 +136,72 @@ def test_num_interactions_adaptive_popsize():
     searcher.step()
     # 10 solutions/batch * 3 = 30 interactions; need >= 100 → 4 batches = 40 solutions
     assert len(searcher.population) == 40
+
+
+def test_state_dict_resume_pgpe():
+    import pickle
+
+    def fresh():
+        prob = make_problem(seed=77)
+        return PGPE(prob, popsize=40, center_learning_rate=0.2, stdev_learning_rate=0.1,
+                    stdev_init=2.0, optimizer="clipup")
+
+    a = fresh()
+    a.run(10)
+    blob = pickle.dumps(a.state_dict())
+
+    b = fresh()
+    b.load_state_dict(pickle.loads(blob))
+    assert b.step_count == 10
+    assert torch.allclose(
+        torch.Tensor.as_subclass(b.status["center"], torch.Tensor),
+        torch.Tensor.as_subclass(a.status["center"], torch.Tensor),
+    )
+    b.run(5)  # resumes cleanly
+    assert b.step_count == 15
+
+
+def test_state_dict_resume_cmaes():
+    from evotorch_amd.algorithms import CMAES
+
+    prob = make_problem(seed=78)
+    a = CMAES(prob, stdev_init=1.0)
+    a.run(10)
+    sd = a.state_dict()
+    prob2 = make_problem(seed=78)
+    b = CMAES(prob2, stdev_init=1.0)
+    b.load_state_dict(sd)
+    assert float(b.status["sigma"]) == pytest.approx(float(a.status["sigma"]))
+
+
+def test_state_dict_resume_ga():
+    from evotorch_amd.algorithms import GeneticAlgorithm
+    from evotorch_amd.operators import GaussianMutation, OnePointCrossOver
+
+    prob = make_problem(seed=79)
+    ga = GeneticAlgorithm(prob, popsize=20, operators=[
+        OnePointCrossOver(prob, tournament_size=2), GaussianMutation(prob, stdev=0.2)])
+    ga.run(5)
+    sd = ga.state_dict()
+    prob2 = make_problem(seed=79)
+    ga2 = GeneticAlgorithm(prob2, popsize=20, operators=[
+        OnePointCrossOver(prob2, tournament_size=2), GaussianMutation(prob2, stdev=0.2)])
+    ga2.load_state_dict(sd)
+    assert torch.allclose(ga2.population.unsafe_values, ga.population.unsafe_values)
+    ga2.run(3)
+
+
+def test_state_dict_includes_obs_norm():
+    from evotorch_amd.neuroevolution import SyntheticRolloutProblem
+
+    prob = SyntheticRolloutProblem(seed=80, episode_length=5)
+    searcher = PGPE(prob, popsize=16, center_learning_rate=0.1, stdev_learning_rate=0.1,
+                    stdev_init=0.1, distributed=True)
+    searcher.run(2)
+    sd = searcher.state_dict()
+    assert sd["obs_norm"]["count"] > 0
+    prob2 = SyntheticRolloutProblem(seed=81, episode_length=5)
+    s2 = PGPE(prob2, popsize=16, center_learning_rate=0.1, stdev_learning_rate=0.1,
+              stdev_init=0.1, distributed=True)
+    s2.load_state_dict(sd)
+    assert prob2.obs_norm.count == prob.obs_norm.count
