@@ -52,6 +52,7 @@ from .misc import (
     modify_tensor,
     modify_vector,
     numpy_copy,
+    make_batched_false_for_vmap,
     split_workload,
     stdev_from_radius,
     storage_ptr,

@@ -70,6 +70,7 @@ __all__ = [
     "make_gaussian_shaped_like",
     "stdev_from_radius",
     "split_workload",
+    "make_batched_false_for_vmap",
     "ErroneousResult",
     "numpy_copy",
     "storage_ptr",
@@ -649,6 +650,15 @@ def split_workload(workload: int, num_pieces: int) -> list:
     base = workload // num_pieces
     extra = workload % num_pieces
     return [base + (1 if i < extra else 0) for i in range(num_pieces)]
+
+
+def make_batched_false_for_vmap(device: Optional[Device] = None) -> torch.Tensor:
+    """A scalar False built from tensor ops rather than a python literal, so
+    that code running under `torch.func.vmap` can use it as the initial value
+    of a flag that is later combined (|=, where) with batched booleans
+    (reference tools/misc.py:2209). A fresh unbatched tensor broadcasts
+    correctly against batched operands inside vmap."""
+    return torch.zeros((), dtype=torch.bool, device=(device if device is None else torch.device(device)))
 
 
 class ErroneousResult:

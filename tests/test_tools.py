@@ -186,3 +186,17 @@ def test_penalty_and_log_barrier():
     assert float(lb) == pytest.approx(0.0)  # log(1) = 0
     lb2 = log_barrier(2.5, "<=", 2.0, penalty_sign="-", inf=100.0)
     assert float(lb2) == pytest.approx(-100.0)
+
+
+def test_make_batched_false_for_vmap():
+    from torch.func import vmap
+
+    from evotorch_amd.utils import make_batched_false_for_vmap
+
+    def f(x):
+        flag = make_batched_false_for_vmap()
+        flag = flag | (x > 0)
+        return torch.where(flag, x, -x)
+
+    out = vmap(f)(torch.tensor([-1.0, 2.0, -3.0]))
+    assert torch.equal(out, torch.tensor([1.0, 2.0, 3.0]))
