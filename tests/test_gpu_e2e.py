@@ -164,3 +164,31 @@ def test_genetic_programming_example_gpu():
     )
     assert result.returncode == 0, result.stderr[-2000:]
     assert "best MSE" in result.stdout
+
+
+@requires_gpu
+def test_state_dict_resume_on_gpu():
+    """Checkpoint from a GPU searcher restores onto a fresh one (cross-device:
+    saved tensors are cpu, load puts them back on cuda)."""
+    from evotorch_amd.algorithms import PGPE
+    from evotorch_amd.neuroevolution import SyntheticRolloutProblem
+
+    def fresh(seed):
+        prob = SyntheticRolloutProblem(seed=90, episode_length=20, device="cuda:0")
+        return PGPE(prob, popsize=64, center_learning_rate=0.15, stdev_learning_rate=0.1,
+                    stdev_init=0.1, distributed=True)
+
+    a = fresh(90)
+    a.run(3)
+    sd = a.state_dict()
+    assert all(not v.is_cuda for v in sd["items"]["distribution"].values())
+
+    b = fresh(90)
+    b.load_state_dict(sd)
+    assert b.step_count == 3
+    center_a = torch.Tensor.as_subclass(a.status["center"], torch.Tensor)
+    center_b = torch.Tensor.as_subclass(b.status["center"], torch.Tensor)
+    assert center_b.is_cuda and torch.allclose(center_a, center_b)
+    assert float(b.problem.obs_norm.count) == float(a.problem.obs_norm.count)
+    b.run(2)
+    assert b.step_count == 5
