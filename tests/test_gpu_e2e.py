@@ -181,7 +181,7 @@ def test_state_dict_resume_on_gpu():
     a = fresh(90)
     a.run(3)
     sd = a.state_dict()
-    assert all(not v.is_cuda for v in sd["items"]["distribution"].values())
+    assert all(not v.is_cuda for v in sd["items"]["distribution"].values() if torch.is_tensor(v))
 
     b = fresh(90)
     b.load_state_dict(sd)
