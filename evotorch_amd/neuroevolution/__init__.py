@@ -2,7 +2,7 @@
 /root/reference/src/evotorch/neuroevolution/__init__.py."""
 
 from .neproblem import NEProblem
-from .runningnorm import ObsNormLayer, RunningNorm
+from .runningnorm import ObsNormLayer, RunningNorm, RunningStat
 from .supervisedne import SupervisedNE
 from .synthetic import SyntheticRolloutProblem
 from .synthetic_env import SyntheticEnvSpec, rollout_eager
@@ -13,6 +13,7 @@ __all__ = [
     "NEProblem",
     "ObsNormLayer",
     "RunningNorm",
+    "RunningStat",
     "SupervisedNE",
     "SyntheticEnvSpec",
     "SyntheticRolloutProblem",

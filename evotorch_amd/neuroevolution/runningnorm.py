@@ -158,3 +158,93 @@ class ObsNormLayer(torch.nn.Module):
         if self._clip is not None:
             y = torch.clamp(y, self._clip[0], self._clip[1])
         return y
+
+
+class RunningStat:
+    """Numpy-side mergeable running mean/stdev (reference
+    net/runningstat.py:25,90). `RunningNorm` is the on-device torch
+    equivalent; this one serves CPU rollout paths (GymNE) where observations
+    arrive as numpy arrays and stats are merged across worker processes."""
+
+    def __init__(self):
+        self.reset()
+
+    def reset(self):
+        self._count = 0
+        self._sum = None
+        self._sum_sq = None
+
+    @property
+    def count(self) -> int:
+        return self._count
+
+    @property
+    def sum(self):
+        return self._sum
+
+    @property
+    def sum_of_squares(self):
+        return self._sum_sq
+
+    @property
+    def mean(self):
+        return self._sum / self._count
+
+    @property
+    def stdev(self):
+        import numpy as np
+
+        var = self._sum_sq / self._count - self.mean**2
+        return np.sqrt(np.clip(var, 1e-8, None))
+
+    def update(self, x):
+        """Accept a single observation vector, a 2-D batch, or another
+        RunningStat (merge)."""
+        import numpy as np
+
+        if isinstance(x, RunningStat):
+            if x._count == 0:
+                return
+            if self._count == 0:
+                self._count = x._count
+                self._sum = x._sum.copy()
+                self._sum_sq = x._sum_sq.copy()
+            else:
+                self._count += x._count
+                self._sum += x._sum
+                self._sum_sq += x._sum_sq
+            return
+        arr = np.asarray(x, dtype=np.float64)
+        if arr.ndim == 1:
+            arr = arr[None, :]
+        if self._count == 0:
+            self._sum = arr.sum(axis=0)
+            self._sum_sq = (arr**2).sum(axis=0)
+        else:
+            self._sum += arr.sum(axis=0)
+            self._sum_sq += (arr**2).sum(axis=0)
+        self._count += arr.shape[0]
+
+    def normalize(self, x):
+        import numpy as np
+
+        arr = np.asarray(x, dtype=np.float64)
+        return (arr - self.mean) / self.stdev
+
+    def to_running_norm(self, rn: "RunningNorm"):
+        """Pour these stats into a torch RunningNorm (device transfer)."""
+        import torch as _torch
+
+        if self._count == 0:
+            return rn
+        rn.update(
+            (
+                _torch.tensor(float(self._count)),
+                _torch.as_tensor(self._sum, dtype=_torch.float32),
+                _torch.as_tensor(self._sum_sq, dtype=_torch.float32),
+            )
+        )
+        return rn
+
+    def __repr__(self):
+        return f"<RunningStat count={self._count}>"

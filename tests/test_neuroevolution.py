@@ -242,3 +242,36 @@ def test_vecenvne_recurrent_policy():
     batch = prob.generate_batch(4)
     prob.evaluate(batch)
     assert batch.evals_are_ready
+
+
+def test_running_stat_numpy_merge():
+    import numpy as np
+
+    from evotorch_amd.neuroevolution import RunningNorm, RunningStat
+
+    rng = np.random.default_rng(0)
+    a_data = rng.normal(2.0, 3.0, size=(50, 4))
+    b_data = rng.normal(-1.0, 0.5, size=(30, 4))
+
+    a = RunningStat()
+    a.update(a_data)
+    b = RunningStat()
+    for row in b_data:
+        b.update(row)  # one-at-a-time path
+    merged = RunningStat()
+    merged.update(a)
+    merged.update(b)
+
+    all_data = np.concatenate([a_data, b_data])
+    assert merged.count == 80
+    np.testing.assert_allclose(merged.mean, all_data.mean(axis=0), rtol=1e-10)
+    np.testing.assert_allclose(merged.stdev, all_data.std(axis=0), rtol=1e-6)
+
+    normed = merged.normalize(all_data)
+    np.testing.assert_allclose(normed.mean(axis=0), 0.0, atol=1e-10)
+
+    # pour into the torch-side RunningNorm
+    rn = RunningNorm(shape=4)
+    merged.to_running_norm(rn)
+    assert float(rn.count) == 80
+    np.testing.assert_allclose(rn.mean.numpy(), all_data.mean(axis=0), rtol=1e-5)
