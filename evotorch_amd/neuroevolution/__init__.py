@@ -30,4 +30,8 @@ def __getattr__(name):
         from .gymne import GymNE
 
         return GymNE
+    if name == "ActClipLayer":
+        from .gymne import ActClipLayer
+
+        return ActClipLayer
     raise AttributeError(f"module {__name__!r} has no attribute {name!r}")
