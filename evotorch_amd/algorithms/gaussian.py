@@ -62,6 +62,7 @@ class GaussianSearchAlgorithm(SearchAlgorithm, SinglePopulationAlgorithmMixin):
         distributed: bool = False,
         popsize_weighted_grad_avg: Optional[bool] = None,
         ensure_even_popsize: bool = False,
+        grad_chunk_rows: Optional[int] = None,
     ):
         problem.ensure_numeric()
         problem.ensure_unbounded()
@@ -112,6 +113,9 @@ class GaussianSearchAlgorithm(SearchAlgorithm, SinglePopulationAlgorithmMixin):
         self._stdev_max_change = opt_length_tensor(stdev_max_change, "stdev_max_change")
 
         self._distributed = bool(distributed)
+        self._grad_chunk_rows = None if grad_chunk_rows is None else int(grad_chunk_rows)
+        if self._grad_chunk_rows is not None and not self._distributed:
+            raise ValueError("grad_chunk_rows (streaming gradients) requires distributed=True")
         if not self._distributed and popsize_weighted_grad_avg is not None:
             raise ValueError("popsize_weighted_grad_avg is only meaningful with distributed=True")
         self._popsize_weighted_grad_avg = popsize_weighted_grad_avg
@@ -153,6 +157,7 @@ class GaussianSearchAlgorithm(SearchAlgorithm, SinglePopulationAlgorithmMixin):
         else:
             self._step_non_distributed()
 
+
     def _step_distributed(self):
         fetched = self.problem.sample_and_compute_gradients(
             self._distribution,
@@ -162,6 +167,7 @@ class GaussianSearchAlgorithm(SearchAlgorithm, SinglePopulationAlgorithmMixin):
             num_interactions=self._num_interactions,
             ranking_method=self._ranking_method,
             ensure_even_popsize=self._ensure_even_popsize,
+            chunk_rows=self._grad_chunk_rows,
         )
         self._update_distribution(fetched["gradients"])
         self._mean_eval = fetched["mean_eval"]
@@ -324,6 +330,7 @@ class PGPE(GaussianSearchAlgorithm):
         obj_index: Optional[int] = None,
         distributed: bool = False,
         popsize_weighted_grad_avg: Optional[bool] = None,
+        grad_chunk_rows: Optional[int] = None,
     ):
         if symmetric:
             self.DISTRIBUTION_TYPE = SymmetricSeparableGaussian
@@ -352,6 +359,7 @@ class PGPE(GaussianSearchAlgorithm):
             distributed=distributed,
             popsize_weighted_grad_avg=popsize_weighted_grad_avg,
             ensure_even_popsize=symmetric,
+            grad_chunk_rows=grad_chunk_rows,
         )
 
 
@@ -385,6 +393,7 @@ class SNES(GaussianSearchAlgorithm):
         obj_index: Optional[int] = None,
         distributed: bool = False,
         popsize_weighted_grad_avg: Optional[bool] = None,
+        grad_chunk_rows: Optional[int] = None,
     ):
         if popsize is None:
             popsize = int(4 + math.floor(3 * math.log(problem.solution_length)))
@@ -421,6 +430,7 @@ class SNES(GaussianSearchAlgorithm):
             obj_index=obj_index,
             distributed=distributed,
             popsize_weighted_grad_avg=popsize_weighted_grad_avg,
+            grad_chunk_rows=grad_chunk_rows,
         )
 
 
@@ -498,6 +508,7 @@ class XNES(GaussianSearchAlgorithm):
         obj_index: Optional[int] = None,
         distributed: bool = False,
         popsize_weighted_grad_avg: Optional[bool] = None,
+        grad_chunk_rows: Optional[int] = None,
     ):
         if popsize is None:
             popsize = int(4 + math.floor(3 * math.log(problem.solution_length)))

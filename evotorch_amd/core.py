@@ -1080,9 +1080,22 @@ class Problem(TensorMakerMixin, Serializable):
         num_interactions: Optional[int] = None,
         popsize_max: Optional[int] = None,
         ensure_even_popsize: bool = False,
+        chunk_rows: Optional[int] = None,
     ) -> dict:
         """Sample a population from `distribution`, evaluate it, and return
         ``{"gradients": ..., "num_solutions": ..., "mean_eval": ...}``.
+
+        Streaming mode (`chunk_rows` given): the N×L population is never
+        materialized. Pass 1 samples and evaluates `chunk_rows` directions
+        at a time keeping only the N fitnesses; utilities are ranked over
+        the full fitness vector; pass 2 REGENERATES each chunk's noise from
+        the same counter-addressed philox stream and accumulates the exact
+        gradient. Memory is O(chunk_rows × L), so separable-Gaussian ES
+        scales to parameter vectors far beyond HBM (L ≈ 10⁸⁺ with small
+        chunks) — a capability the reference does not have (its population
+        is always materialized). Requires a
+        SeparableGaussian-family distribution; `num_interactions` adaptive
+        popsize is not supported in this mode.
 
         SPMD mode (Comm attached, world > 1): each rank samples and
         evaluates ``popsize / world`` solutions using its own slice of the
@@ -1095,7 +1108,9 @@ class Problem(TensorMakerMixin, Serializable):
         obj_index = 0 if obj_index is None else int(obj_index)
         comm = self._comm
         self._before_grad_hook()
-        if comm is not None and comm.world_size > 1:
+        if chunk_rows is not None:
+            result = self._sample_and_compute_gradients_streamed(distribution, int(popsize), obj_index, ranking_method, int(chunk_rows), comm)
+        elif comm is not None and comm.world_size > 1:
             result = self._sample_and_compute_gradients_sharded(distribution, int(popsize), obj_index, ranking_method, comm, num_interactions, popsize_max, ensure_even_popsize)
         else:
             result = self._sample_and_compute_gradients(distribution, int(popsize), obj_index, ranking_method, num_interactions, popsize_max, ensure_even_popsize)
@@ -1168,6 +1183,101 @@ class Problem(TensorMakerMixin, Serializable):
         for k in grads:
             grads[k] = grads[k] * (local_popsize / total)
         comm.all_reduce_container(grads)
+        return {
+            "gradients": grads,
+            "num_solutions": total,
+            "mean_eval": torch.nanmean(all_fit),  # 0-dim tensor: no host sync
+        }
+
+    def _get_stream_batch(self, popsize: int) -> "SolutionBatch":
+        cache = getattr(self, "_stream_batch_cache", None)
+        if cache is None:
+            cache = self._stream_batch_cache = {}
+        batch = cache.get(popsize)
+        if batch is None or batch.device != torch.device(self._device):
+            batch = self.generate_batch(popsize, empty=True)
+            cache.clear()  # at most the main chunk size + one tail size alive
+            cache[popsize] = batch
+        batch.forget_evals()
+        return batch
+
+    def _sample_and_compute_gradients_streamed(
+        self, distribution, popsize: int, obj_index: int, ranking_method, chunk_rows: int, comm
+    ) -> dict:
+        """Two-pass streaming ES gradients — see sample_and_compute_gradients."""
+        if not hasattr(distribution, "compute_gradients_streamed"):
+            raise ValueError(f"{type(distribution).__name__} does not support streamed gradients")
+        from .ops.dispatch import _seed_from_generator
+        from .utils import ranking as ranking_mod
+
+        world = comm.world_size if comm is not None else 1
+        my_rank = comm.rank if comm is not None else 0
+        local_popsize = popsize // world
+        symmetric = bool(getattr(distribution, "_symmetric", False))
+        if symmetric and local_popsize % 2 != 0:
+            local_popsize += 1
+        directions = local_popsize // 2 if symmetric else local_popsize
+        length = self._solution_length
+        # philox counters cover 4 elements each: chunk starts must land on
+        # 4-element boundaries of the flat (direction-major) noise stream
+        if length % 4 != 0 and chunk_rows % 4 != 0:
+            chunk_rows = max(4, (chunk_rows // 4) * 4)
+        chunk_rows = min(chunk_rows, directions)
+        seed = _seed_from_generator(self._generator, self._device)
+
+        fits = torch.empty(local_popsize, dtype=self._eval_dtype, device=self._device)
+
+        def chunk_spans():
+            r0 = 0
+            while r0 < directions:
+                rows = min(chunk_rows, directions - r0)
+                yield r0, rows
+                r0 += rows
+
+        # -- pass 1: sample + evaluate, keep fitnesses only ------------------
+        for r0, rows in chunk_spans():
+            batch = self._get_stream_batch(rows * 2 if symmetric else rows)
+            with record_range("stream_sample"):
+                distribution.fill_counter_addressed(batch.access_values(), seed=seed, elem_offset=r0 * length)
+            with record_range("stream_eval"):
+                self._before_eval_hook(batch)
+                self._evaluate_batch(batch)
+                if self._store_solution_stats:
+                    self._update_solution_stats(batch)
+                self._after_eval_status = self._after_eval_hook.accumulate_dict(batch)
+            evals = batch._evals[:, obj_index]
+            if symmetric:
+                fits[r0 : r0 + rows] = evals[:rows]
+                fits[directions + r0 : directions + r0 + rows] = evals[rows:]
+            else:
+                fits[r0 : r0 + rows] = evals
+
+        # -- global ranking ---------------------------------------------------
+        if comm is not None and world > 1:
+            all_fit = comm.all_gather_vector(fits)
+        else:
+            all_fit = fits
+        sense = self._senses[obj_index]
+        all_utils = ranking_mod.rank(all_fit, ranking_method or "raw", higher_is_better=(sense == "max"))
+        my_utils = all_utils[my_rank * local_popsize : (my_rank + 1) * local_popsize].to(dtype=self._dtype if self._dtype.is_floating_point else torch.float32)
+
+        # -- pass 2: regenerate noise per chunk, accumulate exact gradients --
+        def regen_chunks():
+            for r0, rows in chunk_spans():
+                batch = self._get_stream_batch(rows * 2 if symmetric else rows)
+                with record_range("stream_regen"):
+                    distribution.fill_counter_addressed(batch.access_values(), seed=seed, elem_offset=r0 * length)
+                yield batch._values, r0, rows
+
+        with record_range("stream_grad"):
+            grads = distribution.compute_gradients_streamed(regen_chunks(), my_utils, ranking_used=(ranking_method or "raw"))
+        if comm is not None and world > 1:
+            total = world * local_popsize
+            for k in grads:
+                grads[k] = grads[k] * (local_popsize / total)
+            comm.all_reduce_container(grads)
+        else:
+            total = local_popsize
         return {
             "gradients": grads,
             "num_solutions": total,

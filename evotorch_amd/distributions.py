@@ -201,6 +201,50 @@ class SeparableGaussian(Distribution):
 
         ops.sample_gaussian(out, self.mu, self.sigma, symmetric=self._symmetric, generator=generator)
 
+    def fill_counter_addressed(self, out: torch.Tensor, *, seed: int, elem_offset: int = 0):
+        """Sample a row-block of the virtual population: noise element e is
+        philox(seed, (elem_offset+e)//4), identical however the population
+        is chunked (and identical on CPU and GPU). Enables the streaming
+        large-L gradient path (Problem.sample_and_compute_gradients with
+        chunk_rows) to REGENERATE noise in pass 2 instead of storing the
+        N×L population."""
+        from . import ops
+
+        ops.sample_gaussian(out, self.mu, self.sigma, symmetric=self._symmetric, seed=seed, elem_offset=elem_offset)
+
+    def compute_gradients_streamed(self, chunk_iter, weights: torch.Tensor, *, ranking_used: Optional[str]) -> dict:
+        """Gradients over a chunked (never-materialized) population.
+
+        `chunk_iter` yields (values_chunk, row0, rows) where row0/rows index
+        DIRECTIONS (= plus-rows for the symmetric subclass, plain rows
+        otherwise); `weights` is the full local utility vector. Weighted
+        sums are additive over row-blocks, so raw sums accumulate per chunk
+        and the (global, weight-dependent) normalization is applied once at
+        the end — exactly equal to the unstreamed gradient.
+        """
+        if "parenthood_ratio" in self._parameters:
+            raise ValueError("elite (parenthood_ratio) gradients cannot be streamed")
+        from . import ops
+
+        w = self._centered_weights(weights, ranking_used)
+        d = w.shape[0] // 2 if self._symmetric else w.shape[0]
+        mu_acc = torch.zeros_like(self.mu, dtype=torch.float32)
+        sigma_acc = torch.zeros_like(self.sigma, dtype=torch.float32)
+        for values_chunk, row0, rows in chunk_iter:
+            if self._symmetric:
+                w_c = torch.cat([w[row0 : row0 + rows], w[d + row0 : d + row0 + rows]])
+            else:
+                w_c = w[row0 : row0 + rows]
+            g_mu, g_sigma = ops.es_gradients(values_chunk, self.mu, self.sigma, w_c, symmetric=self._symmetric)
+            mu_acc += g_mu.to(torch.float32)
+            sigma_acc += g_sigma.to(torch.float32)
+        mu_grad = mu_acc.to(self.mu.dtype)
+        sigma_grad = sigma_acc.to(self.sigma.dtype)
+        return {
+            "mu": self._divide_grad("mu", mu_grad, w),
+            "sigma": self._divide_grad("sigma", sigma_grad, w),
+        }
+
     @classmethod
     def functional_sample(cls, num_solutions: int, parameters: dict) -> torch.Tensor:
         """Pure out-of-place sampling (vmap-safe under
@@ -328,6 +372,19 @@ class ExpSeparableGaussian(SeparableGaussian):
 
         mu_grad, sigma_grad = ops.snes_gradients(samples, self.mu, self.sigma, weights)
         return {"mu": mu_grad, "sigma": sigma_grad}
+
+    def compute_gradients_streamed(self, chunk_iter, weights: torch.Tensor, *, ranking_used: Optional[str]) -> dict:
+        from . import ops
+
+        if ranking_used != "nes":
+            weights = weights / weights.abs().sum()
+        mu_acc = torch.zeros_like(self.mu, dtype=torch.float32)
+        sigma_acc = torch.zeros_like(self.sigma, dtype=torch.float32)
+        for values_chunk, row0, rows in chunk_iter:
+            g_mu, g_sigma = ops.snes_gradients(values_chunk, self.mu, self.sigma, weights[row0 : row0 + rows])
+            mu_acc += g_mu.to(torch.float32)
+            sigma_acc += g_sigma.to(torch.float32)
+        return {"mu": mu_acc.to(self.mu.dtype), "sigma": sigma_acc.to(self.sigma.dtype)}
 
     def update_parameters(self, gradients: dict, *, learning_rates: Optional[dict] = None, optimizers: Optional[dict] = None) -> "ExpSeparableGaussian":
         new_mu = self.mu + self._follow_gradient("mu", gradients["mu"], learning_rates=learning_rates, optimizers=optimizers)

@@ -56,11 +56,12 @@ def _box_muller4(r0, r1, r2, r3):
     return np.stack([z0, z1, z2, z3], axis=-1)  # (..., 4)
 
 
-def philox_normals(seed: int, stream_id: int, n_elements: int) -> torch.Tensor:
-    """The first n_elements of stream (seed, stream_id) — matches
-    philox_normal4 element indexing."""
+def philox_normals(seed: int, stream_id: int, n_elements: int, *, idx4_offset: int = 0) -> torch.Tensor:
+    """The first n_elements of stream (seed, stream_id) starting at counter
+    idx4_offset — matches philox_normal4 element indexing (element e of the
+    stream is produced by counter e//4, so idx4_offset = elem_offset//4)."""
     n4 = (n_elements + 3) // 4
-    idx4 = np.arange(n4, dtype=np.uint64)
+    idx4 = np.arange(n4, dtype=np.uint64) + np.uint64(idx4_offset)
     c0 = idx4.astype(np.uint32)
     c1 = (idx4 >> np.uint64(32)).astype(np.uint32)
     c2 = np.full(n4, np.uint32(stream_id & 0xFFFFFFFF), dtype=np.uint32)

@@ -77,8 +77,6 @@ def _allow_eager_on_gpu() -> bool:
 
 _SPLITMIX_INC = 0x9E3779B97F4A7C15
 _seed_fallback_counter = [0x1234ABCD]
-_gen_counters: dict = {}
-
 
 def _splitmix64(x: int) -> int:
     x = (x + _SPLITMIX_INC) & 0xFFFFFFFFFFFFFFFF
@@ -89,16 +87,28 @@ def _splitmix64(x: int) -> int:
 
 
 def _seed_from_generator(generator: Optional[torch.Generator], device: torch.device) -> int:
-    """Derive a fresh 63-bit seed deterministically from the generator's
-    initial seed plus a host-side call counter (splitmix64 mix) — NO device
-    RNG draw, so kernel launches never force a host-device sync."""
+    """Derive a fresh 63-bit seed deterministically from the generator's OWN
+    state — no global bookkeeping, no device sync.
+
+    CUDA generators: PyTorch tracks the philox (seed, offset) pair host-side,
+    so get_offset()/set_offset() advance the generator without touching the
+    GPU; the returned seed is splitmix64(initial_seed xor golden*offset).
+    Two generators seeded identically produce identical seed sequences, and
+    reseeding (manual_seed resets offset to 0) restarts the sequence.
+
+    CPU generators: a single host randint draw (stateful, cheap).
+
+    (An earlier version kept a module dict keyed by id(generator); after a
+    generator was garbage collected its id could be reused by a NEW
+    generator, silently resuming the dead one's counter. Generator-intrinsic
+    state cannot alias.)"""
     if generator is not None:
-        key = id(generator)
-        counter = _gen_counters.get(key)
-        if counter is None:
-            counter = int(generator.initial_seed()) & 0xFFFFFFFFFFFFFFFF
-        _gen_counters[key] = counter + 1
-        return _splitmix64(counter)
+        if generator.device.type == "cuda":
+            off = int(generator.get_offset())
+            generator.set_offset(off + 4)
+            mixed = (int(generator.initial_seed()) ^ ((off + 1) * 0x9E3779B97F4A7C15)) & 0xFFFFFFFFFFFFFFFF
+            return _splitmix64(mixed)
+        return int(torch.randint(0, 2**62, (1,), generator=generator).item())
     _seed_fallback_counter[0] += 1
     return _splitmix64(_seed_fallback_counter[0])
 
@@ -115,22 +125,55 @@ def sample_gaussian(
     *,
     symmetric: bool = False,
     generator: Optional[torch.Generator] = None,
+    seed: Optional[int] = None,
+    elem_offset: int = 0,
 ) -> torch.Tensor:
     """Fill `out` (N×L) with x = mu + sigma * z. With symmetric=True, rows
     [0, N/2) hold mu + sigma*z and rows [N/2, N) the mirrored mu - sigma*z
-    (halves layout — see evotorch_amd/distributions.py docstring)."""
+    (halves layout — see evotorch_amd/distributions.py docstring).
+
+    Counter-addressed mode (`seed` given): noise element e of the virtual
+    full population is philox(seed, counter=(elem_offset+e)//4) — the SAME
+    values regardless of how the population is chunked into row-blocks, and
+    identical on CPU (numpy philox reference) and GPU. This is what the
+    streaming large-L gradient path uses to regenerate noise instead of
+    storing it; `elem_offset` must be a multiple of 4 and counts elements of
+    the first (non-mirrored) half."""
     if out.ndim != 2:
         raise ValueError(f"expected a 2-D population, got shape {tuple(out.shape)}")
+    n = out.shape[0]
+    if symmetric and n % 2 != 0:
+        raise ValueError("symmetric sampling requires even popsize")
+    if seed is not None:
+        if elem_offset % 4 != 0:
+            raise ValueError("elem_offset must be a multiple of 4")
+        if out.device.type == "cuda" and not _allow_eager_on_gpu():
+            mod = hip_required()
+            mod.sample_gaussian(out, mu.to(out.dtype), sigma.to(out.dtype), bool(symmetric), int(seed), int(elem_offset))
+            return out
+        # philox-exact eager reference (matches the kernel bit-for-bit in fp32)
+        from ..neuroevolution.philox_ref import philox_normals
+
+        rows = n // 2 if symmetric else n
+        length = out.shape[1]
+        z = philox_normals(int(seed), 0, rows * length, idx4_offset=elem_offset // 4)
+        z = z.reshape(rows, length).to(device=out.device)
+        mu32 = mu.to(torch.float32)
+        sigma32 = sigma.to(torch.float32)
+        plus = (mu32 + sigma32 * z).to(out.dtype)
+        if symmetric:
+            out[:rows] = plus
+            out[rows:] = (2.0 * mu32 - plus.to(torch.float32)).to(out.dtype)
+        else:
+            out.copy_(plus)
+        return out
     if out.device.type == "cuda" and not _allow_eager_on_gpu():
         mod = hip_required()
-        seed = _seed_from_generator(generator, out.device)
-        mod.sample_gaussian(out, mu.to(out.dtype), sigma.to(out.dtype), bool(symmetric), seed)
+        drawn = _seed_from_generator(generator, out.device)
+        mod.sample_gaussian(out, mu.to(out.dtype), sigma.to(out.dtype), bool(symmetric), drawn, 0)
         return out
     # eager reference
-    n = out.shape[0]
     if symmetric:
-        if n % 2 != 0:
-            raise ValueError("symmetric sampling requires even popsize")
         half = out[: n // 2]
         half.normal_(generator=generator)
         half.mul_(sigma).add_(mu)

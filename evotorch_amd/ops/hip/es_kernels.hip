@@ -54,13 +54,14 @@ template <typename T, bool kSymmetric>
 __global__ void sample_gaussian_kernel(T* __restrict__ out, const T* __restrict__ mu, const T* __restrict__ sigma,
                                        int64_t rows,  // = N (plain) or N/2 (symmetric)
                                        int64_t length, uint64_t seed_in,
-                                       const unsigned long long* __restrict__ seed_ptr) {
+                                       const unsigned long long* __restrict__ seed_ptr,
+                                       uint64_t elem_offset4) {
     const uint64_t seed = resolve_seed(seed_in, seed_ptr);
     const int64_t total4 = (rows * length + 3) / 4;
     for (int64_t idx4 = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; idx4 < total4;
          idx4 += (int64_t)gridDim.x * blockDim.x) {
         float z[4];
-        philox_normal4(seed, 0u, (uint64_t)idx4, z);
+        philox_normal4(seed, 0u, (uint64_t)idx4 + elem_offset4, z);
         const int64_t base = idx4 * 4;
 #pragma unroll
         for (int j = 0; j < 4; ++j) {
@@ -85,13 +86,14 @@ __global__ void sample_gaussian_kernel(T* __restrict__ out, const T* __restrict_
 template <bool kSymmetric>
 __global__ void sample_gaussian_f32x4_kernel(float4* __restrict__ out, const float4* __restrict__ mu,
                                              const float4* __restrict__ sigma, int64_t rows, int64_t length4,
-                                             uint64_t seed_in, const unsigned long long* __restrict__ seed_ptr) {
+                                             uint64_t seed_in, const unsigned long long* __restrict__ seed_ptr,
+                                             uint64_t elem_offset4) {
     const uint64_t seed = resolve_seed(seed_in, seed_ptr);
     const int64_t total4 = rows * length4;
     for (int64_t idx4 = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; idx4 < total4;
          idx4 += (int64_t)gridDim.x * blockDim.x) {
         float z[4];
-        philox_normal4(seed, 0u, (uint64_t)idx4, z);
+        philox_normal4(seed, 0u, (uint64_t)idx4 + elem_offset4, z);
         const int64_t col4 = idx4 % length4;
         const float4 m = mu[col4];
         const float4 s = sigma[col4];
@@ -113,7 +115,9 @@ __global__ void sample_gaussian_f32x4_kernel(float4* __restrict__ out, const flo
 }
 
 void sample_gaussian_impl(torch::Tensor out, torch::Tensor mu, torch::Tensor sigma, bool symmetric, int64_t seed,
-                          const unsigned long long* seed_ptr) {
+                          const unsigned long long* seed_ptr, int64_t elem_offset = 0) {
+    TORCH_CHECK(elem_offset % 4 == 0, "elem_offset must be a multiple of 4 (philox counter granularity)");
+    const uint64_t elem_offset4 = (uint64_t)(elem_offset / 4);
     CHECK_GPU(out); CHECK_CONTIG(out); CHECK_GPU(mu); CHECK_GPU(sigma);
     const int64_t n = out.size(0), length = out.size(1);
     TORCH_CHECK(!symmetric || n % 2 == 0, "symmetric sampling needs even popsize");
@@ -126,11 +130,11 @@ void sample_gaussian_impl(torch::Tensor out, torch::Tensor mu, torch::Tensor sig
         if (symmetric) {
             hipLaunchKernelGGL((sample_gaussian_f32x4_kernel<true>), dim3(blocks), dim3(threads), 0, stream,
                                reinterpret_cast<float4*>(out.data_ptr<float>()), reinterpret_cast<const float4*>(mu.data_ptr<float>()),
-                               reinterpret_cast<const float4*>(sigma.data_ptr<float>()), rows, length / 4, (uint64_t)seed, seed_ptr);
+                               reinterpret_cast<const float4*>(sigma.data_ptr<float>()), rows, length / 4, (uint64_t)seed, seed_ptr, elem_offset4);
         } else {
             hipLaunchKernelGGL((sample_gaussian_f32x4_kernel<false>), dim3(blocks), dim3(threads), 0, stream,
                                reinterpret_cast<float4*>(out.data_ptr<float>()), reinterpret_cast<const float4*>(mu.data_ptr<float>()),
-                               reinterpret_cast<const float4*>(sigma.data_ptr<float>()), rows, length / 4, (uint64_t)seed, seed_ptr);
+                               reinterpret_cast<const float4*>(sigma.data_ptr<float>()), rows, length / 4, (uint64_t)seed, seed_ptr, elem_offset4);
         }
         return;
     }
@@ -138,16 +142,17 @@ void sample_gaussian_impl(torch::Tensor out, torch::Tensor mu, torch::Tensor sig
         using T = scalar_t;
         if (symmetric) {
             hipLaunchKernelGGL((sample_gaussian_kernel<T, true>), dim3(blocks), dim3(threads), 0, stream,
-                               out.data_ptr<T>(), mu.data_ptr<T>(), sigma.data_ptr<T>(), rows, length, (uint64_t)seed, seed_ptr);
+                               out.data_ptr<T>(), mu.data_ptr<T>(), sigma.data_ptr<T>(), rows, length, (uint64_t)seed, seed_ptr, elem_offset4);
         } else {
             hipLaunchKernelGGL((sample_gaussian_kernel<T, false>), dim3(blocks), dim3(threads), 0, stream,
-                               out.data_ptr<T>(), mu.data_ptr<T>(), sigma.data_ptr<T>(), rows, length, (uint64_t)seed, seed_ptr);
+                               out.data_ptr<T>(), mu.data_ptr<T>(), sigma.data_ptr<T>(), rows, length, (uint64_t)seed, seed_ptr, elem_offset4);
         }
     });
 }
 
-void sample_gaussian(torch::Tensor out, torch::Tensor mu, torch::Tensor sigma, bool symmetric, int64_t seed) {
-    sample_gaussian_impl(out, mu, sigma, symmetric, seed, nullptr);
+void sample_gaussian(torch::Tensor out, torch::Tensor mu, torch::Tensor sigma, bool symmetric, int64_t seed,
+                     int64_t elem_offset) {
+    sample_gaussian_impl(out, mu, sigma, symmetric, seed, nullptr, elem_offset);
 }
 
 // Graph-safe variant: the seed lives in `seed_buf` (int64 tensor of 1
