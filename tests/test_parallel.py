@@ -137,6 +137,31 @@ def _body_obs_norm_allreduce(comm, rank, world):
     return prob.obs_norm.count
 
 
+def _body_streamed_pgpe(comm, rank, world):
+    """Sharded + streamed gradients: ranks stay in lockstep and descend."""
+    from evotorch_amd import Problem
+    from evotorch_amd.algorithms import PGPE
+    from evotorch_amd.decorators import vectorized
+
+    @vectorized
+    def sphere(x):
+        return (x**2).sum(-1)
+
+    prob = Problem("min", sphere, solution_length=8, initial_bounds=(-1, 1), seed=70 + rank)
+    prob.use_comm(comm)
+    searcher = PGPE(prob, popsize=40, center_learning_rate=0.2, stdev_learning_rate=0.1,
+                    stdev_init=1.0, distributed=True, center_init=torch.ones(8),
+                    grad_chunk_rows=4)
+    for _ in range(10):
+        searcher.step()
+    center_t = torch.Tensor.as_subclass(searcher.status["center"], torch.Tensor).clone()
+    ref = center_t.clone()
+    comm.broadcast_(ref, src=0)
+    assert torch.allclose(center_t, ref, atol=1e-6), "ranks diverged"
+    assert searcher.status["mean_eval"] < 8.0
+    return center_t.tolist()
+
+
 def _make_dist(prob):
     from evotorch_amd.distributions import SymmetricSeparableGaussian
 
@@ -149,14 +174,14 @@ def _make_dist(prob):
 
 @pytest.mark.parametrize(
     "body",
-    ["_body_collectives", "_body_sharded_evaluate", "_body_distributed_pgpe", "_body_obs_norm_allreduce"],
+    ["_body_collectives", "_body_sharded_evaluate", "_body_distributed_pgpe", "_body_obs_norm_allreduce", "_body_streamed_pgpe"],
 )
 def test_world2(body):
     results = _run_world(body, world=2)
     assert len(results) == 2
     if body == "_body_sharded_evaluate":
         assert results[0] == results[1]  # both ranks hold the full eval vector
-    if body == "_body_distributed_pgpe":
+    if body in ("_body_distributed_pgpe", "_body_streamed_pgpe"):
         assert results[0] == results[1]
 
 
