@@ -37,13 +37,14 @@ if _verbose is not None and _verbose >= 0:
     _logger.setLevel(_verbose)
 
 from . import utils
-from .core import Problem, ProblemBoundEvaluator, Solution, SolutionBatch
+from .core import Problem, ObjectTypedProblemBoundEvaluator, ProblemBoundEvaluator, Solution, SolutionBatch
 
 from . import algorithms, decorators, distributions, logging, models, neuroevolution, operators, ops, optimizers, parallel, testing, tools  # noqa: E402
 
 __all__ = [
     "__version__",
     "Problem",
+    "ObjectTypedProblemBoundEvaluator",
     "ProblemBoundEvaluator",
     "Solution",
     "SolutionBatch",

@@ -2,7 +2,7 @@
 /root/reference/src/evotorch/algorithms/__init__.py."""
 
 from .cmaes import CMAES
-from .ga import Cosyne, GeneticAlgorithm, SteadyStateGA
+from .ga import ExtendedPopulationMixin, Cosyne, GeneticAlgorithm, SteadyStateGA
 from .gaussian import CEM, PGPE, SNES, XNES, GaussianSearchAlgorithm
 from .graphed import GraphedSearch
 from .mapelites import MAPElites

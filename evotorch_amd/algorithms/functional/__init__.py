@@ -14,6 +14,8 @@ runs B independent searches — no object state, vmap-free batching.
 from .funccem import CEMState, cem, cem_ask, cem_tell
 from .funcoptimizers import (
     AdamState,
+    OptimizerFunctions,
+    get_functional_optimizer,
     ClipUpState,
     SGDState,
     adam,

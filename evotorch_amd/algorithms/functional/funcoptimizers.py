@@ -2,11 +2,11 @@
 triples (reference funcadam.py:34, funcclipup.py:31, funcsgd.py:30).
 All state fields are tensors, batched over leading dimensions."""
 
-from typing import NamedTuple, Optional, Union
+from typing import Callable, NamedTuple, Optional, Union
 
 import torch
 
-__all__ = ["AdamState", "adam", "adam_ask", "adam_tell", "ClipUpState", "clipup", "clipup_ask", "clipup_tell", "SGDState", "sgd", "sgd_ask", "sgd_tell"]
+__all__ = ["OptimizerFunctions", "get_functional_optimizer", "AdamState", "adam", "adam_ask", "adam_tell", "ClipUpState", "clipup", "clipup_ask", "clipup_tell", "SGDState", "sgd", "sgd_ask", "sgd_tell"]
 
 
 def _t(x, like: torch.Tensor) -> torch.Tensor:
@@ -123,12 +123,21 @@ _OPTIMIZERS = {
 }
 
 
+class OptimizerFunctions(NamedTuple):
+    """(initialize, ask, tell) triple returned by get_functional_optimizer
+    (reference algorithms/functional/misc.py:26)."""
+
+    initialize: Callable
+    ask: Callable
+    tell: Callable
+
+
 def get_functional_optimizer(name: Union[str, tuple]):
     """Resolve 'adam'/'clipup'/'sgd' (or a custom (init, ask, tell)
     triple) — reference functional/misc.py:26."""
     if isinstance(name, tuple):
-        return name
+        return OptimizerFunctions(*name)
     try:
-        return _OPTIMIZERS[str(name).lower()]
+        return OptimizerFunctions(*_OPTIMIZERS[str(name).lower()])
     except KeyError:
         raise ValueError(f"Unknown functional optimizer {name!r}") from None

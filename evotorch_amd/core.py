@@ -1287,6 +1287,8 @@ class Problem(TensorMakerMixin, Serializable):
     # -- functional adapter ---------------------------------------------------------
 
     def make_callable_evaluator(self, *, obj_index: Optional[int] = None) -> "ProblemBoundEvaluator":
+        if is_dtype_object(self._dtype):
+            return ObjectTypedProblemBoundEvaluator(self, obj_index=obj_index)
         return ProblemBoundEvaluator(self, obj_index=obj_index)
 
     # -- cloning / pickling -----------------------------------------------------------
@@ -1339,3 +1341,31 @@ class ProblemBoundEvaluator:
             out = self(flat)
             return out.reshape(lead + (values.shape[-2],))
         raise ValueError(f"Expected values of ndim >= 2, got shape {tuple(values.shape)}")
+
+class ObjectTypedProblemBoundEvaluator:
+    """Callable evaluator for object-dtype problems (reference
+    core.py:5201): values arrive as an ObjectArray (or any sequence) and
+    evals come back as a 1-D tensor; no extra batch dimensions (object
+    populations cannot be vmapped)."""
+
+    def __init__(self, problem: "Problem", *, obj_index: Optional[int] = None):
+        if not is_dtype_object(problem.dtype):
+            raise TypeError(
+                f"Expected an object-dtype problem, got dtype {problem.dtype}."
+                " Hint: use ProblemBoundEvaluator (make_callable_evaluator) for tensor-dtype problems."
+            )
+        self._problem = problem
+        self._obj_index = 0 if obj_index is None else int(obj_index)
+
+    @property
+    def problem(self) -> "Problem":
+        return self._problem
+
+    def __call__(self, values) -> torch.Tensor:
+        problem = self._problem
+        n = len(values)
+        batch = SolutionBatch(problem, popsize=n, empty=True)
+        for i in range(n):
+            batch.access_values()[i] = values[i]
+        problem.evaluate(batch)
+        return batch._evals[:, self._obj_index].clone()

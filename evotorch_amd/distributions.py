@@ -19,7 +19,7 @@ MI355X:
 """
 
 import math
-from typing import Iterable, Optional, Type
+from typing import Iterable, Optional, Type, NamedTuple
 
 import torch
 
@@ -27,6 +27,9 @@ from .utils import Device, DType, TensorMakerMixin, to_torch_dtype
 from .utils.ranking import rank
 
 __all__ = [
+    "GradsWithSamples",
+    "GradsWithFitnesses",
+    "GradsWithSamplesAndFitnesses",
     "Distribution",
     "SeparableGaussian",
     "SymmetricSeparableGaussian",
@@ -473,6 +476,25 @@ class ExpGaussian(Distribution):
 # ----------------------------------------------------------------------------
 # Functional bridge (reference distributions.py:1023-1622)
 # ----------------------------------------------------------------------------
+
+
+class GradsWithSamples(NamedTuple):
+    """Return type option of functional grad estimators (reference
+    distributions.py: GradsWithSamples)."""
+
+    grads: tuple
+    samples: torch.Tensor
+
+
+class GradsWithFitnesses(NamedTuple):
+    grads: tuple
+    fitnesses: torch.Tensor
+
+
+class GradsWithSamplesAndFitnesses(NamedTuple):
+    grads: tuple
+    samples: torch.Tensor
+    fitnesses: torch.Tensor
 
 
 def make_functional_sampler(distribution_class: Type[Distribution], *, required_parameters: Iterable[str], fixed_parameters: Optional[dict] = None):

@@ -19,7 +19,9 @@ from torch import nn
 from . import layers as _layers
 from .multilayered import MultiLayered
 
-__all__ = ["str_to_net", "NetParsingError"]
+__all__ = [
+    "submodules",
+    "concat_modules","str_to_net", "NetParsingError"]
 
 
 class NetParsingError(Exception):
@@ -101,3 +103,27 @@ def str_to_net(s: str, **constants) -> nn.Module:
     if not isinstance(result, nn.Module):
         raise NetParsingError(f"Expression did not produce an nn.Module but {type(result)}")
     return result
+
+
+def submodules(module: nn.Module) -> list:
+    """Flat list of a MultiLayered/Sequential's children, or [module]
+    (reference net/parser.py: submodules)."""
+    from .multilayered import MultiLayered
+
+    if isinstance(module, (MultiLayered, nn.Sequential)):
+        out = []
+        for child in module.children():
+            out.extend(submodules(child))
+        return out
+    return [module]
+
+
+def concat_modules(*modules: nn.Module) -> nn.Module:
+    """Compose modules left-to-right into one MultiLayered pipeline — the
+    programmatic form of the DSL's `>>` (reference net/parser.py)."""
+    from .multilayered import MultiLayered
+
+    flat = []
+    for m in modules:
+        flat.extend(submodules(m))
+    return MultiLayered(*flat)

@@ -1,7 +1,7 @@
 """Neuroevolution problem domains (L3). Reference parity:
 /root/reference/src/evotorch/neuroevolution/__init__.py."""
 
-from .neproblem import NEProblem
+from .neproblem import BaseNEProblem, NEProblem
 from .runningnorm import ObsNormLayer, RunningNorm, RunningStat
 from .supervisedne import SupervisedNE
 from .synthetic import SyntheticRolloutProblem
@@ -10,6 +10,7 @@ from .vecenv import GymVectorEnvAdapter, SyntheticTorchEnv, VecEnvNE, VecGymNE
 
 __all__ = [
     "GymVectorEnvAdapter",
+    "BaseNEProblem",
     "NEProblem",
     "ObsNormLayer",
     "RunningNorm",

@@ -20,7 +20,16 @@ from ..models import ensure_stateful
 from .neproblem import NEProblem
 from .runningnorm import RunningNorm
 
-__all__ = ["GymNE", "ActClipLayer"]
+__all__ = [
+    "ActClipLayer",
+    "ActClipWrapperModule",
+    "AliveBonusScheduleWrapper",
+    "GymNE",
+    "ObsNormWrapperModule",
+    "ensure_space_types",
+    "reset_env",
+    "take_step_in_env",
+]
 
 
 class ActClipLayer(nn.Module):
@@ -34,6 +43,80 @@ class ActClipLayer(nn.Module):
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         return torch.clamp(x, self.lb, self.ub)
+
+
+# reference net/rl.py exposes these under different names; aliases keep
+# imports from the reference working unchanged
+from .runningnorm import ObsNormLayer as ObsNormWrapperModule  # noqa: E402
+
+ActClipWrapperModule = ActClipLayer
+
+
+def ensure_space_types(env) -> None:
+    """Require Box observation and action spaces (reference
+    gymne.py: ensure_space_types)."""
+    import gymnasium as gym
+
+    if not isinstance(env.observation_space, gym.spaces.Box):
+        raise TypeError(f"Unsupported observation space {env.observation_space!r}: expected gymnasium.spaces.Box")
+    if not isinstance(env.action_space, gym.spaces.Box):
+        raise TypeError(f"Unsupported action space {env.action_space!r}: expected gymnasium.spaces.Box")
+
+
+def reset_env(env):
+    """Version-tolerant env reset: returns the observation alone whether the
+    env follows the old (obs) or new (obs, info) API (reference
+    net/rl.py:63)."""
+    result = env.reset()
+    if isinstance(result, tuple) and len(result) == 2:
+        return result[0]
+    return result
+
+
+def take_step_in_env(env, action):
+    """Version-tolerant env step: normalizes 4-tuple (obs, reward, done,
+    info) and 5-tuple (obs, reward, terminated, truncated, info) step APIs
+    to (obs, reward, done, info) (reference net/rl.py:83)."""
+    result = env.step(action)
+    if len(result) == 5:
+        obs, reward, terminated, truncated, info = result
+        return obs, reward, bool(terminated) or bool(truncated), info
+    return result
+
+
+class AliveBonusScheduleWrapper:
+    """Env wrapper subtracting the constant alive bonus early in training
+    and restoring it on a schedule (t0, t1[, bonus]) — by step t0 nothing
+    is given back, between t0 and t1 the bonus ramps linearly, after t1 the
+    full bonus is added (reference net/rl.py:199)."""
+
+    def __init__(self, env, alive_bonus_schedule):
+        self.env = env
+        if len(alive_bonus_schedule) == 3:
+            self._t0, self._t1, self._bonus = alive_bonus_schedule
+        else:
+            self._t0, self._t1 = alive_bonus_schedule
+            self._bonus = 1.0
+        self._t = 0
+
+    def __getattr__(self, name):
+        return getattr(self.env, name)
+
+    def reset(self, *args, **kwargs):
+        self._t = 0
+        return self.env.reset(*args, **kwargs)
+
+    def step(self, action):
+        result = self.env.step(action)
+        self._t += 1
+        if self._t < self._t0:
+            scale = 0.0
+        elif self._t >= self._t1:
+            scale = 1.0
+        else:
+            scale = (self._t - self._t0) / (self._t1 - self._t0)
+        reward = result[1] + scale * self._bonus
+        return (result[0], reward) + tuple(result[2:])
 
 
 class GymNE(NEProblem):

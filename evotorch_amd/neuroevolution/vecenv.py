@@ -27,7 +27,21 @@ from .neproblem import NEProblem
 from .runningnorm import RunningNorm
 from .synthetic_env import SyntheticEnvSpec
 
-__all__ = ["SyntheticTorchEnv", "GymVectorEnvAdapter", "VecEnvNE", "VecGymNE"]
+__all__ = [
+    "GymVectorEnvAdapter",
+    "SyntheticTorchEnv",
+    "TorchWrapper",
+    "VecEnvNE",
+    "VecGymNE",
+    "array_type",
+    "convert_from_torch",
+    "convert_to_torch",
+    "convert_to_torch_bool",
+    "is_brax_env",
+    "make_brax_env",
+    "make_gym_env",
+    "make_vector_env",
+]
 
 
 class SyntheticTorchEnv:
@@ -265,3 +279,106 @@ class VecEnvNE(NEProblem):
 
 
 VecGymNE = VecEnvNE  # reference-compatible alias (vecgymne.py:95)
+
+
+# -- env-infra helpers (reference net/vecrl.py) ------------------------------
+# brax/jax are not installed in the MI355X image (no jax wheel for ROCm
+# here); the brax entry points stay importable and raise with a clear
+# message, the numpy<->torch plumbing is fully functional.
+
+
+def array_type(x) -> str:
+    """'torch' | 'numpy' | 'scalar' | 'unknown' (reference vecrl.py)."""
+    import numpy as _np
+
+    if isinstance(x, torch.Tensor):
+        return "torch"
+    if isinstance(x, _np.ndarray):
+        return "numpy"
+    if isinstance(x, (int, float, bool, _np.number)):
+        return "scalar"
+    return "unknown"
+
+
+def convert_to_torch(x, *, device=None) -> torch.Tensor:
+    """numpy/scalar/tensor -> torch tensor (zero-copy where possible)."""
+    import numpy as _np
+
+    if isinstance(x, torch.Tensor):
+        t = x
+    elif isinstance(x, _np.ndarray):
+        t = torch.from_numpy(_np.ascontiguousarray(x))
+    else:
+        t = torch.as_tensor(x)
+    return t.to(device) if device is not None else t
+
+
+def convert_to_torch_bool(x, *, device=None) -> torch.Tensor:
+    return convert_to_torch(x, device=device).to(torch.bool)
+
+
+def convert_from_torch(x: torch.Tensor):
+    """torch tensor -> numpy array (cpu copy only if needed)."""
+    return x.detach().cpu().numpy()
+
+
+def is_brax_env(env) -> bool:
+    """True if env comes from brax (not installed in this image — always
+    False unless the user provides brax themselves)."""
+    mod = type(env).__module__ or ""
+    return mod.startswith("brax")
+
+
+def make_brax_env(env_name: str, **kwargs):
+    raise ImportError(
+        "brax requires jax, which has no ROCm wheel in this image."
+        " Use SyntheticTorchEnv / a torch-native batched env (the VecEnvNE"
+        " contract: reset()->obs, step(act)->(obs, reward, done, info)),"
+        " or GymVectorEnvAdapter over gymnasium."
+    )
+
+
+def make_gym_env(env_name: str, **kwargs):
+    """Instantiate a single gymnasium env by name (reference
+    vecrl.py:668)."""
+    import gymnasium as gym
+
+    return gym.make(env_name, **kwargs)
+
+
+def make_vector_env(env_name: str, *, num_envs: int, **kwargs):
+    """Batched env by name: a gymnasium vector env wrapped into the torch
+    contract used by VecEnvNE (reference vecrl.py:764)."""
+    import gymnasium as gym
+
+    vec = gym.make_vec(env_name, num_envs=num_envs, **kwargs)
+    return GymVectorEnvAdapter(vec)
+
+
+class TorchWrapper:
+    """Wraps a classic (numpy-API) env so that reset/step speak torch
+    tensors (reference vecrl.py:362); also normalizes old/new gym step
+    APIs via the gymne helpers."""
+
+    def __init__(self, env, *, device=None):
+        from .gymne import reset_env, take_step_in_env
+
+        self.env = env
+        self._device = device
+        self._reset_env = reset_env
+        self._step_env = take_step_in_env
+
+    def __getattr__(self, name):
+        return getattr(self.env, name)
+
+    def reset(self):
+        return convert_to_torch(self._reset_env(self.env), device=self._device)
+
+    def step(self, action: torch.Tensor):
+        obs, reward, done, info = self._step_env(self.env, convert_from_torch(action))
+        return (
+            convert_to_torch(obs, device=self._device),
+            convert_to_torch(reward, device=self._device),
+            convert_to_torch_bool(done, device=self._device),
+            info,
+        )

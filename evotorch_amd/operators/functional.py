@@ -11,13 +11,17 @@ population arguments run independent batched populations via
 reference), which is how batched searches compose.
 """
 
-from typing import Optional, Union
+from typing import Optional, Union, NamedTuple
 
 import torch
 
 from ..utils import ranking as _ranking
 
 __all__ = [
+    "SelectedParentIndices",
+    "SelectedParentValues",
+    "SelectedParents",
+    "SelectedAndStackedParents",
     "tournament",
     "multi_point_cross_over",
     "one_point_cross_over",
@@ -58,6 +62,31 @@ def _utils_2d(evals: torch.Tensor, objective_sense) -> torch.Tensor:
 # ----------------------------------------------------------------------------
 # selection
 # ----------------------------------------------------------------------------
+
+
+class SelectedParentIndices(NamedTuple):
+    """Pair-selection return types mirroring the reference's tournament API
+    (reference operators/functional.py:557-577)."""
+
+    parent1_indices: torch.Tensor
+    parent2_indices: torch.Tensor
+
+
+class SelectedParentValues(NamedTuple):
+    parent1_values: torch.Tensor
+    parent2_values: torch.Tensor
+
+
+class SelectedParents(NamedTuple):
+    parent1_values: torch.Tensor
+    parent1_evals: torch.Tensor
+    parent2_values: torch.Tensor
+    parent2_evals: torch.Tensor
+
+
+class SelectedAndStackedParents(NamedTuple):
+    parent_values: torch.Tensor
+    parent_evals: torch.Tensor
 
 
 def tournament(

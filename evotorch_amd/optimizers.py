@@ -15,7 +15,7 @@ import torch
 from . import ops
 from .utils import Device, DType, to_torch_dtype
 
-__all__ = ["Adam", "SGD", "ClipUp", "get_optimizer_class"]
+__all__ = ["Adam", "SGD", "ClipUp", "ClipUpParameterGroup", "TorchOptimizer", "get_optimizer_class"]
 
 
 class _Optimizer:
@@ -160,6 +160,54 @@ class ClipUp(_Optimizer):
             momentum=self._momentum,
         )
         return self._velocity.clone() if cloned_result else self._velocity
+
+
+class TorchOptimizer:
+    """Adapter exposing any `torch.optim` optimizer through the framework's
+    `ascent(gradient)` interface (reference optimizers.py:31-100): the
+    gradient-ASCENT direction is handed to a torch optimizer that performs
+    descent on an internal parameter copy; the returned step is the
+    parameter delta.
+
+    Example:
+        opt = TorchOptimizer(torch.optim.RMSprop, solution_length=L,
+                             stepsize=0.01, config={"alpha": 0.95})
+        new_center = center + opt.ascent(grad)
+    """
+
+    def __init__(
+        self,
+        torch_optimizer,
+        *,
+        solution_length: int,
+        stepsize: float,
+        dtype: DType = torch.float32,
+        device: Device = "cpu",
+        config: Optional[dict] = None,
+    ):
+        self._length = int(solution_length)
+        self._dtype = to_torch_dtype(dtype)
+        self._device = torch.device(device)
+        self._params = torch.zeros(self._length, dtype=self._dtype, device=self._device, requires_grad=False)
+        cfg = dict(config or {})
+        cfg["lr"] = float(stepsize)
+        self._opt = torch_optimizer([self._params], **cfg)
+        self.param_groups = self._opt.param_groups
+
+    @property
+    def contained_optimizer(self):
+        return self._opt
+
+    def ascent(self, globalg: torch.Tensor, *, cloned_result: bool = True) -> torch.Tensor:
+        before = self._params.clone()
+        self._params.grad = -globalg.to(dtype=self._dtype, device=self._device).reshape(-1)
+        self._opt.step()
+        self._opt.zero_grad(set_to_none=True)
+        step = self._params - before
+        return step.clone() if cloned_result else step
+
+
+ClipUpParameterGroup = _ParamGroup  # reference exposes the ClipUp param-group view type
 
 
 def get_optimizer_class(s: str, optimizer_config: Optional[dict] = None) -> Union[Type, callable]:

@@ -8,7 +8,8 @@ from typing import Iterable, Optional, Union
 import numpy as np
 import torch
 
-__all__ = ["TestingError", "assert_allclose", "assert_almost_between", "assert_dtype_matches", "assert_shape_matches"]
+__all__ = [
+    "assert_eachclose","TestingError", "assert_allclose", "assert_almost_between", "assert_dtype_matches", "assert_shape_matches"]
 
 
 class TestingError(AssertionError):
@@ -68,3 +69,19 @@ def assert_shape_matches(x, shape):
     for w, g in zip(want, got):
         if w is not None and w != g:
             raise TestingError(f"shape mismatch: expected {want}, got {got}")
+
+
+def assert_eachclose(x: Iterable, value, *, rtol: Optional[float] = None, atol: Optional[float] = None):
+    """Assert that every element of `x` is close to the scalar `value`
+    (reference testing.py: assert_eachclose)."""
+    arr = np.asarray([float(v) for v in np.asarray(x).reshape(-1)])
+    target = np.full_like(arr, float(value))
+    kwargs = {}
+    if rtol is not None:
+        kwargs["rtol"] = rtol
+    if atol is not None:
+        kwargs["atol"] = atol
+    if rtol is None and atol is None:
+        kwargs["atol"] = 1e-8
+    if not np.allclose(arr, target, **kwargs):
+        raise TestingError(f"Elements {arr} are not all close to {value}")

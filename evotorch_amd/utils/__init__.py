@@ -52,6 +52,13 @@ from .misc import (
     modify_tensor,
     modify_vector,
     numpy_copy,
+    pass_info_if_needed,
+    as_tensor,
+    is_tensor_on_cpu,
+    multiply_rows_by_scalars,
+    rowwise_sum,
+    message_from,
+    set_default_logger_config,
     make_batched_false_for_vmap,
     split_workload,
     stdev_from_radius,
@@ -61,9 +68,9 @@ from .misc import (
     to_torch_dtype,
 )
 from .objectarray import ObjectArray, as_object_array
-from .structures import CBag, CDict, CList, CMemory, Structure
+from .structures import CBag, CDict, CList, CMemory, Structure, do_where
 from .tensorframe import TensorFrame
-from .ranking import rank, ranking_method_exists
+from .ranking import centered, linear, nes, normalized, rank, raw
 from .readonlytensor import ReadOnlyTensor, as_read_only_tensor, read_only_tensor
 from .recursiveprintable import RecursivePrintable
 from .tensormaker import TensorMakerMixin

@@ -12,7 +12,9 @@ import torch
 
 from .readonlytensor import ReadOnlyTensor, as_read_only_tensor
 
-__all__ = ["as_immutable", "mutable_copy", "ImmutableContainer", "ImmutableList", "ImmutableSet", "ImmutableDict"]
+__all__ = [
+    "ImmutableSequence",
+    "is_immutable_container_or_tensor","as_immutable", "mutable_copy", "ImmutableContainer", "ImmutableList", "ImmutableSet", "ImmutableDict"]
 
 
 class ImmutableContainer:
@@ -78,6 +80,17 @@ class ImmutableDict(ImmutableContainer, Mapping):
 
     def __repr__(self):
         return f"ImmutableDict({self._data!r})"
+
+
+ImmutableSequence = ImmutableList  # reference alias (immutable.py:195)
+
+
+def is_immutable_container_or_tensor(x) -> bool:
+    """True for ImmutableList/Set/Dict and read-only tensors (reference
+    tools/immutable.py)."""
+    from .readonlytensor import ReadOnlyTensor
+
+    return isinstance(x, (ImmutableContainer, ReadOnlyTensor))
 
 
 def _eq(a, b) -> bool:

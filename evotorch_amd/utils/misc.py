@@ -57,6 +57,12 @@ __all__ = [
     "ensure_tensor_length_and_dtype",
     "expect_none",
     "pass_info_if_needed",
+    "set_default_logger_config",
+    "message_from",
+    "rowwise_sum",
+    "multiply_rows_by_scalars",
+    "is_tensor_on_cpu",
+    "as_tensor",
     "make_tensor",
     "make_empty",
     "make_zeros",
@@ -691,3 +697,68 @@ def to_stdev_init(*, solution_length: int, stdev_init=None, radius_init=None):
     if stdev_init is not None:
         return stdev_init
     return stdev_from_radius(float(radius_init), solution_length)
+
+
+def as_tensor(x, *, dtype: Optional[DType] = None, device: Optional[Device] = None):
+    """torch.as_tensor with the framework's dtype coercion (accepts numpy
+    dtypes, strings, 'object' → ObjectArray passthrough); reference
+    tools/misc.py: as_tensor."""
+    if dtype is not None and is_dtype_object(dtype):
+        from .objectarray import as_object_array
+
+        return as_object_array(x)
+    kwargs = {}
+    if dtype is not None:
+        kwargs["dtype"] = to_torch_dtype(dtype)
+    if device is not None:
+        kwargs["device"] = torch.device(device)
+    return torch.as_tensor(x, **kwargs)
+
+
+def is_tensor_on_cpu(x) -> bool:
+    """True if x is a torch tensor residing on the CPU."""
+    return isinstance(x, torch.Tensor) and x.device.type == "cpu"
+
+
+def multiply_rows_by_scalars(scalars: torch.Tensor, x: torch.Tensor) -> torch.Tensor:
+    """row i of the result = scalars[i] * x[i] (reference tools/misc.py)."""
+    return scalars.unsqueeze(-1) * x
+
+
+def rowwise_sum(x: torch.Tensor) -> torch.Tensor:
+    """Sum over the last dimension, row by row."""
+    return x.sum(dim=-1)
+
+
+def message_from(sender, message: str) -> str:
+    """Prefix a log/exception message with its originating object — used in
+    verbose-mode diagnostics (reference tools/misc.py: message_from)."""
+    name = sender if isinstance(sender, str) else type(sender).__name__
+    return f"[{name}] {message}"
+
+
+def set_default_logger_config(
+    logger_name: str = "evotorch_amd",
+    logger_level=None,
+    override: bool = False,
+):
+    """Install the default stderr handler + level on the framework logger
+    (honors EVOTORCH_AMD_VERBOSE_LEVEL; reference tools/misc.py:2072)."""
+    import logging as _logging
+    import os as _os
+    import sys as _sys
+
+    logger = _logging.getLogger(logger_name)
+    if logger.handlers and not override:
+        return logger
+    if override:
+        for h in list(logger.handlers):
+            logger.removeHandler(h)
+    if logger_level is None:
+        level_name = _os.environ.get("EVOTORCH_AMD_VERBOSE_LEVEL", "INFO").upper()
+        logger_level = getattr(_logging, level_name, _logging.INFO)
+    handler = _logging.StreamHandler(_sys.stderr)
+    handler.setFormatter(_logging.Formatter("[%(asctime)s] %(name)s %(levelname)s: %(message)s"))
+    logger.addHandler(handler)
+    logger.setLevel(logger_level)
+    return logger

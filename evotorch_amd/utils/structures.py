@@ -16,7 +16,7 @@ import torch
 
 from .misc import DType, Device, to_torch_dtype
 
-__all__ = ["CMemory", "Structure", "CDict", "CList", "CBag"]
+__all__ = ["CMemory", "Structure", "CDict", "CList", "CBag", "do_where"]
 
 Numbers = Union[int, float, Iterable, torch.Tensor]
 
