@@ -5,7 +5,7 @@ from .cmaes import CMAES
 from .ga import ExtendedPopulationMixin, Cosyne, GeneticAlgorithm, SteadyStateGA
 from .gaussian import CEM, PGPE, SNES, XNES, GaussianSearchAlgorithm
 from .graphed import GraphedSearch
-from .mapelites import MAPElites
+from .mapelites import make_feature_grid, MAPElites
 from .restarter import IPOP, ModifyingRestart, Restart
 from .searchalgorithm import LazyReporter, LazyStatusDict, SearchAlgorithm, SinglePopulationAlgorithmMixin
 

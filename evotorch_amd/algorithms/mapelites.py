@@ -16,7 +16,7 @@ from ..core import Problem, SolutionBatch
 from .ga import ExtendedPopulationMixin
 from .searchalgorithm import SearchAlgorithm, SinglePopulationAlgorithmMixin
 
-__all__ = ["MAPElites"]
+__all__ = ["MAPElites", "make_feature_grid"]
 
 
 class MAPElites(SearchAlgorithm, SinglePopulationAlgorithmMixin, ExtendedPopulationMixin):
@@ -124,3 +124,7 @@ class MAPElites(SearchAlgorithm, SinglePopulationAlgorithmMixin, ExtendedPopulat
         bad = float("-inf") if sense == "max" else float("inf")
         evals = self._population.access_evals()
         evals[~self._filled, 0] = bad
+
+
+# module-level alias of the staticmethod, mirroring the reference export
+make_feature_grid = MAPElites.make_feature_grid

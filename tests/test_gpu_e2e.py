@@ -192,3 +192,81 @@ def test_state_dict_resume_on_gpu():
     assert float(b.problem.obs_norm.count) == float(a.problem.obs_norm.count)
     b.run(2)
     assert b.step_count == 5
+
+
+@requires_gpu
+def test_misc_searchers_on_gpu():
+    """One compact pass over the searcher families not covered by the
+    dedicated GPU tests: XNES, MAPElites, NSGA-II GA, IPOP restarts, and
+    the functional API — all on cuda:0."""
+    _misc_searchers_body("cuda:0")
+
+
+def _misc_searchers_body(dev):
+    from evotorch_amd import Problem
+    from evotorch_amd.algorithms import IPOP, SNES, XNES, GeneticAlgorithm, MAPElites
+    from evotorch_amd.algorithms.functional import pgpe, pgpe_ask, pgpe_tell
+    from evotorch_amd.decorators import vectorized
+    from evotorch_amd.operators import GaussianMutation, PolynomialMutation, SimulatedBinaryCrossOver
+
+    @vectorized
+    def sphere(x):
+        return (x**2).sum(-1)
+
+    # XNES (full covariance, matrix_exp on rocBLAS/rocSOLVER)
+    prob = Problem("min", sphere, solution_length=16, initial_bounds=(-3, 3), device=dev, seed=1)
+    xnes = XNES(prob, stdev_init=2.0)
+    xnes.run(20)
+    assert float(xnes.status["mean_eval"]) < 16 * 9
+
+    # MAPElites
+    @vectorized
+    def with_features(x):
+        fitness = -(x**2).sum(-1)
+        return fitness, x[:, :2]
+
+    prob = Problem("max", with_features, solution_length=8, initial_bounds=(-2, 2),
+                   device=dev, seed=2, eval_data_length=2)
+    grid = MAPElites.make_feature_grid([-2.0, -2.0], [2.0, 2.0], [6, 6], device=dev)
+    me = MAPElites(prob, operators=[GaussianMutation(prob, stdev=0.3)], feature_grid=grid)
+    me.run(10)
+    assert int(me.filled.sum()) > 10
+
+    # NSGA-II end to end on device (pareto HIP kernels inside argsort on cuda)
+    @vectorized
+    def two_obj(x):
+        f1 = x[:, 0]
+        g = 1 + 9 * x[:, 1:].mean(dim=-1)
+        return torch.stack([f1, g * (1 - torch.sqrt((f1 / g).clamp(min=0)))], dim=-1)
+
+    prob = Problem(["min", "min"], two_obj, solution_length=8, bounds=(0.0, 1.0),
+                   initial_bounds=(0.0, 1.0), device=dev, seed=3)
+    ga = GeneticAlgorithm(prob, popsize=256, operators=[
+        SimulatedBinaryCrossOver(prob, eta=15, tournament_size=2),
+        PolynomialMutation(prob, eta=20, mutation_probability=0.125)])
+    ga.run(60)
+    ranks, _ = ga.population.compute_pareto_ranks(crowdsort=False)
+    assert int((ranks == 0).sum()) > 128
+
+    # IPOP restart wrapper around SNES
+    prob = Problem("min", sphere, solution_length=8, initial_bounds=(-3, 3), device=dev, seed=4)
+    ipop = IPOP(prob, SNES, algorithm_args={"stdev_init": 1.0, "popsize": 20},
+                max_inner_steps=5)
+    ipop.run(12)
+    assert ipop.step_count == 12
+    assert ipop.num_restarts >= 2
+
+    # functional API on device
+    state = pgpe(center_init=torch.ones(8, device=dev) * 2, radius_init=1.0,
+                 center_learning_rate=0.3, stdev_learning_rate=0.1,
+                 optimizer="clipup", optimizer_config={"max_speed": 0.3},
+                 ranking_method="centered", objective_sense="min")
+    for _ in range(30):
+        pop = pgpe_ask(state, popsize=64)
+        fit = sphere(pop)
+        state = pgpe_tell(state, pop, fit)
+    from evotorch_amd.algorithms.functional import get_functional_optimizer
+
+    _, opt_ask, _ = get_functional_optimizer(state.optimizer)
+    center = opt_ask(state.optimizer_state)
+    assert float((center**2).sum()) < 8 * 4
