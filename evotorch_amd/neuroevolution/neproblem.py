@@ -123,7 +123,8 @@ class NEProblem(Problem):
     def _evaluate(self, solution: Solution):
         net = self.parameterize_net(torch.Tensor.as_subclass(solution.values, torch.Tensor))
         fn = self._network_eval_func
-        result = self._evaluate_network(net) if fn is None else fn(net)
+        with torch.no_grad():  # fitness evaluation never needs autograd
+            result = self._evaluate_network(net) if fn is None else fn(net)
         self._write_solution_result(solution, result)
 
     # -- pickling: do not ship the live network -------------------------------

@@ -222,6 +222,7 @@ class VecEnvNE(NEProblem):
 
     # -- the hot loop ---------------------------------------
 
+    @torch.no_grad()  # rollouts never need autograd (user nets may carry requires_grad params)
     def _evaluate_batch(self, batch: SolutionBatch):
         n = len(batch)
         env = self._get_env(n)
