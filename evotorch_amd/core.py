@@ -1109,6 +1109,8 @@ class Problem(TensorMakerMixin, Serializable):
         comm = self._comm
         self._before_grad_hook()
         if chunk_rows is not None:
+            if num_interactions is not None:
+                raise ValueError("chunk_rows (streaming gradients) does not support num_interactions adaptive popsize")
             result = self._sample_and_compute_gradients_streamed(distribution, int(popsize), obj_index, ranking_method, int(chunk_rows), comm)
         elif comm is not None and comm.world_size > 1:
             result = self._sample_and_compute_gradients_sharded(distribution, int(popsize), obj_index, ranking_method, comm, num_interactions, popsize_max, ensure_even_popsize)
