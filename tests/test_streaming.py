@@ -100,3 +100,35 @@ def test_odd_length_chunk_alignment():
     r2 = make_problem(length=7, seed=13).sample_and_compute_gradients(sym_dist(7), 24, ranking_method="centered", chunk_rows=12)
     for k in ("mu", "sigma"):
         assert torch.allclose(r1["gradients"][k], r2["gradients"][k], atol=1e-5)
+
+
+import hypothesis.strategies as st
+from hypothesis import given, settings
+
+
+@settings(max_examples=30, deadline=None)
+@given(
+    length=st.integers(min_value=2, max_value=23),
+    directions=st.integers(min_value=2, max_value=20),
+    chunk=st.integers(min_value=1, max_value=24),
+    ranking=st.sampled_from(["centered", "nes", "raw", "linear"]),
+    divide=st.sampled_from(["num_directions", "num_solutions", "total_weight", None]),
+    seed=st.integers(min_value=0, max_value=10_000),
+)
+def test_streamed_gradient_fuzz(length, directions, chunk, ranking, divide, seed):
+    """Property: for ANY (length, popsize, chunking, ranking, divisor), the
+    streamed gradient equals the single-chunk gradient of the same seed."""
+    popsize = directions * 2
+    params = {"mu": torch.zeros(length), "sigma": torch.ones(length)}
+    if divide is not None:
+        params["divide_mu_grad_by"] = divide
+        params["divide_sigma_grad_by"] = divide
+    prob_a = make_problem(length=length, seed=seed)
+    r_a = prob_a.sample_and_compute_gradients(
+        SymmetricSeparableGaussian(dict(params)), popsize, ranking_method=ranking, chunk_rows=chunk)
+    prob_b = make_problem(length=length, seed=seed)
+    r_b = prob_b.sample_and_compute_gradients(
+        SymmetricSeparableGaussian(dict(params)), popsize, ranking_method=ranking, chunk_rows=directions)
+    for k in ("mu", "sigma"):
+        a, b = r_a["gradients"][k], r_b["gradients"][k]
+        assert torch.allclose(a, b, atol=1e-4, rtol=1e-4), (k, (a - b).abs().max(), length, directions, chunk, ranking, divide)
