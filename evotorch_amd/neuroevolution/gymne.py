@@ -254,4 +254,6 @@ class GymNE(NEProblem):
         layers.append(module)
         if clip_actions and hasattr(self._act_space, "low"):
             layers.append(ActClipLayer(self._act_space.low, self._act_space.high))
-        return nn.Sequential(*layers)
+        policy = nn.Sequential(*layers)
+        policy.requires_grad_(False)  # inference artifact
+        return policy

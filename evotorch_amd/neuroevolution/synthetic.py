@@ -174,4 +174,6 @@ class SyntheticRolloutProblem(Problem):
                 linear.weight.copy_(x[: A * O].reshape(A, O))
                 linear.bias.copy_(x[A * O :])
                 layers += [linear, torch.nn.Hardtanh()]
-        return torch.nn.Sequential(*layers)
+        policy = torch.nn.Sequential(*layers)
+        policy.requires_grad_(False)  # inference artifact
+        return policy

@@ -275,7 +275,8 @@ class VecEnvNE(NEProblem):
         module = self._policy.to_torch_module(torch.as_tensor(x, dtype=torch.float32).cpu())
         if self._obs_norm_enabled and self._obs_norm.has_data:
             norm_cpu = self._obs_norm.to("cpu")
-            return nn.Sequential(norm_cpu.to_layer(), module)
+            module = nn.Sequential(norm_cpu.to_layer(), module)
+        module.requires_grad_(False)  # inference artifact
         return module
 
 
