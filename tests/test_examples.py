@@ -17,6 +17,7 @@ CASES = [
     ("examples/synthetic_humanoid_pgpe.py", ["--generations", "2", "--popsize", "16", "--device", "cpu"]),
     ("examples/mnist30k_distributed.py", ["--generations", "1", "--popsize", "8"]),
     ("examples/genetic_programming.py", ["--generations", "5", "--popsize", "64"]),
+    ("examples/checkpoint_resume.py", []),
 ]
 
 
