@@ -409,3 +409,42 @@ def test_streaming_large_l_smoke():
     # (plus allocator churn); chunked path stays near parameter-vector cost
     assert peak < 2.2, f"peak {peak:.2f} GiB — streaming not effective"
     assert searcher.step_count == 2
+
+
+@requires_gpu
+def test_kernel_fuzz_random_shapes():
+    """Randomized-shape sweep of K1/K3 vs the eager fp32 reference:
+    odd lengths, odd popsizes, both dtypes, both layouts."""
+    import os
+
+    from evotorch_amd.ops import es_gradients, sample_gaussian, snes_gradients
+
+    g = torch.Generator().manual_seed(7)
+    for trial in range(12):
+        length = int(torch.randint(1, 700, (1,), generator=g))
+        rows = int(torch.randint(1, 97, (1,), generator=g))
+        symmetric = bool(torch.rand(1, generator=g) < 0.5)
+        dtype = torch.float32 if torch.rand(1, generator=g) < 0.7 else torch.bfloat16
+        n = rows * 2 if symmetric else rows
+        mu = torch.randn(length, device="cuda:0")
+        sigma = torch.rand(length, device="cuda:0") + 0.3
+
+        out = torch.empty(n, length, device="cuda:0", dtype=dtype)
+        seed = 1000 + trial
+        sample_gaussian(out, mu.to(dtype), sigma.to(dtype), symmetric=symmetric, seed=seed)
+        cpu = torch.empty(n, length, dtype=dtype)
+        sample_gaussian(cpu, mu.cpu().to(dtype), sigma.cpu().to(dtype), symmetric=symmetric, seed=seed)
+        tol = 1e-6 if dtype == torch.float32 else 0.0  # bf16 rounding is deterministic
+        assert torch.allclose(out.cpu().float(), cpu.float(), atol=max(tol, 1e-6) if dtype == torch.float32 else 2e-2), \
+            (trial, length, rows, symmetric, dtype)
+
+        if dtype == torch.float32:
+            weights = torch.randn(n, device="cuda:0")
+            gm, gs = es_gradients(out, mu, sigma, weights, symmetric=symmetric)
+            os.environ["EVOTORCH_AMD_ALLOW_EAGER_GPU"] = "1"
+            try:
+                gm_ref, gs_ref = es_gradients(out, mu, sigma, weights, symmetric=symmetric)
+            finally:
+                del os.environ["EVOTORCH_AMD_ALLOW_EAGER_GPU"]
+            assert torch.allclose(gm, gm_ref, atol=1e-3, rtol=1e-3), (trial, (gm - gm_ref).abs().max())
+            assert torch.allclose(gs, gs_ref, atol=1e-3, rtol=1e-3), (trial, (gs - gs_ref).abs().max())
