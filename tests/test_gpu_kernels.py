@@ -434,9 +434,12 @@ def test_kernel_fuzz_random_shapes():
         sample_gaussian(out, mu.to(dtype), sigma.to(dtype), symmetric=symmetric, seed=seed)
         cpu = torch.empty(n, length, dtype=dtype)
         sample_gaussian(cpu, mu.cpu().to(dtype), sigma.cpu().to(dtype), symmetric=symmetric, seed=seed)
-        tol = 1e-6 if dtype == torch.float32 else 0.0  # bf16 rounding is deterministic
-        assert torch.allclose(out.cpu().float(), cpu.float(), atol=max(tol, 1e-6) if dtype == torch.float32 else 2e-2), \
-            (trial, length, rows, symmetric, dtype)
+        # fp32: the kernel uses fmaf while the cpu reference multiplies then
+        # adds; near mu = -sigma*z cancellation the ABSOLUTE error is
+        # ~ulp(sigma*z) (~2e-6), so compare with an absolute bound.
+        atol = 1e-5 if dtype == torch.float32 else 2e-2
+        assert torch.allclose(out.cpu().float(), cpu.float(), atol=atol), \
+            (trial, length, rows, symmetric, dtype, (out.cpu().float() - cpu.float()).abs().max())
 
         if dtype == torch.float32:
             weights = torch.randn(n, device="cuda:0")
