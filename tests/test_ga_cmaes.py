@@ -284,3 +284,20 @@ def test_object_dtype_ga_with_cut_and_splice():
     assert ga.status["pop_best_eval"] <= first
     best = ga.population.take_best()
     assert isinstance(list(best.values), list)
+
+
+@pytest.mark.parametrize("kwargs", [
+    {"active": False},
+    {"csa_squared": True},
+    {"c_sigma_ratio": 0.5, "damp_sigma_ratio": 2.0},
+    {"stdev_min": 1e-8, "stdev_max": 5.0},
+    {"limit_C_decomposition": False},
+])
+def test_cmaes_variants_converge(kwargs):
+    """Every CMA-ES configuration knob still yields a working optimizer."""
+    prob = Problem("min", sphere, solution_length=8, initial_bounds=(-3, 3), seed=31)
+    searcher = CMAES(prob, stdev_init=2.0, **kwargs)
+    searcher.step()
+    first = float(searcher.status["mean_eval"])
+    searcher.run(60)
+    assert float(searcher.status["mean_eval"]) < first * 0.2, kwargs
