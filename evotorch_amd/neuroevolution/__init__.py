@@ -1,6 +1,7 @@
 """Neuroevolution problem domains (L3). Reference parity:
 /root/reference/src/evotorch/neuroevolution/__init__.py."""
 
+from .deploy import load_policy, save_policy
 from .neproblem import BaseNEProblem, NEProblem
 from .runningnorm import ObsNormLayer, RunningNorm, RunningStat
 from .supervisedne import SupervisedNE
@@ -9,6 +10,8 @@ from .synthetic_env import SyntheticEnvSpec, rollout_eager
 from .vecenv import GymVectorEnvAdapter, SyntheticTorchEnv, VecEnvNE, VecGymNE
 
 __all__ = [
+    "load_policy",
+    "save_policy",
     "GymVectorEnvAdapter",
     "BaseNEProblem",
     "NEProblem",

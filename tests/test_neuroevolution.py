@@ -328,3 +328,22 @@ def test_torch_wrapper_env():
     obs2, reward, done, info = env.step(torch.zeros(2))
     assert float(reward) == 0.25 and done.dtype == torch.bool and bool(done)
     assert convert_to_torch(5.0).ndim == 0
+
+
+def test_save_load_policy_safetensors(tmp_path):
+    from evotorch_amd.neuroevolution import SyntheticRolloutProblem, load_policy, save_policy
+
+    prob = SyntheticRolloutProblem(seed=5, episode_length=4)
+    x = torch.randn(prob.solution_length)
+    policy = prob.to_policy(x)
+    path = save_policy(policy, str(tmp_path / "policy"), metadata={"gens": 10})
+    assert path.endswith(".safetensors")
+
+    fresh = prob.to_policy(torch.zeros(prob.solution_length))
+    load_policy(path, fresh)
+    obs = torch.randn(376)
+    assert torch.allclose(policy(obs), fresh(obs))
+    import json
+
+    sidecar = json.load(open(path + ".json"))
+    assert sidecar["metadata"]["gens"] == 10
