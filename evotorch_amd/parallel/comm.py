@@ -47,7 +47,16 @@ class Comm:
                 backend = "nccl" if torch.cuda.is_available() else "gloo"
             os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
             os.environ.setdefault("MASTER_PORT", "29500")
-            dist.init_process_group(backend=backend, timeout=datetime.timedelta(seconds=timeout_s))
+            init_kwargs = {}
+            if backend == "nccl":
+                # pin the device BEFORE the process group exists: RCCL
+                # builds its communicator on the current device at first
+                # collective, and an explicit device_id avoids the implicit
+                # barrier-on-wrong-device failure mode
+                local_rank = int(os.environ.get("LOCAL_RANK", int(env_rank) % max(torch.cuda.device_count(), 1)))
+                torch.cuda.set_device(local_rank)
+                init_kwargs["device_id"] = torch.device("cuda", local_rank)
+            dist.init_process_group(backend=backend, timeout=datetime.timedelta(seconds=timeout_s), **init_kwargs)
             self._rank = dist.get_rank()
             self._world = dist.get_world_size()
         self._initialized_group = True
