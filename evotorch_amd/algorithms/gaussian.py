@@ -171,6 +171,7 @@ class GaussianSearchAlgorithm(SearchAlgorithm, SinglePopulationAlgorithmMixin):
         )
         self._update_distribution(fetched["gradients"])
         self._mean_eval = fetched["mean_eval"]
+        self.update_status({"num_solutions": int(fetched["num_solutions"])})
 
     def _fill_and_eval_pop(self):
         problem = self.problem

@@ -1147,12 +1147,30 @@ class Problem(TensorMakerMixin, Serializable):
             distribution.sample(out=batch.access_values(), generator=self._generator)
         with record_range("evaluate"):
             self.evaluate(batch)
+        values = batch._values
         fitnesses = batch._evals[:, obj_index]
+        total_popsize = popsize
+        if num_interactions is not None and hasattr(self, "last_eval_interaction_count"):
+            # adaptive popsize (reference core.py:3239-3274): keep sampling
+            # extra sub-batches until the interaction threshold is reached
+            interactions = int(getattr(self, "last_eval_interaction_count", 0) or 0)
+            extra_values, extra_fits = [values], [fitnesses]
+            while interactions < int(num_interactions) and not (popsize_max is not None and total_popsize >= int(popsize_max)):
+                more = SolutionBatch(self, popsize=popsize, device=self._device, empty=True)
+                distribution.sample(out=more.access_values(), generator=self._generator)
+                self.evaluate(more)
+                interactions += int(getattr(self, "last_eval_interaction_count", 0) or 0)
+                extra_values.append(more._values)
+                extra_fits.append(more._evals[:, obj_index])
+                total_popsize += popsize
+            if len(extra_values) > 1:
+                values = torch.cat(extra_values)
+                fitnesses = torch.cat(extra_fits)
         sense = self._senses[obj_index]
-        grads = distribution.compute_gradients(batch._values, fitnesses, objective_sense=sense, ranking_method=ranking_method)
+        grads = distribution.compute_gradients(values, fitnesses, objective_sense=sense, ranking_method=ranking_method)
         return {
             "gradients": grads,
-            "num_solutions": popsize,
+            "num_solutions": total_popsize,
             "mean_eval": torch.nanmean(fitnesses),  # 0-dim tensor: no host sync
         }
 

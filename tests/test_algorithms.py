@@ -223,3 +223,23 @@ def test_pgpe_with_torch_optimizer():
     first = searcher.status["mean_eval"]
     searcher.run(40)
     assert searcher.status["mean_eval"] < first * 0.5
+
+
+def test_num_interactions_adaptive_popsize_distributed():
+    """distributed=True honors num_interactions: extra sub-batches are drawn
+    until the interaction budget is met (reference core.py:3239-3274)."""
+
+    class CountingProblem(Problem):
+        def __init__(self):
+            super().__init__("min", solution_length=4, initial_bounds=(-1, 1), vectorized=True, objective_func=sphere)
+            self.last_eval_interaction_count = 0
+
+        def evaluate(self, batch):
+            self.last_eval_interaction_count = len(batch) * 3
+            super().evaluate(batch)
+
+    prob = CountingProblem()
+    searcher = PGPE(prob, popsize=10, center_learning_rate=0.1, stdev_learning_rate=0.1, stdev_init=1.0,
+                    num_interactions=100, popsize_max=100, distributed=True)
+    searcher.step()
+    assert searcher.status["num_solutions"] == 40  # 4 sub-batches of 10
