@@ -118,6 +118,11 @@ class GaussianSearchAlgorithm(SearchAlgorithm, SinglePopulationAlgorithmMixin):
             raise ValueError("grad_chunk_rows (streaming gradients) requires distributed=True")
         if not self._distributed and popsize_weighted_grad_avg is not None:
             raise ValueError("popsize_weighted_grad_avg is only meaningful with distributed=True")
+        # Accepted for reference compatibility. In the reference, actors may
+        # evaluate different popsizes, making weighted vs unweighted
+        # averaging distinct (gaussian.py:246-271); here every rank owns an
+        # equal popsize/world shard, so the two averages coincide and the
+        # sharded path's local/total scaling is already the weighted form.
         self._popsize_weighted_grad_avg = popsize_weighted_grad_avg
 
         self._population: Optional[SolutionBatch] = None
