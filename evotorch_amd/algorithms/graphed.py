@@ -161,3 +161,15 @@ class GraphedSearch:
     @property
     def mean_eval(self) -> float:
         return float(self._mean_eval_buf)
+
+    def state_dict(self) -> dict:
+        """Checkpoint of the wrapped searcher's current (graph-updated)
+        state — the in-place buffers ARE the distribution parameters, so
+        the inner searcher's state_dict sees the latest values."""
+        return self._searcher.state_dict()
+
+    def load_state_dict(self, state: dict):
+        self._searcher.load_state_dict(state)
+        if self._graph is not None:
+            raise RuntimeError("load_state_dict after capture() is not supported: the captured graph holds the old parameter buffers; re-create the GraphedSearch and capture again")
+        return self
