@@ -18,6 +18,7 @@ CASES = [
     ("examples/mnist30k_distributed.py", ["--generations", "1", "--popsize", "8"]),
     ("examples/genetic_programming.py", ["--generations", "5", "--popsize", "64"]),
     ("examples/checkpoint_resume.py", []),
+    ("examples/lennard_jones_cmaes.py", ["--generations", "30"]),
 ]
 
 
