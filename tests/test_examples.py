@@ -19,6 +19,7 @@ CASES = [
     ("examples/genetic_programming.py", ["--generations", "5", "--popsize", "64"]),
     ("examples/checkpoint_resume.py", []),
     ("examples/lennard_jones_cmaes.py", ["--generations", "30"]),
+    ("examples/mpc_cem_pendulum.py", ["--steps", "90"]),
 ]
 
 
