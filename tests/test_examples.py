@@ -20,6 +20,7 @@ CASES = [
     ("examples/checkpoint_resume.py", []),
     ("examples/lennard_jones_cmaes.py", ["--generations", "30"]),
     ("examples/mpc_cem_pendulum.py", ["--steps", "90"]),
+    ("examples/vqe_snes.py", ["--generations", "250"]),
 ]
 
 
