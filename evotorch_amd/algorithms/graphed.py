@@ -9,6 +9,10 @@ distribution update — into a hipGraph (`torch.cuda.CUDAGraph` maps onto
 hipGraph on ROCm) and replays it per generation: one graph launch instead
 of ~10 kernel launches and zero Python in the loop.
 
+(Green-field MI355X feature: the reference has no equivalent — its
+nearest analogue is the torch.compile/vmap usage in the functional API,
+/root/reference/src/evotorch/algorithms/functional/__init__.py:15-50.)
+
 Graph-safe RNG: sampling uses `sample_gaussian_graphsafe`, which reads its
 philox seed from a device buffer and advances it ON DEVICE (splitmix64
 bump kernel) — a by-value seed would be frozen into the capture and every

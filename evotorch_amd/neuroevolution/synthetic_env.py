@@ -11,6 +11,11 @@ low-rank neural dynamics:
     r   = wr·o' + alive_bonus − act_cost·‖a‖²/A
     o₀  = 0.1·N(0, I)  (philox-deterministic per global member index)
 
+(Green-field MI355X substitute for the reference's brax/gym env plumbing,
+/root/reference/src/evotorch/neuroevolution/net/vecrl.py:616-1664 — the
+rollout CONTRACT matches VecGymNE's batched env, the dynamics are
+synthetic.)
+
 The fused gfx950 kernel `evotorch_amd._C.rollout_linear`
 (evotorch_amd/ops/hip/rollout.hip) runs the entire T-step episode of every
 member with policy weights and env matrices resident in LDS (bf16 storage,
