@@ -123,3 +123,83 @@ def test_take_best_is_truly_best(n, take):
     all_vals = batch.unsafe_evals[:, 0]
     threshold = all_vals.sort().values[take - 1]
     assert bool((best_vals <= threshold + 1e-6).all())
+
+
+@settings(max_examples=20, deadline=None)
+@given(
+    n=st.integers(min_value=2, max_value=20).map(lambda v: v * 2),
+    length=st.integers(min_value=2, max_value=30),
+    seed=st.integers(min_value=0, max_value=9999),
+)
+def test_crossover_children_inherit_columns(n, length, seed):
+    """One/two-point crossover: every child element comes from one of its
+    two parents at the same column."""
+    from evotorch_amd.operators.functional import one_point_cross_over, two_point_cross_over
+
+    g = torch.Generator().manual_seed(seed)
+    parents = torch.randn(n, length, generator=g)
+    for fn in (one_point_cross_over, two_point_cross_over):
+        children = fn(parents, generator=g)
+        half = n // 2
+        p1, p2 = parents[:half], parents[half : 2 * half]
+        for row in range(children.shape[0]):
+            a = p1[row % half]
+            b = p2[row % half]
+            c = children[row]
+            from_parent = (c == a) | (c == b)
+            assert bool(from_parent.all()), (fn.__name__, row)
+
+
+@settings(max_examples=20, deadline=None)
+@given(
+    pairs=st.integers(min_value=1, max_value=20),
+    length=st.integers(min_value=1, max_value=30),
+    eta=st.floats(min_value=1.0, max_value=40.0),
+    seed=st.integers(min_value=0, max_value=9999),
+)
+def test_sbx_preserves_pair_means(pairs, length, eta, seed):
+    """Simulated binary crossover: each child pair's mean equals its parent
+    pair's mean (definitional SBX property)."""
+    from evotorch_amd.operators.functional import simulated_binary_cross_over
+
+    g = torch.Generator().manual_seed(seed)
+    parents = torch.randn(pairs * 2, length, generator=g)
+    children = simulated_binary_cross_over(parents, eta=eta, generator=g)
+    half = pairs
+    p_mean = (parents[:half] + parents[half : 2 * half]) / 2
+    c_mean = (children[:half] + children[half : 2 * half]) / 2
+    assert torch.allclose(p_mean, c_mean, atol=1e-5)
+
+
+@settings(max_examples=15, deadline=None)
+@given(
+    n=st.integers(min_value=2, max_value=30),
+    length=st.integers(min_value=2, max_value=20),
+    seed=st.integers(min_value=0, max_value=9999),
+)
+def test_cosyne_permutation_is_columnwise_permutation(n, length, seed):
+    from evotorch_amd.operators.functional import cosyne_permutation
+
+    g = torch.Generator().manual_seed(seed)
+    values = torch.randn(n, length, generator=g)
+    permuted = cosyne_permutation(values, permute_all=True, generator=g)
+    for col in range(length):
+        assert torch.allclose(values[:, col].sort().values, permuted[:, col].sort().values)
+
+
+@settings(max_examples=15, deadline=None)
+@given(seed=st.integers(min_value=0, max_value=9999))
+def test_operators_respect_bounds(seed):
+    """GaussianMutation/PolynomialMutation/SBX children never leave the
+    problem's bounds (Operator._respect_bounds, reference base.py:75)."""
+    from evotorch_amd.operators import GaussianMutation, PolynomialMutation, SimulatedBinaryCrossOver
+
+    prob = Problem("min", sphere, solution_length=6, bounds=(-1.0, 1.0),
+                   initial_bounds=(-1.0, 1.0), seed=seed)
+    batch = prob.generate_batch(20)
+    prob.evaluate(batch)
+    for op in (GaussianMutation(prob, stdev=5.0), PolynomialMutation(prob, eta=5.0),
+               SimulatedBinaryCrossOver(prob, eta=3.0, tournament_size=2)):
+        out = op(batch)
+        vals = out.unsafe_values if hasattr(out, "unsafe_values") else out
+        assert bool((vals >= -1.0 - 1e-6).all()) and bool((vals <= 1.0 + 1e-6).all()), type(op).__name__
