@@ -243,3 +243,21 @@ def test_num_interactions_adaptive_popsize_distributed():
                     num_interactions=100, popsize_max=100, distributed=True)
     searcher.step()
     assert searcher.status["num_solutions"] == 40  # 4 sub-batches of 10
+
+
+def test_xnes_rotation_invariance_on_ellipsoid():
+    """XNES (full covariance, exponential map) adapts to a rotated
+    ill-conditioned ellipsoid — separable SNES cannot."""
+    d = 8
+    cond = torch.logspace(0, 2.5, d)
+    q, _ = torch.linalg.qr(torch.randn(d, d, generator=torch.Generator().manual_seed(4)))
+
+    @vectorized
+    def rotated_ellipsoid(x):
+        y = x @ q.T
+        return (cond * y**2).sum(-1)
+
+    prob = Problem("min", rotated_ellipsoid, solution_length=d, initial_bounds=(-3, 3), seed=9)
+    searcher = XNES(prob, stdev_init=2.0, popsize=32)
+    searcher.run(500)
+    assert float(searcher.status["pop_best_eval"]) < 1e-2
