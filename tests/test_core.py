@@ -243,3 +243,22 @@ def test_crowding_distances_match_bruteforce():
         finite = torch.isfinite(crowd_ref)
         assert torch.equal(torch.isfinite(crowd), finite)
         assert torch.allclose(crowd[finite], crowd_ref[finite], atol=1e-5)
+
+
+def test_solution_batch_pieces_alias_parent():
+    """split() pieces are views: evaluating pieces evaluates the parent
+    (reference core.py:4603 SolutionBatchPieces)."""
+    prob = make_problem(seed=91)
+    batch = prob.generate_batch(10)
+    pieces = batch.split(3)
+    assert len(pieces) == 3
+    assert sum(len(p) for p in pieces) == 10
+    for piece in pieces:
+        prob.evaluate(piece)
+    assert batch.evals_are_ready
+    # max_size variant
+    pieces2 = batch.split(max_size=4)
+    assert all(len(p) <= 4 for p in pieces2)
+    assert sum(len(p) for p in pieces2) == 10
+    # indexing
+    assert len(pieces2[0]) == len(pieces2[0])
