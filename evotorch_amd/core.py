@@ -804,6 +804,44 @@ class Problem(TensorMakerMixin, Serializable):
         core.py:1763)."""
         return [s == "max" for s in self._senses]
 
+    def normalize_obj_index(self, obj_index: Optional[int] = None) -> int:
+        """Resolve an objective index: None is allowed only for
+        single-objective problems; negative indices wrap (reference
+        core.py:2370)."""
+        if obj_index is None:
+            if len(self._senses) > 1:
+                raise ValueError("obj_index must be given for a multi-objective problem")
+            return 0
+        obj_index = int(obj_index)
+        if obj_index < 0:
+            obj_index += len(self._senses)
+        if not (0 <= obj_index < len(self._senses)):
+            raise IndexError(f"obj_index out of range for {len(self._senses)} objectives")
+        return obj_index
+
+    def compare_solutions(self, a: "Solution", b: "Solution", obj_index: Optional[int] = None) -> float:
+        """Positive if `a` is better, negative if `b` is better, 0 on a tie
+        (both must be evaluated; reference core.py:2402)."""
+        j = self.normalize_obj_index(obj_index)
+        ea = float(a.evals[j])
+        eb = float(b.evals[j])
+        diff = ea - eb
+        return diff if self._senses[j] == "max" else -diff
+
+    def is_better(self, a: "Solution", b: "Solution", obj_index: Optional[int] = None) -> bool:
+        """True iff `a` strictly beats `b` (reference core.py:2432)."""
+        return self.compare_solutions(a, b, obj_index) > 0
+
+    def is_worse(self, a: "Solution", b: "Solution", obj_index: Optional[int] = None) -> bool:
+        """True iff `a` is strictly beaten by `b` (reference core.py:2448)."""
+        return self.compare_solutions(a, b, obj_index) < 0
+
+    @property
+    def is_on_cpu(self) -> bool:
+        """True if the problem's main device is the CPU (reference
+        core.py:1727)."""
+        return self._device.type == "cpu"
+
     def ensure_tensor_length_and_dtype(self, t, *, about: Optional[str] = None, allow_scalar: bool = False):
         """Coerce `t` into a solution-length vector of this problem's dtype
         and device (reference exposes this as a Problem method)."""

@@ -262,3 +262,17 @@ def test_solution_batch_pieces_alias_parent():
     assert sum(len(p) for p in pieces2) == 10
     # indexing
     assert len(pieces2[0]) == len(pieces2[0])
+
+
+def test_solution_comparison_methods():
+    prob = make_problem(seed=92)
+    batch = prob.generate_batch(4)
+    prob.evaluate(batch)
+    order = batch.argsort()
+    best, worst = batch[int(order[0])], batch[int(order[-1])]
+    assert prob.is_better(best, worst)
+    assert prob.is_worse(worst, best)
+    assert prob.compare_solutions(best, worst) > 0
+    assert prob.normalize_obj_index(None) == 0
+    assert prob.normalize_obj_index(-1) == 0
+    assert prob.is_on_cpu
