@@ -611,6 +611,38 @@ class Solution(Serializable):
     def set_values(self, values):
         self._batch.set_values(values, solutions=self._index)
 
+    def access_evals(self) -> torch.Tensor:
+        """Mutable view of this solution's evals row (reference
+        core.py:4858)."""
+        return self._batch.access_evals()[self._index]
+
+    @property
+    def shape(self) -> torch.Size:
+        return torch.Size([self._batch.solution_length]) if self._batch.solution_length is not None else torch.Size([1])
+
+    def size(self) -> torch.Size:
+        return self.shape
+
+    @property
+    def ndim(self) -> int:
+        return 1
+
+    def dim(self) -> int:
+        return self.ndim
+
+    @property
+    def eval_shape(self) -> torch.Size:
+        return torch.Size([self._batch._evals.shape[1]])
+
+    @property
+    def eval_dtype(self):
+        return self._batch.eval_dtype
+
+    def to(self, device) -> "Solution":
+        """Copy of this solution (as a 1-element batch view) on `device`
+        (reference core.py:5010)."""
+        return self.to_batch().to(device)[0]
+
     def set_evals(self, evals, eval_data=None):
         evals = torch.as_tensor(evals, dtype=self._batch._evals.dtype, device=self._batch._evals.device).reshape(-1)
         no = self._batch._num_objs
