@@ -43,6 +43,11 @@ class LazyReporter:
     def has_status_key(self, key: str) -> bool:
         return key in self._getters or key in self._computed
 
+    def is_status_computed(self, key: str) -> bool:
+        """True if the status item is already materialized (reading it will
+        not trigger its getter) — reference searchalgorithm.py:118."""
+        return key in self._computed
+
     def iter_status_keys(self):
         seen = set()
         for k in self._computed:
@@ -166,6 +171,13 @@ class SearchAlgorithm(LazyReporter):
 
     def reset_first_step_datetime(self):
         self._first_step_datetime = None
+
+    @property
+    def is_terminated(self) -> bool:
+        """Whether the searcher reached a terminal state (always False in
+        the base; Restart wrappers consult this) — reference
+        searchalgorithm.py:445."""
+        return False
 
     # -- checkpoint / resume -------------------------------------------------
     # (a green-field addition relative to the reference, which only offers

@@ -47,6 +47,42 @@ class ObjectArray:
     def is_read_only(self) -> bool:
         return self._read_only
 
+    def size(self):
+        """torch.Size-style shape tuple (reference objectarray.py:208)."""
+        import torch
+
+        return torch.Size([len(self)])
+
+    def dim(self) -> int:
+        return 1
+
+    def numel(self) -> int:
+        return len(self)
+
+    def set_item(self, i, x):
+        """Explicit setter (reference objectarray.py:344): same as
+        `self[i] = x` (stores an immutable clone)."""
+        self[i] = x
+
+    def repeat(self, count: int) -> "ObjectArray":
+        """Concatenate `count` copies of this array (reference
+        objectarray.py:244)."""
+        count = int(count)
+        out = ObjectArray(len(self) * count)
+        for rep in range(count):
+            for i in range(len(self)):
+                out[rep * len(self) + i] = self[i]
+        return out
+
+    @staticmethod
+    def from_numpy(arr) -> "ObjectArray":
+        """Build from a numpy object (or any) 1-D array (reference
+        objectarray.py:302)."""
+        out = ObjectArray(len(arr))
+        for i, item in enumerate(arr):
+            out[i] = item
+        return out
+
     @property
     def device(self) -> str:
         return "cpu"
