@@ -176,6 +176,64 @@ class TensorFrame(RecursivePrintable):
             return self.hstack(result, override=override)
         return result
 
+    def cpu(self) -> "TensorFrame":
+        return self.to("cpu")
+
+    def cuda(self, device=None) -> "TensorFrame":
+        return self.to(torch.device("cuda") if device is None else torch.device(device))
+
+    def as_tensor(self, x, *, to_work_with: Optional[str] = None, broadcast_if_scalar: bool = False) -> torch.Tensor:
+        """Coerce `x` to a tensor on this frame's device; with
+        `to_work_with` (a column name or tensor), match that column's dtype
+        and, if `broadcast_if_scalar`, expand a 0-dim result to the
+        column's leading length (reference tensorframe.py:304)."""
+        ref = None
+        if to_work_with is not None:
+            ref = self[to_work_with] if isinstance(to_work_with, str) else to_work_with
+        out = torch.as_tensor(x, dtype=(ref.dtype if ref is not None else None),
+                              device=(ref.device if ref is not None else self.device))
+        if broadcast_if_scalar and out.ndim == 0 and ref is not None:
+            out = out.expand(ref.shape[0])
+        return out
+
+    def join(self, other) -> "TensorFrame":
+        """pandas-style alias of hstack (reference tensorframe.py:1092)."""
+        if isinstance(other, (list, tuple)):
+            out = self
+            for item in other:
+                out = out.hstack(item)
+            return out
+        return self.hstack(other)
+
+    def drop(self, *, columns) -> "TensorFrame":
+        """Drop the named column(s) (reference tensorframe.py:1107)."""
+        names = [columns] if isinstance(columns, str) else list(columns)
+        return self.without_columns(*names)
+
+    def nlargest(self, n: int, columns) -> "TensorFrame":
+        """The n rows with the largest values in the given column
+        (reference tensorframe.py:1060)."""
+        by = columns if isinstance(columns, str) else list(columns)[0]
+        idx = self.argsort(by, descending=True)[: int(n)]
+        return self.pick[idx]
+
+    def nsmallest(self, n: int, columns) -> "TensorFrame":
+        by = columns if isinstance(columns, str) else list(columns)[0]
+        idx = self.argsort(by, descending=False)[: int(n)]
+        return self.pick[idx]
+
+    def with_enforced_device(self, device) -> "TensorFrame":
+        """Shallow copy whose columns are moved to (and future columns
+        coerced onto) `device` (reference tensorframe.py:432)."""
+        out = self.to(device)
+        out._enforced_device = torch.device(device)
+        return out
+
+    def without_enforced_device(self) -> "TensorFrame":
+        out = self.to(self.device)
+        out._enforced_device = None
+        return out
+
     def get_read_only_view(self) -> "TensorFrame":
         return TensorFrame(self._columns, read_only=True)
 
