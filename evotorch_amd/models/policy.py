@@ -55,6 +55,15 @@ class Policy:
         return self._params
 
     @property
+    def parameter_length(self) -> int:  # reference vecrl.py alias
+        return self.parameter_count
+
+    @property
+    def wrapped_module(self) -> nn.Module:
+        """The underlying torch module (reference vecrl.py: wrapped_module)."""
+        return self.net
+
+    @property
     def net(self) -> nn.Module:
         return self._net
 

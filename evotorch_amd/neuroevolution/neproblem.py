@@ -87,6 +87,12 @@ class NEProblem(Problem):
     def parameterized_net(self) -> Optional[nn.Module]:
         return self._instantiated_net
 
+    @property
+    def network_constants(self) -> dict:
+        """Constants fed to str_to_net (obs_length etc.) — reference
+        neproblem.py: network_constants."""
+        return dict(self._network_constants())
+
     def parameterize_net(self, parameters: torch.Tensor) -> nn.Module:
         """Load a flat parameter vector into the cached network
         (reference neproblem.py:342)."""

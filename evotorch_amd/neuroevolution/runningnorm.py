@@ -42,6 +42,24 @@ class RunningNorm:
         return self._device
 
     @property
+    def min_variance(self) -> float:
+        return self._min_variance
+
+    @property
+    def low(self):
+        """Lower clip bound applied after normalization (or None)."""
+        return None if self._clip is None else self._clip[0]
+
+    @property
+    def high(self):
+        return None if self._clip is None else self._clip[1]
+
+    @property
+    def stats(self):
+        """(count, sum, sum_of_squares) — reference runningnorm.py: stats."""
+        return self.stats_triple()
+
+    @property
     def count(self) -> float:
         return float(self._count)
 

@@ -90,6 +90,11 @@ class SupervisedNE(NEProblem):
 
     # -- loss ----------------------------------------------------------------
 
+    def loss(self, y_hat: torch.Tensor, y: torch.Tensor) -> torch.Tensor:
+        """Public loss entry point (reference supervisedne.py:286): uses the
+        provided loss_func or the overridable `_loss`."""
+        return self._loss(y_hat, y)
+
     def _loss(self, y_hat: torch.Tensor, y: torch.Tensor) -> torch.Tensor:
         if self._loss_func is not None:
             return self._loss_func(y_hat, y)

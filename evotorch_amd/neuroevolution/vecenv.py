@@ -269,6 +269,64 @@ class VecEnvNE(NEProblem):
 
     # -- policy export --------------------------------------------------------
 
+    # -- reference-parity accessors (vecgymne.py:590-716) ---------------------
+
+    @property
+    def episode_count(self) -> int:
+        return self._episode_count
+
+    def set_episode_count(self, n: int):
+        self._episode_count = int(n)
+
+    @property
+    def interaction_count(self) -> int:
+        return self._total_interactions
+
+    def set_interaction_count(self, n: int):
+        self._total_interactions = int(n)
+
+    @property
+    def observation_normalization(self) -> bool:
+        return self._obs_norm_enabled
+
+    @property
+    def max_num_envs(self) -> Optional[int]:
+        return getattr(self._env, "num_envs", None) if getattr(self, "_env", None) is not None else None
+
+    def get_env(self):
+        """The underlying batched env (created lazily on first evaluation)."""
+        return getattr(self, "_env", None)
+
+    def get_observation_stats(self) -> RunningNorm:
+        return self._obs_norm
+
+    def set_observation_stats(self, rn: RunningNorm):
+        self._obs_norm = rn.to(self.network_device)
+
+    def update_observation_stats(self, rn) -> RunningNorm:
+        """Merge another RunningNorm (or (count, sum, sumsq) triple) into
+        this problem's stats — the reference's actor sync protocol entry
+        point (vecgymne.py:658-716)."""
+        self._obs_norm.update(rn if not isinstance(rn, RunningNorm) else rn.stats_triple())
+        return self._obs_norm
+
+    def pop_observation_stats(self) -> Optional[RunningNorm]:
+        """Take (and clear) the stats collected since the last pop — what a
+        worker ships to the main process in the reference protocol."""
+        pending = self._pending_stats
+        self._pending_stats = None
+        return pending
+
+    def save_solution(self, solution, path: str):
+        """Persist a solution as a ready-to-run policy: safetensors weights
+        + JSON sidecar (see deploy.save_policy). `solution` may be a
+        Solution or a flat tensor."""
+        from .deploy import save_policy
+
+        x = solution.values if hasattr(solution, "values") else solution
+        policy = self.to_policy(torch.Tensor.as_subclass(torch.as_tensor(x), torch.Tensor))
+        return save_policy(policy, path)
+
     def to_policy(self, x: torch.Tensor) -> nn.Module:
         if self._policy is None:
             self._policy = self.make_functional_policy()
