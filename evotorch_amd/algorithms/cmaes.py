@@ -169,8 +169,9 @@ class CMAES(SearchAlgorithm, SinglePopulationAlgorithmMixin):
 
     # -- stepping -----------------------------------------------------------
 
-    def _sample(self):
-        n, lam = self._n, self._popsize
+    def _sample(self, num_samples: Optional[int] = None):
+        n = self._n
+        lam = self._popsize if num_samples is None else int(num_samples)
         problem = self.problem
         g = problem.generator
         z = torch.empty((lam, n), dtype=self._m.dtype, device=self._m.device)
@@ -181,29 +182,30 @@ class CMAES(SearchAlgorithm, SinglePopulationAlgorithmMixin):
             y = z @ self._A.T  # y ~ N(0, C) with C = A Aᵀ
         return z, y, self._m + self._sigma * y
 
-    def _step(self):
-        problem = self.problem
-        n, lam = self._n, self._popsize
-        z, y, x = self._sample()
-        batch = SolutionBatch(problem, popsize=lam, device=self._m.device, empty=True)
-        batch.access_values().copy_(x)
-        problem.evaluate(batch)
-        self._population = batch
-        order = batch.argsort(obj_index=self._obj_index)
-        z = z[order]
-        y = y[order]
+    def sample_distribution(self, num_samples: Optional[int] = None):
+        """Draw (z, y, x): standard normals, correlated steps y = z Aᵀ, and
+        solutions x = m + σ·y (reference cmaes.py: sample_distribution)."""
+        return self._sample() if num_samples is None else self._sample(num_samples)
 
-        w = self._weights
-        w_pos = w[: self._mu]
-        y_w = w_pos @ y[: self._mu]
-        z_w = w_pos @ z[: self._mu]
+    def get_population_weights(self, order: torch.Tensor) -> torch.Tensor:
+        """Recombination weights aligned to the fitness-sorted population
+        (reference cmaes.py: get_population_weights)."""
+        return self._weights
 
-        # mean update
+    def update_m(self, y_w: torch.Tensor):
+        """Mean update m += c_m σ y_w (reference cmaes.py: update_m)."""
         self._m = self._m + self._c_m * self._sigma * y_w
 
-        # step-size control (CSA)
+    def update_p_sigma(self, z_w: torch.Tensor):
+        """CSA path update (reference cmaes.py: update_p_sigma)."""
         cs = self._c_sigma
         self._p_sigma = (1.0 - cs) * self._p_sigma + math.sqrt(cs * (2.0 - cs) * self._mu_eff) * z_w
+
+    def update_sigma(self):
+        """CSA step-size update with clamping (reference cmaes.py:
+        update_sigma)."""
+        n = self._n
+        cs = self._c_sigma
         ps_norm = torch.linalg.vector_norm(self._p_sigma)
         if self._csa_squared:
             exponent = (cs / self._damp_sigma) * ((ps_norm**2 / n) - 1.0) / 2.0
@@ -214,15 +216,26 @@ class CMAES(SearchAlgorithm, SinglePopulationAlgorithmMixin):
             self._sigma = torch.clamp(self._sigma, min=self._stdev_min)
         if self._stdev_max is not None:
             self._sigma = torch.clamp(self._sigma, max=self._stdev_max)
+        return ps_norm
 
-        # covariance update (hs stays a device tensor: no per-gen host sync)
-        cc = self._c_c
+    def update_p_c(self, y_w: torch.Tensor, ps_norm: torch.Tensor) -> torch.Tensor:
+        """Rank-1 path update; returns the stall indicator hs as a device
+        tensor (no host sync) — reference cmaes.py: update_p_c."""
+        n = self._n
+        cc, cs = self._c_c, self._c_sigma
         hs_threshold = (1.4 + 2.0 / (n + 1.0)) * self._chi_n * math.sqrt(1.0 - (1.0 - cs) ** (2 * (self._steps_count + 1)))
         hs_f = (ps_norm < hs_threshold).to(self._p_c.dtype)
         self._p_c = (1.0 - cc) * self._p_c + hs_f * math.sqrt(cc * (2.0 - cc) * self._mu_eff) * y_w
+        return hs_f
+
+    def update_C(self, z: torch.Tensor, y: torch.Tensor, hs_f: torch.Tensor):
+        """Rank-1 + rank-μ covariance update (active CMA negative-weight
+        scaling included) — reference cmaes.py: update_C."""
+        n = self._n
+        cc = self._c_c
         c1, cmu = self._c_1, self._c_mu
         delta_hs = (1.0 - hs_f) * cc * (2.0 - cc)
-        # active CMA: negative weights are scaled by n/||z_i||^2
+        w = self._weights
         w_adj = w.clone()
         if self._active:
             neg_mask = w < 0
@@ -232,17 +245,47 @@ class CMAES(SearchAlgorithm, SinglePopulationAlgorithmMixin):
             rank_mu = (w_adj.unsqueeze(-1) * y**2).sum(dim=0)
             self._C = (1.0 + c1 * delta_hs - c1 - cmu * w_adj.sum()) * self._C + c1 * self._p_c**2 + cmu * rank_mu
             self._C = torch.clamp(self._C, min=1e-20)
-            self._A = torch.sqrt(self._C)
         else:
             # rank-μ as a GEMM: Yᵀ diag(w) Y  (rocBLAS; MFMA-shaped)
             rank_mu = (y * w_adj.unsqueeze(-1)).T @ y
             rank_one = torch.outer(self._p_c, self._p_c)
             self._C = (1.0 + c1 * delta_hs - c1 - cmu * w_adj.sum()) * self._C + c1 * rank_one + cmu * rank_mu
             self._C = 0.5 * (self._C + self._C.T)
-            self._steps_since_decompose += 1
-            if self._steps_since_decompose >= self._decompose_interval:
-                self._A = self._cholesky(self._C)
-                self._steps_since_decompose = 0
+
+    def decompose_C(self):
+        """Refresh the Cholesky factor A (amortized over
+        `_decompose_interval` generations; separable mode is a plain sqrt)
+        — reference cmaes.py: decompose_C."""
+        if self._separable:
+            self._A = torch.sqrt(self._C)
+            return
+        self._steps_since_decompose += 1
+        if self._steps_since_decompose >= self._decompose_interval:
+            self._A = self._cholesky(self._C)
+            self._steps_since_decompose = 0
+
+    def _step(self):
+        problem = self.problem
+        lam = self._popsize
+        z, y, x = self.sample_distribution()
+        batch = SolutionBatch(problem, popsize=lam, device=self._m.device, empty=True)
+        batch.access_values().copy_(x)
+        problem.evaluate(batch)
+        self._population = batch
+        order = batch.argsort(obj_index=self._obj_index)
+        z = z[order]
+        y = y[order]
+
+        w_pos = self._weights[: self._mu]
+        y_w = w_pos @ y[: self._mu]
+        z_w = w_pos @ z[: self._mu]
+
+        self.update_m(y_w)
+        self.update_p_sigma(z_w)
+        ps_norm = self.update_sigma()
+        hs_f = self.update_p_c(y_w, ps_norm)
+        self.update_C(z, y, hs_f)
+        self.decompose_C()
 
     def _state_items(self) -> dict:
         return {
