@@ -115,7 +115,8 @@ def main():
 
     if rank == 0:
         result = {
-            "metric": "solutions/sec, PGPE Humanoid-v4 linear policy",
+            "metric": ("solutions/sec, PGPE Humanoid-v4 linear policy" if args.policy == "linear"
+                       else "solutions/sec, PGPE Humanoid-v4 MLP-64 policy"),
             "value": solutions_per_sec,
             "unit": "solutions/sec",
             "n_gpus": world if have_gpu else 0,
