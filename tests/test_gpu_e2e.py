@@ -270,3 +270,22 @@ def _misc_searchers_body(dev):
     _, opt_ask, _ = get_functional_optimizer(state.optimizer)
     center = opt_ask(state.optimizer_state)
     assert float((center**2).sum()) < 8 * 4
+
+
+@requires_gpu
+def test_streaming_with_rollout_problem():
+    """Streaming gradients compose with the fused-rollout problem: chunked
+    evaluation feeds obs-norm per chunk and the searcher still learns."""
+    from evotorch_amd.algorithms import PGPE
+    from evotorch_amd.neuroevolution import SyntheticRolloutProblem
+
+    prob = SyntheticRolloutProblem(device="cuda:0", seed=17, episode_length=50)
+    searcher = PGPE(prob, popsize=256, radius_init=2.25, center_learning_rate=0.1,
+                    stdev_learning_rate=0.1, optimizer="clipup",
+                    distributed=True, grad_chunk_rows=32)
+    searcher.step()
+    first = float(searcher.status["mean_eval"])
+    searcher.run(30)
+    last = float(searcher.status["mean_eval"])
+    assert prob.obs_norm.count > 0
+    assert last > first, (first, last)
