@@ -152,3 +152,15 @@ def test_tensorframe_to_pandas():
     tf = TensorFrame({"a": torch.arange(3.0)})
     df = tf.to_pandas()
     assert list(df["a"]) == [0.0, 1.0, 2.0]
+
+
+def test_tensorframe_pandas_style_methods():
+    tf = TensorFrame({"a": torch.tensor([3.0, 1.0, 2.0]), "b": torch.tensor([30.0, 10.0, 20.0])})
+    assert tf.nlargest(2, "a")["a"].tolist() == [3.0, 2.0]
+    assert tf.nsmallest(1, "b")["b"].tolist() == [10.0]
+    assert tf.drop(columns="b").columns == ["a"]
+    joined = tf.drop(columns="b").join(TensorFrame({"c": torch.ones(3)}))
+    assert set(joined.columns) == {"a", "c"}
+    x = tf.as_tensor(5, to_work_with="a", broadcast_if_scalar=True)
+    assert x.shape == (3,) and x.dtype == torch.float32
+    assert tf.cpu().device == torch.device("cpu")

@@ -200,3 +200,20 @@ def test_make_batched_false_for_vmap():
 
     out = vmap(f)(torch.tensor([-1.0, 2.0, -3.0]))
     assert torch.equal(out, torch.tensor([1.0, 2.0, 3.0]))
+
+
+def test_objectarray_method_parity():
+    import numpy as np
+
+    arr = ObjectArray(2)
+    arr[0] = [1, 2]
+    arr[1] = "x"
+    assert arr.numel() == 2
+    assert arr.dim() == 1
+    assert tuple(arr.size()) == (2,)
+    rep = arr.repeat(2)
+    assert len(rep) == 4 and list(rep[2]) == [1, 2]
+    built = ObjectArray.from_numpy(np.array([[5], [6, 7]], dtype=object))
+    assert list(built[1]) == [6, 7]
+    arr.set_item(0, [9])
+    assert list(arr[0]) == [9]
