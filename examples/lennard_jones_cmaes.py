@@ -44,6 +44,7 @@ def main():
     problem = Problem(
         "min", make_potential(args.atoms), solution_length=args.atoms * 3,
         initial_bounds=(-1.0, 1.0), device=args.device, seed=0,
+        store_solution_stats=True,  # keep best/best_eval on GPU devices too
     )
     searcher = CMAES(problem, stdev_init=0.3, popsize=64)
     searcher.run(args.generations)

@@ -31,6 +31,7 @@ def main():
     args = p.parse_args()
 
     problem = Problem("min", rastrigin, solution_length=100, initial_bounds=(-5.12, 5.12),
+                      store_solution_stats=True,
                       device=args.device, seed=1)
     searcher = SNES(problem, popsize=1000, stdev_init=10.0)
     StdOutLogger(searcher, interval=max(1, args.generations // 10))

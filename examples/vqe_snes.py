@@ -94,7 +94,8 @@ def main():
                 state = apply_cz_ladder(state, n)
         return torch.einsum("bi,ij,bj->b", state.conj(), H.to(state.dtype), state).real
 
-    problem = Problem("min", energy, solution_length=layers * n, initial_bounds=(-0.1, 0.1), seed=1)
+    problem = Problem("min", energy, solution_length=layers * n, initial_bounds=(-0.1, 0.1), seed=1,
+                      store_solution_stats=True)
     searcher = SNES(problem, popsize=40, stdev_init=0.5)
     searcher.run(args.generations)
     found = float(searcher.status["best_eval"])
