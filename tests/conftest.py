@@ -19,3 +19,17 @@ def pytest_sessionstart(session):
 @pytest.fixture
 def device():
     return "cuda" if torch.cuda.is_available() else "cpu"
+
+
+try:  # deterministic hypothesis examples: CI runs must not flake
+    from hypothesis import HealthCheck, settings
+
+    settings.register_profile(
+        "ci",
+        derandomize=True,
+        deadline=None,
+        suppress_health_check=[HealthCheck.too_slow],
+    )
+    settings.load_profile("ci")
+except ImportError:
+    pass
