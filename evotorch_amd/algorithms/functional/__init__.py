@@ -29,6 +29,7 @@ from .funcoptimizers import (
     sgd_tell,
 )
 from .funcpgpe import PGPEState, pgpe, pgpe_ask, pgpe_tell
+from .funcsnes import SNESState, snes, snes_ask, snes_tell
 
 __all__ = [
     "AdamState",

@@ -159,3 +159,24 @@ def test_functional_cosyne_permutation():
     out = F.cosyne_permutation(vals, generator=g)
     for col in range(5):
         assert torch.allclose(out[:, col].sort().values, vals[:, col].sort().values)
+
+
+def test_functional_snes_converges_single_and_batched():
+    from evotorch_amd.algorithms.functional import snes, snes_ask, snes_tell
+
+    def sphere(x):
+        return (x**2).sum(-1)
+
+    g = torch.Generator().manual_seed(0)
+    state = snes(center_init=torch.ones(12) * 3, stdev_init=2.0, objective_sense="min")
+    for _ in range(250):
+        pop = snes_ask(state, popsize=30, generator=g)
+        state = snes_tell(state, pop, sphere(pop))
+    assert float((state.center**2).sum()) < 1e-2
+
+    # stacked states = independent batched searches
+    state = snes(center_init=torch.ones(3, 12) * 3, stdev_init=2.0, objective_sense="min")
+    for _ in range(250):
+        pop = snes_ask(state, popsize=30, generator=g)
+        state = snes_tell(state, pop, sphere(pop))
+    assert float((state.center**2).sum(-1).max()) < 1e-1
