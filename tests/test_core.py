@@ -276,3 +276,26 @@ def test_solution_comparison_methods():
     assert prob.normalize_obj_index(None) == 0
     assert prob.normalize_obj_index(-1) == 0
     assert prob.is_on_cpu
+
+
+def test_batch_utility_and_utils():
+    """SolutionBatch.utility (single-objective) and .utils (per-objective
+    stack) — reference core.py:4207."""
+    prob = make_problem(seed=93)
+    batch = prob.generate_batch(8)
+    prob.evaluate(batch)
+    u = batch.utility(ranking_method="centered")
+    assert u.shape == (8,)
+    # best solution gets the highest utility
+    best = int(batch.argbest())
+    assert int(u.argmax()) == best
+
+    @vectorized
+    def two(x):
+        return torch.stack([x.sum(-1), -x.sum(-1)], dim=-1)
+
+    mo = Problem(["min", "max"], two, solution_length=3, initial_bounds=(-1, 1), seed=94)
+    b2 = mo.generate_batch(6)
+    mo.evaluate(b2)
+    us = b2.utils(ranking_method="centered")
+    assert us.shape == (6, 2)
