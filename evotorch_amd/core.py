@@ -753,6 +753,10 @@ class Problem(TensorMakerMixin, Serializable):
         else:
             self._eval_dtype = self._dtype if (self._dtype != object and to_torch_dtype(self._dtype).is_floating_point) else torch.float32
         self._device = torch.device(device) if device is not None else torch.device("cpu")
+        if self._fitness_device is None and getattr(objective_func, "__evotorch_on_aux_device__", False):
+            # @on_aux_device: run fitness on the first accelerator (resolved
+            # now that the main device is known) — reference decorators.py:440
+            self._fitness_device = self.aux_device
         if self.dtype_is_object and self._device.type != "cpu":
             raise ValueError("object-dtype problems must live on cpu")
 

@@ -132,3 +132,22 @@ def test_rowwise_with_scalar_arg():
 
     out = scale_row(torch.ones(5, 2), 3.0)
     assert torch.allclose(out, torch.full((5, 2), 3.0))
+
+
+def test_on_aux_device_moves_fitness_batches():
+    """@on_aux_device resolves to Problem.aux_device at construction, and
+    vectorized fitness batches are delivered on that device."""
+    from evotorch_amd import Problem
+
+    seen = []
+
+    @vectorized
+    @on_aux_device
+    def f(x):
+        seen.append(x.device)
+        return (x**2).sum(-1)
+
+    prob = Problem("min", f, solution_length=4, initial_bounds=(-1, 1), seed=1)
+    batch = prob.generate_batch(3)
+    prob.evaluate(batch)
+    assert seen and seen[0] == prob.aux_device
