@@ -289,3 +289,26 @@ def test_streaming_with_rollout_problem():
     last = float(searcher.status["mean_eval"])
     assert prob.obs_norm.count > 0
     assert last > first, (first, last)
+
+
+@requires_gpu
+def test_on_aux_device_fitness_runs_on_gpu():
+    """A CPU problem whose fitness is @on_aux_device gets cuda batches
+    (the reference's aux_device pattern for heavy fitness on accelerators)."""
+    from evotorch_amd import Problem
+    from evotorch_amd.decorators import on_aux_device, vectorized
+
+    seen = []
+
+    @vectorized
+    @on_aux_device
+    def f(x):
+        seen.append(x.device.type)
+        return (x**2).sum(-1)
+
+    prob = Problem("min", f, solution_length=8, initial_bounds=(-1, 1), seed=1)  # cpu problem
+    batch = prob.generate_batch(16)
+    prob.evaluate(batch)
+    assert seen[0] == "cuda"
+    assert batch.evals_are_ready
+    assert batch.unsafe_evals.device.type == "cpu"  # results land back on the problem device
