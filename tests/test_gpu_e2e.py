@@ -248,6 +248,14 @@ def _misc_searchers_body(dev):
     ranks, _ = ga.population.compute_pareto_ranks(crowdsort=False)
     assert int((ranks == 0).sum()) > 128
 
+    # Cosyne (column permutation operator on device)
+    from evotorch_amd.algorithms import Cosyne
+
+    prob = Problem("min", sphere, solution_length=8, initial_bounds=(-3, 3), device=dev, seed=11)
+    cosyne = Cosyne(prob, popsize=40, tournament_size=4, mutation_stdev=0.3)
+    cosyne.run(10)
+    assert float(cosyne.status["pop_best_eval"]) < float("inf")
+
     # IPOP restart wrapper around SNES
     prob = Problem("min", sphere, solution_length=8, initial_bounds=(-3, 3), device=dev, seed=4)
     ipop = IPOP(prob, SNES, algorithm_args={"stdev_init": 1.0, "popsize": 20},
