@@ -740,7 +740,28 @@ class Problem(TensorMakerMixin, Serializable):
         seed: Optional[int] = None,
         store_solution_stats: Optional[bool] = None,
         vectorized: Optional[bool] = None,
+        num_actors=None,
+        actor_config=None,
+        num_gpus_per_actor=None,
+        num_subbatches=None,
+        subbatch_size=None,
     ):
+        # Ray-era knobs accepted for drop-in migration from the reference;
+        # parallelism here comes from the torchrun/RCCL topology instead
+        # (docs/migrating_from_evotorch.md), so they are ignored with a
+        # warning rather than a TypeError.
+        _ray_knobs = {"num_actors": num_actors, "actor_config": actor_config,
+                      "num_gpus_per_actor": num_gpus_per_actor,
+                      "num_subbatches": num_subbatches, "subbatch_size": subbatch_size}
+        _given = [k for k, v in _ray_knobs.items() if v is not None]
+        if _given:
+            import warnings
+
+            warnings.warn(
+                f"Ignoring Ray-era argument(s) {_given}: this framework parallelizes via "
+                "torchrun + RCCL (problem.use_comm(init_comm())); see docs/migrating_from_evotorch.md",
+                stacklevel=2,
+            )
         self._senses = _normalize_sense(objective_sense)
         self._objective_func = objective_func
         self._vectorized = bool(getattr(objective_func, "__evotorch_vectorized__", False)) if vectorized is None else bool(vectorized)

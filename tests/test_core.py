@@ -299,3 +299,19 @@ def test_batch_utility_and_utils():
     mo.evaluate(b2)
     us = b2.utils(ranking_method="centered")
     assert us.shape == (6, 2)
+
+
+def test_ray_era_kwargs_accepted_with_warning():
+    """Reference code passing num_actors etc. migrates without a TypeError
+    (the knobs are ignored; parallelism comes from the torchrun topology)."""
+    import warnings
+
+    with warnings.catch_warnings(record=True) as w:
+        warnings.simplefilter("always")
+        prob = make_problem_with_kwargs = Problem(
+            "min", sphere, solution_length=4, initial_bounds=(-1, 1),
+            num_actors="max", num_gpus_per_actor=0.5, subbatch_size=50)
+    assert any("Ray-era" in str(x.message) for x in w)
+    batch = prob.generate_batch(4)
+    prob.evaluate(batch)
+    assert batch.evals_are_ready
