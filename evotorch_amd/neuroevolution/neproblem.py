@@ -34,8 +34,18 @@ class NEProblem(Problem):
         seed: Optional[int] = None,
         device=None,
         num_gpus_per_actor=None,  # accepted for API parity; SPMD topology comes from the launcher
+        num_actors=None,
+        actor_config=None,
+        num_subbatches=None,
+        subbatch_size=None,
         store_solution_stats: Optional[bool] = None,
     ):
+        if any(v is not None for v in (num_actors, actor_config, num_subbatches, subbatch_size)):
+            import warnings
+
+            warnings.warn(
+                "Ignoring Ray-era actor argument(s): parallelize via torchrun + RCCL "
+                "(docs/migrating_from_evotorch.md)", stacklevel=2)
         self._network_def = network
         self._network_args = dict(network_args or {})
         self._network_eval_func = network_eval_func
