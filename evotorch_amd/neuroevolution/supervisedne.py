@@ -40,7 +40,16 @@ class SupervisedNE(NEProblem):
         device=None,
         seed: Optional[int] = None,
         vectorized_eval: bool = True,
+        subbatch_size: Optional[int] = None,
+        num_subbatches: Optional[int] = None,
+        num_actors=None,
+        num_gpus_per_actor=None,
+        actor_config=None,
     ):
+        if any(v is not None for v in (num_actors, num_gpus_per_actor, actor_config)):
+            import warnings
+
+            warnings.warn("Ignoring Ray-era actor argument(s): parallelize via torchrun + RCCL", stacklevel=2)
         super().__init__(
             "min" if loss_as_fitness_sign == "min" else "max",
             network,
@@ -58,6 +67,8 @@ class SupervisedNE(NEProblem):
         self._dataloader: Optional[DataLoader] = None
         self._dataloader_iter = None
         self._fmodule = None
+        self._subbatch_size = None if subbatch_size is None else int(subbatch_size)
+        self._num_subbatches = None if num_subbatches is None else int(num_subbatches)
 
     # -- data plumbing -------------------------------------------------------
 
@@ -118,17 +129,27 @@ class SupervisedNE(NEProblem):
         if not (self._common_minibatch and self._vectorized_eval):
             super()._evaluate_batch(batch)
             return
-        # common minibatch + whole-population vmapped forward
+        # common minibatch + whole-population vmapped forward. With
+        # subbatch_size (reference supervisedne.py:334-348, the MNIST30K
+        # config uses 50), each SUBBATCH of solutions shares one fresh
+        # minibatch — a middle ground between per-solution minibatches
+        # (high variance between solutions) and one global minibatch.
         if self._fmodule is None:
             self._fmodule = make_functional_module(self._instantiate_net().to(self.network_device))
         params = batch.access_values(keep_evals=True).to(self.network_device, torch.float32)
-        losses = torch.zeros(len(batch), dtype=torch.float32, device=self.network_device)
+        n = len(batch)
+        sub = self._subbatch_size
+        if sub is None and self._num_subbatches is not None:
+            sub = max(1, (n + self._num_subbatches - 1) // self._num_subbatches)
+        losses = torch.zeros(n, dtype=torch.float32, device=self.network_device)
         with torch.no_grad():
             for _ in range(self._num_minibatches):
-                x, y = self.get_minibatch()
+                for start in range(0, n, sub or n):
+                    stop = min(start + (sub or n), n)
+                    x, y = self.get_minibatch()
 
-                def member_loss(flat):
-                    return self._loss(self._fmodule._single(flat, x), y)
+                    def member_loss(flat):
+                        return self._loss(self._fmodule._single(flat, x), y)
 
-                losses = losses + torch.func.vmap(member_loss, randomness="different")(params)
+                    losses[start:stop] = losses[start:stop] + torch.func.vmap(member_loss, randomness="different")(params[start:stop])
         batch.set_evals((losses / self._num_minibatches).to(batch.device))
