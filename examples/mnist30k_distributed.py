@@ -71,6 +71,7 @@ def main():
         nn.CrossEntropyLoss(),
         minibatch_size=1024,
         common_minibatch=True,
+        subbatch_size=50,       # reference MNIST30K config (Training_MNIST30K.ipynb)
         device=device,
         seed=1 + (comm.rank if comm else 0),
     )
