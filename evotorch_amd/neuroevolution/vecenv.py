@@ -125,13 +125,27 @@ class VecEnvNE(NEProblem):
         device=None,
         seed: Optional[int] = None,
         initial_bounds=(-0.00001, 0.00001),
+        env_config: Optional[dict] = None,
+        max_num_envs: Optional[int] = None,
+        action_noise_stdev: Optional[float] = None,
+        num_episodes: int = 1,
+        num_actors=None,
+        num_gpus_per_actor=None,
+        num_subbatches=None,
+        subbatch_size=None,
+        actor_config=None,
     ):
+        if any(v is not None for v in (num_actors, num_gpus_per_actor, num_subbatches, subbatch_size, actor_config)):
+            import warnings
+
+            warnings.warn("Ignoring Ray-era actor argument(s): parallelize via torchrun + RCCL", stacklevel=2)
+        env_config = dict(env_config or {})
         if isinstance(env, str):
             env_name = env
             def factory(num_envs: int):
                 import gymnasium as gym
 
-                return GymVectorEnvAdapter(gym.make_vec(env_name, num_envs=num_envs), device=device or "cpu")
+                return GymVectorEnvAdapter(gym.make_vec(env_name, num_envs=num_envs, **env_config), device=device or "cpu")
 
             self._env_factory = factory
             self._env = None
@@ -148,6 +162,9 @@ class VecEnvNE(NEProblem):
             self._obs_dim, self._act_dim = env.obs_dim, env.act_dim
 
         self._max_num_steps = max_num_steps
+        self._max_num_envs = None if max_num_envs is None else int(max_num_envs)
+        self._action_noise_stdev = None if action_noise_stdev is None else float(action_noise_stdev)
+        self._num_episodes = max(1, int(num_episodes))
         self._obs_norm_enabled = bool(observation_normalization)
         self._decrease_rewards_by = float(decrease_rewards_by)
         self._alive_bonus_schedule = alive_bonus_schedule
@@ -225,6 +242,27 @@ class VecEnvNE(NEProblem):
     @torch.no_grad()  # rollouts never need autograd (user nets may carry requires_grad params)
     def _evaluate_batch(self, batch: SolutionBatch):
         n = len(batch)
+        if self._max_num_envs is not None and n > self._max_num_envs:
+            # reference vecgymne.py: cap the batched env size and evaluate
+            # the population in env-sized pieces
+            for piece in batch.split(max_size=self._max_num_envs):
+                self._evaluate_batch(piece)
+            return
+        if self._num_episodes > 1:
+            total = torch.zeros(n, dtype=torch.float32)
+            steps = 0
+            for _ in range(self._num_episodes):
+                self._rollout_once(batch)
+                total = total + torch.Tensor.as_subclass(batch._evals[:, 0], torch.Tensor).to("cpu", torch.float32)
+                steps += self.last_eval_interaction_count
+                batch.forget_evals()
+            batch.set_evals((total / self._num_episodes).to(self._eval_dtype).to(batch.device))
+            self.last_eval_interaction_count = steps
+            return
+        self._rollout_once(batch)
+
+    def _rollout_once(self, batch: SolutionBatch):
+        n = len(batch)
         env = self._get_env(n)
         device = self.network_device
         if self._policy is None:
@@ -249,6 +287,8 @@ class VecEnvNE(NEProblem):
             else:
                 obs_in = obs
             actions = policy(obs_in)
+            if self._action_noise_stdev is not None:
+                actions = actions + self._action_noise_stdev * torch.randn_like(actions)
             obs, reward, done = env.step(actions)
             obs = obs.to(device)
             bonus = self._alive_bonus(t)
@@ -291,7 +331,7 @@ class VecEnvNE(NEProblem):
 
     @property
     def max_num_envs(self) -> Optional[int]:
-        return getattr(self._env, "num_envs", None) if getattr(self, "_env", None) is not None else None
+        return self._max_num_envs
 
     def get_env(self):
         """The underlying batched env (created lazily on first evaluation)."""

@@ -347,3 +347,23 @@ def test_save_load_policy_safetensors(tmp_path):
 
     sidecar = json.load(open(path + ".json"))
     assert sidecar["metadata"]["gens"] == 10
+
+
+def test_vecenvne_max_num_envs_and_episodes():
+    """max_num_envs splits the population into env-sized pieces;
+    num_episodes averages fitness over repeated rollouts; action noise
+    perturbs actions (reference vecgymne.py ctor knobs)."""
+    from evotorch_amd.neuroevolution import SyntheticTorchEnv, VecEnvNE
+
+    def factory(n):
+        return SyntheticTorchEnv(num_envs=n, obs_dim=6, act_dim=2, episode_length=5)
+
+    prob = VecEnvNE(factory, "Linear(obs_length, act_length)", seed=3,
+                    max_num_envs=4, num_episodes=2, action_noise_stdev=0.05,
+                    observation_normalization=False)
+    batch = prob.generate_batch(10)
+    prob.evaluate(batch)
+    assert batch.evals_are_ready
+    assert prob.max_num_envs == 4
+    # averaged over 2 episodes of 5 steps each, pieces of <= 4 envs
+    assert prob.interaction_count == 10 * 5 * 2
