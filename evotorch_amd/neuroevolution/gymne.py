@@ -136,7 +136,15 @@ class GymNE(NEProblem):
         action_noise_stdev: Optional[float] = None,
         initial_bounds=(-0.00001, 0.00001),
         seed: Optional[int] = None,
+        num_actors=None,
+        actor_config=None,
+        num_subbatches=None,
+        subbatch_size=None,
     ):
+        if any(v is not None for v in (num_actors, actor_config, num_subbatches, subbatch_size)):
+            import warnings
+
+            warnings.warn("Ignoring Ray-era actor argument(s): parallelize via torchrun + RCCL", stacklevel=2)
         try:
             import gymnasium  # noqa: F401
         except ImportError as e:
