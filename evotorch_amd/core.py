@@ -1204,6 +1204,8 @@ class Problem(TensorMakerMixin, Serializable):
         comm = self._comm
         self._before_grad_hook()
         if chunk_rows is not None:
+            if int(chunk_rows) < 1:
+                raise ValueError(f"chunk_rows must be >= 1, got {chunk_rows}")
             if num_interactions is not None:
                 raise ValueError("chunk_rows (streaming gradients) does not support num_interactions adaptive popsize")
             result = self._sample_and_compute_gradients_streamed(distribution, int(popsize), obj_index, ranking_method, int(chunk_rows), comm)
