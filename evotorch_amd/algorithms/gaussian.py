@@ -464,6 +464,7 @@ class CEM(GaussianSearchAlgorithm):
         obj_index: Optional[int] = None,
         distributed: bool = False,
         popsize_weighted_grad_avg: Optional[bool] = None,
+        grad_chunk_rows: Optional[int] = None,
     ):
         self.DISTRIBUTION_PARAMS = {"parenthood_ratio": float(parenthood_ratio)}
         super().__init__(
@@ -485,6 +486,7 @@ class CEM(GaussianSearchAlgorithm):
             obj_index=obj_index,
             distributed=distributed,
             popsize_weighted_grad_avg=popsize_weighted_grad_avg,
+            grad_chunk_rows=grad_chunk_rows,
         )
 
 

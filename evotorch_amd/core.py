@@ -1324,6 +1324,11 @@ class Problem(TensorMakerMixin, Serializable):
         """Two-pass streaming ES gradients — see sample_and_compute_gradients."""
         if not hasattr(distribution, "compute_gradients_streamed"):
             raise ValueError(f"{type(distribution).__name__} does not support streamed gradients")
+        if "parenthood_ratio" in getattr(distribution, "parameters", {}) and comm is not None and comm.world_size > 1:
+            raise ValueError(
+                "streamed CEM elite gradients are single-rank only: the elite mean/std "
+                "merge is not a popsize-weighted average (shard the population without chunk_rows instead)"
+            )
         from .ops.dispatch import _seed_from_generator
         from .utils import ranking as ranking_mod
 

@@ -215,6 +215,33 @@ class SeparableGaussian(Distribution):
 
         ops.sample_gaussian(out, self.mu, self.sigma, symmetric=self._symmetric, seed=seed, elem_offset=elem_offset)
 
+    def _compute_elite_gradients_streamed(self, chunk_iter, weights: torch.Tensor) -> dict:
+        """CEM elite gradients without materializing the population: the
+        elite SET is fixed by the full utility vector, and elite mean/std
+        reduce to masked (Σx, Σx², k) accumulated chunk by chunk —
+        algebraically identical to `_elite_gradients`."""
+        num_samples = weights.shape[0]
+        num_elites = math.floor(num_samples * float(self._parameters["parenthood_ratio"]))
+        elite_idx = weights.argsort(descending=True)[:num_elites]
+        is_elite = torch.zeros(num_samples, dtype=torch.bool, device=weights.device)
+        is_elite[elite_idx] = True
+        sum_x = torch.zeros_like(self.mu, dtype=torch.float64)
+        sum_x2 = torch.zeros_like(self.mu, dtype=torch.float64)
+        for values_chunk, row0, rows in chunk_iter:
+            mask = is_elite[row0 : row0 + rows]
+            if bool(mask.any()):
+                selected = values_chunk[mask].to(torch.float64)
+                sum_x += selected.sum(dim=0)
+                sum_x2 += (selected**2).sum(dim=0)
+        k = max(num_elites, 1)
+        mean = sum_x / k
+        var = (sum_x2 - k * mean**2) / max(k - 1, 1)
+        std = torch.sqrt(torch.clamp(var, min=0.0))
+        return {
+            "mu": (mean - self.mu.to(torch.float64)).to(self.mu.dtype),
+            "sigma": (std - self.sigma.to(torch.float64)).to(self.sigma.dtype),
+        }
+
     def compute_gradients_streamed(self, chunk_iter, weights: torch.Tensor, *, ranking_used: Optional[str]) -> dict:
         """Gradients over a chunked (never-materialized) population.
 
@@ -225,9 +252,10 @@ class SeparableGaussian(Distribution):
         and the (global, weight-dependent) normalization is applied once at
         the end — exactly equal to the unstreamed gradient.
         """
-        if "parenthood_ratio" in self._parameters:
-            raise ValueError("elite (parenthood_ratio) gradients cannot be streamed")
         from . import ops
+
+        if "parenthood_ratio" in self._parameters:
+            return self._compute_elite_gradients_streamed(chunk_iter, weights)
 
         w = self._centered_weights(weights, ranking_used)
         d = w.shape[0] // 2 if self._symmetric else w.shape[0]
@@ -375,6 +403,33 @@ class ExpSeparableGaussian(SeparableGaussian):
 
         mu_grad, sigma_grad = ops.snes_gradients(samples, self.mu, self.sigma, weights)
         return {"mu": mu_grad, "sigma": sigma_grad}
+
+    def _compute_elite_gradients_streamed(self, chunk_iter, weights: torch.Tensor) -> dict:
+        """CEM elite gradients without materializing the population: the
+        elite SET is fixed by the full utility vector, and elite mean/std
+        reduce to masked (Σx, Σx², k) accumulated chunk by chunk —
+        algebraically identical to `_elite_gradients`."""
+        num_samples = weights.shape[0]
+        num_elites = math.floor(num_samples * float(self._parameters["parenthood_ratio"]))
+        elite_idx = weights.argsort(descending=True)[:num_elites]
+        is_elite = torch.zeros(num_samples, dtype=torch.bool, device=weights.device)
+        is_elite[elite_idx] = True
+        sum_x = torch.zeros_like(self.mu, dtype=torch.float64)
+        sum_x2 = torch.zeros_like(self.mu, dtype=torch.float64)
+        for values_chunk, row0, rows in chunk_iter:
+            mask = is_elite[row0 : row0 + rows]
+            if bool(mask.any()):
+                selected = values_chunk[mask].to(torch.float64)
+                sum_x += selected.sum(dim=0)
+                sum_x2 += (selected**2).sum(dim=0)
+        k = max(num_elites, 1)
+        mean = sum_x / k
+        var = (sum_x2 - k * mean**2) / max(k - 1, 1)
+        std = torch.sqrt(torch.clamp(var, min=0.0))
+        return {
+            "mu": (mean - self.mu.to(torch.float64)).to(self.mu.dtype),
+            "sigma": (std - self.sigma.to(torch.float64)).to(self.sigma.dtype),
+        }
 
     def compute_gradients_streamed(self, chunk_iter, weights: torch.Tensor, *, ranking_used: Optional[str]) -> dict:
         from . import ops
