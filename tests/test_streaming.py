@@ -60,10 +60,13 @@ def test_streamed_gradients_plain_separable():
         assert torch.allclose(r1["gradients"][k], r2["gradients"][k], atol=1e-5)
 
 
-def test_streamed_rejects_elite_and_foreign_distributions():
-    elite = SeparableGaussian({"mu": torch.zeros(4), "sigma": torch.ones(4), "parenthood_ratio": 0.5})
+def test_streamed_rejects_unsupported_configs():
+    # elite streaming is single-rank only; num_interactions unsupported; chunk_rows >= 1
+    dist = SymmetricSeparableGaussian({"mu": torch.zeros(4), "sigma": torch.ones(4)})
     with pytest.raises(ValueError):
-        make_problem(length=4).sample_and_compute_gradients(elite, 8, chunk_rows=4)
+        make_problem(length=4).sample_and_compute_gradients(dist, 8, chunk_rows=0)
+    with pytest.raises(ValueError):
+        make_problem(length=4).sample_and_compute_gradients(dist, 8, chunk_rows=4, num_interactions=100)
 
 
 def test_pgpe_with_streaming_converges():
