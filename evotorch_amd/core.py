@@ -1324,11 +1324,7 @@ class Problem(TensorMakerMixin, Serializable):
         """Two-pass streaming ES gradients — see sample_and_compute_gradients."""
         if not hasattr(distribution, "compute_gradients_streamed"):
             raise ValueError(f"{type(distribution).__name__} does not support streamed gradients")
-        if "parenthood_ratio" in getattr(distribution, "parameters", {}) and comm is not None and comm.world_size > 1:
-            raise ValueError(
-                "streamed CEM elite gradients are single-rank only: the elite mean/std "
-                "merge is not a popsize-weighted average (shard the population without chunk_rows instead)"
-            )
+        elite_mode = "parenthood_ratio" in getattr(distribution, "parameters", {})
         from .ops.dispatch import _seed_from_generator
         from .utils import ranking as ranking_mod
 
@@ -1392,13 +1388,33 @@ class Problem(TensorMakerMixin, Serializable):
                 yield batch._values, r0, rows
 
         with record_range("stream_grad"):
-            grads = distribution.compute_gradients_streamed(regen_chunks(), my_utils, ranking_used=(ranking_method or "raw"))
-        if comm is not None and world > 1:
-            total = world * local_popsize
-            for k in grads:
-                grads[k] = grads[k] * (local_popsize / total)
-            comm.all_reduce_container(grads)
-        else:
+            if elite_mode and comm is not None and world > 1:
+                # sharded CEM: elite SET chosen from the GLOBAL utilities;
+                # each rank accumulates its local elite members' (Σx, Σx²)
+                # and one all-reduce of the sums (not a weighted gradient
+                # average — elite stats are non-linear in the shards)
+                import math as _math
+
+                total = world * local_popsize
+                num_elites = _math.floor(total * float(distribution.parameters["parenthood_ratio"]))
+                global_elite = torch.zeros(total, dtype=torch.bool, device=all_utils.device)
+                global_elite[all_utils.argsort(descending=True)[:num_elites]] = True
+                my_elite = global_elite[my_rank * local_popsize : (my_rank + 1) * local_popsize]
+                sum_x, sum_x2 = distribution.accumulate_elite_sums_streamed(regen_chunks(), my_elite)
+                sums = {"x": sum_x, "x2": sum_x2}
+                comm.all_reduce_container(sums)
+                grads = distribution.finalize_elite_gradients(
+                    sums["x"].to(torch.float64), sums["x2"].to(torch.float64), num_elites)
+            else:
+                grads = distribution.compute_gradients_streamed(regen_chunks(), my_utils, ranking_used=(ranking_method or "raw"))
+                if comm is not None and world > 1:
+                    total = world * local_popsize
+                    for k in grads:
+                        grads[k] = grads[k] * (local_popsize / total)
+                    comm.all_reduce_container(grads)
+                else:
+                    total = local_popsize
+        if comm is None or world <= 1:
             total = local_popsize
         return {
             "gradients": grads,

@@ -215,16 +215,9 @@ class SeparableGaussian(Distribution):
 
         ops.sample_gaussian(out, self.mu, self.sigma, symmetric=self._symmetric, seed=seed, elem_offset=elem_offset)
 
-    def _compute_elite_gradients_streamed(self, chunk_iter, weights: torch.Tensor) -> dict:
-        """CEM elite gradients without materializing the population: the
-        elite SET is fixed by the full utility vector, and elite mean/std
-        reduce to masked (Σx, Σx², k) accumulated chunk by chunk —
-        algebraically identical to `_elite_gradients`."""
-        num_samples = weights.shape[0]
-        num_elites = math.floor(num_samples * float(self._parameters["parenthood_ratio"]))
-        elite_idx = weights.argsort(descending=True)[:num_elites]
-        is_elite = torch.zeros(num_samples, dtype=torch.bool, device=weights.device)
-        is_elite[elite_idx] = True
+    def accumulate_elite_sums_streamed(self, chunk_iter, is_elite: torch.Tensor):
+        """Masked (Σx, Σx²) over elite rows, chunk by chunk (fp64 accum).
+        `is_elite` flags LOCAL rows; shards all-reduce the returned sums."""
         sum_x = torch.zeros_like(self.mu, dtype=torch.float64)
         sum_x2 = torch.zeros_like(self.mu, dtype=torch.float64)
         for values_chunk, row0, rows in chunk_iter:
@@ -233,7 +226,12 @@ class SeparableGaussian(Distribution):
                 selected = values_chunk[mask].to(torch.float64)
                 sum_x += selected.sum(dim=0)
                 sum_x2 += (selected**2).sum(dim=0)
-        k = max(num_elites, 1)
+        return sum_x, sum_x2
+
+    def finalize_elite_gradients(self, sum_x: torch.Tensor, sum_x2: torch.Tensor, num_elites: int) -> dict:
+        """Turn global elite (Σx, Σx², k) into the CEM gradients —
+        algebraically identical to `_elite_gradients`."""
+        k = max(int(num_elites), 1)
         mean = sum_x / k
         var = (sum_x2 - k * mean**2) / max(k - 1, 1)
         std = torch.sqrt(torch.clamp(var, min=0.0))
@@ -241,6 +239,16 @@ class SeparableGaussian(Distribution):
             "mu": (mean - self.mu.to(torch.float64)).to(self.mu.dtype),
             "sigma": (std - self.sigma.to(torch.float64)).to(self.sigma.dtype),
         }
+
+    def _compute_elite_gradients_streamed(self, chunk_iter, weights: torch.Tensor) -> dict:
+        """Single-population form: elite set from the full utility vector."""
+        num_samples = weights.shape[0]
+        num_elites = math.floor(num_samples * float(self._parameters["parenthood_ratio"]))
+        elite_idx = weights.argsort(descending=True)[:num_elites]
+        is_elite = torch.zeros(num_samples, dtype=torch.bool, device=weights.device)
+        is_elite[elite_idx] = True
+        sum_x, sum_x2 = self.accumulate_elite_sums_streamed(chunk_iter, is_elite)
+        return self.finalize_elite_gradients(sum_x, sum_x2, num_elites)
 
     def compute_gradients_streamed(self, chunk_iter, weights: torch.Tensor, *, ranking_used: Optional[str]) -> dict:
         """Gradients over a chunked (never-materialized) population.
@@ -404,16 +412,9 @@ class ExpSeparableGaussian(SeparableGaussian):
         mu_grad, sigma_grad = ops.snes_gradients(samples, self.mu, self.sigma, weights)
         return {"mu": mu_grad, "sigma": sigma_grad}
 
-    def _compute_elite_gradients_streamed(self, chunk_iter, weights: torch.Tensor) -> dict:
-        """CEM elite gradients without materializing the population: the
-        elite SET is fixed by the full utility vector, and elite mean/std
-        reduce to masked (Σx, Σx², k) accumulated chunk by chunk —
-        algebraically identical to `_elite_gradients`."""
-        num_samples = weights.shape[0]
-        num_elites = math.floor(num_samples * float(self._parameters["parenthood_ratio"]))
-        elite_idx = weights.argsort(descending=True)[:num_elites]
-        is_elite = torch.zeros(num_samples, dtype=torch.bool, device=weights.device)
-        is_elite[elite_idx] = True
+    def accumulate_elite_sums_streamed(self, chunk_iter, is_elite: torch.Tensor):
+        """Masked (Σx, Σx²) over elite rows, chunk by chunk (fp64 accum).
+        `is_elite` flags LOCAL rows; shards all-reduce the returned sums."""
         sum_x = torch.zeros_like(self.mu, dtype=torch.float64)
         sum_x2 = torch.zeros_like(self.mu, dtype=torch.float64)
         for values_chunk, row0, rows in chunk_iter:
@@ -422,7 +423,12 @@ class ExpSeparableGaussian(SeparableGaussian):
                 selected = values_chunk[mask].to(torch.float64)
                 sum_x += selected.sum(dim=0)
                 sum_x2 += (selected**2).sum(dim=0)
-        k = max(num_elites, 1)
+        return sum_x, sum_x2
+
+    def finalize_elite_gradients(self, sum_x: torch.Tensor, sum_x2: torch.Tensor, num_elites: int) -> dict:
+        """Turn global elite (Σx, Σx², k) into the CEM gradients —
+        algebraically identical to `_elite_gradients`."""
+        k = max(int(num_elites), 1)
         mean = sum_x / k
         var = (sum_x2 - k * mean**2) / max(k - 1, 1)
         std = torch.sqrt(torch.clamp(var, min=0.0))
@@ -430,6 +436,16 @@ class ExpSeparableGaussian(SeparableGaussian):
             "mu": (mean - self.mu.to(torch.float64)).to(self.mu.dtype),
             "sigma": (std - self.sigma.to(torch.float64)).to(self.sigma.dtype),
         }
+
+    def _compute_elite_gradients_streamed(self, chunk_iter, weights: torch.Tensor) -> dict:
+        """Single-population form: elite set from the full utility vector."""
+        num_samples = weights.shape[0]
+        num_elites = math.floor(num_samples * float(self._parameters["parenthood_ratio"]))
+        elite_idx = weights.argsort(descending=True)[:num_elites]
+        is_elite = torch.zeros(num_samples, dtype=torch.bool, device=weights.device)
+        is_elite[elite_idx] = True
+        sum_x, sum_x2 = self.accumulate_elite_sums_streamed(chunk_iter, is_elite)
+        return self.finalize_elite_gradients(sum_x, sum_x2, num_elites)
 
     def compute_gradients_streamed(self, chunk_iter, weights: torch.Tensor, *, ranking_used: Optional[str]) -> dict:
         from . import ops

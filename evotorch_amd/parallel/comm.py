@@ -118,7 +118,12 @@ class Comm:
         if not self.active:
             return container
         keys = sorted(container.keys())
-        flats = [container[k].reshape(-1).to(torch.float32) for k in keys]
+        # reduce in the widest dtype present (fp64 stats stay fp64)
+        widest = torch.float32
+        for k in keys:
+            if container[k].dtype == torch.float64:
+                widest = torch.float64
+        flats = [container[k].reshape(-1).to(widest) for k in keys]
         buf = torch.cat(flats)
         self.all_reduce_(buf, op=op)
         offset = 0

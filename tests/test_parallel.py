@@ -162,6 +162,30 @@ def _body_streamed_pgpe(comm, rank, world):
     return center_t.tolist()
 
 
+def _body_streamed_cem(comm, rank, world):
+    """Sharded + streamed CEM: global elite set, (Σx, Σx²) all-reduce."""
+    from evotorch_amd import Problem
+    from evotorch_amd.algorithms import CEM
+    from evotorch_amd.decorators import vectorized
+
+    @vectorized
+    def sphere(x):
+        return (x**2).sum(-1)
+
+    prob = Problem("min", sphere, solution_length=8, initial_bounds=(-1, 1), seed=80 + rank)
+    prob.use_comm(comm)
+    searcher = CEM(prob, popsize=48, parenthood_ratio=0.25, stdev_init=1.5,
+                   center_init=torch.ones(8) * 2, distributed=True, grad_chunk_rows=6)
+    for _ in range(15):
+        searcher.step()
+    center_t = torch.Tensor.as_subclass(searcher.status["center"], torch.Tensor).clone()
+    ref = center_t.clone()
+    comm.broadcast_(ref, src=0)
+    assert torch.allclose(center_t, ref, atol=1e-6), "ranks diverged"
+    assert searcher.status["mean_eval"] < 8 * 4
+    return center_t.tolist()
+
+
 def _make_dist(prob):
     from evotorch_amd.distributions import SymmetricSeparableGaussian
 
@@ -174,14 +198,14 @@ def _make_dist(prob):
 
 @pytest.mark.parametrize(
     "body",
-    ["_body_collectives", "_body_sharded_evaluate", "_body_distributed_pgpe", "_body_obs_norm_allreduce", "_body_streamed_pgpe"],
+    ["_body_collectives", "_body_sharded_evaluate", "_body_distributed_pgpe", "_body_obs_norm_allreduce", "_body_streamed_pgpe", "_body_streamed_cem"],
 )
 def test_world2(body):
     results = _run_world(body, world=2)
     assert len(results) == 2
     if body == "_body_sharded_evaluate":
         assert results[0] == results[1]  # both ranks hold the full eval vector
-    if body in ("_body_distributed_pgpe", "_body_streamed_pgpe"):
+    if body in ("_body_distributed_pgpe", "_body_streamed_pgpe", "_body_streamed_cem"):
         assert results[0] == results[1]
 
 
