@@ -157,3 +157,24 @@ def test_streamed_cem_elite_gradients():
     first = float(searcher.status["mean_eval"])
     searcher.run(50)
     assert float(searcher.status["mean_eval"]) < first * 0.3
+
+
+def test_streaming_with_per_solution_problem():
+    """Streaming composes with non-vectorized (per-solution _evaluate)
+    problems — pass 1 uses the normal evaluation pipeline."""
+
+    class PerSolutionProblem(Problem):
+        def __init__(self):
+            super().__init__("min", solution_length=6, initial_bounds=(-1, 1), seed=5)
+
+        def _evaluate(self, solution):
+            values = torch.Tensor.as_subclass(solution.values, torch.Tensor)
+            solution.set_evaluation(float((values**2).sum()))
+
+    prob = PerSolutionProblem()
+    searcher = PGPE(prob, popsize=24, center_learning_rate=0.3, stdev_learning_rate=0.1,
+                    stdev_init=1.5, distributed=True, grad_chunk_rows=4)
+    searcher.step()
+    first = float(searcher.status["mean_eval"])
+    searcher.run(30)
+    assert float(searcher.status["mean_eval"]) < first * 0.5
