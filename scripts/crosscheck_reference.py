@@ -111,6 +111,25 @@ def main():
     for k in sorted(g_ref):
         report.append((f"ExpSep-nes/{k}", (g_ref[k] - g_my[k]).abs().max().item()))
 
+    # constraint penalization helpers
+    from evotorch.tools import constraints as ref_constraints
+
+    from evotorch_amd.utils import constraints as my_constraints
+
+    cx = torch.randn(64, dtype=torch.float64) * 3
+    for op in ("<=", ">="):
+        a = ref_constraints.violation(lhs=cx, comparison=op, rhs=1.0)
+        b = my_constraints.violation(lhs=cx, comparison=op, rhs=1.0)
+        report.append((f"violation/{op}", (a - b).abs().max().item()))
+        a = ref_constraints.log_barrier(lhs=cx, comparison=op, rhs=1.0, penalty_sign="-")
+        b = my_constraints.log_barrier(lhs=cx, comparison=op, rhs=1.0, penalty_sign="-")
+        finite = torch.isfinite(a)
+        assert torch.equal(torch.isfinite(b), finite)  # same -inf pattern
+        report.append((f"log_barrier/{op}", (a[finite] - b[finite]).abs().max().item() if finite.any() else 0.0))
+    a = ref_constraints.penalty(lhs=cx, comparison="<=", rhs=1.0, penalty_sign="-", linear=2.0, step=5.0, exp=2.0, exp_inf=100.0)
+    b = my_constraints.penalty(lhs=cx, comparison="<=", rhs=1.0, penalty_sign="-", linear=2.0, step=5.0, exp=2.0, exp_inf=100.0)
+    report.append(("penalty", (a - b).abs().max().item()))
+
     g_seq = [torch.randn(L, dtype=torch.float64) for _ in range(5)]
     pairs = [
         ("ClipUp/5steps",
