@@ -111,6 +111,21 @@ def main():
     for k in sorted(g_ref):
         report.append((f"ExpSep-nes/{k}", (g_ref[k] - g_my[k]).abs().max().item()))
 
+    # stdev-control clamping (modify_tensor drives PGPE's stdev_max_change)
+    from evotorch.tools.misc import modify_tensor as ref_modify
+    from evotorch.tools.misc import stdev_from_radius as ref_sfr
+
+    from evotorch_amd.utils import modify_tensor as my_modify
+    from evotorch_amd.utils import stdev_from_radius as my_sfr
+
+    orig = torch.rand(40, dtype=torch.float64) + 0.5
+    target = orig * torch.empty(40, dtype=torch.float64).uniform_(0.3, 3.0)
+    for kwargs in (dict(lb=0.6, ub=2.0), dict(max_change=0.2), dict(lb=0.7, ub=1.8, max_change=0.3)):
+        a = ref_modify(orig, target, **kwargs)
+        b = my_modify(orig, target, **kwargs)
+        report.append((f"modify_tensor/{sorted(kwargs)}", (a - b).abs().max().item()))
+    report.append(("stdev_from_radius", abs(ref_sfr(2.25, 6409) - my_sfr(2.25, 6409))))
+
     # constraint penalization helpers
     from evotorch.tools import constraints as ref_constraints
 
