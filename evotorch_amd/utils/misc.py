@@ -338,32 +338,33 @@ def modify_tensor(
     max_change: Optional[RealOrVector] = None,
     in_place: bool = False,
 ) -> torch.Tensor:
-    """Move `original` towards `target`, clamping the per-element change to
-    ``max_change`` (absolute if max_change is a plain vector, relative to
-    ``|original|`` if it is a scalar/vector interpreted that way by the
-    caller) and the result into ``[lb, ub]``.
+    """Move `original` towards `target` under clamping (reference
+    tools/misc.py:711): `max_change` is RELATIVE — the allowed change is
+    ``|original| * max_change`` — and it TIGHTENS [lb, ub] (so with a
+    degenerate intersection the upper bound wins, exactly like the
+    reference's max-then-min clamp order)."""
+    if lb is None and ub is None and max_change is None:
+        if in_place:
+            original[:] = target
+            return original
+        return target.clone()
 
-    Semantics follow the reference's `modify_tensor`
-    (/root/reference/src/evotorch/tools/misc.py:711): `max_change` is
-    *relative*: the allowed change is ``|original| * max_change``.
-    """
+    def to_tensor(x, default):
+        if x is None:
+            x = default
+        return torch.as_tensor(x, dtype=original.dtype, device=original.device)
+
+    lo = to_tensor(lb, float("-inf"))
+    hi = to_tensor(ub, float("inf"))
     if max_change is not None:
-        max_change = torch.as_tensor(max_change, dtype=original.dtype, device=original.device)
-        allowed = original.abs() * max_change
-        lo = original - allowed
-        hi = original + allowed
-        result = torch.clamp(target, lo, hi)
-    else:
-        result = target
-    if lb is not None:
-        result = torch.max(result, torch.as_tensor(lb, dtype=original.dtype, device=original.device))
-    if ub is not None:
-        result = torch.min(result, torch.as_tensor(ub, dtype=original.dtype, device=original.device))
+        allowed = original.abs() * to_tensor(max_change, 0.0)
+        lo = torch.max(lo, original - allowed)
+        hi = torch.min(hi, original + allowed)
+    result = torch.min(torch.max(target, lo), hi)
     if in_place:
-        original.copy_(result)
+        original[:] = result
         return original
     return result
-
 
 def modify_vector(*args, **kwargs) -> torch.Tensor:
     """Alias of :func:`modify_tensor` (reference parity)."""
