@@ -340,27 +340,35 @@ def modify_tensor(
 ) -> torch.Tensor:
     """Move `original` towards `target` under clamping (reference
     tools/misc.py:711): `max_change` is RELATIVE — the allowed change is
-    ``|original| * max_change`` — and it TIGHTENS [lb, ub] (so with a
-    degenerate intersection the upper bound wins, exactly like the
-    reference's max-then-min clamp order)."""
+    ``|original| * max_change`` — and it TIGHTENS [lb, ub], with the
+    reference's max-then-min order (the upper bound wins on a degenerate
+    intersection). hipGraph-capture-safe: python-scalar bounds stay
+    scalars (no host-to-device copies on the hot path)."""
     if lb is None and ub is None and max_change is None:
         if in_place:
             original[:] = target
             return original
         return target.clone()
 
-    def to_tensor(x, default):
-        if x is None:
-            x = default
-        return torch.as_tensor(x, dtype=original.dtype, device=original.device)
+    def scalar_or_tensor(x):
+        return x if (x is None or isinstance(x, torch.Tensor)) else float(x)
 
-    lo = to_tensor(lb, float("-inf"))
-    hi = to_tensor(ub, float("inf"))
+    lo = scalar_or_tensor(lb)
+    hi = scalar_or_tensor(ub)
     if max_change is not None:
-        allowed = original.abs() * to_tensor(max_change, 0.0)
-        lo = torch.max(lo, original - allowed)
-        hi = torch.min(hi, original + allowed)
-    result = torch.min(torch.max(target, lo), hi)
+        mc = max_change if isinstance(max_change, torch.Tensor) else float(max_change)
+        allowed = original.abs() * mc
+        lo_t = original - allowed
+        hi_t = original + allowed
+        if lo is not None:
+            lo_t = torch.max(lo_t, lo) if isinstance(lo, torch.Tensor) else torch.clamp(lo_t, min=lo)
+        if hi is not None:
+            hi_t = torch.min(hi_t, hi) if isinstance(hi, torch.Tensor) else torch.clamp(hi_t, max=hi)
+        result = torch.min(torch.max(target, lo_t), hi_t)
+    else:
+        # torch.clamp applies max(min_val) before min(max_val) — same
+        # upper-bound-wins semantics as the reference
+        result = torch.clamp(target, min=lo, max=hi)
     if in_place:
         original[:] = result
         return original
