@@ -261,3 +261,46 @@ def test_xnes_rotation_invariance_on_ellipsoid():
     searcher = XNES(prob, stdev_init=2.0, popsize=32)
     searcher.run(500)
     assert float(searcher.status["pop_best_eval"]) < 1e-2
+
+
+def test_random_config_chaos():
+    """Randomized searcher configurations (seeded) all construct and step
+    without crashing — guards kwarg plumbing across the whole family."""
+    import random
+
+    from evotorch_amd.algorithms import CEM, CMAES, XNES
+
+    random.seed(0)
+    for trial in range(25):
+        length = random.choice([2, 3, 7, 16, 33])
+        prob = Problem("min" if random.random() < 0.7 else "max", sphere,
+                       solution_length=length, initial_bounds=(-2, 2), seed=trial)
+        kind = random.choice(["pgpe", "snes", "cem", "xnes", "cmaes"])
+        if kind == "pgpe":
+            searcher = PGPE(prob, popsize=random.choice([4, 10, 25]) * 2,
+                            center_learning_rate=random.uniform(0.01, 0.5),
+                            stdev_learning_rate=random.uniform(0.01, 0.3),
+                            stdev_init=random.uniform(0.1, 3.0),
+                            symmetric=random.random() < 0.8,
+                            optimizer=random.choice(["clipup", "adam", "sgd", None]),
+                            ranking_method=random.choice(["centered", "linear", "nes", "raw", None]),
+                            distributed=random.random() < 0.4)
+        elif kind == "snes":
+            searcher = SNES(prob, stdev_init=random.uniform(0.1, 3.0),
+                            popsize=random.choice([None, 8, 30]),
+                            distributed=random.random() < 0.4)
+        elif kind == "cem":
+            searcher = CEM(prob, popsize=random.choice([20, 50]),
+                           parenthood_ratio=random.uniform(0.1, 0.5),
+                           stdev_init=random.uniform(0.5, 3.0),
+                           stdev_max_change=random.choice([None, 0.2, 1.0]))
+        elif kind == "xnes":
+            searcher = XNES(prob, stdev_init=random.uniform(0.5, 2.0), popsize=random.choice([None, 12]))
+        else:
+            searcher = CMAES(prob, stdev_init=random.uniform(0.5, 2.0),
+                             popsize=random.choice([None, 8, 24]),
+                             separable=random.random() < 0.3,
+                             active=random.random() < 0.7)
+        searcher.run(3)
+        assert searcher.step_count == 3
+        float(searcher.status["mean_eval"])
