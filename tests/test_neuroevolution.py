@@ -367,3 +367,20 @@ def test_vecenvne_max_num_envs_and_episodes():
     assert prob.max_num_envs == 4
     # averaged over 2 episodes of 5 steps each, pieces of <= 4 envs
     assert prob.interaction_count == 10 * 5 * 2
+
+
+def test_stateful_module_and_multilayered():
+    from evotorch_amd.models import MultiLayered, ensure_stateful
+    from evotorch_amd.models.layers import RNN
+
+    net = MultiLayered(RNN(4, 6), torch.nn.Linear(6, 2))
+    stateful = ensure_stateful(net)
+    x = torch.randn(4)
+    y1 = stateful(x)
+    assert y1.shape == (2,)
+    # hidden state threads: second call differs from a fresh reset
+    y2 = stateful(x)
+    stateful.reset()
+    y1_again = stateful(x)
+    assert torch.allclose(y1, y1_again, atol=1e-6)
+    assert not torch.allclose(y1, y2, atol=1e-6)

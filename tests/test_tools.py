@@ -217,3 +217,40 @@ def test_objectarray_method_parity():
     assert list(built[1]) == [6, 7]
     arr.set_item(0, [9])
     assert list(arr[0]) == [9]
+
+
+def test_tensor_maker_mixin_bound_factories():
+    """Problem's make_* factories are bound to its dtype/device/generator
+    (reference tools/tensormaker.py:27)."""
+    from evotorch_amd import Problem
+    from evotorch_amd.decorators import vectorized
+
+    @vectorized
+    def f(x):
+        return (x**2).sum(-1)
+
+    prob = Problem("min", f, solution_length=4, initial_bounds=(-1, 1), seed=3, dtype=torch.float64)
+    z = prob.make_zeros(5)
+    assert z.dtype == torch.float64 and z.shape == (5,)
+    g1 = prob.make_gaussian(6)
+    g2 = prob.make_gaussian(6)
+    assert g1.dtype == torch.float64
+    assert not torch.equal(g1, g2)  # generator advances
+    u = prob.make_uniform(4, lb=-2.0, ub=-1.0)
+    assert bool((u >= -2.0).all() and (u <= -1.0).all())
+    e = prob.make_empty(3)
+    assert e.shape == (3,)
+    i = prob.make_I(3)
+    assert torch.equal(i, torch.eye(3, dtype=torch.float64))
+
+
+def test_recursive_printable_and_profiling_noop():
+    from evotorch_amd.utils import CList
+    from evotorch_amd.utils.profiling import record_range
+
+    lst = CList(max_length=4, batch_size=2)
+    assert "CList" in repr(lst)
+
+    with record_range("cpu-noop"):  # no roctx on CPU: must be a clean no-op
+        x = torch.ones(3).sum()
+    assert float(x) == 3.0
