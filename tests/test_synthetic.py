@@ -119,3 +119,24 @@ def test_mlp_policy_eager_and_to_policy():
     act = policy(torch.randn(prob.spec.obs_dim))
     assert act.shape == (prob.spec.act_dim,)
     assert float(act.abs().max()) <= 1.0
+
+
+def test_pickling_logger_with_policy_export(tmp_path):
+    """PicklingLogger(make_policy_from=...) embeds a ready-to-run policy in
+    the checkpoint (reference logging.py:128-180)."""
+    from evotorch_amd.algorithms import PGPE
+    from evotorch_amd.logging import PicklingLogger
+    from evotorch_amd.neuroevolution import SyntheticRolloutProblem
+
+    prob = SyntheticRolloutProblem(seed=9, episode_length=5)
+    searcher = PGPE(prob, popsize=8, center_learning_rate=0.1,
+                    stdev_learning_rate=0.1, stdev_init=0.1)
+    logger = PicklingLogger(searcher, interval=2, directory=str(tmp_path),
+                            make_policy_from="center", verbose=False)
+    searcher.run(2)
+    payload = logger.unpickle_last_file()
+    policy = payload["policy"]
+    obs = torch.randn(prob._spec.obs_dim)
+    act = policy(obs)
+    assert act.shape == (prob._spec.act_dim,)
+    assert bool((act.abs() <= 1.0 + 1e-6).all())
