@@ -315,3 +315,39 @@ def test_ray_era_kwargs_accepted_with_warning():
     batch = prob.generate_batch(4)
     prob.evaluate(batch)
     assert batch.evals_are_ready
+
+
+def test_eval_data_roundtrip():
+    """eval_data columns ride along with fitnesses (reference core.py:
+    eval_data_length); both batch- and solution-level setters agree."""
+
+    @vectorized
+    def with_data(x):
+        fit = (x**2).sum(-1)
+        return fit, x[:, :2] * 10
+
+    prob = Problem("min", with_data, solution_length=4, initial_bounds=(-1, 1),
+                   eval_data_length=2, seed=44)
+    batch = prob.generate_batch(5)
+    prob.evaluate(batch)
+    evals = batch.access_evals()
+    assert evals.shape == (5, 3)  # 1 objective + 2 data columns
+    assert torch.allclose(evals[:, 1:], batch.unsafe_values[:, :2] * 10)
+
+    sol = batch[0]
+    assert sol.evals.shape == (3,)
+    sol.set_evals(torch.tensor([7.0]), torch.tensor([1.0, 2.0]))
+    assert batch.unsafe_evals[0].tolist() == [7.0, 1.0, 2.0]
+
+
+def test_batch_set_values_subset():
+    prob = make_problem(seed=45)
+    batch = prob.generate_batch(6)
+    prob.evaluate(batch)
+    new_rows = torch.zeros(2, batch.unsafe_values.shape[1])
+    batch.set_values(new_rows, solutions=torch.tensor([1, 3]))
+    assert bool((batch.unsafe_values[1] == 0).all())
+    assert bool((batch.unsafe_values[3] == 0).all())
+    # touched rows forget their evals, untouched rows keep them
+    assert torch.isnan(batch.unsafe_evals[1, 0])
+    assert not torch.isnan(batch.unsafe_evals[0, 0])
