@@ -237,3 +237,32 @@ def test_bench_entry_torchrun_world2(tmp_path):
     assert payload["config"]["global_batch"] == 32
     assert payload["config"]["parallelism"] == "dp2"
     assert payload["value"] > 0
+
+
+def test_bench_json_contract_single_process(tmp_path):
+    """bench.py's output line carries every field the driver contract
+    requires, with sane types (single-process CPU run)."""
+    import json
+    import os
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    result = subprocess.run(
+        [sys.executable, os.path.join(repo, "bench.py"), "--steps", "2", "--warmup", "1",
+         "--popsize-per-gpu", "16", "--episode-length", "4"],
+        capture_output=True, text=True, timeout=300,
+    )
+    assert result.returncode == 0, result.stderr[-2000:]
+    line = result.stdout.strip().splitlines()[-1]
+    d = json.loads(line)
+    for key, typ in [("metric", str), ("value", float), ("unit", str), ("n_gpus", int),
+                     ("steps", int), ("warmup", int), ("ms_per_step", float),
+                     ("higher_is_better", bool), ("scaling", str), ("dtype", str),
+                     ("data", str), ("config", dict)]:
+        assert key in d and isinstance(d[key], typ), (key, type(d.get(key)))
+    assert d["vs_baseline"] is None  # BASELINE.json publishes no numbers
+    assert d["steps"] == 2 and d["warmup"] == 1
+    assert d["scaling"] == "weak"
+    assert "Humanoid-v4 linear policy" in d["metric"]
+    assert d["config"]["global_batch"] == 16
