@@ -254,3 +254,16 @@ def test_recursive_printable_and_profiling_noop():
     with record_range("cpu-noop"):  # no roctx on CPU: must be a clean no-op
         x = torch.ones(3).sum()
     assert float(x) == 3.0
+
+
+def test_graft_entry_contract():
+    """__graft_entry__ exposes build() and smoke() (the driver contract)."""
+    import importlib.util
+    import os
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    spec = importlib.util.spec_from_file_location("graft_entry", os.path.join(repo, "__graft_entry__.py"))
+    mod = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(mod)
+    assert callable(mod.build)
+    assert callable(mod.smoke)
