@@ -838,6 +838,8 @@ class Problem(TensorMakerMixin, Serializable):
         length = self._solution_length
         lb = ensure_tensor_length_and_dtype(lb, length, self._dtype, about="lower bound", device=self._device)
         ub = ensure_tensor_length_and_dtype(ub, length, self._dtype, about="upper bound", device=self._device)
+        if bool((lb > ub).any()):
+            raise ValueError(f"Invalid bounds: lower bound exceeds upper bound ({lb} > {ub})")
         return lb, ub
 
     @property

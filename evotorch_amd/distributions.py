@@ -173,6 +173,12 @@ class SeparableGaussian(Distribution):
     PARAMETER_NDIMS = {"mu": 1, "sigma": 1}
 
     def __init__(self, parameters: dict, *, solution_length: Optional[int] = None, dtype=None, device=None):
+        missing = self.MANDATORY_PARAMETERS - set(parameters)
+        if missing:
+            raise ValueError(f"{type(self).__name__} is missing mandatory parameter(s): {sorted(missing)}")
+        unknown = set(parameters) - self.MANDATORY_PARAMETERS - self.OPTIONAL_PARAMETERS
+        if unknown:
+            raise ValueError(f"{type(self).__name__} got unknown parameter(s): {sorted(unknown)}")
         (mu_len,) = parameters["mu"].shape
         (sigma_len,) = parameters["sigma"].shape
         if mu_len != sigma_len:

@@ -351,3 +351,22 @@ def test_batch_set_values_subset():
     # touched rows forget their evals, untouched rows keep them
     assert torch.isnan(batch.unsafe_evals[1, 0])
     assert not torch.isnan(batch.unsafe_evals[0, 0])
+
+
+def test_clear_error_messages():
+    """Common misuses raise targeted errors, not obscure tracebacks."""
+    with pytest.raises(ValueError):
+        Problem("sideways", sphere, solution_length=4, initial_bounds=(-1, 1))  # bad sense
+    with pytest.raises(ValueError):
+        prob_bad = Problem("min", sphere, solution_length=4, initial_bounds=(3, -3))
+        prob_bad.generate_batch(2)
+    prob = make_problem()
+    with pytest.raises(IndexError):
+        prob.normalize_obj_index(5)
+    batch = prob.generate_batch(4)
+    with pytest.raises(IndexError):
+        batch[10]
+    with pytest.raises(ValueError):
+        from evotorch_amd.distributions import SeparableGaussian
+
+        SeparableGaussian({"mu": torch.zeros(3)})  # sigma missing
