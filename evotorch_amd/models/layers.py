@@ -228,7 +228,11 @@ class StructuredControlNet(nn.Module):
 
 class LocomotorNet(nn.Module):
     """StructuredControlNet variant whose nonlinear stream is a sum of
-    sinusoids (reference layers.py:487)."""
+    sinusoids (reference layers.py:487).
+
+    Note: the sinusoid phase advances an internal time buffer per forward
+    call, so this layer suits per-solution rollouts (GymNE-style); for
+    vmapped population rollouts use an explicit-state design instead."""
 
     def __init__(self, *, in_features: int, out_features: int, bias: bool = True, num_sinusoids: int = 16):
         super().__init__()
