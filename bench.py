@@ -36,7 +36,8 @@ def parse_args():
     p.add_argument("--steps", type=int, default=20)
     p.add_argument("--warmup", type=int, default=3)
     p.add_argument("--popsize-per-gpu", type=int, default=4000)
-    p.add_argument("--episode-length", type=int, default=200)
+    p.add_argument("--episode-length", type=int, default=1000,
+                   help="episode steps per rollout; 1000 matches the reference flagship config (BASELINE.md row 3)")
     p.add_argument("--seed", type=int, default=1)
     p.add_argument("--policy", choices=["linear", "mlp64"], default="linear",
                    help="linear = Humanoid-v4 linear policy (the headline metric); mlp64 = the paper's MLP-64-tanh brax config")
@@ -53,13 +54,11 @@ def main():
     rank = int(os.environ.get("RANK", "0"))
     have_gpu = torch.cuda.is_available()
 
-    comm = init_comm() if world > 1 else None
-    if comm is not None:
-        device = comm.device
-    elif have_gpu:
-        device = torch.device("cuda", 0)
-    else:
-        device = torch.device("cpu")
+    # one Comm for every world size: N=1 runs the same SPMD code path
+    # (counter-addressed sampling, global ranking, fused all-reduce with
+    # no-op collectives) as the 8-GPU scaling run
+    comm = init_comm()
+    device = comm.device
 
     total_popsize = args.popsize_per_gpu * world  # weak scaling
     radius_init = 2.25
@@ -73,8 +72,7 @@ def main():
         observation_normalization=True,
         policy_hidden=64 if args.policy == "mlp64" else 0,
     )
-    if comm is not None:
-        problem.use_comm(comm)
+    problem.use_comm(comm)
 
     searcher = PGPE(
         problem,
@@ -89,8 +87,7 @@ def main():
     )
 
     def sync():
-        if comm is not None:
-            comm.barrier()
+        comm.barrier()
         if have_gpu:
             torch.cuda.synchronize()
 
@@ -105,7 +102,7 @@ def main():
     elapsed = time.perf_counter() - t0
 
     # max over ranks
-    if comm is not None:
+    if world > 1:
         t = torch.tensor([elapsed], dtype=torch.float64, device=device if have_gpu else "cpu")
         comm.all_reduce_(t, op="max")
         elapsed = float(t)

@@ -105,6 +105,11 @@ def pgpe_tell(state: PGPEState, values: torch.Tensor, evals: torch.Tensor) -> PG
 
     n = values.shape[-2]
     if state.symmetric:
+        # zero-center raw/nes-style weights here too (same rule as the
+        # non-symmetric branch and the object API; round-1 ADVICE fix —
+        # uncentered weights bias the sigma gradient's (w⁺+w⁻)/2 terms)
+        if state.ranking_method not in ("centered", "normalized"):
+            weights = weights - weights.mean(dim=-1, keepdim=True)
         d = n // 2
         noises = values[..., :d, :] - center.unsqueeze(-2)
         w_plus = weights[..., :d]

@@ -189,6 +189,14 @@ class GaussianSearchAlgorithm(SearchAlgorithm, SinglePopulationAlgorithmMixin):
         # Adaptive popsize (Toklu et al. 2020): keep sampling sub-batches
         # until the interaction threshold is reached or popsize_max is hit
         # (reference core.py:3239-3274).
+        if not hasattr(problem, "last_eval_interaction_count"):
+            import warnings
+
+            warnings.warn(
+                f"num_interactions={self._num_interactions} is set but {type(problem).__name__} does not "
+                "report interaction counts (no `last_eval_interaction_count`); adaptive popsize is disabled.",
+                stacklevel=2,
+            )
         interactions = 0
         batches = []
         total = 0
@@ -553,4 +561,5 @@ class XNES(GaussianSearchAlgorithm):
             obj_index=obj_index,
             distributed=distributed,
             popsize_weighted_grad_avg=popsize_weighted_grad_avg,
+            grad_chunk_rows=grad_chunk_rows,
         )

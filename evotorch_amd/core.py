@@ -825,6 +825,10 @@ class Problem(TensorMakerMixin, Serializable):
 
         # SPMD comm (attached lazily; replaces the reference's actor pool)
         self._comm = None
+        self._spmd_seed_base = 0
+        self._spmd_gen_counter = 0
+        self._defer_fused_reduce = False
+        self._fused_reduce_requests: list = []
 
     # -- configuration properties ---------------------------------------------
 
@@ -1000,9 +1004,46 @@ class Problem(TensorMakerMixin, Serializable):
     def use_comm(self, comm) -> "Problem":
         """Attach an `evotorch_amd.parallel.Comm`; subsequent `evaluate`
         calls shard the population rows across its ranks and all-gather the
-        evals, and `sample_and_compute_gradients` runs in SPMD mode."""
+        evals, and `sample_and_compute_gradients` runs in SPMD mode.
+
+        A shared seed chain is established here (one int64 broadcast from
+        rank 0): SPMD population sampling is counter-addressed from this
+        chain, so the virtual population — and the whole search trajectory
+        for a deterministic fitness function — is IDENTICAL for any world
+        size that divides the popsize (SURVEY.md §7 "RNG discipline"). Rank
+        seeds need not match; rank 0's generator decides."""
+        from .ops.dispatch import _seed_from_generator
+
         self._comm = comm
+        base = _seed_from_generator(self._generator, self._device)
+        t = torch.tensor([base], dtype=torch.int64)
+        comm.broadcast_(t, src=0)
+        self._spmd_seed_base = int(t.item()) & 0x7FFFFFFFFFFFFFFF
+        self._spmd_gen_counter = 0
         return self
+
+    def _next_spmd_seed(self) -> int:
+        """Next seed of the shared chain (all ranks advance in lockstep)."""
+        from .ops.dispatch import _splitmix64
+
+        self._spmd_gen_counter += 1
+        return _splitmix64(self._spmd_seed_base ^ (self._spmd_gen_counter * 0x9E3779B97F4A7C15))
+
+    def request_fused_reduce(self, tensor: torch.Tensor, callback) -> None:
+        """All-reduce `tensor` across ranks and hand the result to
+        `callback`. During a sharded gradient generation the reduction is
+        DEFERRED and fused into the gradient all-reduce (one collective
+        carries gradients + obs-norm stats — the per-generation collective
+        count stays at 2); outside that window it runs immediately."""
+        comm = self._comm
+        if comm is None or comm.world_size <= 1:
+            callback(tensor)
+            return
+        if getattr(self, "_defer_fused_reduce", False):
+            self._fused_reduce_requests.append((tensor, callback))
+        else:
+            comm.all_reduce_(tensor)
+            callback(tensor)
 
     @property
     def comm(self):
@@ -1211,7 +1252,11 @@ class Problem(TensorMakerMixin, Serializable):
             if num_interactions is not None:
                 raise ValueError("chunk_rows (streaming gradients) does not support num_interactions adaptive popsize")
             result = self._sample_and_compute_gradients_streamed(distribution, int(popsize), obj_index, ranking_method, int(chunk_rows), comm)
-        elif comm is not None and comm.world_size > 1:
+        elif comm is not None:
+            # SPMD path for ANY world size (world 1 degenerates to local
+            # no-op collectives): one code path from the 1-GPU bench to the
+            # 8-GPU scaling run, counter-addressed so the trajectory is
+            # world-size invariant.
             result = self._sample_and_compute_gradients_sharded(distribution, int(popsize), obj_index, ranking_method, comm, num_interactions, popsize_max, ensure_even_popsize)
         else:
             result = self._sample_and_compute_gradients(distribution, int(popsize), obj_index, ranking_method, num_interactions, popsize_max, ensure_even_popsize)
@@ -1249,6 +1294,13 @@ class Problem(TensorMakerMixin, Serializable):
         values = batch._values
         fitnesses = batch._evals[:, obj_index]
         total_popsize = popsize
+        if num_interactions is not None and not hasattr(self, "last_eval_interaction_count") and not getattr(self, "_warned_no_interactions", False):
+            self._warned_no_interactions = True
+            _logger.warning(
+                "num_interactions=%s is set but %s does not report interaction counts "
+                "(no `last_eval_interaction_count`); the adaptive-popsize loop is disabled.",
+                num_interactions, type(self).__name__,
+            )
         if num_interactions is not None and hasattr(self, "last_eval_interaction_count"):
             # adaptive popsize (reference core.py:3239-3274): keep sampling
             # extra sub-batches until the interaction threshold is reached
@@ -1273,38 +1325,147 @@ class Problem(TensorMakerMixin, Serializable):
             "mean_eval": torch.nanmean(fitnesses),  # 0-dim tensor: no host sync
         }
 
+    def _eval_shard_batch(self, batch: "SolutionBatch"):
+        """Evaluate a rank-local shard with hooks/stats, deferring any
+        cross-rank stat reductions into the gradient all-reduce."""
+        self._before_eval_hook(batch)
+        self._defer_fused_reduce = True
+        try:
+            self._evaluate_batch(batch)
+            if self._store_solution_stats:
+                self._update_solution_stats(batch)
+            self._after_eval_status = self._after_eval_hook.accumulate_dict(batch)
+        finally:
+            self._defer_fused_reduce = False
+
+    def _fused_allreduce_and_finalize(self, comm, sums: dict) -> dict:
+        """ONE all-reduce carrying the gradient sums plus any deferred stat
+        reductions (obs-norm (count, Σ, Σ²) etc.) — per-generation
+        collective count stays at 2: fitness gather + this."""
+        pending = self._fused_reduce_requests
+        self._fused_reduce_requests = []
+        container = dict(sums)
+        for i, (tensor, _cb) in enumerate(pending):
+            container[f"__fused{i}"] = tensor
+        comm.all_reduce_container(container)
+        for i, (_tensor, cb) in enumerate(pending):
+            cb(container.pop(f"__fused{i}"))
+        return container
+
     def _sample_and_compute_gradients_sharded(
         self, distribution, popsize: int, obj_index: int, ranking_method, comm, num_interactions=None, popsize_max=None, ensure_even_popsize: bool = False
     ) -> dict:
-        world = comm.world_size
-        local_popsize = popsize // world
-        if ensure_even_popsize and local_popsize % 2 != 0:
-            local_popsize += 1
-        batch = self._get_cached_grad_batch(local_popsize, self._device)
-        distribution.sample(out=batch.access_values(), generator=self._generator)
-        self._before_eval_hook(batch)
-        self._evaluate_batch(batch)
-        if self._store_solution_stats:
-            self._update_solution_stats(batch)
-        self._after_eval_status = self._after_eval_hook.accumulate_dict(batch)
-        local_fit = batch._evals[:, obj_index]
-        # Global ranking: gather all fitnesses (N floats — latency-bound, tiny)
-        all_fit = comm.all_gather_vector(local_fit)
-        sense = self._senses[obj_index]
-        from .utils import ranking
+        """SPMD ES generation (the RCCL collapse of the reference's actor
+        round-trip, core.py:2762-3074 → SURVEY.md §3.3).
 
-        all_utils = ranking.rank(all_fit, ranking_method or "raw", higher_is_better=(sense == "max"))
-        my_utils = all_utils[comm.rank * local_popsize : (comm.rank + 1) * local_popsize].to(dtype=local_fit.dtype)
-        grads = distribution._compute_gradients(batch._values, my_utils, ranking_used=(ranking_method or "raw"))
-        # Partial gradients sum over local samples; merge & normalize by the
-        # global popsize with one fused all-reduce.
-        total = world * local_popsize
-        for k in grads:
-            grads[k] = grads[k] * (local_popsize / total)
-        comm.all_reduce_container(grads)
+        Rank r samples directions [r·D/W, (r+1)·D/W) of a GLOBAL virtual
+        population, counter-addressed from the shared seed chain — the
+        population (hence the trajectory, for deterministic fitness fns) is
+        identical for every world size W dividing the popsize. Utilities
+        are ranked globally after one all-gather of the fitness vector;
+        gradients are raw per-rank partial sums merged by one fused
+        all-reduce, with every weight-dependent normalization computed from
+        the GLOBAL utility vector (round-1 ADVICE medium fix).
+
+        Adaptive popsize (`num_interactions`, reference core.py:3239-3274):
+        all ranks keep sampling synchronized extra rounds until the GLOBAL
+        interaction count (one scalar all-reduce per round) reaches the
+        threshold or `popsize_max` is hit.
+        """
+        world = comm.world_size
+        rank = comm.rank
+        symmetric = bool(getattr(distribution, "_symmetric", False))
+        local_popsize = popsize // world
+        if (ensure_even_popsize or symmetric) and local_popsize % 2 != 0:
+            local_popsize += 1
+        dirs_local = local_popsize // 2 if symmetric else local_popsize
+        dirs_global = dirs_local * world
+        ranking_used = ranking_method or "raw"
+        sense = self._senses[obj_index]
+        from .utils import ranking as ranking_mod
+
+        if num_interactions is not None and not hasattr(self, "last_eval_interaction_count") and not getattr(self, "_warned_no_interactions", False):
+            self._warned_no_interactions = True
+            _logger.warning(
+                "num_interactions=%s is set but %s does not report interaction counts "
+                "(no `last_eval_interaction_count`); the adaptive-popsize loop is disabled.",
+                num_interactions, type(self).__name__,
+            )
+        if num_interactions is not None and not hasattr(self, "last_eval_interaction_count"):
+            num_interactions = None
+
+        round_values: List[torch.Tensor] = []
+        round_fits: List[torch.Tensor] = []
+        total_popsize = 0
+        interactions = 0
+        while True:
+            seed = self._next_spmd_seed()
+            if len(round_values) == 0:
+                batch = self._get_cached_grad_batch(local_popsize, self._device)
+            else:
+                batch = SolutionBatch(self, popsize=local_popsize, device=self._device, empty=True)
+            with record_range("sample"):
+                distribution.fill_counter_addressed(batch.access_values(), seed=seed, row_offset=rank * dirs_local)
+            with record_range("evaluate"):
+                self._eval_shard_batch(batch)
+            round_values.append(batch._values)
+            with record_range("fit_gather"):
+                round_fits.append(comm.all_gather_vector(batch._evals[:, obj_index].contiguous()))
+            total_popsize += local_popsize * world
+            if num_interactions is None:
+                break
+            local_count = int(getattr(self, "last_eval_interaction_count", 0) or 0)
+            count_t = torch.tensor([local_count], dtype=torch.int64)
+            comm.all_reduce_(count_t)
+            interactions += int(count_t.item())
+            if interactions >= int(num_interactions):
+                break
+            if popsize_max is not None and total_popsize >= int(popsize_max):
+                break
+
+        all_fit = round_fits[0] if len(round_fits) == 1 else torch.cat(round_fits)
+        all_utils = ranking_mod.rank(all_fit, ranking_used, higher_is_better=(sense == "max"))
+        elite_mode = "parenthood_ratio" in getattr(distribution, "parameters", {})
+
+        with record_range("partial_grads"):
+            if elite_mode:
+                num_elites = math.floor(total_popsize * float(distribution.parameters["parenthood_ratio"]))
+                global_elite = torch.zeros(total_popsize, dtype=torch.bool, device=all_utils.device)
+                global_elite[all_utils.argsort(descending=True)[:num_elites]] = True
+                # local-row elite mask: global row of my local row i in round
+                # k is k·local·world + rank·local + i (gather is rank-major)
+                mask = torch.zeros(len(round_values) * local_popsize, dtype=torch.bool, device=all_utils.device)
+                for k in range(len(round_values)):
+                    base = k * local_popsize * world + rank * local_popsize
+                    mask[k * local_popsize : (k + 1) * local_popsize] = global_elite[base : base + local_popsize]
+                sum_x, sum_x2 = distribution.accumulate_elite_sums_streamed(
+                    ((values, k * local_popsize, local_popsize) for k, values in enumerate(round_values)), mask
+                )
+                sums = {"x": sum_x, "x2": sum_x2}
+            else:
+                w_all = distribution.prepare_weights_global(all_utils, ranking_used)
+                w_dtype = self._dtype if self._dtype.is_floating_point else torch.float32
+                sums = None
+                for k, values in enumerate(round_values):
+                    base = k * local_popsize * world + rank * local_popsize
+                    my_w = w_all[base : base + local_popsize].to(dtype=w_dtype)
+                    part = distribution.partial_grad_sums(values, my_w)
+                    if sums is None:
+                        sums = part
+                    else:
+                        for key in sums:
+                            sums[key] = sums[key] + part[key]
+
+        with record_range("grad_allreduce"):
+            sums = self._fused_allreduce_and_finalize(comm, sums)
+
+        if elite_mode:
+            grads = distribution.finalize_elite_gradients(sums["x"].to(torch.float64), sums["x2"].to(torch.float64), num_elites)
+        else:
+            grads = distribution.finalize_shard_gradients(sums, w_all, ranking_used)
         return {
             "gradients": grads,
-            "num_solutions": total,
+            "num_solutions": total_popsize,
             "mean_eval": torch.nanmean(all_fit),  # 0-dim tensor: no host sync
         }
 
@@ -1323,8 +1484,16 @@ class Problem(TensorMakerMixin, Serializable):
     def _sample_and_compute_gradients_streamed(
         self, distribution, popsize: int, obj_index: int, ranking_method, chunk_rows: int, comm
     ) -> dict:
-        """Two-pass streaming ES gradients — see sample_and_compute_gradients."""
-        if not hasattr(distribution, "compute_gradients_streamed"):
+        """Two-pass streaming ES gradients — see sample_and_compute_gradients.
+
+        Uses the same shard-gradient protocol as the non-streamed SPMD path
+        (prepare weights globally → raw partial sums per chunk → one fused
+        all-reduce → finalize with global normalizers), with the chunk axis
+        playing the role of extra shards. Rank r owns directions
+        [r·D/W, (r+1)·D/W) of the global virtual population, addressed by
+        per-row philox streams from the shared seed chain — regenerable in
+        pass 2 without storing the N×L population."""
+        if not hasattr(distribution, "fill_counter_addressed"):
             raise ValueError(f"{type(distribution).__name__} does not support streamed gradients")
         elite_mode = "parenthood_ratio" in getattr(distribution, "parameters", {})
         from .ops.dispatch import _seed_from_generator
@@ -1337,13 +1506,10 @@ class Problem(TensorMakerMixin, Serializable):
         if symmetric and local_popsize % 2 != 0:
             local_popsize += 1
         directions = local_popsize // 2 if symmetric else local_popsize
-        length = self._solution_length
-        # philox counters cover 4 elements each: chunk starts must land on
-        # 4-element boundaries of the flat (direction-major) noise stream
-        if length % 4 != 0 and chunk_rows % 4 != 0:
-            chunk_rows = max(4, (chunk_rows // 4) * 4)
+        my_dir0 = my_rank * directions
+        # stream-per-row philox addressing: any chunk boundary is exact
         chunk_rows = min(chunk_rows, directions)
-        seed = _seed_from_generator(self._generator, self._device)
+        seed = self._next_spmd_seed() if comm is not None else _seed_from_generator(self._generator, self._device)
 
         fits = torch.empty(local_popsize, dtype=self._eval_dtype, device=self._device)
 
@@ -1358,13 +1524,9 @@ class Problem(TensorMakerMixin, Serializable):
         for r0, rows in chunk_spans():
             batch = self._get_stream_batch(rows * 2 if symmetric else rows)
             with record_range("stream_sample"):
-                distribution.fill_counter_addressed(batch.access_values(), seed=seed, elem_offset=r0 * length)
+                distribution.fill_counter_addressed(batch.access_values(), seed=seed, row_offset=my_dir0 + r0)
             with record_range("stream_eval"):
-                self._before_eval_hook(batch)
-                self._evaluate_batch(batch)
-                if self._store_solution_stats:
-                    self._update_solution_stats(batch)
-                self._after_eval_status = self._after_eval_hook.accumulate_dict(batch)
+                self._eval_shard_batch(batch)
             evals = batch._evals[:, obj_index]
             if symmetric:
                 fits[r0 : r0 + rows] = evals[:rows]
@@ -1378,46 +1540,53 @@ class Problem(TensorMakerMixin, Serializable):
         else:
             all_fit = fits
         sense = self._senses[obj_index]
-        all_utils = ranking_mod.rank(all_fit, ranking_method or "raw", higher_is_better=(sense == "max"))
-        my_utils = all_utils[my_rank * local_popsize : (my_rank + 1) * local_popsize].to(dtype=self._dtype if self._dtype.is_floating_point else torch.float32)
+        ranking_used = ranking_method or "raw"
+        all_utils = ranking_mod.rank(all_fit, ranking_used, higher_is_better=(sense == "max"))
+        total = world * local_popsize
 
         # -- pass 2: regenerate noise per chunk, accumulate exact gradients --
         def regen_chunks():
             for r0, rows in chunk_spans():
                 batch = self._get_stream_batch(rows * 2 if symmetric else rows)
                 with record_range("stream_regen"):
-                    distribution.fill_counter_addressed(batch.access_values(), seed=seed, elem_offset=r0 * length)
+                    distribution.fill_counter_addressed(batch.access_values(), seed=seed, row_offset=my_dir0 + r0)
                 yield batch._values, r0, rows
 
         with record_range("stream_grad"):
-            if elite_mode and comm is not None and world > 1:
-                # sharded CEM: elite SET chosen from the GLOBAL utilities;
-                # each rank accumulates its local elite members' (Σx, Σx²)
-                # and one all-reduce of the sums (not a weighted gradient
-                # average — elite stats are non-linear in the shards)
-                import math as _math
-
-                total = world * local_popsize
-                num_elites = _math.floor(total * float(distribution.parameters["parenthood_ratio"]))
+            if elite_mode:
+                # elite SET chosen from the GLOBAL utilities; each rank
+                # accumulates its local elite members' (Σx, Σx²); the sums
+                # (not a weighted gradient average — elite stats are
+                # non-linear in the shards) go through the fused all-reduce
+                num_elites = math.floor(total * float(distribution.parameters["parenthood_ratio"]))
                 global_elite = torch.zeros(total, dtype=torch.bool, device=all_utils.device)
                 global_elite[all_utils.argsort(descending=True)[:num_elites]] = True
                 my_elite = global_elite[my_rank * local_popsize : (my_rank + 1) * local_popsize]
                 sum_x, sum_x2 = distribution.accumulate_elite_sums_streamed(regen_chunks(), my_elite)
                 sums = {"x": sum_x, "x2": sum_x2}
-                comm.all_reduce_container(sums)
+                if comm is not None:
+                    sums = self._fused_allreduce_and_finalize(comm, sums)
                 grads = distribution.finalize_elite_gradients(
                     sums["x"].to(torch.float64), sums["x2"].to(torch.float64), num_elites)
             else:
-                grads = distribution.compute_gradients_streamed(regen_chunks(), my_utils, ranking_used=(ranking_method or "raw"))
-                if comm is not None and world > 1:
-                    total = world * local_popsize
-                    for k in grads:
-                        grads[k] = grads[k] * (local_popsize / total)
-                    comm.all_reduce_container(grads)
-                else:
-                    total = local_popsize
-        if comm is None or world <= 1:
-            total = local_popsize
+                w_all = distribution.prepare_weights_global(all_utils, ranking_used)
+                w_dtype = self._dtype if self._dtype.is_floating_point else torch.float32
+                my_w = w_all[my_rank * local_popsize : (my_rank + 1) * local_popsize].to(dtype=w_dtype)
+                sums = None
+                for values_chunk, r0, rows in regen_chunks():
+                    if symmetric:
+                        w_c = torch.cat([my_w[r0 : r0 + rows], my_w[directions + r0 : directions + r0 + rows]])
+                    else:
+                        w_c = my_w[r0 : r0 + rows]
+                    part = distribution.partial_grad_sums(values_chunk, w_c)
+                    if sums is None:
+                        sums = part
+                    else:
+                        for key in sums:
+                            sums[key] = sums[key] + part[key]
+                if comm is not None:
+                    sums = self._fused_allreduce_and_finalize(comm, sums)
+                grads = distribution.finalize_shard_gradients(sums, w_all, ranking_used)
         return {
             "gradients": grads,
             "num_solutions": total,

@@ -126,6 +126,34 @@ class Distribution(TensorMakerMixin):
     def _compute_gradients(self, samples: torch.Tensor, weights: torch.Tensor, ranking_used: Optional[str]) -> dict:
         raise NotImplementedError
 
+    # -- SPMD shard-gradient protocol ----------------------------------------
+    #
+    # The sharded path (Problem._sample_and_compute_gradients_sharded) splits
+    # the weight-dependent gradient into three phases so every normalization
+    # uses GLOBAL quantities (a per-shard normalization would silently change
+    # the merged gradient — ADVICE.md round-1 medium finding):
+    #
+    #   w_all = prepare_weights_global(all_utils, ranking)   # global center/normalize
+    #   sums  = partial_grad_sums(local_samples, w_local)    # raw sums, linear in w
+    #   <all-reduce the sums across ranks>
+    #   grads = finalize_shard_gradients(sums, w_all, ranking)  # global divisors
+    #
+    # partial_grad_sums MUST be additive over row blocks of the population so
+    # the all-reduced sum equals the single-process gradient exactly.
+
+    def prepare_weights_global(self, all_weights: torch.Tensor, ranking_used: Optional[str]) -> torch.Tensor:
+        """Weight preprocessing computed over the GLOBAL utility vector."""
+        return all_weights
+
+    def partial_grad_sums(self, samples: torch.Tensor, weights: torch.Tensor) -> dict:
+        """Raw, un-normalized gradient sums over a row block (linear in the
+        weights; additive across blocks)."""
+        raise NotImplementedError
+
+    def finalize_shard_gradients(self, sums: dict, all_weights: torch.Tensor, ranking_used: Optional[str]) -> dict:
+        """Apply weight-dependent normalization using GLOBAL weights."""
+        return sums
+
     # -- updates ------------------------------------------------------------
 
     def _follow_gradient(self, param_name: str, grad: torch.Tensor, *, learning_rates: Optional[dict] = None, optimizers: Optional[dict] = None) -> torch.Tensor:
@@ -210,16 +238,17 @@ class SeparableGaussian(Distribution):
 
         ops.sample_gaussian(out, self.mu, self.sigma, symmetric=self._symmetric, generator=generator)
 
-    def fill_counter_addressed(self, out: torch.Tensor, *, seed: int, elem_offset: int = 0):
-        """Sample a row-block of the virtual population: noise element e is
-        philox(seed, (elem_offset+e)//4), identical however the population
-        is chunked (and identical on CPU and GPU). Enables the streaming
-        large-L gradient path (Problem.sample_and_compute_gradients with
-        chunk_rows) to REGENERATE noise in pass 2 instead of storing the
-        N×L population."""
+    def fill_counter_addressed(self, out: torch.Tensor, *, seed: int, row_offset: int = 0):
+        """Sample a row-block of the virtual population: direction row d
+        draws from philox stream `row_offset + d`, identical however the
+        population is partitioned (and identical on CPU and GPU). This is
+        what makes (a) the streaming large-L gradient path able to
+        REGENERATE noise in pass 2 instead of storing the N×L population,
+        and (b) SPMD sharded sampling produce the same virtual population
+        for any world size."""
         from . import ops
 
-        ops.sample_gaussian(out, self.mu, self.sigma, symmetric=self._symmetric, seed=seed, elem_offset=elem_offset)
+        ops.sample_gaussian(out, self.mu, self.sigma, symmetric=self._symmetric, seed=seed, row_offset=row_offset)
 
     def accumulate_elite_sums_streamed(self, chunk_iter, is_elite: torch.Tensor):
         """Masked (Σx, Σx²) over elite rows, chunk by chunk (fp64 accum).
@@ -244,50 +273,6 @@ class SeparableGaussian(Distribution):
         return {
             "mu": (mean - self.mu.to(torch.float64)).to(self.mu.dtype),
             "sigma": (std - self.sigma.to(torch.float64)).to(self.sigma.dtype),
-        }
-
-    def _compute_elite_gradients_streamed(self, chunk_iter, weights: torch.Tensor) -> dict:
-        """Single-population form: elite set from the full utility vector."""
-        num_samples = weights.shape[0]
-        num_elites = math.floor(num_samples * float(self._parameters["parenthood_ratio"]))
-        elite_idx = weights.argsort(descending=True)[:num_elites]
-        is_elite = torch.zeros(num_samples, dtype=torch.bool, device=weights.device)
-        is_elite[elite_idx] = True
-        sum_x, sum_x2 = self.accumulate_elite_sums_streamed(chunk_iter, is_elite)
-        return self.finalize_elite_gradients(sum_x, sum_x2, num_elites)
-
-    def compute_gradients_streamed(self, chunk_iter, weights: torch.Tensor, *, ranking_used: Optional[str]) -> dict:
-        """Gradients over a chunked (never-materialized) population.
-
-        `chunk_iter` yields (values_chunk, row0, rows) where row0/rows index
-        DIRECTIONS (= plus-rows for the symmetric subclass, plain rows
-        otherwise); `weights` is the full local utility vector. Weighted
-        sums are additive over row-blocks, so raw sums accumulate per chunk
-        and the (global, weight-dependent) normalization is applied once at
-        the end — exactly equal to the unstreamed gradient.
-        """
-        from . import ops
-
-        if "parenthood_ratio" in self._parameters:
-            return self._compute_elite_gradients_streamed(chunk_iter, weights)
-
-        w = self._centered_weights(weights, ranking_used)
-        d = w.shape[0] // 2 if self._symmetric else w.shape[0]
-        mu_acc = torch.zeros_like(self.mu, dtype=torch.float32)
-        sigma_acc = torch.zeros_like(self.sigma, dtype=torch.float32)
-        for values_chunk, row0, rows in chunk_iter:
-            if self._symmetric:
-                w_c = torch.cat([w[row0 : row0 + rows], w[d + row0 : d + row0 + rows]])
-            else:
-                w_c = w[row0 : row0 + rows]
-            g_mu, g_sigma = ops.es_gradients(values_chunk, self.mu, self.sigma, w_c, symmetric=self._symmetric)
-            mu_acc += g_mu.to(torch.float32)
-            sigma_acc += g_sigma.to(torch.float32)
-        mu_grad = mu_acc.to(self.mu.dtype)
-        sigma_grad = sigma_acc.to(self.sigma.dtype)
-        return {
-            "mu": self._divide_grad("mu", mu_grad, w),
-            "sigma": self._divide_grad("sigma", sigma_grad, w),
         }
 
     @classmethod
@@ -340,6 +325,23 @@ class SeparableGaussian(Distribution):
         return {
             "mu": self._divide_grad("mu", mu_grad, weights),
             "sigma": self._divide_grad("sigma", sigma_grad, weights),
+        }
+
+    # -- SPMD shard-gradient protocol (see Distribution) ---------------------
+
+    def prepare_weights_global(self, all_weights: torch.Tensor, ranking_used: Optional[str]) -> torch.Tensor:
+        return self._centered_weights(all_weights, ranking_used)
+
+    def partial_grad_sums(self, samples: torch.Tensor, weights: torch.Tensor) -> dict:
+        from . import ops
+
+        mu_grad, sigma_grad = ops.es_gradients(samples, self.mu, self.sigma, weights, symmetric=self._symmetric)
+        return {"mu": mu_grad.to(torch.float32), "sigma": sigma_grad.to(torch.float32)}
+
+    def finalize_shard_gradients(self, sums: dict, all_weights: torch.Tensor, ranking_used: Optional[str]) -> dict:
+        return {
+            "mu": self._divide_grad("mu", sums["mu"].to(self.mu.dtype), all_weights),
+            "sigma": self._divide_grad("sigma", sums["sigma"].to(self.sigma.dtype), all_weights),
         }
 
     def update_parameters(self, gradients: dict, *, learning_rates: Optional[dict] = None, optimizers: Optional[dict] = None) -> "SeparableGaussian":
@@ -418,53 +420,23 @@ class ExpSeparableGaussian(SeparableGaussian):
         mu_grad, sigma_grad = ops.snes_gradients(samples, self.mu, self.sigma, weights)
         return {"mu": mu_grad, "sigma": sigma_grad}
 
-    def accumulate_elite_sums_streamed(self, chunk_iter, is_elite: torch.Tensor):
-        """Masked (Σx, Σx²) over elite rows, chunk by chunk (fp64 accum).
-        `is_elite` flags LOCAL rows; shards all-reduce the returned sums."""
-        sum_x = torch.zeros_like(self.mu, dtype=torch.float64)
-        sum_x2 = torch.zeros_like(self.mu, dtype=torch.float64)
-        for values_chunk, row0, rows in chunk_iter:
-            mask = is_elite[row0 : row0 + rows]
-            if bool(mask.any()):
-                selected = values_chunk[mask].to(torch.float64)
-                sum_x += selected.sum(dim=0)
-                sum_x2 += (selected**2).sum(dim=0)
-        return sum_x, sum_x2
+    # -- SPMD shard-gradient protocol: SNES normalizes by the GLOBAL
+    # |w| sum (the round-1 per-shard normalization bug) and applies no
+    # divisors afterwards.
 
-    def finalize_elite_gradients(self, sum_x: torch.Tensor, sum_x2: torch.Tensor, num_elites: int) -> dict:
-        """Turn global elite (Σx, Σx², k) into the CEM gradients —
-        algebraically identical to `_elite_gradients`."""
-        k = max(int(num_elites), 1)
-        mean = sum_x / k
-        var = (sum_x2 - k * mean**2) / max(k - 1, 1)
-        std = torch.sqrt(torch.clamp(var, min=0.0))
-        return {
-            "mu": (mean - self.mu.to(torch.float64)).to(self.mu.dtype),
-            "sigma": (std - self.sigma.to(torch.float64)).to(self.sigma.dtype),
-        }
+    def prepare_weights_global(self, all_weights: torch.Tensor, ranking_used: Optional[str]) -> torch.Tensor:
+        if ranking_used != "nes":
+            all_weights = all_weights / all_weights.abs().sum()
+        return all_weights
 
-    def _compute_elite_gradients_streamed(self, chunk_iter, weights: torch.Tensor) -> dict:
-        """Single-population form: elite set from the full utility vector."""
-        num_samples = weights.shape[0]
-        num_elites = math.floor(num_samples * float(self._parameters["parenthood_ratio"]))
-        elite_idx = weights.argsort(descending=True)[:num_elites]
-        is_elite = torch.zeros(num_samples, dtype=torch.bool, device=weights.device)
-        is_elite[elite_idx] = True
-        sum_x, sum_x2 = self.accumulate_elite_sums_streamed(chunk_iter, is_elite)
-        return self.finalize_elite_gradients(sum_x, sum_x2, num_elites)
-
-    def compute_gradients_streamed(self, chunk_iter, weights: torch.Tensor, *, ranking_used: Optional[str]) -> dict:
+    def partial_grad_sums(self, samples: torch.Tensor, weights: torch.Tensor) -> dict:
         from . import ops
 
-        if ranking_used != "nes":
-            weights = weights / weights.abs().sum()
-        mu_acc = torch.zeros_like(self.mu, dtype=torch.float32)
-        sigma_acc = torch.zeros_like(self.sigma, dtype=torch.float32)
-        for values_chunk, row0, rows in chunk_iter:
-            g_mu, g_sigma = ops.snes_gradients(values_chunk, self.mu, self.sigma, weights[row0 : row0 + rows])
-            mu_acc += g_mu.to(torch.float32)
-            sigma_acc += g_sigma.to(torch.float32)
-        return {"mu": mu_acc.to(self.mu.dtype), "sigma": sigma_acc.to(self.sigma.dtype)}
+        mu_grad, sigma_grad = ops.snes_gradients(samples, self.mu, self.sigma, weights)
+        return {"mu": mu_grad.to(torch.float32), "sigma": sigma_grad.to(torch.float32)}
+
+    def finalize_shard_gradients(self, sums: dict, all_weights: torch.Tensor, ranking_used: Optional[str]) -> dict:
+        return {"mu": sums["mu"].to(self.mu.dtype), "sigma": sums["sigma"].to(self.sigma.dtype)}
 
     def update_parameters(self, gradients: dict, *, learning_rates: Optional[dict] = None, optimizers: Optional[dict] = None) -> "ExpSeparableGaussian":
         new_mu = self.mu + self._follow_gradient("mu", gradients["mu"], learning_rates=learning_rates, optimizers=optimizers)
@@ -523,6 +495,19 @@ class ExpGaussian(Distribution):
         out.normal_(generator=generator)
         out.copy_(self.to_global_coordinates(out))
 
+    def fill_counter_addressed(self, out: torch.Tensor, *, seed: int, row_offset: int = 0):
+        """Counter-addressed full-covariance sampling: row r draws its
+        standard normals from philox stream `row_offset + r` and maps them
+        through A — world-size-invariant like the separable family (the
+        SPMD sharded path requires this entry point)."""
+        from . import ops
+
+        z = torch.empty_like(out)
+        zeros = torch.zeros_like(self.mu)
+        ones = torch.ones_like(self.mu)
+        ops.sample_gaussian(z, zeros, ones, symmetric=False, seed=seed, row_offset=row_offset)
+        out.copy_(self.to_global_coordinates(z))
+
     def _compute_gradients(self, samples: torch.Tensor, weights: torch.Tensor, ranking_used: Optional[str]) -> dict:
         z = self.to_local_coordinates(samples)
         if ranking_used not in ("centered", "normalized"):
@@ -532,6 +517,24 @@ class ExpGaussian(Distribution):
         # MFMA-shaped (K5/K6 in SURVEY.md §2.9), served by rocBLAS.
         M_grad = (z * weights.unsqueeze(-1)).T @ z - weights.sum() * self.eye
         return {"d": d_grad, "M": M_grad}
+
+    # -- SPMD shard-gradient protocol: both d and M are linear in w, so raw
+    # local sums all-reduce exactly; centering uses the GLOBAL mean.
+
+    def prepare_weights_global(self, all_weights: torch.Tensor, ranking_used: Optional[str]) -> torch.Tensor:
+        if ranking_used not in ("centered", "normalized"):
+            all_weights = all_weights - all_weights.mean()
+        return all_weights
+
+    def partial_grad_sums(self, samples: torch.Tensor, weights: torch.Tensor) -> dict:
+        z = self.to_local_coordinates(samples)
+        d_grad = torch.mv(z.T, weights)
+        M_grad = (z * weights.unsqueeze(-1)).T @ z - weights.sum() * self.eye
+        return {"d": d_grad.to(torch.float32), "M": M_grad.to(torch.float32)}
+
+    def finalize_shard_gradients(self, sums: dict, all_weights: torch.Tensor, ranking_used: Optional[str]) -> dict:
+        dt = self.mu.dtype
+        return {"d": sums["d"].to(dt), "M": sums["M"].to(dt)}
 
     def update_parameters(self, gradients: dict, *, learning_rates: Optional[dict] = None, optimizers: Optional[dict] = None) -> "ExpGaussian":
         learning_rates = dict(learning_rates or {})

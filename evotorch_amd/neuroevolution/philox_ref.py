@@ -6,7 +6,7 @@ integer level."""
 import numpy as np
 import torch
 
-__all__ = ["philox4x32_10", "philox_normal_rows", "philox_normals"]
+__all__ = ["philox4x32_10", "philox_normal_rows", "philox_normals", "philox_normals_2d"]
 
 _M0 = np.uint32(0xD2511F53)
 _M1 = np.uint32(0xCD9E8D57)
@@ -68,6 +68,25 @@ def philox_normals(seed: int, stream_id: int, n_elements: int, *, idx4_offset: i
     c3 = np.zeros(n4, dtype=np.uint32)
     r0, r1, r2, r3 = philox4x32_10(c0, c1, c2, c3, np.uint32(seed & 0xFFFFFFFF), np.uint32((seed >> 32) & 0xFFFFFFFF))
     z = _box_muller4(r0, r1, r2, r3).reshape(-1)[:n_elements]
+    return torch.from_numpy(np.ascontiguousarray(z))
+
+
+def philox_normals_2d(seed: int, row_offset: int, rows: int, length: int) -> torch.Tensor:
+    """(rows, length) normals with STREAM-PER-ROW addressing: row r draws
+    from stream `row_offset + r`, counter c covering columns [4c, 4c+4).
+    Matches the K1 counter-addressed sampling kernel
+    (ops/hip/es_kernels.hip::sample_gaussian_kernel) for any row partition
+    of a virtual population: generating rows [a, b) here equals slicing
+    rows [a, b) of the full population."""
+    len4 = (length + 3) // 4
+    col4 = np.tile(np.arange(len4, dtype=np.uint64), rows)
+    stream = np.repeat(np.arange(row_offset, row_offset + rows, dtype=np.uint64), len4)
+    c0 = col4.astype(np.uint32)
+    c1 = (col4 >> np.uint64(32)).astype(np.uint32)
+    c2 = stream.astype(np.uint32)
+    c3 = np.zeros_like(c2)
+    r0, r1, r2, r3 = philox4x32_10(c0, c1, c2, c3, np.uint32(seed & 0xFFFFFFFF), np.uint32((seed >> 32) & 0xFFFFFFFF))
+    z = _box_muller4(r0, r1, r2, r3).reshape(rows, len4 * 4)[:, :length]
     return torch.from_numpy(np.ascontiguousarray(z))
 
 

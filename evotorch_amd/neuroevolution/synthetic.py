@@ -134,7 +134,10 @@ class SyntheticRolloutProblem(Problem):
 
     def _merge_pending_stats(self):
         """Fold the rollout's (count, Σ, Σ²) into the running norm — with an
-        attached Comm this is ONE all-reduce across ranks (P5)."""
+        attached Comm this is ONE all-reduce across ranks (P5), and during a
+        sharded gradient generation it is FUSED into the gradient all-reduce
+        (Problem.request_fused_reduce) so a generation costs exactly two
+        collectives: fitness all-gather + fused all-reduce."""
         if self._pending_stats is None or not self._obs_norm_enabled:
             self._pending_stats = None
             return
@@ -143,11 +146,15 @@ class SyntheticRolloutProblem(Problem):
         comm = self._comm
         if comm is not None and comm.world_size > 1:
             packed = torch.cat([torch.tensor([count], dtype=torch.float32, device=s.device), s.reshape(-1), ss.reshape(-1)])
-            comm.all_reduce_(packed)
-            count = packed[0]  # stays a device tensor: no host sync
-            s = packed[1 : 1 + self._spec.obs_dim]
-            ss = packed[1 + self._spec.obs_dim :]
-        self._obs_norm.update((count, s, ss))
+            O = self._spec.obs_dim
+
+            def merge(reduced: torch.Tensor):
+                # count stays a device tensor: no host sync
+                self._obs_norm.update((reduced[0], reduced[1 : 1 + O], reduced[1 + O :]))
+
+            self.request_fused_reduce(packed, merge)
+        else:
+            self._obs_norm.update((count, s, ss))
 
     # -- policy export -------------------------------------------------------
 

@@ -232,8 +232,13 @@ class VecEnvNE(NEProblem):
         if comm is not None and comm.world_size > 1:
             c, s, ss = pending.stats_triple()
             packed = torch.cat([c.to(s.device, s.dtype).reshape(1), s.reshape(-1), ss.reshape(-1)])
-            comm.all_reduce_(packed)
-            self._obs_norm.update((packed[0], packed[1 : 1 + self._obs_dim], packed[1 + self._obs_dim :]))
+            # During a sharded gradient generation this reduction is fused
+            # into the gradient all-reduce (Problem.request_fused_reduce):
+            # one collective carries gradients + (count, Σ, Σ²).
+            def merge(reduced: torch.Tensor):
+                self._obs_norm.update((reduced[0], reduced[1 : 1 + self._obs_dim], reduced[1 + self._obs_dim :]))
+
+            self.request_fused_reduce(packed, merge)
         else:
             self._obs_norm.update(pending)
 

@@ -126,38 +126,34 @@ def sample_gaussian(
     symmetric: bool = False,
     generator: Optional[torch.Generator] = None,
     seed: Optional[int] = None,
-    elem_offset: int = 0,
+    row_offset: int = 0,
 ) -> torch.Tensor:
     """Fill `out` (N×L) with x = mu + sigma * z. With symmetric=True, rows
     [0, N/2) hold mu + sigma*z and rows [N/2, N) the mirrored mu - sigma*z
     (halves layout — see evotorch_amd/distributions.py docstring).
 
-    Counter-addressed mode (`seed` given): noise element e of the virtual
-    full population is philox(seed, counter=(elem_offset+e)//4) — the SAME
-    values regardless of how the population is chunked into row-blocks, and
-    identical on CPU (numpy philox reference) and GPU. This is what the
-    streaming large-L gradient path uses to regenerate noise instead of
-    storing it; `elem_offset` must be a multiple of 4 and counts elements of
-    the first (non-mirrored) half."""
+    Counter-addressed mode (`seed` given): row r of `out` (a direction row
+    for symmetric sampling) draws from philox stream `row_offset + r` with
+    the counter walking the row — the SAME values regardless of how the
+    virtual population is partitioned into row blocks (sharding across
+    ranks, or chunking in the streaming large-L gradient path), and
+    identical on CPU (numpy philox reference) and GPU."""
     if out.ndim != 2:
         raise ValueError(f"expected a 2-D population, got shape {tuple(out.shape)}")
     n = out.shape[0]
     if symmetric and n % 2 != 0:
         raise ValueError("symmetric sampling requires even popsize")
     if seed is not None:
-        if elem_offset % 4 != 0:
-            raise ValueError("elem_offset must be a multiple of 4")
         if out.device.type == "cuda" and not _allow_eager_on_gpu():
             mod = hip_required()
-            mod.sample_gaussian(out, mu.to(out.dtype), sigma.to(out.dtype), bool(symmetric), int(seed), int(elem_offset))
+            mod.sample_gaussian(out, mu.to(out.dtype), sigma.to(out.dtype), bool(symmetric), int(seed), int(row_offset))
             return out
         # philox-exact eager reference (matches the kernel bit-for-bit in fp32)
-        from ..neuroevolution.philox_ref import philox_normals
+        from ..neuroevolution.philox_ref import philox_normals_2d
 
         rows = n // 2 if symmetric else n
         length = out.shape[1]
-        z = philox_normals(int(seed), 0, rows * length, idx4_offset=elem_offset // 4)
-        z = z.reshape(rows, length).to(device=out.device)
+        z = philox_normals_2d(int(seed), int(row_offset), rows, length).to(device=out.device)
         mu32 = mu.to(torch.float32)
         sigma32 = sigma.to(torch.float32)
         plus = (mu32 + sigma32 * z).to(out.dtype)

@@ -2,7 +2,7 @@
 #include <torch/extension.h>
 
 namespace ea {
-void sample_gaussian(torch::Tensor out, torch::Tensor mu, torch::Tensor sigma, bool symmetric, int64_t seed, int64_t elem_offset);
+void sample_gaussian(torch::Tensor out, torch::Tensor mu, torch::Tensor sigma, bool symmetric, int64_t seed, int64_t row_offset);
 void sample_gaussian_graphsafe(torch::Tensor out, torch::Tensor mu, torch::Tensor sigma, bool symmetric,
                                torch::Tensor seed_buf);
 std::vector<torch::Tensor> es_gradients(torch::Tensor samples, torch::Tensor mu, torch::Tensor sigma,
@@ -23,7 +23,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.doc() = "evotorch_amd gfx950 (MI355X / CDNA4) HIP kernels";
     m.def("sample_gaussian", &ea::sample_gaussian, "K1: philox Gaussian population sampling (plain/antithetic)",
           py::arg("out"), py::arg("mu"), py::arg("sigma"), py::arg("symmetric"), py::arg("seed"),
-          py::arg("elem_offset") = 0);
+          py::arg("row_offset") = 0);
     m.def("sample_gaussian_graphsafe", &ea::sample_gaussian_graphsafe,
           "K1 (hipGraph-safe): seed read from and advanced in device memory");
     m.def("es_gradients", &ea::es_gradients, "K3: fused (mu, sigma) ES gradient reduction");

@@ -50,30 +50,40 @@ __global__ void bump_seed_kernel(unsigned long long* seed_ptr) {
     *seed_ptr = (z ^ (z >> 31)) & 0x7FFFFFFFFFFFFFFFull;
 }
 
+// Counter addressing is STREAM-PER-ROW: row r of the (virtual, possibly
+// rank-sharded) population draws from philox stream `row_offset + r` with
+// the counter walking the row's columns (counter c covers columns
+// [4c, 4c+4)). Any row partition of the population is therefore
+// regenerable exactly, for ANY solution length — the property the SPMD
+// sharded sampling and the streaming large-L gradient path both rely on
+// (SURVEY.md §7 "RNG discipline"). CPU reference:
+// evotorch_amd/neuroevolution/philox_ref.py::philox_normals_2d.
 template <typename T, bool kSymmetric>
 __global__ void sample_gaussian_kernel(T* __restrict__ out, const T* __restrict__ mu, const T* __restrict__ sigma,
                                        int64_t rows,  // = N (plain) or N/2 (symmetric)
                                        int64_t length, uint64_t seed_in,
                                        const unsigned long long* __restrict__ seed_ptr,
-                                       uint64_t elem_offset4) {
+                                       int64_t row_offset) {
     const uint64_t seed = resolve_seed(seed_in, seed_ptr);
-    const int64_t total4 = (rows * length + 3) / 4;
+    const int64_t len4 = (length + 3) / 4;
+    const int64_t total4 = rows * len4;
     for (int64_t idx4 = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; idx4 < total4;
          idx4 += (int64_t)gridDim.x * blockDim.x) {
+        const int64_t row = idx4 / len4;
+        const int64_t col4 = idx4 - row * len4;
         float z[4];
-        philox_normal4(seed, 0u, (uint64_t)idx4 + elem_offset4, z);
-        const int64_t base = idx4 * 4;
+        philox_normal4(seed, (uint32_t)(row + row_offset), (uint64_t)col4, z);
+        const int64_t base_col = col4 * 4;
 #pragma unroll
         for (int j = 0; j < 4; ++j) {
-            const int64_t e = base + j;
-            if (e >= rows * length) break;
-            const int64_t col = e % length;
+            const int64_t col = base_col + j;
+            if (col >= length) break;
             const float m = static_cast<float>(mu[col]);
             const float s = static_cast<float>(sigma[col]);
             const float plus = fmaf(s, z[j], m);
-            out[e] = static_cast<T>(plus);
+            out[row * length + col] = static_cast<T>(plus);
             if (kSymmetric) {
-                out[e + rows * length] = static_cast<T>(2.0f * m - plus);
+                out[(row + rows) * length + col] = static_cast<T>(2.0f * m - plus);
             }
         }
     }
@@ -87,14 +97,15 @@ template <bool kSymmetric>
 __global__ void sample_gaussian_f32x4_kernel(float4* __restrict__ out, const float4* __restrict__ mu,
                                              const float4* __restrict__ sigma, int64_t rows, int64_t length4,
                                              uint64_t seed_in, const unsigned long long* __restrict__ seed_ptr,
-                                             uint64_t elem_offset4) {
+                                             int64_t row_offset) {
     const uint64_t seed = resolve_seed(seed_in, seed_ptr);
     const int64_t total4 = rows * length4;
     for (int64_t idx4 = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; idx4 < total4;
          idx4 += (int64_t)gridDim.x * blockDim.x) {
+        const int64_t row = idx4 / length4;
+        const int64_t col4 = idx4 - row * length4;
         float z[4];
-        philox_normal4(seed, 0u, (uint64_t)idx4 + elem_offset4, z);
-        const int64_t col4 = idx4 % length4;
+        philox_normal4(seed, (uint32_t)(row + row_offset), (uint64_t)col4, z);
         const float4 m = mu[col4];
         const float4 s = sigma[col4];
         float4 plus;
@@ -115,15 +126,15 @@ __global__ void sample_gaussian_f32x4_kernel(float4* __restrict__ out, const flo
 }
 
 void sample_gaussian_impl(torch::Tensor out, torch::Tensor mu, torch::Tensor sigma, bool symmetric, int64_t seed,
-                          const unsigned long long* seed_ptr, int64_t elem_offset = 0) {
-    TORCH_CHECK(elem_offset % 4 == 0, "elem_offset must be a multiple of 4 (philox counter granularity)");
-    const uint64_t elem_offset4 = (uint64_t)(elem_offset / 4);
+                          const unsigned long long* seed_ptr, int64_t row_offset = 0) {
+    TORCH_CHECK(row_offset >= 0, "row_offset must be non-negative");
     CHECK_GPU(out); CHECK_CONTIG(out); CHECK_GPU(mu); CHECK_GPU(sigma);
     const int64_t n = out.size(0), length = out.size(1);
     TORCH_CHECK(!symmetric || n % 2 == 0, "symmetric sampling needs even popsize");
     const int64_t rows = symmetric ? n / 2 : n;
     const int threads = 256;
-    const int64_t total4 = (rows * length + 3) / 4;
+    const int64_t len4 = (length + 3) / 4;
+    const int64_t total4 = rows * len4;
     // 2048 blocks (8/CU) saturate: deeper grids measured identical (the
     // SQ_WAIT:BUSY ~12:1 is philox->Box-Muller dependency latency plus the
     // write pipe, not occupancy starvation)
@@ -133,11 +144,11 @@ void sample_gaussian_impl(torch::Tensor out, torch::Tensor mu, torch::Tensor sig
         if (symmetric) {
             hipLaunchKernelGGL((sample_gaussian_f32x4_kernel<true>), dim3(blocks), dim3(threads), 0, stream,
                                reinterpret_cast<float4*>(out.data_ptr<float>()), reinterpret_cast<const float4*>(mu.data_ptr<float>()),
-                               reinterpret_cast<const float4*>(sigma.data_ptr<float>()), rows, length / 4, (uint64_t)seed, seed_ptr, elem_offset4);
+                               reinterpret_cast<const float4*>(sigma.data_ptr<float>()), rows, length / 4, (uint64_t)seed, seed_ptr, row_offset);
         } else {
             hipLaunchKernelGGL((sample_gaussian_f32x4_kernel<false>), dim3(blocks), dim3(threads), 0, stream,
                                reinterpret_cast<float4*>(out.data_ptr<float>()), reinterpret_cast<const float4*>(mu.data_ptr<float>()),
-                               reinterpret_cast<const float4*>(sigma.data_ptr<float>()), rows, length / 4, (uint64_t)seed, seed_ptr, elem_offset4);
+                               reinterpret_cast<const float4*>(sigma.data_ptr<float>()), rows, length / 4, (uint64_t)seed, seed_ptr, row_offset);
         }
         return;
     }
@@ -145,17 +156,17 @@ void sample_gaussian_impl(torch::Tensor out, torch::Tensor mu, torch::Tensor sig
         using T = scalar_t;
         if (symmetric) {
             hipLaunchKernelGGL((sample_gaussian_kernel<T, true>), dim3(blocks), dim3(threads), 0, stream,
-                               out.data_ptr<T>(), mu.data_ptr<T>(), sigma.data_ptr<T>(), rows, length, (uint64_t)seed, seed_ptr, elem_offset4);
+                               out.data_ptr<T>(), mu.data_ptr<T>(), sigma.data_ptr<T>(), rows, length, (uint64_t)seed, seed_ptr, row_offset);
         } else {
             hipLaunchKernelGGL((sample_gaussian_kernel<T, false>), dim3(blocks), dim3(threads), 0, stream,
-                               out.data_ptr<T>(), mu.data_ptr<T>(), sigma.data_ptr<T>(), rows, length, (uint64_t)seed, seed_ptr, elem_offset4);
+                               out.data_ptr<T>(), mu.data_ptr<T>(), sigma.data_ptr<T>(), rows, length, (uint64_t)seed, seed_ptr, row_offset);
         }
     });
 }
 
 void sample_gaussian(torch::Tensor out, torch::Tensor mu, torch::Tensor sigma, bool symmetric, int64_t seed,
-                     int64_t elem_offset) {
-    sample_gaussian_impl(out, mu, sigma, symmetric, seed, nullptr, elem_offset);
+                     int64_t row_offset) {
+    sample_gaussian_impl(out, mu, sigma, symmetric, seed, nullptr, row_offset);
 }
 
 // Graph-safe variant: the seed lives in `seed_buf` (int64 tensor of 1
