@@ -329,29 +329,32 @@ class TestMlpRollout:
 
 @requires_gpu
 def test_counter_addressed_sampling_matches_cpu_reference():
-    """GPU counter-addressed sampling (seed + elem_offset) is bit-equal in
-    fp32 to the numpy philox reference, for any chunking."""
+    """GPU counter-addressed sampling (seed + row_offset, stream-per-row
+    philox) is bit-equal in fp32 to the numpy philox reference, for any row
+    partition — including an odd solution length (L % 4 != 0), where the
+    old flat addressing could not align chunk starts."""
     from evotorch_amd.ops import sample_gaussian
 
-    L, N = 12, 8
-    torch.manual_seed(0)
-    mu = torch.randn(L, device="cuda:0")
-    sigma = torch.rand(L, device="cuda:0") + 0.5
+    for L in (12, 7):  # multiple-of-4 fast path and odd-length tail path
+        N = 8
+        torch.manual_seed(0)
+        mu = torch.randn(L, device="cuda:0")
+        sigma = torch.rand(L, device="cuda:0") + 0.5
 
-    full = torch.empty(N, L, device="cuda:0")
-    sample_gaussian(full, mu, sigma, symmetric=True, seed=777)
+        full = torch.empty(N, L, device="cuda:0")
+        sample_gaussian(full, mu, sigma, symmetric=True, seed=777)
 
-    cpu_full = torch.empty(N, L)
-    sample_gaussian(cpu_full, mu.cpu(), sigma.cpu(), symmetric=True, seed=777)
-    assert torch.allclose(full.cpu(), cpu_full, atol=1e-6), (full.cpu() - cpu_full).abs().max()
+        cpu_full = torch.empty(N, L)
+        sample_gaussian(cpu_full, mu.cpu(), sigma.cpu(), symmetric=True, seed=777)
+        assert torch.allclose(full.cpu(), cpu_full, atol=1e-6), (full.cpu() - cpu_full).abs().max()
 
-    # chunked regeneration equals the whole on device
-    d = N // 2
-    for r0, rows in [(0, 1), (1, 3)]:
-        chunk = torch.empty(rows * 2, L, device="cuda:0")
-        sample_gaussian(chunk, mu, sigma, symmetric=True, seed=777, elem_offset=r0 * L)
-        assert torch.equal(chunk[:rows], full[r0 : r0 + rows])
-        assert torch.equal(chunk[rows:], full[d + r0 : d + r0 + rows])
+        # chunked regeneration equals the whole on device
+        d = N // 2
+        for r0, rows in [(0, 1), (1, 3)]:
+            chunk = torch.empty(rows * 2, L, device="cuda:0")
+            sample_gaussian(chunk, mu, sigma, symmetric=True, seed=777, row_offset=r0)
+            assert torch.equal(chunk[:rows], full[r0 : r0 + rows])
+            assert torch.equal(chunk[rows:], full[d + r0 : d + r0 + rows])
 
 
 @requires_gpu
