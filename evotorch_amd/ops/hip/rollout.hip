@@ -446,6 +446,10 @@ static void launch_rollout(int n_blocks, int block, size_t lds_bytes, hipStream_
     hipLaunchKernelGGL((rollout_linear_kernel<kGroup, kMembers>), dim3(n_blocks), dim3(block), lds_bytes, stream, args);
 }
 
+void rollout_v7(torch::Tensor params, torch::Tensor env_blob, torch::Tensor obs_stats_out, torch::Tensor fitness,
+                int64_t obs_dim, int64_t act_dim, int64_t rank, int64_t steps, double alive_bonus, double act_cost,
+                int64_t init_seed, int64_t member_offset);  // rollout_v7.hip
+
 torch::Tensor rollout_linear(torch::Tensor params, torch::Tensor env_blob, torch::Tensor obs_stats_out,
                              int64_t obs_dim, int64_t act_dim, int64_t rank, int64_t steps, double alive_bonus,
                              double act_cost, int64_t init_seed, int64_t member_offset, int64_t policy_hidden) {
@@ -459,6 +463,13 @@ torch::Tensor rollout_linear(torch::Tensor params, torch::Tensor env_blob, torch
     TORCH_CHECK(O % 2 == 0, "obs_dim must be even (bf16x2 packing)");
     TORCH_CHECK(H % 2 == 0, "policy_hidden must be even (bf16x2 packing)");
     auto fitness = torch::empty({n}, params.options());
+    // v7 (MFMA, 16 members/block) serves the linear flagship geometry;
+    // v6 covers MLP policies and off-geometry envs.
+    if (H == 0 && R == 16 && O <= 384 && A <= 17 && !getenv("EVOTORCH_AMD_ROLLOUT_V6")) {
+        rollout_v7(params, env_blob, obs_stats_out, fitness, obs_dim, act_dim, rank, steps, alive_bonus, act_cost,
+                   init_seed, member_offset);
+        return fitness;
+    }
 
     RolloutArgs args;
     args.params = params.data_ptr<float>();

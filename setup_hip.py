@@ -25,6 +25,7 @@ setup(
                 os.path.join(SRC, "bindings.cpp"),
                 os.path.join(SRC, "es_kernels.hip"),
                 os.path.join(SRC, "rollout.hip"),
+                os.path.join(SRC, "rollout_v7.hip"),
                 os.path.join(SRC, "pareto.hip"),
             ],
             extra_compile_args={
