@@ -465,7 +465,7 @@ torch::Tensor rollout_linear(torch::Tensor params, torch::Tensor env_blob, torch
     auto fitness = torch::empty({n}, params.options());
     // v7 (MFMA, 16 members/block) serves the linear flagship geometry;
     // v6 covers MLP policies and off-geometry envs.
-    if (H == 0 && R == 16 && O <= 384 && A <= 17 && !getenv("EVOTORCH_AMD_ROLLOUT_V6")) {
+    if (H == 0 && R == 16 && O == 376 && A == 17 && !getenv("EVOTORCH_AMD_ROLLOUT_V6")) {
         rollout_v7(params, env_blob, obs_stats_out, fitness, obs_dim, act_dim, rank, steps, alive_bonus, act_cost,
                    init_seed, member_offset);
         return fitness;

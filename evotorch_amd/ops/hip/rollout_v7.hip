@@ -54,15 +54,19 @@ struct RolloutV7Args {
     unsigned long long init_seed;
 };
 
-// Compile-time geometry: O padded to OP (multiple of 128 so 8 waves split
-// OP/16 tiles evenly), R == 16 (one GEMM1 tile), A <= 31 (fits hact K=64).
-template <int OP, int A_MAX>
+// Compile-time geometry (all guards constant-fold; a runtime-guarded
+// W-load was measured to demote w_frag to scratch memory): O padded to OP
+// (multiple of 128 so 8 waves split OP/16 tiles evenly), R == 16 (one
+// GEMM1 tile), A <= 31 (fits hact K=64).
+template <int O, int A>
 __global__ __launch_bounds__(512, 1) void rollout_v7_kernel(RolloutV7Args args) {
     constexpr int kMembers = 16;
+    constexpr int OP = (O + 127) / 128 * 128;
+    constexpr int A_MAX = A;
+    constexpr int R = 16;
     constexpr int kTiles = OP / 16;           // GEMM2 output tiles
     constexpr int kTilesPerWave = kTiles / 8; // = 3 at OP=384
     constexpr int kChunk = OP / 32;           // policy obs columns per lane (12)
-    const int O = args.obs_dim, A = args.act_dim, R = args.rank;
     const int tid = threadIdx.x;
     const int lane = tid & 63;
     const int wave = tid >> 6;
@@ -82,6 +86,8 @@ __global__ __launch_bounds__(512, 1) void rollout_v7_kernel(RolloutV7Args args) 
     float* istd_l = mean_l + OP;                         // [OP]
     float* wave_fit = istd_l + OP;                       // [8][16] per-wave fitness partials
     float* actsq_l = wave_fit + 8 * 16;                  // [16]
+    __bf16* v_l = reinterpret_cast<__bf16*>(actsq_l + 16);  // [16][OP] V (GEMM1 B-operand)
+    __bf16* ud_l = v_l + kMembers * OP;                  // [OP][64] k-major [U;D2] (GEMM2 B)
 
     const long RO = (long)R * O, AO = (long)A * O;
     const float* eV = args.env_blob;
@@ -107,44 +113,27 @@ __global__ __launch_bounds__(512, 1) void rollout_v7_kernel(RolloutV7Args args) 
     }
     if (tid < kMembers) actsq_l[tid] = 0.0f;
 
-    // ---- static B-operand register fragments --------------------------------
-    // GEMM1 (wave 0 only): B[k=o][col=r] = V[r][o]; 12 k-slices of 32.
-    bf16x8_t v_frag[OP / 32];
-    if (wave == 0) {
-        const int col = lane & 15;           // r
-        const int k0 = (lane >> 4) * 8;      // k within slice
-#pragma unroll
-        for (int s = 0; s < OP / 32; ++s) {
-#pragma unroll
-            for (int i = 0; i < 8; ++i) {
-                const int o = s * 32 + k0 + i;
-                v_frag[s][i] = (col < R && o < O) ? f2b7(eV[(long)col * O + o]) : f2b7(0.0f);
-            }
-        }
+    // ---- static B-operands ---------------------------------------------------
+    // GEMM1 B (V) lives in LDS — keeping it in registers on every wave
+    // (it is only used by wave 0) was what pushed the kernel over the
+    // 256-VGPR budget and demoted w_frag to scratch.
+    for (int j = tid; j < kMembers * OP; j += 512) {
+        const int r = j / OP, o = j % OP;
+        v_l[j] = (r < R && o < O) ? f2b7(eV[(long)r * O + o]) : f2b7(0.0f);
     }
-    // GEMM2: B[k][col=o] = k<16 ? U_T[k][o] : (k-16<A ? D2_T[k-16][o] : 0);
-    // per wave: kTilesPerWave tiles × 2 k-slices.
-    bf16x8_t ud_frag[kTilesPerWave][2];
-    {
-        const int col_in_tile = lane & 15;
-        const int k0 = (lane >> 4) * 8;
-#pragma unroll
-        for (int tw = 0; tw < kTilesPerWave; ++tw) {
-            const int o = (wave * kTilesPerWave + tw) * 16 + col_in_tile;
-#pragma unroll
-            for (int s = 0; s < 2; ++s) {
-#pragma unroll
-                for (int i = 0; i < 8; ++i) {
-                    const int k = s * 32 + k0 + i;
-                    float v = 0.0f;
-                    if (o < O) {
-                        if (k < R) v = eU[(long)k * O + o];
-                        else if (k - R < A) v = eD2[(long)(k - R) * O + o];
-                    }
-                    ud_frag[tw][s][i] = f2b7(v);
-                }
-            }
+    // GEMM2 B ([U;D2]) in LDS, k-major [o][k] so a lane's 8-consecutive-k
+    // fragment is one aligned 16 B read. (Register-resident B-fragments
+    // for BOTH GEMMs pushed past the 256-VGPR/8-wave cap and demoted the
+    // policy weights to scratch — LDS B costs ~49 KB/step of bandwidth
+    // but keeps w_frag in registers, which dominates.)
+    for (int j = tid; j < OP * 64; j += 512) {
+        const int o = j / 64, k = j % 64;
+        float v = 0.0f;
+        if (o < O) {
+            if (k < R) v = eU[(long)k * O + o];
+            else if (k - R < A) v = eD2[(long)(k - R) * O + o];
         }
+        ud_l[j] = f2b7(v);
     }
 
     // ---- per-member policy weights in registers -----------------------------
@@ -156,7 +145,7 @@ __global__ __launch_bounds__(512, 1) void rollout_v7_kernel(RolloutV7Args args) 
     {
         const float* W = args.params + (long)(base_member + min(my_member, live - 1)) * ((long)A * O + A);
         const int cbase = l32 * kChunk;
-#pragma unroll 4
+#pragma unroll
         for (int a = 0; a < A_MAX; ++a) {
 #pragma unroll
             for (int p = 0; p < kChunk / 2; ++p) {
@@ -216,7 +205,7 @@ __global__ __launch_bounds__(512, 1) void rollout_v7_kernel(RolloutV7Args args) 
 #pragma unroll
                 for (int p = 0; p < kChunk / 2; ++p) {
                     const bf16x2_t o2 = *reinterpret_cast<const bf16x2_t*>(on + 2 * p);
-#pragma unroll 4
+#pragma unroll
                     for (int a = 0; a < A_MAX; ++a) {
                         acc[a] = __builtin_amdgcn_fdot2_f32_bf16(w_frag[a][p], o2, acc[a], false);
                     }
@@ -224,12 +213,12 @@ __global__ __launch_bounds__(512, 1) void rollout_v7_kernel(RolloutV7Args args) 
             }
 #pragma unroll
             for (int off = 16; off > 0; off >>= 1) {
-#pragma unroll 4
+#pragma unroll
                 for (int a = 0; a < A_MAX; ++a) acc[a] += __shfl_down(acc[a], off, 32);
             }
             if (l32 == 0 && my_member < live) {
                 float sq = 0.0f;
-#pragma unroll 4
+#pragma unroll
                 for (int a = 0; a < A_MAX; ++a) {
                     if (a >= A) break;
                     const float av = fminf(fmaxf(acc[a] + b_l[my_member * A_MAX + a], -1.0f), 1.0f);
@@ -245,7 +234,8 @@ __global__ __launch_bounds__(512, 1) void rollout_v7_kernel(RolloutV7Args args) 
 #pragma unroll
             for (int s = 0; s < OP / 32; ++s) {
                 const bf16x8_t a_frag = *reinterpret_cast<const bf16x8_t*>(obs_l + g2_row * OP + s * 32 + g2_k0);
-                acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, v_frag[s], acc, 0, 0, 0);
+                const bf16x8_t b_frag = *reinterpret_cast<const bf16x8_t*>(v_l + c_col * OP + s * 32 + g2_k0);
+                acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, b_frag, acc, 0, 0, 0);
             }
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
@@ -262,9 +252,12 @@ __global__ __launch_bounds__(512, 1) void rollout_v7_kernel(RolloutV7Args args) 
             floatx4_t out[kTilesPerWave];
 #pragma unroll
             for (int tw = 0; tw < kTilesPerWave; ++tw) {
+                const int o = (wave * kTilesPerWave + tw) * 16 + c_col;
+                const bf16x8_t b0 = *reinterpret_cast<const bf16x8_t*>(ud_l + o * 64 + g2_k0);
+                const bf16x8_t b1 = *reinterpret_cast<const bf16x8_t*>(ud_l + o * 64 + 32 + g2_k0);
                 floatx4_t acc = {0.f, 0.f, 0.f, 0.f};
-                acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, ud_frag[tw][0], acc, 0, 0, 0);
-                acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, ud_frag[tw][1], acc, 0, 0, 0);
+                acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc, 0, 0, 0);
+                acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b1, acc, 0, 0, 0);
                 out[tw] = acc;
             }
 #pragma unroll
@@ -351,17 +344,18 @@ void rollout_v7(torch::Tensor params, torch::Tensor env_blob, torch::Tensor obs_
     args.act_cost = (float)act_cost;
     args.init_seed = (unsigned long long)init_seed;
 
-    constexpr int OP = 384, A_MAX = 17;
-    const size_t lds = (size_t)(2 * 16 * OP + 16 * 64) * 2 + (size_t)(16 * A_MAX + 4 * OP + 8 * 16 + 16) * 4;
+    constexpr int O_T = 376, A_T = 17, OP = 384;
+    TORCH_CHECK(O == O_T && A == A_T, "rollout v7 instantiated for the Humanoid geometry (obs 376, act 17)");
+    const size_t lds = (size_t)(3 * 16 * OP + 16 * 64 + OP * 64) * 2 + (size_t)(16 * A_T + 4 * OP + 8 * 16 + 16) * 4;
     static bool attr_set7 = false;
     if (!attr_set7) {
-        (void)hipFuncSetAttribute(reinterpret_cast<const void*>(&rollout_v7_kernel<OP, A_MAX>),
+        (void)hipFuncSetAttribute(reinterpret_cast<const void*>(&rollout_v7_kernel<O_T, A_T>),
                                   hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
         attr_set7 = true;
     }
     auto stream = at::cuda::getCurrentCUDAStream();
     const int blocks = (n + 15) / 16;
-    hipLaunchKernelGGL((rollout_v7_kernel<OP, A_MAX>), dim3(blocks), dim3(512), lds, stream, args);
+    hipLaunchKernelGGL((rollout_v7_kernel<O_T, A_T>), dim3(blocks), dim3(512), lds, stream, args);
 }
 
 }  // namespace ea
