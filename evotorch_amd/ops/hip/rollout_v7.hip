@@ -42,6 +42,35 @@ typedef __attribute__((ext_vector_type(4))) float floatx4_t;
 __device__ __forceinline__ __bf16 f2b7(float v) { return (__bf16)v; }
 __device__ __forceinline__ float b2f7(__bf16 v) { return (float)v; }
 
+// tanh via the hardware exp unit (v_exp_f32): ~1e-6 relative error —
+// far below the bf16 quantization (2^-8) every value passes through.
+// libm tanhf was 12 calls/lane/step of branchy code on the hot path.
+__device__ __forceinline__ float tanh_fast(float x) {
+    const float xc = fminf(fmaxf(x, -15.0f), 15.0f);
+    const float e = __expf(2.0f * xc);
+    return (e - 1.0f) / (e + 1.0f);
+}
+
+// One step of a 32-lane sum reduce on the VALU pipe via DPP modifiers —
+// __shfl_down lowers to ds_bpermute_b32 (LDS pipe), and the policy
+// phase's 17 accumulators × 5 levels were 85 LDS-pipe ops per lane per
+// step, fighting the dot-product LDS reads. After row_shr 1/2/4/8 +
+// row_bcast:15 the 32-lane totals sit in lanes 31 and 63.
+template <int kCtrl>
+__device__ __forceinline__ float dpp_add(float x) {
+    const int moved = __builtin_amdgcn_update_dpp(0, __builtin_bit_cast(int, x), kCtrl, 0xf, 0xf, true);
+    return x + __builtin_bit_cast(float, moved);
+}
+
+__device__ __forceinline__ float reduce32_dpp(float x) {
+    x = dpp_add<0x111>(x);  // row_shr:1
+    x = dpp_add<0x112>(x);  // row_shr:2
+    x = dpp_add<0x114>(x);  // row_shr:4
+    x = dpp_add<0x118>(x);  // row_shr:8
+    x = dpp_add<0x142>(x);  // row_bcast:15 → lanes 31/63 hold the 32-lane sums
+    return x;
+}
+
 struct RolloutV7Args {
     const float* params;     // [n][A*O + A]
     const float* env_blob;   // V[R][O] U_T[R][O] D2_T[A][O] c[O] wr[O] mean[O] std[O]
@@ -212,11 +241,8 @@ __global__ __launch_bounds__(512, 1) void rollout_v7_kernel(RolloutV7Args args) 
                 }
             }
 #pragma unroll
-            for (int off = 16; off > 0; off >>= 1) {
-#pragma unroll
-                for (int a = 0; a < A_MAX; ++a) acc[a] += __shfl_down(acc[a], off, 32);
-            }
-            if (l32 == 0 && my_member < live) {
+            for (int a = 0; a < A_MAX; ++a) acc[a] = reduce32_dpp(acc[a]);
+            if (l32 == 31 && my_member < live) {
                 float sq = 0.0f;
 #pragma unroll
                 for (int a = 0; a < A_MAX; ++a) {
@@ -272,7 +298,7 @@ __global__ __launch_bounds__(512, 1) void rollout_v7_kernel(RolloutV7Args args) 
 #pragma unroll
                 for (int r = 0; r < 4; ++r) {
                     const int m = c_row0 + r;
-                    const float o_new = tanhf(out[tw][r] + cv);
+                    const float o_new = tanh_fast(out[tw][r] + cv);
                     fit_part[r] = fmaf(wrv, o_new, fit_part[r]);
                     if (m < live && col_in) {
                         ssum += o_new;
@@ -303,7 +329,7 @@ __global__ __launch_bounds__(512, 1) void rollout_v7_kernel(RolloutV7Args args) 
 #pragma unroll
         for (int r = 0; r < 4; ++r) wave_fit[wave * 16 + c_row0 + r] = fit_part[r];
     }
-    if (l32 == 0 && my_member < live) actsq_l[my_member] = actsq_total;
+    if (l32 == 31 && my_member < live) actsq_l[my_member] = actsq_total;
     __syncthreads();
     if (tid < live) {
         float total = 0.0f;
