@@ -93,6 +93,10 @@ __global__ __launch_bounds__(512, 1) void rollout_v7_kernel(RolloutV7Args args) 
     constexpr int OP = (O + 127) / 128 * 128;
     constexpr int A_MAX = A;
     constexpr int R = 16;
+    constexpr int OPS = OP + 8;               // obs-row LDS stride: breaks the
+                                              // 768 B ≡ 0 (mod 64 dwords) bank
+                                              // aliasing of 16-row fragment reads
+    constexpr int KS = 72;                    // hact/ud row stride (vs 64), same reason
     constexpr int kTiles = OP / 16;           // GEMM2 output tiles
     constexpr int kTilesPerWave = kTiles / 8; // = 3 at OP=384
     constexpr int kChunk = OP / 32;           // policy obs columns per lane (12)
@@ -106,9 +110,9 @@ __global__ __launch_bounds__(512, 1) void rollout_v7_kernel(RolloutV7Args args) 
     // ---- LDS ----
     extern __shared__ unsigned char lds7[];
     __bf16* obs_l = reinterpret_cast<__bf16*>(lds7);     // [16][OP] raw (quantized) obs
-    __bf16* obsn_l = obs_l + kMembers * OP;              // [16][OP] normalized obs
-    __bf16* hact_l = obsn_l + kMembers * OP;             // [16][64]: k<16 h, 16..16+A act, rest 0
-    float* b_l = reinterpret_cast<float*>(hact_l + kMembers * 64);  // [16][A_MAX] bias
+    __bf16* obsn_l = obs_l + kMembers * OPS;             // [16][OPS] normalized obs
+    __bf16* hact_l = obsn_l + kMembers * OPS;            // [16][KS]: k<16 h, 16..16+A act, rest 0
+    float* b_l = reinterpret_cast<float*>(hact_l + kMembers * KS);  // [16][A_MAX] bias
     float* c_l = b_l + kMembers * A_MAX;                 // [OP]
     float* wr_l = c_l + OP;                              // [OP]
     float* mean_l = wr_l + OP;                           // [OP]
@@ -116,7 +120,7 @@ __global__ __launch_bounds__(512, 1) void rollout_v7_kernel(RolloutV7Args args) 
     float* wave_fit = istd_l + OP;                       // [8][16] per-wave fitness partials
     float* actsq_l = wave_fit + 8 * 16;                  // [16]
     __bf16* v_l = reinterpret_cast<__bf16*>(actsq_l + 16);  // [16][OP] V (GEMM1 B-operand)
-    __bf16* ud_l = v_l + kMembers * OP;                  // [OP][64] k-major [U;D2] (GEMM2 B)
+    __bf16* ud_l = v_l + kMembers * OPS;                 // [OP][KS] k-major [U;D2] (GEMM2 B)
 
     const long RO = (long)R * O, AO = (long)A * O;
     const float* eV = args.env_blob;
@@ -135,7 +139,7 @@ __global__ __launch_bounds__(512, 1) void rollout_v7_kernel(RolloutV7Args args) 
         mean_l[j] = in ? e_mean[j] : 0.0f;
         istd_l[j] = in ? 1.0f / e_std[j] : 0.0f;
     }
-    for (int j = tid; j < kMembers * 64; j += 512) hact_l[j] = f2b7(0.0f);
+    for (int j = tid; j < kMembers * KS; j += 512) hact_l[j] = f2b7(0.0f);
     for (int j = tid; j < kMembers * A_MAX; j += 512) {
         const int m = j / A_MAX, a = j % A_MAX;
         b_l[j] = (m < live && a < A) ? args.params[(long)(base_member + m) * (AO + A) + AO + a] : 0.0f;
@@ -146,8 +150,8 @@ __global__ __launch_bounds__(512, 1) void rollout_v7_kernel(RolloutV7Args args) 
     // GEMM1 B (V) lives in LDS — keeping it in registers on every wave
     // (it is only used by wave 0) was what pushed the kernel over the
     // 256-VGPR budget and demoted w_frag to scratch.
-    for (int j = tid; j < kMembers * OP; j += 512) {
-        const int r = j / OP, o = j % OP;
+    for (int j = tid; j < kMembers * OPS; j += 512) {
+        const int r = j / OPS, o = j % OPS;
         v_l[j] = (r < R && o < O) ? f2b7(eV[(long)r * O + o]) : f2b7(0.0f);
     }
     // GEMM2 B ([U;D2]) in LDS, k-major [o][k] so a lane's 8-consecutive-k
@@ -155,8 +159,8 @@ __global__ __launch_bounds__(512, 1) void rollout_v7_kernel(RolloutV7Args args) 
     // for BOTH GEMMs pushed past the 256-VGPR/8-wave cap and demoted the
     // policy weights to scratch — LDS B costs ~49 KB/step of bandwidth
     // but keeps w_frag in registers, which dominates.)
-    for (int j = tid; j < OP * 64; j += 512) {
-        const int o = j / 64, k = j % 64;
+    for (int j = tid; j < OP * KS; j += 512) {
+        const int o = j / KS, k = j % KS;
         float v = 0.0f;
         if (o < O) {
             if (k < R) v = eU[(long)k * O + o];
@@ -188,7 +192,7 @@ __global__ __launch_bounds__(512, 1) void rollout_v7_kernel(RolloutV7Args args) 
     }
 
     // ---- initial observations (philox stream per global member) -------------
-    for (int j = tid; j < kMembers * OP; j += 512) obs_l[j] = f2b7(0.0f);
+    for (int j = tid; j < kMembers * OPS; j += 512) obs_l[j] = f2b7(0.0f);
     __syncthreads();
     {
         const int per_member4 = (O + 3) / 4;
@@ -201,14 +205,14 @@ __global__ __launch_bounds__(512, 1) void rollout_v7_kernel(RolloutV7Args args) 
 #pragma unroll
             for (int u = 0; u < 4; ++u) {
                 const int j = j4 * 4 + u;
-                if (j < O) obs_l[m * OP + j] = f2b7(0.1f * z[u]);
+                if (j < O) obs_l[m * OPS + j] = f2b7(0.1f * z[u]);
             }
         }
     }
     __syncthreads();
-    for (int j = tid; j < kMembers * OP; j += 512) {
-        const int jo = j % OP;
-        obsn_l[j] = f2b7((b2f7(obs_l[j]) - mean_l[jo]) * istd_l[jo]);
+    for (int j = tid; j < kMembers * OPS; j += 512) {
+        const int jo = j % OPS;
+        obsn_l[j] = (jo < OP) ? f2b7((b2f7(obs_l[j]) - mean_l[jo]) * istd_l[jo]) : f2b7(0.0f);
     }
     __syncthreads();
 
@@ -230,7 +234,7 @@ __global__ __launch_bounds__(512, 1) void rollout_v7_kernel(RolloutV7Args args) 
 #pragma unroll
             for (int a = 0; a < A_MAX; ++a) acc[a] = 0.0f;
             if (my_member < live) {
-                const __bf16* on = obsn_l + my_member * OP + l32 * kChunk;
+                const __bf16* on = obsn_l + my_member * OPS + l32 * kChunk;
 #pragma unroll
                 for (int p = 0; p < kChunk / 2; ++p) {
                     const bf16x2_t o2 = *reinterpret_cast<const bf16x2_t*>(on + 2 * p);
@@ -248,7 +252,7 @@ __global__ __launch_bounds__(512, 1) void rollout_v7_kernel(RolloutV7Args args) 
                 for (int a = 0; a < A_MAX; ++a) {
                     if (a >= A) break;
                     const float av = fminf(fmaxf(acc[a] + b_l[my_member * A_MAX + a], -1.0f), 1.0f);
-                    hact_l[my_member * 64 + 16 + a] = f2b7(av);
+                    hact_l[my_member * KS + 16 + a] = f2b7(av);
                     sq = fmaf(av, av, sq);
                 }
                 actsq_total += sq;
@@ -259,28 +263,28 @@ __global__ __launch_bounds__(512, 1) void rollout_v7_kernel(RolloutV7Args args) 
             floatx4_t acc = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
             for (int s = 0; s < OP / 32; ++s) {
-                const bf16x8_t a_frag = *reinterpret_cast<const bf16x8_t*>(obs_l + g2_row * OP + s * 32 + g2_k0);
-                const bf16x8_t b_frag = *reinterpret_cast<const bf16x8_t*>(v_l + c_col * OP + s * 32 + g2_k0);
+                const bf16x8_t a_frag = *reinterpret_cast<const bf16x8_t*>(obs_l + g2_row * OPS + s * 32 + g2_k0);
+                const bf16x8_t b_frag = *reinterpret_cast<const bf16x8_t*>(v_l + c_col * OPS + s * 32 + g2_k0);
                 acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, b_frag, acc, 0, 0, 0);
             }
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
                 // D: col = h-index, row = member
-                hact_l[(c_row0 + r) * 64 + c_col] = f2b7(acc[r]);
+                hact_l[(c_row0 + r) * KS + c_col] = f2b7(acc[r]);
             }
         }
         __syncthreads();
 
         // ===== GEMM2: o' = tanh(hact @ [U;D2] + c); fused epilogue =====
         {
-            const bf16x8_t a0 = *reinterpret_cast<const bf16x8_t*>(hact_l + g2_row * 64 + g2_k0);
-            const bf16x8_t a1 = *reinterpret_cast<const bf16x8_t*>(hact_l + g2_row * 64 + 32 + g2_k0);
+            const bf16x8_t a0 = *reinterpret_cast<const bf16x8_t*>(hact_l + g2_row * KS + g2_k0);
+            const bf16x8_t a1 = *reinterpret_cast<const bf16x8_t*>(hact_l + g2_row * KS + 32 + g2_k0);
             floatx4_t out[kTilesPerWave];
 #pragma unroll
             for (int tw = 0; tw < kTilesPerWave; ++tw) {
                 const int o = (wave * kTilesPerWave + tw) * 16 + c_col;
-                const bf16x8_t b0 = *reinterpret_cast<const bf16x8_t*>(ud_l + o * 64 + g2_k0);
-                const bf16x8_t b1 = *reinterpret_cast<const bf16x8_t*>(ud_l + o * 64 + 32 + g2_k0);
+                const bf16x8_t b0 = *reinterpret_cast<const bf16x8_t*>(ud_l + o * KS + g2_k0);
+                const bf16x8_t b1 = *reinterpret_cast<const bf16x8_t*>(ud_l + o * KS + 32 + g2_k0);
                 floatx4_t acc = {0.f, 0.f, 0.f, 0.f};
                 acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc, 0, 0, 0);
                 acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b1, acc, 0, 0, 0);
@@ -305,8 +309,8 @@ __global__ __launch_bounds__(512, 1) void rollout_v7_kernel(RolloutV7Args args) 
                         ssq = fmaf(o_new, o_new, ssq);
                     }
                     const __bf16 ob = f2b7(o_new);
-                    obs_l[m * OP + col] = ob;
-                    obsn_l[m * OP + col] = f2b7((b2f7(ob) - mv) * iv);
+                    obs_l[m * OPS + col] = ob;
+                    obsn_l[m * OPS + col] = f2b7((b2f7(ob) - mv) * iv);
                 }
                 stat_sum[tw] += ssum;
                 stat_sumsq[tw] += ssq;
@@ -372,7 +376,7 @@ void rollout_v7(torch::Tensor params, torch::Tensor env_blob, torch::Tensor obs_
 
     constexpr int O_T = 376, A_T = 17, OP = 384;
     TORCH_CHECK(O == O_T && A == A_T, "rollout v7 instantiated for the Humanoid geometry (obs 376, act 17)");
-    const size_t lds = (size_t)(3 * 16 * OP + 16 * 64 + OP * 64) * 2 + (size_t)(16 * A_T + 4 * OP + 8 * 16 + 16) * 4;
+    const size_t lds = (size_t)(3 * 16 * (OP + 8) + 16 * 72 + OP * 72) * 2 + (size_t)(16 * A_T + 4 * OP + 8 * 16 + 16) * 4;
     static bool attr_set7 = false;
     if (!attr_set7) {
         (void)hipFuncSetAttribute(reinterpret_cast<const void*>(&rollout_v7_kernel<O_T, A_T>),
