@@ -1,25 +1,35 @@
 // v7 rollout: MFMA whole-episode rollout for the LINEAR flagship policy
-// (K10+K11) — 16 members per workgroup.
+// (K10+K11).
 //
 // Why v7 beats v6 (2 members/block, all-v_dot2): v6 is LDS-bound — every
 // dot2 reads 8 B of LDS for 4 FLOPs (~220 KB LDS traffic per block-step).
 // The env dynamics (V·o, Uᵀh + D2ᵀa) have SHARED matrices across the
-// population, i.e. real GEMMs once ≥16 members sit in one block:
+// population, i.e. real GEMMs once many members sit in one block:
 //
-//   h(16×16)    = obs(16×384) @ Vᵀ(384×16)          GEMM1, 12× mfma
-//   o'(16×384)  = [h|act](16×64) @ [U;D2](64×384)    GEMM2, 48× mfma
+//   h(M×16)    = obs(M×384) @ Vᵀ(384×16)          GEMM1 on mfma
+//   o'(M×384)  = [h|act](M×64) @ [U;D2](64×384)    GEMM2 on mfma
 //
-// run on v_mfma_f32_16x16x32_bf16 with the STATIC B-operands (V, U, D2)
-// pre-loaded into per-lane register fragments once per episode — zero LDS
-// traffic for them in the steady state. The per-member policy GEMV
+// run on v_mfma_f32_16x16x32_bf16. The per-member policy GEMV
 // (act = W·obsn, no shared operand — not MFMA-shaped) keeps per-member
 // weights in REGISTERS (one 32-lane half-wave per member, 12 obs columns
-// per lane, 17 accumulators) so each obsn element is read from LDS once
-// per step instead of 17 times. Per-step LDS traffic drops ~3× and the
-// member count per block rises 8×.
+// per lane, 17 accumulators, DPP-reduced on the VALU pipe).
+//
+// Occupancy (the v7.2→v7.3 lesson, measured via SQ_WAIT_ANY:SQ_BUSY ≈ 5:1):
+// a single 8-wave block per CU exposes every barrier and LDS-latency chain
+// of the 1000-step loop — there is nothing else to run. v7.3 uses 4-wave
+// (256-thread) blocks of 8 members with ≤ 80 KB LDS and ≤ 256 VGPRs, so
+// TWO independent blocks share each CU and cover each other's stalls; the
+// GEMM2 B-operand ([U;D2]) lives in register fragments (48 VGPRs), the
+// GEMM1 B-operand (V) in LDS.
+//
+// LDS row strides are padded (OPS = OP+8, KS = 72) — the natural 768 B /
+// 128 B strides are ≡ 0 (mod 64 dwords), serializing 16-row fragment
+// reads up to 16-way (measured: SQ_LDS_BANK_CONFLICT ≈ 2× SQ_BUSY).
 //
 // Numerics contract: identical to v6 / rollout_eager (bf16 operands, fp32
-// accumulate, quantize-then-normalize obs).
+// accumulate, quantize-then-normalize obs); tanh via the hardware exp
+// unit (~1e-6 rel. error, far below the bf16 quantization of every
+// operand).
 //
 // Layout facts verified on-device (native_probes/mfma_probe.hip):
 //   A(16×32): row = lane&15,  k = (lane>>4)*8 + i
@@ -30,12 +40,13 @@
 #include <torch/extension.h>
 #include <ATen/cuda/CUDAContext.h>
 
+#include <cstdlib>
+
 #include "philox.h"
 
 namespace ea {
 
 typedef __attribute__((ext_vector_type(2))) __bf16 bf16x2_t;
-typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4_t;
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8_t;
 typedef __attribute__((ext_vector_type(4))) float floatx4_t;
 
@@ -51,10 +62,10 @@ __device__ __forceinline__ float tanh_fast(float x) {
     return (e - 1.0f) / (e + 1.0f);
 }
 
-// One step of a 32-lane sum reduce on the VALU pipe via DPP modifiers —
-// __shfl_down lowers to ds_bpermute_b32 (LDS pipe), and the policy
-// phase's 17 accumulators × 5 levels were 85 LDS-pipe ops per lane per
-// step, fighting the dot-product LDS reads. After row_shr 1/2/4/8 +
+// 32-lane sum reduce on the VALU pipe via DPP modifiers — __shfl_down
+// lowers to ds_bpermute_b32 (LDS pipe), and the policy phase's 17
+// accumulators × 5 levels were 85 LDS-pipe ops per lane per step,
+// fighting the dot-product LDS reads. After row_shr 1/2/4/8 +
 // row_bcast:15 the 32-lane totals sit in lanes 31 and 63.
 template <int kCtrl>
 __device__ __forceinline__ float dpp_add(float x) {
@@ -81,46 +92,48 @@ struct RolloutV7Args {
     int obs_dim, act_dim, rank, steps;
     float alive_bonus, act_cost;
     unsigned long long init_seed;
+    int skip_mask;  // perf probe only (EVOTORCH_AMD_V7_SKIP): 1=policy 2=GEMM1 4=GEMM2
 };
 
 // Compile-time geometry (all guards constant-fold; a runtime-guarded
-// W-load was measured to demote w_frag to scratch memory): O padded to OP
-// (multiple of 128 so 8 waves split OP/16 tiles evenly), R == 16 (one
-// GEMM1 tile), A <= 31 (fits hact K=64).
-template <int O, int A>
-__global__ __launch_bounds__(512, 1) void rollout_v7_kernel(RolloutV7Args args) {
-    constexpr int kMembers = 16;
+// W-load was measured to demote w_frag to scratch memory). kWaves ∈ {4, 8}:
+// members per block = 2·kWaves.
+template <int O, int A, int kWaves>
+__global__ __launch_bounds__(64 * kWaves, 1) void rollout_v7_kernel(RolloutV7Args args) {
+    constexpr int kThreads = 64 * kWaves;
+    constexpr int kM = 2 * kWaves;            // members per block
     constexpr int OP = (O + 127) / 128 * 128;
+    constexpr int OPS = OP + 8;               // padded obs-row stride (bank decorrelation)
+    constexpr int KS = 72;                    // padded hact row stride (vs 64)
     constexpr int A_MAX = A;
     constexpr int R = 16;
-    constexpr int OPS = OP + 8;               // obs-row LDS stride: breaks the
-                                              // 768 B ≡ 0 (mod 64 dwords) bank
-                                              // aliasing of 16-row fragment reads
-    constexpr int KS = 72;                    // hact/ud row stride (vs 64), same reason
     constexpr int kTiles = OP / 16;           // GEMM2 output tiles
-    constexpr int kTilesPerWave = kTiles / 8; // = 3 at OP=384
+    constexpr int kTilesPerWave = kTiles / kWaves;
     constexpr int kChunk = OP / 32;           // policy obs columns per lane (12)
     const int tid = threadIdx.x;
     const int lane = tid & 63;
     const int wave = tid >> 6;
-    const int base_member = blockIdx.x * kMembers;
+    const int base_member = blockIdx.x * kM;
     if (base_member >= args.n_members) return;
-    const int live = min(kMembers, args.n_members - base_member);
+    const int live = min(kM, args.n_members - base_member);
 
     // ---- LDS ----
     extern __shared__ unsigned char lds7[];
-    __bf16* obs_l = reinterpret_cast<__bf16*>(lds7);     // [16][OP] raw (quantized) obs
-    __bf16* obsn_l = obs_l + kMembers * OPS;             // [16][OPS] normalized obs
-    __bf16* hact_l = obsn_l + kMembers * OPS;            // [16][KS]: k<16 h, 16..16+A act, rest 0
-    float* b_l = reinterpret_cast<float*>(hact_l + kMembers * KS);  // [16][A_MAX] bias
-    float* c_l = b_l + kMembers * A_MAX;                 // [OP]
+    __bf16* obs_l = reinterpret_cast<__bf16*>(lds7);     // [kM][OPS] raw (quantized) obs
+    __bf16* obsn_l = obs_l + kM * OPS;                   // [kM][OPS] normalized obs
+    __bf16* hact_l = obsn_l + kM * OPS;                  // [16][KS]: k<16 h, 16..16+A act, rest 0
+    __bf16* v_l = hact_l + 16 * KS;                      // [16][OPS] V (GEMM1 B-operand)
+    float* b_l = reinterpret_cast<float*>(v_l + 16 * OPS);  // [kM][A_MAX] bias
+    float* c_l = b_l + kM * A_MAX;                       // [OP]
     float* wr_l = c_l + OP;                              // [OP]
     float* mean_l = wr_l + OP;                           // [OP]
     float* istd_l = mean_l + OP;                         // [OP]
-    float* wave_fit = istd_l + OP;                       // [8][16] per-wave fitness partials
-    float* actsq_l = wave_fit + 8 * 16;                  // [16]
-    __bf16* v_l = reinterpret_cast<__bf16*>(actsq_l + 16);  // [16][OP] V (GEMM1 B-operand)
-    __bf16* ud_l = v_l + kMembers * OPS;                 // [OP][KS] k-major [U;D2] (GEMM2 B)
+    float* wave_fit = istd_l + OP;                       // [kWaves][16] fitness partials
+    float* actsq_l = wave_fit + kWaves * 16;             // [kM]
+    // 8-wave variant: GEMM2 B in LDS (register fragments would exceed the
+    // 256-VGPR cap of a 512-thread block); 4-wave variant: in registers
+    // (2 blocks/CU needs <= 80 KB LDS).
+    __bf16* ud_l = reinterpret_cast<__bf16*>(actsq_l + kM);  // [OP][KS] (kWaves==8 only)
 
     const long RO = (long)R * O, AO = (long)A * O;
     const float* eV = args.env_blob;
@@ -132,41 +145,62 @@ __global__ __launch_bounds__(512, 1) void rollout_v7_kernel(RolloutV7Args args) 
     const float* e_std = e_mean + O;
 
     // ---- stage vectors (pads: c=0, wr=0, mean=0, istd=0 → obsn pad = 0) ----
-    for (int j = tid; j < OP; j += 512) {
+    for (int j = tid; j < OP; j += kThreads) {
         const bool in = j < O;
         c_l[j] = in ? e_c[j] : 0.0f;
         wr_l[j] = in ? e_wr[j] : 0.0f;
         mean_l[j] = in ? e_mean[j] : 0.0f;
         istd_l[j] = in ? 1.0f / e_std[j] : 0.0f;
     }
-    for (int j = tid; j < kMembers * KS; j += 512) hact_l[j] = f2b7(0.0f);
-    for (int j = tid; j < kMembers * A_MAX; j += 512) {
+    for (int j = tid; j < 16 * KS; j += kThreads) hact_l[j] = f2b7(0.0f);
+    for (int j = tid; j < kM * A_MAX; j += kThreads) {
         const int m = j / A_MAX, a = j % A_MAX;
         b_l[j] = (m < live && a < A) ? args.params[(long)(base_member + m) * (AO + A) + AO + a] : 0.0f;
     }
-    if (tid < kMembers) actsq_l[tid] = 0.0f;
+    if (tid < kM) actsq_l[tid] = 0.0f;
 
-    // ---- static B-operands ---------------------------------------------------
-    // GEMM1 B (V) lives in LDS — keeping it in registers on every wave
-    // (it is only used by wave 0) was what pushed the kernel over the
-    // 256-VGPR budget and demoted w_frag to scratch.
-    for (int j = tid; j < kMembers * OPS; j += 512) {
+    // GEMM1 B (V) in LDS (only wave 0 reads it — registers on every wave
+    // would blow the VGPR budget and demote w_frag to scratch).
+    for (int j = tid; j < 16 * OPS; j += kThreads) {
         const int r = j / OPS, o = j % OPS;
         v_l[j] = (r < R && o < O) ? f2b7(eV[(long)r * O + o]) : f2b7(0.0f);
     }
-    // GEMM2 B ([U;D2]) in LDS, k-major [o][k] so a lane's 8-consecutive-k
-    // fragment is one aligned 16 B read. (Register-resident B-fragments
-    // for BOTH GEMMs pushed past the 256-VGPR/8-wave cap and demoted the
-    // policy weights to scratch — LDS B costs ~49 KB/step of bandwidth
-    // but keeps w_frag in registers, which dominates.)
-    for (int j = tid; j < OP * KS; j += 512) {
-        const int o = j / KS, k = j % KS;
-        float v = 0.0f;
-        if (o < O) {
-            if (k < R) v = eU[(long)k * O + o];
-            else if (k - R < A) v = eD2[(long)(k - R) * O + o];
+
+    const int g2_row = lane & 15;              // A-frag row (member) for GEMM2/GEMM1
+    const int g2_k0 = (lane >> 4) * 8;
+    const int c_col = lane & 15;               // C-frag col
+    const int c_row0 = (lane >> 4) * 4;        // C-frag first row (member)
+
+    // GEMM2 B ([U;D2]): register fragments (4-wave) or k-major LDS (8-wave).
+    bf16x8_t ud_frag[kWaves == 4 ? kTilesPerWave : 1][2];
+    if constexpr (kWaves == 4) {
+#pragma unroll
+        for (int tw = 0; tw < kTilesPerWave; ++tw) {
+            const int o = (wave * kTilesPerWave + tw) * 16 + c_col;
+#pragma unroll
+            for (int sl = 0; sl < 2; ++sl) {
+#pragma unroll
+                for (int i = 0; i < 8; ++i) {
+                    const int k = sl * 32 + g2_k0 + i;
+                    float v = 0.0f;
+                    if (o < O) {
+                        if (k < R) v = eU[(long)k * O + o];
+                        else if (k - R < A) v = eD2[(long)(k - R) * O + o];
+                    }
+                    ud_frag[tw][sl][i] = f2b7(v);
+                }
+            }
         }
-        ud_l[j] = f2b7(v);
+    } else {
+        for (int j = tid; j < OP * KS; j += kThreads) {
+            const int o = j / KS, k = j % KS;
+            float v = 0.0f;
+            if (o < O) {
+                if (k < R) v = eU[(long)k * O + o];
+                else if (k - R < A) v = eD2[(long)(k - R) * O + o];
+            }
+            ud_l[j] = f2b7(v);
+        }
     }
 
     // ---- per-member policy weights in registers -----------------------------
@@ -192,11 +226,11 @@ __global__ __launch_bounds__(512, 1) void rollout_v7_kernel(RolloutV7Args args) 
     }
 
     // ---- initial observations (philox stream per global member) -------------
-    for (int j = tid; j < kMembers * OPS; j += 512) obs_l[j] = f2b7(0.0f);
+    for (int j = tid; j < kM * OPS; j += kThreads) obs_l[j] = f2b7(0.0f);
     __syncthreads();
     {
         const int per_member4 = (O + 3) / 4;
-        for (int idx = tid; idx < kMembers * per_member4; idx += 512) {
+        for (int idx = tid; idx < kM * per_member4; idx += kThreads) {
             const int m = idx / per_member4;
             const int j4 = idx % per_member4;
             if (m >= live) continue;
@@ -210,7 +244,7 @@ __global__ __launch_bounds__(512, 1) void rollout_v7_kernel(RolloutV7Args args) 
         }
     }
     __syncthreads();
-    for (int j = tid; j < kMembers * OPS; j += 512) {
+    for (int j = tid; j < kM * OPS; j += kThreads) {
         const int jo = j % OPS;
         obsn_l[j] = (jo < OP) ? f2b7((b2f7(obs_l[j]) - mean_l[jo]) * istd_l[jo]) : f2b7(0.0f);
     }
@@ -218,22 +252,32 @@ __global__ __launch_bounds__(512, 1) void rollout_v7_kernel(RolloutV7Args args) 
 
     // ---- episode state in registers ----
     float fit_part[4] = {0.f, 0.f, 0.f, 0.f};  // member rows (lane>>4)*4+reg of GEMM2
-    float actsq_total = 0.0f;                  // on l32==0 lanes: my_member's Σa²
+    float actsq_total = 0.0f;                  // on l32==31 lanes: my_member's Σa²
     float stat_sum[kTilesPerWave] = {};        // per owned obs column
     float stat_sumsq[kTilesPerWave] = {};
 
-    const int g2_row = lane & 15;              // A-frag row (member) for GEMM2/GEMM1
-    const int g2_k0 = (lane >> 4) * 8;
-    const int c_col = lane & 15;               // C-frag col
-    const int c_row0 = (lane >> 4) * 4;        // C-frag first row (member)
+    // A-frag rows beyond the stored kM member rows alias row 0 (safe reads;
+    // their outputs are masked in every epilogue).
+    const int a_row = (g2_row < kM) ? g2_row : 0;
 
     for (int t = 0; t < args.steps; ++t) {
+        // ===== GEMM1 (wave 0): h = obs @ Vᵀ — issued before the policy VALU
+        // work so the MFMA chain executes on the MAI pipe underneath it =====
+        floatx4_t h_acc = {0.f, 0.f, 0.f, 0.f};
+        if (wave == 0 && !(args.skip_mask & 2)) {
+#pragma unroll
+            for (int s = 0; s < OP / 32; ++s) {
+                const bf16x8_t a_frag = *reinterpret_cast<const bf16x8_t*>(obs_l + a_row * OPS + s * 32 + g2_k0);
+                const bf16x8_t b_frag = *reinterpret_cast<const bf16x8_t*>(v_l + c_col * OPS + s * 32 + g2_k0);
+                h_acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, b_frag, h_acc, 0, 0, 0);
+            }
+        }
         // ===== policy: act = clamp(W · obsn + b) — per-member half-waves =====
         {
             float acc[A_MAX];
 #pragma unroll
             for (int a = 0; a < A_MAX; ++a) acc[a] = 0.0f;
-            if (my_member < live) {
+            if (my_member < live && !(args.skip_mask & 1)) {
                 const __bf16* on = obsn_l + my_member * OPS + l32 * kChunk;
 #pragma unroll
                 for (int p = 0; p < kChunk / 2; ++p) {
@@ -250,7 +294,6 @@ __global__ __launch_bounds__(512, 1) void rollout_v7_kernel(RolloutV7Args args) 
                 float sq = 0.0f;
 #pragma unroll
                 for (int a = 0; a < A_MAX; ++a) {
-                    if (a >= A) break;
                     const float av = fminf(fmaxf(acc[a] + b_l[my_member * A_MAX + a], -1.0f), 1.0f);
                     hact_l[my_member * KS + 16 + a] = f2b7(av);
                     sq = fmaf(av, av, sq);
@@ -258,41 +301,33 @@ __global__ __launch_bounds__(512, 1) void rollout_v7_kernel(RolloutV7Args args) 
                 actsq_total += sq;
             }
         }
-        // ===== GEMM1 (wave 0): h = obs @ Vᵀ =====
         if (wave == 0) {
-            floatx4_t acc = {0.f, 0.f, 0.f, 0.f};
-#pragma unroll
-            for (int s = 0; s < OP / 32; ++s) {
-                const bf16x8_t a_frag = *reinterpret_cast<const bf16x8_t*>(obs_l + g2_row * OPS + s * 32 + g2_k0);
-                const bf16x8_t b_frag = *reinterpret_cast<const bf16x8_t*>(v_l + c_col * OPS + s * 32 + g2_k0);
-                acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, b_frag, acc, 0, 0, 0);
-            }
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
                 // D: col = h-index, row = member
-                hact_l[(c_row0 + r) * KS + c_col] = f2b7(acc[r]);
+                const int m = c_row0 + r;
+                if (m < kM) hact_l[m * KS + c_col] = f2b7(h_acc[r]);
             }
         }
         __syncthreads();
 
         // ===== GEMM2: o' = tanh(hact @ [U;D2] + c); fused epilogue =====
-        {
+        if (!(args.skip_mask & 4)) {
             const bf16x8_t a0 = *reinterpret_cast<const bf16x8_t*>(hact_l + g2_row * KS + g2_k0);
             const bf16x8_t a1 = *reinterpret_cast<const bf16x8_t*>(hact_l + g2_row * KS + 32 + g2_k0);
-            floatx4_t out[kTilesPerWave];
-#pragma unroll
-            for (int tw = 0; tw < kTilesPerWave; ++tw) {
-                const int o = (wave * kTilesPerWave + tw) * 16 + c_col;
-                const bf16x8_t b0 = *reinterpret_cast<const bf16x8_t*>(ud_l + o * KS + g2_k0);
-                const bf16x8_t b1 = *reinterpret_cast<const bf16x8_t*>(ud_l + o * KS + 32 + g2_k0);
-                floatx4_t acc = {0.f, 0.f, 0.f, 0.f};
-                acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc, 0, 0, 0);
-                acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b1, acc, 0, 0, 0);
-                out[tw] = acc;
-            }
 #pragma unroll
             for (int tw = 0; tw < kTilesPerWave; ++tw) {
                 const int col = (wave * kTilesPerWave + tw) * 16 + c_col;
+                floatx4_t acc = {0.f, 0.f, 0.f, 0.f};
+                if constexpr (kWaves == 4) {
+                    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, ud_frag[tw][0], acc, 0, 0, 0);
+                    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, ud_frag[tw][1], acc, 0, 0, 0);
+                } else {
+                    const bf16x8_t b0 = *reinterpret_cast<const bf16x8_t*>(ud_l + col * KS + g2_k0);
+                    const bf16x8_t b1 = *reinterpret_cast<const bf16x8_t*>(ud_l + col * KS + 32 + g2_k0);
+                    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc, 0, 0, 0);
+                    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b1, acc, 0, 0, 0);
+                }
                 const float cv = c_l[col];
                 const float wrv = wr_l[col];
                 const float mv = mean_l[col];
@@ -302,15 +337,17 @@ __global__ __launch_bounds__(512, 1) void rollout_v7_kernel(RolloutV7Args args) 
 #pragma unroll
                 for (int r = 0; r < 4; ++r) {
                     const int m = c_row0 + r;
-                    const float o_new = tanh_fast(out[tw][r] + cv);
+                    const float o_new = tanh_fast(acc[r] + cv);
                     fit_part[r] = fmaf(wrv, o_new, fit_part[r]);
                     if (m < live && col_in) {
                         ssum += o_new;
                         ssq = fmaf(o_new, o_new, ssq);
                     }
-                    const __bf16 ob = f2b7(o_new);
-                    obs_l[m * OPS + col] = ob;
-                    obsn_l[m * OPS + col] = f2b7((b2f7(ob) - mv) * iv);
+                    if (m < kM) {
+                        const __bf16 ob = f2b7(o_new);
+                        obs_l[m * OPS + col] = ob;
+                        obsn_l[m * OPS + col] = f2b7((b2f7(ob) - mv) * iv);
+                    }
                 }
                 stat_sum[tw] += ssum;
                 stat_sumsq[tw] += ssq;
@@ -323,7 +360,7 @@ __global__ __launch_bounds__(512, 1) void rollout_v7_kernel(RolloutV7Args args) 
     // Deterministic fitness reduction (no float atomics — the kernel must
     // be bitwise run-to-run reproducible): (1) shuffle-reduce fit_part
     // across the 16 lanes of each row segment, (2) stage per-wave partials
-    // in LDS, (3) one thread per member sums the 8 waves in fixed order.
+    // in LDS, (3) one thread per member sums the waves in fixed order.
 #pragma unroll
     for (int off = 8; off > 0; off >>= 1) {
 #pragma unroll
@@ -338,7 +375,7 @@ __global__ __launch_bounds__(512, 1) void rollout_v7_kernel(RolloutV7Args args) 
     if (tid < live) {
         float total = 0.0f;
 #pragma unroll
-        for (int w = 0; w < 8; ++w) total += wave_fit[w * 16 + tid];
+        for (int w = 0; w < kWaves; ++w) total += wave_fit[w * 16 + tid];
         args.fitness_out[base_member + tid] =
             total + args.alive_bonus * (float)args.steps - args.act_cost * actsq_l[tid] / (float)A;
     }
@@ -352,14 +389,30 @@ __global__ __launch_bounds__(512, 1) void rollout_v7_kernel(RolloutV7Args args) 
     }
 }
 
+template <int O_T, int A_T, int kWaves>
+static void launch_v7(const RolloutV7Args& args, int n, hipStream_t stream) {
+    constexpr int OP = (O_T + 127) / 128 * 128;
+    constexpr int kM = 2 * kWaves;
+    size_t lds = (size_t)(2 * kM * (OP + 8) + 16 * 72 + 16 * (OP + 8)) * 2 +
+                 (size_t)(kM * A_T + 4 * OP + kWaves * 16 + kM) * 4;
+    if (kWaves == 8) lds += (size_t)OP * 72 * 2;  // ud_l
+    static bool attr_set7[2] = {false, false};
+    const int slot = (kWaves == 8) ? 1 : 0;
+    if (!attr_set7[slot]) {
+        (void)hipFuncSetAttribute(reinterpret_cast<const void*>(&rollout_v7_kernel<O_T, A_T, kWaves>),
+                                  hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
+        attr_set7[slot] = true;
+    }
+    const int blocks = (n + kM - 1) / kM;
+    hipLaunchKernelGGL((rollout_v7_kernel<O_T, A_T, kWaves>), dim3(blocks), dim3(64 * kWaves), lds, stream, args);
+}
+
 void rollout_v7(torch::Tensor params, torch::Tensor env_blob, torch::Tensor obs_stats_out, torch::Tensor fitness,
                 int64_t obs_dim, int64_t act_dim, int64_t rank, int64_t steps, double alive_bonus, double act_cost,
                 int64_t init_seed, int64_t member_offset) {
     const int n = (int)params.size(0);
     const int O = (int)obs_dim, A = (int)act_dim, R = (int)rank;
     TORCH_CHECK(R == 16, "rollout v7 requires rank 16");
-    TORCH_CHECK(A <= 17, "rollout v7 compiled for act_dim <= 17");
-    TORCH_CHECK(O <= 384, "rollout v7 compiled for obs_dim <= 384");
 
     RolloutV7Args args;
     args.params = params.data_ptr<float>();
@@ -374,18 +427,17 @@ void rollout_v7(torch::Tensor params, torch::Tensor env_blob, torch::Tensor obs_
     args.act_cost = (float)act_cost;
     args.init_seed = (unsigned long long)init_seed;
 
-    constexpr int O_T = 376, A_T = 17, OP = 384;
+    constexpr int O_T = 376, A_T = 17;
     TORCH_CHECK(O == O_T && A == A_T, "rollout v7 instantiated for the Humanoid geometry (obs 376, act 17)");
-    const size_t lds = (size_t)(3 * 16 * (OP + 8) + 16 * 72 + OP * 72) * 2 + (size_t)(16 * A_T + 4 * OP + 8 * 16 + 16) * 4;
-    static bool attr_set7 = false;
-    if (!attr_set7) {
-        (void)hipFuncSetAttribute(reinterpret_cast<const void*>(&rollout_v7_kernel<O_T, A_T>),
-                                  hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
-        attr_set7 = true;
-    }
     auto stream = at::cuda::getCurrentCUDAStream();
-    const int blocks = (n + 15) / 16;
-    hipLaunchKernelGGL((rollout_v7_kernel<O_T, A_T>), dim3(blocks), dim3(512), lds, stream, args);
+    args.skip_mask = 0;
+    if (const char* skp = getenv("EVOTORCH_AMD_V7_SKIP")) args.skip_mask = atoi(skp);  // perf probe only
+    const char* wenv = getenv("EVOTORCH_AMD_ROLLOUT_V7_WAVES");
+    if (wenv && atoi(wenv) == 4) {
+        launch_v7<O_T, A_T, 4>(args, n, stream);
+    } else {
+        launch_v7<O_T, A_T, 8>(args, n, stream);
+    }
 }
 
 }  // namespace ea
