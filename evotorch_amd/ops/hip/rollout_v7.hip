@@ -59,7 +59,9 @@ __device__ __forceinline__ float b2f7(__bf16 v) { return (float)v; }
 __device__ __forceinline__ float tanh_fast(float x) {
     const float xc = fminf(fmaxf(x, -15.0f), 15.0f);
     const float e = __expf(2.0f * xc);
-    return (e - 1.0f) / (e + 1.0f);
+    // v_rcp_f32 (~1 ulp) instead of an IEEE divide: the result is bf16-
+    // quantized immediately, so the 2^-23-level rcp error is invisible
+    return (e - 1.0f) * __builtin_amdgcn_rcpf(e + 1.0f);
 }
 
 // 32-lane sum reduce on the VALU pipe via DPP modifiers — __shfl_down
@@ -121,7 +123,7 @@ __global__ __launch_bounds__(64 * kWaves, 1) void rollout_v7_kernel(RolloutV7Arg
     extern __shared__ unsigned char lds7[];
     __bf16* obs_l = reinterpret_cast<__bf16*>(lds7);     // [kM][OPS] raw (quantized) obs
     __bf16* obsn_l = obs_l + kM * OPS;                   // [kM][OPS] normalized obs
-    __bf16* hact_l = obsn_l + kM * OPS;                  // [16][KS]: k<16 h, 16..16+A act, rest 0
+    __bf16* hact_l = obsn_l + kM * OPS;                  // [16][KS]: k<16 h, k=16..31 act[0..16), slot 32 = act[16]
     __bf16* v_l = hact_l + 16 * KS;                      // [16][OPS] V (GEMM1 B-operand)
     float* b_l = reinterpret_cast<float*>(v_l + 16 * OPS);  // [kM][A_MAX] bias
     float* c_l = b_l + kM * A_MAX;                       // [OP]
@@ -130,10 +132,11 @@ __global__ __launch_bounds__(64 * kWaves, 1) void rollout_v7_kernel(RolloutV7Arg
     float* istd_l = mean_l + OP;                         // [OP]
     float* wave_fit = istd_l + OP;                       // [kWaves][16] fitness partials
     float* actsq_l = wave_fit + kWaves * 16;             // [kM]
+    float* d2last_l = actsq_l + kM;                      // [OP] D2_T row A-1 (rank-1 epilogue term)
     // 8-wave variant: GEMM2 B in LDS (register fragments would exceed the
     // 256-VGPR cap of a 512-thread block); 4-wave variant: in registers
     // (2 blocks/CU needs <= 80 KB LDS).
-    __bf16* ud_l = reinterpret_cast<__bf16*>(actsq_l + kM);  // [OP][KS] (kWaves==8 only)
+    __bf16* ud_l = reinterpret_cast<__bf16*>(d2last_l + OP);  // [OP][40] k-major (kWaves==8 only)
 
     const long RO = (long)R * O, AO = (long)A * O;
     const float* eV = args.env_blob;
@@ -151,6 +154,9 @@ __global__ __launch_bounds__(64 * kWaves, 1) void rollout_v7_kernel(RolloutV7Arg
         wr_l[j] = in ? e_wr[j] : 0.0f;
         mean_l[j] = in ? e_mean[j] : 0.0f;
         istd_l[j] = in ? 1.0f / e_std[j] : 0.0f;
+        // act[A-1]'s D2 row is applied as a rank-1 epilogue term (bf16-
+        // quantized like the mfma operands) so the GEMM K stays 32
+        d2last_l[j] = in ? b2f7(f2b7(eD2[(long)(A - 1) * O + j])) : 0.0f;
     }
     for (int j = tid; j < 16 * KS; j += kThreads) hact_l[j] = f2b7(0.0f);
     for (int j = tid; j < kM * A_MAX; j += kThreads) {
@@ -171,33 +177,33 @@ __global__ __launch_bounds__(64 * kWaves, 1) void rollout_v7_kernel(RolloutV7Arg
     const int c_col = lane & 15;               // C-frag col
     const int c_row0 = (lane >> 4) * 4;        // C-frag first row (member)
 
-    // GEMM2 B ([U;D2]): register fragments (4-wave) or k-major LDS (8-wave).
-    bf16x8_t ud_frag[kWaves == 4 ? kTilesPerWave : 1][2];
+    // GEMM2 B ([U;D2 rows 0..15]): K = 32 exactly (R=16 h-rows + 16 action
+    // rows; action A-1 is the rank-1 epilogue term). Register fragments
+    // (4-wave) or k-major LDS with padded stride 40 (8-wave).
+    constexpr int KU = 40;  // ud_l row stride: 32 + 8 pad (bank decorrelation)
+    bf16x8_t ud_frag[kWaves == 4 ? kTilesPerWave : 1];
     if constexpr (kWaves == 4) {
 #pragma unroll
         for (int tw = 0; tw < kTilesPerWave; ++tw) {
             const int o = (wave * kTilesPerWave + tw) * 16 + c_col;
 #pragma unroll
-            for (int sl = 0; sl < 2; ++sl) {
-#pragma unroll
-                for (int i = 0; i < 8; ++i) {
-                    const int k = sl * 32 + g2_k0 + i;
-                    float v = 0.0f;
-                    if (o < O) {
-                        if (k < R) v = eU[(long)k * O + o];
-                        else if (k - R < A) v = eD2[(long)(k - R) * O + o];
-                    }
-                    ud_frag[tw][sl][i] = f2b7(v);
+            for (int i = 0; i < 8; ++i) {
+                const int k = g2_k0 + i;
+                float v = 0.0f;
+                if (o < O) {
+                    if (k < R) v = eU[(long)k * O + o];
+                    else if (k - R < A - 1) v = eD2[(long)(k - R) * O + o];
                 }
+                ud_frag[tw][i] = f2b7(v);
             }
         }
     } else {
-        for (int j = tid; j < OP * KS; j += kThreads) {
-            const int o = j / KS, k = j % KS;
+        for (int j = tid; j < OP * KU; j += kThreads) {
+            const int o = j / KU, k = j % KU;
             float v = 0.0f;
-            if (o < O) {
+            if (o < O && k < 32) {
                 if (k < R) v = eU[(long)k * O + o];
-                else if (k - R < A) v = eD2[(long)(k - R) * O + o];
+                else if (k - R < A - 1) v = eD2[(long)(k - R) * O + o];
             }
             ud_l[j] = f2b7(v);
         }
@@ -273,11 +279,11 @@ __global__ __launch_bounds__(64 * kWaves, 1) void rollout_v7_kernel(RolloutV7Arg
             }
         }
         // ===== policy: act = clamp(W · obsn + b) — per-member half-waves =====
-        {
+        if (!(args.skip_mask & 1)) {
             float acc[A_MAX];
 #pragma unroll
             for (int a = 0; a < A_MAX; ++a) acc[a] = 0.0f;
-            if (my_member < live && !(args.skip_mask & 1)) {
+            if (my_member < live) {
                 const __bf16* on = obsn_l + my_member * OPS + l32 * kChunk;
 #pragma unroll
                 for (int p = 0; p < kChunk / 2; ++p) {
@@ -295,7 +301,7 @@ __global__ __launch_bounds__(64 * kWaves, 1) void rollout_v7_kernel(RolloutV7Arg
 #pragma unroll
                 for (int a = 0; a < A_MAX; ++a) {
                     const float av = fminf(fmaxf(acc[a] + b_l[my_member * A_MAX + a], -1.0f), 1.0f);
-                    hact_l[my_member * KS + 16 + a] = f2b7(av);
+                    hact_l[my_member * KS + ((a < 16) ? (16 + a) : 32)] = f2b7(av);
                     sq = fmaf(av, av, sq);
                 }
                 actsq_total += sq;
@@ -309,25 +315,29 @@ __global__ __launch_bounds__(64 * kWaves, 1) void rollout_v7_kernel(RolloutV7Arg
                 if (m < kM) hact_l[m * KS + c_col] = f2b7(h_acc[r]);
             }
         }
-        __syncthreads();
+        if (!(args.skip_mask & 8)) __syncthreads();
 
-        // ===== GEMM2: o' = tanh(hact @ [U;D2] + c); fused epilogue =====
+        // ===== GEMM2: o' = tanh(hact @ [U;D2] + act16·d2last + c) =====
         if (!(args.skip_mask & 4)) {
             const bf16x8_t a0 = *reinterpret_cast<const bf16x8_t*>(hact_l + g2_row * KS + g2_k0);
-            const bf16x8_t a1 = *reinterpret_cast<const bf16x8_t*>(hact_l + g2_row * KS + 32 + g2_k0);
+            // per-member act[A-1] for the rank-1 term, indexed by C rows
+            float a16[4];
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int m = c_row0 + r;
+                a16[r] = b2f7(hact_l[((m < kM) ? m : 0) * KS + 32]);
+            }
 #pragma unroll
             for (int tw = 0; tw < kTilesPerWave; ++tw) {
                 const int col = (wave * kTilesPerWave + tw) * 16 + c_col;
                 floatx4_t acc = {0.f, 0.f, 0.f, 0.f};
                 if constexpr (kWaves == 4) {
-                    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, ud_frag[tw][0], acc, 0, 0, 0);
-                    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, ud_frag[tw][1], acc, 0, 0, 0);
+                    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, ud_frag[tw], acc, 0, 0, 0);
                 } else {
-                    const bf16x8_t b0 = *reinterpret_cast<const bf16x8_t*>(ud_l + col * KS + g2_k0);
-                    const bf16x8_t b1 = *reinterpret_cast<const bf16x8_t*>(ud_l + col * KS + 32 + g2_k0);
+                    const bf16x8_t b0 = *reinterpret_cast<const bf16x8_t*>(ud_l + col * KU + g2_k0);
                     acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc, 0, 0, 0);
-                    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b1, acc, 0, 0, 0);
                 }
+                const float d2l = d2last_l[col];
                 const float cv = c_l[col];
                 const float wrv = wr_l[col];
                 const float mv = mean_l[col];
@@ -337,7 +347,7 @@ __global__ __launch_bounds__(64 * kWaves, 1) void rollout_v7_kernel(RolloutV7Arg
 #pragma unroll
                 for (int r = 0; r < 4; ++r) {
                     const int m = c_row0 + r;
-                    const float o_new = tanh_fast(acc[r] + cv);
+                    const float o_new = tanh_fast(fmaf(a16[r], d2l, acc[r]) + cv);
                     fit_part[r] = fmaf(wrv, o_new, fit_part[r]);
                     if (m < live && col_in) {
                         ssum += o_new;
@@ -353,7 +363,7 @@ __global__ __launch_bounds__(64 * kWaves, 1) void rollout_v7_kernel(RolloutV7Arg
                 stat_sumsq[tw] += ssq;
             }
         }
-        __syncthreads();
+        if (!(args.skip_mask & 8)) __syncthreads();
     }
 
     // ---- wrap-up ----
@@ -394,8 +404,8 @@ static void launch_v7(const RolloutV7Args& args, int n, hipStream_t stream) {
     constexpr int OP = (O_T + 127) / 128 * 128;
     constexpr int kM = 2 * kWaves;
     size_t lds = (size_t)(2 * kM * (OP + 8) + 16 * 72 + 16 * (OP + 8)) * 2 +
-                 (size_t)(kM * A_T + 4 * OP + kWaves * 16 + kM) * 4;
-    if (kWaves == 8) lds += (size_t)OP * 72 * 2;  // ud_l
+                 (size_t)(kM * A_T + 5 * OP + kWaves * 16 + kM) * 4;
+    if (kWaves == 8) lds += (size_t)OP * 40 * 2;  // ud_l
     static bool attr_set7[2] = {false, false};
     const int slot = (kWaves == 8) ? 1 : 0;
     if (!attr_set7[slot]) {
