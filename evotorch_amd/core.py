@@ -1029,6 +1029,64 @@ class Problem(TensorMakerMixin, Serializable):
         self._spmd_gen_counter += 1
         return _splitmix64(self._spmd_seed_base ^ (self._spmd_gen_counter * 0x9E3779B97F4A7C15))
 
+    def _peek_spmd_seed(self, ahead: int = 1) -> int:
+        """Look ahead in the shared seed chain without advancing it (the
+        side-stream noise pre-generation needs next generation's seed)."""
+        from .ops.dispatch import _splitmix64
+
+        return _splitmix64(self._spmd_seed_base ^ ((self._spmd_gen_counter + ahead) * 0x9E3779B97F4A7C15))
+
+    # -- side-stream noise pre-generation (SURVEY.md §2.8 P2 overlap) --------
+    #
+    # The philox noise of generation g+1 does not depend on the updated
+    # distribution parameters, so it is generated on a side HIP stream
+    # CONCURRENTLY with generation g's evaluation and gradient all-reduce;
+    # after the parameter update only a cheap bandwidth-bound affine pass
+    # (x = mu ± sigma·z) sits on the critical path. Bitwise-identical to
+    # direct counter-addressed sampling (both run fmaf(sigma, z, mu)).
+
+    def _sample_with_pregen(self, distribution, values: torch.Tensor, seed: int, row_offset: int, dirs: int):
+        from . import ops
+
+        pg = getattr(self, "_pregen", None)
+        if (
+            pg is not None
+            and pg["seed"] == seed
+            and pg["row_offset"] == row_offset
+            and pg["z"].shape == (dirs, self._solution_length)
+        ):
+            torch.cuda.current_stream().wait_event(pg["ready"])
+            distribution.fill_from_noise(values, pg["z"])
+        else:
+            distribution.fill_counter_addressed(values, seed=seed, row_offset=row_offset)
+
+    def _schedule_pregen(self, dirs: int, row_offset: int, seed: int):
+        from . import ops
+
+        length = self._solution_length
+        if getattr(self, "_pregen_stream", None) is None:
+            self._pregen_stream = torch.cuda.Stream(device=self._device)
+        stream = self._pregen_stream
+        pg = getattr(self, "_pregen", None)
+        if pg is None or pg["z"].shape != (dirs, length):
+            pg = {
+                "z": torch.empty((dirs, length), dtype=torch.float32, device=self._device),
+                "zero": torch.zeros(length, dtype=torch.float32, device=self._device),
+                "one": torch.ones(length, dtype=torch.float32, device=self._device),
+                "ready": torch.cuda.Event(),
+            }
+        # the side stream must not overwrite z while the main stream's
+        # affine pass is still reading it
+        consumed = torch.cuda.Event()
+        consumed.record()
+        with torch.cuda.stream(stream):
+            stream.wait_event(consumed)
+            ops.sample_gaussian(pg["z"], pg["zero"], pg["one"], symmetric=False, seed=seed, row_offset=row_offset)
+            pg["ready"].record()
+        pg["seed"] = seed
+        pg["row_offset"] = row_offset
+        self._pregen = pg
+
     def request_fused_reduce(self, tensor: torch.Tensor, callback) -> None:
         """All-reduce `tensor` across ranks and hand the result to
         `callback`. During a sharded gradient generation the reduction is
@@ -1404,8 +1462,20 @@ class Problem(TensorMakerMixin, Serializable):
                 batch = self._get_cached_grad_batch(local_popsize, self._device)
             else:
                 batch = SolutionBatch(self, popsize=local_popsize, device=self._device, empty=True)
+            use_overlap = (
+                torch.device(self._device).type == "cuda"
+                and hasattr(distribution, "fill_from_noise")
+                and num_interactions is None  # adaptive rounds consume extra seeds
+            )
             with record_range("sample"):
-                distribution.fill_counter_addressed(batch.access_values(), seed=seed, row_offset=rank * dirs_local)
+                if use_overlap:
+                    self._sample_with_pregen(distribution, batch.access_values(), seed, rank * dirs_local, dirs_local)
+                    # pre-generate NEXT generation's noise on the side
+                    # stream; it runs underneath the evaluation below and
+                    # the gradient all-reduce after it
+                    self._schedule_pregen(dirs_local, rank * dirs_local, self._peek_spmd_seed(1))
+                else:
+                    distribution.fill_counter_addressed(batch.access_values(), seed=seed, row_offset=rank * dirs_local)
             with record_range("evaluate"):
                 self._eval_shard_batch(batch)
             round_values.append(batch._values)
@@ -1605,7 +1675,7 @@ class Problem(TensorMakerMixin, Serializable):
     def _get_cloned_state(self, *, memo: dict) -> dict:
         state = {}
         for k, v in self.__dict__.items():
-            if k in ("_generator", "_comm", "_grad_batch_cache"):
+            if k in ("_generator", "_comm", "_grad_batch_cache", "_pregen", "_pregen_stream"):
                 state[k] = None
             else:
                 state[k] = deep_clone(v, otherwise_deepcopy=True, memo=memo)

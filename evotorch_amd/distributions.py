@@ -250,6 +250,14 @@ class SeparableGaussian(Distribution):
 
         ops.sample_gaussian(out, self.mu, self.sigma, symmetric=self._symmetric, seed=seed, row_offset=row_offset)
 
+    def fill_from_noise(self, out: torch.Tensor, z: torch.Tensor):
+        """Apply x = mu + sigma*z from pre-generated standard normals
+        (side-stream overlap path) — bitwise equal to fill_counter_addressed
+        when z holds the same counter-addressed philox draws."""
+        from . import ops
+
+        ops.affine_from_noise(out, z, self.mu, self.sigma, symmetric=self._symmetric)
+
     def accumulate_elite_sums_streamed(self, chunk_iter, is_elite: torch.Tensor):
         """Masked (Σx, Σx²) over elite rows, chunk by chunk (fp64 accum).
         `is_elite` flags LOCAL rows; shards all-reduce the returned sums."""

@@ -21,6 +21,7 @@ from typing import Optional, Tuple
 import torch
 
 __all__ = [
+    "affine_from_noise",
     "hip_available",
     "hip_required",
     "load_hip",
@@ -177,6 +178,35 @@ def sample_gaussian(
     else:
         out.normal_(generator=generator)
         out.mul_(sigma).add_(mu)
+    return out
+
+
+def affine_from_noise(
+    out: torch.Tensor,
+    z: torch.Tensor,
+    mu: torch.Tensor,
+    sigma: torch.Tensor,
+    *,
+    symmetric: bool,
+) -> torch.Tensor:
+    """x = mu + sigma*z from PRE-GENERATED fp32 standard normals (z filled
+    counter-addressed on a side stream, hidden behind evaluation and the
+    gradient all-reduce). Bitwise-identical to `sample_gaussian` with the
+    same seed/row_offset: both compute fmaf(sigma, z, mu) in fp32."""
+    if out.device.type == "cuda" and not _allow_eager_on_gpu():
+        mod = hip_required()
+        mod.affine_from_noise(out, z, mu.to(out.dtype), sigma.to(out.dtype), bool(symmetric))
+        return out
+    n = out.shape[0]
+    rows = n // 2 if symmetric else n
+    mu32 = mu.to(torch.float32)
+    s32 = sigma.to(torch.float32)
+    plus = (mu32 + s32 * z.reshape(rows, -1)).to(out.dtype)
+    if symmetric:
+        out[:rows] = plus
+        out[rows:] = (2.0 * mu32 - plus.to(torch.float32)).to(out.dtype)
+    else:
+        out.copy_(plus)
     return out
 
 

@@ -5,6 +5,7 @@ namespace ea {
 void sample_gaussian(torch::Tensor out, torch::Tensor mu, torch::Tensor sigma, bool symmetric, int64_t seed, int64_t row_offset);
 void sample_gaussian_graphsafe(torch::Tensor out, torch::Tensor mu, torch::Tensor sigma, bool symmetric,
                                torch::Tensor seed_buf);
+void affine_from_noise(torch::Tensor out, torch::Tensor z, torch::Tensor mu, torch::Tensor sigma, bool symmetric);
 std::vector<torch::Tensor> es_gradients(torch::Tensor samples, torch::Tensor mu, torch::Tensor sigma,
                                         torch::Tensor weights, bool symmetric);
 std::vector<torch::Tensor> snes_gradients(torch::Tensor samples, torch::Tensor mu, torch::Tensor sigma,
@@ -26,6 +27,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("row_offset") = 0);
     m.def("sample_gaussian_graphsafe", &ea::sample_gaussian_graphsafe,
           "K1 (hipGraph-safe): seed read from and advanced in device memory");
+    m.def("affine_from_noise", &ea::affine_from_noise,
+          "K1 overlap: mu + sigma*z from pre-generated side-stream noise");
     m.def("es_gradients", &ea::es_gradients, "K3: fused (mu, sigma) ES gradient reduction");
     m.def("snes_gradients", &ea::snes_gradients, "K3: SNES raw-noise gradient reduction");
     m.def("clipup_step", &ea::clipup_step, "K4: fused ClipUp velocity update (no host sync)");

@@ -182,6 +182,49 @@ void sample_gaussian_graphsafe(torch::Tensor out, torch::Tensor mu, torch::Tenso
     hipLaunchKernelGGL(bump_seed_kernel, dim3(1), dim3(1), 0, stream, ptr);
 }
 
+// Apply x = mu + sigma*z from PRE-GENERATED standard normals (the noise
+// was filled on a side stream, hidden behind evaluation / collectives;
+// only this cheap bandwidth-bound affine pass sits on the critical path
+// after the distribution update — SURVEY.md §2.8 P2 overlap).
+// Symmetric: z has N/2 rows; out rows [0,N/2) = mu+sigma*z, mirrored below.
+template <typename T, bool kSymmetric>
+__global__ void affine_from_noise_kernel(T* __restrict__ out, const float* __restrict__ z,
+                                         const T* __restrict__ mu, const T* __restrict__ sigma,
+                                         int64_t rows, int64_t length) {
+    const int64_t total = rows * length;
+    for (int64_t e = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; e < total;
+         e += (int64_t)gridDim.x * blockDim.x) {
+        const int64_t col = e % length;
+        const float m = static_cast<float>(mu[col]);
+        const float s = static_cast<float>(sigma[col]);
+        const float plus = fmaf(s, z[e], m);
+        out[e] = static_cast<T>(plus);
+        if (kSymmetric) out[e + total] = static_cast<T>(2.0f * m - plus);
+    }
+}
+
+void affine_from_noise(torch::Tensor out, torch::Tensor z, torch::Tensor mu, torch::Tensor sigma, bool symmetric) {
+    CHECK_GPU(out); CHECK_CONTIG(out); CHECK_GPU(z); CHECK_CONTIG(z);
+    const int64_t n = out.size(0), length = out.size(1);
+    TORCH_CHECK(!symmetric || n % 2 == 0, "symmetric affine needs even popsize");
+    const int64_t rows = symmetric ? n / 2 : n;
+    TORCH_CHECK(z.numel() == rows * length, "noise buffer shape mismatch");
+    TORCH_CHECK(z.scalar_type() == at::ScalarType::Float, "noise must be fp32");
+    const int threads = 256;
+    const int blocks = (int)std::min<int64_t>((rows * length + threads - 1) / threads, 256 * 8);
+    auto stream = at::cuda::getCurrentCUDAStream();
+    AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::BFloat16, at::ScalarType::Half, out.scalar_type(), "affine_from_noise", [&] {
+        using T = scalar_t;
+        if (symmetric) {
+            hipLaunchKernelGGL((affine_from_noise_kernel<T, true>), dim3(blocks), dim3(threads), 0, stream,
+                               out.data_ptr<T>(), z.data_ptr<float>(), mu.data_ptr<T>(), sigma.data_ptr<T>(), rows, length);
+        } else {
+            hipLaunchKernelGGL((affine_from_noise_kernel<T, false>), dim3(blocks), dim3(threads), 0, stream,
+                               out.data_ptr<T>(), z.data_ptr<float>(), mu.data_ptr<T>(), sigma.data_ptr<T>(), rows, length);
+        }
+    });
+}
+
 // ---------------------------------------------------------------------------
 // K3: fused ES gradient reductions (N x L -> L), fp32 accumulation
 // ---------------------------------------------------------------------------
