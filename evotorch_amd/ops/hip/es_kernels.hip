@@ -263,13 +263,10 @@ __global__ void es_grad_kernel(const T* __restrict__ x, const T* __restrict__ mu
             acc_sigma = fmaf(wv, raw * raw - 1.0f, acc_sigma);
         }
     }
-    if (row_chunks == 1) {
-        mu_grad[col] = acc_mu;
-        sigma_grad[col] = acc_sigma;
-    } else {
-        atomicAdd(&mu_grad[col], acc_mu);
-        atomicAdd(&sigma_grad[col], acc_sigma);
-    }
+    // per-chunk partial rows (summed by a deterministic torch reduction) —
+    // float atomicAdd would make the gradient run-to-run order-dependent
+    mu_grad[(int64_t)chunk * length + col] = acc_mu;
+    sigma_grad[(int64_t)chunk * length + col] = acc_sigma;
 }
 
 template <GradMode kMode>
@@ -284,8 +281,8 @@ std::vector<torch::Tensor> es_grad_launch(torch::Tensor samples, torch::Tensor m
     // fill the chip: >= ~1024 blocks total
     int row_chunks = 1;
     if (col_blocks < 1024) row_chunks = (int)std::min<int64_t>((1024 + col_blocks - 1) / col_blocks, std::max<int64_t>(rows / 8, 1));
-    auto mu_grad = (row_chunks == 1) ? torch::empty({length}, opts) : torch::zeros({length}, opts);
-    auto sigma_grad = (row_chunks == 1) ? torch::empty({length}, opts) : torch::zeros({length}, opts);
+    auto mu_grad = torch::empty({row_chunks, length}, opts);
+    auto sigma_grad = torch::empty({row_chunks, length}, opts);
     auto stream = at::cuda::getCurrentCUDAStream();
     AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::BFloat16, at::ScalarType::Half, samples.scalar_type(), "es_gradients", [&] {
         using T = scalar_t;
@@ -293,7 +290,10 @@ std::vector<torch::Tensor> es_grad_launch(torch::Tensor samples, torch::Tensor m
                            samples.data_ptr<T>(), mu.data_ptr<T>(), sigma.data_ptr<T>(), weights.data_ptr<T>(),
                            mu_grad.data_ptr<float>(), sigma_grad.data_ptr<float>(), rows, n, length, row_chunks);
     });
-    return {mu_grad.to(samples.scalar_type()), sigma_grad.to(samples.scalar_type())};
+    if (row_chunks > 1) {
+        return {mu_grad.sum(0).to(samples.scalar_type()), sigma_grad.sum(0).to(samples.scalar_type())};
+    }
+    return {mu_grad.reshape({length}).to(samples.scalar_type()), sigma_grad.reshape({length}).to(samples.scalar_type())};
 }
 
 std::vector<torch::Tensor> es_gradients(torch::Tensor samples, torch::Tensor mu, torch::Tensor sigma,
@@ -312,7 +312,7 @@ std::vector<torch::Tensor> snes_gradients(torch::Tensor samples, torch::Tensor m
 // ---------------------------------------------------------------------------
 
 template <typename T>
-__global__ void sumsq_kernel(const T* __restrict__ v, float* __restrict__ out, int64_t n) {
+__global__ void sumsq_kernel(const T* __restrict__ v, float* __restrict__ partials, int64_t n) {
     __shared__ float scratch[8];
     float acc = 0.0f;
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n; i += (int64_t)gridDim.x * blockDim.x) {
@@ -320,14 +320,25 @@ __global__ void sumsq_kernel(const T* __restrict__ v, float* __restrict__ out, i
         acc = fmaf(x, x, acc);
     }
     acc = block_reduce_sum<false>(acc, scratch);
-    if (threadIdx.x == 0) atomicAdd(out, acc);
+    if (threadIdx.x == 0) partials[blockIdx.x] = acc;
+}
+
+// Deterministic per-block partial combine (float atomicAdd would make
+// ClipUp's norms — and hence the whole trajectory — run-to-run
+// order-dependent): one block sums `count` partials in fixed order.
+__global__ void reduce_partials_kernel(const float* __restrict__ partials, float* __restrict__ out, int count) {
+    __shared__ float scratch[8];
+    float acc = 0.0f;
+    for (int i = threadIdx.x; i < count; i += blockDim.x) acc += partials[i];
+    acc = block_reduce_sum<false>(acc, scratch);
+    if (threadIdx.x == 0) *out = acc;
 }
 
 // velocity = momentum * velocity + grad * (step_size / ||grad||); also
-// accumulates ||velocity||^2 for the clip pass.
+// accumulates per-block ||velocity||^2 partials for the clip pass.
 template <typename T>
 __global__ void clipup_update_kernel(T* __restrict__ velocity, const T* __restrict__ grad,
-                                     const float* __restrict__ gnorm_sq, float* __restrict__ vnorm_sq,
+                                     const float* __restrict__ gnorm_sq, float* __restrict__ vnorm_partials,
                                      float step_size, float momentum, int64_t n) {
     __shared__ float scratch[8];
     const float gnorm = sqrtf(fmaxf(*gnorm_sq, 1e-30f));
@@ -339,7 +350,7 @@ __global__ void clipup_update_kernel(T* __restrict__ velocity, const T* __restri
         acc = fmaf(v, v, acc);
     }
     acc = block_reduce_sum<false>(acc, scratch);
-    if (threadIdx.x == 0) atomicAdd(vnorm_sq, acc);
+    if (threadIdx.x == 0) vnorm_partials[blockIdx.x] = acc;
 }
 
 template <typename T>
@@ -358,15 +369,19 @@ void clipup_step(torch::Tensor velocity, torch::Tensor grad, double step_size, d
     const int64_t n = velocity.numel();
     const int threads = 256;
     const int blocks = (int)std::min<int64_t>((n + threads - 1) / threads, 1024);
-    auto norms = torch::zeros({2}, velocity.options().dtype(torch::kFloat32));
+    // [0] gnorm², [1] vnorm², [2..2+blocks) per-block partials
+    auto scratch = torch::empty({2 + blocks}, velocity.options().dtype(torch::kFloat32));
     auto stream = at::cuda::getCurrentCUDAStream();
     AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::BFloat16, at::ScalarType::Half, velocity.scalar_type(), "clipup_step", [&] {
         using T = scalar_t;
-        float* gnorm_sq = norms.data_ptr<float>();
+        float* gnorm_sq = scratch.data_ptr<float>();
         float* vnorm_sq = gnorm_sq + 1;
-        hipLaunchKernelGGL((sumsq_kernel<T>), dim3(blocks), dim3(threads), 0, stream, grad.data_ptr<T>(), gnorm_sq, n);
+        float* partials = gnorm_sq + 2;
+        hipLaunchKernelGGL((sumsq_kernel<T>), dim3(blocks), dim3(threads), 0, stream, grad.data_ptr<T>(), partials, n);
+        hipLaunchKernelGGL(reduce_partials_kernel, dim3(1), dim3(256), 0, stream, partials, gnorm_sq, blocks);
         hipLaunchKernelGGL((clipup_update_kernel<T>), dim3(blocks), dim3(threads), 0, stream, velocity.data_ptr<T>(),
-                           grad.data_ptr<T>(), gnorm_sq, vnorm_sq, (float)step_size, (float)momentum, n);
+                           grad.data_ptr<T>(), gnorm_sq, partials, (float)step_size, (float)momentum, n);
+        hipLaunchKernelGGL(reduce_partials_kernel, dim3(1), dim3(256), 0, stream, partials, vnorm_sq, blocks);
         hipLaunchKernelGGL((clipup_clip_kernel<T>), dim3(blocks), dim3(threads), 0, stream, velocity.data_ptr<T>(),
                            vnorm_sq, (float)max_speed, n);
     });

@@ -427,11 +427,17 @@ __global__ __launch_bounds__(512, 2) void rollout_linear_kernel(RolloutArgs args
                 total + args.alive_bonus * args.steps - args.act_cost * act_total / (float)A;
         }
     }
+    // per-(block, member) stat slices: the j-loop visits each (m, jo) pair
+    // exactly once, so plain stores are race-free and — unlike the float
+    // atomicAdd they replace — run-to-run deterministic; the launcher sums
+    // the partial slices with one torch reduction.
     for (int j = tid; j < live_members * O; j += blockDim.x) {
         const int slot = j >= (int)blockDim.x;
+        const int m = j / O;
         const int jo = j % O;
-        atomicAdd(&args.obs_stats_out[jo], stat_sum[slot]);
-        atomicAdd(&args.obs_stats_out[O + jo], stat_sumsq[slot]);
+        float* stats = args.obs_stats_out + ((int64_t)blockIdx.x * kMembers + m) * 2 * O;
+        stats[jo] = stat_sum[slot];
+        stats[O + jo] = stat_sumsq[slot];
     }
 }
 
@@ -466,8 +472,11 @@ torch::Tensor rollout_linear(torch::Tensor params, torch::Tensor env_blob, torch
     // v7 (MFMA, 16 members/block) serves the linear flagship geometry;
     // v6 covers MLP policies and off-geometry envs.
     if (H == 0 && R == 16 && O == 376 && A == 17 && !getenv("EVOTORCH_AMD_ROLLOUT_V6")) {
-        rollout_v7(params, env_blob, obs_stats_out, fitness, obs_dim, act_dim, rank, steps, alive_bonus, act_cost,
+        const int n_blocks7 = (n + 15) / 16;
+        auto stat_partials = torch::zeros({(int64_t)n_blocks7, 2 * (int64_t)O}, params.options());
+        rollout_v7(params, env_blob, stat_partials, fitness, obs_dim, act_dim, rank, steps, alive_bonus, act_cost,
                    init_seed, member_offset);
+        obs_stats_out.add_(stat_partials.sum(0));
         return fitness;
     }
 
@@ -504,11 +513,14 @@ torch::Tensor rollout_linear(torch::Tensor params, torch::Tensor env_blob, torch
 
     auto stream = at::cuda::getCurrentCUDAStream();
     const int n_blocks = (n + members - 1) / members;
+    auto stat_partials = torch::zeros({(int64_t)n_blocks * members, 2 * (int64_t)O}, params.options());
+    args.obs_stats_out = stat_partials.data_ptr<float>();
     if (members == 2) {
         launch_rollout<8, 2>(n_blocks, block, lds_bytes, stream, args);
     } else {
         launch_rollout<8, 1>(n_blocks, block, lds_bytes, stream, args);
     }
+    obs_stats_out.add_(stat_partials.sum(0));
     return fitness;
 }
 

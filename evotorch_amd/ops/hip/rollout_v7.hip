@@ -88,7 +88,7 @@ struct RolloutV7Args {
     const float* params;     // [n][A*O + A]
     const float* env_blob;   // V[R][O] U_T[R][O] D2_T[A][O] c[O] wr[O] mean[O] std[O]
     float* fitness_out;      // [n]
-    float* obs_stats_out;    // [2][O]
+    float* obs_stats_out;    // [blocks][2][O] per-block partials (deterministic sum in the launcher)
     int n_members;
     long member_offset;
     int obs_dim, act_dim, rank, steps;
@@ -389,12 +389,16 @@ __global__ __launch_bounds__(64 * kWaves, 1) void rollout_v7_kernel(RolloutV7Arg
         args.fitness_out[base_member + tid] =
             total + args.alive_bonus * (float)args.steps - args.act_cost * actsq_l[tid] / (float)A;
     }
+    // per-block stat slice: each col has exactly ONE writing lane per
+    // block, so plain stores suffice — float atomicAdd across blocks
+    // would make the obs-norm stats run-to-run order-dependent
+    float* stats = args.obs_stats_out + (int64_t)blockIdx.x * 2 * O;
 #pragma unroll
     for (int tw = 0; tw < kTilesPerWave; ++tw) {
         const int col = (wave * kTilesPerWave + tw) * 16 + c_col;
         if (col < O) {
-            atomicAdd(&args.obs_stats_out[col], stat_sum[tw]);
-            atomicAdd(&args.obs_stats_out[O + col], stat_sumsq[tw]);
+            stats[col] = stat_sum[tw];
+            stats[O + col] = stat_sumsq[tw];
         }
     }
 }
