@@ -389,16 +389,22 @@ __global__ __launch_bounds__(64 * kWaves, 1) void rollout_v7_kernel(RolloutV7Arg
         args.fitness_out[base_member + tid] =
             total + args.alive_bonus * (float)args.steps - args.act_cost * actsq_l[tid] / (float)A;
     }
-    // per-block stat slice: each col has exactly ONE writing lane per
-    // block, so plain stores suffice — float atomicAdd across blocks
-    // would make the obs-norm stats run-to-run order-dependent
+    // per-block stat slice: the four 16-lane groups of a wave share each
+    // col (they hold different member rows), so reduce across them with
+    // fixed-order shuffles, then lanes 0-15 store — plain stores, no
+    // float atomics, run-to-run deterministic
     float* stats = args.obs_stats_out + (int64_t)blockIdx.x * 2 * O;
 #pragma unroll
     for (int tw = 0; tw < kTilesPerWave; ++tw) {
+        float ss = stat_sum[tw], sq = stat_sumsq[tw];
+        ss += __shfl_down(ss, 32, 64);
+        ss += __shfl_down(ss, 16, 64);
+        sq += __shfl_down(sq, 32, 64);
+        sq += __shfl_down(sq, 16, 64);
         const int col = (wave * kTilesPerWave + tw) * 16 + c_col;
-        if (col < O) {
-            stats[col] = stat_sum[tw];
-            stats[O + col] = stat_sumsq[tw];
+        if (lane < 16 && col < O) {
+            stats[col] = ss;
+            stats[O + col] = sq;
         }
     }
 }
