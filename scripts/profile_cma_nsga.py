@@ -1,5 +1,7 @@
 """Round-2 profiling driver: CMA-ES d=4096 and NSGA-II popsize-32k
 per-kernel attribution."""
+import os, sys
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import sys
 import time
 
