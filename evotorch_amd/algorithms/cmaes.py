@@ -246,11 +246,14 @@ class CMAES(SearchAlgorithm, SinglePopulationAlgorithmMixin):
             self._C = (1.0 + c1 * delta_hs - c1 - cmu * w_adj.sum()) * self._C + c1 * self._p_c**2 + cmu * rank_mu
             self._C = torch.clamp(self._C, min=1e-20)
         else:
-            # rank-μ as a GEMM: Yᵀ diag(w) Y  (rocBLAS; MFMA-shaped)
-            rank_mu = (y * w_adj.unsqueeze(-1)).T @ y
-            rank_one = torch.outer(self._p_c, self._p_c)
-            self._C = (1.0 + c1 * delta_hs - c1 - cmu * w_adj.sum()) * self._C + c1 * rank_one + cmu * rank_mu
-            self._C = 0.5 * (self._C + self._C.T)
+            # fused K5 kernel: one pass over C applying all three terms
+            # with exact symmetry (ops/hip/cma.hip; eager torch chain on
+            # CPU); the hs stall correction and Σw stay device-side.
+            from .. import ops
+
+            self._C = self._C.contiguous()
+            ops.cma_update_c_(self._C, y, w_adj, self._p_c, torch.as_tensor(hs_f, device=y.device),
+                              c1=float(c1), cmu=float(cmu), cc=float(cc))
 
     def decompose_C(self):
         """Refresh the Cholesky factor A (amortized over

@@ -6,6 +6,7 @@ See `evotorch_amd/ops/dispatch.py` for the routing contract and
 
 from .dispatch import (
     affine_from_noise,
+    cma_update_c_,
     clipup_step_,
     domination_counts,
     es_gradients,
@@ -20,6 +21,7 @@ from .dispatch import (
 
 __all__ = [
     "affine_from_noise",
+    "cma_update_c_",
     "clipup_step_",
     "es_gradients",
     "fused_adam_step_",

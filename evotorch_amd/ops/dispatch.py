@@ -32,6 +32,7 @@ __all__ = [
     "fused_adam_step_",
     "pareto_ranks",
     "domination_counts",
+    "cma_update_c_",
 ]
 
 _hip_module = None
@@ -363,3 +364,42 @@ def pareto_ranks(utils: torch.Tensor) -> torch.Tensor:
 
     ranks, _ = _compute_pareto_ranks_eager(utils, crowdsort=False)
     return ranks
+
+
+# ============================================================================
+# K5 — fused CMA-ES covariance update
+# ============================================================================
+
+
+def cma_update_c_(
+    C: torch.Tensor,
+    y: torch.Tensor,
+    w_adj: torch.Tensor,
+    p_c: torch.Tensor,
+    hs_f: torch.Tensor,
+    *,
+    c1: float,
+    cmu: float,
+    cc: float,
+) -> torch.Tensor:
+    """In-place C = scale·C + c1·pc pcᵀ + cμ·Yᵀdiag(w)Y, exactly
+    symmetric, with scale = 1 + c1·(1−hs)·cc·(2−cc) − c1 − cμ·Σw computed
+    from DEVICE scalars (no host sync per generation).
+
+    On GPU this is ONE fused pass over C (ops/hip/cma.hip — see that file
+    for why fp32 VALU, not MFMA, is the right tool at CMA-ES shapes); the
+    eager reference is the reference-parity torch chain (cmaes.py:547-553)
+    plus explicit symmetrization."""
+    wsum = w_adj.sum().to(torch.float32)
+    if C.device.type == "cuda" and not _allow_eager_on_gpu():
+        mod = hip_required()
+        mod.cma_update_c(C, y.contiguous().to(torch.float32), w_adj.to(torch.float32), p_c.to(torch.float32),
+                         hs_f.reshape(1).to(torch.float32), wsum.reshape(1), float(c1), float(cmu), float(cc))
+        return C
+    delta_hs = (1.0 - hs_f) * cc * (2.0 - cc)
+    scale = 1.0 + c1 * delta_hs - c1 - cmu * wsum
+    rank_mu = (y * w_adj.unsqueeze(-1)).T @ y
+    rank_one = torch.outer(p_c, p_c)
+    C.mul_(scale).add_(rank_one, alpha=c1).add_(rank_mu, alpha=cmu)
+    C.copy_(0.5 * (C + C.T))
+    return C

@@ -16,6 +16,8 @@ void adam_step(torch::Tensor step_out, torch::Tensor grad, torch::Tensor m, torc
 torch::Tensor rollout_linear(torch::Tensor params, torch::Tensor env_blob, torch::Tensor obs_stats_out,
                              int64_t obs_dim, int64_t act_dim, int64_t rank, int64_t steps, double alive_bonus,
                              double act_cost, int64_t init_seed, int64_t member_offset, int64_t policy_hidden);
+void cma_update_c(torch::Tensor C, torch::Tensor Y, torch::Tensor w, torch::Tensor pc, torch::Tensor hs_f,
+                  torch::Tensor wsum, double c1, double cmu, double cc);
 torch::Tensor domination_counts(torch::Tensor utils);
 torch::Tensor pareto_ranks(torch::Tensor utils);
 }  // namespace ea
@@ -39,6 +41,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           pybind11::arg("obs_dim"), pybind11::arg("act_dim"), pybind11::arg("rank"), pybind11::arg("steps"),
           pybind11::arg("alive_bonus"), pybind11::arg("act_cost"), pybind11::arg("init_seed"),
           pybind11::arg("member_offset"), pybind11::arg("policy_hidden") = 0);
+    m.def("cma_update_c", &ea::cma_update_c,
+          "K5: fused CMA-ES covariance update (scale*C + c1*pc pc^T + cmu*Y^T diag(w) Y, exact symmetry)");
     m.def("domination_counts", &ea::domination_counts, "K7: NSGA-II domination counts (no N x N matrix)");
     m.def("pareto_ranks", &ea::pareto_ranks, "K7: NSGA-II non-dominated sorting by front peeling");
 }

@@ -53,7 +53,7 @@ def main():
             return torch.stack([f1, f2], dim=-1)
 
         for N in (8192, 32768):
-            prob = Problem(["min", "min"], multi, solution_length=64, initial_bounds=(0, 1), seed=2, device="cuda:0")
+            prob = Problem(["min", "min"], multi, solution_length=64, initial_bounds=(0, 1), bounds=(0.0, 1.0), seed=2, device="cuda:0")
             ga = GeneticAlgorithm(
                 prob, popsize=N,
                 operators=[

@@ -27,6 +27,7 @@ setup(
                 os.path.join(SRC, "rollout.hip"),
                 os.path.join(SRC, "rollout_v7.hip"),
                 os.path.join(SRC, "pareto.hip"),
+                os.path.join(SRC, "cma.hip"),
             ],
             extra_compile_args={
                 "cxx": ["-O3", "-std=c++17"],

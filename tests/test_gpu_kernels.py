@@ -521,3 +521,45 @@ def test_sharded_world1_pgpe_with_overlap_improves():
     assert evals1[-1] < evals1[0]
     assert evals1 == evals2, "SPMD GPU trajectory must be deterministic"
     assert torch.equal(c1, c2)
+
+
+@requires_gpu
+def test_cma_update_c_matches_eager():
+    """K5 fused covariance update vs the eager torch chain; exact symmetry
+    by construction."""
+    from evotorch_amd.ops import cma_update_c_
+
+    torch.manual_seed(4)
+    d, lam = 300, 48
+    B = torch.randn(d, d)
+    C0 = (B @ B.T) / d + torch.eye(d)
+    y = torch.randn(lam, d)
+    w = torch.cat([torch.rand(lam // 2) + 0.1, -0.1 * torch.rand(lam - lam // 2)])
+    pc = torch.randn(d)
+    for hs in (0.0, 1.0):
+        hs_t = torch.tensor(hs)
+        C_cpu = C0.clone()
+        cma_update_c_(C_cpu, y, w, pc, hs_t, c1=0.02, cmu=0.05, cc=0.1)
+        C_gpu = C0.clone().cuda()
+        cma_update_c_(C_gpu, y.cuda(), w.cuda(), pc.cuda(), hs_t.cuda(), c1=0.02, cmu=0.05, cc=0.1)
+        assert torch.allclose(C_gpu.cpu(), C_cpu, rtol=1e-4, atol=1e-4)
+        assert torch.equal(C_gpu, C_gpu.T), "fused update must be exactly symmetric"
+
+
+@requires_gpu
+def test_cma_update_c_odd_dim_tail():
+    """d not a multiple of the 64-tile: boundary tiles guarded."""
+    from evotorch_amd.ops import cma_update_c_
+
+    torch.manual_seed(9)
+    d, lam = 130, 20
+    C0 = torch.eye(d) * 2.0
+    y = torch.randn(lam, d)
+    w = torch.rand(lam) / lam
+    pc = torch.randn(d)
+    hs = torch.tensor(1.0)
+    C_cpu = C0.clone()
+    cma_update_c_(C_cpu, y, w, pc, hs, c1=0.1, cmu=0.2, cc=0.3)
+    C_gpu = C0.clone().cuda()
+    cma_update_c_(C_gpu, y.cuda(), w.cuda(), pc.cuda(), hs.cuda(), c1=0.1, cmu=0.2, cc=0.3)
+    assert torch.allclose(C_gpu.cpu(), C_cpu, rtol=1e-4, atol=1e-4)
