@@ -75,19 +75,22 @@ def _domination_matrix(utils: torch.Tensor) -> torch.Tensor:
     return ge & gt
 
 
-def _compute_pareto_ranks(utils: torch.Tensor, crowdsort: bool = True):
+def _compute_pareto_ranks(utils: torch.Tensor, crowdsort: bool = True, min_assigned: Optional[int] = None):
     """Device-dispatching wrapper: HIP front-peel kernels on ROCm (K7),
-    eager masked reductions on CPU."""
+    eager masked reductions on CPU. `min_assigned` allows the peel to
+    stop early once that many solutions hold final ranks (exact for
+    take_best(n): the boundary front is always fully peeled, so the
+    first n of the (rank, crowding) order are unchanged)."""
     if utils.device.type == "cuda":
         from . import ops
 
-        ranks = ops.pareto_ranks(utils)
+        ranks = ops.pareto_ranks(utils, min_assigned)
         crowd = _crowding_distances(utils, ranks) if crowdsort else None
         return ranks, crowd
-    return _compute_pareto_ranks_eager(utils, crowdsort)
+    return _compute_pareto_ranks_eager(utils, crowdsort, min_assigned=min_assigned)
 
 
-def _compute_pareto_ranks_eager(utils: torch.Tensor, crowdsort: bool = True):
+def _compute_pareto_ranks_eager(utils: torch.Tensor, crowdsort: bool = True, min_assigned: Optional[int] = None):
     """Non-dominated ranking by iterative front peeling over domination
     counts — formulated as masked reductions so the only host sync is the
     loop-termination check (SURVEY.md §7 hard-parts note on K7).
@@ -95,6 +98,7 @@ def _compute_pareto_ranks_eager(utils: torch.Tensor, crowdsort: bool = True):
     Returns (ranks, crowd) where ranks[i] is the index of i's pareto front
     (0 = best front) and crowd[i] is the crowding distance (or None)."""
     n = utils.shape[0]
+    stop_at = n if (min_assigned is None or min_assigned <= 0) else min(int(min_assigned), n)
     dom = _domination_matrix(utils)
     dom_count = dom.sum(dim=0).to(torch.int64)  # how many dominate me
     ranks = torch.full((n,), -1, dtype=torch.int64, device=utils.device)
@@ -102,6 +106,9 @@ def _compute_pareto_ranks_eager(utils: torch.Tensor, crowdsort: bool = True):
     front_index = 0
     domf = dom.to(torch.int64)
     while not bool(assigned.all()):
+        if int(assigned.sum()) >= stop_at:
+            ranks[~assigned] = front_index  # beyond-last lump (early stop)
+            break
         current = (dom_count == 0) & (~assigned)
         if not bool(current.any()):
             # numerical corner: break ties by assigning the rest to one front
@@ -413,9 +420,9 @@ class SolutionBatch(Serializable):
         """Best n solutions as a new batch (pareto-based for multi-objective
         when obj_index is None); with n omitted, the single best Solution."""
         if n is None:
-            indices = self.argsort(obj_index=obj_index)
+            indices = self.argsort(obj_index=obj_index, min_assigned=1)
             return self[int(indices[0])]
-        indices = self.argsort(obj_index=obj_index)[:n]
+        indices = self.argsort(obj_index=obj_index, min_assigned=int(n))[:n]
         return self.take(indices)
 
     def split(self, num_pieces: Optional[int] = None, *, max_size: Optional[int] = None) -> "SolutionBatchPieces":
@@ -463,11 +470,13 @@ class SolutionBatch(Serializable):
                 utils[:, j] = -utils[:, j]
         return utils
 
-    def argsort(self, obj_index: Optional[int] = None) -> torch.Tensor:
+    def argsort(self, obj_index: Optional[int] = None, *, min_assigned: Optional[int] = None) -> torch.Tensor:
         """Indices from best to worst. For a multi-objective batch without
-        obj_index, sorts by (pareto rank, -crowding distance)."""
+        obj_index, sorts by (pareto rank, -crowding distance);
+        `min_assigned` lets the non-dominated sort stop after the fronts
+        covering that many solutions (exact for head-of-order uses)."""
         if self._num_objs > 1 and obj_index is None:
-            ranks, crowd = self.compute_pareto_ranks(crowdsort=True)
+            ranks, crowd = self.compute_pareto_ranks(crowdsort=True, min_assigned=min_assigned)
             # lexicographic (pareto rank asc, crowding distance desc) via
             # integer keys: position-in-crowd-order breaks ties within a front
             n = ranks.shape[0]
@@ -491,9 +500,9 @@ class SolutionBatch(Serializable):
         evals = self._evals[:, j]
         return evals.argmin() if self._senses[j] == "max" else evals.argmax()
 
-    def compute_pareto_ranks(self, crowdsort: bool = True):
+    def compute_pareto_ranks(self, crowdsort: bool = True, min_assigned: Optional[int] = None):
         utils = self._utils_for_sorting()
-        return _compute_pareto_ranks(utils, crowdsort=crowdsort)
+        return _compute_pareto_ranks(utils, crowdsort=crowdsort, min_assigned=min_assigned)
 
     def arg_pareto_sort(self, crowdsort: bool = True):
         """List of fronts (each a tensor of indices), best front first

@@ -355,14 +355,17 @@ def domination_counts(utils: torch.Tensor) -> torch.Tensor:
     return dom.sum(dim=0).to(torch.int64)
 
 
-def pareto_ranks(utils: torch.Tensor) -> torch.Tensor:
+def pareto_ranks(utils: torch.Tensor, min_assigned: Optional[int] = None) -> torch.Tensor:
     """Front index per solution (0 = non-dominated front), computed on GPU
-    by count + front peeling without the N x N matrix."""
+    by count + front peeling without the N x N matrix. With
+    `min_assigned`, peeling stops once that many solutions hold final
+    ranks (take_best(n) only needs the fronts crossing n); the rest get a
+    shared beyond-last rank."""
     if utils.device.type == "cuda" and not _allow_eager_on_gpu():
-        return hip_required().pareto_ranks(utils)
+        return hip_required().pareto_ranks(utils, int(min_assigned or 0))
     from ..core import _compute_pareto_ranks_eager
 
-    ranks, _ = _compute_pareto_ranks_eager(utils, crowdsort=False)
+    ranks, _ = _compute_pareto_ranks_eager(utils, crowdsort=False, min_assigned=min_assigned)
     return ranks
 
 

@@ -19,7 +19,7 @@ torch::Tensor rollout_linear(torch::Tensor params, torch::Tensor env_blob, torch
 void cma_update_c(torch::Tensor C, torch::Tensor Y, torch::Tensor w, torch::Tensor pc, torch::Tensor hs_f,
                   torch::Tensor wsum, double c1, double cmu, double cc);
 torch::Tensor domination_counts(torch::Tensor utils);
-torch::Tensor pareto_ranks(torch::Tensor utils);
+torch::Tensor pareto_ranks(torch::Tensor utils, int64_t min_assigned);
 }  // namespace ea
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -44,5 +44,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("cma_update_c", &ea::cma_update_c,
           "K5: fused CMA-ES covariance update (scale*C + c1*pc pc^T + cmu*Y^T diag(w) Y, exact symmetry)");
     m.def("domination_counts", &ea::domination_counts, "K7: NSGA-II domination counts (no N x N matrix)");
-    m.def("pareto_ranks", &ea::pareto_ranks, "K7: NSGA-II non-dominated sorting by front peeling");
+    m.def("pareto_ranks", &ea::pareto_ranks, "K7: NSGA-II non-dominated sorting by front peeling",
+          py::arg("utils"), py::arg("min_assigned") = 0);
 }

@@ -129,11 +129,15 @@ torch::Tensor domination_counts(torch::Tensor utils) {
     return counts;
 }
 
-torch::Tensor pareto_ranks(torch::Tensor utils) {
+torch::Tensor pareto_ranks(torch::Tensor utils, int64_t min_assigned) {
+    // min_assigned > 0: stop peeling once that many solutions hold final
+    // ranks (take_best(n) only needs the fronts crossing n); stragglers
+    // get a beyond-last rank. 0 = full sort.
     TORCH_CHECK(utils.is_cuda() && utils.dim() == 2, "utils must be a 2-D ROCm tensor");
     auto utils_f = utils.to(torch::kFloat32).contiguous();
     const int64_t n = utils_f.size(0);
     const int m = (int)utils_f.size(1);
+    if (min_assigned <= 0 || min_assigned > n) min_assigned = n;
     auto counts = domination_counts(utils_f);
     auto ranks = torch::full({n}, -1, utils_f.options().dtype(torch::kInt64));
     const int threads = 256;
@@ -159,6 +163,12 @@ torch::Tensor pareto_ranks(torch::Tensor utils) {
         }
         const int64_t remaining = (counts >= 0).sum().item<int64_t>();  // one sync per batch
         if (remaining == 0) break;
+        if (n - remaining >= min_assigned) {
+            // enough solutions carry final ranks; lump the rest one past
+            // the last peeled front (they sort after every real front)
+            ranks.masked_fill_(counts >= 0, front_index);
+            break;
+        }
         // numerical corner: no zero-count candidates among the remaining —
         // assign the stragglers and stop
         const int64_t candidates = (counts == 0).sum().item<int64_t>();
