@@ -6,8 +6,9 @@ observation normalization (RunningNorm here plays the RunningStat role),
 alive-bonus removal via `decrease_rewards_by`, episode/interaction
 counters, and `to_policy` exporting ObsNorm + ActClip wrapped modules.
 
-gymnasium is an optional dependency (absent in the offline image); the
-import is deferred to construction time.
+Environments come from real gymnasium when installed, otherwise from the
+vendored classic-control registry (`gym_compat`) — the rollout loop runs
+real environment dynamics either way.
 """
 
 from typing import Callable, Optional, Union
@@ -54,13 +55,22 @@ ActClipWrapperModule = ActClipLayer
 
 def ensure_space_types(env) -> None:
     """Require Box observation and action spaces (reference
-    gymne.py: ensure_space_types)."""
-    import gymnasium as gym
+    gymne.py: ensure_space_types). Accepts both real gymnasium Box spaces
+    and the vendored gym_compat.Box."""
+    from . import gym_compat
 
-    if not isinstance(env.observation_space, gym.spaces.Box):
-        raise TypeError(f"Unsupported observation space {env.observation_space!r}: expected gymnasium.spaces.Box")
-    if not isinstance(env.action_space, gym.spaces.Box):
-        raise TypeError(f"Unsupported action space {env.action_space!r}: expected gymnasium.spaces.Box")
+    box_types = [gym_compat.Box]
+    try:
+        import gymnasium as gym
+
+        box_types.append(gym.spaces.Box)
+    except ImportError:
+        pass
+    box_types = tuple(box_types)
+    if not isinstance(env.observation_space, box_types):
+        raise TypeError(f"Unsupported observation space {env.observation_space!r}: expected a Box space")
+    if not isinstance(env.action_space, box_types):
+        raise TypeError(f"Unsupported action space {env.action_space!r}: expected a Box space")
 
 
 def reset_env(env):
@@ -145,10 +155,6 @@ class GymNE(NEProblem):
             import warnings
 
             warnings.warn("Ignoring Ray-era actor argument(s): parallelize via torchrun + RCCL", stacklevel=2)
-        try:
-            import gymnasium  # noqa: F401
-        except ImportError as e:
-            raise ImportError("GymNE requires gymnasium, which is not installed in this image; use VecEnvNE with a torch-native env instead") from e
         if env is None:
             env = env_name
         if env is None:
@@ -180,10 +186,13 @@ class GymNE(NEProblem):
         return {"obs_length": self._obs_dim, "act_length": self._act_dim, "obs_space": getattr(self._env, "observation_space", None), "act_space": self._act_space}
 
     def _make_env(self):
-        import gymnasium as gym
+        # real gymnasium when installed, the vendored classic-control
+        # registry (gym_compat) otherwise — the rollout loop runs real
+        # environment dynamics either way
+        from . import gym_compat
 
         if isinstance(self._env_def, str):
-            return gym.make(self._env_def, **self._env_config)
+            return gym_compat.make(self._env_def, **self._env_config)
         return self._env_def(**self._env_config)
 
     @property

@@ -80,9 +80,10 @@ class SyntheticTorchEnv:
 
 
 class GymVectorEnvAdapter:
-    """Wraps a gymnasium (Sync/Async)VectorEnv into the torch contract
-    (reference net/vecrl.py:362 TorchWrapper). Import-guarded: gymnasium
-    is optional and absent in the offline image."""
+    """Wraps a vector env with the gymnasium 5-tuple step API (real
+    gymnasium's, or gym_compat.SyncVectorEnv's with per-env autoreset +
+    info batching) into the torch contract (reference net/vecrl.py:362
+    TorchWrapper)."""
 
     def __init__(self, env, *, device="cpu"):
         import numpy as np  # noqa: F401
@@ -143,9 +144,9 @@ class VecEnvNE(NEProblem):
         if isinstance(env, str):
             env_name = env
             def factory(num_envs: int):
-                import gymnasium as gym
+                from . import gym_compat
 
-                return GymVectorEnvAdapter(gym.make_vec(env_name, num_envs=num_envs, **env_config), device=device or "cpu")
+                return GymVectorEnvAdapter(gym_compat.make_vec(env_name, num_envs=num_envs, **env_config), device=device or "cpu")
 
             self._env_factory = factory
             self._env = None
@@ -444,19 +445,21 @@ def make_brax_env(env_name: str, **kwargs):
 
 
 def make_gym_env(env_name: str, **kwargs):
-    """Instantiate a single gymnasium env by name (reference
-    vecrl.py:668)."""
-    import gymnasium as gym
+    """Instantiate a single env by name (reference vecrl.py:668): real
+    gymnasium when installed, the vendored classic-control registry
+    otherwise."""
+    from . import gym_compat
 
-    return gym.make(env_name, **kwargs)
+    return gym_compat.make(env_name, **kwargs)
 
 
 def make_vector_env(env_name: str, *, num_envs: int, **kwargs):
-    """Batched env by name: a gymnasium vector env wrapped into the torch
-    contract used by VecEnvNE (reference vecrl.py:764)."""
-    import gymnasium as gym
+    """Batched env by name, wrapped into the torch contract used by
+    VecEnvNE (reference vecrl.py:764): real gymnasium when installed,
+    the vendored autoresetting SyncVectorEnv otherwise."""
+    from . import gym_compat
 
-    vec = gym.make_vec(env_name, num_envs=num_envs, **kwargs)
+    vec = gym_compat.make_vec(env_name, num_envs=num_envs, **kwargs)
     return GymVectorEnvAdapter(vec)
 
 
