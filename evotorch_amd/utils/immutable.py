@@ -110,6 +110,9 @@ def as_immutable(x: Any) -> Any:
         return x
     if isinstance(x, torch.Tensor):
         return as_read_only_tensor(x.clone())
+    if hasattr(x, "get_read_only_view") and hasattr(x, "is_read_only"):
+        # TensorFrame (and compatible containers): immutable = read-only view
+        return x if x.is_read_only else x.get_read_only_view()
     if isinstance(x, np.ndarray):
         if x.dtype == object:
             return ImmutableList(list(x))

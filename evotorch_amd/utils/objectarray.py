@@ -104,7 +104,7 @@ class ObjectArray:
 
     def _check_writable(self):
         if self._read_only:
-            raise RuntimeError("This ObjectArray is read-only")
+            raise ValueError("This ObjectArray is read-only")
 
     def __getitem__(self, i):
         if isinstance(i, slice):
@@ -153,10 +153,41 @@ class ObjectArray:
             yield self[i]
 
     def clone(self, *, memo: Optional[dict] = None) -> "ObjectArray":
+        """Deep clone: the elements themselves are cloned (reference
+        objectarray.py:404 — `x.clone()[0] is not x[0]` for container
+        elements), and the clone is always writable."""
+        if memo is None:
+            memo = {}
+        if id(self) in memo:
+            return memo[id(self)]
         result = ObjectArray(len(self))
+        memo[id(self)] = result
         for i in range(len(self)):
-            result._data[i] = self._data[self._start + i]
+            item = self._data[self._start + i]
+            result._data[i] = as_immutable(mutable_copy(item)) if item is not None else None
         return result
+
+    def __copy__(self) -> "ObjectArray":
+        return self.clone()
+
+    def __deepcopy__(self, memo: Optional[dict]) -> "ObjectArray":
+        return self.clone(memo=memo if memo is not None else {})
+
+    def untyped_storage(self):
+        return self.storage()
+
+    def storage(self):
+        """torch-like storage handle exposing data_ptr() (reference
+        objectarray.py:479; used by tools.storage_ptr)."""
+
+        class _Storage:
+            def __init__(self, ptr):
+                self._ptr = ptr
+
+            def data_ptr(self):
+                return self._ptr
+
+        return _Storage(self._data.ctypes.data)
 
     def numpy(self) -> np.ndarray:
         out = np.empty(len(self), dtype=object)

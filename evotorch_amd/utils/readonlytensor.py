@@ -14,7 +14,7 @@ __all__ = ["ReadOnlyTensor", "read_only_tensor", "as_read_only_tensor"]
 
 
 def _err(op: str):
-    raise RuntimeError(f"ReadOnlyTensor does not allow the in-place/mutating operation {op!r}. Use `.clone()` to obtain a mutable copy.")
+    raise TypeError(f"ReadOnlyTensor does not allow the in-place/mutating operation {op!r}. Use `.clone()` to obtain a mutable copy.")
 
 
 class ReadOnlyTensor(torch.Tensor):

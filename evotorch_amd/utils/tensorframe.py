@@ -17,30 +17,55 @@ __all__ = ["TensorFrame"]
 
 
 class _PickIndexer:
+    """frame.pick[rows] / frame.pick[rows, cols] — getter returns a new
+    (copied) frame; SETTER writes in place into the frame's own storage
+    (reference tensorframe.py:1270: `tbl.pick[1:4, "X"] = ...`)."""
+
     def __init__(self, frame: "TensorFrame"):
         self._frame = frame
+
+    @staticmethod
+    def _normalize_cols(frame: "TensorFrame", cols) -> list:
+        if isinstance(cols, str):
+            return [cols]
+        if isinstance(cols, slice):
+            return list(frame._columns.keys())[cols]
+        return list(cols)
 
     def __getitem__(self, index) -> "TensorFrame":
         frame = self._frame
         if isinstance(index, tuple) and len(index) == 2:
             rows, cols = index
-            if isinstance(cols, str):
-                cols = [cols]
-            sub = TensorFrame({c: frame._columns[c] for c in cols})
+            names = self._normalize_cols(frame, cols)
+            sub = TensorFrame({c: frame._columns[c] for c in cols} if False else {c: frame._columns[c] for c in names})
             return sub.pick[rows]
         if isinstance(index, (int,)):
             index = slice(index, index + 1)
-        new_cols = {}
-        for name, col in frame._columns.items():
-            if isinstance(index, torch.Tensor) and index.dtype == torch.bool:
-                new_cols[name] = col[index]
-            else:
-                new_cols[name] = col[index]
-        return TensorFrame(new_cols)
+        return TensorFrame({name: col[index] for name, col in frame._columns.items()})
+
+    def __setitem__(self, index, value):
+        frame = self._frame
+        if frame._read_only:
+            raise TypeError("This TensorFrame is read-only")
+        if isinstance(index, tuple) and len(index) == 2:
+            rows, cols = index
+            names = self._normalize_cols(frame, cols)
+        else:
+            rows = index
+            names = list(frame._columns.keys())
+        if isinstance(value, TensorFrame):
+            value = value._columns
+        if isinstance(value, Mapping):
+            for name in names:
+                frame._columns[name][rows] = torch.as_tensor(value[name])
+            return
+        if len(names) != 1:
+            raise ValueError("Assigning a single tensor requires exactly one target column")
+        frame._columns[names[0]][rows] = torch.as_tensor(value)
 
 
 class TensorFrame(RecursivePrintable):
-    def __init__(self, data: Optional[Union[Mapping, "TensorFrame"]] = None, *, read_only: bool = False, device=None, **kwargs):
+    def __init__(self, data: Optional[Union[Mapping, "TensorFrame"]] = None, *, read_only: bool = False, device=None, _copy: bool = True, **kwargs):
         if isinstance(data, TensorFrame):
             columns = dict(data._columns)
         elif data is not None:
@@ -51,7 +76,13 @@ class TensorFrame(RecursivePrintable):
         self._columns = {}
         n = None
         for name, col in columns.items():
-            t = torch.as_tensor(col) if not isinstance(col, torch.Tensor) else col
+            if isinstance(col, torch.Tensor):
+                # own our storage: pandas-style isolation — in-place writes
+                # through pick[...] must never touch the caller's tensor
+                t = torch.Tensor.as_subclass(col, torch.Tensor)
+                t = t.clone() if _copy else t
+            else:
+                t = torch.as_tensor(col)
             if device is not None:
                 t = t.to(device)
             if t.ndim == 0:
@@ -65,6 +96,10 @@ class TensorFrame(RecursivePrintable):
                     raise ValueError(f"Column {name!r} has {t.shape[0]} rows, expected {n}")
             self._columns[str(name)] = t
         self._read_only = bool(read_only)
+        if read_only:
+            from .readonlytensor import as_read_only_tensor
+
+            self._columns = {k: as_read_only_tensor(v) for k, v in self._columns.items()}
 
     # -- basics --------------------------------------------------------------
 
@@ -94,7 +129,7 @@ class TensorFrame(RecursivePrintable):
 
     def __setitem__(self, name: str, value):
         if self._read_only:
-            raise RuntimeError("This TensorFrame is read-only")
+            raise TypeError("This TensorFrame is read-only")
         value = torch.as_tensor(value)
         if value.ndim == 0 and len(self) > 0:
             value = value.expand(len(self)).clone()
@@ -119,10 +154,12 @@ class TensorFrame(RecursivePrintable):
     # -- transforms -----------------------------------------------------------
 
     def with_columns(self, **kwargs) -> "TensorFrame":
+        """New frame with the given columns added or replaced; the
+        read-only flag is preserved (reference tensorframe.py)."""
         new = dict(self._columns)
         for k, v in kwargs.items():
             new[k] = torch.as_tensor(v)
-        return TensorFrame(new)
+        return TensorFrame(new, read_only=self._read_only)
 
     def without_columns(self, *names: str) -> "TensorFrame":
         return TensorFrame({k: v for k, v in self._columns.items() if k not in names})
@@ -151,6 +188,10 @@ class TensorFrame(RecursivePrintable):
     def vstack(self, other: "TensorFrame") -> "TensorFrame":
         if set(self.columns) != set(other.columns):
             raise ValueError("vstack requires identical column sets")
+        for k in self._columns:
+            a, b = self._columns[k], other._columns[k]
+            if a.ndim != b.ndim or a.shape[1:] != b.shape[1:]:
+                raise ValueError(f"Column {k!r}: row shapes {tuple(a.shape[1:])} vs {tuple(b.shape[1:])} cannot vstack")
         return TensorFrame({k: torch.cat([self._columns[k], other._columns[k]], dim=0) for k in self._columns})
 
     def each(self, fn: Callable, *, join: bool = False, override: bool = False) -> "TensorFrame":
@@ -235,10 +276,15 @@ class TensorFrame(RecursivePrintable):
         return out
 
     def get_read_only_view(self) -> "TensorFrame":
-        return TensorFrame(self._columns, read_only=True)
+        """Read-only view SHARING storage: mutating the original is
+        reflected, writes through the view raise TypeError (the columns
+        are ReadOnlyTensors)."""
+        return TensorFrame(self._columns, read_only=True, _copy=False)
 
     def clone(self, *, memo: Optional[dict] = None) -> "TensorFrame":
-        return TensorFrame({k: v.clone() for k, v in self._columns.items()})
+        """Mutable deep copy (cloning a read-only frame yields a writable
+        one, reference parity)."""
+        return TensorFrame({k: torch.Tensor.as_subclass(v, torch.Tensor).clone() for k, v in self._columns.items()}, _copy=False)
 
     def to_pandas(self):
         import pandas as pd

@@ -144,7 +144,7 @@ def test_tensorframe_each():
 
 def test_tensorframe_read_only():
     tf = TensorFrame({"a": torch.zeros(3)}).get_read_only_view()
-    with pytest.raises(RuntimeError):
+    with pytest.raises(TypeError):  # reference raises TypeError
         tf["b"] = torch.ones(3)
 
 
@@ -164,3 +164,144 @@ def test_tensorframe_pandas_style_methods():
     x = tf.as_tensor(5, to_work_with="a", broadcast_if_scalar=True)
     assert x.shape == (3,) and x.dtype == torch.float32
     assert tf.cpu().device == torch.device("cpu")
+
+
+# -- reference-parity behavior set (ported from the reference's
+# test_tensorframe.py / test_objectarray.py BEHAVIORS, original code) -------
+
+
+def test_tensorframe_pick_setter_isolated_from_source():
+    src = torch.tensor([1.0, 2, 3, 4, 5])
+    tf = TensorFrame({"X": src, "Y": src * 10})
+    tf.pick[1:4, "X"] = torch.tensor([-2.0, -3, -4])
+    assert tf.X.tolist() == [1, -2, -3, -4, 5]
+    assert src.tolist() == [1, 2, 3, 4, 5]  # ctor copied; source untouched
+
+
+@pytest.mark.parametrize("rhs_as_frame", [False, True])
+def test_tensorframe_pick_setter_multicolumn(rhs_as_frame):
+    src = torch.tensor([1.0, 2, 3, 4, 5])
+    tf = TensorFrame({"X": src.clone(), "Y": (src * 10).clone()})
+    rhs = {"X": torch.tensor([-2.0, -3, -4]), "Y": torch.tensor([-20.0, -30, -40])}
+    tf.pick[1:4, ["X", "Y"]] = TensorFrame(rhs) if rhs_as_frame else rhs
+    assert tf.X.tolist() == [1, -2, -3, -4, 5]
+    assert tf.Y.tolist() == [10, -20, -30, -40, 50]
+
+
+def test_tensorframe_each_under_vmap_with_scalar_column():
+    """The reference's batched-operations contract: a whole frame pipeline
+    (ctor + each with a dict row fn + scalar column) composes under an
+    outer torch.func.vmap."""
+    torch.manual_seed(0)
+    bx = torch.randn(2, 3, 2)
+    by = torch.randn(2, 3)
+
+    def run(x, y):
+        tf = TensorFrame(dict(X=x, Y=y, Z=True))
+
+        def per_row(row):
+            assert row["X"].shape == (2,)
+            assert row["Y"].shape == ()
+            assert row["Z"].shape == ()
+            return {"OUT": row["Z"] * (torch.max(row["X"]) + row["Y"])}
+
+        return tf.each(per_row).OUT
+
+    out = torch.func.vmap(run)(bx, by)
+    want = torch.func.vmap(torch.func.vmap(torch.max))(bx) + by
+    assert torch.allclose(out, want, atol=1e-5)
+
+
+def test_tensorframe_pick_column_slicers_and_bool_mask():
+    tf = TensorFrame(dict(A=[1.0, 2, 3, 4], B=[[10.0, 20], [30, 40], [50, 60], [70, 80]], C=[-1.0, -2, -3, -4]))
+    sub = tf.pick[[1, 3]]
+    assert sub.columns == ["A", "B", "C"] and sub.A.tolist() == [2, 4]
+    assert sub.B.tolist() == [[30, 40], [70, 80]]
+    sub = tf.pick[[1, 3], "A"]
+    assert sub.columns == ["A"]
+    for slicer in (slice(None, 2), tf.C > -3):
+        assert tf.pick[slicer, "A"].A.tolist() == [1, 2]
+        got = tf.pick[slicer, ["A", "C"]]
+        assert got.columns == ["A", "C"] and got.C.tolist() == [-1, -2]
+        full = tf.pick[slicer, slice(None)]
+        assert full.columns == ["A", "B", "C"]
+
+
+def test_tensorframe_vstack_failures():
+    a = TensorFrame(dict(A=[1.0, 2, 3], B=[4.0, 5, 6]))
+    with pytest.raises(ValueError):
+        a.vstack(TensorFrame(dict(A=[1.0, 2, 3], B=[[4.0, 5], [6, 7], [8, 9]])))
+    with pytest.raises(ValueError):
+        a.vstack(TensorFrame(dict(A=[1.0, 2, 3], C=[1.0, 2, 3])))
+
+
+def test_tensorframe_read_only_full_protocol():
+    tf = TensorFrame(dict(A=[1.0, 2, 3]))
+    tf["B"] = 4  # scalar broadcast
+    assert tf.B.tolist() == [4, 4, 4]
+    ro = tf.get_read_only_view()
+    with pytest.raises(TypeError):
+        ro["C"] = 5
+    with pytest.raises(TypeError):
+        ro["A"][:] = 2.0  # columns of a read-only view are read-only
+    cloned = ro.clone()
+    assert not cloned.is_read_only
+    cloned["C"] = 5
+    cloned["A"][0] = 9.0
+    # storing a frame in an ObjectArray makes it immutable (read-only view)
+    from evotorch_amd.utils.objectarray import ObjectArray
+
+    arr = ObjectArray(1)
+    arr[0] = cloned
+    assert arr[0].is_read_only
+    with pytest.raises(TypeError):
+        arr[0]["D"] = 10
+
+
+def test_tensorframe_with_columns_preserves_read_only():
+    tf = TensorFrame(dict(A=[1.0, 2, 3])).with_columns(B=[4.0, 5, 6], C=7)
+    assert tf.C.tolist() == [7, 7, 7]
+    ro = tf.get_read_only_view().with_columns(A=[10.0, 20, 30], Z=[100.0, 200, 300])
+    assert ro.is_read_only
+    assert ro.A.tolist() == [10, 20, 30] and ro.Z.tolist() == [100, 200, 300]
+
+
+def test_objectarray_copy_and_deepcopy_protocols():
+    from copy import copy, deepcopy
+
+    from evotorch_amd.utils.objectarray import ObjectArray
+
+    x = ObjectArray(10)
+    x[:] = [0 for _ in range(10)]
+    for maker in (copy, deepcopy, lambda a: a.clone()):
+        y = maker(x)
+        assert all(a == b for a, b in zip(x, y))
+        y[:] = [1 for _ in range(10)]
+        assert all(a != b for a, b in zip(x, y))  # independent storage
+
+
+def test_objectarray_clone_clones_elements():
+    from evotorch_amd.utils.objectarray import ObjectArray
+
+    x = ObjectArray(2)
+    x[0] = [1, 2]
+    x[1] = [3, 4]
+    y = x.clone()
+    assert x[0] == y[0] and x[1] == y[1]
+    assert x[0] is not y[0] and x[1] is not y[0]
+
+
+def test_objectarray_storage_ptr_shared_by_views():
+    from evotorch_amd.utils.misc import storage_ptr
+    from evotorch_amd.utils.objectarray import ObjectArray
+
+    x = ObjectArray(10)
+    x[:] = range(10)
+    y = x[3:5]
+    assert storage_ptr(x) == storage_ptr(y)
+    y[:] = [0, 0]
+    assert x[3] == 0 and x[4] == 0
+    ro = x.get_read_only_view()
+    assert storage_ptr(ro) == storage_ptr(x)
+    with pytest.raises(ValueError):
+        ro[0] = 9

@@ -142,7 +142,7 @@ def test_object_array_immutability_and_views():
     view[0] = "hello"
     assert arr[1] == "hello"
     ro = arr.get_read_only_view()
-    with pytest.raises(RuntimeError):
+    with pytest.raises(ValueError):
         ro[0] = 5
 
 
