@@ -420,4 +420,97 @@ void adam_step(torch::Tensor step_out, torch::Tensor grad, torch::Tensor m, torc
     });
 }
 
+
+// ---------------------------------------------------------------------------
+// K2: fused fitness ranking -> utility map (reference
+// tools/ranking.py:24-216). One launch replaces the torch chain
+// (rocPRIM radix sort + scatter + 3-4 elementwise maps): a single-block
+// bitonic sort over LDS (N <= 8192 padded to a power of two) followed by
+// the utility map written back through the sorted index payload.
+// method: 0 = centered (rank/(n-1) - 0.5), 1 = linear (rank/(n-1)),
+//         2 = nes (log-utilities, normalized to sum to ~0).
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(1024) void fused_rank_kernel(const float* __restrict__ fit, float* __restrict__ out,
+                                                          int n, int p, int method, bool higher_better) {
+    extern __shared__ unsigned char rank_lds[];
+    float* keys = reinterpret_cast<float*>(rank_lds);
+    int* idx = reinterpret_cast<int*>(keys + p);
+    float* scratch = reinterpret_cast<float*>(idx + p);  // [32] for the nes sum
+
+    const int tid = threadIdx.x;
+    for (int i = tid; i < p; i += blockDim.x) {
+        keys[i] = (i < n) ? (higher_better ? fit[i] : -fit[i]) : INFINITY;
+        idx[i] = i;
+    }
+    __syncthreads();
+    // bitonic sort ascending: position 0 = worst
+    for (int k = 2; k <= p; k <<= 1) {
+        for (int j = k >> 1; j > 0; j >>= 1) {
+            for (int i = tid; i < p; i += blockDim.x) {
+                const int ixj = i ^ j;
+                if (ixj > i) {
+                    const bool up = (i & k) == 0;
+                    const float a = keys[i], b = keys[ixj];
+                    // NaN-robust: order NaNs last so they rank as best-key
+                    // (matches torch argsort's NaN-is-largest behavior)
+                    const bool swap = up ? (b < a || (isnan(a) && !isnan(b)))
+                                         : (a < b || (isnan(b) && !isnan(a)));
+                    if (swap) {
+                        keys[i] = b; keys[ixj] = a;
+                        const int t = idx[i]; idx[i] = idx[ixj]; idx[ixj] = t;
+                    }
+                }
+            }
+            __syncthreads();
+        }
+    }
+    if (method == 2) {
+        // NES log-utilities need their global sum before the final map
+        float partial = 0.0f;
+        for (int i = tid; i < n; i += blockDim.x) {
+            const float rank_from_best = (float)(n - i);
+            const float u = fmaxf(__logf((float)n / 2.0f + 1.0f) - __logf(rank_from_best), 0.0f);
+            keys[i] = u;  // reuse the key slot for the raw utility
+            partial += u;
+        }
+        __syncthreads();
+        // block sum (1024 threads -> 32 warp partials -> one value)
+        const int lane = tid & 63, wave = tid >> 6;
+        for (int off = 32; off > 0; off >>= 1) partial += __shfl_down(partial, off, 64);
+        if (lane == 0) scratch[wave] = partial;
+        __syncthreads();
+        if (tid == 0) {
+            float total = 0.0f;
+            for (int w = 0; w < (int)(blockDim.x >> 6); ++w) total += scratch[w];
+            scratch[0] = total;
+        }
+        __syncthreads();
+        const float denom = scratch[0];
+        const float inv_n = 1.0f / (float)n;
+        for (int i = tid; i < n; i += blockDim.x) out[idx[i]] = keys[i] / denom - inv_n;
+        return;
+    }
+    const float inv = (n > 1) ? 1.0f / (float)(n - 1) : 0.0f;
+    for (int i = tid; i < n; i += blockDim.x) {
+        const float r = (float)i;
+        out[idx[i]] = (method == 0) ? (r * inv - 0.5f) : (r * inv);
+    }
+}
+
+torch::Tensor fused_rank(torch::Tensor fitnesses, int64_t method, bool higher_better) {
+    TORCH_CHECK(fitnesses.is_cuda() && fitnesses.dim() == 1, "fused_rank expects a 1-D ROCm tensor");
+    auto fit = fitnesses.to(torch::kFloat32).contiguous();
+    const int n = (int)fit.size(0);
+    TORCH_CHECK(n >= 1 && n <= 8192, "fused_rank supports 1 <= n <= 8192");
+    int p = 1;
+    while (p < n) p <<= 1;
+    auto out = torch::empty({n}, fit.options());
+    const size_t lds = (size_t)p * 8 + 32 * 4;
+    auto stream = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(fused_rank_kernel, dim3(1), dim3(1024), lds, stream, fit.data_ptr<float>(),
+                       out.data_ptr<float>(), n, p, (int)method, higher_better);
+    return out;
+}
+
 }  // namespace ea

@@ -54,6 +54,14 @@ typedef __attribute__((ext_vector_type(2))) __bf16 bf16x2;
 __device__ __forceinline__ __bf16 f2b(float v) { return (__bf16)v; }
 __device__ __forceinline__ float b2f(__bf16 v) { return (float)v; }
 
+// tanh via the hardware exp unit (~1e-6 rel error, far below the bf16
+// quantization of every operand; libm tanhf is branchy on the hot path)
+__device__ __forceinline__ float tanh_fast6(float x) {
+    const float xc = fminf(fmaxf(x, -15.0f), 15.0f);
+    const float e = __expf(2.0f * xc);
+    return (e - 1.0f) * __builtin_amdgcn_rcpf(e + 1.0f);
+}
+
 // runtime member-index -> pointer select WITHOUT a runtime-indexed array
 // (which would demote to scratch memory — common-mistake #20); for
 // kMembers<=2 this folds to a single v_cndmask per operand.
@@ -356,7 +364,7 @@ __global__ __launch_bounds__(512, 2) void rollout_linear_kernel(RolloutArgs args
                     if (descs[r].member == m) actsq_part[m] = fmaf(a, a, actsq_part[m]);      \
                 }                                                                             \
             } else if (descs[r].kind == 2) {                                                  \
-                *descs[r].dst = f2b(tanhf(acc[r] + *descs[r].bias));                          \
+                *descs[r].dst = f2b(tanh_fast6(acc[r] + *descs[r].bias));                          \
             } else {                                                                          \
                 *descs[r].dst = f2b(acc[r]);                                                  \
             }                                                                                 \
@@ -394,7 +402,7 @@ __global__ __launch_bounds__(512, 2) void rollout_linear_kernel(RolloutArgs args
                 dacc = __builtin_amdgcn_fdot2_f32_bf16(
                     D2_pair[p * O + jo], *reinterpret_cast<const bf16x2*>(av + 2 * p), dacc, false);
             }
-            const float o_new = tanhf(uacc + dacc);
+            const float o_new = tanh_fast6(uacc + dacc);
 #pragma unroll
             for (int mm = 0; mm < kMembers; ++mm) {
                 if (m == mm) fit_part[mm] = fmaf(wr_l[jo], o_new, fit_part[mm]);

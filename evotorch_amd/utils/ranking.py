@@ -92,8 +92,15 @@ def ranking_method_exists(method: str) -> bool:
     return method in rankers
 
 
+_FUSED_METHODS = {"centered": 0, "linear": 1, "nes": 2}
+
+
 def rank(fitnesses: torch.Tensor, ranking_method: Optional[str] = "raw", *, higher_is_better: bool) -> torch.Tensor:
-    """Apply the named ranking method; ``None`` means raw."""
+    """Apply the named ranking method; ``None`` means raw.
+
+    On ROCm, sort-based methods for 1-D fitness vectors up to 8192 run as
+    ONE fused HIP kernel (bitonic sort + utility map, K2 in SURVEY.md
+    §2.9) instead of the ~6-dispatch torch chain."""
     if ranking_method is None:
         ranking_method = "raw"
     try:
@@ -101,4 +108,15 @@ def rank(fitnesses: torch.Tensor, ranking_method: Optional[str] = "raw", *, high
     except KeyError:
         raise ValueError(f"Unknown ranking method {ranking_method!r}; available: {sorted(rankers)}") from None
     fitnesses = torch.as_tensor(fitnesses)
+    if (
+        fitnesses.is_cuda
+        and fitnesses.ndim == 1
+        and 1 < fitnesses.shape[0] <= 8192
+        and ranking_method in _FUSED_METHODS
+    ):
+        from ..ops.dispatch import _allow_eager_on_gpu, hip_required
+
+        if not _allow_eager_on_gpu():
+            mod = hip_required()
+            return mod.fused_rank(fitnesses, _FUSED_METHODS[ranking_method], bool(higher_is_better))
     return f(fitnesses, higher_is_better=higher_is_better)
