@@ -88,11 +88,19 @@ class MAPElites(SearchAlgorithm, SinglePopulationAlgorithmMixin, ExtendedPopulat
         features = evals[:, 1 : 1 + nf]  # (N, F)
         lo = self._feature_grid[:, :, 0].unsqueeze(1)  # (C, 1, F)
         hi = self._feature_grid[:, :, 1].unsqueeze(1)
-        f = features.unsqueeze(0)  # (1, N, F)
-        inside = ((f >= lo) & (f <= hi)).all(dim=-1)  # (C, N)
         sense = problem.senses[0]
         utils = fitness if sense == "max" else -fitness
         utils = torch.nan_to_num(utils.to(torch.float32), nan=float("-inf"))
+        if features.is_cuda:
+            # K9 HIP kernel: streamed best-in-box argmax, O(C + N) memory
+            # (the eager broadcast below materializes a (C, N) matrix)
+            from ..ops.dispatch import _allow_eager_on_gpu, hip_required
+
+            if not _allow_eager_on_gpu():
+                best_idx, any_valid = hip_required().mapelites_assign(self._feature_grid, features, utils)
+                return best_idx, any_valid
+        f = features.unsqueeze(0)  # (1, N, F)
+        inside = ((f >= lo) & (f <= hi)).all(dim=-1)  # (C, N)
         masked = torch.where(inside, utils.unsqueeze(0), torch.full_like(utils, float("-inf")).unsqueeze(0).expand_as(inside))
         best_idx = masked.argmax(dim=1)  # (C,)
         any_valid = inside.any(dim=1)

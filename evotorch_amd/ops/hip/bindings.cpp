@@ -21,6 +21,7 @@ void cma_update_c(torch::Tensor C, torch::Tensor Y, torch::Tensor w, torch::Tens
 torch::Tensor fused_rank(torch::Tensor fitnesses, int64_t method, bool higher_better);
 torch::Tensor domination_counts(torch::Tensor utils);
 torch::Tensor pareto_ranks(torch::Tensor utils, int64_t min_assigned);
+std::vector<torch::Tensor> mapelites_assign(torch::Tensor grid, torch::Tensor feats, torch::Tensor utils);
 }  // namespace ea
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -49,4 +50,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("domination_counts", &ea::domination_counts, "K7: NSGA-II domination counts (no N x N matrix)");
     m.def("pareto_ranks", &ea::pareto_ranks, "K7: NSGA-II non-dominated sorting by front peeling",
           py::arg("utils"), py::arg("min_assigned") = 0);
+    m.def("mapelites_assign", &ea::mapelites_assign,
+          "K9: MAPElites cell assignment, O(C+N) memory (streamed best-in-box argmax)");
 }

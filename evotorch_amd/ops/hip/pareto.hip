@@ -180,4 +180,70 @@ torch::Tensor pareto_ranks(torch::Tensor utils, int64_t min_assigned) {
     return ranks;
 }
 
+// ---------------------------------------------------------------------------
+// K9: MAPElites cell assignment — for every hypergrid
+// cell, the best solution whose features fall inside the cell box
+// (reference mapelites.py:24-68). The eager path materializes a (C, N)
+// inside-mask + masked utility matrix (O(C·N) memory — 800 MB at
+// C=10k × N=20k); this kernel streams the solutions once per cell thread
+// with O(C + N) memory. Feature rows are read wave-uniformly (broadcast,
+// L2-cached) like the domination kernels above.
+// ---------------------------------------------------------------------------
+
+constexpr int kMaxFeat = 16;
+
+__global__ void mapelites_assign_kernel(const float* __restrict__ grid,   // [C][F][2]
+                                        const float* __restrict__ feats,  // [N][F]
+                                        const float* __restrict__ utils,  // [N] higher is better
+                                        int64_t* __restrict__ best_idx,   // [C]
+                                        bool* __restrict__ any_valid,     // [C]
+                                        int64_t c_total, int64_t n, int f) {
+    const int64_t c = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+    if (c >= c_total) return;
+    float lo[kMaxFeat], hi[kMaxFeat];
+    for (int k = 0; k < f; ++k) {
+        lo[k] = grid[(c * f + k) * 2];
+        hi[k] = grid[(c * f + k) * 2 + 1];
+    }
+    float best_u = -INFINITY;
+    int64_t best = 0;
+    bool found = false;
+    for (int64_t i = 0; i < n; ++i) {
+        bool inside = true;
+        for (int k = 0; k < f; ++k) {
+            const float x = feats[i * f + k];
+            inside &= (x >= lo[k]) & (x <= hi[k]);
+        }
+        const float u = utils[i];
+        // strict > keeps the FIRST best row on ties (torch argmax parity)
+        if (inside && (!found || u > best_u)) {
+            best_u = u;
+            best = i;
+            found = true;
+        }
+    }
+    best_idx[c] = best;
+    any_valid[c] = found;
+}
+
+std::vector<torch::Tensor> mapelites_assign(torch::Tensor grid, torch::Tensor feats, torch::Tensor utils) {
+    TORCH_CHECK(grid.is_cuda() && grid.dim() == 3 && grid.size(2) == 2, "grid must be (C, F, 2) on ROCm");
+    TORCH_CHECK(feats.dim() == 2 && utils.dim() == 1, "feats (N, F), utils (N)");
+    const int f = (int)grid.size(1);
+    TORCH_CHECK(f <= kMaxFeat, "at most ", kMaxFeat, " feature dimensions");
+    auto grid_f = grid.to(torch::kFloat32).contiguous();
+    auto feats_f = feats.to(torch::kFloat32).contiguous();
+    auto utils_f = utils.to(torch::kFloat32).contiguous();
+    const int64_t c_total = grid.size(0), n = feats.size(0);
+    auto best = torch::zeros({c_total}, grid.options().dtype(torch::kInt64));
+    auto valid = torch::zeros({c_total}, grid.options().dtype(torch::kBool));
+    const int threads = 256;
+    const int blocks = (int)((c_total + threads - 1) / threads);
+    auto stream = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(mapelites_assign_kernel, dim3(blocks), dim3(threads), 0, stream, grid_f.data_ptr<float>(),
+                       feats_f.data_ptr<float>(), utils_f.data_ptr<float>(), best.data_ptr<int64_t>(),
+                       valid.data_ptr<bool>(), c_total, n, f);
+    return {best, valid};
+}
+
 }  // namespace ea

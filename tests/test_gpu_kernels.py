@@ -590,3 +590,23 @@ def test_rank_dispatch_uses_fused_kernel():
     ref = ranking.centered(fit.cpu(), higher_is_better=True)
     assert torch.allclose(out.cpu(), ref.to(torch.float32), atol=1e-5)
     assert float(out.sum().abs()) < 1e-3  # centered utilities sum to ~0
+
+
+@requires_gpu
+def test_mapelites_assign_matches_eager():
+    """K9 streamed cell assignment vs the eager (C, N) broadcast."""
+    import evotorch_amd._C as C
+
+    torch.manual_seed(3)
+    cells, n, f = 64, 500, 3
+    centers = torch.rand(cells, f, device="cuda")
+    grid = torch.stack([centers - 0.15, centers + 0.15], dim=-1)
+    feats = torch.rand(n, f, device="cuda")
+    utils = torch.randn(n, device="cuda")
+    best, valid = C.mapelites_assign(grid, feats, utils)
+    inside = ((feats.unsqueeze(0) >= grid[:, :, 0].unsqueeze(1)) & (feats.unsqueeze(0) <= grid[:, :, 1].unsqueeze(1))).all(-1)
+    masked = torch.where(inside, utils.unsqueeze(0), torch.full_like(utils, float("-inf")).unsqueeze(0).expand_as(inside))
+    ref_best = masked.argmax(dim=1)
+    ref_valid = inside.any(dim=1)
+    assert torch.equal(valid, ref_valid)
+    assert torch.equal(best[ref_valid], ref_best[ref_valid])
