@@ -19,10 +19,11 @@ bump kernel) — a by-value seed would be frozen into the capture and every
 replay would resample the same population.
 
 Supported: non-distributed GaussianSearchAlgorithm searchers (PGPE / SNES
-/ CEM via SeparableGaussian-family distributions) on a ROCm device, with a
-capture-safe (pure-tensor) fitness function, fixed popsize, and either a
-plain center learning rate or a ClipUp optimizer (Adam keeps host-side
-step counts and is rejected). Do not interleave plain `searcher.step()`
+/ CEM — the elite top-k sort is shape-static and captures fine) on a ROCm
+device, with a capture-safe (pure-tensor) fitness function, fixed
+popsize, and a plain center learning rate, ClipUp, or Adam (the Adam step
+count lives in a device buffer advanced on device, so each replay applies
+the right bias correction). Do not interleave plain `searcher.step()`
 calls after capture — the graph is bound to the distribution's parameter
 buffers; re-create the GraphedSearch if you need to.
 """
@@ -32,7 +33,7 @@ from typing import Optional
 import torch
 
 from ..distributions import ExpSeparableGaussian, SeparableGaussian, SymmetricSeparableGaussian
-from ..optimizers import ClipUp
+from ..optimizers import Adam, ClipUp
 from ..utils import ranking as _ranking
 from ..utils.misc import modify_tensor
 from .gaussian import GaussianSearchAlgorithm
@@ -51,11 +52,9 @@ class GraphedSearch:
         dist = searcher._distribution
         if not isinstance(dist, (SeparableGaussian, SymmetricSeparableGaussian, ExpSeparableGaussian)):
             raise TypeError(f"Unsupported distribution for graph capture: {type(dist).__name__}")
-        if "parenthood_ratio" in dist.parameters:
-            raise ValueError("CEM's elite-sort update is not graph-captured yet; use PGPE/SNES")
         opt = searcher._optimizer
-        if opt is not None and not isinstance(opt, ClipUp):
-            raise TypeError("Graph capture supports no optimizer or ClipUp (Adam keeps host-side state)")
+        if opt is not None and not isinstance(opt, (ClipUp, Adam)):
+            raise TypeError("Graph capture supports no optimizer, ClipUp, or Adam")
         problem = searcher.problem
         if problem.device.type != "cuda":
             raise ValueError("GraphedSearch needs a ROCm device problem")
@@ -80,6 +79,11 @@ class GraphedSearch:
         # scratch for the stdev-control clamp (needs the pre-update sigma)
         self._old_sigma = dist.parameters["sigma"].clone()
         self._mean_eval_buf = torch.zeros((), dtype=problem.eval_dtype, device=problem.device)
+        self._adam_t_buf = None
+        if isinstance(opt, Adam):
+            # graph-safe device-side step counter (see adam_step_graphsafe)
+            self._adam_t_buf = torch.tensor([int(opt._t)], dtype=torch.int64, device=problem.device)
+            self._adam_step_out = torch.zeros_like(dist.parameters["mu"])
 
     # -- one in-place generation (everything stays in fixed buffers) --------
 
@@ -102,7 +106,20 @@ class GraphedSearch:
         grads = dist._compute_gradients(values, weights, ranking_used=method)
 
         # in-place center update
-        if self._opt is not None:
+        if isinstance(self._opt, Adam):
+            self._C.adam_step_graphsafe(
+                self._adam_step_out,
+                grads["mu"].to(self._adam_step_out.dtype),
+                self._opt._m,
+                self._opt._v,
+                self._adam_t_buf,
+                self._opt._stepsize,
+                self._opt._beta1,
+                self._opt._beta2,
+                self._opt._epsilon,
+            )
+            mu.add_(self._adam_step_out)
+        elif self._opt is not None:
             from .. import ops
 
             ops.clipup_step_(

@@ -13,6 +13,8 @@ std::vector<torch::Tensor> snes_gradients(torch::Tensor samples, torch::Tensor m
 void clipup_step(torch::Tensor velocity, torch::Tensor grad, double step_size, double max_speed, double momentum);
 void adam_step(torch::Tensor step_out, torch::Tensor grad, torch::Tensor m, torch::Tensor v, int64_t step_count,
                double stepsize, double beta1, double beta2, double epsilon);
+void adam_step_graphsafe(torch::Tensor step_out, torch::Tensor grad, torch::Tensor m, torch::Tensor v,
+                         torch::Tensor t_buf, double stepsize, double beta1, double beta2, double epsilon);
 torch::Tensor rollout_linear(torch::Tensor params, torch::Tensor env_blob, torch::Tensor obs_stats_out,
                              int64_t obs_dim, int64_t act_dim, int64_t rank, int64_t steps, double alive_bonus,
                              double act_cost, int64_t init_seed, int64_t member_offset, int64_t policy_hidden);
@@ -37,6 +39,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("snes_gradients", &ea::snes_gradients, "K3: SNES raw-noise gradient reduction");
     m.def("clipup_step", &ea::clipup_step, "K4: fused ClipUp velocity update (no host sync)");
     m.def("adam_step", &ea::adam_step, "K4: fused Adam ascent step");
+    m.def("adam_step_graphsafe", &ea::adam_step_graphsafe,
+          "K4 (hipGraph-safe): Adam with device-side step counter");
     m.def("rollout_linear", &ea::rollout_linear,
           "K10+K11: fused policy episode rollout (linear or MLP-H, synthetic env)",
           pybind11::arg("params"), pybind11::arg("env_blob"), pybind11::arg("obs_stats_out"),

@@ -403,6 +403,48 @@ __global__ void adam_kernel(T* __restrict__ step_out, const T* __restrict__ grad
     }
 }
 
+// Graph-safe Adam: the step count lives in a device buffer and is
+// advanced ON DEVICE, so hipGraph replays apply the correct bias
+// correction each generation (a by-value step count would freeze it).
+template <typename T>
+__global__ void adam_graphsafe_kernel(T* __restrict__ step_out, const T* __restrict__ grad, T* __restrict__ m,
+                                      T* __restrict__ v, const long long* __restrict__ t_buf, float stepsize,
+                                      float beta1, float beta2, float epsilon, int64_t n) {
+    const float t = (float)(*t_buf + 1);
+    const float bias1 = 1.0f - powf(beta1, t);
+    const float bias2 = 1.0f - powf(beta2, t);
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n; i += (int64_t)gridDim.x * blockDim.x) {
+        const float g = static_cast<float>(grad[i]);
+        const float m_new = fmaf(beta1, static_cast<float>(m[i]), (1.0f - beta1) * g);
+        const float v_new = fmaf(beta2, static_cast<float>(v[i]), (1.0f - beta2) * g * g);
+        m[i] = static_cast<T>(m_new);
+        v[i] = static_cast<T>(v_new);
+        step_out[i] = static_cast<T>(stepsize * (m_new / bias1) / (sqrtf(v_new / bias2) + epsilon));
+    }
+}
+
+__global__ void bump_step_kernel(long long* t_buf) { *t_buf += 1; }
+
+void adam_step_graphsafe(torch::Tensor step_out, torch::Tensor grad, torch::Tensor m, torch::Tensor v,
+                         torch::Tensor t_buf, double stepsize, double beta1, double beta2, double epsilon) {
+    CHECK_GPU(step_out);
+    TORCH_CHECK(t_buf.is_cuda() && t_buf.scalar_type() == at::ScalarType::Long && t_buf.numel() >= 1,
+                "t_buf must be a cuda int64 tensor");
+    const int64_t n = step_out.numel();
+    const int threads = 256;
+    const int blocks = (int)std::min<int64_t>((n + threads - 1) / threads, 2048);
+    auto stream = at::cuda::getCurrentCUDAStream();
+    AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::BFloat16, at::ScalarType::Half, step_out.scalar_type(), "adam_graphsafe", [&] {
+        using T = scalar_t;
+        hipLaunchKernelGGL((adam_graphsafe_kernel<T>), dim3(blocks), dim3(threads), 0, stream, step_out.data_ptr<T>(),
+                           grad.data_ptr<T>(), m.data_ptr<T>(), v.data_ptr<T>(),
+                           reinterpret_cast<const long long*>(t_buf.data_ptr<int64_t>()), (float)stepsize,
+                           (float)beta1, (float)beta2, (float)epsilon, n);
+    });
+    hipLaunchKernelGGL(bump_step_kernel, dim3(1), dim3(1), 0, stream,
+                       reinterpret_cast<long long*>(t_buf.data_ptr<int64_t>()));
+}
+
 void adam_step(torch::Tensor step_out, torch::Tensor grad, torch::Tensor m, torch::Tensor v, int64_t step_count,
                double stepsize, double beta1, double beta2, double epsilon) {
     CHECK_GPU(step_out);

@@ -320,3 +320,58 @@ def test_on_aux_device_fitness_runs_on_gpu():
     assert seen[0] == "cuda"
     assert batch.evals_are_ready
     assert batch.unsafe_evals.device.type == "cpu"  # results land back on the problem device
+
+
+@requires_gpu
+def test_graphed_cem_and_adam():
+    """Graph capture extended to CEM (elite top-k is shape-static) and to
+    Adam via the device-side step counter."""
+    import torch
+
+    from evotorch_amd import Problem
+    from evotorch_amd.algorithms import CEM, PGPE, GraphedSearch
+    from evotorch_amd.decorators import vectorized
+
+    @vectorized
+    def sphere(x):
+        return (x**2).sum(-1)
+
+    prob = Problem("min", sphere, solution_length=30, initial_bounds=(-1, 1), seed=1, device="cuda:0")
+    cem = CEM(prob, popsize=64, parenthood_ratio=0.25, stdev_init=2.0,
+              center_init=torch.full((30,), 3.0))
+    g = GraphedSearch(cem)
+    g.capture()
+    first = g.mean_eval
+    g.run(60)
+    assert g.mean_eval < first * 0.2
+
+    prob2 = Problem("min", sphere, solution_length=30, initial_bounds=(-1, 1), seed=2, device="cuda:0")
+    pgpe = PGPE(prob2, popsize=64, center_learning_rate=0.1, stdev_learning_rate=0.05,
+                stdev_init=1.0, center_init=torch.full((30,), 3.0), optimizer="adam",
+                ranking_method="centered")
+    g2 = GraphedSearch(pgpe)
+    g2.capture()
+    first2 = g2.mean_eval
+    g2.run(100)
+    assert g2.mean_eval < first2 * 0.5
+
+
+@requires_gpu
+def test_graphsafe_adam_matches_host_adam():
+    """adam_step_graphsafe (device step counter) must track the host-count
+    kernel step for step for several iterations."""
+    import evotorch_amd._C as C
+    import torch
+
+    torch.manual_seed(0)
+    n = 257
+    m1 = torch.zeros(n, device="cuda"); v1 = torch.zeros(n, device="cuda")
+    m2 = torch.zeros(n, device="cuda"); v2 = torch.zeros(n, device="cuda")
+    t_buf = torch.zeros(1, dtype=torch.int64, device="cuda")
+    out1 = torch.empty(n, device="cuda"); out2 = torch.empty(n, device="cuda")
+    for t in range(1, 6):
+        g = torch.randn(n, device="cuda")
+        C.adam_step(out1, g, m1, v1, t, 1e-2, 0.9, 0.999, 1e-8)
+        C.adam_step_graphsafe(out2, g, m2, v2, t_buf, 1e-2, 0.9, 0.999, 1e-8)
+        assert torch.allclose(out1, out2, rtol=1e-5, atol=1e-7), t
+    assert int(t_buf) == 5
