@@ -384,3 +384,25 @@ def test_stateful_module_and_multilayered():
     y1_again = stateful(x)
     assert torch.allclose(y1, y1_again, atol=1e-6)
     assert not torch.allclose(y1, y2, atol=1e-6)
+
+
+def test_supervised_mnist30k_config_descends():
+    """The BASELINE row-4 configuration shape (MNIST30K convnet,
+    common-minibatch subbatched PGPE+Adam) runs end to end on synthetic
+    MNIST-shaped data and reduces the training loss."""
+    import subprocess
+    import sys
+    import json
+    import os
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    result = subprocess.run(
+        [sys.executable, os.path.join(repo, "scripts", "bench_supervised.py"),
+         "--steps", "10", "--warmup", "1", "--popsize", "64", "--minibatch", "128",
+         "--subbatch", "32", "--data-size", "512"],
+        capture_output=True, text=True, timeout=600, cwd=repo,
+    )
+    assert result.returncode == 0, result.stderr[-2000:]
+    d = json.loads(result.stdout.strip().splitlines()[-1])
+    assert d["config"]["final_mean_loss"] < d["config"]["first_mean_loss"] + 0.3
+    assert 28000 < int(d["config"]["model"].split("(")[1].split()[0]) < 30000
