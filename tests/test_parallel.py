@@ -326,6 +326,17 @@ def test_world_size_invariant_trajectory(body):
     assert torch.allclose(c1, c2, atol=1e-5, rtol=1e-5), f"world-1 vs world-2 trajectories differ:\n{c1}\n{c2}"
 
 
+def test_world4_matches_world1():
+    """Same invariance at world size 4 (the popsize-40 shards stay even,
+    mirroring a 4-GPU slice of the driver's scaling ladder)."""
+    r1 = _run_world("_body_traj_pgpe", world=1)
+    r4 = _run_world("_body_traj_pgpe", world=4)
+    assert all(r4[0] == r4[i] for i in range(4)), "ranks diverged"
+    c1 = torch.tensor(r1[0], dtype=torch.float64)
+    c4 = torch.tensor(r4[0], dtype=torch.float64)
+    assert torch.allclose(c1, c4, atol=1e-5, rtol=1e-5)
+
+
 def test_bench_entry_torchrun_world2(tmp_path):
     """The driver launches bench.py via torch.distributed.run; validate that
     exact entry path (world 2, gloo on CPU) end to end."""
