@@ -15,6 +15,29 @@ from evotorch_amd.neuroevolution import SyntheticRolloutProblem
 
 
 def main():
+    # v7 flagship (linear, T=1000) through the SPMD path (comm world-1 ==
+    # the code the driver's scaling run executes), incl. pregen overlap
+    from evotorch_amd.parallel.comm import Comm
+
+    comm = Comm(device=torch.device("cuda", 0))
+    prob_lin = SyntheticRolloutProblem(device="cuda:0", seed=3, episode_length=1000)
+    prob_lin.use_comm(comm)
+    r = 2.25
+    s_lin = PGPE(prob_lin, popsize=4000, radius_init=r, center_learning_rate=0.75 * r / 15,
+                 stdev_learning_rate=0.1, optimizer="clipup", optimizer_config={"max_speed": r / 15},
+                 distributed=True)
+    t0 = time.perf_counter()
+    for g in range(1000):
+        s_lin.step()
+        if (g + 1) % 250 == 0:
+            me = float(s_lin.status["mean_eval"])
+            assert math.isfinite(me), f"non-finite mean_eval at gen {g+1}"
+            print(f"v7 linear T=1000 gen {g+1}: mean={me:.1f}", flush=True)
+    torch.cuda.synchronize()
+    dt = time.perf_counter() - t0
+    print(f"PGPE v7 linear 1000-gen soak (SPMD world-1 + overlap): {dt:.1f}s, "
+          f"{1000*4000/dt:,.0f} sol/s sustained, final {float(s_lin.status['mean_eval']):.1f}")
+
     prob = SyntheticRolloutProblem(device="cuda:0", seed=7, episode_length=200, policy_hidden=64)
     r = 2.25
     s = PGPE(prob, popsize=4000, radius_init=r, center_learning_rate=0.75 * r / 15,
