@@ -56,6 +56,23 @@ __device__ __forceinline__ float b2f(__bf16 v) { return (float)v; }
 
 // tanh via the hardware exp unit (~1e-6 rel error, far below the bf16
 // quantization of every operand; libm tanhf is branchy on the hot path)
+// 8-lane group sum on the VALU pipe: after row_shr 1/2/4 the TOP lane of
+// each 8-lane group (glane == 7) holds the group sum (__shfl_down lowers
+// to ds_bpermute_b32 on the LDS pipe — the v7 kernel measured that
+// contention; same fix here).
+template <int kCtrl6>
+__device__ __forceinline__ float dpp_add6(float x) {
+    const int moved = __builtin_amdgcn_update_dpp(0, __builtin_bit_cast(int, x), kCtrl6, 0xf, 0xf, true);
+    return x + __builtin_bit_cast(float, moved);
+}
+
+__device__ __forceinline__ float group8_sum_dpp(float x) {
+    x = dpp_add6<0x111>(x);  // row_shr:1
+    x = dpp_add6<0x112>(x);  // row_shr:2
+    x = dpp_add6<0x114>(x);  // row_shr:4
+    return x;                // valid in the top lane of each 8-lane group
+}
+
 __device__ __forceinline__ float tanh_fast6(float x) {
     const float xc = fminf(fmaxf(x, -15.0f), 15.0f);
     const float e = __expf(2.0f * xc);
@@ -344,17 +361,14 @@ __global__ __launch_bounds__(512, 2) void rollout_linear_kernel(RolloutArgs args
             }                                                                                 \
             acc[r] = a0 + a1;                                                                 \
         }                                                                                     \
-        _Pragma("unroll") for (int offset = kGroup / 2; offset > 0; offset >>= 1) {           \
-            _Pragma("unroll") for (int r = 0; r < (NR); ++r) {                                \
-                /* validity is GROUP-uniform, so this divergence masks      */                \
-                /* whole 8-lane groups: invalid rounds issue no bpermutes   */                \
-                if (descs[r].valid) acc[r] += __shfl_down(acc[r], offset, kWaveSize);         \
-            }                                                                                 \
+        _Pragma("unroll") for (int r = 0; r < (NR); ++r) {                                    \
+            /* VALU DPP group reduce; group sums land in glane == kGroup-1 */                 \
+            if (descs[r].valid) acc[r] = group8_sum_dpp(acc[r]);                              \
         }                                                                                     \
     }
 
 #define DOT_EPILOGUE(descs, NR, acc)                                                          \
-    if (glane == 0) {                                                                         \
+    if (glane == kGroup - 1) {                                                                \
         _Pragma("unroll") for (int r = 0; r < (NR); ++r) {                                    \
             if (!descs[r].valid) continue;                                                    \
             if (descs[r].kind == 1) {                                                         \
