@@ -14,11 +14,8 @@
 // more O(N²·M) pass — host sync only once per front.
 
 #include <hip/hip_runtime.h>
-#include <hip/hip_cooperative_groups.h>
 #include <torch/extension.h>
 #include <ATen/cuda/CUDAContext.h>
-
-namespace cg = cooperative_groups;
 
 namespace ea {
 
@@ -132,69 +129,6 @@ torch::Tensor domination_counts(torch::Tensor utils) {
     return counts;
 }
 
-// Whole peel in ONE cooperative launch: 500-front populations cost ~1500
-// kernel launches + ~30 host syncs under the batched peel (measured
-// ~24 ms of the 35 ms total at N=65k); here the front loop runs device-
-// side with grid-wide syncs, so the peel costs just its arithmetic (each
-// solution is subtracted exactly once -> one more O(N^2 M) pass).
-__global__ void peel_all_coop_kernel(const float* __restrict__ utils, int* __restrict__ counts,
-                                     int64_t* __restrict__ ranks, int* __restrict__ front_list,
-                                     int* __restrict__ scalars,  // [0] front_count, [1] assigned
-                                     int64_t n, int m, int64_t min_assigned) {
-    cg::grid_group grid = cg::this_grid();
-    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    const int64_t tid0 = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
-    int64_t front_index = 0;
-    while (true) {
-        if (tid0 == 0) scalars[0] = 0;
-        grid.sync();
-        for (int64_t j = tid0; j < n; j += stride) {
-            if (counts[j] == 0) {
-                const int pos = atomicAdd(&scalars[0], 1);
-                front_list[pos] = (int)j;
-                ranks[j] = front_index;
-                counts[j] = -1;
-            }
-        }
-        grid.sync();
-        const int fc = scalars[0];
-        if (tid0 == 0) scalars[1] += fc;
-        grid.sync();
-        const int assigned = scalars[1];
-        ++front_index;
-        if (fc == 0 || assigned >= n || (int64_t)assigned >= min_assigned) {
-            // stragglers (early stop, or a numerical corner with no
-            // zero-count candidates) share one beyond-last front
-            for (int64_t j = tid0; j < n; j += stride) {
-                if (counts[j] >= 0) ranks[j] = front_index;
-            }
-            return;
-        }
-        for (int64_t j = tid0; j < n; j += stride) {
-            if (counts[j] < 0) continue;
-            float mine[kMaxObjCache];
-            const bool cached = m <= kMaxObjCache;
-            if (cached) {
-                for (int k = 0; k < m; ++k) mine[k] = utils[j * m + k];
-            }
-            int removed = 0;
-            for (int t = 0; t < fc; ++t) {
-                const float* other = utils + (int64_t)front_list[t] * m;
-                const float* me = cached ? mine : (utils + j * m);
-                bool ge_all = true, gt_any = false;
-                for (int k = 0; k < m; ++k) {
-                    const float o = other[k];
-                    ge_all &= (o >= me[k]);
-                    gt_any |= (o > me[k]);
-                }
-                removed += (ge_all && gt_any) ? 1 : 0;
-            }
-            counts[j] -= removed;
-        }
-        grid.sync();
-    }
-}
-
 torch::Tensor pareto_ranks(torch::Tensor utils, int64_t min_assigned) {
     // min_assigned > 0: stop peeling once that many solutions hold final
     // ranks (take_best(n) only needs the fronts crossing n); stragglers
@@ -209,39 +143,6 @@ torch::Tensor pareto_ranks(torch::Tensor utils, int64_t min_assigned) {
     const int threads = 256;
     const int blocks = (int)((n + threads - 1) / threads);
     auto stream = at::cuda::getCurrentCUDAStream();
-
-    // cooperative single-launch peel when the grid can be co-resident
-    static int coop_blocks_per_cu = -1;
-    static int num_cus = 0;
-    if (coop_blocks_per_cu < 0) {
-        hipDeviceProp_t prop;
-        (void)hipGetDeviceProperties(&prop, 0);
-        num_cus = prop.multiProcessorCount;
-        int per_cu = 0;
-        if (hipOccupancyMaxActiveBlocksPerMultiprocessor(&per_cu, reinterpret_cast<const void*>(&peel_all_coop_kernel), threads, 0) != hipSuccess || !prop.cooperativeLaunch) {
-            per_cu = 0;
-        }
-        coop_blocks_per_cu = per_cu;
-    }
-    const int coop_max = coop_blocks_per_cu * num_cus;
-    if (coop_max > 0) {
-        const int coop_blocks = std::min(blocks, coop_max);
-        auto front_list = torch::empty({n}, utils_f.options().dtype(torch::kInt32));
-        auto scalars = torch::zeros({2}, utils_f.options().dtype(torch::kInt32));
-        const float* utils_p = utils_f.data_ptr<float>();
-        int* counts_p = counts.data_ptr<int>();
-        int64_t* ranks_p = ranks.data_ptr<int64_t>();
-        int* list_p = front_list.data_ptr<int>();
-        int* scalars_p = scalars.data_ptr<int>();
-        int64_t n_arg = n;
-        int m_arg = m;
-        int64_t min_arg = min_assigned;
-        void* args[] = {&utils_p, &counts_p, &ranks_p, &list_p, &scalars_p, &n_arg, &m_arg, &min_arg};
-        hipError_t err = hipLaunchCooperativeKernel(reinterpret_cast<const void*>(&peel_all_coop_kernel),
-                                                    dim3(coop_blocks), dim3(threads), args, 0, stream);
-        if (err == hipSuccess) return ranks;
-        (void)hipGetLastError();  // clear; fall back to the batched peel
-    }
     // Peel in blind batches of kPeelBatch fronts with ONE host sync per
     // batch (a sync per front costs hundreds of round-trips at large N:
     // a random 16k population can have hundreds of fronts). A peel with

@@ -76,24 +76,4 @@ def main():
             time_block(f"  evaluate N={N}", lambda: prob.evaluate(batch), iters=10)
 
 
-if sys.argv[1:2] != ["nsga_fine"]:
-    main()
-
-
-def nsga_fine():
-    """Fine-grained: counts vs peel vs crowding at 32k/65k rows."""
-    import evotorch_amd._C as C
-    from evotorch_amd.core import _crowding_distances
-
-    for N in (32768, 65536):
-        utils = torch.randn(N, 2, device="cuda")
-        time_block(f"  domination_counts N={N}", lambda: C.domination_counts(utils), iters=10)
-        time_block(f"  pareto_ranks full N={N}", lambda: C.pareto_ranks(utils, 0), iters=10)
-        time_block(f"  pareto_ranks min_assigned=N/2 N={N}", lambda: C.pareto_ranks(utils, N // 2), iters=10)
-        ranks = C.pareto_ranks(utils, 0)
-        nf = int(ranks.max()) + 1
-        time_block(f"  crowding N={N} (fronts={nf})", lambda: _crowding_distances(utils, ranks), iters=10)
-
-
-if sys.argv[1:2] == ["nsga_fine"]:
-    nsga_fine()
+main()
