@@ -278,7 +278,7 @@ class CMAES(SearchAlgorithm, SinglePopulationAlgorithmMixin):
         self._steps_since_decompose += 1
         interval = self._decompose_interval
         if interval >= 4 and self._device_is_gpu():
-            if self._steps_since_decompose == max(1, interval // 2):
+            if self._steps_since_decompose >= max(1, interval // 2) and self._pending_A is None:
                 # snapshot + async factorization on the side stream
                 if self._decomp_stream is None:
                     self._decomp_stream = torch.cuda.Stream(device=self._C.device)

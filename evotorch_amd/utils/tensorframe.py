@@ -37,7 +37,7 @@ class _PickIndexer:
         if isinstance(index, tuple) and len(index) == 2:
             rows, cols = index
             names = self._normalize_cols(frame, cols)
-            sub = TensorFrame({c: frame._columns[c] for c in cols} if False else {c: frame._columns[c] for c in names})
+            sub = TensorFrame({c: frame._columns[c] for c in names})
             return sub.pick[rows]
         if isinstance(index, (int,)):
             index = slice(index, index + 1)
