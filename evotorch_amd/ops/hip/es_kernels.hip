@@ -486,44 +486,27 @@ __global__ __launch_bounds__(1024) void fused_rank_kernel(const float* __restric
         idx[i] = i;
     }
     __syncthreads();
-
-#define BITONIC_PAIR(i)                                                               \
-    {                                                                                 \
-        const int ixj = (i) ^ j;                                                      \
-        if (ixj > (i)) {                                                              \
-            const bool up = ((i) & k) == 0;                                           \
-            const float a = keys[i], b = keys[ixj];                                   \
-            /* NaN-robust: NaNs order last = best-key (torch argsort parity) */       \
-            const bool swap = up ? (b < a || (isnan(a) && !isnan(b)))                 \
-                                 : (a < b || (isnan(b) && !isnan(a)));                \
-            if (swap) {                                                               \
-                keys[i] = b; keys[ixj] = a;                                           \
-                const int t = idx[i]; idx[i] = idx[ixj]; idx[ixj] = t;                \
-            }                                                                         \
-        }                                                                             \
-    }
-
-    // Wave-local bitonic stages: the block barrier was the whole cost
-    // (78 barriers at p=4096 ≈ 100 µs). Each wave owns an aligned
-    // `range`-sized element block; a stage with j < range only compares
-    // within blocks, so it needs no s_barrier — intra-wave LDS
-    // read-after-write is ordered by the compiler's lgkmcnt waits (the
-    // wave_barrier is a compiler reordering fence, not an instruction).
-    const int range = p / (int)(blockDim.x >> 6);  // elements per wave (pow2)
-    const int wv = tid >> 6, ln = tid & 63;
+    // bitonic sort ascending: position 0 = worst
     for (int k = 2; k <= p; k <<= 1) {
         for (int j = k >> 1; j > 0; j >>= 1) {
-            if (j >= range) {
-                for (int i = tid; i < p; i += blockDim.x) BITONIC_PAIR(i);
-                __syncthreads();
-            } else {
-                for (int i = wv * range + ln; i < (wv + 1) * range; i += 64) BITONIC_PAIR(i);
-                __builtin_amdgcn_wave_barrier();
+            for (int i = tid; i < p; i += blockDim.x) {
+                const int ixj = i ^ j;
+                if (ixj > i) {
+                    const bool up = (i & k) == 0;
+                    const float a = keys[i], b = keys[ixj];
+                    // NaN-robust: order NaNs last so they rank as best-key
+                    // (matches torch argsort's NaN-is-largest behavior)
+                    const bool swap = up ? (b < a || (isnan(a) && !isnan(b)))
+                                         : (a < b || (isnan(b) && !isnan(a)));
+                    if (swap) {
+                        keys[i] = b; keys[ixj] = a;
+                        const int t = idx[i]; idx[i] = idx[ixj]; idx[ixj] = t;
+                    }
+                }
             }
+            __syncthreads();
         }
     }
-    __syncthreads();
-#undef BITONIC_PAIR
     if (method == 2) {
         // NES log-utilities need their global sum before the final map
         float partial = 0.0f;
