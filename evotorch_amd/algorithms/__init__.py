@@ -6,7 +6,7 @@ from .ga import ExtendedPopulationMixin, Cosyne, GeneticAlgorithm, SteadyStateGA
 from .gaussian import CEM, PGPE, SNES, XNES, GaussianSearchAlgorithm
 from .graphed import GraphedSearch
 from .mapelites import make_feature_grid, MAPElites
-from .restarter import IPOP, ModifyingRestart, Restart
+from .restarter import BIPOP, IPOP, ModifyingRestart, Restart
 from .searchalgorithm import LazyReporter, LazyStatusDict, SearchAlgorithm, SinglePopulationAlgorithmMixin
 
 try:

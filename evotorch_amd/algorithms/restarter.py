@@ -13,7 +13,7 @@ import torch
 from ..core import Problem
 from .searchalgorithm import SearchAlgorithm
 
-__all__ = ["Restart", "ModifyingRestart", "IPOP"]
+__all__ = ["BIPOP", "IPOP", "ModifyingRestart", "Restart"]
 
 
 class Restart(SearchAlgorithm):
@@ -118,4 +118,63 @@ class IPOP(ModifyingRestart):
     def _on_restart(self):
         if "popsize" in self._algorithm_args and self._algorithm_args["popsize"]:
             self._algorithm_args["popsize"] = int(self._algorithm_args["popsize"] * self._popsize_multiplier)
+        super()._on_restart()
+
+
+class BIPOP(ModifyingRestart):
+    """BIPOP restart strategy (Hansen 2009): restarts alternate between a
+    LARGE regime whose population doubles each time (IPOP-style) and a
+    SMALL regime whose population is drawn as
+    ``λ_default · (λ_large / (2·λ_default)) ** U[0,1]²`` with at most half
+    the large regime's evaluation budget. Each restart enters the regime
+    that has consumed fewer evaluations so far.
+
+    The reference only reaches BIPOP through the external `cma` package
+    (`PyCMAES(..., cma_options)`); this is a native implementation usable
+    with any popsize-parameterized searcher (typically `CMAES`).
+    """
+
+    def __init__(
+        self,
+        problem: Problem,
+        algorithm_factory: Callable[..., SearchAlgorithm],
+        *,
+        algorithm_args: Optional[dict] = None,
+        seed: Optional[int] = None,
+        **kwargs,
+    ):
+        algorithm_args = dict(algorithm_args or {})
+        if not algorithm_args.get("popsize"):
+            import math
+
+            algorithm_args["popsize"] = int(4 + math.floor(3 * math.log(problem.solution_length)))
+        super().__init__(problem, algorithm_factory, algorithm_args=algorithm_args, **kwargs)
+        self._default_popsize = int(algorithm_args["popsize"])
+        self._large_popsize = self._default_popsize
+        self._large_evals = 0
+        self._small_evals = 0
+        self._regime = "large"  # the first run counts as a large run
+        self._rng = torch.Generator().manual_seed(seed if seed is not None else 0)
+        self.add_status_getters({
+            "regime": lambda: self._regime,
+            "current_popsize": lambda: int(self._algorithm_args.get("popsize", 0)),
+        })
+
+    def _on_restart(self):
+        spent = self._inner_steps * int(self._algorithm_args.get("popsize", self._default_popsize))
+        if self._regime == "large":
+            self._large_evals += spent
+        else:
+            self._small_evals += spent
+        if self._small_evals < self._large_evals:
+            # small regime: randomized popsize in [λ_default, λ_large/2]
+            self._regime = "small"
+            u = float(torch.rand((), generator=self._rng))
+            ratio = max(1.0, self._large_popsize / (2.0 * self._default_popsize))
+            popsize = int(self._default_popsize * (ratio ** (u * u)))
+            self._algorithm_args["popsize"] = max(popsize, 4)
+        else:
+            self._regime = "large"
+            self._large_popsize = int(self._large_popsize * 2)
+            self._algorithm_args["popsize"] = self._large_popsize
         super()._on_restart()

@@ -326,3 +326,40 @@ def test_cmaes_rotation_invariance_on_ellipsoid():
         finals[rotated] = float(s.status["pop_best_eval"])
     assert finals[True] < 1e-3, finals
     assert finals[False] < 1e-3, finals
+
+
+def test_bipop_restart_regimes():
+    """Native BIPOP: restarts alternate large (doubling popsize) and small
+    (randomized sub-default-to-half-large popsize) regimes by spent
+    evaluation budget, and the search still descends."""
+    import torch
+
+    from evotorch_amd import Problem
+    from evotorch_amd.algorithms import BIPOP, CMAES
+    from evotorch_amd.decorators import vectorized
+
+    @vectorized
+    def sphere(x):
+        return (x**2).sum(-1)
+
+    prob = Problem("min", sphere, solution_length=6, initial_bounds=(-3, 3), seed=2)
+    meta = BIPOP(
+        prob,
+        lambda p, **kw: CMAES(p, stdev_init=1.0, **kw),
+        algorithm_args={"popsize": 8},
+        max_inner_steps=5,  # force frequent restarts
+        seed=7,
+    )
+    regimes = []
+    popsizes = []
+    for _ in range(40):
+        meta.step()
+        regimes.append(meta.status["regime"])
+        popsizes.append(meta.status["current_popsize"])
+    assert meta.num_restarts >= 6
+    assert "small" in regimes and "large" in regimes
+    large_sizes = sorted({p for r, p in zip(regimes, popsizes) if r == "large"})
+    assert len(large_sizes) >= 2 and large_sizes[1] == large_sizes[0] * 2  # doubling
+    small_sizes = {p for r, p in zip(regimes, popsizes) if r == "small"}
+    assert all(4 <= p <= max(large_sizes) for p in small_sizes)
+    assert meta.status["mean_eval"] < 54.0  # descending on sphere (E||x||² = 6·9)
