@@ -37,8 +37,10 @@ def main():
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     have_gpu = torch.cuda.is_available()
-    comm = init_comm() if world > 1 else None
-    device = comm.device if comm is not None else (torch.device("cuda", 0) if have_gpu else torch.device("cpu"))
+    # one Comm for every world size: world-1 runs the same SPMD path
+    # (counter-addressed sampling + side-stream noise pre-generation)
+    comm = init_comm()
+    device = comm.device
 
     L = args.length
     g = torch.Generator(device="cpu").manual_seed(7)
@@ -59,8 +61,7 @@ def main():
         return d.sum(-1)
 
     prob = Problem("min", quadratic, solution_length=L, initial_bounds=(-1, 1), device=device, seed=1 + rank)
-    if comm is not None:
-        prob.use_comm(comm)
+    prob.use_comm(comm)
     searcher = PGPE(
         prob,
         popsize=args.popsize_per_gpu * world,
@@ -73,8 +74,7 @@ def main():
     )
 
     def sync():
-        if comm is not None:
-            comm.barrier()
+        comm.barrier()
         if have_gpu:
             torch.cuda.synchronize()
 
