@@ -86,7 +86,7 @@ def main():
         searcher.step()
     sync()
     dt = time.perf_counter() - t0
-    if comm is not None:
+    if world > 1:
         t = torch.tensor([dt], dtype=torch.float64, device=device if have_gpu else "cpu")
         comm.all_reduce_(t, op="max")
         dt = float(t)
