@@ -1475,6 +1475,13 @@ class Problem(TensorMakerMixin, Serializable):
                 torch.device(self._device).type == "cuda"
                 and hasattr(distribution, "fill_from_noise")
                 and num_interactions is None  # adaptive rounds consume extra seeds
+                # The overlap trades ~1.5x sampling HBM traffic (pregen z
+                # write + affine read/write vs one fused write) for hiding
+                # the philox latency — profitable when sampling is
+                # latency-bound (small/medium populations), a net LOSS when
+                # the population is tens of GB and everything is
+                # bandwidth-bound (measured 82 vs 69 ms/gen at 50 GB).
+                and local_popsize * self._solution_length * 4 <= (1 << 30)
             )
             with record_range("sample"):
                 if use_overlap:
