@@ -133,10 +133,7 @@ __global__ __launch_bounds__(64 * kWaves, 1) void rollout_v7_kernel(RolloutV7Arg
     float* wave_fit = istd_l + OP;                       // [kWaves][16] fitness partials
     float* actsq_l = wave_fit + kWaves * 16;             // [kM]
     float* d2last_l = actsq_l + kM;                      // [OP] D2_T row A-1 (rank-1 epilogue term)
-    // 8-wave variant: GEMM2 B in LDS (register fragments would exceed the
-    // 256-VGPR cap of a 512-thread block); 4-wave variant: in registers
-    // (2 blocks/CU needs <= 80 KB LDS).
-    __bf16* ud_l = reinterpret_cast<__bf16*>(d2last_l + OP);  // [OP][40] k-major (kWaves==8 only)
+    __bf16* ud_l = reinterpret_cast<__bf16*>(d2last_l + OP);  // [OP][40] k-major GEMM2 B
 
     const long RO = (long)R * O, AO = (long)A * O;
     const float* eV = args.env_blob;
@@ -178,35 +175,18 @@ __global__ __launch_bounds__(64 * kWaves, 1) void rollout_v7_kernel(RolloutV7Arg
     const int c_row0 = (lane >> 4) * 4;        // C-frag first row (member)
 
     // GEMM2 B ([U;D2 rows 0..15]): K = 32 exactly (R=16 h-rows + 16 action
-    // rows; action A-1 is the rank-1 epilogue term). Register fragments
-    // (4-wave) or k-major LDS with padded stride 40 (8-wave).
+    // rows; action A-1 is the rank-1 epilogue term), k-major LDS with
+    // padded stride 40. Register fragments would push the 4-wave variant
+    // past 256 VGPRs and forfeit its 2-blocks/CU occupancy.
     constexpr int KU = 40;  // ud_l row stride: 32 + 8 pad (bank decorrelation)
-    bf16x8_t ud_frag[kWaves == 4 ? kTilesPerWave : 1];
-    if constexpr (kWaves == 4) {
-#pragma unroll
-        for (int tw = 0; tw < kTilesPerWave; ++tw) {
-            const int o = (wave * kTilesPerWave + tw) * 16 + c_col;
-#pragma unroll
-            for (int i = 0; i < 8; ++i) {
-                const int k = g2_k0 + i;
-                float v = 0.0f;
-                if (o < O) {
-                    if (k < R) v = eU[(long)k * O + o];
-                    else if (k - R < A - 1) v = eD2[(long)(k - R) * O + o];
-                }
-                ud_frag[tw][i] = f2b7(v);
-            }
+    for (int j = tid; j < OP * KU; j += kThreads) {
+        const int o = j / KU, k = j % KU;
+        float v = 0.0f;
+        if (o < O && k < 32) {
+            if (k < R) v = eU[(long)k * O + o];
+            else if (k - R < A - 1) v = eD2[(long)(k - R) * O + o];
         }
-    } else {
-        for (int j = tid; j < OP * KU; j += kThreads) {
-            const int o = j / KU, k = j % KU;
-            float v = 0.0f;
-            if (o < O && k < 32) {
-                if (k < R) v = eU[(long)k * O + o];
-                else if (k - R < A - 1) v = eD2[(long)(k - R) * O + o];
-            }
-            ud_l[j] = f2b7(v);
-        }
+        ud_l[j] = f2b7(v);
     }
 
     // ---- per-member policy weights in registers -----------------------------
@@ -331,12 +311,8 @@ __global__ __launch_bounds__(64 * kWaves, 1) void rollout_v7_kernel(RolloutV7Arg
             for (int tw = 0; tw < kTilesPerWave; ++tw) {
                 const int col = (wave * kTilesPerWave + tw) * 16 + c_col;
                 floatx4_t acc = {0.f, 0.f, 0.f, 0.f};
-                if constexpr (kWaves == 4) {
-                    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, ud_frag[tw], acc, 0, 0, 0);
-                } else {
-                    const bf16x8_t b0 = *reinterpret_cast<const bf16x8_t*>(ud_l + col * KU + g2_k0);
-                    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc, 0, 0, 0);
-                }
+                const bf16x8_t b0 = *reinterpret_cast<const bf16x8_t*>(ud_l + col * KU + g2_k0);
+                acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc, 0, 0, 0);
                 const float d2l = d2last_l[col];
                 const float cv = c_l[col];
                 const float wrv = wr_l[col];
@@ -413,9 +389,8 @@ template <int O_T, int A_T, int kWaves>
 static void launch_v7(const RolloutV7Args& args, int n, hipStream_t stream) {
     constexpr int OP = (O_T + 127) / 128 * 128;
     constexpr int kM = 2 * kWaves;
-    size_t lds = (size_t)(2 * kM * (OP + 8) + 16 * 72 + 16 * (OP + 8)) * 2 +
+    size_t lds = (size_t)(2 * kM * (OP + 8) + 16 * 72 + 16 * (OP + 8) + OP * 40) * 2 +
                  (size_t)(kM * A_T + 5 * OP + kWaves * 16 + kM) * 4;
-    if (kWaves == 8) lds += (size_t)OP * 40 * 2;  // ud_l
     static bool attr_set7[2] = {false, false};
     const int slot = (kWaves == 8) ? 1 : 0;
     if (!attr_set7[slot]) {
