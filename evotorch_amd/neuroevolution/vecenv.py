@@ -38,9 +38,11 @@ __all__ = [
     "convert_to_torch",
     "convert_to_torch_bool",
     "is_brax_env",
+    "jax_to_torch",
     "make_brax_env",
     "make_gym_env",
     "make_vector_env",
+    "torch_to_jax",
 ]
 
 
@@ -426,6 +428,25 @@ def convert_to_torch_bool(x, *, device=None) -> torch.Tensor:
 def convert_from_torch(x: torch.Tensor):
     """torch tensor -> numpy array (cpu copy only if needed)."""
     return x.detach().cpu().numpy()
+
+
+def jax_to_torch(x):
+    """Zero-copy jax→torch via the DLPack protocol (reference
+    net/vecrl.py:53-81). Requires jax (no ROCm jax wheel ships in this
+    image; with a user-provided jax the bridge works as-is)."""
+    return torch.from_dlpack(x)
+
+
+def torch_to_jax(x: torch.Tensor):
+    """Zero-copy torch→jax via DLPack (reference net/vecrl.py:53-81)."""
+    try:
+        import jax
+    except ImportError as e:
+        raise ImportError(
+            "torch_to_jax requires jax, which has no ROCm wheel in this image; "
+            "provide your own jax build to use the brax bridge"
+        ) from e
+    return jax.dlpack.from_dlpack(torch.utils.dlpack.to_dlpack(x.detach()))
 
 
 def is_brax_env(env) -> bool:
