@@ -326,6 +326,52 @@ def test_world_size_invariant_trajectory(body):
     assert torch.allclose(c1, c2, atol=1e-5, rtol=1e-5), f"world-1 vs world-2 trajectories differ:\n{c1}\n{c2}"
 
 
+def _body_nsga2_sharded(comm, rank, world):
+    """GA-family sharding contract: identical seeds on every rank, the
+    searcher's own RNG reproduces the same populations, and evaluate()
+    shards fitness rows across ranks (P1). The full NSGA-II trajectory
+    must agree across ranks AND with a single-process run."""
+    import torch as _t
+
+    from evotorch_amd import Problem
+    from evotorch_amd.algorithms import GeneticAlgorithm
+    from evotorch_amd.decorators import vectorized
+    from evotorch_amd.operators import PolynomialMutation, SimulatedBinaryCrossOver
+
+    @vectorized
+    def zdt1ish(x):
+        f1 = x[..., 0]
+        g = 1.0 + 9.0 * x[..., 1:].mean(-1)
+        f2 = g * (1.0 - _t.sqrt(_t.clamp(f1 / g, min=0.0)))
+        return _t.stack([f1, f2], dim=-1)
+
+    prob = Problem(["min", "min"], zdt1ish, solution_length=6, initial_bounds=(0.0, 1.0),
+                   bounds=(0.0, 1.0), seed=99)  # SAME seed on all ranks (GA contract)
+    prob.use_comm(comm)
+    ga = GeneticAlgorithm(
+        prob, popsize=24,
+        operators=[
+            SimulatedBinaryCrossOver(prob, tournament_size=3, cross_over_rate=1.0, eta=8),
+            PolynomialMutation(prob, eta=20, mutation_probability=0.25),
+        ],
+    )
+    for _ in range(6):
+        ga.step()
+    vals = _t.Tensor.as_subclass(ga.population.values, _t.Tensor).clone()
+    ref = vals.clone()
+    comm.broadcast_(ref.reshape(-1), src=0)
+    assert _t.allclose(vals, ref, atol=1e-6), "ranks diverged"
+    return vals.reshape(-1).tolist()
+
+
+def test_nsga2_sharded_evaluate_matches_single_process():
+    r1 = _run_world("_body_nsga2_sharded", world=1)
+    r2 = _run_world("_body_nsga2_sharded", world=2)
+    v1 = torch.tensor(r1[0], dtype=torch.float64)
+    v2 = torch.tensor(r2[0], dtype=torch.float64)
+    assert torch.allclose(v1, v2, atol=1e-6), "sharded NSGA-II diverged from single-process"
+
+
 def test_world4_matches_world1():
     """Same invariance at world size 4 (the popsize-40 shards stay even,
     mirroring a 4-GPU slice of the driver's scaling ladder)."""
