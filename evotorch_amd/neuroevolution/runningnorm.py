@@ -112,6 +112,21 @@ class RunningNorm:
         if x.ndim == len(self._shape):
             x = x.unsqueeze(0)
         if mask is not None:
+            if x.is_cuda:
+                # Arithmetic masking: boolean indexing (x[mask]) calls
+                # nonzero() under the hood, which host-syncs on GPU to size
+                # the output — fatal in a per-env-step loop. Multiply-by-mask
+                # keeps shapes static and the stream free. (If the batch is
+                # nonempty but the mask all-False, has_data flips True with
+                # zero counts; mean/stdev clamp the count so normalize stays
+                # finite.)
+                m = mask.to(self._device, self._dtype).reshape(x.shape[0], *([1] * (x.ndim - 1)))
+                xm = x * m
+                self._count += mask.sum()
+                self._sum += xm.sum(dim=0)
+                self._sum_sq += (xm * x).sum(dim=0)
+                self._has_data = self._has_data or x.shape[0] > 0
+                return
             x = x[mask]
         self._count += x.shape[0]
         self._sum += x.sum(dim=0)

@@ -287,7 +287,13 @@ class VecEnvNE(NEProblem):
         active = torch.ones(n, dtype=torch.bool, device=device)
         pending = RunningNorm(shape=self._obs_dim, device=device)
         max_steps = self._max_num_steps or getattr(env.spec, "episode_length", None) or 1000
-        steps_done = 0
+        # On GPU the loop body must not host-sync: steps accumulate in a
+        # device scalar, stateless policies skip the (masked) reset, and the
+        # all-done early-exit probe runs every kCheck steps instead of every
+        # step (each bool()/int() read drains the whole HIP pipeline).
+        on_gpu = fitness.is_cuda
+        check_every = 16 if on_gpu else 1
+        steps_dev = torch.zeros((), dtype=torch.int64, device=device)
         for t in range(max_steps):
             if self._obs_norm_enabled:
                 pending.update(obs, mask=active)
@@ -302,13 +308,14 @@ class VecEnvNE(NEProblem):
             bonus = self._alive_bonus(t)
             step_reward = reward.to(device) - self._decrease_rewards_by + bonus
             fitness = fitness + step_reward * active
-            steps_done += int(active.sum())
-            newly_done = done.to(device) & active
-            if bool(newly_done.any()):
-                policy.reset(newly_done)
-            active = active & ~done.to(device)
-            if not bool(active.any()):
+            steps_dev += active.sum()
+            done = done.to(device)
+            if policy.h is not None:
+                policy.reset(done & active)
+            active = active & ~done
+            if (t + 1) % check_every == 0 and not bool(active.any()):
                 break
+        steps_done = int(steps_dev)
         batch.set_evals(fitness.to(self._eval_dtype).to(batch.device))
         self._pending_stats = pending
         self.last_eval_interaction_count = steps_done
