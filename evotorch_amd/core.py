@@ -1691,7 +1691,7 @@ class Problem(TensorMakerMixin, Serializable):
     def _get_cloned_state(self, *, memo: dict) -> dict:
         state = {}
         for k, v in self.__dict__.items():
-            if k in ("_generator", "_comm", "_grad_batch_cache", "_pregen", "_pregen_stream"):
+            if k in ("_generator", "_comm", "_grad_batch_cache", "_pregen", "_pregen_stream", "_graph_state"):
                 state[k] = None
             else:
                 state[k] = deep_clone(v, otherwise_deepcopy=True, memo=memo)

@@ -64,6 +64,12 @@ class SyntheticTorchEnv:
         self._obs: Optional[torch.Tensor] = None
         self._t = 0
 
+    # Cross-step device state for hipGraph capture (VecEnvNE
+    # use_hip_graph=True): the capture machinery re-homes these tensors
+    # into static buffers. The host counter self._t is restored by
+    # reset(), which runs before every capture/replay.
+    graph_state_attrs = ("_obs",)
+
     def reset(self, seed: int = 0) -> torch.Tensor:
         self._obs = self.spec.initial_obs(self.num_envs, 0, int(seed), device=self.device)
         self._t = 0
@@ -132,6 +138,7 @@ class VecEnvNE(NEProblem):
         max_num_envs: Optional[int] = None,
         action_noise_stdev: Optional[float] = None,
         num_episodes: int = 1,
+        use_hip_graph: bool = False,
         num_actors=None,
         num_gpus_per_actor=None,
         num_subbatches=None,
@@ -181,6 +188,8 @@ class VecEnvNE(NEProblem):
             store_solution_stats=False,
         )
         self._obs_norm = RunningNorm(shape=self._obs_dim, device=self.network_device)
+        self._use_hip_graph = bool(use_hip_graph)
+        self._graph_state: Optional[dict] = None
         self._policy: Optional[Policy] = None
         self.last_eval_interaction_count = 0
         self._total_interactions = 0
@@ -269,31 +278,16 @@ class VecEnvNE(NEProblem):
             return
         self._rollout_once(batch)
 
-    def _rollout_once(self, batch: SolutionBatch):
-        n = len(batch)
-        env = self._get_env(n)
-        device = self.network_device
-        if self._policy is None:
-            self._policy = self.make_functional_policy()
-        policy = self._policy
-        params = batch.access_values(keep_evals=True).to(device, torch.float32)
-        policy.set_parameters(params)
-
-        from ..ops.dispatch import _seed_from_generator
-
-        episode_seed = _seed_from_generator(self._generator, self._device) & 0x7FFFFFFF
-        obs = env.reset(seed=episode_seed).to(device)
-        fitness = torch.zeros(n, dtype=torch.float32, device=device)
-        active = torch.ones(n, dtype=torch.bool, device=device)
-        pending = RunningNorm(shape=self._obs_dim, device=device)
-        max_steps = self._max_num_steps or getattr(env.spec, "episode_length", None) or 1000
-        # On GPU the loop body must not host-sync: steps accumulate in a
-        # device scalar, stateless policies skip the (masked) reset, and the
-        # all-done early-exit probe runs every kCheck steps instead of every
-        # step (each bool()/int() read drains the whole HIP pipeline).
-        on_gpu = fitness.is_cuda
-        check_every = 16 if on_gpu else 1
-        steps_dev = torch.zeros((), dtype=torch.int64, device=device)
+    def _episode_body(self, env, policy, obs, active, fitness, steps_dev, pending, max_steps: int, *, early_exit: bool):
+        """One episode over the batched env, accumulating IN PLACE into
+        `fitness`/`steps_dev`/`pending`. On GPU the body never host-syncs:
+        steps accumulate in a device scalar, stateless policies skip the
+        (masked) recurrent reset, and the all-done early-exit probe runs
+        every kCheck steps instead of every step (each bool()/int() read
+        drains the whole HIP pipeline). With early_exit=False the body is
+        hipGraph-capturable (no data-dependent host control flow at all)."""
+        device = fitness.device
+        check_every = 16 if fitness.is_cuda else 1
         for t in range(max_steps):
             if self._obs_norm_enabled:
                 pending.update(obs, mask=active)
@@ -307,20 +301,159 @@ class VecEnvNE(NEProblem):
             obs = obs.to(device)
             bonus = self._alive_bonus(t)
             step_reward = reward.to(device) - self._decrease_rewards_by + bonus
-            fitness = fitness + step_reward * active
-            steps_dev += active.sum()
+            fitness.add_(step_reward * active)
+            steps_dev.add_(active.sum())
             done = done.to(device)
             if policy.h is not None:
                 policy.reset(done & active)
             active = active & ~done
-            if (t + 1) % check_every == 0 and not bool(active.any()):
+            if early_exit and (t + 1) % check_every == 0 and not bool(active.any()):
                 break
-        steps_done = int(steps_dev)
+
+    def _rollout_once(self, batch: SolutionBatch):
+        n = len(batch)
+        env = self._get_env(n)
+        device = self.network_device
+        if self._policy is None:
+            self._policy = self.make_functional_policy()
+        policy = self._policy
+        params = batch.access_values(keep_evals=True).to(device, torch.float32)
+
+        from ..ops.dispatch import _seed_from_generator
+
+        episode_seed = _seed_from_generator(self._generator, self._device) & 0x7FFFFFFF
+        max_steps = self._max_num_steps or getattr(env.spec, "episode_length", None) or 1000
+
+        if self._use_hip_graph and device.type == "cuda" and not (self._obs_norm_enabled and not self._obs_norm.has_data):
+            # (The has_data guard runs the FIRST generation eagerly so the
+            # capture does not bake normalize()'s no-data identity branch.)
+            fitness, steps_done, pending = self._rollout_graphed(env, policy, params, episode_seed, n, max_steps)
+        else:
+            policy.set_parameters(params)
+            obs = env.reset(seed=episode_seed).to(device)
+            fitness = torch.zeros(n, dtype=torch.float32, device=device)
+            active = torch.ones(n, dtype=torch.bool, device=device)
+            pending = RunningNorm(shape=self._obs_dim, device=device)
+            steps_dev = torch.zeros((), dtype=torch.int64, device=device)
+            self._episode_body(env, policy, obs, active, fitness, steps_dev, pending, max_steps, early_exit=True)
+            steps_done = int(steps_dev)
         batch.set_evals(fitness.to(self._eval_dtype).to(batch.device))
-        self._pending_stats = pending
+        if self._pending_stats is None:
+            self._pending_stats = pending
+        else:
+            # multi-episode evaluations merge at the hook; fold, don't drop
+            self._pending_stats.update(pending)
         self.last_eval_interaction_count = steps_done
         self._total_interactions += steps_done
         self._episode_count += n
+
+    # -- hipGraph-captured rollouts -------------------------------------------
+
+    def _rollout_graphed(self, env, policy, params, episode_seed: int, n: int, max_steps: int):
+        """Replay the whole T-step episode as ONE hipGraph.
+
+        The eager loop pays per-step host costs (vmap dispatch, dozens of
+        small kernel launches) that dwarf the device work for small nets;
+        capturing the unrolled episode removes all of it — one launch per
+        generation. Requirements (opt-in via use_hip_graph=True):
+        - torch-native env on the problem's CUDA device with static shapes;
+        - cross-step device state declared in `env.graph_state_attrs`
+          (host-side counters must be restored by env.reset, which runs
+          before every replay);
+        - fixed horizon: the graph always runs max_steps (done rows are
+          masked out exactly like the eager path, so fitness/steps/stats
+          are identical, but no early exit happens).
+        The capture is keyed by (n, max_steps, policy length); changing
+        popsize re-captures."""
+        gr = self._graph_state
+        key = (n, max_steps, int(params.shape[-1]), bool(self._obs_norm_enabled))
+        if gr is None or gr["key"] != key or gr["env"] is not env:
+            gr = self._capture_rollout_graph(env, policy, params, episode_seed, n, max_steps, key)
+            self._graph_state = gr
+        else:
+            gr["params_buf"].copy_(params)
+            self._graph_prepare(gr, episode_seed)
+        gr["graph"].replay()
+        pending = RunningNorm(shape=self._obs_dim, device=params.device)
+        if self._obs_norm_enabled:
+            gr["pending"]._has_data = True
+            pending.update(gr["pending"])  # snapshot: the static buffers are re-zeroed by the next replay
+        return gr["fitness"].clone(), int(gr["steps"]), pending
+
+    def _graph_prepare(self, gr: dict, episode_seed: int):
+        """Reset the env eagerly and re-home its cross-step state into the
+        capture-time static buffers (env.step REBINDS state attributes, so
+        after any episode they point at graph-internal tensors)."""
+        env = gr["env"]
+        obs0 = env.reset(seed=episode_seed).to(gr["fitness"].device)
+        for attr, buf in gr["state_bufs"].items():
+            buf.copy_(getattr(env, attr))
+            setattr(env, attr, buf)
+        if gr["obs_attr"] is None:
+            gr["obs_buf"].copy_(obs0)
+
+    def _capture_rollout_graph(self, env, policy, params, episode_seed: int, n: int, max_steps: int, key):
+        device = self.network_device
+        obs0 = env.reset(seed=episode_seed)
+        if not (torch.is_tensor(obs0) and obs0.is_cuda):
+            raise ValueError(
+                "use_hip_graph=True needs a torch-native env living on the GPU device "
+                "(numpy/gymnasium adapter envs cannot be hipGraph-captured)"
+            )
+        attrs = getattr(env, "graph_state_attrs", None)
+        if attrs is None:
+            attrs = ("_obs",) if torch.is_tensor(getattr(env, "_obs", None)) else ()
+        originals = {a: getattr(env, a) for a in attrs}
+        state_bufs = {a: v.detach().clone() for a, v in originals.items()}
+        obs_attr = next((a for a in attrs if originals[a] is obs0), None)
+        obs_buf = state_bufs[obs_attr] if obs_attr is not None else obs0.detach().clone()
+
+        params_buf = params.detach().clone()
+        policy.set_parameters(params_buf)
+        pending = RunningNorm(shape=self._obs_dim, device=device)
+        fitness = torch.zeros(n, dtype=torch.float32, device=device)
+        active = torch.ones(n, dtype=torch.bool, device=device)
+        steps = torch.zeros((), dtype=torch.int64, device=device)
+        gr = {
+            "key": key,
+            "env": env,
+            "state_bufs": state_bufs,
+            "obs_attr": obs_attr,
+            "obs_buf": obs_buf,
+            "params_buf": params_buf,
+            "pending": pending,
+            "fitness": fitness,
+            "steps": steps,
+            "graph": None,
+        }
+
+        def body():
+            pending._count.zero_()
+            pending._sum.zero_()
+            pending._sum_sq.zero_()
+            fitness.zero_()
+            steps.zero_()
+            active.fill_(True)
+            policy.reset()  # recurrent state (if any) re-initializes inside the capture
+            self._episode_body(env, policy, obs_buf, active, fitness, steps, pending, max_steps, early_exit=False)
+
+        self._graph_prepare(gr, episode_seed)
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            for _ in range(2):  # warm up allocator/BLAS workspaces off the capture
+                body()
+                self._graph_prepare(gr, episode_seed)
+        torch.cuda.current_stream().wait_stream(side)
+        torch.cuda.synchronize()
+        graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(graph):
+            body()
+        gr["graph"] = graph
+        # capture records without executing: prepare() state is still intact,
+        # so the caller's replay() right after this produces this
+        # generation's real result
+        return gr
 
     # -- policy export --------------------------------------------------------
 

@@ -33,7 +33,7 @@ def main():
     T = 200
     fused = SyntheticRolloutProblem(device="cuda:0", seed=1, episode_length=T)
     ms_fused = time_searcher(fused)
-    print(f"fused kernel path:   {ms_fused:7.2f} ms/gen  ({2048/ms_fused*1000:,.0f} sol/s)")
+    print(f"fused kernel path:      {ms_fused:8.2f} ms/gen  ({2048/ms_fused*1000:,.0f} sol/s)")
 
     general = VecEnvNE(
         lambda n: SyntheticTorchEnv(num_envs=n, episode_length=T, device="cuda:0"),
@@ -43,8 +43,19 @@ def main():
         max_num_steps=T,
     )
     ms_gen = time_searcher(general, steps=5, warmup=1)
-    print(f"general vmapped path:{ms_gen:7.2f} ms/gen  ({2048/ms_gen*1000:,.0f} sol/s)")
-    print(f"fused advantage: {ms_gen/ms_fused:.1f}x")
+    print(f"general vmapped path:   {ms_gen:8.2f} ms/gen  ({2048/ms_gen*1000:,.0f} sol/s)")
+
+    graphed = VecEnvNE(
+        lambda n: SyntheticTorchEnv(num_envs=n, episode_length=T, device="cuda:0"),
+        "Linear(obs_length, act_length)",
+        device="cuda:0",
+        seed=1,
+        max_num_steps=T,
+        use_hip_graph=True,
+    )
+    ms_graph = time_searcher(graphed, steps=10, warmup=2)
+    print(f"general + hipGraph:     {ms_graph:8.2f} ms/gen  ({2048/ms_graph*1000:,.0f} sol/s)")
+    print(f"fused vs general: {ms_gen/ms_fused:.1f}x   graph vs general: {ms_gen/ms_graph:.1f}x   fused vs graph: {ms_graph/ms_fused:.1f}x")
 
 
 main()

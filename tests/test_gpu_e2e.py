@@ -375,3 +375,46 @@ def test_graphsafe_adam_matches_host_adam():
         C.adam_step_graphsafe(out2, g, m2, v2, t_buf, 1e-2, 0.9, 0.999, 1e-8)
         assert torch.allclose(out1, out2, rtol=1e-5, atol=1e-7), t
     assert int(t_buf) == 5
+
+
+@requires_gpu
+def test_vecenvne_hip_graph_matches_eager():
+    """use_hip_graph=True replays the whole episode as one hipGraph; the
+    trajectory must match the eager loop (same ops, same order — the first
+    generation runs eagerly by design while obs-norm has no data)."""
+    from evotorch_amd.algorithms import PGPE
+    from evotorch_amd.neuroevolution import SyntheticTorchEnv, VecEnvNE
+
+    def env_factory(num_envs):
+        return SyntheticTorchEnv(num_envs=num_envs, episode_length=24, obs_dim=32, act_dim=8, rank=8, device="cuda:0")
+
+    def run(use_graph):
+        prob = VecEnvNE(env_factory, "Linear(obs_length, 16) >> Tanh() >> Linear(16, act_length)",
+                        device="cuda:0", seed=7, use_hip_graph=use_graph)
+        searcher = PGPE(prob, popsize=64, center_learning_rate=0.05, stdev_learning_rate=0.1,
+                        stdev_init=0.1, distributed=True)
+        searcher.run(4)
+        return searcher.status["mean_eval"], searcher._distribution.mu.clone(), prob.interaction_count
+
+    eval_e, mu_e, steps_e = run(False)
+    eval_g, mu_g, steps_g = run(True)
+    assert steps_e == steps_g
+    torch.testing.assert_close(mu_g, mu_e, rtol=1e-4, atol=1e-5)
+    assert abs(float(eval_g) - float(eval_e)) < 1e-2 * max(1.0, abs(float(eval_e)))
+
+
+@requires_gpu
+def test_vecenvne_hip_graph_recaptures_on_popsize_change():
+    from evotorch_amd.neuroevolution import SyntheticTorchEnv, VecEnvNE
+    from evotorch_amd.core import SolutionBatch
+
+    def env_factory(num_envs):
+        return SyntheticTorchEnv(num_envs=num_envs, episode_length=12, obs_dim=32, act_dim=8, rank=8, device="cuda:0")
+
+    prob = VecEnvNE(env_factory, "Linear(obs_length, act_length)", device="cuda:0", seed=5,
+                    use_hip_graph=True, observation_normalization=False)
+    for n in (16, 16, 32, 16):
+        batch = SolutionBatch(prob, popsize=n, device="cuda:0")
+        batch.access_values()[:] = 0.01 * torch.randn(n, prob.solution_length, device="cuda:0")
+        prob.evaluate(batch)
+        assert batch.evals_are_ready
