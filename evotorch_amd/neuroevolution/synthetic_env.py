@@ -92,12 +92,22 @@ class SyntheticEnvSpec:
         return torch.cat([cache[1], tail]).contiguous()
 
     def initial_obs(self, n_members: int, member_offset: int, init_seed: int, device=None) -> torch.Tensor:
-        """0.1 * N(0,1), matching the kernel's philox stream exactly."""
-        from .philox_ref import philox_normal_rows
+        """0.1 * N(0,1), matching the kernel's philox stream exactly.
 
+        Routed through ops.sample_gaussian (mu=0, sigma=0.1): the HIP K1
+        kernel on GPU, the vectorized numpy philox reference on CPU —
+        bitwise-identical values either way (fmaf(0.1, z, 0) == 0.1*z),
+        and no per-row Python loop (env.reset was costing ~150 ms per
+        generation at 2048 envs through the scalar reference)."""
+        from ..ops import sample_gaussian
+
+        dev = torch.device(device or self.device)
         O = self.obs_dim
-        z = philox_normal_rows(init_seed, member_offset, n_members, O)
-        return (0.1 * z).to(device or self.device)
+        out = torch.empty(n_members, O, dtype=torch.float32, device=dev)
+        mu = torch.zeros(O, dtype=torch.float32, device=dev)
+        sigma = torch.full((O,), 0.1, dtype=torch.float32, device=dev)
+        sample_gaussian(out, mu, sigma, seed=int(init_seed), row_offset=int(member_offset))
+        return out
 
 
 def rollout_eager(

@@ -288,10 +288,25 @@ class VecEnvNE(NEProblem):
         hipGraph-capturable (no data-dependent host control flow at all)."""
         device = fitness.device
         check_every = 16 if fitness.is_cuda else 1
+        if self._obs_norm_enabled and self._obs_norm.has_data:
+            # obs-norm statistics are frozen for the whole episode (updates
+            # merge after evaluation), so hoist mean/stdev out of the step
+            # loop; inside a graph capture this recomputes once per REPLAY
+            # from the in-place-updated buffers.
+            norm_mean = self._obs_norm.mean.to(device)
+            norm_stdev = self._obs_norm.stdev.to(device)
+        else:
+            norm_mean = norm_stdev = None
+        norm_clip = getattr(self._obs_norm, "_clip", None)
         for t in range(max_steps):
             if self._obs_norm_enabled:
                 pending.update(obs, mask=active)
-                obs_in = self._obs_norm.normalize(obs)
+                if norm_mean is None:
+                    obs_in = obs
+                else:
+                    obs_in = (obs - norm_mean) / norm_stdev
+                    if norm_clip is not None:
+                        obs_in = torch.clamp(obs_in, norm_clip[0], norm_clip[1])
             else:
                 obs_in = obs
             actions = policy(obs_in)
