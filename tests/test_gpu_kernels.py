@@ -404,13 +404,20 @@ def test_streaming_large_l_smoke():
                    seed=3, device="cuda:0")
     searcher = PGPE(prob, popsize=32, center_learning_rate=0.05, stdev_learning_rate=0.05,
                     stdev_init=0.1, distributed=True, grad_chunk_rows=4)
+    # isolate from residue of earlier tests (e.g. hipGraph private pools
+    # whose owners are awaiting garbage collection)
+    import gc
+
+    gc.collect()
+    torch.cuda.empty_cache()
+    base = torch.cuda.memory_allocated() / 2**30
     torch.cuda.reset_peak_memory_stats()
     searcher.step()
     searcher.step()
-    peak = torch.cuda.max_memory_allocated() / 2**30
+    peak = torch.cuda.max_memory_allocated() / 2**30 - base
     # materialized would need 32*20e6*4 = 2.4 GB for the population alone
     # (plus allocator churn); chunked path stays near parameter-vector cost
-    assert peak < 2.2, f"peak {peak:.2f} GiB — streaming not effective"
+    assert peak < 2.2, f"peak {peak:.2f} GiB above baseline — streaming not effective"
     assert searcher.step_count == 2
 
 
