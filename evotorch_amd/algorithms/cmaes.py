@@ -17,6 +17,7 @@ from typing import Optional
 
 import torch
 
+from .. import ops
 from ..core import Problem, SolutionBatch
 from ..utils import RealOrVector, to_stdev_init
 from ..utils.misc import ensure_tensor_length_and_dtype
@@ -344,17 +345,38 @@ class CMAES(SearchAlgorithm, SinglePopulationAlgorithmMixin):
         return self._m.device.type == "cuda"
 
     @staticmethod
-    def _blocked_cholesky(C: torch.Tensor, block: int = 512) -> torch.Tensor:
+    def _blocked_cholesky(C: torch.Tensor, block: int = 512, panel: int = 128) -> torch.Tensor:
         """Right-looking blocked Cholesky composed from rocBLAS trsm + gemm
-        (the O(n³) mass) with rocSOLVER only on the small diagonal panels.
-        rocSOLVER's monolithic potrf at d=4096 runs at ~1.8 TFLOP/s on
-        MI355X; the trailing syrk-style updates through rocBLAS run the
-        same flops several times faster."""
+        (the O(n³) mass) with the hand in-LDS panel kernel
+        (ops.potrf_tile_) on the ≤128-wide diagonal panels. rocSOLVER's
+        monolithic potrf at d=4096 runs at ~1.8 TFLOP/s on MI355X and its
+        unblocked potf2 chain of tiny kernels dominated the CMA-ES
+        refresh profile; this keeps the trailing syrk-style updates in
+        rocBLAS and leaves NO rocSOLVER kernel in the factorization.
+        Non-PD input raises (after one device read at the end), matching
+        torch.linalg.cholesky for the caller's jitter-retry path."""
         A = C.clone()
         n = A.shape[0]
+        on_gpu = A.is_cuda
+        info = torch.zeros(1, dtype=torch.int32, device=A.device) if on_gpu else None
         for k in range(0, n, block):
             e = min(k + block, n)
-            A[k:e, k:e] = torch.linalg.cholesky(A[k:e, k:e])
+            if on_gpu:
+                blk = A[k:e, k:e]
+                m = e - k
+                for p in range(0, m, panel):
+                    q = min(p + panel, m)
+                    ops.potrf_tile_(blk[p:q, p:q], info)
+                    if q < m:
+                        # trailing of the diagonal block (the .mT views read
+                        # only our L; the panel's stale upper half is inert)
+                        blk[q:, p:q] = torch.linalg.solve_triangular(
+                            blk[p:q, p:q].mT, blk[q:, p:q], upper=True, left=False
+                        )
+                        Lp = blk[q:, p:q]
+                        blk[q:, q:] -= Lp @ Lp.T
+            else:
+                A[k:e, k:e] = torch.linalg.cholesky(A[k:e, k:e])
             if e < n:
                 # L21 = A21 · L11⁻ᵀ  ⇔  X · L11ᵀ = A21 (right triangular solve)
                 A[e:, k:e] = torch.linalg.solve_triangular(
@@ -362,6 +384,8 @@ class CMAES(SearchAlgorithm, SinglePopulationAlgorithmMixin):
                 )
                 L21 = A[e:, k:e]
                 A[e:, e:] -= L21 @ L21.T
+        if info is not None and int(info.item()) != 0:
+            raise RuntimeError(f"cholesky: matrix not positive-definite (panel column {int(info.item()) - 1})")
         return torch.tril(A)
 
     def _cholesky(self, C: torch.Tensor) -> torch.Tensor:

@@ -15,6 +15,7 @@ from .dispatch import (
     hip_required,
     load_hip,
     pareto_ranks,
+    potrf_tile_,
     sample_gaussian,
     snes_gradients,
 )
@@ -31,5 +32,6 @@ __all__ = [
     "sample_gaussian",
     "snes_gradients",
     "pareto_ranks",
+    "potrf_tile_",
     "domination_counts",
 ]

@@ -406,3 +406,29 @@ def cma_update_c_(
     C.mul_(scale).add_(rank_one, alpha=c1).add_(rank_mu, alpha=cmu)
     C.copy_(0.5 * (C + C.T))
     return C
+
+
+def potrf_tile_(A: torch.Tensor, info: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """In-place lower Cholesky of one SPD panel (n ≤ 128). On GPU this is
+    ONE kernel with the panel LDS-resident (ops/hip/cma.hip
+    potrf_panel_kernel) — the building block that removes rocSOLVER's
+    small-potf2 chain from the blocked factorization
+    (algorithms/cmaes.py _blocked_cholesky). `A` may be a strided view
+    (a diagonal block of a larger matrix); its upper triangle is left
+    untouched. A non-positive pivot at column j sets `info` (a device
+    int32 scalar) to j+1 instead of raising — callers check once per
+    factorization, not per panel.
+
+    Eager/CPU reference: torch.linalg.cholesky (raises on failure, which
+    the CPU callers rely on; it also zeroes the upper triangle, so
+    callers must not rely on the upper half either way)."""
+    if A.ndim != 2 or A.shape[0] != A.shape[1]:
+        raise ValueError(f"expected a square panel, got {tuple(A.shape)}")
+    if A.device.type == "cuda" and not _allow_eager_on_gpu():
+        mod = hip_required()
+        if info is None:
+            info = torch.zeros(1, dtype=torch.int32, device=A.device)
+        mod.potrf_tile(A, info)
+        return A
+    A.copy_(torch.linalg.cholesky(A))
+    return A

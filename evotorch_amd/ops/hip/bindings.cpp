@@ -20,6 +20,7 @@ torch::Tensor rollout_linear(torch::Tensor params, torch::Tensor env_blob, torch
                              double act_cost, int64_t init_seed, int64_t member_offset, int64_t policy_hidden);
 void cma_update_c(torch::Tensor C, torch::Tensor Y, torch::Tensor w, torch::Tensor pc, torch::Tensor hs_f,
                   torch::Tensor wsum, double c1, double cmu, double cc);
+void potrf_tile(torch::Tensor A, torch::Tensor info);
 torch::Tensor fused_rank(torch::Tensor fitnesses, int64_t method, bool higher_better);
 torch::Tensor domination_counts(torch::Tensor utils);
 torch::Tensor pareto_ranks(torch::Tensor utils, int64_t min_assigned);
@@ -49,6 +50,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           pybind11::arg("member_offset"), pybind11::arg("policy_hidden") = 0);
     m.def("cma_update_c", &ea::cma_update_c,
           "K5: fused CMA-ES covariance update (scale*C + c1*pc pc^T + cmu*Y^T diag(w) Y, exact symmetry)");
+    m.def("potrf_tile", &ea::potrf_tile,
+          "K5b: in-LDS Cholesky of one <=128x128 SPD diagonal panel (in place, device info flag)");
     m.def("fused_rank", &ea::fused_rank,
           "K2: fused bitonic ranking + utility map (centered/linear/nes) in one launch");
     m.def("domination_counts", &ea::domination_counts, "K7: NSGA-II domination counts (no N x N matrix)");

@@ -90,6 +90,85 @@ __global__ __launch_bounds__(256) void cma_update_c_kernel(
     }
 }
 
+// ---------------------------------------------------------------------------
+// In-LDS Cholesky of one <=128x128 SPD diagonal panel (reference reaches
+// this through torch.linalg.cholesky -> rocSOLVER, whose unblocked potf2
+// runs a chain of tiny kernels that dominated the d=4096 CMA-ES profile —
+// profiles/cma_phase_attribution_r2.log). ONE launch: stage the panel
+// into LDS (row stride 132 dwords keeps column walks off the 64-bank
+// alias), right-looking rank-1 factorization with the active column
+// mirrored in a flat buffer (conflict-free broadcast reads in the
+// trailing update), write back the lower triangle. The blocked driver
+// (cmaes.py _blocked_cholesky) keeps rocBLAS trsm/gemm for the O(n³)
+// mass and calls this for every diagonal panel, so no rocSOLVER kernel
+// remains in the factorization.
+
+constexpr int kPotrfMax = 128;
+constexpr int kPotrfPad = kPotrfMax + 4;
+
+__global__ __launch_bounds__(256) void potrf_panel_kernel(float* __restrict__ A, long lda, int n,
+                                                          int* __restrict__ info) {
+    extern __shared__ float T[];  // kPotrfMax * kPotrfPad + kPotrfMax floats
+    float* colj = T + kPotrfMax * kPotrfPad;
+    const int tid = threadIdx.x;
+    for (int e = tid; e < n * n; e += 256) {
+        const int r = e / n, c = e % n;
+        T[r * kPotrfPad + c] = A[(long)r * lda + c];
+    }
+    __syncthreads();
+    for (int j = 0; j < n; ++j) {
+        // every thread derives the pivot locally (no single-writer round
+        // trip). T[j][j] is NOT overwritten here: waves read d at their own
+        // pace after the previous barrier, so the commit of sqrt(d) waits
+        // until after the next barrier (nothing reads T[j][j] again before
+        // the write-back barrier).
+        const float d = T[j * kPotrfPad + j];
+        const float s = sqrtf(d);
+        const float inv = 1.0f / s;
+        for (int i = j + 1 + tid; i < n; i += 256) {
+            const float v = T[i * kPotrfPad + j] * inv;
+            T[i * kPotrfPad + j] = v;
+            colj[i] = v;
+        }
+        __syncthreads();
+        if (tid == 0) {
+            if (!(d > 0.0f)) atomicCAS(info, 0, j + 1);
+            T[j * kPotrfPad + j] = s;
+        }
+        const int m = n - j - 1;
+        for (int e = tid; e < m * m; e += 256) {
+            const int r = j + 1 + e / m, c = j + 1 + e % m;
+            if (r >= c) T[r * kPotrfPad + c] = fmaf(-colj[r], colj[c], T[r * kPotrfPad + c]);
+        }
+        __syncthreads();
+    }
+    for (int e = tid; e < n * n; e += 256) {
+        const int r = e / n, c = e % n;
+        if (r >= c) A[(long)r * lda + c] = T[r * kPotrfPad + c];
+    }
+}
+
+void potrf_tile(torch::Tensor A, torch::Tensor info) {
+    CHECK_GPU_C(A);
+    TORCH_CHECK(A.dim() == 2 && A.size(0) == A.size(1), "A must be square");
+    TORCH_CHECK(A.size(0) <= kPotrfMax, "panel larger than ", kPotrfMax);
+    TORCH_CHECK(A.stride(1) == 1, "A rows must be contiguous");
+    TORCH_CHECK(A.scalar_type() == at::ScalarType::Float, "A must be fp32");
+    TORCH_CHECK(info.is_cuda() && info.scalar_type() == at::ScalarType::Int && info.numel() >= 1,
+                "info must be a device int32 scalar");
+    const int n = (int)A.size(0);
+    const size_t lds = (size_t)(kPotrfMax * kPotrfPad + kPotrfMax) * sizeof(float);
+    static bool attr_set = false;
+    if (!attr_set) {
+        (void)hipFuncSetAttribute(reinterpret_cast<const void*>(&potrf_panel_kernel),
+                                  hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
+        attr_set = true;
+    }
+    auto stream = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(potrf_panel_kernel, dim3(1), dim3(256), lds, stream, A.data_ptr<float>(),
+                       (long)A.stride(0), n, info.data_ptr<int>());
+}
+
 void cma_update_c(torch::Tensor C, torch::Tensor Y, torch::Tensor w, torch::Tensor pc, torch::Tensor hs_f,
                   torch::Tensor wsum, double c1, double cmu, double cc) {
     CHECK_GPU_C(C);

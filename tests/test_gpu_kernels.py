@@ -617,3 +617,52 @@ def test_mapelites_assign_matches_eager():
     ref_valid = inside.any(dim=1)
     assert torch.equal(valid, ref_valid)
     assert torch.equal(best[ref_valid], ref_best[ref_valid])
+
+
+@requires_gpu
+def test_potrf_tile_matches_torch():
+    """K5b panel Cholesky vs torch.linalg.cholesky: sizes across the
+    range, a strided diagonal-block view, and the non-PD info flag."""
+    from evotorch_amd.ops import potrf_tile_
+
+    torch.manual_seed(11)
+    info = torch.zeros(1, dtype=torch.int32, device="cuda:0")
+    for n in (1, 7, 37, 64, 100, 128):
+        A = torch.randn(n, n, device="cuda:0")
+        C = A @ A.T + n * torch.eye(n, device="cuda:0")
+        ref = torch.linalg.cholesky(C)
+        out = C.clone()
+        info.zero_()
+        potrf_tile_(out, info)
+        assert int(info.item()) == 0
+        torch.testing.assert_close(torch.tril(out), ref, rtol=3e-5, atol=3e-5)
+    # strided view: a diagonal block inside a larger matrix
+    M = torch.randn(200, 200, device="cuda:0")
+    M = M @ M.T + 200 * torch.eye(200, device="cuda:0")
+    blk = M[64:192, 64:192]
+    ref = torch.linalg.cholesky(blk.clone())
+    info.zero_()
+    potrf_tile_(blk, info)
+    assert int(info.item()) == 0
+    torch.testing.assert_close(torch.tril(blk), ref, rtol=3e-5, atol=3e-5)
+    # non-positive pivot reports its column in info
+    B = torch.eye(16, device="cuda:0")
+    B[5, 5] = -1.0
+    info.zero_()
+    potrf_tile_(B, info)
+    assert int(info.item()) == 6
+
+
+@requires_gpu
+def test_blocked_cholesky_gpu_matches_torch():
+    from evotorch_amd.algorithms.cmaes import CMAES
+
+    torch.manual_seed(12)
+    n = 1500  # not a multiple of the 512 block or the 128 panel
+    A = torch.randn(n, n, device="cuda:0")
+    C = A @ A.T + n * torch.eye(n, device="cuda:0")
+    L1 = CMAES._blocked_cholesky(C)
+    L2 = torch.linalg.cholesky(C)
+    torch.testing.assert_close(L1, L2, rtol=1e-3, atol=1e-3)
+    with pytest.raises(RuntimeError):
+        CMAES._blocked_cholesky(-torch.eye(700, device="cuda:0"))
