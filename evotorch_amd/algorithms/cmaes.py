@@ -150,6 +150,7 @@ class CMAES(SearchAlgorithm, SinglePopulationAlgorithmMixin):
         self._decomp_stream = None  # side-stream pipelined potrf state
         self._decomp_ready = None
         self._pending_A = None
+        self._decomp_info = None  # deferred non-PD flag of the async factorization
         self._pending_info = None
         self._population: Optional[SolutionBatch] = None
         SinglePopulationAlgorithmMixin.__init__(self, exclude={"mean_eval"})
@@ -280,16 +281,26 @@ class CMAES(SearchAlgorithm, SinglePopulationAlgorithmMixin):
         interval = self._decompose_interval
         if interval >= 4 and self._device_is_gpu():
             if self._steps_since_decompose >= max(1, interval // 2) and self._pending_A is None:
-                # snapshot + async factorization on the side stream
+                # deferred non-PD check of the PREVIOUS async factorization:
+                # by now its side-stream work finished long ago, so the read
+                # costs no wait; on failure, repair synchronously through
+                # the jitter path
+                if self._decomp_info is not None and int(self._decomp_info.item()) != 0:
+                    self._decomp_info.zero_()
+                    self._A = self._cholesky(self._C)
+                # snapshot + async factorization on the side stream (no
+                # host sync anywhere: the failure flag stays on device)
                 if self._decomp_stream is None:
                     self._decomp_stream = torch.cuda.Stream(device=self._C.device)
                     self._decomp_ready = torch.cuda.Event()
+                if self._decomp_info is None:
+                    self._decomp_info = torch.zeros(1, dtype=torch.int32, device=self._C.device)
                 snap_ready = torch.cuda.Event()
                 snap_ready.record()
                 with torch.cuda.stream(self._decomp_stream):
                     self._decomp_stream.wait_event(snap_ready)
                     c_snap = self._C.clone()
-                    self._pending_A = self._cholesky(c_snap)
+                    self._pending_A = self._cholesky(c_snap, info_out=self._decomp_info)
                     self._decomp_ready.record()
             if self._steps_since_decompose >= interval and self._pending_A is not None:
                 torch.cuda.current_stream().wait_event(self._decomp_ready)
@@ -345,7 +356,9 @@ class CMAES(SearchAlgorithm, SinglePopulationAlgorithmMixin):
         return self._m.device.type == "cuda"
 
     @staticmethod
-    def _blocked_cholesky(C: torch.Tensor, block: int = 512, panel: int = 128) -> torch.Tensor:
+    def _blocked_cholesky(
+        C: torch.Tensor, block: int = 512, panel: int = 128, info_out: Optional[torch.Tensor] = None
+    ) -> torch.Tensor:
         """Right-looking blocked Cholesky composed from rocBLAS trsm + gemm
         (the O(n³) mass) with the hand in-LDS panel kernel
         (ops.potrf_tile_) on the ≤128-wide diagonal panels. rocSOLVER's
@@ -354,11 +367,18 @@ class CMAES(SearchAlgorithm, SinglePopulationAlgorithmMixin):
         refresh profile; this keeps the trailing syrk-style updates in
         rocBLAS and leaves NO rocSOLVER kernel in the factorization.
         Non-PD input raises (after one device read at the end), matching
-        torch.linalg.cholesky for the caller's jitter-retry path."""
+        torch.linalg.cholesky for the caller's jitter-retry path — UNLESS
+        `info_out` is given, in which case the failure flag is left in
+        that device scalar and nothing host-syncs (for the side-stream
+        pipelined refresh, which checks it one interval later)."""
         A = C.clone()
         n = A.shape[0]
         on_gpu = A.is_cuda
-        info = torch.zeros(1, dtype=torch.int32, device=A.device) if on_gpu else None
+        if info_out is not None:
+            info = info_out
+            info.zero_()
+        else:
+            info = torch.zeros(1, dtype=torch.int32, device=A.device) if on_gpu else None
         for k in range(0, n, block):
             e = min(k + block, n)
             if on_gpu:
@@ -384,14 +404,14 @@ class CMAES(SearchAlgorithm, SinglePopulationAlgorithmMixin):
                 )
                 L21 = A[e:, k:e]
                 A[e:, e:] -= L21 @ L21.T
-        if info is not None and int(info.item()) != 0:
+        if info_out is None and info is not None and int(info.item()) != 0:
             raise RuntimeError(f"cholesky: matrix not positive-definite (panel column {int(info.item()) - 1})")
         return torch.tril(A)
 
-    def _cholesky(self, C: torch.Tensor) -> torch.Tensor:
+    def _cholesky(self, C: torch.Tensor, info_out: Optional[torch.Tensor] = None) -> torch.Tensor:
         try:
             if self._device_is_gpu() and C.shape[0] > 1024:
-                return self._blocked_cholesky(C)
+                return self._blocked_cholesky(C, info_out=info_out)
             return torch.linalg.cholesky(C)
         except Exception:
             # regularize on failure
