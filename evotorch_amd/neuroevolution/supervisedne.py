@@ -67,6 +67,7 @@ class SupervisedNE(NEProblem):
         self._dataloader: Optional[DataLoader] = None
         self._dataloader_iter = None
         self._fmodule = None
+        self._net_obj = None
         self._subbatch_size = None if subbatch_size is None else int(subbatch_size)
         self._num_subbatches = None if num_subbatches is None else int(num_subbatches)
 
@@ -135,21 +136,35 @@ class SupervisedNE(NEProblem):
         # minibatch — a middle ground between per-solution minibatches
         # (high variance between solutions) and one global minibatch.
         if self._fmodule is None:
-            self._fmodule = make_functional_module(self._instantiate_net().to(self.network_device))
+            self._net_obj = self._instantiate_net().to(self.network_device)
+            self._fmodule = make_functional_module(self._net_obj)
         params = batch.access_values(keep_evals=True).to(self.network_device, torch.float32)
         n = len(batch)
         sub = self._subbatch_size
         if sub is None and self._num_subbatches is not None:
             sub = max(1, (n + self._num_subbatches - 1) // self._num_subbatches)
         losses = torch.zeros(n, dtype=torch.float32, device=self.network_device)
+        # population_forward protocol: a net may provide its own
+        # population-batched forward `population_forward(params (G, P),
+        # x) -> (G, N, *out)` — e.g. convnets mapping member weights onto
+        # channel GROUPS so the shared minibatch stays in one
+        # MIOpen-friendly layout instead of vmap's grouped/naive conv
+        # lowering plus per-op reshapes (2.1x on the MNIST30K benchmark —
+        # scripts/bench_supervised.py).
+        pop_fwd = getattr(self._net_obj, "population_forward", None)
         with torch.no_grad():
             for _ in range(self._num_minibatches):
                 for start in range(0, n, sub or n):
                     stop = min(start + (sub or n), n)
                     x, y = self.get_minibatch()
 
-                    def member_loss(flat):
-                        return self._loss(self._fmodule._single(flat, x), y)
+                    if pop_fwd is not None:
+                        preds = pop_fwd(params[start:stop], x)
+                        chunk = torch.func.vmap(lambda p: self._loss(p, y))(preds)
+                    else:
+                        def member_loss(flat):
+                            return self._loss(self._fmodule._single(flat, x), y)
 
-                    losses[start:stop] = losses[start:stop] + torch.func.vmap(member_loss, randomness="different")(params[start:stop])
+                        chunk = torch.func.vmap(member_loss, randomness="different")(params[start:stop])
+                    losses[start:stop] = losses[start:stop] + chunk
         batch.set_evals((losses / self._num_minibatches).to(batch.device))

@@ -50,6 +50,37 @@ class MNIST30KNet(nn.Module):
         x = self.norm(x.flatten(start_dim=-3))
         return self.out(x)
 
+    def population_forward(self, params: torch.Tensor, x: torch.Tensor) -> torch.Tensor:
+        """SupervisedNE population_forward protocol: the whole member
+        subbatch in ONE channel-grouped pass. Member weights become
+        channel groups (conv1: one plain conv with G·16 output channels
+        over the SHARED minibatch; conv2: groups=G), so the data stays in
+        a single MIOpen-friendly NCHW layout — no vmap grouped/naive conv
+        lowering, no per-op member-dim reshapes. Exactly the same math as
+        vmapping `forward` over members (same parameter flattening order:
+        conv1.w, conv1.b, conv2.w, conv2.b, out.w, out.b)."""
+        import torch.nn.functional as F
+
+        G, N = params.shape[0], x.shape[0]
+        i = 0
+
+        def take(k):
+            nonlocal i
+            v = params[:, i : i + k]
+            i += k
+            return v
+
+        w1 = take(400).reshape(G * 16, 1, 5, 5)
+        b1 = take(16).reshape(G * 16)
+        w2 = take(12800).reshape(G * 32, 16, 5, 5)
+        b2 = take(32).reshape(G * 32)
+        w3 = take(15680).reshape(G, 10, 1568)
+        b3 = take(10)
+        h = F.max_pool2d(F.relu(F.conv2d(x, w1, b1, padding=2)), 2)
+        h = F.max_pool2d(F.relu(F.conv2d(h, w2, b2, padding=2, groups=G)), 2)
+        f = F.layer_norm(h.reshape(N, G, 1568), (1568,))
+        return torch.baddbmm(b3.unsqueeze(1), f.transpose(0, 1), w3.transpose(1, 2))
+
 
 def synthetic_mnist(n: int, seed: int = 0):
     """MNIST-shaped synthetic classification: 10 random 28×28 prototypes +

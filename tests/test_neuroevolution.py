@@ -406,3 +406,47 @@ def test_supervised_mnist30k_config_descends():
     d = json.loads(result.stdout.strip().splitlines()[-1])
     assert d["config"]["final_mean_loss"] < d["config"]["first_mean_loss"] + 0.3
     assert 28000 < int(d["config"]["model"].split("(")[1].split()[0]) < 30000
+
+
+def test_supervisedne_population_forward_protocol():
+    """A net providing `population_forward` must yield the same losses as
+    the vmapped per-member path (MNIST30KNet's grouped-conv
+    implementation, small shapes, CPU)."""
+    import os
+    import sys as _sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    _sys.path.insert(0, os.path.join(repo, "scripts"))
+    from bench_supervised import MNIST30KNet, synthetic_mnist
+
+    from evotorch_amd.core import SolutionBatch
+    from evotorch_amd.neuroevolution import SupervisedNE
+
+    def make(problem_net):
+        return SupervisedNE(
+            synthetic_mnist(64, seed=5),
+            problem_net,
+            loss_func=lambda y_hat, y: torch.nn.functional.cross_entropy(y_hat, y),
+            minibatch_size=16,
+            common_minibatch=True,
+            subbatch_size=3,
+            seed=9,
+        )
+
+    class PlainNet(MNIST30KNet):
+        population_forward = None  # force the vmap path
+
+    torch.manual_seed(2)
+    prob_pop = make(lambda: MNIST30KNet())
+    prob_vmap = make(lambda: PlainNet())
+    vals = 0.1 * torch.randn(7, prob_pop.solution_length)
+
+    def losses(prob):
+        b = SolutionBatch(prob, popsize=7)
+        b.access_values()[:] = vals
+        prob.evaluate(b)
+        return torch.Tensor.as_subclass(b.evals[:, 0], torch.Tensor).clone()
+
+    lp = losses(prob_pop)
+    lv = losses(prob_vmap)
+    torch.testing.assert_close(lp, lv, rtol=1e-5, atol=1e-6)
