@@ -147,9 +147,9 @@ class SupervisedNE(NEProblem):
         # population_forward protocol: a net may provide its own
         # population-batched forward `population_forward(params (G, P),
         # x) -> (G, N, *out)` — e.g. convnets mapping member weights onto
-        # channel GROUPS so the shared minibatch stays in one
-        # MIOpen-friendly layout instead of vmap's grouped/naive conv
-        # lowering plus per-op reshapes (2.1x on the MNIST30K benchmark —
+        # channel GROUPS so the shared minibatch stays in one NCHW layout
+        # end-to-end, skipping vmap's per-op member-dim reshapes (1.16x
+        # on the MNIST30K benchmark, which is otherwise conv-FLOP-bound —
         # scripts/bench_supervised.py).
         pop_fwd = getattr(self._net_obj, "population_forward", None)
         with torch.no_grad():
