@@ -12,7 +12,8 @@ closes that gap:
   `spaces.Box` / `spaces.Discrete`, `make(name)` over a registry of
   classic-control environments written here from their textbook dynamics
   (cart-pole per Barto, Sutton & Anderson 1983; torque-limited pendulum
-  swing-up), and `SyncVectorEnv` with SAME-STEP AUTORESET: when a sub-env
+  swing-up; mountain car per Moore 1990, discrete and continuous; acrobot
+  per Sutton 1996 with RK4), and `SyncVectorEnv` with SAME-STEP AUTORESET: when a sub-env
   ends, the batched step returns the freshly reset observation while
   `info["final_observation"][i]` carries the terminal one — the contract
   of the reference's own vector env.
@@ -28,9 +29,12 @@ from typing import Callable, List, Optional, Sequence
 import numpy as np
 
 __all__ = [
+    "AcrobotEnv",
     "Box",
     "CartPoleEnv",
     "Discrete",
+    "MountainCarContinuousEnv",
+    "MountainCarEnv",
     "PendulumEnv",
     "SyncVectorEnv",
     "have_real_gymnasium",
@@ -226,6 +230,186 @@ class PendulumEnv(_EnvBase):
         return self._obs(), -cost, False, truncated, {}
 
 
+class MountainCarEnv(_EnvBase):
+    """Under-powered car in a valley (Moore 1990 dynamics).
+
+    Observation: [position, velocity]; action: Discrete(3) push
+    left/none/right; reward −1 per step; terminates at the right hilltop
+    (position ≥ 0.5); truncates at 200 steps."""
+
+    MIN_POS, MAX_POS = -1.2, 0.6
+    MAX_SPEED = 0.07
+    GOAL_POS = 0.5
+    FORCE = 0.001
+    GRAVITY = 0.0025
+
+    def __init__(self, max_episode_steps: int = 200):
+        super().__init__()
+        lo = np.array([self.MIN_POS, -self.MAX_SPEED], dtype=np.float32)
+        hi = np.array([self.MAX_POS, self.MAX_SPEED], dtype=np.float32)
+        self.observation_space = Box(lo, hi)
+        self.action_space = Discrete(3)
+        self._max_steps = int(max_episode_steps)
+        self._pos = 0.0
+        self._vel = 0.0
+        self._t = 0
+
+    def _obs(self):
+        return np.array([self._pos, self._vel], dtype=np.float32)
+
+    def reset(self, *, seed: Optional[int] = None, options=None):
+        self._seed(seed)
+        self._pos = float(self._rng.uniform(-0.6, -0.4))
+        self._vel = 0.0
+        self._t = 0
+        return self._obs(), {}
+
+    def step(self, action):
+        self._vel += (int(action) - 1) * self.FORCE - self.GRAVITY * math.cos(3.0 * self._pos)
+        self._vel = float(np.clip(self._vel, -self.MAX_SPEED, self.MAX_SPEED))
+        self._pos = float(np.clip(self._pos + self._vel, self.MIN_POS, self.MAX_POS))
+        if self._pos <= self.MIN_POS and self._vel < 0.0:
+            self._vel = 0.0  # inelastic left wall
+        self._t += 1
+        terminated = bool(self._pos >= self.GOAL_POS)
+        truncated = bool(self._t >= self._max_steps)
+        return self._obs(), -1.0, terminated, truncated, {}
+
+
+class MountainCarContinuousEnv(_EnvBase):
+    """Continuous-torque mountain car.
+
+    Observation: [position, velocity]; action: Box(−1, 1) engine force;
+    reward +100 on reaching the goal minus 0.1·u² per step; terminates at
+    position ≥ 0.45; truncates at 999 steps."""
+
+    MIN_POS, MAX_POS = -1.2, 0.6
+    MAX_SPEED = 0.07
+    GOAL_POS = 0.45
+    POWER = 0.0015
+    GRAVITY = 0.0025
+
+    def __init__(self, max_episode_steps: int = 999):
+        super().__init__()
+        lo = np.array([self.MIN_POS, -self.MAX_SPEED], dtype=np.float32)
+        hi = np.array([self.MAX_POS, self.MAX_SPEED], dtype=np.float32)
+        self.observation_space = Box(lo, hi)
+        self.action_space = Box(np.array([-1.0], dtype=np.float32), np.array([1.0], dtype=np.float32))
+        self._max_steps = int(max_episode_steps)
+        self._pos = 0.0
+        self._vel = 0.0
+        self._t = 0
+
+    def _obs(self):
+        return np.array([self._pos, self._vel], dtype=np.float32)
+
+    def reset(self, *, seed: Optional[int] = None, options=None):
+        self._seed(seed)
+        self._pos = float(self._rng.uniform(-0.6, -0.4))
+        self._vel = 0.0
+        self._t = 0
+        return self._obs(), {}
+
+    def step(self, action):
+        u = float(np.clip(np.asarray(action).reshape(-1)[0], -1.0, 1.0))
+        self._vel += u * self.POWER - self.GRAVITY * math.cos(3.0 * self._pos)
+        self._vel = float(np.clip(self._vel, -self.MAX_SPEED, self.MAX_SPEED))
+        self._pos = float(np.clip(self._pos + self._vel, self.MIN_POS, self.MAX_POS))
+        if self._pos <= self.MIN_POS and self._vel < 0.0:
+            self._vel = 0.0
+        self._t += 1
+        terminated = bool(self._pos >= self.GOAL_POS)
+        truncated = bool(self._t >= self._max_steps)
+        reward = (100.0 if terminated else 0.0) - 0.1 * u * u
+        return self._obs(), reward, terminated, truncated, {}
+
+
+class AcrobotEnv(_EnvBase):
+    """Two-link underactuated pendulum swing-up (Sutton 1996 dynamics,
+    book variant, RK4 integration).
+
+    Observation: [cos θ1, sin θ1, cos θ2, sin θ2, θ̇1, θ̇2]; action:
+    Discrete(3) torque {−1, 0, +1} on the SECOND joint; reward −1 per
+    step; terminates when the tip rises above one link length
+    (−cos θ1 − cos(θ1+θ2) > 1); truncates at 500 steps."""
+
+    DT = 0.2
+    M1 = M2 = 1.0
+    L1 = 1.0
+    LC1 = LC2 = 0.5
+    I1 = I2 = 1.0
+    G = 9.8
+    MAX_VEL1 = 4.0 * math.pi
+    MAX_VEL2 = 9.0 * math.pi
+
+    def __init__(self, max_episode_steps: int = 500):
+        super().__init__()
+        hi = np.array([1.0, 1.0, 1.0, 1.0, self.MAX_VEL1, self.MAX_VEL2], dtype=np.float32)
+        self.observation_space = Box(-hi, hi)
+        self.action_space = Discrete(3)
+        self._max_steps = int(max_episode_steps)
+        self._s = np.zeros(4)
+        self._t = 0
+
+    def _obs(self):
+        t1, t2, d1, d2 = self._s
+        return np.array([math.cos(t1), math.sin(t1), math.cos(t2), math.sin(t2), d1, d2], dtype=np.float32)
+
+    def reset(self, *, seed: Optional[int] = None, options=None):
+        self._seed(seed)
+        self._s = self._rng.uniform(-0.1, 0.1, size=4)
+        self._t = 0
+        return self._obs(), {}
+
+    def _dsdt(self, s, tau):
+        m1, m2, l1, lc1, lc2, i1, i2, g = self.M1, self.M2, self.L1, self.LC1, self.LC2, self.I1, self.I2, self.G
+        t1, t2, dt1, dt2 = s
+        d1 = m1 * lc1**2 + m2 * (l1**2 + lc2**2 + 2 * l1 * lc2 * math.cos(t2)) + i1 + i2
+        d2 = m2 * (lc2**2 + l1 * lc2 * math.cos(t2)) + i2
+        phi2 = m2 * lc2 * g * math.cos(t1 + t2 - math.pi / 2.0)
+        phi1 = (
+            -m2 * l1 * lc2 * dt2**2 * math.sin(t2)
+            - 2 * m2 * l1 * lc2 * dt2 * dt1 * math.sin(t2)
+            + (m1 * lc1 + m2 * l1) * g * math.cos(t1 - math.pi / 2.0)
+            + phi2
+        )
+        # the "book" formulation (Sutton & Barto): solve the second joint
+        # acceleration first, then back-substitute
+        ddt2 = (tau + d2 / d1 * phi1 - m2 * l1 * lc2 * dt1**2 * math.sin(t2) - phi2) / (
+            m2 * lc2**2 + i2 - d2**2 / d1
+        )
+        ddt1 = -(d2 * ddt2 + phi1) / d1
+        return np.array([dt1, dt2, ddt1, ddt2])
+
+    @staticmethod
+    def _wrap(x, lo, hi):
+        rng = hi - lo
+        while x > hi:
+            x -= rng
+        while x < lo:
+            x += rng
+        return x
+
+    def step(self, action):
+        tau = float(int(action) - 1)
+        s = self._s
+        # one RK4 step of the autonomous dynamics with constant torque
+        k1 = self._dsdt(s, tau)
+        k2 = self._dsdt(s + 0.5 * self.DT * k1, tau)
+        k3 = self._dsdt(s + 0.5 * self.DT * k2, tau)
+        k4 = self._dsdt(s + self.DT * k3, tau)
+        ns = s + self.DT / 6.0 * (k1 + 2 * k2 + 2 * k3 + k4)
+        ns[0] = self._wrap(ns[0], -math.pi, math.pi)
+        ns[1] = self._wrap(ns[1], -math.pi, math.pi)
+        ns[2] = float(np.clip(ns[2], -self.MAX_VEL1, self.MAX_VEL1))
+        ns[3] = float(np.clip(ns[3], -self.MAX_VEL2, self.MAX_VEL2))
+        self._s = ns
+        self._t += 1
+        terminated = bool(-math.cos(ns[0]) - math.cos(ns[1] + ns[0]) > 1.0)
+        truncated = bool(self._t >= self._max_steps)
+        return self._obs(), -1.0, terminated, truncated, {}
+
+
 # ---------------------------------------------------------------------------
 # registry + make
 # ---------------------------------------------------------------------------
@@ -233,6 +417,9 @@ class PendulumEnv(_EnvBase):
 _REGISTRY = {
     "CartPole-v1": lambda **kw: CartPoleEnv(**kw),
     "Pendulum-v1": lambda **kw: PendulumEnv(**kw),
+    "MountainCar-v0": lambda **kw: MountainCarEnv(**kw),
+    "MountainCarContinuous-v0": lambda **kw: MountainCarContinuousEnv(**kw),
+    "Acrobot-v1": lambda **kw: AcrobotEnv(**kw),
 }
 
 

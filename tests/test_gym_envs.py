@@ -234,3 +234,74 @@ class TestVecEnvAdapter:
         ensure_space_types(PendulumEnv())
         with pytest.raises(TypeError):
             ensure_space_types(CartPoleEnv())  # Discrete action space
+
+
+class TestMoreClassicControl:
+    def test_mountaincar_api_and_goal(self):
+        from evotorch_amd.neuroevolution.gym_compat import MountainCarEnv
+
+        env = MountainCarEnv()
+        obs, _ = env.reset(seed=3)
+        assert obs.shape == (2,) and -0.6 <= obs[0] <= -0.4 and obs[1] == 0.0
+        for _ in range(50):
+            obs, r, term, trunc, _ = env.step(2)
+            assert r == -1.0
+            assert env.observation_space.contains(obs)
+        # drive to the goal by teleporting near it with speed
+        env._pos, env._vel = 0.49, 0.07
+        obs, r, term, trunc, _ = env.step(2)
+        assert term
+
+    def test_mountaincar_continuous_reward_and_goal(self):
+        from evotorch_amd.neuroevolution.gym_compat import MountainCarContinuousEnv
+
+        env = MountainCarContinuousEnv()
+        env.reset(seed=4)
+        _, r, term, _, _ = env.step([0.5])
+        assert not term and abs(r - (-0.1 * 0.25)) < 1e-9
+        env._pos, env._vel = 0.449, 0.07
+        _, r, term, _, _ = env.step([1.0])
+        assert term and r > 99.0
+
+    def test_acrobot_dynamics_and_termination(self):
+        import math
+
+        from evotorch_amd.neuroevolution.gym_compat import AcrobotEnv
+
+        env = AcrobotEnv()
+        obs, _ = env.reset(seed=5)
+        assert obs.shape == (6,)
+        # cos/sin pairs are unit-norm
+        assert abs(obs[0] ** 2 + obs[1] ** 2 - 1.0) < 1e-6
+        total = 0.0
+        for _ in range(30):
+            obs, r, term, trunc, _ = env.step(0)
+            total += r
+            assert env.observation_space.contains(obs)
+            assert not term  # hanging near the bottom cannot terminate this fast
+        assert total == -30.0
+        # a state with the tip high above the bar terminates
+        env._s = np.array([math.pi, 0.0, 0.0, 0.0])
+        _, _, term, _, _ = env.step(1)
+        assert term
+
+    def test_vectorized_and_make(self):
+        from evotorch_amd.neuroevolution import gym_compat
+
+        for name, act in (("MountainCar-v0", 1), ("Acrobot-v1", 2)):
+            env = gym_compat.make_vec(name, num_envs=3)
+            obs, _ = env.reset(seed=7)
+            assert obs.shape[0] == 3
+            for _ in range(5):
+                obs, r, term, trunc, info = env.step(np.full(3, act, dtype=np.int64))
+            assert obs.shape[0] == 3
+
+    def test_gymne_on_acrobot(self):
+        from evotorch_amd.algorithms import PGPE
+        from evotorch_amd.neuroevolution import GymNE
+
+        prob = GymNE("Acrobot-v1", "Linear(obs_length, act_length)", num_episodes=1,
+                     episode_length=40, seed=11)
+        s = PGPE(prob, popsize=8, center_learning_rate=0.2, stdev_learning_rate=0.1, radius_init=0.5)
+        s.run(2)
+        assert s.status["iter"] == 2
