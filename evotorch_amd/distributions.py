@@ -165,6 +165,21 @@ class Distribution(TensorMakerMixin):
             return float(learning_rates[param_name]) * grad
         raise ValueError(f"No learning rate or optimizer given for parameter {param_name!r}")
 
+    def _stepped_param(
+        self, param_name: str, current: torch.Tensor, grad: torch.Tensor, learning_rates: Optional[dict], optimizers: Optional[dict]
+    ) -> torch.Tensor:
+        """current + step WITHOUT materializing the step tensor when a
+        plain learning rate drives the parameter (one fused add), and
+        without the optimizer's defensive clone (the sum is the only
+        consumer). At extreme L every L-sized temporary counts — the
+        billion-parameter streaming mode is bounded by how many of these
+        are alive at the update's peak."""
+        if optimizers is not None and param_name in optimizers and optimizers[param_name] is not None:
+            return current + optimizers[param_name].ascent(grad, cloned_result=False)
+        if learning_rates is not None and param_name in learning_rates:
+            return torch.add(current, grad, alpha=float(learning_rates[param_name]))
+        raise ValueError(f"No learning rate or optimizer given for parameter {param_name!r}")
+
     def update_parameters(self, gradients: dict, *, learning_rates: Optional[dict] = None, optimizers: Optional[dict] = None) -> "Distribution":
         raise NotImplementedError
 
@@ -353,8 +368,8 @@ class SeparableGaussian(Distribution):
         }
 
     def update_parameters(self, gradients: dict, *, learning_rates: Optional[dict] = None, optimizers: Optional[dict] = None) -> "SeparableGaussian":
-        new_mu = self.mu + self._follow_gradient("mu", gradients["mu"], learning_rates=learning_rates, optimizers=optimizers)
-        new_sigma = self.sigma + self._follow_gradient("sigma", gradients["sigma"], learning_rates=learning_rates, optimizers=optimizers)
+        new_mu = self._stepped_param("mu", self.mu, gradients["mu"], learning_rates, optimizers)
+        new_sigma = self._stepped_param("sigma", self.sigma, gradients["sigma"], learning_rates, optimizers)
         return self.modified_copy(mu=new_mu, sigma=new_sigma)
 
     def relative_entropy(self, other: "SeparableGaussian") -> float:
@@ -447,8 +462,11 @@ class ExpSeparableGaussian(SeparableGaussian):
         return {"mu": sums["mu"].to(self.mu.dtype), "sigma": sums["sigma"].to(self.sigma.dtype)}
 
     def update_parameters(self, gradients: dict, *, learning_rates: Optional[dict] = None, optimizers: Optional[dict] = None) -> "ExpSeparableGaussian":
-        new_mu = self.mu + self._follow_gradient("mu", gradients["mu"], learning_rates=learning_rates, optimizers=optimizers)
-        new_sigma = self.sigma * torch.exp(0.5 * self._follow_gradient("sigma", gradients["sigma"], learning_rates=learning_rates, optimizers=optimizers))
+        new_mu = self._stepped_param("mu", self.mu, gradients["mu"], learning_rates, optimizers)
+        # the exponential sigma step owns a fresh tensor (it never aliases
+        # optimizer state here), so transform it in place
+        step = self._follow_gradient("sigma", gradients["sigma"], learning_rates=learning_rates, optimizers=optimizers)
+        new_sigma = self.sigma * step.mul_(0.5).exp_()
         return self.modified_copy(mu=new_mu, sigma=new_sigma)
 
 

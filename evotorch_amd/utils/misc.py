@@ -356,15 +356,25 @@ def modify_tensor(
     lo = scalar_or_tensor(lb)
     hi = scalar_or_tensor(ub)
     if max_change is not None:
+        # two L-sized temporaries total (lo_t doubles as the result) — at
+        # billion-parameter L this clamp runs inside the update's memory
+        # peak, so every temporary counts
         mc = max_change if isinstance(max_change, torch.Tensor) else float(max_change)
-        allowed = original.abs() * mc
+        allowed = original.abs().mul_(mc)
         lo_t = original - allowed
-        hi_t = original + allowed
+        hi_t = allowed.add_(original)  # `allowed` is dead after this line
         if lo is not None:
-            lo_t = torch.max(lo_t, lo) if isinstance(lo, torch.Tensor) else torch.clamp(lo_t, min=lo)
+            if isinstance(lo, torch.Tensor):
+                torch.max(lo_t, lo, out=lo_t)
+            else:
+                lo_t.clamp_(min=lo)
         if hi is not None:
-            hi_t = torch.min(hi_t, hi) if isinstance(hi, torch.Tensor) else torch.clamp(hi_t, max=hi)
-        result = torch.min(torch.max(target, lo_t), hi_t)
+            if isinstance(hi, torch.Tensor):
+                torch.min(hi_t, hi, out=hi_t)
+            else:
+                hi_t.clamp_(max=hi)
+        torch.max(target, lo_t, out=lo_t)
+        result = torch.min(lo_t, hi_t, out=lo_t)
     else:
         # torch.clamp applies max(min_val) before min(max_val) — same
         # upper-bound-wins semantics as the reference
