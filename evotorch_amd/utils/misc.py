@@ -375,10 +375,15 @@ def modify_tensor(
                 hi_t.clamp_(max=hi)
         torch.max(target, lo_t, out=lo_t)
         result = torch.min(lo_t, hi_t, out=lo_t)
-    else:
+    elif isinstance(lo, torch.Tensor) == isinstance(hi, torch.Tensor) or lo is None or hi is None:
         # torch.clamp applies max(min_val) before min(max_val) — same
         # upper-bound-wins semantics as the reference
         result = torch.clamp(target, min=lo, max=hi)
+    else:
+        # mixed tensor/scalar bounds: clamp() rejects the combination, so
+        # apply max-then-min explicitly (same upper-bound-wins order)
+        result = torch.max(target, lo) if isinstance(lo, torch.Tensor) else torch.clamp(target, min=lo)
+        result = torch.min(result, hi) if isinstance(hi, torch.Tensor) else result.clamp_(max=hi)
     if in_place:
         original[:] = result
         return original

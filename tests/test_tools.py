@@ -267,3 +267,50 @@ def test_graft_entry_contract():
     spec.loader.exec_module(mod)
     assert callable(mod.build)
     assert callable(mod.smoke)
+
+
+def test_modify_tensor_matches_naive_reference():
+    """The temporary-slimmed (out=/in-place) modify_tensor against a naive
+    expression of the same semantics, across bound type combinations."""
+    from evotorch_amd.utils import modify_tensor
+
+    def naive(original, target, lb, ub, max_change):
+        lo_t, hi_t = None, None
+        if max_change is not None:
+            allowed = original.abs() * max_change
+            lo_t = original - allowed
+            hi_t = original + allowed
+        if lb is not None:
+            lb_t = lb if isinstance(lb, torch.Tensor) else torch.full_like(original, float(lb))
+            lo_t = lb_t if lo_t is None else torch.max(lo_t, lb_t)
+        if ub is not None:
+            ub_t = ub if isinstance(ub, torch.Tensor) else torch.full_like(original, float(ub))
+            hi_t = ub_t if hi_t is None else torch.min(hi_t, ub_t)
+        result = target
+        if lo_t is not None:
+            result = torch.max(result, lo_t)
+        if hi_t is not None:
+            result = torch.min(result, hi_t)  # upper bound wins, as in the reference
+        return result
+
+    g = torch.Generator().manual_seed(123)
+    for trial in range(40):
+        n = int(torch.randint(1, 50, (1,), generator=g))
+        original = torch.randn(n, generator=g)
+        target = original + torch.randn(n, generator=g)
+        lb = [None, -0.5, torch.randn(n, generator=g) - 1.0][trial % 3]
+        ub = [None, 0.5, torch.randn(n, generator=g) + 1.0][(trial // 3) % 3]
+        mc = [None, 0.25, torch.rand(n, generator=g) * 0.5][(trial // 9) % 3]
+        if lb is None and ub is None and mc is None:
+            continue
+        orig_copy = original.clone()
+        tgt_copy = target.clone()
+        got = modify_tensor(original, target, lb=lb, ub=ub, max_change=mc)
+        want = naive(orig_copy, tgt_copy, lb, ub, mc)
+        torch.testing.assert_close(got, want, rtol=0, atol=0)
+        assert torch.equal(original, orig_copy), "modify_tensor must not mutate original"
+        assert torch.equal(target, tgt_copy), "modify_tensor must not mutate target"
+        # in_place=True writes the same values into original
+        got_ip = modify_tensor(original, target, lb=lb, ub=ub, max_change=mc, in_place=True)
+        assert got_ip is original and torch.equal(original, want)
+        original.copy_(orig_copy)
