@@ -19,6 +19,7 @@ evaluated"; basic slicing returns shared-memory views.
 
 import logging
 import math
+import os
 from typing import Any, Callable, Iterable, List, Optional, Union
 
 import numpy as np
@@ -755,22 +756,28 @@ class Problem(TensorMakerMixin, Serializable):
         num_subbatches=None,
         subbatch_size=None,
     ):
-        # Ray-era knobs accepted for drop-in migration from the reference;
-        # parallelism here comes from the torchrun/RCCL topology instead
-        # (docs/migrating_from_evotorch.md), so they are ignored with a
-        # warning rather than a TypeError.
-        _ray_knobs = {"num_actors": num_actors, "actor_config": actor_config,
-                      "num_gpus_per_actor": num_gpus_per_actor,
-                      "num_subbatches": num_subbatches, "subbatch_size": subbatch_size}
-        _given = [k for k, v in _ray_knobs.items() if v is not None]
+        # num_actors / num_subbatches / subbatch_size are honored: they
+        # drive a multiprocessing evaluation pool (parallel/evalpool.py —
+        # the reference's Ray actor system on the standard library) for
+        # CPU-side per-solution fitness. GPU/actor-placement knobs are
+        # meaningless here (GPU scaling is torchrun + RCCL,
+        # docs/parallelism.md) and are ignored with a warning.
+        _gpu_knobs = {"actor_config": actor_config, "num_gpus_per_actor": num_gpus_per_actor}
+        _given = [k for k, v in _gpu_knobs.items() if v is not None]
         if _given:
             import warnings
 
             warnings.warn(
-                f"Ignoring Ray-era argument(s) {_given}: this framework parallelizes via "
+                f"Ignoring Ray-era argument(s) {_given}: GPU parallelism here comes from "
                 "torchrun + RCCL (problem.use_comm(init_comm())); see docs/migrating_from_evotorch.md",
                 stacklevel=2,
             )
+        if num_actors in ("max", "num_cpus"):
+            num_actors = os.cpu_count() or 1
+        self._num_actors = None if num_actors is None else max(0, int(num_actors))
+        self._num_subbatches = None if num_subbatches is None else int(num_subbatches)
+        self._subbatch_size = None if subbatch_size is None else int(subbatch_size)
+        self._eval_pool = None
         self._senses = _normalize_sense(objective_sense)
         self._objective_func = objective_func
         self._vectorized = bool(getattr(objective_func, "__evotorch_vectorized__", False)) if vectorized is None else bool(vectorized)
@@ -1225,6 +1232,8 @@ class Problem(TensorMakerMixin, Serializable):
         comm = self._comm
         if comm is not None and comm.world_size > 1 and len(batch) >= comm.world_size:
             self._evaluate_sharded(batch, comm)
+        elif self._use_eval_pool(batch):
+            self._eval_pool.evaluate_into(self, batch)
         else:
             self._evaluate_batch(batch)
         if self._store_solution_stats:
@@ -1236,6 +1245,27 @@ class Problem(TensorMakerMixin, Serializable):
         my_piece = pieces[comm.rank]
         self._evaluate_batch(my_piece)
         comm.all_gather_rows(batch.unsafe_evals, [pieces.indices_of(i) for i in range(len(pieces))])
+
+    # -- process-pool evaluation (the reference's Ray actor system) -----------
+
+    def _use_eval_pool(self, batch: SolutionBatch) -> bool:
+        if not self._num_actors or self._num_actors < 2 or len(batch) < 2:
+            return False
+        if self._device.type != "cpu":
+            return False  # GPU problems scale via torchrun + RCCL instead
+        if self._eval_pool is None:
+            from .parallel.evalpool import EvalPool
+
+            self._eval_pool = EvalPool(
+                self, self._num_actors, num_subbatches=self._num_subbatches, subbatch_size=self._subbatch_size
+            )
+        return True
+
+    def kill_actors(self) -> None:
+        """Shut down the evaluation worker pool (reference core.py:2117)."""
+        pool, self._eval_pool = self._eval_pool, None
+        if pool is not None:
+            pool.close()
 
     def _update_solution_stats(self, batch: SolutionBatch):
         nobj = len(self._senses)
@@ -1691,7 +1721,7 @@ class Problem(TensorMakerMixin, Serializable):
     def _get_cloned_state(self, *, memo: dict) -> dict:
         state = {}
         for k, v in self.__dict__.items():
-            if k in ("_generator", "_comm", "_grad_batch_cache", "_pregen", "_pregen_stream", "_graph_state"):
+            if k in ("_generator", "_comm", "_grad_batch_cache", "_pregen", "_pregen_stream", "_graph_state", "_eval_pool"):
                 state[k] = None
             else:
                 state[k] = deep_clone(v, otherwise_deepcopy=True, memo=memo)

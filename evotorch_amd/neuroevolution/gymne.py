@@ -151,10 +151,6 @@ class GymNE(NEProblem):
         num_subbatches=None,
         subbatch_size=None,
     ):
-        if any(v is not None for v in (num_actors, actor_config, num_subbatches, subbatch_size)):
-            import warnings
-
-            warnings.warn("Ignoring Ray-era actor argument(s): parallelize via torchrun + RCCL", stacklevel=2)
         if env is None:
             env = env_name
         if env is None:
@@ -173,11 +169,13 @@ class GymNE(NEProblem):
         self._alive_bonus_schedule = alive_bonus_schedule
         self._action_noise_stdev = action_noise_stdev
         self._obs_norm_enabled = bool(observation_normalization)
-        super().__init__("max", network, network_args=network_args, initial_bounds=initial_bounds, seed=seed, store_solution_stats=False)
+        super().__init__("max", network, network_args=network_args, initial_bounds=initial_bounds, seed=seed, store_solution_stats=False,
+                         num_actors=num_actors, actor_config=actor_config, num_subbatches=num_subbatches, subbatch_size=subbatch_size)
         self._obs_norm = RunningNorm(shape=self._obs_dim, device="cpu")
         self.last_eval_interaction_count = 0
         self._total_interactions = 0
         self._episode_count = 0
+        self._pending_stats: Optional[RunningNorm] = None
         self.after_eval_hook.append(
             lambda b: {"total_interaction_count": self._total_interactions, "total_episode_count": self._episode_count}
         )
@@ -202,6 +200,29 @@ class GymNE(NEProblem):
     def observation_normalization_data(self):
         return {"mean": self._obs_norm.mean, "stdev": self._obs_norm.stdev, "count": self._obs_norm.count}
 
+    # -- actor sync protocol (reference gymne.py / core.py:2239-2340) ---------
+
+    def get_observation_stats(self) -> RunningNorm:
+        return self._obs_norm
+
+    def set_observation_stats(self, rn) -> None:
+        """Replace this problem's stats (main → worker sync)."""
+        self._obs_norm.reset()
+        self._obs_norm.update(rn if not isinstance(rn, RunningNorm) else rn.stats_triple())
+
+    def update_observation_stats(self, rn) -> RunningNorm:
+        """Merge another RunningNorm (or (count, sum, sumsq) triple) into
+        this problem's stats (worker → main sync)."""
+        self._obs_norm.update(rn if not isinstance(rn, RunningNorm) else rn.stats_triple())
+        return self._obs_norm
+
+    def pop_observation_stats(self) -> Optional[RunningNorm]:
+        """Take (and clear) the stats collected since the last pop — what a
+        worker ships back to the main process."""
+        pending = self._pending_stats
+        self._pending_stats = None
+        return pending
+
     def _alive_bonus(self, t: int) -> float:
         if self._alive_bonus_schedule is None:
             return 0.0
@@ -224,6 +245,9 @@ class GymNE(NEProblem):
         while True:
             obs_t = torch.as_tensor(np.asarray(obs), dtype=torch.float32)
             if self._obs_norm_enabled:
+                if self._pending_stats is None:
+                    self._pending_stats = RunningNorm(shape=self._obs_dim, device="cpu")
+                self._pending_stats.update(obs_t)  # delta since last pop (actor sync protocol)
                 obs_in = self._obs_norm.update_and_normalize(obs_t)
             else:
                 obs_in = obs_t

@@ -40,12 +40,6 @@ class NEProblem(Problem):
         subbatch_size=None,
         store_solution_stats: Optional[bool] = None,
     ):
-        if any(v is not None for v in (num_actors, actor_config, num_subbatches, subbatch_size)):
-            import warnings
-
-            warnings.warn(
-                "Ignoring Ray-era actor argument(s): parallelize via torchrun + RCCL "
-                "(docs/migrating_from_evotorch.md)", stacklevel=2)
         self._network_def = network
         self._network_args = dict(network_args or {})
         self._network_eval_func = network_eval_func
@@ -63,6 +57,11 @@ class NEProblem(Problem):
             eval_data_length=eval_data_length,
             seed=seed,
             store_solution_stats=store_solution_stats,
+            num_actors=num_actors,
+            actor_config=actor_config,
+            num_gpus_per_actor=num_gpus_per_actor,
+            num_subbatches=num_subbatches,
+            subbatch_size=subbatch_size,
         )
         self._instantiated_net = net.to(self.network_device)
 

@@ -46,10 +46,9 @@ class SupervisedNE(NEProblem):
         num_gpus_per_actor=None,
         actor_config=None,
     ):
-        if any(v is not None for v in (num_actors, num_gpus_per_actor, actor_config)):
-            import warnings
-
-            warnings.warn("Ignoring Ray-era actor argument(s): parallelize via torchrun + RCCL", stacklevel=2)
+        # num_subbatches/subbatch_size keep their SupervisedNE meaning
+        # (solutions per common minibatch) and are consumed here, not by
+        # the evaluation pool.
         super().__init__(
             "min" if loss_as_fitness_sign == "min" else "max",
             network,
@@ -57,6 +56,9 @@ class SupervisedNE(NEProblem):
             initial_bounds=initial_bounds,
             device=device,
             seed=seed,
+            num_actors=num_actors,
+            num_gpus_per_actor=num_gpus_per_actor,
+            actor_config=actor_config,
         )
         self._dataset = dataset
         self._loss_func = loss_func

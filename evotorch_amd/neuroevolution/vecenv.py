@@ -145,10 +145,6 @@ class VecEnvNE(NEProblem):
         subbatch_size=None,
         actor_config=None,
     ):
-        if any(v is not None for v in (num_actors, num_gpus_per_actor, num_subbatches, subbatch_size, actor_config)):
-            import warnings
-
-            warnings.warn("Ignoring Ray-era actor argument(s): parallelize via torchrun + RCCL", stacklevel=2)
         env_config = dict(env_config or {})
         if isinstance(env, str):
             env_name = env
@@ -186,6 +182,11 @@ class VecEnvNE(NEProblem):
             device=device,
             seed=seed,
             store_solution_stats=False,
+            num_actors=num_actors,
+            num_gpus_per_actor=num_gpus_per_actor,
+            actor_config=actor_config,
+            num_subbatches=num_subbatches,
+            subbatch_size=subbatch_size,
         )
         self._obs_norm = RunningNorm(shape=self._obs_dim, device=self.network_device)
         self._use_hip_graph = bool(use_hip_graph)

@@ -1,0 +1,72 @@
+"""Process-pool evaluation (`num_actors`): the reference's Ray actor
+system rebuilt on multiprocessing + cloudpickle (parallel/evalpool.py)."""
+
+import torch
+
+from evotorch_amd import Problem
+
+
+def test_pool_matches_serial_and_supports_lambdas():
+    def make(num_actors=None):
+        return Problem("min", lambda x: float((x - 0.25).pow(2).sum()),
+                       solution_length=6, initial_bounds=(-1, 1), seed=11, num_actors=num_actors)
+
+    serial = make()
+    pooled = make(num_actors=2)
+    vals = torch.randn(torch.Generator().manual_seed(3) and 9, 6)
+    torch.manual_seed(3)
+    vals = torch.randn(9, 6)
+
+    def evaluate(prob):
+        b = prob.generate_batch(9)
+        b.access_values()[:] = vals
+        prob.evaluate(b)
+        return torch.Tensor.as_subclass(b.evals[:, 0], torch.Tensor).clone()
+
+    try:
+        e_serial = evaluate(serial)
+        e_pooled = evaluate(pooled)
+        torch.testing.assert_close(e_pooled, e_serial)
+        assert pooled._eval_pool is not None
+        # subsequent evaluations reuse the same pool
+        pool = pooled._eval_pool
+        evaluate(pooled)
+        assert pooled._eval_pool is pool
+    finally:
+        pooled.kill_actors()
+    assert pooled._eval_pool is None
+
+
+def test_pool_subbatch_size_and_vectorized_fitness():
+    from evotorch_amd.decorators import vectorized
+
+    @vectorized
+    def f(x):
+        return (x**2).sum(-1)
+
+    prob = Problem("min", f, solution_length=4, initial_bounds=(-1, 1), seed=7,
+                   num_actors=2, subbatch_size=3)
+    try:
+        b = prob.generate_batch(10)
+        prob.evaluate(b)
+        vals = torch.Tensor.as_subclass(b.access_values(keep_evals=True), torch.Tensor)
+        expected = (vals**2).sum(-1)
+        torch.testing.assert_close(torch.Tensor.as_subclass(b.evals[:, 0], torch.Tensor), expected)
+    finally:
+        prob.kill_actors()
+
+
+def test_pool_gymne_merges_counters_and_obs_stats():
+    from evotorch_amd.algorithms import PGPE
+    from evotorch_amd.neuroevolution import GymNE
+
+    prob = GymNE("CartPole-v1", "Linear(obs_length, act_length)", num_episodes=1,
+                 episode_length=25, seed=5, num_actors=2, observation_normalization=True)
+    try:
+        searcher = PGPE(prob, popsize=8, center_learning_rate=0.2, stdev_learning_rate=0.1, radius_init=0.5)
+        searcher.run(2)
+        assert searcher.status["iter"] == 2
+        assert prob._total_interactions > 0
+        assert prob.obs_norm.has_data  # worker stats merged back
+    finally:
+        prob.kill_actors()
