@@ -12,6 +12,7 @@ REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 CASES = [
     ("examples/rastrigin_snes.py", ["--generations", "3"]),
     ("examples/cartpole_gymne.py", ["--generations", "3"]),
+    ("examples/parallel_cpu_actors.py", ["--generations", "2", "--popsize", "8"]),
     ("examples/multiobjective_nsga2.py", []),
     ("examples/mapelites_illumination.py", []),
     ("examples/functional_api_batched.py", []),
