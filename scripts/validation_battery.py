@@ -13,7 +13,7 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import torch
 
 from evotorch_amd import Problem
-from evotorch_amd.algorithms import CMAES, PGPE, SNES, GeneticAlgorithm, GraphedSearch
+from evotorch_amd.algorithms import CEM, CMAES, PGPE, SNES, XNES, Cosyne, GeneticAlgorithm, GraphedSearch, MAPElites
 from evotorch_amd.decorators import vectorized
 from evotorch_amd.neuroevolution import SyntheticRolloutProblem
 from evotorch_amd.operators import GaussianMutation, PolynomialMutation, SimulatedBinaryCrossOver
@@ -104,6 +104,47 @@ def main():
     frac = float((ranks == 0).float().mean())
     report("NSGA-II ZDT1 d=12 pop=512 (HIP pareto sort)", 100, dt, frac)
     print(f"    (fraction of population on the non-dominated front: {frac:.2f})")
+
+    # 6. XNES full covariance d=256 (rocBLAS exp-map updates)
+    prob = Problem("min", sphere, solution_length=256, initial_bounds=(-3, 3), device=DEVICE, seed=6)
+    s = XNES(prob, stdev_init=1.0, popsize=64)
+    t0 = time.perf_counter()
+    s.run(400)
+    report("XNES sphere d=256 pop=64", 400, time.perf_counter() - t0, s.status["pop_best_eval"])
+
+    # 7. CEM d=1000
+    prob = Problem("min", sphere, solution_length=1000, initial_bounds=(-3, 3), device=DEVICE, seed=7)
+    s = CEM(prob, stdev_init=2.0, popsize=500, parenthood_ratio=0.2)
+    t0 = time.perf_counter()
+    s.run(300)
+    report("CEM sphere d=1000 pop=500", 300, time.perf_counter() - t0, s.status["pop_best_eval"])
+
+    # 8. Cosyne Rastrigin d=30
+    prob = Problem("min", rastrigin, solution_length=30, initial_bounds=(-5.12, 5.12), device=DEVICE, seed=8)
+    s = Cosyne(prob, popsize=128, tournament_size=4, mutation_stdev=0.3)
+    t0 = time.perf_counter()
+    s.run(300)
+    report("Cosyne Rastrigin d=30 pop=128", 300, time.perf_counter() - t0, s.status["pop_best_eval"])
+
+    # 9. MAPElites on Rastrigin with 2 descriptor dims (K9 assignment kernel)
+    from evotorch_amd.algorithms import make_feature_grid
+
+    @vectorized
+    def rastrigin_with_feats(x):
+        f = 10 * x.shape[-1] + (x**2 - 10 * torch.cos(2 * math.pi * x)).sum(-1)
+        return torch.stack([f, x[:, 0], x[:, 1]], dim=-1)
+
+    prob = Problem("min", rastrigin_with_feats, solution_length=16, initial_bounds=(-5.12, 5.12),
+                   device=DEVICE, seed=9, eval_data_length=2)
+    grid = make_feature_grid(lower_bounds=[-5.12, -5.12], upper_bounds=[5.12, 5.12], num_bins=20, device=DEVICE)
+    s = MAPElites(prob, feature_grid=grid, re_evaluate=False,
+                  operators=[GaussianMutation(prob, stdev=0.3)])
+    t0 = time.perf_counter()
+    s.run(200)
+    dt = time.perf_counter() - t0
+    filled = float(s.filled.float().mean())
+    report("MAPElites Rastrigin 20x20 grid", 200, dt, filled)
+    print(f"    (fraction of grid cells filled: {filled:.2f})")
 
 
 if __name__ == "__main__":
