@@ -37,3 +37,25 @@ print(f"graphed VecEnvNE: 300 gens in {el:.1f}s ({el/300*1000:.1f} ms/gen), mean
 assert last > first + 1.0
 print(f"mem allocated {torch.cuda.memory_allocated()/2**20:.0f} MiB")
 print("soak ok")
+
+# extended flagship soak: 10k generations of the v7 SPMD path at T=200
+from evotorch_amd.neuroevolution import SyntheticRolloutProblem
+from evotorch_amd.parallel import init_comm
+prob = SyntheticRolloutProblem(device="cuda:0", seed=2, episode_length=200)
+comm = init_comm()
+prob.use_comm(comm)
+r = 2.25
+s2 = PGPE(prob, popsize=4000, radius_init=r, center_learning_rate=0.75*r/15, stdev_learning_rate=0.1,
+          optimizer="clipup", optimizer_config={"max_speed": r/15}, distributed=True)
+for _ in range(50):
+    s2.step()
+torch.cuda.synchronize()
+t0 = time.perf_counter()
+for _ in range(10000):
+    s2.step()
+torch.cuda.synchronize()
+el = time.perf_counter() - t0
+me = float(s2.status["mean_eval"])
+print(f"flagship 10k-gen soak: {el:.1f}s ({10000/el:.0f} gens/s, {4000*10000/el/1e6:.2f}M sol/s), mean_eval {me:.1f}")
+assert me > 400.0 and el == el
+print("extended soak ok")
