@@ -58,6 +58,14 @@ class GraphedSearch:
         problem = searcher.problem
         if problem.device.type != "cuda":
             raise ValueError("GraphedSearch needs a ROCm device problem")
+        comm = getattr(problem, "_comm", None)
+        if comm is not None and comm.world_size > 1:
+            raise ValueError("GraphedSearch cannot capture collectives; detach the Comm (world 1) first")
+        # problems with host-derived per-eval randomness switch it to a
+        # device seed chain so episode seeds vary across replays
+        enable_graph_mode = getattr(problem, "enable_graph_mode", None)
+        if enable_graph_mode is not None:
+            enable_graph_mode()
         self._searcher = searcher
         self._problem = problem
         self._dist = dist
@@ -98,6 +106,9 @@ class GraphedSearch:
 
         self._C.sample_gaussian_graphsafe(values, mu, sigma, dist._symmetric, self._seed_buf)
         problem._evaluate_batch(pop)
+        merge_stats = getattr(problem, "_merge_pending_stats", None)
+        if merge_stats is not None:
+            merge_stats()  # world-1: in-place obs-norm accumulation, capture-safe
         fitnesses = pop.unsafe_evals[:, searcher._obj_index]
         self._mean_eval_buf.copy_(fitnesses.mean())
         sense = problem.senses[searcher._obj_index]

@@ -77,13 +77,28 @@ class SyntheticRolloutProblem(Problem):
 
     # -- evaluation ----------------------------------------------------------
 
+    def enable_graph_mode(self) -> None:
+        """Switch episode seeding to a DEVICE splitmix64 chain so that
+        `_evaluate_batch` is hipGraph-capturable (a host-derived seed
+        would be frozen into the capture and every replay would rerun the
+        same episodes). Called by GraphedSearch; the chain starts from
+        the problem's own generator."""
+        if self._device.type != "cuda":
+            return
+        if getattr(self, "_graph_seed_buf", None) is None:
+            from ..ops.dispatch import _seed_from_generator
+
+            seed0 = _seed_from_generator(self._generator, self._device)
+            self._graph_seed_buf = torch.tensor([seed0], dtype=torch.int64, device=self._device)
+
     def _evaluate_batch(self, batch: SolutionBatch):
         values = batch.access_values(keep_evals=True)
         n = len(batch)
         spec = self._spec
         from ..ops.dispatch import _seed_from_generator
 
-        init_seed = _seed_from_generator(self._generator, self._device) & 0x7FFFFFFF
+        seed_buf = getattr(self, "_graph_seed_buf", None)
+        init_seed = 0 if seed_buf is not None else _seed_from_generator(self._generator, self._device) & 0x7FFFFFFF
         member_offset = 0
         comm = self._comm
         if comm is not None and comm.world_size > 1:
@@ -97,6 +112,8 @@ class SyntheticRolloutProblem(Problem):
             mean, std = self._norm_mean_std()
             blob = spec.env_blob(mean, std, device=values.device)
             obs_stats = torch.zeros(2 * spec.obs_dim, dtype=torch.float32, device=values.device)
+            if seed_buf is not None:
+                mod.bump_seed(seed_buf)  # fresh episode seed per replay, on device
             fitness = mod.rollout_linear(
                 values.contiguous(),
                 blob,
@@ -110,6 +127,7 @@ class SyntheticRolloutProblem(Problem):
                 init_seed,
                 member_offset,
                 spec.policy_hidden,
+                seed_buf,
             )
             triple = (float(n) * spec.episode_length, obs_stats[: spec.obs_dim], obs_stats[spec.obs_dim :])
         else:

@@ -17,7 +17,9 @@ void adam_step_graphsafe(torch::Tensor step_out, torch::Tensor grad, torch::Tens
                          torch::Tensor t_buf, double stepsize, double beta1, double beta2, double epsilon);
 torch::Tensor rollout_linear(torch::Tensor params, torch::Tensor env_blob, torch::Tensor obs_stats_out,
                              int64_t obs_dim, int64_t act_dim, int64_t rank, int64_t steps, double alive_bonus,
-                             double act_cost, int64_t init_seed, int64_t member_offset, int64_t policy_hidden);
+                             double act_cost, int64_t init_seed, int64_t member_offset, int64_t policy_hidden,
+                             c10::optional<torch::Tensor> seed_buf);
+void bump_seed(torch::Tensor seed_buf);
 void cma_update_c(torch::Tensor C, torch::Tensor Y, torch::Tensor w, torch::Tensor pc, torch::Tensor hs_f,
                   torch::Tensor wsum, double c1, double cmu, double cc);
 void potrf_tile(torch::Tensor A, torch::Tensor info);
@@ -47,7 +49,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           pybind11::arg("params"), pybind11::arg("env_blob"), pybind11::arg("obs_stats_out"),
           pybind11::arg("obs_dim"), pybind11::arg("act_dim"), pybind11::arg("rank"), pybind11::arg("steps"),
           pybind11::arg("alive_bonus"), pybind11::arg("act_cost"), pybind11::arg("init_seed"),
-          pybind11::arg("member_offset"), pybind11::arg("policy_hidden") = 0);
+          pybind11::arg("member_offset"), pybind11::arg("policy_hidden") = 0,
+          pybind11::arg("seed_buf") = c10::optional<torch::Tensor>());
+    m.def("bump_seed", &ea::bump_seed,
+          "advance a device splitmix64 seed chain by one step (hipGraph-safe episode seeds)");
     m.def("cma_update_c", &ea::cma_update_c,
           "K5: fused CMA-ES covariance update (scale*C + c1*pc pc^T + cmu*Y^T diag(w) Y, exact symmetry)");
     m.def("potrf_tile", &ea::potrf_tile,

@@ -182,6 +182,17 @@ void sample_gaussian_graphsafe(torch::Tensor out, torch::Tensor mu, torch::Tenso
     hipLaunchKernelGGL(bump_seed_kernel, dim3(1), dim3(1), 0, stream, ptr);
 }
 
+// Advance a device seed chain by one splitmix64 step (standalone entry for
+// graph-safe consumers like the rollout kernels: bump, then launch with
+// seed_buf — replays draw fresh episode seeds).
+void bump_seed(torch::Tensor seed_buf) {
+    TORCH_CHECK(seed_buf.is_cuda() && seed_buf.scalar_type() == at::ScalarType::Long && seed_buf.numel() >= 1,
+                "seed_buf must be a cuda int64 tensor");
+    auto* ptr = reinterpret_cast<unsigned long long*>(seed_buf.data_ptr<int64_t>());
+    auto stream = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(bump_seed_kernel, dim3(1), dim3(1), 0, stream, ptr);
+}
+
 // Apply x = mu + sigma*z from PRE-GENERATED standard normals (the noise
 // was filled on a side stream, hidden behind evaluation / collectives;
 // only this cheap bandwidth-bound affine pass sits on the critical path

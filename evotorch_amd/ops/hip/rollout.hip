@@ -97,6 +97,7 @@ struct RolloutArgs {
     int obs_dim, act_dim, rank, steps, hidden;
     float alive_bonus, act_cost;
     unsigned long long init_seed;
+    const unsigned long long* seed_ptr;  // device episode seed (hipGraph-safe); overrides init_seed
 };
 
 // one group-reduced dot output: 8 lanes compute row·vec, lane 0 of the
@@ -221,9 +222,10 @@ __global__ __launch_bounds__(512, 2) void rollout_linear_kernel(RolloutArgs args
         for (int i = tid; i < R_PAD; i += blockDim.x) h_b[m][i] = f2b(0.0f);
         for (int i = tid; i < A_PAD; i += blockDim.x) act_b[m][i] = f2b(0.0f);
         const unsigned long long gmember = (unsigned long long)(args.member_offset + base_member + m);
+        const unsigned long long iseed = args.seed_ptr ? *args.seed_ptr : args.init_seed;
         for (int j4 = tid; j4 * 4 < O; j4 += blockDim.x) {
             float z[4];
-            philox_normal4(args.init_seed, (uint32_t)gmember, (uint64_t)j4, z);
+            philox_normal4(iseed, (uint32_t)gmember, (uint64_t)j4, z);
 #pragma unroll
             for (int u = 0; u < 4; ++u) {
                 const int j = j4 * 4 + u;
@@ -476,11 +478,18 @@ static void launch_rollout(int n_blocks, int block, size_t lds_bytes, hipStream_
 
 void rollout_v7(torch::Tensor params, torch::Tensor env_blob, torch::Tensor obs_stats_out, torch::Tensor fitness,
                 int64_t obs_dim, int64_t act_dim, int64_t rank, int64_t steps, double alive_bonus, double act_cost,
-                int64_t init_seed, int64_t member_offset);  // rollout_v7.hip
+                int64_t init_seed, int64_t member_offset, const unsigned long long* seed_ptr);  // rollout_v7.hip
 
 torch::Tensor rollout_linear(torch::Tensor params, torch::Tensor env_blob, torch::Tensor obs_stats_out,
                              int64_t obs_dim, int64_t act_dim, int64_t rank, int64_t steps, double alive_bonus,
-                             double act_cost, int64_t init_seed, int64_t member_offset, int64_t policy_hidden) {
+                             double act_cost, int64_t init_seed, int64_t member_offset, int64_t policy_hidden,
+                             c10::optional<torch::Tensor> seed_buf) {
+    const unsigned long long* seed_ptr = nullptr;
+    if (seed_buf.has_value()) {
+        TORCH_CHECK(seed_buf->is_cuda() && seed_buf->scalar_type() == at::ScalarType::Long && seed_buf->numel() >= 1,
+                    "seed_buf must be a device int64 scalar");
+        seed_ptr = reinterpret_cast<const unsigned long long*>(seed_buf->data_ptr<int64_t>());
+    }
     CHECK_GPU(params);
     TORCH_CHECK(params.is_contiguous() && params.dim() == 2, "params must be contiguous [N][L]");
     TORCH_CHECK(params.scalar_type() == at::ScalarType::Float, "params must be fp32");
@@ -497,7 +506,7 @@ torch::Tensor rollout_linear(torch::Tensor params, torch::Tensor env_blob, torch
         const int n_blocks7 = (n + 15) / 16;
         auto stat_partials = torch::zeros({(int64_t)n_blocks7, 2 * (int64_t)O}, params.options());
         rollout_v7(params, env_blob, stat_partials, fitness, obs_dim, act_dim, rank, steps, alive_bonus, act_cost,
-                   init_seed, member_offset);
+                   init_seed, member_offset, seed_ptr);
         obs_stats_out.add_(stat_partials.sum(0));
         return fitness;
     }
@@ -514,6 +523,7 @@ torch::Tensor rollout_linear(torch::Tensor params, torch::Tensor env_blob, torch
     args.alive_bonus = (float)alive_bonus;
     args.act_cost = (float)act_cost;
     args.init_seed = (unsigned long long)init_seed;
+    args.seed_ptr = seed_ptr;
 
     const int block = 512;
     const int A_PAD = (A + 2) & ~1, R_PAD = (R + 2) & ~1;

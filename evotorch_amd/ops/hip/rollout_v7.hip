@@ -94,6 +94,7 @@ struct RolloutV7Args {
     int obs_dim, act_dim, rank, steps;
     float alive_bonus, act_cost;
     unsigned long long init_seed;
+    const unsigned long long* seed_ptr;  // device episode seed (hipGraph-safe); overrides init_seed
     int skip_mask;  // perf probe only (EVOTORCH_AMD_V7_SKIP): 1=policy 2=GEMM1 4=GEMM2
 };
 
@@ -221,7 +222,8 @@ __global__ __launch_bounds__(64 * kWaves, 1) void rollout_v7_kernel(RolloutV7Arg
             const int j4 = idx % per_member4;
             if (m >= live) continue;
             float z[4];
-            philox_normal4(args.init_seed, (uint32_t)(args.member_offset + base_member + m), (uint64_t)j4, z);
+            const unsigned long long iseed = args.seed_ptr ? *args.seed_ptr : args.init_seed;
+            philox_normal4(iseed, (uint32_t)(args.member_offset + base_member + m), (uint64_t)j4, z);
 #pragma unroll
             for (int u = 0; u < 4; ++u) {
                 const int j = j4 * 4 + u;
@@ -404,12 +406,13 @@ static void launch_v7(const RolloutV7Args& args, int n, hipStream_t stream) {
 
 void rollout_v7(torch::Tensor params, torch::Tensor env_blob, torch::Tensor obs_stats_out, torch::Tensor fitness,
                 int64_t obs_dim, int64_t act_dim, int64_t rank, int64_t steps, double alive_bonus, double act_cost,
-                int64_t init_seed, int64_t member_offset) {
+                int64_t init_seed, int64_t member_offset, const unsigned long long* seed_ptr) {
     const int n = (int)params.size(0);
     const int O = (int)obs_dim, A = (int)act_dim, R = (int)rank;
     TORCH_CHECK(R == 16, "rollout v7 requires rank 16");
 
     RolloutV7Args args;
+    args.seed_ptr = seed_ptr;
     args.params = params.data_ptr<float>();
     args.env_blob = env_blob.data_ptr<float>();
     args.fitness_out = fitness.data_ptr<float>();

@@ -418,3 +418,28 @@ def test_vecenvne_hip_graph_recaptures_on_popsize_change():
         batch.access_values()[:] = 0.01 * torch.randn(n, prob.solution_length, device="cuda:0")
         prob.evaluate(batch)
         assert batch.evals_are_ready
+
+
+@requires_gpu
+def test_graphed_flagship_rollout():
+    """GraphedSearch over the fused-rollout flagship problem: the whole
+    PGPE generation (sampling, rollout kernel, ranking, gradients,
+    ClipUp, obs-norm merge) replays as ONE hipGraph. Episode seeds come
+    from a device splitmix chain (bumped in-graph), so replays run fresh
+    episodes and observation statistics keep accumulating."""
+    from evotorch_amd.algorithms import PGPE, GraphedSearch
+    from evotorch_amd.neuroevolution import SyntheticRolloutProblem
+
+    prob = SyntheticRolloutProblem(device="cuda:0", seed=11, episode_length=50)
+    r = 2.25
+    s = PGPE(prob, popsize=512, radius_init=r, center_learning_rate=0.75 * r / 15,
+             stdev_learning_rate=0.1, optimizer="clipup", optimizer_config={"max_speed": r / 15})
+    g = GraphedSearch(s, generations_per_capture=5)
+    g.capture()
+    seed_before = int(prob._graph_seed_buf.item())
+    count_before = prob.obs_norm.count
+    first = float(g.mean_eval)
+    g.run(100)
+    assert int(prob._graph_seed_buf.item()) != seed_before  # in-graph bump ran
+    assert prob.obs_norm.count > count_before  # stats merged inside the graph
+    assert float(g.mean_eval) > first + 20.0  # it optimizes
