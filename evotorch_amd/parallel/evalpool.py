@@ -57,6 +57,10 @@ def _worker_init(payload: bytes, base_seed: int, counter) -> None:
     _WORKER_PROBLEM = problem
 
 
+def _worker_ping():
+    return _WORKER_PROBLEM is not None
+
+
 def _worker_eval(args):
     """Evaluate one contiguous piece of the population; return (piece
     index, evals, aux sync data)."""
@@ -114,6 +118,22 @@ class EvalPool:
         ctx = mp.get_context("spawn")
         counter = ctx.Value("i", 0)
         self._pool = ctx.Pool(self._num_actors, initializer=_worker_init, initargs=(payload, int(base_seed) & 0x7FFFFFFF, counter))
+        # readiness probe: a worker whose spawn/init keeps failing makes the
+        # pool respawn forever and evaluation would HANG; surface it instead.
+        # (The classic cause: the launching script is not importable — run
+        # pool-using code under `if __name__ == "__main__":`, like any
+        # multiprocessing or DataLoader-workers program.)
+        timeout_s = float(os.environ.get("EVOTORCH_AMD_POOL_INIT_TIMEOUT", "120"))
+        try:
+            self._pool.apply_async(_worker_ping).get(timeout=timeout_s)
+        except mp.TimeoutError:
+            self.close()
+            raise RuntimeError(
+                f"evaluation workers failed to come up within {timeout_s:.0f}s "
+                "(num_actors pool). Most common cause: the launching script is not "
+                "importable by spawned processes — guard the entry point with "
+                "`if __name__ == '__main__':` (the standard multiprocessing requirement)."
+            ) from None
 
     def _piece_size(self, n: int) -> int:
         if self._subbatch_size is not None:

@@ -70,3 +70,30 @@ def test_pool_gymne_merges_counters_and_obs_stats():
         assert prob.obs_norm.has_data  # worker stats merged back
     finally:
         prob.kill_actors()
+
+
+def test_pool_object_dtype_problem():
+    """Variable-length (object dtype) problems evaluate through the pool:
+    ObjectArray pieces ship via cloudpickle, per-solution `_evaluate` runs
+    in the workers."""
+    from evotorch_amd import Problem
+
+    class VarLenProblem(Problem):
+        def __init__(self, **kw):
+            super().__init__("min", dtype=object, seed=3, **kw)
+
+        def _fill(self, values):
+            for i in range(len(values)):
+                values[i] = [float(i), float(i + 1)]
+
+        def _evaluate(self, solution):
+            solution.set_evaluation(float(sum(solution.values)))
+
+    prob = VarLenProblem(num_actors=2)
+    try:
+        b = prob.generate_batch(6)
+        prob.evaluate(b)
+        got = [float(x) for x in b.evals[:, 0]]
+        assert got == [1.0, 3.0, 5.0, 7.0, 9.0, 11.0]
+    finally:
+        prob.kill_actors()
