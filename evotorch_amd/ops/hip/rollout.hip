@@ -478,8 +478,7 @@ static void launch_rollout(int n_blocks, int block, size_t lds_bytes, hipStream_
 
 void rollout_v7(torch::Tensor params, torch::Tensor env_blob, torch::Tensor obs_stats_out, torch::Tensor fitness,
                 int64_t obs_dim, int64_t act_dim, int64_t rank, int64_t steps, double alive_bonus, double act_cost,
-                int64_t init_seed, int64_t member_offset, const unsigned long long* seed_ptr,
-                bool use_v8);  // rollout_v7.hip
+                int64_t init_seed, int64_t member_offset, const unsigned long long* seed_ptr);  // rollout_v7.hip
 
 torch::Tensor rollout_linear(torch::Tensor params, torch::Tensor env_blob, torch::Tensor obs_stats_out,
                              int64_t obs_dim, int64_t act_dim, int64_t rank, int64_t steps, double alive_bonus,
@@ -504,11 +503,10 @@ torch::Tensor rollout_linear(torch::Tensor params, torch::Tensor env_blob, torch
     // v7 (MFMA, 16 members/block) serves the linear flagship geometry;
     // v6 covers MLP policies and off-geometry envs.
     if (H == 0 && R == 16 && O == 376 && A == 17 && !getenv("EVOTORCH_AMD_ROLLOUT_V6")) {
-        const bool use_v8 = getenv("EVOTORCH_AMD_ROLLOUT_V8") != nullptr;
-        const int n_pblocks = use_v8 ? (n + 7) / 8 : (n + 15) / 16;
-        auto stat_partials = torch::zeros({(int64_t)n_pblocks, 2 * (int64_t)O}, params.options());
+        const int n_blocks7 = (n + 15) / 16;
+        auto stat_partials = torch::zeros({(int64_t)n_blocks7, 2 * (int64_t)O}, params.options());
         rollout_v7(params, env_blob, stat_partials, fitness, obs_dim, act_dim, rank, steps, alive_bonus, act_cost,
-                   init_seed, member_offset, seed_ptr, use_v8);
+                   init_seed, member_offset, seed_ptr);
         obs_stats_out.add_(stat_partials.sum(0));
         return fitness;
     }

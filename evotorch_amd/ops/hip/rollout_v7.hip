@@ -387,263 +387,6 @@ __global__ __launch_bounds__(64 * kWaves, 1) void rollout_v7_kernel(RolloutV7Arg
     }
 }
 
-// ===========================================================================
-// v8: ONE MEMBER PER WAVE, zero barriers in the step loop.
-//
-// v7's PMC profile (profiles/pmc_v7_final_summary.txt) shows
-// SQ_WAIT:SQ_BUSY ≈ 5.4 — the 16-members-per-block design serializes on
-// its two block barriers and LDS round-trips every step, and with one
-// block per CU (228 VGPRs) nothing covers those stalls. v8 removes the
-// synchronization instead of hiding it: each WAVE owns one member for
-// the whole episode, with
-//   * the policy weights (17×6 bf16x2 = 51 VGPRs) AND the dynamics
-//     matrix [U;D2] (17 k-pairs × 6 cols = 102 VGPRs) in registers,
-//   * the observation held in 3 bf16x2 registers per lane (6 columns),
-//   * V read from LDS (the only LDS traffic in the loop: 48 dwords per
-//     lane-step, conflict-free),
-//   * the 33 per-member dots reduced with DPP and broadcast through
-//     __builtin_amdgcn_readlane (SGPRs) — no LDS, no barrier, and the
-//     dynamics update runs as v_dot2 with a uniform bf16x2 k-pair
-//     operand.
-// A wave never waits for another wave; the two co-resident waves per
-// SIMD cover each other's VALU/trans latencies. Numerics contract is
-// v6/v7's: bf16 operands, fp32 accumulation, quantize-then-normalize.
-// ===========================================================================
-
-template <int O, int A>
-__global__ __launch_bounds__(512, 1) void rollout_v8_kernel(RolloutV7Args args) {
-    constexpr int kWaves8 = 8;                   // members per block
-    constexpr int OP = (O + 127) / 128 * 128;    // 384: padded cols (zero-filled)
-    constexpr int kPairs = OP / 2 / 64;          // bf16x2 col-pairs per lane (3)
-    constexpr int kCols = 2 * kPairs;            // columns per lane (6)
-    constexpr int R = 16;
-    constexpr int K = R + A;                     // 33 dynamics rows [U;D2]
-    constexpr int KQ = (K + 1) / 2;              // 17 k-pairs (last padded 0)
-    constexpr int VS = OP + 8;                   // LDS V row stride (bf16)
-
-    const int tid = threadIdx.x;
-    const int lane = tid & 63;
-    const int wave = tid >> 6;
-    const int member = blockIdx.x * kWaves8 + wave;
-    const bool live = member < args.n_members;
-    const int mem_clamped = live ? member : 0;
-
-    extern __shared__ unsigned char lds8[];
-    __bf16* v_l = reinterpret_cast<__bf16*>(lds8);            // [R][VS]
-    float* mean_l = reinterpret_cast<float*>(v_l + R * VS);   // [OP]
-    float* istd_l = mean_l + OP;                              // [OP]
-    float* c_l = istd_l + OP;                                 // [OP]
-    float* wr_l = c_l + OP;                                   // [OP]
-    float* sstat = wr_l + OP;                                 // [2][kWaves8][OP] end-of-episode only
-
-    const float* e_V = args.env_blob;
-    const float* e_M = args.env_blob + (size_t)R * O;         // [U_T(16); D2_T(17)] rows × O
-    const float* e_c = args.env_blob + (size_t)(2 * R + A) * O;
-    const float* e_wr = e_c + O;
-    const float* e_mean = e_wr + O;
-    const float* e_std = e_mean + O;
-
-    for (int j = tid; j < R * VS; j += 512) {
-        const int r = j / VS, cc = j % VS;
-        v_l[j] = (cc < O) ? f2b7(e_V[(size_t)r * O + cc]) : f2b7(0.0f);
-    }
-    // per-column epilogue params live in LDS, not registers: holding the
-    // 24 fp32 per-lane copies measured VGPR spills (the step loop's
-    // register demand sits right at the 256 ceiling)
-    for (int j = tid; j < OP; j += 512) {
-        const bool in = j < O;
-        mean_l[j] = in ? e_mean[j] : 0.0f;
-        istd_l[j] = in ? 1.0f / e_std[j] : 0.0f;
-        c_l[j] = in ? e_c[j] : 0.0f;
-        wr_l[j] = in ? e_wr[j] : 0.0f;
-    }
-
-    // ---- policy weights + dynamics matrix in registers ----------------------
-    bf16x2_t w2[A][kPairs];
-    {
-        const float* W = args.params + (long)mem_clamped * ((long)A * O + A);
-#pragma unroll
-        for (int a = 0; a < A; ++a) {
-#pragma unroll
-            for (int p = 0; p < kPairs; ++p) {
-                const int c0 = 2 * (p * 64 + lane);
-                bf16x2_t w;
-                w.x = (c0 < O) ? f2b7(W[(long)a * O + c0]) : f2b7(0.0f);
-                w.y = (c0 + 1 < O) ? f2b7(W[(long)a * O + c0 + 1]) : f2b7(0.0f);
-                w2[a][p] = w;
-            }
-        }
-    }
-    float bias[A];
-    {
-        const float* B = args.params + (long)mem_clamped * ((long)A * O + A) + (long)A * O;
-#pragma unroll
-        for (int a = 0; a < A; ++a)
-            // wave-uniform: force into SGPRs (17 VGPRs saved)
-            bias[a] = __builtin_bit_cast(float, __builtin_amdgcn_readfirstlane(__builtin_bit_cast(int, B[a])));
-    }
-    bf16x2_t ud2[KQ][kCols];  // (M[2q][col], M[2q+1][col])
-#pragma unroll
-    for (int q = 0; q < KQ; ++q) {
-#pragma unroll
-        for (int p = 0; p < kPairs; ++p) {
-#pragma unroll
-            for (int e = 0; e < 2; ++e) {
-                const int col = 2 * (p * 64 + lane) + e;
-                bf16x2_t m;
-                m.x = (col < O) ? f2b7(e_M[(size_t)(2 * q) * O + col]) : f2b7(0.0f);
-                m.y = (col < O && 2 * q + 1 < K) ? f2b7(e_M[(size_t)(2 * q + 1) * O + col]) : f2b7(0.0f);
-                ud2[q][2 * p + e] = m;
-            }
-        }
-    }
-
-    float ssum[kCols] = {}, ssq[kCols] = {};
-    float fit_acc = 0.0f;
-    float asq_total = 0.0f;
-    __syncthreads();  // v_l + per-column params staged (the ONLY pre-loop barrier)
-
-    // ---- initial observation (philox stream per global member) --------------
-    const unsigned long long iseed = args.seed_ptr ? *args.seed_ptr : args.init_seed;
-    const int odd = lane & 1;  // which half of the philox quad this lane's pair is
-    bf16x2_t obs2[kPairs], obsn2[kPairs];
-#pragma unroll
-    for (int p = 0; p < kPairs; ++p) {
-        const int pi = p * 64 + lane;
-        float z[4];
-        philox_normal4(iseed, (uint32_t)(args.member_offset + member), (uint64_t)(pi >> 1), z);
-        bf16x2_t o, onr;
-#pragma unroll
-        for (int e = 0; e < 2; ++e) {
-            const int col = 2 * pi + e;
-            // quad element index is e + 2*(lane&1): constexpr candidates +
-            // a branchless select (a runtime z[col&3] would demote z to
-            // scratch — the v7.0 lesson)
-            const float zv = odd ? z[e + 2] : z[e];
-            const __bf16 ob = (col < O) ? f2b7(0.1f * zv) : f2b7(0.0f);
-            const float onf = (b2f7(ob) - mean_l[col < OP ? col : 0]) * istd_l[col < OP ? col : 0];
-            if (e == 0) { o.x = ob; onr.x = f2b7(onf); } else { o.y = ob; onr.y = f2b7(onf); }
-        }
-        obs2[p] = o;
-        obsn2[p] = onr;
-    }
-
-    for (int t = 0; t < args.steps; ++t) {
-        // ---- the 33 per-member dots, computed in k order and packed into
-        // uniform bf16x2 k-pairs on the fly (holding separate h[16]/act[17]
-        // arrays measured 151 VGPR spills — only two scalars are live here).
-        // k < R: dynamics row (V from LDS · raw obs); k >= R: policy row
-        // (registers · normalized obs) with bias + clamp epilogue. ----
-        bf16x2_t hact2[KQ];
-#pragma unroll
-        for (int q = 0; q < KQ; ++q) {
-            float kv[2];
-#pragma unroll
-            for (int e = 0; e < 2; ++e) {
-                const int k = 2 * q + e;
-                if (k >= K) {
-                    kv[e] = 0.0f;
-                    continue;
-                }
-                float acc = 0.0f;
-                if (k < R) {
-#pragma unroll
-                    for (int p = 0; p < kPairs; ++p) {
-                        const bf16x2_t v = *reinterpret_cast<const bf16x2_t*>(v_l + k * VS + 2 * (p * 64 + lane));
-                        acc = __builtin_amdgcn_fdot2_f32_bf16(v, obs2[p], acc, false);
-                    }
-                } else {
-#pragma unroll
-                    for (int p = 0; p < kPairs; ++p)
-                        acc = __builtin_amdgcn_fdot2_f32_bf16(w2[k - R][p], obsn2[p], acc, false);
-                }
-                acc = reduce32_dpp(acc);
-                float tot = __builtin_bit_cast(float, __builtin_amdgcn_readlane(__builtin_bit_cast(int, acc), 31)) +
-                            __builtin_bit_cast(float, __builtin_amdgcn_readlane(__builtin_bit_cast(int, acc), 63));
-                if (k >= R) {
-                    tot = fminf(fmaxf(tot + bias[k - R], -1.0f), 1.0f);
-                    asq_total = fmaf(tot, tot, asq_total);
-                }
-                kv[e] = tot;
-            }
-            bf16x2_t hp;
-            hp.x = f2b7(kv[0]);
-            hp.y = f2b7(kv[1]);
-            hact2[q] = hp;
-        }
-        // ---- o' = tanh(hact @ [U;D2] + c): per-lane columns ----
-#pragma unroll
-        for (int p = 0; p < kPairs; ++p) {
-            const int col0 = 2 * (p * 64 + lane);
-            bf16x2_t o, onr;
-#pragma unroll
-            for (int e = 0; e < 2; ++e) {
-                const int j = 2 * p + e;
-                float acc = 0.0f;
-#pragma unroll
-                for (int q = 0; q < KQ; ++q) acc = __builtin_amdgcn_fdot2_f32_bf16(ud2[q][j], hact2[q], acc, false);
-                const float o_new = tanh_fast(acc + c_l[col0 + e]);
-                fit_acc = fmaf(wr_l[col0 + e], o_new, fit_acc);
-                ssum[j] += o_new;
-                ssq[j] = fmaf(o_new, o_new, ssq[j]);
-                const __bf16 ob = f2b7(o_new);
-                const float onf = (b2f7(ob) - mean_l[col0 + e]) * istd_l[col0 + e];
-                if (e == 0) { o.x = ob; onr.x = f2b7(onf); } else { o.y = ob; onr.y = f2b7(onf); }
-            }
-            obs2[p] = o;
-            obsn2[p] = onr;
-        }
-    }
-
-    // ---- wrap-up: fitness (deterministic wave reduce) ----
-    {
-        float f = reduce32_dpp(fit_acc);
-        const float total = __builtin_bit_cast(float, __builtin_amdgcn_readlane(__builtin_bit_cast(int, f), 31)) +
-                            __builtin_bit_cast(float, __builtin_amdgcn_readlane(__builtin_bit_cast(int, f), 63));
-        if (lane == 0 && live)
-            args.fitness_out[member] =
-                total + args.alive_bonus * (float)args.steps - args.act_cost * asq_total / (float)A;
-    }
-    // ---- block stat partial: per-wave slices summed in fixed order ----
-#pragma unroll
-    for (int p = 0; p < kPairs; ++p) {
-#pragma unroll
-        for (int e = 0; e < 2; ++e) {
-            const int col = 2 * (p * 64 + lane) + e;
-            if (col < OP) {
-                sstat[(size_t)wave * OP + col] = live ? ssum[2 * p + e] : 0.0f;
-                sstat[(size_t)(kWaves8 + wave) * OP + col] = live ? ssq[2 * p + e] : 0.0f;
-            }
-        }
-    }
-    __syncthreads();
-    float* stats = args.obs_stats_out + (int64_t)blockIdx.x * 2 * O;
-    for (int col = tid; col < O; col += 512) {
-        float s = 0.0f, q = 0.0f;
-#pragma unroll
-        for (int w = 0; w < kWaves8; ++w) {
-            s += sstat[(size_t)w * OP + col];
-            q += sstat[(size_t)(kWaves8 + w) * OP + col];
-        }
-        stats[col] = s;
-        stats[O + col] = q;
-    }
-}
-
-template <int O_T, int A_T>
-static void launch_v8(const RolloutV7Args& args, int n, hipStream_t stream) {
-    constexpr int OP = (O_T + 127) / 128 * 128;
-    const size_t lds = (size_t)16 * (OP + 8) * 2 + (size_t)4 * OP * 4 + (size_t)2 * 8 * OP * 4;
-    static bool attr_set8 = false;
-    if (!attr_set8) {
-        (void)hipFuncSetAttribute(reinterpret_cast<const void*>(&rollout_v8_kernel<O_T, A_T>),
-                                  hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
-        attr_set8 = true;
-    }
-    const int blocks = (n + 7) / 8;
-    hipLaunchKernelGGL((rollout_v8_kernel<O_T, A_T>), dim3(blocks), dim3(512), lds, stream, args);
-}
-
 template <int O_T, int A_T, int kWaves>
 static void launch_v7(const RolloutV7Args& args, int n, hipStream_t stream) {
     constexpr int OP = (O_T + 127) / 128 * 128;
@@ -663,7 +406,7 @@ static void launch_v7(const RolloutV7Args& args, int n, hipStream_t stream) {
 
 void rollout_v7(torch::Tensor params, torch::Tensor env_blob, torch::Tensor obs_stats_out, torch::Tensor fitness,
                 int64_t obs_dim, int64_t act_dim, int64_t rank, int64_t steps, double alive_bonus, double act_cost,
-                int64_t init_seed, int64_t member_offset, const unsigned long long* seed_ptr, bool use_v8) {
+                int64_t init_seed, int64_t member_offset, const unsigned long long* seed_ptr) {
     const int n = (int)params.size(0);
     const int O = (int)obs_dim, A = (int)act_dim, R = (int)rank;
     TORCH_CHECK(R == 16, "rollout v7 requires rank 16");
@@ -687,10 +430,6 @@ void rollout_v7(torch::Tensor params, torch::Tensor env_blob, torch::Tensor obs_
     auto stream = at::cuda::getCurrentCUDAStream();
     args.skip_mask = 0;
     if (const char* skp = getenv("EVOTORCH_AMD_V7_SKIP")) args.skip_mask = atoi(skp);  // perf probe only
-    if (use_v8) {
-        launch_v8<O_T, A_T>(args, n, stream);
-        return;
-    }
     const char* wenv = getenv("EVOTORCH_AMD_ROLLOUT_V7_WAVES");
     if (wenv && atoi(wenv) == 4) {
         launch_v7<O_T, A_T, 4>(args, n, stream);
