@@ -253,12 +253,21 @@ __global__ __launch_bounds__(64 * kWaves, 1) void rollout_v7_kernel(RolloutV7Arg
         // work so the MFMA chain executes on the MAI pipe underneath it =====
         floatx4_t h_acc = {0.f, 0.f, 0.f, 0.f};
         if (wave == 0 && !(args.skip_mask & 2)) {
+            // two accumulator chains: a single accumulator serializes 12
+            // dependent MFMAs (~32 cycles each) on wave 0, which is the
+            // barrier straggler (it also runs its policy share)
+            floatx4_t h_acc1 = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
-            for (int s = 0; s < OP / 32; ++s) {
-                const bf16x8_t a_frag = *reinterpret_cast<const bf16x8_t*>(obs_l + a_row * OPS + s * 32 + g2_k0);
-                const bf16x8_t b_frag = *reinterpret_cast<const bf16x8_t*>(v_l + c_col * OPS + s * 32 + g2_k0);
-                h_acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, b_frag, h_acc, 0, 0, 0);
+            for (int s = 0; s < OP / 32; s += 2) {
+                const bf16x8_t a0 = *reinterpret_cast<const bf16x8_t*>(obs_l + a_row * OPS + s * 32 + g2_k0);
+                const bf16x8_t b0 = *reinterpret_cast<const bf16x8_t*>(v_l + c_col * OPS + s * 32 + g2_k0);
+                h_acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, h_acc, 0, 0, 0);
+                const bf16x8_t a1 = *reinterpret_cast<const bf16x8_t*>(obs_l + a_row * OPS + (s + 1) * 32 + g2_k0);
+                const bf16x8_t b1 = *reinterpret_cast<const bf16x8_t*>(v_l + c_col * OPS + (s + 1) * 32 + g2_k0);
+                h_acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b1, h_acc1, 0, 0, 0);
             }
+#pragma unroll
+            for (int r = 0; r < 4; ++r) h_acc[r] += h_acc1[r];
         }
         // ===== policy: act = clamp(W · obsn + b) — per-member half-waves =====
         if (!(args.skip_mask & 1)) {
