@@ -834,6 +834,7 @@ class Problem(TensorMakerMixin, Serializable):
         self._after_eval_hook = Hook()
         self._before_grad_hook = Hook()
         self._after_grad_hook = Hook()
+        self._remote_hook = Hook()
 
         # Counters
         self._after_eval_status: dict = {}
@@ -1010,6 +1011,15 @@ class Problem(TensorMakerMixin, Serializable):
     @property
     def after_grad_hook(self) -> Hook:
         return self._after_grad_hook
+
+    @property
+    def remote_hook(self) -> Hook:
+        """Runs once on each evaluation worker's problem clone right after
+        it is constructed (reference core.py:142, :2229 — the actor-side
+        initialization hook). Register callables taking the problem; they
+        execute inside the `num_actors` pool workers, never on the main
+        process."""
+        return self._remote_hook
 
     @property
     def status(self) -> dict:

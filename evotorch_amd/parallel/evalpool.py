@@ -54,6 +54,9 @@ def _worker_init(payload: bytes, base_seed: int, counter) -> None:
         problem._seed = seed
         problem._generator = torch.Generator(device=problem._device)
         problem._generator.manual_seed(seed)
+    hook = getattr(problem, "remote_hook", None)
+    if hook is not None:
+        hook(problem)  # reference core.py:142 — actor-side init hook
     _WORKER_PROBLEM = problem
 
 

@@ -97,3 +97,32 @@ def test_pool_object_dtype_problem():
         assert got == [1.0, 3.0, 5.0, 7.0, 9.0, 11.0]
     finally:
         prob.kill_actors()
+
+
+def test_remote_hook_runs_on_workers():
+    """remote_hook executes on each worker's problem clone (not on the
+    main process) — observable through the fitness it configures."""
+    from evotorch_amd import Problem
+
+    class OffsetProblem(Problem):
+        def __init__(self, **kw):
+            super().__init__("min", solution_length=3, initial_bounds=(-1, 1), seed=2, **kw)
+            self.offset = 0.0
+
+        def _evaluate(self, solution):
+            solution.set_evaluation(float(solution.values.sum()) + self.offset)
+
+    def set_offset(problem):
+        problem.offset = 100.0
+
+    prob = OffsetProblem(num_actors=2)
+    prob.remote_hook.append(set_offset)
+    try:
+        b = prob.generate_batch(4)
+        vals = torch.Tensor.as_subclass(b.access_values(keep_evals=True), torch.Tensor).clone()
+        prob.evaluate(b)
+        expected = vals.sum(-1) + 100.0
+        torch.testing.assert_close(torch.Tensor.as_subclass(b.evals[:, 0], torch.Tensor), expected)
+        assert prob.offset == 0.0  # main-process problem untouched
+    finally:
+        prob.kill_actors()
