@@ -36,3 +36,26 @@ def test_example_runs(script, args, tmp_path):
         timeout=420,
     )
     assert result.returncode == 0, f"{script} failed:\n{result.stderr[-2000:]}"
+
+
+def test_bench_json_contract():
+    """bench.py must print ONE JSON line with the driver-contract fields
+    (metric/value/unit/n_gpus/steps/warmup/ms_per_step/higher_is_better/
+    scaling/vs_baseline/dtype/data/config) — run tiny on CPU."""
+    import json
+
+    out = subprocess.run(
+        [sys.executable, os.path.join(REPO, "bench.py"), "--steps", "1", "--warmup", "0",
+         "--popsize", "8", "--episode-length", "5"],
+        capture_output=True, text=True, timeout=600,
+    )
+    assert out.returncode == 0, out.stderr[-2000:]
+    line = out.stdout.strip().splitlines()[-1]
+    d = json.loads(line)
+    for key in ("metric", "value", "unit", "n_gpus", "steps", "warmup", "ms_per_step",
+                "higher_is_better", "scaling", "vs_baseline", "dtype", "data", "config"):
+        assert key in d, f"missing {key}"
+    assert d["steps"] == 1 and d["warmup"] == 0
+    assert d["value"] > 0 and d["ms_per_step"] > 0
+    assert d["scaling"] in ("weak", "strong")
+    assert isinstance(d["config"], dict) and "model" in d["config"]
