@@ -191,6 +191,7 @@ class VecEnvNE(NEProblem):
         self._obs_norm = RunningNorm(shape=self._obs_dim, device=self.network_device)
         self._use_hip_graph = bool(use_hip_graph)
         self._graph_state: Optional[dict] = None
+        self._env_cache: Optional[dict] = None
         self._policy: Optional[Policy] = None
         self.last_eval_interaction_count = 0
         self._total_interactions = 0
@@ -215,8 +216,16 @@ class VecEnvNE(NEProblem):
             return self._env
         if self._env_factory is None:
             raise ValueError(f"The fixed env has {self._env.num_envs} rows but the batch needs {num_envs}; provide an env factory instead")
-        self._env = self._env_factory(num_envs)
-        return self._env
+        # cache per size: max_num_envs splitting an uneven population
+        # alternates between two sizes every generation
+        if self._env_cache is None:
+            self._env_cache = {}
+        env = self._env_cache.get(num_envs)
+        if env is None:
+            env = self._env_factory(num_envs)
+            self._env_cache[num_envs] = env
+        self._env = env
+        return env
 
     def _alive_bonus(self, t: int) -> float:
         if self._alive_bonus_schedule is None:
@@ -379,13 +388,16 @@ class VecEnvNE(NEProblem):
         - fixed horizon: the graph always runs max_steps (done rows are
           masked out exactly like the eager path, so fitness/steps/stats
           are identical, but no early exit happens).
-        The capture is keyed by (n, max_steps, policy length); changing
-        popsize re-captures."""
-        gr = self._graph_state
+        Captures are cached per (n, max_steps, policy length) so workloads
+        that alternate batch sizes (e.g. max_num_envs splitting an uneven
+        population) do not re-capture every call."""
+        if self._graph_state is None:
+            self._graph_state = {}
         key = (n, max_steps, int(params.shape[-1]), bool(self._obs_norm_enabled))
-        if gr is None or gr["key"] != key or gr["env"] is not env:
+        gr = self._graph_state.get(key)
+        if gr is None or gr["env"] is not env:
             gr = self._capture_rollout_graph(env, policy, params, episode_seed, n, max_steps, key)
-            self._graph_state = gr
+            self._graph_state[key] = gr
         else:
             gr["params_buf"].copy_(params)
             self._graph_prepare(gr, episode_seed)
