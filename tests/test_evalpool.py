@@ -126,3 +126,24 @@ def test_remote_hook_runs_on_workers():
         assert prob.offset == 0.0  # main-process problem untouched
     finally:
         prob.kill_actors()
+
+
+def test_pool_parallelizes_distributed_gradient_evaluation():
+    """PGPE(distributed=True) without a Comm routes its population
+    evaluation through Problem.evaluate — so the num_actors pool
+    parallelizes the expensive part of the gradient step too (the
+    reference's actor-parallel `sample_and_compute_gradients` use case)."""
+    from evotorch_amd.algorithms import PGPE
+    from evotorch_amd.neuroevolution import GymNE
+
+    prob = GymNE("CartPole-v1", "Linear(obs_length, act_length)", num_episodes=1,
+                 episode_length=20, seed=6, num_actors=2)
+    try:
+        s = PGPE(prob, popsize=8, center_learning_rate=0.2, stdev_learning_rate=0.1,
+                 radius_init=0.5, distributed=True)
+        s.run(2)
+        assert s.status["iter"] == 2
+        assert prob._eval_pool is not None  # the pool actually engaged
+        assert prob._total_interactions > 0
+    finally:
+        prob.kill_actors()
