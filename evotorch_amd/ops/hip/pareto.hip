@@ -88,6 +88,11 @@ __global__ void compact_front_kernel(int* __restrict__ counts, int64_t* __restri
 
 __global__ void reset_count_kernel(int* __restrict__ front_count) { *front_count = 0; }
 
+__global__ void zero_counts_kernel(int* __restrict__ front_counts, int batch) {
+    const int j = blockIdx.x * blockDim.x + threadIdx.x;
+    if (j < batch) front_counts[j] = 0;
+}
+
 __global__ void subtract_front_kernel(const float* __restrict__ utils, int* __restrict__ counts,
                                       const int* __restrict__ front_list, const int* __restrict__ front_count,
                                       int64_t n, int m) {
@@ -148,17 +153,22 @@ torch::Tensor pareto_ranks(torch::Tensor utils, int64_t min_assigned) {
     // a random 16k population can have hundreds of fronts). A peel with
     // an empty front is a no-op, so over-issuing is safe.
     auto front_list = torch::empty({n}, utils_f.options().dtype(torch::kInt32));
-    auto front_count = torch::zeros({1}, utils_f.options().dtype(torch::kInt32));
+    // per-front counter slots: zeroing them in ONE launch per batch (not
+    // one reset launch per front) cuts the peel's kernel launches by a
+    // third — at 32k population with ~350 fronts the loop is LAUNCH-bound
     constexpr int kPeelBatch = 16;
+    auto front_counts = torch::zeros({kPeelBatch}, utils_f.options().dtype(torch::kInt32));
     int64_t front_index = 0;
     while (front_index <= n) {
+        hipLaunchKernelGGL(zero_counts_kernel, dim3(1), dim3(kPeelBatch), 0, stream, front_counts.data_ptr<int>(),
+                           kPeelBatch);
         for (int k = 0; k < kPeelBatch; ++k) {
-            hipLaunchKernelGGL(reset_count_kernel, dim3(1), dim3(1), 0, stream, front_count.data_ptr<int>());
+            int* slot = front_counts.data_ptr<int>() + k;
             hipLaunchKernelGGL(compact_front_kernel, dim3(blocks), dim3(threads), 0, stream, counts.data_ptr<int>(),
-                               ranks.data_ptr<int64_t>(), front_list.data_ptr<int>(), front_count.data_ptr<int>(), n,
+                               ranks.data_ptr<int64_t>(), front_list.data_ptr<int>(), slot, n,
                                front_index);
             hipLaunchKernelGGL(subtract_front_kernel, dim3(blocks), dim3(threads), 0, stream, utils_f.data_ptr<float>(),
-                               counts.data_ptr<int>(), front_list.data_ptr<int>(), front_count.data_ptr<int>(), n, m);
+                               counts.data_ptr<int>(), front_list.data_ptr<int>(), slot, n, m);
             ++front_index;
         }
         const int64_t remaining = (counts >= 0).sum().item<int64_t>();  // one sync per batch
