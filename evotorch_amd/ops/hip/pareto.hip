@@ -160,57 +160,41 @@ torch::Tensor pareto_ranks(torch::Tensor utils, int64_t min_assigned) {
     constexpr int kPeelBatch = 16;
     auto front_counts = torch::zeros({kPeelBatch}, utils_f.options().dtype(torch::kInt32));
     auto front_index_dev = torch::zeros({1}, utils_f.options().dtype(torch::kInt64));
-    // Graph capture is illegal on the NULL/default stream, so the peel
-    // runs on a process-lifetime side stream joined to the torch stream
-    // with events on batch boundaries (the per-batch host sync exists
-    // anyway). Events/stream are created once and deliberately leaked.
-    static hipStream_t peel_stream = nullptr;
-    static hipEvent_t ev_in = nullptr, ev_out = nullptr;
-    if (peel_stream == nullptr) {
-        (void)hipStreamCreateWithFlags(&peel_stream, hipStreamNonBlocking);
-        (void)hipEventCreateWithFlags(&ev_in, hipEventDisableTiming);
-        (void)hipEventCreateWithFlags(&ev_out, hipEventDisableTiming);
-    }
     hipGraph_t peel_graph = nullptr;
     hipGraphExec_t peel_exec = nullptr;
     auto issue_batch = [&]() {
-        hipLaunchKernelGGL(zero_counts_kernel, dim3(1), dim3(kPeelBatch), 0, peel_stream, front_counts.data_ptr<int>(),
+        hipLaunchKernelGGL(zero_counts_kernel, dim3(1), dim3(kPeelBatch), 0, stream, front_counts.data_ptr<int>(),
                            kPeelBatch);
         for (int k = 0; k < kPeelBatch; ++k) {
             int* slot = front_counts.data_ptr<int>() + k;
-            hipLaunchKernelGGL(compact_front_kernel, dim3(blocks), dim3(threads), 0, peel_stream, counts.data_ptr<int>(),
+            hipLaunchKernelGGL(compact_front_kernel, dim3(blocks), dim3(threads), 0, stream, counts.data_ptr<int>(),
                                ranks.data_ptr<int64_t>(), front_list.data_ptr<int>(), slot, n,
                                reinterpret_cast<const long long*>(front_index_dev.data_ptr<int64_t>()));
-            hipLaunchKernelGGL(subtract_front_kernel, dim3(blocks), dim3(threads), 0, peel_stream,
-                               utils_f.data_ptr<float>(), counts.data_ptr<int>(), front_list.data_ptr<int>(), slot, n,
-                               m);
-            hipLaunchKernelGGL(bump_front_index_kernel, dim3(1), dim3(1), 0, peel_stream,
+            hipLaunchKernelGGL(subtract_front_kernel, dim3(blocks), dim3(threads), 0, stream, utils_f.data_ptr<float>(),
+                               counts.data_ptr<int>(), front_list.data_ptr<int>(), slot, n, m);
+            hipLaunchKernelGGL(bump_front_index_kernel, dim3(1), dim3(1), 0, stream,
                                reinterpret_cast<long long*>(front_index_dev.data_ptr<int64_t>()));
         }
     };
     int64_t front_index = 0;  // host mirror (advances in lock-step)
     int batch_no = 0;
     while (front_index <= n) {
-        (void)hipEventRecord(ev_in, stream);
-        (void)hipStreamWaitEvent(peel_stream, ev_in, 0);
         if (batch_no == 0) {
             issue_batch();  // first batch eager (small populations never pay capture cost)
         } else {
             if (peel_exec == nullptr && batch_no == 1) {
-                if (hipStreamBeginCapture(peel_stream, hipStreamCaptureModeThreadLocal) == hipSuccess) {
+                if (hipStreamBeginCapture(stream, hipStreamCaptureModeThreadLocal) == hipSuccess) {
                     issue_batch();
-                    if (hipStreamEndCapture(peel_stream, &peel_graph) == hipSuccess)
+                    if (hipStreamEndCapture(stream, &peel_graph) == hipSuccess)
                         (void)hipGraphInstantiate(&peel_exec, peel_graph, nullptr, nullptr, 0);
                 }
             }
             if (peel_exec != nullptr) {
-                (void)hipGraphLaunch(peel_exec, peel_stream);
+                (void)hipGraphLaunch(peel_exec, stream);
             } else {
                 issue_batch();  // capture unavailable: stay eager
             }
         }
-        (void)hipEventRecord(ev_out, peel_stream);
-        (void)hipStreamWaitEvent(stream, ev_out, 0);
         ++batch_no;
         front_index += kPeelBatch;
         const int64_t remaining = (counts >= 0).sum().item<int64_t>();  // one sync per batch
