@@ -72,18 +72,19 @@ __global__ void domination_counts_kernel(const float* __restrict__ utils, int* _
 
 __global__ void compact_front_kernel(int* __restrict__ counts, int64_t* __restrict__ ranks,
                                      int* __restrict__ front_list, int* __restrict__ front_count, int64_t n,
-                                     const long long* __restrict__ front_index_ptr) {
+                                     int64_t front_index) {
     const int64_t j = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+    if (j == 0 && front_index == 0) {
+        // nothing: front_count reset is handled by reset_count_kernel
+    }
     if (j >= n) return;
     if (counts[j] == 0) {
         const int pos = atomicAdd(front_count, 1);
         front_list[pos] = (int)j;
-        ranks[j] = (int64_t)(*front_index_ptr);
+        ranks[j] = front_index;
         counts[j] = -1;  // assigned marker
     }
 }
-
-__global__ void bump_front_index_kernel(long long* __restrict__ front_index_ptr) { ++*front_index_ptr; }
 
 __global__ void reset_count_kernel(int* __restrict__ front_count) { *front_count = 0; }
 
@@ -152,51 +153,24 @@ torch::Tensor pareto_ranks(torch::Tensor utils, int64_t min_assigned) {
     // a random 16k population can have hundreds of fronts). A peel with
     // an empty front is a no-op, so over-issuing is safe.
     auto front_list = torch::empty({n}, utils_f.options().dtype(torch::kInt32));
-    // The peel loop is LAUNCH-bound at large N (hundreds of fronts x 2
-    // small kernels): per-front counter slots remove the reset launch,
-    // the front index lives in DEVICE memory (bumped by a kernel), and
-    // from the second batch on the whole kPeelBatch-front batch replays
-    // as ONE hipGraph. One host sync per batch checks completion.
+    // per-front counter slots: zeroing them in ONE launch per batch (not
+    // one reset launch per front) cuts the peel's kernel launches by a
+    // third — at 32k population with ~350 fronts the loop is LAUNCH-bound
     constexpr int kPeelBatch = 16;
     auto front_counts = torch::zeros({kPeelBatch}, utils_f.options().dtype(torch::kInt32));
-    auto front_index_dev = torch::zeros({1}, utils_f.options().dtype(torch::kInt64));
-    hipGraph_t peel_graph = nullptr;
-    hipGraphExec_t peel_exec = nullptr;
-    auto issue_batch = [&]() {
+    int64_t front_index = 0;
+    while (front_index <= n) {
         hipLaunchKernelGGL(zero_counts_kernel, dim3(1), dim3(kPeelBatch), 0, stream, front_counts.data_ptr<int>(),
                            kPeelBatch);
         for (int k = 0; k < kPeelBatch; ++k) {
             int* slot = front_counts.data_ptr<int>() + k;
             hipLaunchKernelGGL(compact_front_kernel, dim3(blocks), dim3(threads), 0, stream, counts.data_ptr<int>(),
                                ranks.data_ptr<int64_t>(), front_list.data_ptr<int>(), slot, n,
-                               reinterpret_cast<const long long*>(front_index_dev.data_ptr<int64_t>()));
+                               front_index);
             hipLaunchKernelGGL(subtract_front_kernel, dim3(blocks), dim3(threads), 0, stream, utils_f.data_ptr<float>(),
                                counts.data_ptr<int>(), front_list.data_ptr<int>(), slot, n, m);
-            hipLaunchKernelGGL(bump_front_index_kernel, dim3(1), dim3(1), 0, stream,
-                               reinterpret_cast<long long*>(front_index_dev.data_ptr<int64_t>()));
+            ++front_index;
         }
-    };
-    int64_t front_index = 0;  // host mirror (advances in lock-step)
-    int batch_no = 0;
-    while (front_index <= n) {
-        if (batch_no == 0) {
-            issue_batch();  // first batch eager (small populations never pay capture cost)
-        } else {
-            if (peel_exec == nullptr && batch_no == 1) {
-                if (hipStreamBeginCapture(stream, hipStreamCaptureModeThreadLocal) == hipSuccess) {
-                    issue_batch();
-                    if (hipStreamEndCapture(stream, &peel_graph) == hipSuccess)
-                        (void)hipGraphInstantiate(&peel_exec, peel_graph, nullptr, nullptr, 0);
-                }
-            }
-            if (peel_exec != nullptr) {
-                (void)hipGraphLaunch(peel_exec, stream);
-            } else {
-                issue_batch();  // capture unavailable: stay eager
-            }
-        }
-        ++batch_no;
-        front_index += kPeelBatch;
         const int64_t remaining = (counts >= 0).sum().item<int64_t>();  // one sync per batch
         if (remaining == 0) break;
         if (n - remaining >= min_assigned) {
@@ -213,8 +187,6 @@ torch::Tensor pareto_ranks(torch::Tensor utils, int64_t min_assigned) {
             break;
         }
     }
-    if (peel_exec != nullptr) (void)hipGraphExecDestroy(peel_exec);
-    if (peel_graph != nullptr) (void)hipGraphDestroy(peel_graph);
     return ranks;
 }
 
