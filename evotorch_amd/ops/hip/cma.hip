@@ -95,7 +95,7 @@ __global__ __launch_bounds__(256) void cma_update_c_kernel(
 // this through torch.linalg.cholesky -> rocSOLVER, whose unblocked potf2
 // runs a chain of tiny kernels that dominated the d=4096 CMA-ES profile —
 // profiles/cma_phase_attribution_r2.log). ONE launch: stage the panel
-// into LDS (row stride 132 dwords keeps column walks off the 64-bank
+// into LDS (odd row stride 133 dwords keeps column walks off the 64-bank
 // alias), right-looking rank-1 factorization with the active column
 // mirrored in a flat buffer (conflict-free broadcast reads in the
 // trailing update), write back the lower triangle. The blocked driver
@@ -104,7 +104,7 @@ __global__ __launch_bounds__(256) void cma_update_c_kernel(
 // remains in the factorization.
 
 constexpr int kPotrfMax = 128;
-constexpr int kPotrfPad = kPotrfMax + 4;
+constexpr int kPotrfPad = kPotrfMax + 5;  // odd dword stride: 133 mod 64 = 5 (measured 0.51 conflict cycles/LDS instr at +4)
 
 constexpr int kPotrfBw = 8;        // micro-panel width (rank of each trailing update)
 constexpr int kPotrfPPad = kPotrfBw + 1;  // stripe mirror row stride (conflict-free column reads)
