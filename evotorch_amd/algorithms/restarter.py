@@ -54,6 +54,13 @@ class Restart(SearchAlgorithm):
     def num_restarts(self) -> int:
         return self._num_restarts
 
+    @property
+    def algorithm_args(self) -> dict:
+        """The LIVE keyword arguments the next restart will construct the
+        inner searcher with — `ModifyingRestart`'s modify callback (and
+        subclasses like IPOP) mutate this dict in place."""
+        return self._algorithm_args
+
     def _current_args(self) -> dict:
         return dict(self._algorithm_args)
 
