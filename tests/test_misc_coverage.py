@@ -118,3 +118,29 @@ def test_mlflow_logger_with_stub_client(monkeypatch):
     s.run(2)
     assert calls and all(c[0] == "r2" for c in calls)
     assert any(c[1] == "mean_eval" for c in calls)
+
+
+def test_small_dsl_layers():
+    """Clip/Bin/Slice/Round/Apply — the reference's utility layers, both
+    directly and through the str_to_net DSL."""
+    from evotorch_amd.models import str_to_net
+    from evotorch_amd.models.layers import Apply, Bin, Clip, Round, Slice
+
+    x = torch.tensor([-2.0, -0.2, 0.4, 3.0])
+    torch.testing.assert_close(Clip(-1.0, 1.0)(x), torch.tensor([-1.0, -0.2, 0.4, 1.0]))
+    b = Bin(-1.0, 1.0)(x)
+    assert set(b.tolist()) <= {-1.0, 1.0}
+    torch.testing.assert_close(Slice(1, 3)(x), torch.tensor([-0.2, 0.4]))
+    torch.testing.assert_close(Round(1)(torch.tensor([0.123, 0.678])), torch.tensor([0.1, 0.7]))
+    torch.testing.assert_close(Apply("tanh")(x), torch.tanh(x))
+    net = str_to_net("Linear(4, 3) >> Clip(-0.5, 0.5)")
+    y = net(torch.randn(4))
+    assert y.abs().max() <= 0.5
+
+
+def test_act_clip_layer_and_policy_export_box():
+    from evotorch_amd.neuroevolution.gymne import ActClipLayer
+
+    layer = ActClipLayer([-1.0, -2.0], [1.0, 2.0])
+    out = layer(torch.tensor([[-3.0, 5.0], [0.5, -0.5]]))
+    torch.testing.assert_close(out, torch.tensor([[-1.0, 2.0], [0.5, -0.5]]))
