@@ -383,6 +383,18 @@ def test_world4_matches_world1():
     assert torch.allclose(c1, c4, atol=1e-5, rtol=1e-5)
 
 
+def test_world5_matches_world1():
+    """Non-power-of-two world size (5 ranks, 8-direction shards of the
+    40-member virtual population): catches any hidden power-of-two
+    assumptions in the counter-addressed sharding."""
+    r1 = _run_world("_body_traj_pgpe", world=1)
+    r5 = _run_world("_body_traj_pgpe", world=5)
+    assert all(r5[0] == r5[i] for i in range(5)), "ranks diverged"
+    c1 = torch.tensor(r1[0], dtype=torch.float64)
+    c5 = torch.tensor(r5[0], dtype=torch.float64)
+    assert torch.allclose(c1, c5, atol=1e-5, rtol=1e-5)
+
+
 def test_bench_entry_torchrun_world2(tmp_path):
     """The driver launches bench.py via torch.distributed.run; validate that
     exact entry path (world 2, gloo on CPU) end to end."""
