@@ -144,3 +144,20 @@ def test_act_clip_layer_and_policy_export_box():
     layer = ActClipLayer([-1.0, -2.0], [1.0, 2.0])
     out = layer(torch.tensor([[-3.0, 5.0], [0.5, -0.5]]))
     torch.testing.assert_close(out, torch.tensor([[-1.0, 2.0], [0.5, -0.5]]))
+
+
+@pytest.mark.parametrize("dt", [torch.float64, torch.bfloat16, torch.float16])
+def test_nondefault_problem_dtypes(dt):
+    """float64 / bf16 / fp16 problems run a PGPE loop end to end (the
+    reference supports arbitrary float dtypes on Problem)."""
+    from evotorch_amd.algorithms import PGPE
+
+    @vectorized
+    def f(x):
+        return (x.float() ** 2).sum(-1)
+
+    p = Problem("min", f, solution_length=6, initial_bounds=(-1, 1), seed=1, dtype=dt)
+    s = PGPE(p, popsize=8, center_learning_rate=0.1, stdev_learning_rate=0.1, stdev_init=0.5)
+    s.run(3)
+    assert s.status["iter"] == 3
+    assert p.generate_batch(2).access_values().dtype == dt
