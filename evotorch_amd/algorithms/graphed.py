@@ -167,10 +167,19 @@ class GraphedSearch:
                 self._step_body()
                 self._searcher._steps_count += 1
         torch.cuda.current_stream().wait_stream(side)
-        self._graph = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self._graph):
-            for _ in range(self._gens_per_capture):
-                self._step_body()
+        # no cyclic GC during capture: freeing stale CUDA garbage on
+        # non-captured streams mid-capture aborts the HIP runtime
+        import gc
+
+        gc.collect()
+        gc.disable()
+        try:
+            self._graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(self._graph):
+                for _ in range(self._gens_per_capture):
+                    self._step_body()
+        finally:
+            gc.enable()
         self._searcher._steps_count += self._gens_per_capture
         return self
 

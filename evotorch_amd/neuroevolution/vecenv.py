@@ -474,9 +474,20 @@ class VecEnvNE(NEProblem):
                 self._graph_prepare(gr, episode_seed)
         torch.cuda.current_stream().wait_stream(side)
         torch.cuda.synchronize()
-        graph = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(graph):
-            body()
+        # The cyclic GC must not run DURING capture: collecting stale CUDA
+        # garbage (dead graphs/tensors from earlier work) mid-capture frees
+        # device memory on non-captured streams, which aborts the HIP
+        # runtime. Drain it now, then hold it off until capture ends.
+        import gc
+
+        gc.collect()
+        gc.disable()
+        try:
+            graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(graph):
+                body()
+        finally:
+            gc.enable()
         gr["graph"] = graph
         # capture records without executing: prepare() state is still intact,
         # so the caller's replay() right after this produces this
