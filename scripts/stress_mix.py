@@ -21,7 +21,7 @@ D = "cuda:0"
 gc.set_threshold(50, 5, 5)  # aggressive cyclic GC to provoke lifetime bugs
 for cycle in range(6):
     prob = SyntheticRolloutProblem(device=D, seed=cycle, episode_length=30)
-    s = PGPE(prob, popsize=256, center_learning_rate=0.1, stdev_learning_rate=0.1, radius_init=1.0, distributed=True)
+    s = PGPE(prob, popsize=256, center_learning_rate=0.1, stdev_learning_rate=0.1, radius_init=1.0)
     g = GraphedSearch(s, generations_per_capture=4)
     g.capture(); g.run(20)
 
