@@ -160,7 +160,11 @@ class GraphedSearch:
     # -- capture & replay ----------------------------------------------------
 
     def capture(self):
-        side = torch.cuda.Stream()
+        from ..neuroevolution.vecenv import _capture_stream
+
+        # persistent per-device capture stream: keeps the BLAS workspace
+        # OUT of the graph's private pool (see _capture_stream docstring)
+        side = _capture_stream(self._problem.device)
         side.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(side):
             for _ in range(self._warmup):
@@ -175,7 +179,7 @@ class GraphedSearch:
         gc.disable()
         try:
             self._graph = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(self._graph):
+            with torch.cuda.graph(self._graph, stream=side):
                 for _ in range(self._gens_per_capture):
                     self._step_body()
         finally:
