@@ -108,3 +108,22 @@ def test_mapelites_tiny_grid_gpu():
     s = MAPElites(prob, feature_grid=grid, re_evaluate=False, operators=[GaussianMutation(prob, stdev=0.2)])
     s.run(4)
     assert s.status["iter"] == 4
+
+
+@requires_gpu
+@pytest.mark.parametrize("dt", [torch.bfloat16, torch.float16, torch.float64])
+def test_nondefault_dtypes_gpu(dt):
+    """Non-fp32 problem dtypes through the kernel-backed PGPE loop."""
+    from evotorch_amd import Problem
+    from evotorch_amd.algorithms import PGPE
+    from evotorch_amd.decorators import vectorized
+
+    @vectorized
+    def f(x):
+        return (x.float() ** 2).sum(-1)
+
+    prob = Problem("min", f, solution_length=33, initial_bounds=(-1, 1), seed=3, device="cuda:0", dtype=dt)
+    s = PGPE(prob, popsize=16, center_learning_rate=0.1, stdev_learning_rate=0.1, stdev_init=0.5, distributed=True)
+    s.run(3)
+    assert s.status["iter"] == 3
+    assert torch.isfinite(torch.as_tensor(float(s.status["mean_eval"])))
