@@ -1,6 +1,7 @@
 """Process-pool evaluation (`num_actors`): the reference's Ray actor
 system rebuilt on multiprocessing + cloudpickle (parallel/evalpool.py)."""
 
+import pytest
 import torch
 
 from evotorch_amd import Problem
@@ -145,5 +146,28 @@ def test_pool_parallelizes_distributed_gradient_evaluation():
         assert s.status["iter"] == 2
         assert prob._eval_pool is not None  # the pool actually engaged
         assert prob._total_interactions > 0
+    finally:
+        prob.kill_actors()
+
+
+def test_pool_survives_worker_exception():
+    """A fitness error inside a worker propagates to the caller, and the
+    pool stays usable for the next evaluation."""
+    def touchy(x):
+        if float(x[0]) > 1e8:
+            raise ValueError("boom")
+        return float((x**2).sum())
+
+    prob = Problem("min", touchy, solution_length=3, initial_bounds=(-1, 1), seed=4, num_actors=2)
+    try:
+        b = prob.generate_batch(6)
+        prob.evaluate(b)  # normal values: fine
+        bad = prob.generate_batch(4)
+        bad.access_values()[0, 0] = 1e9
+        with pytest.raises(Exception):
+            prob.evaluate(bad)
+        ok = prob.generate_batch(6)
+        prob.evaluate(ok)  # pool still healthy
+        assert bool(torch.isfinite(torch.Tensor.as_subclass(ok.evals[:, 0], torch.Tensor)).all())
     finally:
         prob.kill_actors()
