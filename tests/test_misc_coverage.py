@@ -161,3 +161,32 @@ def test_nondefault_problem_dtypes(dt):
     s.run(3)
     assert s.status["iter"] == 3
     assert p.generate_batch(2).access_values().dtype == dt
+
+
+def test_testing_helpers_surface():
+    """The user-facing assertion helpers (reference testing.py:100-273)."""
+    from evotorch_amd.testing import (
+        TestingError,
+        assert_allclose,
+        assert_almost_between,
+        assert_dtype_matches,
+        assert_eachclose,
+        assert_shape_matches,
+    )
+
+    a = torch.tensor([1.0, 2.0, 3.0])
+    assert_allclose(a, [1.0, 2.0, 3.0], atol=1e-8)
+    with pytest.raises(TestingError):
+        assert_allclose(a, [1.0, 2.0, 4.0], atol=1e-3)
+    assert_almost_between(a, 0.5, 3.5)
+    with pytest.raises(TestingError):
+        assert_almost_between(a, 1.5, 3.5)
+    assert_dtype_matches(a, torch.float32)
+    with pytest.raises(TestingError):
+        assert_dtype_matches(a, torch.int64)
+    assert_shape_matches(torch.zeros(2, 3), (2, 3))
+    with pytest.raises(TestingError):
+        assert_shape_matches(torch.zeros(2, 3), (3, 2))
+    assert_eachclose(torch.full((4,), 7.0), 7.0)
+    with pytest.raises(TestingError):
+        assert_eachclose(torch.tensor([7.0, 8.0]), 7.0)
