@@ -479,6 +479,9 @@ static void launch_rollout(int n_blocks, int block, size_t lds_bytes, hipStream_
 void rollout_v7(torch::Tensor params, torch::Tensor env_blob, torch::Tensor obs_stats_out, torch::Tensor fitness,
                 int64_t obs_dim, int64_t act_dim, int64_t rank, int64_t steps, double alive_bonus, double act_cost,
                 int64_t init_seed, int64_t member_offset, const unsigned long long* seed_ptr);  // rollout_v7.hip
+void rollout_m7(torch::Tensor params, torch::Tensor env_blob, torch::Tensor obs_stats_out, torch::Tensor fitness,
+                int64_t obs_dim, int64_t act_dim, int64_t rank, int64_t steps, double alive_bonus, double act_cost,
+                int64_t init_seed, int64_t member_offset, const unsigned long long* seed_ptr);  // rollout_v7.hip
 
 torch::Tensor rollout_linear(torch::Tensor params, torch::Tensor env_blob, torch::Tensor obs_stats_out,
                              int64_t obs_dim, int64_t act_dim, int64_t rank, int64_t steps, double alive_bonus,
@@ -502,6 +505,14 @@ torch::Tensor rollout_linear(torch::Tensor params, torch::Tensor env_blob, torch
     auto fitness = torch::empty({n}, params.options());
     // v7 (MFMA, 16 members/block) serves the linear flagship geometry;
     // v6 covers MLP policies and off-geometry envs.
+    if (H == 64 && R == 16 && O == 376 && A == 17 && getenv("EVOTORCH_AMD_ROLLOUT_M7")) {
+        const int n_pblocks = (n + 3) / 4;
+        auto stat_partials = torch::zeros({(int64_t)n_pblocks, 2 * (int64_t)O}, params.options());
+        rollout_m7(params, env_blob, stat_partials, fitness, obs_dim, act_dim, rank, steps, alive_bonus, act_cost,
+                   init_seed, member_offset, seed_ptr);
+        obs_stats_out.add_(stat_partials.sum(0));
+        return fitness;
+    }
     if (H == 0 && R == 16 && O == 376 && A == 17 && !getenv("EVOTORCH_AMD_ROLLOUT_V6")) {
         const int n_blocks7 = (n + 15) / 16;
         auto stat_partials = torch::zeros({(int64_t)n_blocks7, 2 * (int64_t)O}, params.options());

@@ -396,6 +396,319 @@ __global__ __launch_bounds__(64 * kWaves, 1) void rollout_v7_kernel(RolloutV7Arg
     }
 }
 
+
+// ===========================================================================
+// m7: MLP-64 rollout — 2-wave TEAMS per member, W1 register-resident.
+//
+// v6 runs MLP members with 8-lane group dots + block barriers and
+// measures ~10:1 WAIT:BUSY (36 ms/gen at the reference MLP-64 config).
+// Here each member is owned by a PAIR of waves: wave half h of the pair
+// holds hidden rows [32h, 32h+32) of W1 in REGISTERS (32 rows x 6
+// columns per lane = 96 VGPRs — a full per-member W1 cannot fit one
+// wave, but half can), the observation lives in 3 bf16x2 registers per
+// lane, and the layer-2 weights sit k-sliced per lane (17 bf16 = 9
+// VGPRs). Per step: 32 col-sliced hidden dots per wave (DPP-reduced),
+// one LDS exchange of the quantized h1 halves, the 17 action dots
+// k-sliced over lanes, and the v8-style per-lane-column dynamics update
+// with [U;D2] k-pairs from LDS. Two block barriers per step; all four
+// members of a block run the same schedule, so barriers stay balanced.
+// Numerics contract identical to v6/rollout_eager (bf16 operands, fp32
+// accumulation, quantize-then-normalize, tanh on hidden and output).
+// Env-gated (EVOTORCH_AMD_ROLLOUT_M7) until it beats v6 end to end.
+// ===========================================================================
+
+template <int O, int A, int H>
+__global__ __launch_bounds__(512, 1) void rollout_m7_kernel(RolloutV7Args args) {
+    constexpr int kM = 4;                        // members per block (2 waves each)
+    constexpr int OP = (O + 127) / 128 * 128;    // 384 padded cols
+    constexpr int kPairs = OP / 2 / 64;          // 3 bf16x2 col-pairs per lane
+    constexpr int kCols = 2 * kPairs;            // 6 cols per lane
+    constexpr int HH = H / 2;                    // hidden rows per wave (32)
+    constexpr int R = 16;
+    constexpr int K = R + A;                     // 33 dynamics rows
+    constexpr int KQ = (K + 1) / 2;              // 17 k-pairs
+    constexpr int VS = OP + 8;
+    constexpr int US = OP + 8;
+
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wave = tid >> 6;
+    const int member_slot = wave >> 1;           // 0..3 within the block
+    const int half = wave & 1;                   // which half of W1 this wave owns
+    const int member = blockIdx.x * kM + member_slot;
+    const bool live = member < args.n_members;
+    const int mem_clamped = live ? member : 0;
+
+    extern __shared__ unsigned char ldsm[];
+    __bf16* v_l = reinterpret_cast<__bf16*>(ldsm);               // [R][VS]
+    bf16x2_t* ud_l = reinterpret_cast<bf16x2_t*>(v_l + R * VS);  // [KQ][US]
+    float* mean_l = reinterpret_cast<float*>(ud_l + KQ * US);    // [OP]
+    float* istd_l = mean_l + OP;                                 // [OP]
+    float* c_l = istd_l + OP;                                    // [OP]
+    float* wr_l = c_l + OP;                                      // [OP]
+    __bf16* h1_l = reinterpret_cast<__bf16*>(wr_l + OP);         // [kM][H] quantized hidden
+    float* act_l = reinterpret_cast<float*>(h1_l + kM * H);      // [kM][A + 1] actions (pad)
+    float* sstat = act_l + kM * (A + 1);                         // [2][kM][OP] end only
+
+    const float* e_V = args.env_blob;
+    const float* e_M = args.env_blob + (size_t)R * O;
+    const float* e_c = args.env_blob + (size_t)(2 * R + A) * O;
+    const float* e_wr = e_c + O;
+    const float* e_mean = e_wr + O;
+    const float* e_std = e_mean + O;
+
+    for (int j = tid; j < R * VS; j += 512) {
+        const int r = j / VS, cc = j % VS;
+        v_l[j] = (cc < O) ? f2b7(e_V[(size_t)r * O + cc]) : f2b7(0.0f);
+    }
+    for (int j = tid; j < KQ * US; j += 512) {
+        const int q = j / US, cc = j % US;
+        bf16x2_t m;
+        m.x = (cc < O) ? f2b7(e_M[(size_t)(2 * q) * O + cc]) : f2b7(0.0f);
+        m.y = (cc < O && 2 * q + 1 < K) ? f2b7(e_M[(size_t)(2 * q + 1) * O + cc]) : f2b7(0.0f);
+        ud_l[j] = m;
+    }
+    for (int j = tid; j < OP; j += 512) {
+        const bool in = j < O;
+        mean_l[j] = in ? e_mean[j] : 0.0f;
+        istd_l[j] = in ? 1.0f / e_std[j] : 0.0f;
+        c_l[j] = in ? e_c[j] : 0.0f;
+        wr_l[j] = in ? e_wr[j] : 0.0f;
+    }
+
+    // ---- per-member weights ----
+    // params layout (synthetic_env.py): W1[H][O] b1[H] W2[A][H] b2[A]
+    const long plen = (long)H * O + H + (long)A * H + A;
+    const float* P = args.params + (long)mem_clamped * plen;
+    bf16x2_t w1[HH][kPairs];  // my half's rows x my col slice
+#pragma unroll
+    for (int r = 0; r < HH; ++r) {
+#pragma unroll
+        for (int p = 0; p < kPairs; ++p) {
+            const int c0 = 2 * (p * 64 + lane);
+            const long row = (long)(half * HH + r) * O;
+            bf16x2_t w;
+            w.x = (c0 < O) ? f2b7(P[row + c0]) : f2b7(0.0f);
+            w.y = (c0 + 1 < O) ? f2b7(P[row + c0 + 1]) : f2b7(0.0f);
+            w1[r][p] = w;
+        }
+    }
+    float b1[HH];
+#pragma unroll
+    for (int r = 0; r < HH; ++r)
+        b1[r] = __builtin_bit_cast(float, __builtin_amdgcn_readfirstlane(__builtin_bit_cast(int, P[(long)H * O + half * HH + r])));
+    // W2 k-sliced: lane l holds W2[a][l] for all a (h1 index = lane)
+    __bf16 w2[A];
+#pragma unroll
+    for (int a = 0; a < A; ++a) w2[a] = (lane < H) ? f2b7(P[(long)H * O + H + (long)a * H + lane]) : f2b7(0.0f);
+    float b2[A];
+#pragma unroll
+    for (int a = 0; a < A; ++a)
+        b2[a] = __builtin_bit_cast(float, __builtin_amdgcn_readfirstlane(__builtin_bit_cast(int, P[(long)H * O + H + (long)A * H + a])));
+
+    float ssum[kCols] = {}, ssq[kCols] = {};
+    float fit_acc = 0.0f;
+    float asq_total = 0.0f;
+    __syncthreads();
+
+    // ---- initial observation ----
+    const unsigned long long iseed = args.seed_ptr ? *args.seed_ptr : args.init_seed;
+    const int odd = lane & 1;
+    bf16x2_t obs2[kPairs], obsn2[kPairs];
+#pragma unroll
+    for (int p = 0; p < kPairs; ++p) {
+        const int pi = p * 64 + lane;
+        float z[4];
+        philox_normal4(iseed, (uint32_t)(args.member_offset + member), (uint64_t)(pi >> 1), z);
+        bf16x2_t o, onr;
+#pragma unroll
+        for (int e = 0; e < 2; ++e) {
+            const int col = 2 * pi + e;
+            const float zv = odd ? z[e + 2] : z[e];
+            const __bf16 ob = (col < O) ? f2b7(0.1f * zv) : f2b7(0.0f);
+            const float onf = (b2f7(ob) - mean_l[col < OP ? col : 0]) * istd_l[col < OP ? col : 0];
+            if (e == 0) { o.x = ob; onr.x = f2b7(onf); } else { o.y = ob; onr.y = f2b7(onf); }
+        }
+        obs2[p] = o;
+        obsn2[p] = onr;
+    }
+
+    for (int t = 0; t < args.steps; ++t) {
+        // ---- hidden half: 32 col-sliced dots, DPP-reduced, quantized to LDS ----
+        {
+            float hacc[HH];
+#pragma unroll
+            for (int r = 0; r < HH; ++r) {
+                float acc = 0.0f;
+#pragma unroll
+                for (int p = 0; p < kPairs; ++p) acc = __builtin_amdgcn_fdot2_f32_bf16(w1[r][p], obsn2[p], acc, false);
+                hacc[r] = acc;
+            }
+#pragma unroll
+            for (int r = 0; r < HH; ++r) {
+                float s = reduce32_dpp(hacc[r]);
+                const float tot =
+                    __builtin_bit_cast(float, __builtin_amdgcn_readlane(__builtin_bit_cast(int, s), 31)) +
+                    __builtin_bit_cast(float, __builtin_amdgcn_readlane(__builtin_bit_cast(int, s), 63));
+                if (lane == r)  // one writer per row: lane r stores row (half*HH + r)
+                    h1_l[member_slot * H + half * HH + r] = f2b7(tanh_fast(tot + b1[r]));
+            }
+        }
+        __syncthreads();
+        // ---- actions: h1 k-sliced per lane (h index = lane), 17 reductions;
+        // wave half 0 computes them and publishes to LDS ----
+        if (half == 0) {
+            const float h1v = (lane < H) ? b2f7(h1_l[member_slot * H + lane]) : 0.0f;
+#pragma unroll
+            for (int a = 0; a < A; ++a) {
+                float part = b2f7(w2[a]) * h1v;
+                float s = reduce32_dpp(part);
+                const float tot =
+                    __builtin_bit_cast(float, __builtin_amdgcn_readlane(__builtin_bit_cast(int, s), 31)) +
+                    __builtin_bit_cast(float, __builtin_amdgcn_readlane(__builtin_bit_cast(int, s), 63));
+                if (lane == a) {
+                    const float av = fminf(fmaxf(tot + b2[a], -1.0f), 1.0f);
+                    act_l[member_slot * (A + 1) + a] = av;
+                }
+            }
+        }
+        __syncthreads();
+        // ---- dynamics: h = V @ obs (16 dots, both waves share the work:
+        // wave half h does rows [8h, 8h+8)) then o' per lane-column ----
+        float hdyn[8];
+#pragma unroll
+        for (int rr = 0; rr < 8; ++rr) {
+            const int r = half * 8 + rr;
+            float acc = 0.0f;
+#pragma unroll
+            for (int p = 0; p < kPairs; ++p) {
+                const bf16x2_t v = *reinterpret_cast<const bf16x2_t*>(v_l + r * VS + 2 * (p * 64 + lane));
+                acc = __builtin_amdgcn_fdot2_f32_bf16(v, obs2[p], acc, false);
+            }
+            float s = reduce32_dpp(acc);
+            hdyn[rr] = __builtin_bit_cast(float, __builtin_amdgcn_readlane(__builtin_bit_cast(int, s), 31)) +
+                       __builtin_bit_cast(float, __builtin_amdgcn_readlane(__builtin_bit_cast(int, s), 63));
+        }
+        // publish my 8 dynamics values through act_l's tail? No: build hact
+        // pairs locally — each wave needs ALL 33 k values; exchange the h
+        // halves through LDS (reuse h1_l tail is occupied; use act_l? sizes
+        // differ). Simplest: a small LDS patch after act_l is sstat — safe
+        // to reuse DURING the loop since sstat is only written at wrap-up.
+        {
+            float* hx = sstat;  // [kM][16] scratch (fits well inside sstat)
+#pragma unroll
+            for (int rr = 0; rr < 8; ++rr)
+                if (lane == rr) hx[member_slot * 16 + half * 8 + rr] = hdyn[rr];  // no runtime index
+        }
+        __syncthreads();
+        {
+            bf16x2_t hact2[KQ];
+            const float* hx = sstat;
+            const float aq_sum_unused = 0.0f;
+            (void)aq_sum_unused;
+#pragma unroll
+            for (int q = 0; q < KQ; ++q) {
+                float kv[2];
+#pragma unroll
+                for (int e = 0; e < 2; ++e) {
+                    const int k = 2 * q + e;
+                    if (k >= K) { kv[e] = 0.0f; continue; }
+                    kv[e] = (k < R) ? hx[member_slot * 16 + k] : act_l[member_slot * (A + 1) + (k - R)];
+                }
+                bf16x2_t hp;
+                hp.x = f2b7(kv[0]);
+                hp.y = f2b7(kv[1]);
+                hact2[q] = hp;
+            }
+#pragma unroll
+            for (int p = 0; p < kPairs; ++p) {
+                const int col0 = 2 * (p * 64 + lane);
+                bf16x2_t o, onr;
+#pragma unroll
+                for (int e = 0; e < 2; ++e) {
+                    const int j = 2 * p + e;
+                    float acc = 0.0f;
+#pragma unroll
+                    for (int q = 0; q < KQ; ++q)
+                        acc = __builtin_amdgcn_fdot2_f32_bf16(ud_l[(size_t)q * US + col0 + e], hact2[q], acc, false);
+                    const float o_new = tanh_fast(acc + c_l[col0 + e]);
+                    if (half == 0) {  // one wave of the pair owns fitness/stats
+                        fit_acc = fmaf(wr_l[col0 + e], o_new, fit_acc);
+                        ssum[j] += o_new;
+                        ssq[j] = fmaf(o_new, o_new, ssq[j]);
+                    }
+                    const __bf16 ob = f2b7(o_new);
+                    const float onf = (b2f7(ob) - mean_l[col0 + e]) * istd_l[col0 + e];
+                    if (e == 0) { o.x = ob; onr.x = f2b7(onf); } else { o.y = ob; onr.y = f2b7(onf); }
+                }
+                obs2[p] = o;
+                obsn2[p] = onr;
+            }
+            if (half == 0 && lane < A) {
+                const float av = act_l[member_slot * (A + 1) + lane];
+                asq_total = fmaf(av, av, asq_total);
+            }
+        }
+        __syncthreads();  // hx scratch / act_l reuse next step
+    }
+
+    // ---- fitness wrap-up (wave half 0 of each member) ----
+    if (half == 0) {
+        float f = reduce32_dpp(fit_acc);
+        float total = __builtin_bit_cast(float, __builtin_amdgcn_readlane(__builtin_bit_cast(int, f), 31)) +
+                      __builtin_bit_cast(float, __builtin_amdgcn_readlane(__builtin_bit_cast(int, f), 63));
+        float aq = reduce32_dpp(asq_total);
+        float aq_total = __builtin_bit_cast(float, __builtin_amdgcn_readlane(__builtin_bit_cast(int, aq), 31)) +
+                         __builtin_bit_cast(float, __builtin_amdgcn_readlane(__builtin_bit_cast(int, aq), 63));
+        if (lane == 0 && live)
+            args.fitness_out[member] =
+                total + args.alive_bonus * (float)args.steps - args.act_cost * aq_total / (float)A;
+    }
+    __syncthreads();
+    // ---- block stat partial (sstat was scratch during the loop; rebuilt now) ----
+    if (half == 0) {  // one writer per member slot (the stats-owning wave)
+#pragma unroll
+        for (int p = 0; p < kPairs; ++p) {
+#pragma unroll
+            for (int e = 0; e < 2; ++e) {
+                const int col = 2 * (p * 64 + lane) + e;
+                if (col < OP) {
+                    sstat[(size_t)member_slot * OP + col] = live ? ssum[2 * p + e] : 0.0f;
+                    sstat[(size_t)(kM + member_slot) * OP + col] = live ? ssq[2 * p + e] : 0.0f;
+                }
+            }
+        }
+    }
+    __syncthreads();
+    float* stats = args.obs_stats_out + (int64_t)blockIdx.x * 2 * O;
+    for (int col = tid; col < O; col += 512) {
+        float s = 0.0f, q = 0.0f;
+#pragma unroll
+        for (int w = 0; w < kM; ++w) {
+            s += sstat[(size_t)w * OP + col];
+            q += sstat[(size_t)(kM + w) * OP + col];
+        }
+        stats[col] = s;
+        stats[O + col] = q;
+    }
+}
+
+template <int O_T, int A_T, int H_T>
+static void launch_m7(const RolloutV7Args& args, int n, hipStream_t stream) {
+    constexpr int OP = (O_T + 127) / 128 * 128;
+    constexpr int KQ = (16 + A_T + 1) / 2;
+    const size_t lds = (size_t)16 * (OP + 8) * 2 + (size_t)KQ * (OP + 8) * 4 + (size_t)4 * OP * 4 +
+                       (size_t)4 * H_T * 2 + (size_t)4 * (A_T + 1) * 4 + (size_t)2 * 4 * OP * 4;
+    static bool attr_m7 = false;
+    if (!attr_m7) {
+        (void)hipFuncSetAttribute(reinterpret_cast<const void*>(&rollout_m7_kernel<O_T, A_T, H_T>),
+                                  hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
+        attr_m7 = true;
+    }
+    const int blocks = (n + 3) / 4;
+    hipLaunchKernelGGL((rollout_m7_kernel<O_T, A_T, H_T>), dim3(blocks), dim3(512), lds, stream, args);
+}
+
 template <int O_T, int A_T, int kWaves>
 static void launch_v7(const RolloutV7Args& args, int n, hipStream_t stream) {
     constexpr int OP = (O_T + 127) / 128 * 128;
@@ -445,6 +758,31 @@ void rollout_v7(torch::Tensor params, torch::Tensor env_blob, torch::Tensor obs_
     } else {
         launch_v7<O_T, A_T, 8>(args, n, stream);
     }
+}
+
+
+void rollout_m7(torch::Tensor params, torch::Tensor env_blob, torch::Tensor obs_stats_out, torch::Tensor fitness,
+                int64_t obs_dim, int64_t act_dim, int64_t rank, int64_t steps, double alive_bonus, double act_cost,
+                int64_t init_seed, int64_t member_offset, const unsigned long long* seed_ptr) {
+    const int n = (int)params.size(0);
+    TORCH_CHECK((int)rank == 16 && (int)obs_dim == 376 && (int)act_dim == 17,
+                "rollout m7 is instantiated for the Humanoid geometry");
+    RolloutV7Args args;
+    args.params = params.data_ptr<float>();
+    args.env_blob = env_blob.data_ptr<float>();
+    args.fitness_out = fitness.data_ptr<float>();
+    args.obs_stats_out = obs_stats_out.data_ptr<float>();
+    args.n_members = n;
+    args.member_offset = (long)member_offset;
+    args.obs_dim = (int)obs_dim; args.act_dim = (int)act_dim; args.rank = (int)rank;
+    args.steps = (int)steps;
+    args.alive_bonus = (float)alive_bonus;
+    args.act_cost = (float)act_cost;
+    args.init_seed = (unsigned long long)init_seed;
+    args.seed_ptr = seed_ptr;
+    args.skip_mask = 0;
+    auto stream = at::cuda::getCurrentCUDAStream();
+    launch_m7<376, 17, 64>(args, n, stream);
 }
 
 }  // namespace ea
