@@ -505,8 +505,10 @@ torch::Tensor rollout_linear(torch::Tensor params, torch::Tensor env_blob, torch
     auto fitness = torch::empty({n}, params.options());
     // v7 (MFMA, 16 members/block) serves the linear flagship geometry;
     // v6 covers MLP policies and off-geometry envs.
-    if (H == 64 && R == 16 && O == 376 && A == 17 && getenv("EVOTORCH_AMD_ROLLOUT_M7")) {
-        const int n_pblocks = (n + 3) / 4;
+    if (H == 64 && R == 16 && O == 376 && A == 17 && !getenv("EVOTORCH_AMD_ROLLOUT_V6")) {
+        const char* km_env = getenv("EVOTORCH_AMD_M7_MEMBERS");
+        const int m7_members = (km_env && atoi(km_env) == 4) ? 4 : 2;
+        const int n_pblocks = (n + m7_members - 1) / m7_members;
         auto stat_partials = torch::zeros({(int64_t)n_pblocks, 2 * (int64_t)O}, params.options());
         rollout_m7(params, env_blob, stat_partials, fitness, obs_dim, act_dim, rank, steps, alive_bonus, act_cost,
                    init_seed, member_offset, seed_ptr);

@@ -414,12 +414,15 @@ __global__ __launch_bounds__(64 * kWaves, 1) void rollout_v7_kernel(RolloutV7Arg
 // members of a block run the same schedule, so barriers stay balanced.
 // Numerics contract identical to v6/rollout_eager (bf16 operands, fp32
 // accumulation, quantize-then-normalize, tanh on hidden and output).
-// Env-gated (EVOTORCH_AMD_ROLLOUT_M7) until it beats v6 end to end.
+// Default for the MLP-64 flagship geometry (29.4 vs v6's 36.4 ms at
+// T=1000 popsize 4000, 1.24x); EVOTORCH_AMD_ROLLOUT_V6 forces v6, and
+// EVOTORCH_AMD_M7_MEMBERS=4 selects the single-block 4-member variant
+// (measured slower: one 8-wave block per CU exposes the barrier chain).
 // ===========================================================================
 
-template <int O, int A, int H>
-__global__ __launch_bounds__(512, 1) void rollout_m7_kernel(RolloutV7Args args) {
-    constexpr int kM = 4;                        // members per block (2 waves each)
+template <int O, int A, int H, int kM>
+__global__ __launch_bounds__(kM * 128, (kM == 2) ? 2 : 1) void rollout_m7_kernel(RolloutV7Args args) {
+    constexpr int kThreads = kM * 128;           // 2 waves per member
     constexpr int OP = (O + 127) / 128 * 128;    // 384 padded cols
     constexpr int kPairs = OP / 2 / 64;          // 3 bf16x2 col-pairs per lane
     constexpr int kCols = 2 * kPairs;            // 6 cols per lane
@@ -457,18 +460,18 @@ __global__ __launch_bounds__(512, 1) void rollout_m7_kernel(RolloutV7Args args) 
     const float* e_mean = e_wr + O;
     const float* e_std = e_mean + O;
 
-    for (int j = tid; j < R * VS; j += 512) {
+    for (int j = tid; j < R * VS; j += kThreads) {
         const int r = j / VS, cc = j % VS;
         v_l[j] = (cc < O) ? f2b7(e_V[(size_t)r * O + cc]) : f2b7(0.0f);
     }
-    for (int j = tid; j < KQ * US; j += 512) {
+    for (int j = tid; j < KQ * US; j += kThreads) {
         const int q = j / US, cc = j % US;
         bf16x2_t m;
         m.x = (cc < O) ? f2b7(e_M[(size_t)(2 * q) * O + cc]) : f2b7(0.0f);
         m.y = (cc < O && 2 * q + 1 < K) ? f2b7(e_M[(size_t)(2 * q + 1) * O + cc]) : f2b7(0.0f);
         ud_l[j] = m;
     }
-    for (int j = tid; j < OP; j += 512) {
+    for (int j = tid; j < OP; j += kThreads) {
         const bool in = j < O;
         mean_l[j] = in ? e_mean[j] : 0.0f;
         istd_l[j] = in ? 1.0f / e_std[j] : 0.0f;
@@ -536,69 +539,56 @@ __global__ __launch_bounds__(512, 1) void rollout_m7_kernel(RolloutV7Args args) 
     for (int t = 0; t < args.steps; ++t) {
         // ---- hidden half: 32 col-sliced dots, DPP-reduced, quantized to LDS ----
         {
-            float hacc[HH];
+            // fused dot+reduce per row: row r+1's dot2s issue underneath
+            // row r's DPP chain, and no 32-deep accumulator array stays
+            // live (the separated two-loop form held 32 extra VGPRs).
+            // v_readlane scalarization measured as the dominant stall
+            // (SALU round-trips); one cross-half shuffle puts the 64-lane
+            // total in lane 31, which writes directly.
 #pragma unroll
             for (int r = 0; r < HH; ++r) {
                 float acc = 0.0f;
 #pragma unroll
                 for (int p = 0; p < kPairs; ++p) acc = __builtin_amdgcn_fdot2_f32_bf16(w1[r][p], obsn2[p], acc, false);
-                hacc[r] = acc;
+                float s = reduce32_dpp(acc);
+                s += __shfl_xor(s, 32, 64);
+                if (lane == 31)
+                    h1_l[member_slot * H + half * HH + r] = f2b7(tanh_fast(s + b1[r]));
             }
+        }
+        // ---- dynamics h = V @ obs is independent of h1: same phase, no
+        // extra barrier (published into the sstat-scratch h-exchange) ----
+        {
+            float* hx = sstat;  // [kM][16] scratch; sstat proper is wrap-up-only
 #pragma unroll
-            for (int r = 0; r < HH; ++r) {
-                float s = reduce32_dpp(hacc[r]);
-                const float tot =
-                    __builtin_bit_cast(float, __builtin_amdgcn_readlane(__builtin_bit_cast(int, s), 31)) +
-                    __builtin_bit_cast(float, __builtin_amdgcn_readlane(__builtin_bit_cast(int, s), 63));
-                if (lane == r)  // one writer per row: lane r stores row (half*HH + r)
-                    h1_l[member_slot * H + half * HH + r] = f2b7(tanh_fast(tot + b1[r]));
+            for (int rr = 0; rr < 8; ++rr) {
+                const int r = half * 8 + rr;
+                float acc = 0.0f;
+#pragma unroll
+                for (int p = 0; p < kPairs; ++p) {
+                    const bf16x2_t v = *reinterpret_cast<const bf16x2_t*>(v_l + r * VS + 2 * (p * 64 + lane));
+                    acc = __builtin_amdgcn_fdot2_f32_bf16(v, obs2[p], acc, false);
+                }
+                float s = reduce32_dpp(acc);
+                s += __shfl_xor(s, 32, 64);
+                if (lane == 31) hx[member_slot * 16 + half * 8 + rr] = s;
             }
         }
         __syncthreads();
-        // ---- actions: h1 k-sliced per lane (h index = lane), 17 reductions;
-        // wave half 0 computes them and publishes to LDS ----
-        if (half == 0) {
+        // ---- actions: h1 k-sliced per lane (h index = lane); the 17
+        // reductions split across BOTH halves of the member's wave pair ----
+        {
             const float h1v = (lane < H) ? b2f7(h1_l[member_slot * H + lane]) : 0.0f;
+            constexpr int kA0 = (A + 1) / 2;  // half 0: a < kA0; half 1: the rest
 #pragma unroll
             for (int a = 0; a < A; ++a) {
+                if ((half == 0) != (a < kA0)) continue;
                 float part = b2f7(w2[a]) * h1v;
                 float s = reduce32_dpp(part);
-                const float tot =
-                    __builtin_bit_cast(float, __builtin_amdgcn_readlane(__builtin_bit_cast(int, s), 31)) +
-                    __builtin_bit_cast(float, __builtin_amdgcn_readlane(__builtin_bit_cast(int, s), 63));
-                if (lane == a) {
-                    const float av = fminf(fmaxf(tot + b2[a], -1.0f), 1.0f);
-                    act_l[member_slot * (A + 1) + a] = av;
-                }
+                s += __shfl_xor(s, 32, 64);
+                if (lane == 31)
+                    act_l[member_slot * (A + 1) + a] = fminf(fmaxf(s + b2[a], -1.0f), 1.0f);
             }
-        }
-        __syncthreads();
-        // ---- dynamics: h = V @ obs (16 dots, both waves share the work:
-        // wave half h does rows [8h, 8h+8)) then o' per lane-column ----
-        float hdyn[8];
-#pragma unroll
-        for (int rr = 0; rr < 8; ++rr) {
-            const int r = half * 8 + rr;
-            float acc = 0.0f;
-#pragma unroll
-            for (int p = 0; p < kPairs; ++p) {
-                const bf16x2_t v = *reinterpret_cast<const bf16x2_t*>(v_l + r * VS + 2 * (p * 64 + lane));
-                acc = __builtin_amdgcn_fdot2_f32_bf16(v, obs2[p], acc, false);
-            }
-            float s = reduce32_dpp(acc);
-            hdyn[rr] = __builtin_bit_cast(float, __builtin_amdgcn_readlane(__builtin_bit_cast(int, s), 31)) +
-                       __builtin_bit_cast(float, __builtin_amdgcn_readlane(__builtin_bit_cast(int, s), 63));
-        }
-        // publish my 8 dynamics values through act_l's tail? No: build hact
-        // pairs locally — each wave needs ALL 33 k values; exchange the h
-        // halves through LDS (reuse h1_l tail is occupied; use act_l? sizes
-        // differ). Simplest: a small LDS patch after act_l is sstat — safe
-        // to reuse DURING the loop since sstat is only written at wrap-up.
-        {
-            float* hx = sstat;  // [kM][16] scratch (fits well inside sstat)
-#pragma unroll
-            for (int rr = 0; rr < 8; ++rr)
-                if (lane == rr) hx[member_slot * 16 + half * 8 + rr] = hdyn[rr];  // no runtime index
         }
         __syncthreads();
         {
@@ -681,7 +671,7 @@ __global__ __launch_bounds__(512, 1) void rollout_m7_kernel(RolloutV7Args args) 
     }
     __syncthreads();
     float* stats = args.obs_stats_out + (int64_t)blockIdx.x * 2 * O;
-    for (int col = tid; col < O; col += 512) {
+    for (int col = tid; col < O; col += kThreads) {
         float s = 0.0f, q = 0.0f;
 #pragma unroll
         for (int w = 0; w < kM; ++w) {
@@ -693,20 +683,21 @@ __global__ __launch_bounds__(512, 1) void rollout_m7_kernel(RolloutV7Args args) 
     }
 }
 
-template <int O_T, int A_T, int H_T>
+template <int O_T, int A_T, int H_T, int kM_T>
 static void launch_m7(const RolloutV7Args& args, int n, hipStream_t stream) {
     constexpr int OP = (O_T + 127) / 128 * 128;
     constexpr int KQ = (16 + A_T + 1) / 2;
     const size_t lds = (size_t)16 * (OP + 8) * 2 + (size_t)KQ * (OP + 8) * 4 + (size_t)4 * OP * 4 +
-                       (size_t)4 * H_T * 2 + (size_t)4 * (A_T + 1) * 4 + (size_t)2 * 4 * OP * 4;
-    static bool attr_m7 = false;
-    if (!attr_m7) {
-        (void)hipFuncSetAttribute(reinterpret_cast<const void*>(&rollout_m7_kernel<O_T, A_T, H_T>),
+                       (size_t)kM_T * H_T * 2 + (size_t)kM_T * (A_T + 1) * 4 + (size_t)2 * kM_T * OP * 4;
+    static bool attr_m7[2] = {false, false};
+    const int slot = (kM_T == 2) ? 0 : 1;
+    if (!attr_m7[slot]) {
+        (void)hipFuncSetAttribute(reinterpret_cast<const void*>(&rollout_m7_kernel<O_T, A_T, H_T, kM_T>),
                                   hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
-        attr_m7 = true;
+        attr_m7[slot] = true;
     }
-    const int blocks = (n + 3) / 4;
-    hipLaunchKernelGGL((rollout_m7_kernel<O_T, A_T, H_T>), dim3(blocks), dim3(512), lds, stream, args);
+    const int blocks = (n + kM_T - 1) / kM_T;
+    hipLaunchKernelGGL((rollout_m7_kernel<O_T, A_T, H_T, kM_T>), dim3(blocks), dim3(kM_T * 128), lds, stream, args);
 }
 
 template <int O_T, int A_T, int kWaves>
@@ -782,7 +773,12 @@ void rollout_m7(torch::Tensor params, torch::Tensor env_blob, torch::Tensor obs_
     args.seed_ptr = seed_ptr;
     args.skip_mask = 0;
     auto stream = at::cuda::getCurrentCUDAStream();
-    launch_m7<376, 17, 64>(args, n, stream);
+    const char* km = getenv("EVOTORCH_AMD_M7_MEMBERS");
+    if (km && atoi(km) == 4) {
+        launch_m7<376, 17, 64, 4>(args, n, stream);
+    } else {
+        launch_m7<376, 17, 64, 2>(args, n, stream);  // 2 blocks/CU cover stalls
+    }
 }
 
 }  // namespace ea
