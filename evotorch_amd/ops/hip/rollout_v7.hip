@@ -449,9 +449,10 @@ __global__ __launch_bounds__(kM * 128, (kM == 2) ? 2 : 1) void rollout_m7_kernel
     float* istd_l = mean_l + OP;                                 // [OP]
     float* c_l = istd_l + OP;                                    // [OP]
     float* wr_l = c_l + OP;                                      // [OP]
-    __bf16* h1_l = reinterpret_cast<__bf16*>(wr_l + OP);         // [kM][H] quantized hidden
-    float* act_l = reinterpret_cast<float*>(h1_l + kM * H);      // [kM][A + 1] actions (pad)
-    float* sstat = act_l + kM * (A + 1);                         // [2][kM][OP] end only
+    __bf16* h1_l = reinterpret_cast<__bf16*>(wr_l + OP);         // [2][kM][H] hidden (step-parity banks)
+    float* act_l = reinterpret_cast<float*>(h1_l + 2 * kM * H);  // [2][kM][A + 1] actions
+    float* hx_l = act_l + 2 * kM * (A + 1);                      // [2][kM][16] dynamics exchange
+    float* sstat = hx_l + 2 * kM * 16;                           // [2][kM][OP] end only
 
     const float* e_V = args.env_blob;
     const float* e_M = args.env_blob + (size_t)R * O;
@@ -537,6 +538,11 @@ __global__ __launch_bounds__(kM * 128, (kM == 2) ? 2 : 1) void rollout_m7_kernel
     }
 
     for (int t = 0; t < args.steps; ++t) {
+        // step-parity banks let the NEXT step's phase-A writes start
+        // without a loop-end barrier (2 barriers per step, not 3)
+        __bf16* h1b = h1_l + (t & 1) * kM * H;
+        float* actb = act_l + (t & 1) * kM * (A + 1);
+        float* hxb = hx_l + (t & 1) * kM * 16;
         // ---- hidden half: 32 col-sliced dots, DPP-reduced, quantized to LDS ----
         {
             // fused dot+reduce per row: row r+1's dot2s issue underneath
@@ -553,13 +559,13 @@ __global__ __launch_bounds__(kM * 128, (kM == 2) ? 2 : 1) void rollout_m7_kernel
                 float s = reduce32_dpp(acc);
                 s += __shfl_xor(s, 32, 64);
                 if (lane == 31)
-                    h1_l[member_slot * H + half * HH + r] = f2b7(tanh_fast(s + b1[r]));
+                    h1b[member_slot * H + half * HH + r] = f2b7(tanh_fast(s + b1[r]));
             }
         }
         // ---- dynamics h = V @ obs is independent of h1: same phase, no
         // extra barrier (published into the sstat-scratch h-exchange) ----
         {
-            float* hx = sstat;  // [kM][16] scratch; sstat proper is wrap-up-only
+            float* hx = hxb;
 #pragma unroll
             for (int rr = 0; rr < 8; ++rr) {
                 const int r = half * 8 + rr;
@@ -578,7 +584,7 @@ __global__ __launch_bounds__(kM * 128, (kM == 2) ? 2 : 1) void rollout_m7_kernel
         // ---- actions: h1 k-sliced per lane (h index = lane); the 17
         // reductions split across BOTH halves of the member's wave pair ----
         {
-            const float h1v = (lane < H) ? b2f7(h1_l[member_slot * H + lane]) : 0.0f;
+            const float h1v = (lane < H) ? b2f7(h1b[member_slot * H + lane]) : 0.0f;
             constexpr int kA0 = (A + 1) / 2;  // half 0: a < kA0; half 1: the rest
 #pragma unroll
             for (int a = 0; a < A; ++a) {
@@ -587,15 +593,13 @@ __global__ __launch_bounds__(kM * 128, (kM == 2) ? 2 : 1) void rollout_m7_kernel
                 float s = reduce32_dpp(part);
                 s += __shfl_xor(s, 32, 64);
                 if (lane == 31)
-                    act_l[member_slot * (A + 1) + a] = fminf(fmaxf(s + b2[a], -1.0f), 1.0f);
+                    actb[member_slot * (A + 1) + a] = fminf(fmaxf(s + b2[a], -1.0f), 1.0f);
             }
         }
         __syncthreads();
         {
             bf16x2_t hact2[KQ];
-            const float* hx = sstat;
-            const float aq_sum_unused = 0.0f;
-            (void)aq_sum_unused;
+            const float* hx = hxb;
 #pragma unroll
             for (int q = 0; q < KQ; ++q) {
                 float kv[2];
@@ -603,7 +607,7 @@ __global__ __launch_bounds__(kM * 128, (kM == 2) ? 2 : 1) void rollout_m7_kernel
                 for (int e = 0; e < 2; ++e) {
                     const int k = 2 * q + e;
                     if (k >= K) { kv[e] = 0.0f; continue; }
-                    kv[e] = (k < R) ? hx[member_slot * 16 + k] : act_l[member_slot * (A + 1) + (k - R)];
+                    kv[e] = (k < R) ? hx[member_slot * 16 + k] : actb[member_slot * (A + 1) + (k - R)];
                 }
                 bf16x2_t hp;
                 hp.x = f2b7(kv[0]);
@@ -635,12 +639,13 @@ __global__ __launch_bounds__(kM * 128, (kM == 2) ? 2 : 1) void rollout_m7_kernel
                 obsn2[p] = onr;
             }
             if (half == 0 && lane < A) {
-                const float av = act_l[member_slot * (A + 1) + lane];
+                const float av = actb[member_slot * (A + 1) + lane];
                 asq_total = fmaf(av, av, asq_total);
             }
         }
-        __syncthreads();  // hx scratch / act_l reuse next step
+        // no loop-end barrier: the next step writes the OTHER parity bank
     }
+    __syncthreads();
 
     // ---- fitness wrap-up (wave half 0 of each member) ----
     if (half == 0) {
@@ -688,7 +693,8 @@ static void launch_m7(const RolloutV7Args& args, int n, hipStream_t stream) {
     constexpr int OP = (O_T + 127) / 128 * 128;
     constexpr int KQ = (16 + A_T + 1) / 2;
     const size_t lds = (size_t)16 * (OP + 8) * 2 + (size_t)KQ * (OP + 8) * 4 + (size_t)4 * OP * 4 +
-                       (size_t)kM_T * H_T * 2 + (size_t)kM_T * (A_T + 1) * 4 + (size_t)2 * kM_T * OP * 4;
+                       (size_t)2 * kM_T * H_T * 2 + (size_t)2 * kM_T * (A_T + 1) * 4 +
+                       (size_t)2 * kM_T * 16 * 4 + (size_t)2 * kM_T * OP * 4;
     static bool attr_m7[2] = {false, false};
     const int slot = (kM_T == 2) ? 0 : 1;
     if (!attr_m7[slot]) {

@@ -34,5 +34,5 @@ for i in range(6):
                        spec.alive_bonus, spec.act_cost, 7 + i, 0, spec.policy_hidden)
 torch.cuda.synchronize()
 ms = (time.perf_counter() - t0) / 6 * 1000
-which = "m7" if os.environ.get("EVOTORCH_AMD_ROLLOUT_M7") else "v6"
+which = "v6" if os.environ.get("EVOTORCH_AMD_ROLLOUT_V6") else "m7"
 print(f"{which} MLP-64 rollout T=1000 popsize 4000: {ms:.2f} ms ({4000/ms*1000:,.0f} sol/s kernel-only)")
