@@ -127,3 +127,31 @@ def test_nondefault_dtypes_gpu(dt):
     s.run(3)
     assert s.status["iter"] == 3
     assert torch.isfinite(torch.as_tensor(float(s.status["mean_eval"])))
+
+
+@requires_gpu
+def test_m7_mlp_rollout_bitwise_deterministic():
+    """The m7 MLP kernel has no atomics and fixed-order reductions: two
+    identical launches must agree BITWISE (same contract as v7)."""
+    from evotorch_amd import ops
+    from evotorch_amd.neuroevolution.synthetic_env import SyntheticEnvSpec
+
+    mod = ops.hip_required()
+    spec = SyntheticEnvSpec(episode_length=100, device="cuda", policy_hidden=64)
+    torch.manual_seed(9)
+    params = 0.1 * torch.randn(37, spec.solution_length, device="cuda")
+    mean = torch.zeros(spec.obs_dim, device="cuda")
+    std = torch.ones(spec.obs_dim, device="cuda")
+    blob = spec.env_blob(mean, std, device="cuda")
+
+    def run():
+        stats = torch.zeros(2 * spec.obs_dim, device="cuda")
+        fit = mod.rollout_linear(params, blob, stats, spec.obs_dim, spec.act_dim, spec.rank,
+                                 spec.episode_length, spec.alive_bonus, spec.act_cost, 55, 0,
+                                 spec.policy_hidden)
+        return fit, stats
+
+    f1, s1 = run()
+    f2, s2 = run()
+    assert torch.equal(f1, f2)
+    assert torch.equal(s1, s2)
