@@ -443,3 +443,21 @@ def test_graphed_flagship_rollout():
     assert int(prob._graph_seed_buf.item()) != seed_before  # in-graph bump ran
     assert prob.obs_norm.count > count_before  # stats merged inside the graph
     assert float(g.mean_eval) > first + 20.0  # it optimizes
+
+
+@requires_gpu
+def test_graphed_mlp_flagship_rollout():
+    """GraphedSearch over the MLP-64 rollout problem (m7 kernel) with the
+    device episode-seed chain."""
+    from evotorch_amd.algorithms import PGPE, GraphedSearch
+    from evotorch_amd.neuroevolution import SyntheticRolloutProblem
+
+    prob = SyntheticRolloutProblem(device="cuda:0", seed=13, episode_length=40, policy_hidden=64)
+    r = 2.25
+    s = PGPE(prob, popsize=128, radius_init=r, center_learning_rate=0.75 * r / 15,
+             stdev_learning_rate=0.1, optimizer="clipup", optimizer_config={"max_speed": r / 15})
+    g = GraphedSearch(s, generations_per_capture=5)
+    g.capture()
+    first = float(g.mean_eval)
+    g.run(60)
+    assert float(g.mean_eval) > first + 10.0
