@@ -489,11 +489,17 @@ __global__ __launch_bounds__(kM * 128, (kM == 2) ? 2 : 1) void rollout_m7_kernel
     for (int r = 0; r < HH; ++r) {
 #pragma unroll
         for (int p = 0; p < kPairs; ++p) {
-            const int c0 = 2 * (p * 64 + lane);
+            // lane owns STRIDE-64 single columns (lane + 64*(2p+e)) — the
+            // two columns sharing a bf16x2 are 64 apart. fdot2 only needs
+            // the same pairing in both operands; adjacent-pair ownership
+            // made every GEMM2 [U;D2] read stride-2 per lane (measured
+            // 0.99 LDS conflict cycles per LDS instruction).
+            const int cx = lane + 64 * (2 * p);
+            const int cy = lane + 64 * (2 * p + 1);
             const long row = (long)(half * HH + r) * O;
             bf16x2_t w;
-            w.x = (c0 < O) ? f2b7(P[row + c0]) : f2b7(0.0f);
-            w.y = (c0 + 1 < O) ? f2b7(P[row + c0 + 1]) : f2b7(0.0f);
+            w.x = (cx < O) ? f2b7(P[row + cx]) : f2b7(0.0f);
+            w.y = (cy < O) ? f2b7(P[row + cy]) : f2b7(0.0f);
             w1[r][p] = w;
         }
     }
@@ -517,18 +523,19 @@ __global__ __launch_bounds__(kM * 128, (kM == 2) ? 2 : 1) void rollout_m7_kernel
 
     // ---- initial observation ----
     const unsigned long long iseed = args.seed_ptr ? *args.seed_ptr : args.init_seed;
-    const int odd = lane & 1;
     bf16x2_t obs2[kPairs], obsn2[kPairs];
 #pragma unroll
     for (int p = 0; p < kPairs; ++p) {
-        const int pi = p * 64 + lane;
-        float z[4];
-        philox_normal4(iseed, (uint32_t)(args.member_offset + member), (uint64_t)(pi >> 1), z);
         bf16x2_t o, onr;
 #pragma unroll
         for (int e = 0; e < 2; ++e) {
-            const int col = 2 * pi + e;
-            const float zv = odd ? z[e + 2] : z[e];
+            const int col = lane + 64 * (2 * p + e);
+            float z[4];
+            philox_normal4(iseed, (uint32_t)(args.member_offset + member), (uint64_t)(col >> 2), z);
+            // quad element = col & 3 = lane & 3 (64 ≡ 0 mod 4): branchless
+            const float z01 = (lane & 1) ? z[1] : z[0];
+            const float z23 = (lane & 1) ? z[3] : z[2];
+            const float zv = (lane & 2) ? z23 : z01;
             const __bf16 ob = (col < O) ? f2b7(0.1f * zv) : f2b7(0.0f);
             const float onf = (b2f7(ob) - mean_l[col < OP ? col : 0]) * istd_l[col < OP ? col : 0];
             if (e == 0) { o.x = ob; onr.x = f2b7(onf); } else { o.y = ob; onr.y = f2b7(onf); }
@@ -572,7 +579,9 @@ __global__ __launch_bounds__(kM * 128, (kM == 2) ? 2 : 1) void rollout_m7_kernel
                 float acc = 0.0f;
 #pragma unroll
                 for (int p = 0; p < kPairs; ++p) {
-                    const bf16x2_t v = *reinterpret_cast<const bf16x2_t*>(v_l + r * VS + 2 * (p * 64 + lane));
+                    bf16x2_t v;
+                    v.x = v_l[r * VS + lane + 64 * (2 * p)];
+                    v.y = v_l[r * VS + lane + 64 * (2 * p + 1)];
                     acc = __builtin_amdgcn_fdot2_f32_bf16(v, obs2[p], acc, false);
                 }
                 float s = reduce32_dpp(acc);
@@ -616,23 +625,23 @@ __global__ __launch_bounds__(kM * 128, (kM == 2) ? 2 : 1) void rollout_m7_kernel
             }
 #pragma unroll
             for (int p = 0; p < kPairs; ++p) {
-                const int col0 = 2 * (p * 64 + lane);
                 bf16x2_t o, onr;
 #pragma unroll
                 for (int e = 0; e < 2; ++e) {
+                    const int col = lane + 64 * (2 * p + e);
                     const int j = 2 * p + e;
                     float acc = 0.0f;
 #pragma unroll
                     for (int q = 0; q < KQ; ++q)
-                        acc = __builtin_amdgcn_fdot2_f32_bf16(ud_l[(size_t)q * US + col0 + e], hact2[q], acc, false);
-                    const float o_new = tanh_fast(acc + c_l[col0 + e]);
+                        acc = __builtin_amdgcn_fdot2_f32_bf16(ud_l[(size_t)q * US + col], hact2[q], acc, false);
+                    const float o_new = tanh_fast(acc + c_l[col]);
                     if (half == 0) {  // one wave of the pair owns fitness/stats
-                        fit_acc = fmaf(wr_l[col0 + e], o_new, fit_acc);
+                        fit_acc = fmaf(wr_l[col], o_new, fit_acc);
                         ssum[j] += o_new;
                         ssq[j] = fmaf(o_new, o_new, ssq[j]);
                     }
                     const __bf16 ob = f2b7(o_new);
-                    const float onf = (b2f7(ob) - mean_l[col0 + e]) * istd_l[col0 + e];
+                    const float onf = (b2f7(ob) - mean_l[col]) * istd_l[col];
                     if (e == 0) { o.x = ob; onr.x = f2b7(onf); } else { o.y = ob; onr.y = f2b7(onf); }
                 }
                 obs2[p] = o;
@@ -666,7 +675,7 @@ __global__ __launch_bounds__(kM * 128, (kM == 2) ? 2 : 1) void rollout_m7_kernel
         for (int p = 0; p < kPairs; ++p) {
 #pragma unroll
             for (int e = 0; e < 2; ++e) {
-                const int col = 2 * (p * 64 + lane) + e;
+                const int col = lane + 64 * (2 * p + e);
                 if (col < OP) {
                     sstat[(size_t)member_slot * OP + col] = live ? ssum[2 * p + e] : 0.0f;
                     sstat[(size_t)(kM + member_slot) * OP + col] = live ? ssq[2 * p + e] : 0.0f;
