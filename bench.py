@@ -33,8 +33,12 @@ import torch
 def parse_args():
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
-    p.add_argument("--steps", type=int, default=20)
-    p.add_argument("--warmup", type=int, default=3)
+    # defaults: on a GPU, enough steps that the timed region spans >=30 s
+    # (strong SMI/telemetry evidence) while still finishing in ~a minute;
+    # tiny on CPU where a generation costs seconds
+    on_gpu_default = torch.cuda.is_available()
+    p.add_argument("--steps", type=int, default=10000 if on_gpu_default else 4)
+    p.add_argument("--warmup", type=int, default=100 if on_gpu_default else 1)
     p.add_argument("--popsize-per-gpu", type=int, default=4000)
     p.add_argument("--episode-length", type=int, default=1000,
                    help="episode steps per rollout; 1000 matches the reference flagship config (BASELINE.md row 3)")
