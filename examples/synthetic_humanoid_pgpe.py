@@ -24,9 +24,12 @@ def main():
     p.add_argument("--device", default="cuda:0" if torch.cuda.is_available() else "cpu")
     p.add_argument("--generations", type=int, default=200)
     p.add_argument("--popsize", type=int, default=4000)
+    p.add_argument("--policy", choices=["linear", "mlp64"], default="linear",
+                   help="mlp64 = the reference paper's MLP-64-tanh brax architecture (m7 kernel)")
     args = p.parse_args()
 
-    problem = SyntheticRolloutProblem(device=args.device, seed=1, episode_length=200)
+    problem = SyntheticRolloutProblem(device=args.device, seed=1, episode_length=200,
+                                      policy_hidden=64 if args.policy == "mlp64" else 0)
     radius = 2.25
     max_speed = radius / 15.0
     searcher = PGPE(
