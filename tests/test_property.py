@@ -203,3 +203,53 @@ def test_operators_respect_bounds(seed):
         out = op(batch)
         vals = out.unsafe_values if hasattr(out, "unsafe_values") else out
         assert bool((vals >= -1.0 - 1e-6).all()) and bool((vals <= 1.0 + 1e-6).all()), type(op).__name__
+
+
+@settings(max_examples=25, deadline=None)
+@given(
+    rows=st.integers(min_value=2, max_value=16),
+    length=st.integers(min_value=1, max_value=24),
+    split=st.integers(min_value=1, max_value=15),
+    base=st.integers(min_value=0, max_value=10_000),
+    seed=st.integers(min_value=0, max_value=2**31 - 1),
+)
+def test_counter_addressed_sampling_partition_invariant(rows, length, split, base, seed):
+    """Counter-addressed sampling (K1's stream-per-row mode, the foundation
+    of rank sharding and the streaming two-pass gradient): regenerating ANY
+    row partition with the matching row_offset reproduces the full
+    population's rows exactly — on the CPU philox reference path here,
+    bitwise vs the HIP kernel in the GPU suite."""
+    from evotorch_amd import ops
+
+    split = min(split, rows - 1)
+    mu = torch.randn(length)
+    sigma = torch.rand(length) + 0.1
+
+    full = torch.empty(rows, length)
+    ops.sample_gaussian(full, mu, sigma, seed=seed, row_offset=base)
+
+    top = torch.empty(split, length)
+    bottom = torch.empty(rows - split, length)
+    ops.sample_gaussian(top, mu, sigma, seed=seed, row_offset=base)
+    ops.sample_gaussian(bottom, mu, sigma, seed=seed, row_offset=base + split)
+
+    assert torch.equal(full[:split], top)
+    assert torch.equal(full[split:], bottom)
+
+
+@settings(max_examples=15, deadline=None)
+@given(
+    pairs=st.integers(min_value=1, max_value=8),
+    length=st.integers(min_value=1, max_value=16),
+    seed=st.integers(min_value=0, max_value=2**31 - 1),
+)
+def test_counter_addressed_symmetric_mirror(pairs, length, seed):
+    """Symmetric counter-addressed sampling fills the halves layout with an
+    exact mirror: row i + pairs == 2*mu - row i."""
+    from evotorch_amd import ops
+
+    mu = torch.randn(length)
+    sigma = torch.rand(length) + 0.1
+    out = torch.empty(2 * pairs, length)
+    ops.sample_gaussian(out, mu, sigma, symmetric=True, seed=seed)
+    assert torch.allclose(out[pairs:], 2.0 * mu - out[:pairs], atol=1e-6)
