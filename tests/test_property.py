@@ -253,3 +253,49 @@ def test_counter_addressed_symmetric_mirror(pairs, length, seed):
     out = torch.empty(2 * pairs, length)
     ops.sample_gaussian(out, mu, sigma, symmetric=True, seed=seed)
     assert torch.allclose(out[pairs:], 2.0 * mu - out[:pairs], atol=1e-6)
+
+
+@settings(max_examples=20, deadline=None)
+@given(n=st.integers(min_value=2, max_value=64))
+def test_centered_ranking_sums_to_zero(n):
+    """Centered ranking (PGPE's default fitness shaping) is zero-sum, so the
+    mu-gradient is translation-invariant in the fitnesses."""
+    from evotorch_amd.utils.ranking import rank
+
+    fit = torch.randn(n)
+    w = rank(fit, "centered", higher_is_better=True)
+    assert abs(float(w.sum())) < 1e-5
+    shifted = rank(fit + 123.456, "centered", higher_is_better=True)
+    assert torch.allclose(w, shifted)
+
+
+@settings(max_examples=20, deadline=None)
+@given(
+    n=st.integers(min_value=2, max_value=40),
+    pieces=st.integers(min_value=1, max_value=6),
+    seed=st.integers(min_value=0, max_value=9999),
+)
+def test_solutionbatch_split_concat_roundtrip(n, pieces, seed):
+    from evotorch_amd.core import SolutionBatch
+
+    pieces = min(pieces, n)
+    prob = Problem("min", sphere, solution_length=5, initial_bounds=(-1, 1), seed=seed)
+    batch = prob.generate_batch(n)
+    prob.evaluate(batch)
+    rebuilt = SolutionBatch.cat(batch.split(pieces))
+    assert torch.equal(rebuilt.values.as_subclass(torch.Tensor), batch.values.as_subclass(torch.Tensor))
+    assert torch.equal(rebuilt.evals.as_subclass(torch.Tensor), batch.evals.as_subclass(torch.Tensor))
+
+
+@settings(max_examples=20, deadline=None)
+@given(seed=st.integers(min_value=0, max_value=9999))
+def test_modify_tensor_is_idempotent(seed):
+    """Applying the stdev-control clamp twice equals applying it once."""
+    from evotorch_amd.utils.misc import modify_tensor
+
+    g = torch.Generator().manual_seed(seed)
+    original = torch.rand(12, generator=g) + 0.5
+    target = original + torch.randn(12, generator=g)
+    once = modify_tensor(original, target, lb=0.1, ub=2.0, max_change=0.2)
+    twice = modify_tensor(original, once, lb=0.1, ub=2.0, max_change=0.2)
+    assert torch.allclose(once, twice)
